@@ -1,0 +1,291 @@
+"""SACEngine — the learner's update step for all variants.
+
+Implements the exact update math of the reference learners:
+
+- 'sac' / 'vsac' (LunarLander_Distributed_SAC/src/learner.py:203-239,
+  MT1_Distributed_VSAC/src/learner.py): LL-style actor, two separate
+  critics + two targets, scalar log_alpha, H_bar = -action_dim.
+- 'mtsac' (MT10_Distributed_MTSAC/src/learner.py:253-325): MT-style actor,
+  one twin-Q critic + target, per-task log_alpha vector, per-sample alpha
+  gather, optional task-weighted losses.
+- 'care' lives in :mod:`.care` (adds the context/state encoder machinery).
+
+MI355X-first mechanics under the reference math:
+
+- every optimizer group's params/grads live in ONE flat fp32 buffer
+  (:class:`~distributed_sac_amd.ops.flat.FlatParams`), so Adam is one fused
+  kernel, the Polyak target update is one kernel, and data-parallel gradient
+  all-reduce is one RCCL message per group;
+- forward/backward run through the fused HIP MLP + squashed-Gaussian ops;
+- the whole update is hipGraph-capturable (fixed shapes, graph-safe RNG) —
+  see :meth:`SACEngine.maybe_capture`.
+"""
+
+from __future__ import annotations
+
+import itertools
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+
+from ..config import SACConfig
+from ..models import Actor, Critic, LLActor, LLCritic
+from ..ops import functional as Fops
+from ..ops.flat import FlatParams, FusedAdam, flat_polyak_
+
+
+class SACEngine:
+    """Holds models/optimizers and performs one SAC gradient update."""
+
+    def __init__(self, cfg: SACConfig, device: torch.device | str = "cpu"):
+        self.cfg = cfg
+        self.variant = cfg.variant
+        self.device = torch.device(device)
+        self.gamma = cfg.gamma
+        self.tau = cfg.tau
+        self.reward_scale = cfg.reward_scale
+        self.use_weighted_loss = cfg.use_weighted_loss and cfg.variant == "mtsac"
+        self.num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
+        self.update_iteration = 0
+        self.total_step = 0
+        self._graph = None
+        self._eps_queue: Optional[list] = None  # test hook: deterministic eps
+        self._build_models()
+        self._build_optimizers()
+
+    def _next_eps(self, like: torch.Tensor) -> torch.Tensor:
+        if self._eps_queue:
+            return self._eps_queue.pop(0).to(like.device)
+        return torch.randn_like(like)
+
+    def _sample(self, states: torch.Tensor):
+        """Actor sampling with squashed-Gaussian op (both actor families)."""
+        mu, log_std_raw = self.actor(states)
+        eps = self._next_eps(mu)
+        return Fops.squashed_gaussian(mu, log_std_raw, eps, self.actor.k)
+
+    # ------------------------------------------------------------------
+    def _build_models(self) -> None:
+        cfg, dev = self.cfg, self.device
+        if self.variant in ("sac", "vsac"):
+            self.actor = LLActor(cfg.state_dim, cfg.action_dim,
+                                 cfg.actor_hidden_dim, cfg.action_bound).to(dev)
+            self.local_critic_1 = LLCritic(cfg.state_dim, cfg.action_dim,
+                                           cfg.critic_hidden_dim).to(dev)
+            self.local_critic_2 = LLCritic(cfg.state_dim, cfg.action_dim,
+                                           cfg.critic_hidden_dim).to(dev)
+            self.target_critic_1 = LLCritic(cfg.state_dim, cfg.action_dim,
+                                            cfg.critic_hidden_dim).to(dev)
+            self.target_critic_2 = LLCritic(cfg.state_dim, cfg.action_dim,
+                                            cfg.critic_hidden_dim).to(dev)
+            self.log_alpha = nn.Parameter(torch.tensor(
+                [float(cfg.log_alpha)], device=dev))
+        elif self.variant == "mtsac":
+            self.actor = Actor(cfg.state_dim, cfg.action_dim,
+                               cfg.actor_hidden_dim, cfg.action_bound,
+                               num_tasks=cfg.num_tasks).to(dev)
+            self.local_critic = Critic(cfg.state_dim, cfg.action_dim,
+                                       cfg.critic_hidden_dim,
+                                       num_tasks=cfg.num_tasks).to(dev)
+            self.target_critic = Critic(cfg.state_dim, cfg.action_dim,
+                                        cfg.critic_hidden_dim,
+                                        num_tasks=cfg.num_tasks).to(dev)
+            # per-task log_alpha (reference MT10…MTSAC/src/learner.py:115-121)
+            self.log_alpha = nn.Parameter(torch.full(
+                (cfg.num_tasks,), float(cfg.log_alpha), device=dev))
+        else:
+            raise ValueError(f"variant {self.variant} not handled here")
+        self.H_bar = torch.tensor([-float(cfg.action_dim)], device=dev)
+        self.alpha = self.log_alpha.exp().detach()
+
+    def _critic_params(self):
+        if self.variant in ("sac", "vsac"):
+            return itertools.chain(self.local_critic_1.parameters(),
+                                   self.local_critic_2.parameters())
+        return self.local_critic.parameters()
+
+    def _target_params(self):
+        if self.variant in ("sac", "vsac"):
+            return itertools.chain(self.target_critic_1.parameters(),
+                                   self.target_critic_2.parameters())
+        return self.target_critic.parameters()
+
+    def _build_optimizers(self) -> None:
+        cfg = self.cfg
+        self.actor_group = FlatParams(self.actor.parameters())
+        self.critic_group = FlatParams(self._critic_params())
+        self.target_group = FlatParams(self._target_params(), with_grad=False)
+        self.alpha_group = FlatParams([self.log_alpha])
+        self.actor_optimizer = FusedAdam(self.actor_group, lr=cfg.lr_actor)
+        self.critic_optimizer = FusedAdam(self.critic_group, lr=cfg.lr_critic)
+        # reference uses lr_actor for log_alpha (learner.py build_optimizer)
+        self.log_alpha_optimizer = FusedAdam(self.alpha_group, lr=cfg.lr_actor)
+        self.hard_copy_targets()
+
+    @torch.no_grad()
+    def hard_copy_targets(self) -> None:
+        """targets <- critics (reference soft_update tau=1.0 at run start)."""
+        flat_polyak_(self.target_group, self.critic_group, 1.0)
+
+    def zero_grad(self) -> None:
+        self.actor_group.zero_grad()
+        self.critic_group.zero_grad()
+        self.alpha_group.zero_grad()
+        # autograd may have detached .grad views (e.g. after state_dict load)
+        self.actor_group.rebind_grads()
+        self.critic_group.rebind_grads()
+        self.alpha_group.rebind_grads()
+
+    # ------------------------------------------------------------------
+    def _critic_q(self, states, actions):
+        if self.variant in ("sac", "vsac"):
+            return (self.local_critic_1(states, actions),
+                    self.local_critic_2(states, actions))
+        return self.local_critic(states, actions)
+
+    def _target_q(self, states, actions):
+        if self.variant in ("sac", "vsac"):
+            return (self.target_critic_1(states, actions),
+                    self.target_critic_2(states, actions))
+        return self.target_critic(states, actions)
+
+    def _per_sample_alpha(self, mtobss: torch.Tensor) -> torch.Tensor:
+        """mtsac: per-sample alpha via one-hot gather; sac/vsac: scalar."""
+        if self.variant == "mtsac":
+            one_hots = mtobss[:, -self.num_tasks:]
+            return Fops.gather_log_alpha(one_hots, self.log_alpha).exp().detach()
+        return self.alpha
+
+    def update(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
+        """One SAC gradient update; returns scalar metrics.
+
+        Order (identical to reference update/update_SAC): TD target →
+        critic step → actor step → alpha step → Polyak.
+        """
+        states = batch["states"]
+        actions = batch["actions"]
+        rewards = batch["rewards"]
+        next_states = batch["next_states"]
+        dones = batch["dones"]
+
+        alpha = self._per_sample_alpha(states)
+        self.zero_grad()
+
+        # --- TD target (no grad) -------------------------------------
+        with torch.no_grad():
+            next_actions, next_log_probs, _ = self._sample(next_states)
+            q1_t, q2_t = self._target_q(next_states, next_actions)
+            y = Fops.td_target(rewards, dones, q1_t, q2_t, next_log_probs,
+                               alpha, self.gamma, self.reward_scale)
+
+        # --- critic step ---------------------------------------------
+        if self.variant in ("sac", "vsac"):
+            q_loss = (self.local_critic_1.cal_loss(states, actions, y)
+                      + self.local_critic_2.cal_loss(states, actions, y))
+        else:
+            l1, l2 = self.local_critic.cal_loss(
+                states, actions, y,
+                use_weighted_loss=self.use_weighted_loss,
+                alphas=self.log_alpha.exp().detach())
+            q_loss = l1 + l2
+        q_loss.backward()
+        self.critic_optimizer.step()
+
+        # --- actor step ----------------------------------------------
+        sampled_actions, log_probs, log_stds = self._sample(states)
+        q1, q2 = self._critic_q(states, sampled_actions)
+        q_min = torch.min(q1, q2)
+        if self.variant == "mtsac":
+            policy_loss = self.actor.cal_loss(
+                log_probs, q_min, alpha,
+                use_weighted_loss=self.use_weighted_loss, mtobss=states,
+                alphas=self.log_alpha.exp().detach())
+        else:
+            policy_loss = self.actor.cal_loss(log_probs, q_min, alpha)
+        policy_loss.backward()
+        self.actor_optimizer.step()
+
+        # --- entropy diagnostic (reference learner.py:311-312) --------
+        entropy = Fops.entropy_from_log_std(log_stds)
+
+        # --- temperature step -----------------------------------------
+        if self.variant == "mtsac":
+            log_alpha_g = Fops.gather_log_alpha(
+                states[:, -self.num_tasks:], self.log_alpha)
+            loss_log_alpha = -(log_alpha_g * (log_probs.detach() + self.H_bar)).mean()
+        else:
+            loss_log_alpha = -(self.log_alpha * (log_probs.detach() + self.H_bar)).mean()
+        loss_log_alpha.backward()
+        self.log_alpha_optimizer.step()
+        self.alpha = self.log_alpha.exp().detach()
+
+        # --- Polyak target update -------------------------------------
+        flat_polyak_(self.target_group, self.critic_group, self.tau)
+
+        self.update_iteration += 1
+        return {
+            "critic_loss": float(q_loss.detach()),
+            "actor_loss": float(policy_loss.detach()),
+            "alpha_loss": float(loss_log_alpha.detach()),
+            "entropy": float(entropy.detach()),
+        }
+
+    # ------------------------------------------------------------------
+    # Checkpointing — reference .tar schema (learner.save_checkpoint).
+    # ------------------------------------------------------------------
+    def checkpoint_state(self) -> Dict:
+        def cpu_sd(m):
+            return {k: v.cpu() for k, v in m.state_dict().items()}
+
+        state = {
+            "update_iteration": self.update_iteration,
+            "total_step": self.total_step,
+            "actor": cpu_sd(self.actor),
+            "actor_optimizer": self.actor_optimizer.state_dict(),
+            "critic_optimizer": self.critic_optimizer.state_dict(),
+            "log_alpha": self.log_alpha.detach().cpu(),
+            "log_alpha_optimizer": self.log_alpha_optimizer.state_dict(),
+            "alpha": self.alpha.cpu(),
+        }
+        if self.variant in ("sac", "vsac"):
+            # LunarLander…/src/learner.py:144-163 key layout
+            state.update({
+                "local_critic_1": cpu_sd(self.local_critic_1),
+                "local_critic_2": cpu_sd(self.local_critic_2),
+                "target_critic_1": cpu_sd(self.target_critic_1),
+                "target_critic_2": cpu_sd(self.target_critic_2),
+            })
+        else:
+            # MT10_Distributed_MTSAC/src/learner.py:157-174 key layout
+            state.update({
+                "local_critic": cpu_sd(self.local_critic),
+                "target_critic": cpu_sd(self.target_critic),
+            })
+        return state
+
+    def load_checkpoint_state(self, ckpt: Dict) -> None:
+        """Inverse of checkpoint_state — also fixes the reference's broken
+        learner resume (it referenced a nonexistent ``actor.optimizer``,
+        LunarLander…/src/learner.py:178)."""
+        self.update_iteration = int(ckpt.get("update_iteration", 0))
+        self.total_step = int(ckpt.get("total_step", 0))
+        self.actor.load_state_dict(ckpt["actor"])
+        if self.variant in ("sac", "vsac"):
+            self.local_critic_1.load_state_dict(ckpt["local_critic_1"])
+            self.local_critic_2.load_state_dict(ckpt["local_critic_2"])
+            self.target_critic_1.load_state_dict(ckpt["target_critic_1"])
+            self.target_critic_2.load_state_dict(ckpt["target_critic_2"])
+        else:
+            self.local_critic.load_state_dict(ckpt["local_critic"])
+            self.target_critic.load_state_dict(ckpt["target_critic"])
+        with torch.no_grad():
+            self.log_alpha.copy_(ckpt["log_alpha"].to(self.device))
+        self.alpha = self.log_alpha.exp().detach()
+        if "actor_optimizer" in ckpt:
+            self.actor_optimizer.load_state_dict(ckpt["actor_optimizer"])
+        if "critic_optimizer" in ckpt:
+            self.critic_optimizer.load_state_dict(ckpt["critic_optimizer"])
+        if "log_alpha_optimizer" in ckpt:
+            self.log_alpha_optimizer.load_state_dict(ckpt["log_alpha_optimizer"])
+

@@ -1,0 +1,176 @@
+"""Device-resident sharded replay buffer.
+
+MI355X-first re-design of the reference's replay layer
+(LunarLander_Distributed_SAC/src/replay_buffer.py:13-77 single deque;
+MT10_Distributed_MTSAC/src/replay_buffers.py:13-107 per-task deques):
+
+- Transitions live as SoA ring buffers in device memory.  The reference's
+  1e6-transition MT10 buffer is ~250 MB fp32 — trivial against 288 GB HBM3E
+  per GPU, so the whole buffer is GPU-resident and a minibatch sample is a
+  device-side gather with NO host round-trip (replaces SURVEY §2.6 K12:
+  np.vstack + torch.from_numpy().to(device) per update).
+- Ingest crosses the PCIe/host boundary once, batched: numpy transition
+  blocks are staged into a reusable pinned buffer and copied with one
+  async H2D per field (overlappable with the update stream).
+- MT variants shard per task with stratified sampling (batch//num_tasks
+  from each shard, shuffled concat) and ``len = min over shards`` —
+  reference replay_buffers.py:37-41,67-100 semantics.
+
+Deviation from the reference, by design: minibatch indices are drawn with
+replacement (torch.randint) instead of ``random.sample``'s without-
+replacement draw — at batch 1280 from >=5000 entries the collision rate is
+<15% of samples and statistically immaterial for SAC, while avoiding a
+1e6-element randperm per update.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Sequence
+
+import numpy as np
+import torch
+
+FIELDS = ("states", "actions", "rewards", "next_states", "dones")
+
+
+class ReplayShard:
+    """One SoA ring buffer (one task shard)."""
+
+    def __init__(self, capacity: int, state_dim: int, action_dim: int,
+                 device: torch.device | str = "cpu"):
+        self.capacity = int(capacity)
+        self.device = torch.device(device)
+        self.state_dim = state_dim
+        self.action_dim = action_dim
+        dev = self.device
+        self.states = torch.zeros(self.capacity, state_dim, device=dev)
+        self.actions = torch.zeros(self.capacity, action_dim, device=dev)
+        self.rewards = torch.zeros(self.capacity, 1, device=dev)
+        self.next_states = torch.zeros(self.capacity, state_dim, device=dev)
+        self.dones = torch.zeros(self.capacity, 1, device=dev)
+        self.write_ptr = 0
+        self.size = 0
+
+    def __len__(self) -> int:
+        return self.size
+
+    @torch.no_grad()
+    def append(self, states: torch.Tensor, actions: torch.Tensor,
+               rewards: torch.Tensor, next_states: torch.Tensor,
+               dones: torch.Tensor) -> None:
+        """Append a block of n transitions (device tensors), wrapping."""
+        n = states.shape[0]
+        if n == 0:
+            return
+        if n >= self.capacity:  # keep only the newest `capacity`
+            states, actions = states[-self.capacity:], actions[-self.capacity:]
+            rewards = rewards[-self.capacity:]
+            next_states, dones = next_states[-self.capacity:], dones[-self.capacity:]
+            n = self.capacity
+        idx = torch.arange(self.write_ptr, self.write_ptr + n,
+                           device=self.device) % self.capacity
+        self.states.index_copy_(0, idx, states)
+        self.actions.index_copy_(0, idx, actions)
+        self.rewards.index_copy_(0, idx, rewards.reshape(n, 1))
+        self.next_states.index_copy_(0, idx, next_states)
+        self.dones.index_copy_(0, idx, dones.reshape(n, 1))
+        self.write_ptr = (self.write_ptr + n) % self.capacity
+        self.size = min(self.size + n, self.capacity)
+
+    @torch.no_grad()
+    def sample_indices(self, n: int,
+                       generator: Optional[torch.Generator] = None) -> torch.Tensor:
+        return torch.randint(0, self.size, (n,), device=self.device,
+                             generator=generator)
+
+    @torch.no_grad()
+    def gather(self, idx: torch.Tensor) -> Dict[str, torch.Tensor]:
+        return {
+            "states": self.states.index_select(0, idx),
+            "actions": self.actions.index_select(0, idx),
+            "rewards": self.rewards.index_select(0, idx),
+            "next_states": self.next_states.index_select(0, idx),
+            "dones": self.dones.index_select(0, idx),
+        }
+
+    @torch.no_grad()
+    def sample(self, n: int, generator: Optional[torch.Generator] = None):
+        return self.gather(self.sample_indices(n, generator))
+
+
+class ShardedReplay:
+    """Per-task shards + stratified sampling (MT semantics) with a pinned
+    staging path for host-produced transitions.
+
+    With ``num_tasks == 1`` this is the single-buffer LL/VSAC replay.
+    """
+
+    def __init__(self, buffer_size: int, num_tasks: int, state_dim: int,
+                 action_dim: int, device: torch.device | str = "cpu",
+                 seed: Optional[int] = None):
+        self.num_tasks = num_tasks
+        self.device = torch.device(device)
+        per_task = int(buffer_size) // num_tasks  # reference replay_buffers.py:37-41
+        self.shards: List[ReplayShard] = [
+            ReplayShard(per_task, state_dim, action_dim, device)
+            for _ in range(num_tasks)]
+        self.generator = None
+        if seed is not None:
+            self.generator = torch.Generator(device=self.device)
+            self.generator.manual_seed(seed)
+        self._pinned: Dict[str, torch.Tensor] = {}
+
+    def __len__(self) -> int:
+        """min over shards (reference replay_buffers.__len__:102-107)."""
+        return min(len(s) for s in self.shards)
+
+    @property
+    def total_size(self) -> int:
+        return sum(len(s) for s in self.shards)
+
+    def _stage(self, name: str, arr: np.ndarray) -> torch.Tensor:
+        """numpy -> pinned host tensor -> async device copy."""
+        t = torch.from_numpy(np.ascontiguousarray(arr, dtype=np.float32))
+        if self.device.type == "cuda":
+            key = name
+            buf = self._pinned.get(key)
+            if buf is None or buf.numel() < t.numel():
+                buf = torch.empty(max(t.numel(), 4096), pin_memory=True)
+                self._pinned[key] = buf
+            staged = buf[: t.numel()].view_as(t.reshape(-1)).view(t.shape)
+            staged.copy_(t)
+            return staged.to(self.device, non_blocking=True)
+        return t
+
+    @torch.no_grad()
+    def append_numpy(self, states, actions, rewards, next_states, dones,
+                     task_idx: int = 0) -> None:
+        """Ingest a block of host transitions into one task shard."""
+        s = self._stage("s", states)
+        a = self._stage("a", actions)
+        r = self._stage("r", np.asarray(rewards).reshape(-1, 1))
+        ns = self._stage("ns", next_states)
+        d = self._stage("d", np.asarray(dones, dtype=np.float32).reshape(-1, 1))
+        self.shards[task_idx].append(s, a, r, ns, d)
+
+    @torch.no_grad()
+    def append(self, task_idx: int, **fields: torch.Tensor) -> None:
+        self.shards[task_idx].append(
+            fields["states"], fields["actions"], fields["rewards"],
+            fields["next_states"], fields["dones"])
+
+    @torch.no_grad()
+    def sample(self, batch_size: int) -> Dict[str, torch.Tensor]:
+        """Stratified across shards, shuffled concat (reference
+        replay_buffers.sample:67-100); single shard = plain uniform."""
+        if self.num_tasks == 1:
+            return self.shards[0].sample(batch_size, self.generator)
+        per = batch_size // self.num_tasks
+        parts = [s.sample(per, self.generator) for s in self.shards]
+        out: Dict[str, torch.Tensor] = {}
+        perm = torch.randperm(per * self.num_tasks, device=self.device,
+                              generator=self.generator)
+        for f in FIELDS:
+            cat = torch.cat([p[f] for p in parts], dim=0)
+            out[f] = cat.index_select(0, perm)
+        return out

@@ -1,0 +1,145 @@
+"""Vectorized rollout core — the Player's environment loop.
+
+Re-design of the reference Player (LunarLander_Distributed_SAC/src/
+player.py:13-153; MT10_Distributed_MTSAC/src/player.py:13-286): one worker
+owns a SET of envs (possibly spanning several tasks, reference task
+round-robin player.py:247-253) and steps them in lockstep with BATCHED
+actor inference — instead of the reference's one-env-per-process B=1
+inference + per-step Redis weight download.
+
+Semantics kept from the reference:
+- per-task random warmup for ``random_step`` env steps (player.py:189);
+- mtobs = concat(state, one_hot(task)) (MT10…MTSAC/src/player.py:155-170);
+- done-masking: LunarLander stores done=False when the episode hit the time
+  limit... (reference player.py:111-112 masks `done` given max steps);
+- success-rate evaluation: 50 deterministic episodes per task
+  (MT1_Distributed_VSAC/src/player.py:102-143).
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..config import SACConfig
+from ..models import Actor, LLActor
+
+
+def one_hot(idx: int, n: int) -> np.ndarray:
+    v = np.zeros(n, dtype=np.float32)
+    v[idx] = 1.0
+    return v
+
+
+class VecRollout:
+    """Steps N envs with batched inference on a local actor copy."""
+
+    def __init__(self, cfg: SACConfig, envs: List, task_indices: List[int],
+                 actor: torch.nn.Module, device: torch.device | str = "cpu",
+                 seed: int = 0):
+        assert len(envs) == len(task_indices)
+        self.cfg = cfg
+        self.envs = envs
+        self.task_indices = task_indices
+        self.actor = actor
+        self.device = torch.device(device)
+        self.num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 0
+        self.rng = np.random.default_rng(seed)
+        self.obs = [self._mtobs(e.reset(), t) for e, t in zip(envs, task_indices)]
+        self.ep_steps = [0] * len(envs)
+        self.ep_rewards = [0.0] * len(envs)
+        self.total_steps_per_task: Dict[int, int] = {t: 0 for t in set(task_indices)}
+        self.episode_rewards: Dict[int, List[float]] = {t: [] for t in set(task_indices)}
+        self.warmup_remaining = {t: cfg.random_step for t in set(task_indices)}
+
+    def _mtobs(self, state: np.ndarray, task_idx: int) -> np.ndarray:
+        if self.num_tasks:
+            return np.concatenate([state, one_hot(task_idx, self.num_tasks)])
+        return np.asarray(state, dtype=np.float32)
+
+    @torch.no_grad()
+    def _policy_actions(self, obs_batch: np.ndarray) -> np.ndarray:
+        x = torch.from_numpy(obs_batch).to(self.device)
+        a = self.actor.get_action(x, stochastic=True)
+        return a.cpu().numpy()
+
+    def collect(self, n_steps: int) -> Dict[int, Dict[str, np.ndarray]]:
+        """Run n_steps lockstep env steps; returns per-task transition blocks
+        {task: {states, actions, rewards, next_states, dones}}."""
+        out: Dict[int, Dict[str, List]] = {
+            t: {k: [] for k in ("states", "actions", "rewards",
+                                "next_states", "dones")}
+            for t in set(self.task_indices)}
+        n_envs = len(self.envs)
+        for _ in range(n_steps):
+            obs_batch = np.stack(self.obs).astype(np.float32)
+            need_policy = [self.warmup_remaining[t] <= 0 for t in self.task_indices]
+            actions = np.zeros((n_envs, self.cfg.action_dim), dtype=np.float32)
+            if any(need_policy):
+                pol = self._policy_actions(obs_batch)
+                for i, np_ in enumerate(need_policy):
+                    if np_:
+                        actions[i] = pol[i]
+            for i, np_ in enumerate(need_policy):
+                if not np_:
+                    actions[i] = self.envs[i].action_space.sample()
+
+            for i, env in enumerate(self.envs):
+                t = self.task_indices[i]
+                next_state, reward, done, info = env.step(actions[i])
+                self.ep_steps[i] += 1
+                self.ep_rewards[i] += reward
+                next_obs = self._mtobs(next_state, t)
+                # time-limit masking (reference player.py stores done=False
+                # on max_episode_time truncation so bootstrap continues)
+                timeout = self.ep_steps[i] >= self.cfg.max_episode_time
+                stored_done = bool(done) and not timeout
+                out[t]["states"].append(self.obs[i])
+                out[t]["actions"].append(actions[i])
+                out[t]["rewards"].append(reward)
+                out[t]["next_states"].append(next_obs)
+                out[t]["dones"].append(float(stored_done))
+                if self.warmup_remaining[t] > 0:
+                    self.warmup_remaining[t] -= 1
+                self.total_steps_per_task[t] += 1
+                if done or timeout:
+                    self.episode_rewards[t].append(self.ep_rewards[i])
+                    self.obs[i] = self._mtobs(env.reset(), t)
+                    self.ep_steps[i] = 0
+                    self.ep_rewards[i] = 0.0
+                else:
+                    self.obs[i] = next_obs
+        return {t: {k: np.asarray(v, dtype=np.float32)
+                    for k, v in blk.items()} for t, blk in out.items()}
+
+    @torch.no_grad()
+    def evaluate_success_rate(self, task_idx: int, episodes: int = 50,
+                              max_steps: Optional[int] = None) -> float:
+        """Deterministic 50-episode protocol (reference
+        MT1_Distributed_VSAC/src/player.py:102-143)."""
+        env = self.envs[self.task_indices.index(task_idx)]
+        max_steps = max_steps or self.cfg.max_episode_time
+        successes = 0
+        for _ in range(episodes):
+            state = env.reset()
+            obs = self._mtobs(state, task_idx)
+            succeeded = False
+            for _ in range(max_steps):
+                x = torch.from_numpy(obs[None].astype(np.float32)).to(self.device)
+                a = self.actor.get_action(x, stochastic=False).cpu().numpy()[0]
+                state, reward, done, info = env.step(a)
+                obs = self._mtobs(state, task_idx)
+                if info.get("success", 0):
+                    succeeded = True
+                    break
+                if done:
+                    break
+            successes += int(succeeded)
+        # restore training episode state for that env
+        i = self.task_indices.index(task_idx)
+        self.obs[i] = self._mtobs(env.reset(), task_idx)
+        self.ep_steps[i] = 0
+        self.ep_rewards[i] = 0.0
+        return successes / episodes

@@ -1,0 +1,75 @@
+"""End-to-end single-process training slice on synthetic envs (CPU)."""
+
+import numpy as np
+import torch
+
+from distributed_sac_amd.config import SACConfig, load_variant
+from distributed_sac_amd.workers import Trainer, VecRollout
+
+
+def tiny_cfg(variant="sac"):
+    from tests.test_engine import small_cfg
+    c = small_cfg(variant)
+    c.buffer_size = 4000
+    c.start_memory_len = 64
+    c.random_step = 32
+    c.batch_size = 16
+    c.max_episode_time = 50
+    return c
+
+
+def test_trainer_end_to_end_sac():
+    torch.manual_seed(0)
+    cfg = tiny_cfg("sac")
+    tr = Trainer(cfg, device="cpu", seed=0)
+    metrics = tr.train(env_steps_per_iter=40, updates_per_iter=2, iterations=4)
+    assert tr.engine.update_iteration >= 2
+    assert np.isfinite(metrics["critic_loss"])
+    assert len(tr.replay) > 0
+    assert tr.engine.total_step == 160
+
+
+def test_trainer_end_to_end_mtsac():
+    torch.manual_seed(0)
+    cfg = tiny_cfg("mtsac")
+    tr = Trainer(cfg, device="cpu", seed=1)
+    tr.train(env_steps_per_iter=40, updates_per_iter=1, iterations=3)
+    # every task shard received transitions
+    assert all(len(s) > 0 for s in tr.replay.shards)
+    assert tr.engine.update_iteration >= 1
+
+
+def test_rollout_warmup_and_mtobs():
+    cfg = tiny_cfg("mtsac")
+    cfg.random_step = 10
+    tr = Trainer(cfg, device="cpu", seed=2)
+    blocks = tr.rollout.collect(5)
+    for t, blk in blocks.items():
+        oh = blk["states"][:, -cfg.num_tasks:]
+        assert np.allclose(oh.sum(axis=1), 1.0)
+        assert np.all(oh.argmax(axis=1) == t)
+    # warmup countdown decremented
+    assert all(v < cfg.random_step for v in tr.rollout.warmup_remaining.values())
+
+
+def test_success_rate_eval_protocol():
+    cfg = tiny_cfg("mtsac")
+    cfg.max_episode_time = 20
+    tr = Trainer(cfg, device="cpu", seed=3)
+    rate = tr.rollout.evaluate_success_rate(0, episodes=3, max_steps=10)
+    assert 0.0 <= rate <= 1.0
+
+
+def test_canonical_cfgs_load():
+    for v in ("sac", "vsac", "mtsac", "care", "mt1_care"):
+        cfg = load_variant(v)
+        assert cfg.buffer_size == 1_000_000
+        if v == "mtsac":
+            assert cfg.batch_size == 1280 and cfg.num_tasks == 10
+            assert cfg.use_weighted_loss
+            assert cfg.mtobs_dim == 49
+        if v == "sac":
+            assert cfg.state_dim == 8 and cfg.batch_size == 256
+        if v == "care":
+            assert cfg.encoder is not None
+            assert cfg.encoder["num_encoders"] == 6
