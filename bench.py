@@ -1,0 +1,184 @@
+#!/usr/bin/env python3
+"""Flagship benchmark — learner grad-steps/sec on the MT10-MTSAC config.
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+    (for N>1 launched via torch.distributed.run, one rank per GPU over RCCL)
+
+Measures the BASELINE.json metric ("learner grad-steps/sec + actor
+env-steps/sec, LunarLander SAC & MT10-MTSAC") on the MT10-MTSAC shipped
+config: batch 1280, 39+10-d mtobs, 4-d actions, 3x400 MLPs, twin critics,
+per-task alpha, weighted loss — synthetic transitions, random-init weights,
+fp32 end-to-end (the reference's precision; fp32 >= bf16).  Each timed step
+is ONE full SAC gradient update (TD target, critic fwd+bwd+Adam, actor
+fwd+bwd+Adam, alpha fwd+bwd+Adam, Polyak) — nothing is skipped or cached.
+For N>1 the learner is data-parallel: flat-bucket RCCL all-reduce of
+critic/actor/alpha gradients per update (weak scaling: per-GPU batch fixed).
+
+`value` = aggregate grad-steps/sec = N * K / max-over-ranks(elapsed).
+`env_steps_per_sec` (auxiliary, untimed region) = synthetic-env rollout
+throughput of one vectorized player with batched policy inference.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=300)
+    p.add_argument("--warmup", type=int, default=30)
+    p.add_argument("--config", type=str, default="mtsac",
+                   choices=["mtsac", "sac", "vsac"])
+    p.add_argument("--device", type=str, default=None)
+    p.add_argument("--skip-rollout-probe", action="store_true")
+    return p.parse_args()
+
+
+def prefill_replay(replay, cfg, per_shard: int, device, seed: int):
+    g = torch.Generator(device=device)
+    g.manual_seed(seed)
+    num_tasks = len(replay.shards)
+    for t, shard in enumerate(replay.shards):
+        n = per_shard
+        states = torch.randn(n, cfg.mtobs_dim, device=device, generator=g)
+        next_states = torch.randn(n, cfg.mtobs_dim, device=device, generator=g)
+        if num_tasks > 1:
+            oh = torch.zeros(n, num_tasks, device=device)
+            oh[:, t] = 1.0
+            states[:, -num_tasks:] = oh
+            next_states[:, -num_tasks:] = oh
+        shard.append(
+            states,
+            (torch.rand(n, cfg.action_dim, device=device, generator=g) * 2 - 1),
+            torch.randn(n, 1, device=device, generator=g),
+            next_states,
+            (torch.rand(n, 1, device=device, generator=g) < 0.02).float(),
+        )
+
+
+def rollout_probe(cfg, device, n_envs: int = 16, steps: int = 100) -> float:
+    """Aux metric: env-steps/sec of one vectorized synthetic-env player."""
+    from distributed_sac_amd.workers.trainer import Trainer
+    import copy
+    c = copy.deepcopy(cfg)
+    c.random_step = 0  # measure policy-inference path, not warmup sampling
+    num_tasks = c.num_tasks if c.variant in ("mtsac", "care") else 1
+    per_task = max(1, n_envs // num_tasks)
+    tr = Trainer(c, device=device, envs_per_task=per_task, seed=1234)
+    tr.collect_steps(10)  # warm
+    t0 = time.perf_counter()
+    pushed = tr.collect_steps(steps)
+    dt = time.perf_counter() - t0
+    return pushed / dt
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world)
+
+    if args.device:
+        device = args.device
+    elif torch.cuda.is_available():
+        device = f"cuda:{local_rank}"
+        torch.cuda.set_device(device)
+    else:
+        device = "cpu"
+
+    from distributed_sac_amd.algo import SACEngine
+    from distributed_sac_amd.config import load_variant
+    from distributed_sac_amd.parallel import DataParallelGroup
+    from distributed_sac_amd.replay import ShardedReplay
+
+    cfg = load_variant(args.config)
+    cfg.device = device
+    torch.manual_seed(1000 + rank)
+
+    ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
+    engine = SACEngine(cfg, device)
+    if ddp is not None:
+        engine.attach_ddp(ddp)
+
+    num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
+    replay = ShardedReplay(cfg.buffer_size, num_tasks, cfg.mtobs_dim,
+                           cfg.action_dim, device=device, seed=42 + rank)
+    prefill_replay(replay, cfg, per_shard=max(4096, cfg.batch_size),
+                   device=device, seed=77 + rank)
+
+    env_rate = None
+    if rank == 0 and not args.skip_rollout_probe:
+        env_rate = rollout_probe(cfg, "cpu")
+
+    def sync():
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+
+    def one_step():
+        batch = replay.sample(cfg.batch_size)
+        engine.update(batch)
+
+    for _ in range(args.warmup):
+        one_step()
+
+    if ddp is not None:
+        ddp.barrier()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_step()
+    sync()
+    elapsed = time.perf_counter() - t0
+    if ddp is not None:
+        elapsed = ddp.max_scalar(elapsed)
+        ddp.barrier()
+
+    steps_per_sec = args.steps / elapsed
+    value = n_gpus * steps_per_sec
+    if rank == 0:
+        result = {
+            "metric": "learner grad-steps/sec (aggregate over GPUs), MT10-MTSAC"
+                      if args.config == "mtsac" else
+                      f"learner grad-steps/sec (aggregate over GPUs), {args.config}",
+            "value": round(value, 2),
+            "unit": "grad_steps/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            # reference publishes no grad-step rate for these exact configs
+            # (MT10-MTSAC rate logs stripped; LL rate only
+            # derivable): nearest derived reference rate is MT1-CARE
+            # 5.1 grad-steps/s @ B=1024 on a GTX 1080.
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "env_steps_per_sec": (round(env_rate, 1) if env_rate else None),
+            "config": {
+                "model": ("MT10-MTSAC (mtobs 49, act 4, actor 3x400, "
+                          "twin critic 3x400, per-task alpha, weighted loss)"
+                          if args.config == "mtsac" else args.config),
+                "global_batch": cfg.batch_size * n_gpus,
+                "seq_len": 1,
+                "parallelism": f"dp{n_gpus}",
+                "batch_per_gpu": cfg.batch_size,
+                "update": "full SAC step: TD target + critic/actor/alpha "
+                          "fwd+bwd+fusedAdam + Polyak",
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

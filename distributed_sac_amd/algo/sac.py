@@ -51,6 +51,7 @@ class SACEngine:
         self.total_step = 0
         self._graph = None
         self._eps_queue: Optional[list] = None  # test hook: deterministic eps
+        self.ddp = None  # optional DataParallelGroup (set via attach_ddp)
         self._build_models()
         self._build_optimizers()
 
@@ -123,6 +124,16 @@ class SACEngine:
         self.log_alpha_optimizer = FusedAdam(self.alpha_group, lr=cfg.lr_actor)
         self.hard_copy_targets()
 
+    def attach_ddp(self, ddp) -> None:
+        """Join a data-parallel group: sync replicas, then every update
+        all-reduces the flat gradient buffers (one RCCL message each)."""
+        self.ddp = ddp
+        if ddp is not None and ddp.enabled:
+            ddp.broadcast_params(self.actor_group.flat_data)
+            ddp.broadcast_params(self.critic_group.flat_data)
+            ddp.broadcast_params(self.alpha_group.flat_data)
+            self.hard_copy_targets()
+
     @torch.no_grad()
     def hard_copy_targets(self) -> None:
         """targets <- critics (reference soft_update tau=1.0 at run start)."""
@@ -190,6 +201,8 @@ class SACEngine:
                 alphas=self.log_alpha.exp().detach())
             q_loss = l1 + l2
         q_loss.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()
 
         # --- actor step ----------------------------------------------
@@ -204,6 +217,8 @@ class SACEngine:
         else:
             policy_loss = self.actor.cal_loss(log_probs, q_min, alpha)
         policy_loss.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
         self.actor_optimizer.step()
 
         # --- entropy diagnostic (reference learner.py:311-312) --------
@@ -217,6 +232,8 @@ class SACEngine:
         else:
             loss_log_alpha = -(self.log_alpha * (log_probs.detach() + self.H_bar)).mean()
         loss_log_alpha.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
         self.log_alpha_optimizer.step()
         self.alpha = self.log_alpha.exp().detach()
 

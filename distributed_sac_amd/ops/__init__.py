@@ -24,6 +24,7 @@ def _try_load():
     if _native is not None or _native_err is not None:
         return
     try:
+        import torch  # noqa: F401 — extension links against torch's libs
         _native = importlib.import_module("distributed_sac_amd.ops._hip_ops")
     except Exception as e:  # pragma: no cover - depends on build state
         _native_err = e
