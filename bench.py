@@ -39,6 +39,8 @@ def parse_args():
                    choices=["mtsac", "sac", "vsac"])
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--skip-rollout-probe", action="store_true")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture of the update step")
     return p.parse_args()
 
 
@@ -123,9 +125,23 @@ def main():
         if device.startswith("cuda"):
             torch.cuda.synchronize()
 
+    graphed = False
+    if device.startswith("cuda") and not args.no_graph:
+        try:
+            engine.capture(replay, cfg.batch_size)
+            graphed = True
+        except Exception as e:  # pragma: no cover
+            import sys
+            print(f"[bench] hipGraph capture failed, eager fallback: {e!r}",
+                  file=sys.stderr)
+
     def one_step():
-        batch = replay.sample(cfg.batch_size)
-        engine.update(batch)
+        if graphed:
+            engine.graphed_update()
+        else:
+            engine.update_tensors(replay.sample(cfg.batch_size,
+                                                graph_safe=True))
+            engine.update_iteration += 1
 
     for _ in range(args.warmup):
         one_step()
@@ -175,6 +191,7 @@ def main():
                 "batch_per_gpu": cfg.batch_size,
                 "update": "full SAC step: TD target + critic/actor/alpha "
                           "fwd+bwd+fusedAdam + Polyak",
+                "hipgraph": graphed,
             },
         }
         print(json.dumps(result))

@@ -169,7 +169,14 @@ class SACEngine:
         return self.alpha
 
     def update(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
-        """One SAC gradient update; returns scalar metrics.
+        """One SAC gradient update; returns scalar metrics (syncs)."""
+        out = self.update_tensors(batch)
+        self.update_iteration += 1
+        return {k: float(v.detach()) for k, v in out.items()}
+
+    def update_tensors(self, batch: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        """One SAC gradient update, returning tensor metrics (no host sync
+        — hipGraph-capturable).
 
         Order (identical to reference update/update_SAC): TD target →
         critic step → actor step → alpha step → Polyak.
@@ -240,13 +247,40 @@ class SACEngine:
         # --- Polyak target update -------------------------------------
         flat_polyak_(self.target_group, self.critic_group, self.tau)
 
-        self.update_iteration += 1
         return {
-            "critic_loss": float(q_loss.detach()),
-            "actor_loss": float(policy_loss.detach()),
-            "alpha_loss": float(loss_log_alpha.detach()),
-            "entropy": float(entropy.detach()),
+            "critic_loss": q_loss.detach(),
+            "actor_loss": policy_loss.detach(),
+            "alpha_loss": loss_log_alpha.detach(),
+            "entropy": entropy.detach(),
         }
+
+    # ------------------------------------------------------------------
+    # hipGraph capture: the ENTIRE update (sample gather, forward, backward,
+    # three fused Adam steps, Polyak) replays as one captured graph —
+    # per-step host cost collapses to one hipGraphLaunch.
+    # ------------------------------------------------------------------
+    def capture(self, replay, batch_size: int, warmup_iters: int = 3):
+        assert self.device.type == "cuda", "capture needs a GPU"
+        torch.cuda.synchronize(self.device)
+        side = torch.cuda.Stream(self.device)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                self.update_tensors(replay.sample(batch_size, graph_safe=True))
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        torch.cuda.synchronize(self.device)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            self._graph_metrics = self.update_tensors(
+                replay.sample(batch_size, graph_safe=True))
+        self._graph = graph
+        return graph
+
+    def graphed_update(self) -> Dict[str, torch.Tensor]:
+        """Replay the captured update (tensor metrics refresh in place)."""
+        self._graph.replay()
+        self.update_iteration += 1
+        return self._graph_metrics
 
     # ------------------------------------------------------------------
     # Checkpointing — reference .tar schema (learner.save_checkpoint).

@@ -333,6 +333,32 @@ __global__ __launch_bounds__(256) void k_adam(
   p[i] -= step_size * mi / (sqrtf(vi) * inv_sqrt_bc2 + eps);
 }
 
+// K9b: graph-capturable Adam — the step counter lives on-device so a
+// captured update's bias correction advances across hipGraph replays.
+// k_adam_prolog: step += 1; coeffs = {step_size, inv_sqrt_bc2}.
+__global__ void k_adam_prolog(float* __restrict__ state, float lr, float b1,
+                              float b2) {
+  const float step = state[0] + 1.f;
+  state[0] = step;
+  state[1] = lr / (1.f - __powf(b1, step));          // step_size
+  state[2] = 1.f / sqrtf(1.f - __powf(b2, step));    // inv_sqrt_bc2
+}
+
+__global__ __launch_bounds__(256) void k_adam_dev(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ m, float* __restrict__ v,
+    const float* __restrict__ state, long n, float b1, float b2, float eps) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const float step_size = state[1], inv_sqrt_bc2 = state[2];
+  const float gi = g[i];
+  const float mi = b1 * m[i] + (1.f - b1) * gi;
+  const float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+  m[i] = mi;
+  v[i] = vi;
+  p[i] -= step_size * mi / (sqrtf(vi) * inv_sqrt_bc2 + eps);
+}
+
 // ---------------------------------------------------------------------------
 // K10: t = (1-tau)*t + tau*s over flat buffers.
 // ---------------------------------------------------------------------------
@@ -473,6 +499,22 @@ static void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      (float)eps);
 }
 
+static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+                           torch::Tensor v, torch::Tensor state, double lr,
+                           double b1, double b2, double eps) {
+  CHECK_IN(p); CHECK_IN(state);
+  TORCH_CHECK(state.numel() >= 3, "state = {step, step_size, inv_sqrt_bc2}");
+  const long n = p.numel();
+  hipLaunchKernelGGL(k_adam_prolog, dim3(1), dim3(1), 0, cur_stream(),
+                     state.data_ptr<float>(), (float)lr, (float)b1,
+                     (float)b2);
+  hipLaunchKernelGGL(k_adam_dev, dim3((n + 255) / 256), dim3(256), 0,
+                     cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     state.data_ptr<float>(), n, (float)b1, (float)b2,
+                     (float)eps);
+}
+
 static void polyak_(torch::Tensor t, torch::Tensor s, double tau) {
   CHECK_IN(t);
   const long n = t.numel();
@@ -490,5 +532,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("squashed_gaussian_bwd", &squashed_gaussian_bwd);
   mod.def("td_target", &td_target);
   mod.def("adam_step_", &adam_step_);
+  mod.def("adam_step_dev_", &adam_step_dev_);
   mod.def("polyak_", &polyak_);
 }

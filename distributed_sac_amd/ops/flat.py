@@ -112,25 +112,44 @@ class FusedAdam:
         self.lr = float(lr)
         self.betas = (float(betas[0]), float(betas[1]))
         self.eps = float(eps)
-        self.step_count = 0
+        self._step_count = 0
         self.exp_avg = torch.zeros_like(group.flat_data)
         self.exp_avg_sq = torch.zeros_like(group.flat_data)
+        # hipGraph-capturable path: the step counter + bias-correction
+        # coefficients live on-device ({step, step_size, inv_sqrt_bc2}) so a
+        # captured update keeps correct Adam bias correction across replays.
+        self._dev_state = None
+        if group.flat_data.is_cuda and native_enabled() and has_native():
+            self._dev_state = torch.zeros(3, device=group.flat_data.device)
+
+    @property
+    def step_count(self) -> int:
+        if self._dev_state is not None:
+            return int(self._dev_state[0].item())
+        return self._step_count
+
+    @step_count.setter
+    def step_count(self, v: int) -> None:
+        self._step_count = int(v)
+        if self._dev_state is not None:
+            with torch.no_grad():
+                self._dev_state[0] = float(v)
 
     def zero_grad(self) -> None:
         self.group.zero_grad()
 
     @torch.no_grad()
     def step(self) -> None:
-        self.step_count += 1
         p, g = self.group.flat_data, self.group.flat_grad
         m, v = self.exp_avg, self.exp_avg_sq
         b1, b2 = self.betas
-        if p.is_cuda and native_enabled() and has_native():
-            native().adam_step_(p, g, m, v, self.step_count,
-                                self.lr, b1, b2, self.eps)
+        if self._dev_state is not None:
+            native().adam_step_dev_(p, g, m, v, self._dev_state,
+                                    self.lr, b1, b2, self.eps)
             return
-        bc1 = 1 - b1 ** self.step_count
-        bc2 = 1 - b2 ** self.step_count
+        self._step_count += 1
+        bc1 = 1 - b1 ** self._step_count
+        bc2 = 1 - b2 ** self._step_count
         m.mul_(b1).add_(g, alpha=1 - b1)
         v.mul_(b2).addcmul_(g, g, value=1 - b2)
         denom = (v.sqrt() / math.sqrt(bc2)).add_(self.eps)

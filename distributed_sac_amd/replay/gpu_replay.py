@@ -50,6 +50,9 @@ class ReplayShard:
         self.dones = torch.zeros(self.capacity, 1, device=dev)
         self.write_ptr = 0
         self.size = 0
+        # device-resident size for hipGraph-safe sampling (updated by
+        # append OUTSIDE any captured region; read inside the graph)
+        self.size_dev = torch.zeros(1, device=dev)
 
     def __len__(self) -> int:
         return self.size
@@ -76,10 +79,18 @@ class ReplayShard:
         self.dones.index_copy_(0, idx, dones.reshape(n, 1))
         self.write_ptr = (self.write_ptr + n) % self.capacity
         self.size = min(self.size + n, self.capacity)
+        self.size_dev.fill_(float(self.size))
 
     @torch.no_grad()
     def sample_indices(self, n: int,
-                       generator: Optional[torch.Generator] = None) -> torch.Tensor:
+                       generator: Optional[torch.Generator] = None,
+                       graph_safe: bool = False) -> torch.Tensor:
+        if graph_safe:
+            # hipGraph-capturable draw: default-generator rand (philox with
+            # graph-managed offsets) scaled by the device-resident size.
+            # rand < 1.0 strictly and sizes < 2^24, so floor(r*size) < size.
+            r = torch.rand(n, device=self.device)
+            return (r * self.size_dev).long()
         return torch.randint(0, self.size, (n,), device=self.device,
                              generator=generator)
 
@@ -94,8 +105,9 @@ class ReplayShard:
         }
 
     @torch.no_grad()
-    def sample(self, n: int, generator: Optional[torch.Generator] = None):
-        return self.gather(self.sample_indices(n, generator))
+    def sample(self, n: int, generator: Optional[torch.Generator] = None,
+               graph_safe: bool = False):
+        return self.gather(self.sample_indices(n, generator, graph_safe))
 
 
 class ShardedReplay:
@@ -160,17 +172,25 @@ class ShardedReplay:
             fields["next_states"], fields["dones"])
 
     @torch.no_grad()
-    def sample(self, batch_size: int) -> Dict[str, torch.Tensor]:
+    def sample(self, batch_size: int,
+               graph_safe: bool = False) -> Dict[str, torch.Tensor]:
         """Stratified across shards, shuffled concat (reference
-        replay_buffers.sample:67-100); single shard = plain uniform."""
+        replay_buffers.sample:67-100); single shard = plain uniform.
+
+        graph_safe=True uses the hipGraph-capturable index draw and skips
+        the concat shuffle (every consumer of the batch is
+        permutation-invariant: all losses are batch means)."""
+        gen = None if graph_safe else self.generator
         if self.num_tasks == 1:
-            return self.shards[0].sample(batch_size, self.generator)
+            return self.shards[0].sample(batch_size, gen, graph_safe)
         per = batch_size // self.num_tasks
-        parts = [s.sample(per, self.generator) for s in self.shards]
+        parts = [s.sample(per, gen, graph_safe) for s in self.shards]
         out: Dict[str, torch.Tensor] = {}
-        perm = torch.randperm(per * self.num_tasks, device=self.device,
-                              generator=self.generator)
+        perm = None
+        if not graph_safe:
+            perm = torch.randperm(per * self.num_tasks, device=self.device,
+                                  generator=gen)
         for f in FIELDS:
             cat = torch.cat([p[f] for p in parts], dim=0)
-            out[f] = cat.index_select(0, perm)
+            out[f] = cat.index_select(0, perm) if perm is not None else cat
         return out
