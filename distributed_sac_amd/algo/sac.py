@@ -100,17 +100,62 @@ class SACEngine:
         self.H_bar = torch.tensor([-float(cfg.action_dim)], device=dev)
         self.alpha = self.log_alpha.exp().detach()
 
-    def _critic_params(self):
+    @staticmethod
+    def _critic_linears(c1_or_critic, c2=None):
+        """Per-layer Linear pairs (Q1_i, Q2_i) across both critic styles."""
+        if c2 is not None:  # LL style: two modules
+            l1 = [c1_or_critic.first_layer] + list(c1_or_critic.layer_module)
+            l2 = [c2.first_layer] + list(c2.layer_module)
+        else:  # MT style: one module with Q_function_1/2 Sequentials
+            import torch.nn as nn_
+            l1 = [m for m in c1_or_critic.Q_function_1 if isinstance(m, nn_.Linear)]
+            l2 = [m for m in c1_or_critic.Q_function_2 if isinstance(m, nn_.Linear)]
+        return list(zip(l1, l2))
+
+    def _critic_layer_pairs(self, target: bool = False):
         if self.variant in ("sac", "vsac"):
-            return itertools.chain(self.local_critic_1.parameters(),
-                                   self.local_critic_2.parameters())
-        return self.local_critic.parameters()
+            a = self.target_critic_1 if target else self.local_critic_1
+            b = self.target_critic_2 if target else self.local_critic_2
+            return self._critic_linears(a, b)
+        return self._critic_linears(self.target_critic if target
+                                    else self.local_critic)
+
+    def _critic_params(self):
+        """Interleaved per-layer order [Q1.w_i, Q2.w_i, Q1.b_i, Q2.b_i] so
+        the flat buffer contains contiguous stacked [2,N,K] weights for the
+        grouped twin-GEMM kernels."""
+        out = []
+        for l1, l2 in self._critic_layer_pairs():
+            out += [l1.weight, l2.weight, l1.bias, l2.bias]
+        return out
 
     def _target_params(self):
-        if self.variant in ("sac", "vsac"):
-            return itertools.chain(self.target_critic_1.parameters(),
-                                   self.target_critic_2.parameters())
-        return self.target_critic.parameters()
+        out = []
+        for l1, l2 in self._critic_layer_pairs(target=True):
+            out += [l1.weight, l2.weight, l1.bias, l2.bias]
+        return out
+
+    def _build_twin_stacks(self, group, pairs, with_grad: bool):
+        """Stacked [2,N,K]/[2,N] leaf views into a flat buffer, with .grad
+        views into the flat gradient so autograd accumulates straight into
+        the all-reduce/Adam buffer."""
+        ws, bs = [], []
+        for l1, l2 in pairs:
+            for (pa, pb, shape, dest) in (
+                    (l1.weight, l2.weight, l1.weight.shape, ws),
+                    (l1.bias, l2.bias, l1.bias.shape, bs)):
+                ia = next(i for i, q in enumerate(group.params) if q is pa)
+                ib = next(i for i, q in enumerate(group.params) if q is pb)
+                off = group.offsets[ia]
+                n = pa.numel()
+                assert group.offsets[ib] == off + n, "stacking needs adjacency"
+                view = group.flat_data[off:off + 2 * n].view(2, *shape)
+                stack = view.detach()
+                if with_grad:
+                    stack.requires_grad_(True)
+                    stack.grad = group.flat_grad[off:off + 2 * n].view(2, *shape)
+                dest.append(stack)
+        return ws, bs
 
     def _build_optimizers(self) -> None:
         cfg = self.cfg
@@ -122,6 +167,11 @@ class SACEngine:
         self.critic_optimizer = FusedAdam(self.critic_group, lr=cfg.lr_critic)
         # reference uses lr_actor for log_alpha (learner.py build_optimizer)
         self.log_alpha_optimizer = FusedAdam(self.alpha_group, lr=cfg.lr_actor)
+        self._twin_local = self._build_twin_stacks(
+            self.critic_group, self._critic_layer_pairs(), with_grad=True)
+        self._twin_target = self._build_twin_stacks(
+            self.target_group, self._critic_layer_pairs(target=True),
+            with_grad=False)
         self.hard_copy_targets()
 
     def attach_ddp(self, ddp) -> None:
@@ -149,13 +199,23 @@ class SACEngine:
         self.alpha_group.rebind_grads()
 
     # ------------------------------------------------------------------
+    def _use_fused(self, t: torch.Tensor) -> bool:
+        from ..ops import has_native, native_enabled
+        return t.is_cuda and native_enabled() and has_native()
+
     def _critic_q(self, states, actions):
+        if self._use_fused(states):
+            x = torch.cat([states, actions], dim=-1)
+            return Fops.twin_mlp_forward(x, *self._twin_local)
         if self.variant in ("sac", "vsac"):
             return (self.local_critic_1(states, actions),
                     self.local_critic_2(states, actions))
         return self.local_critic(states, actions)
 
     def _target_q(self, states, actions):
+        if self._use_fused(states):
+            x = torch.cat([states, actions], dim=-1)
+            return Fops.twin_mlp_forward(x, *self._twin_target)
         if self.variant in ("sac", "vsac"):
             return (self.target_critic_1(states, actions),
                     self.target_critic_2(states, actions))
@@ -181,6 +241,8 @@ class SACEngine:
         Order (identical to reference update/update_SAC): TD target →
         critic step → actor step → alpha step → Polyak.
         """
+        if self._use_fused(batch["states"]):
+            return self._update_tensors_fused(batch)
         states = batch["states"]
         actions = batch["actions"]
         rewards = batch["rewards"]
@@ -247,6 +309,58 @@ class SACEngine:
         # --- Polyak target update -------------------------------------
         flat_polyak_(self.target_group, self.critic_group, self.tau)
 
+        return {
+            "critic_loss": q_loss.detach(),
+            "actor_loss": policy_loss.detach(),
+            "alpha_loss": loss_log_alpha.detach(),
+            "entropy": entropy.detach(),
+        }
+
+    def _update_tensors_fused(self, batch):
+        """GPU path: same math/order as update_tensors, with the loss heads,
+        per-sample alpha gather and entropy fused into single kernels
+        (ops.functional.critic_loss / actor_alpha_loss) and the twin-Q
+        critics running as grouped GEMMs."""
+        from ..ops import native
+        states = batch["states"]
+        actions = batch["actions"]
+        rewards = batch["rewards"]
+        next_states = batch["next_states"]
+        dones = batch["dones"]
+        T = self.num_tasks
+        use_w = self.use_weighted_loss
+
+        self.zero_grad()
+        with torch.no_grad():
+            next_actions, next_log_probs, _ = self._sample(next_states)
+            q1_t, q2_t = self._target_q(next_states, next_actions)
+            y = native().td_target_mt(
+                rewards, dones, q1_t, q2_t, next_log_probs, states,
+                self.log_alpha.detach(), T, self.gamma, self.reward_scale)
+
+        q1, q2 = self._critic_q(states, actions)
+        l1, l2 = Fops.critic_loss(q1, q2, y, states, self.log_alpha.detach(),
+                                  T, use_w)
+        q_loss = l1 + l2
+        q_loss.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self.critic_optimizer.step()
+
+        sampled_actions, log_probs, log_stds = self._sample(states)
+        aq1, aq2 = self._critic_q(states, sampled_actions)
+        policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(
+            aq1, aq2, log_probs, log_stds, states, self.log_alpha, T, use_w,
+            float(self.H_bar))
+        (policy_loss + loss_log_alpha).backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
+            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+        self.actor_optimizer.step()
+        self.log_alpha_optimizer.step()
+        self.alpha = self.log_alpha.exp().detach()
+
+        flat_polyak_(self.target_group, self.critic_group, self.tau)
         return {
             "critic_loss": q_loss.detach(),
             "actor_loss": policy_loss.detach(),

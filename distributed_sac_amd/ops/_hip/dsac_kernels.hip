@@ -1,24 +1,34 @@
 // dsac_kernels.hip — CDNA4 (gfx950) kernels for the SAC hot path.
 //
 // Replaces the PyTorch op-sequences inventoried in SURVEY.md §2.6:
-//   K1  linear_act_fwd       — GEMM + bias + ReLU (MFMA f32 16x16x4)
-//   K8  linear_bwd_dx/dwdb   — backward GEMMs with fused ReLU masking +
-//                              fused bias-grad column reduction
+//   K1  linear_act_fwd[_g]   — GEMM + bias + ReLU (MFMA f32 16x16x4),
+//                              grouped over grid.z (twin critics in one
+//                              launch, shared activations)
+//   K8  linear_bwd_dx/dwdb   — backward GEMMs with fused ReLU masking;
+//                              dW/db is split-K over the batch (partials +
+//                              reduce) so the grid fills 256 CUs
 //   K4  squashed_gaussian_*  — fused clamp/exp/rsample/tanh/log-prob (+bwd)
 //   K5  td_target            — Bellman backup elementwise
-//   K9  adam_step_           — fused Adam over one flat parameter buffer
+//   K6  critic/actor/alpha loss — single-workgroup fused loss reductions
+//                              (incl. softmax(-alpha) task weights, per-task
+//                              alpha gather, entropy) with analytic backward
+//   K9  adam_step_[dev_]     — fused Adam over one flat parameter buffer
 //   K10 polyak_              — fused soft target update over flat buffers
+//   K12 replay_sample        — stratified gather from the HBM-resident
+//                              sharded replay in ONE kernel
 //
 // Design notes (see /opt/skills guides):
-// - fp32 end-to-end like the reference (no mixed precision); GEMMs use the
-//   exact f32-input MFMA v_mfma_f32_16x16x4_f32 (155 TF on MI355X — far
-//   above what these latency-bound tiny GEMMs need, at fp32-exact numerics).
+// - fp32 end-to-end like the reference; GEMMs use the exact f32-input MFMA
+//   v_mfma_f32_16x16x4_f32 (155 TF peak — these latency-bound tiny GEMMs
+//   are grid/launch-bound, not FLOP-bound, so exact fp32 costs nothing).
 // - tiles are 64x64x32 with LDS staging; row pads chosen so the MFMA
 //   fragment gathers are LDS-bank-conflict-free (+2 on 32-wide rows:
 //   bank = (34*r + k) % 32 = (2r + k) % 32 distinct for r<16, k<2;
 //   +16 on 64-wide rows: (80*m + j) % 32 = (16m + j) % 32 distinct).
-// - one workgroup = 4 waves, each wave owns a 32x32 output sub-tile as a
-//   2x2 grid of 16x16 MFMA fragments (wave64 per-wave MFMA, not warp32).
+// - one workgroup = 4 waves (wave64), each wave owns a 32x32 output
+//   sub-tile as a 2x2 grid of 16x16 MFMA fragments.
+// - everything is hipGraph-capturable: no host-side state in the hot path
+//   (Adam step counter lives on-device).
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
@@ -29,8 +39,6 @@
 #include <vector>
 
 using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
-
-#define DEV_INLINE __device__ __forceinline__
 
 static constexpr int BM = 64;   // batch-tile rows
 static constexpr int BN = 64;   // out-tile cols
@@ -46,14 +54,21 @@ static inline hipStream_t cur_stream() {
 }
 
 // ---------------------------------------------------------------------------
-// K1: y[M,N] = act(x[M,K] @ w[N,K]^T + b[N])      act: 0=none, 1=relu
+// K1: y[g,M,N] = act(x[M,K] @ w[g,N,K]^T + b[g,N])   act: 0=none, 1=relu
+// x is SHARED across groups (twin critics consume the same activations).
+// grid: (ceil(M/64), ceil(N/64), G)
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_linear_act_fwd(
     const float* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ b, float* __restrict__ y,
-    int M, int N, int K, int act) {
+    int M, int N, int K, int act, long xgs) {
   __shared__ float sx[BM][PAD_K];
   __shared__ float sw[BN][PAD_K];
+  const long g = blockIdx.z;
+  x += g * xgs;            // 0 = activations shared across groups
+  w += g * (long)N * K;
+  b += g * (long)N;
+  y += g * (long)M * N;
   const int m0 = blockIdx.x * BM, n0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
@@ -105,15 +120,24 @@ __global__ __launch_bounds__(256) void k_linear_act_fwd(
 }
 
 // ---------------------------------------------------------------------------
-// K8a: dx[M,K] = (dy * mask)[M,N] @ w[N,K]   (mask = yout > 0 when act==1)
+// K8a: dx[M,K] = sum_g (dy*mask)[g,M,N] @ w[g,N,K]  (mask = yout>0 if act)
+// The G-sum is the twin-critic case: both Qs consume the same x, so dL/dx
+// accumulates over groups inside the K-loop (no extra kernel or atomics).
 // grid: (ceil(M/64), ceil(K/64))
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_linear_bwd_dx(
     const float* __restrict__ dy, const float* __restrict__ w,
     const float* __restrict__ yout, float* __restrict__ dx,
-    int M, int N, int K, int act) {
+    int M, int N, int K, int act, int G) {
   __shared__ float sdy[BM][PAD_K];   // [m][n-slice]
   __shared__ float sw[BK][PAD_N];    // [n-slice][k]
+  // per-group mode: grid.z = G_outer with G==1 inside; summed mode:
+  // grid.z == 1 and the g-loop below accumulates over G groups.
+  const long z = blockIdx.z;
+  dy += z * (long)M * N;
+  w += z * (long)N * K;
+  yout += z * (long)M * N;
+  dx += z * (long)M * K;
   const int m0 = blockIdx.x * BM, c0 = blockIdx.y * BN;  // c over K
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
@@ -121,39 +145,44 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
   const int fi = lane & 15, fk = lane >> 4;
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
-  for (int n0 = 0; n0 < N; n0 += BK) {
+  for (int g = 0; g < G; ++g) {
+    const float* dyg = dy + (long)g * M * N;
+    const float* wg = w + (long)g * N * K;
+    const float* yg = yout + (long)g * M * N;
+    for (int n0 = 0; n0 < N; n0 += BK) {
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {  // dy tile: 64x32
-      const int idx = tid * 8 + j;
-      const int r = idx >> 5, c = idx & 31;
-      const int gm = m0 + r, gn = n0 + c;
-      float v = 0.f;
-      if (gm < M && gn < N) {
-        v = dy[(long)gm * N + gn];
-        if (act == 1 && yout[(long)gm * N + gn] <= 0.f) v = 0.f;
+      for (int j = 0; j < 8; ++j) {  // dy tile: 64x32
+        const int idx = tid * 8 + j;
+        const int r = idx >> 5, c = idx & 31;
+        const int gm = m0 + r, gn = n0 + c;
+        float v = 0.f;
+        if (gm < M && gn < N) {
+          v = dyg[(long)gm * N + gn];
+          if (act == 1 && yg[(long)gm * N + gn] <= 0.f) v = 0.f;
+        }
+        sdy[r][c] = v;
       }
-      sdy[r][c] = v;
-    }
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {  // w tile: 32x64, k fast
-      const int idx = tid * 8 + j;
-      const int r = idx >> 6, c = idx & 63;
-      const int gn = n0 + r, gk = c0 + c;
-      sw[r][c] = (gn < N && gk < K) ? w[(long)gn * K + gk] : 0.f;
-    }
-    __syncthreads();
+      for (int j = 0; j < 8; ++j) {  // w tile: 32x64, k fast
+        const int idx = tid * 8 + j;
+        const int r = idx >> 6, c = idx & 63;
+        const int gn = n0 + r, gk = c0 + c;
+        sw[r][c] = (gn < N && gk < K) ? wg[(long)gn * K + gk] : 0.f;
+      }
+      __syncthreads();
 #pragma unroll
-    for (int kk = 0; kk < BK; kk += 4) {
-      const float a0 = sdy[wr + fi][kk + fk];
-      const float a1 = sdy[wr + 16 + fi][kk + fk];
-      const float b0 = sw[kk + fk][wc + fi];
-      const float b1 = sw[kk + fk][wc + 16 + fi];
-      acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
-      acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01, 0, 0, 0);
-      acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
-      acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
+      for (int kk = 0; kk < BK; kk += 4) {
+        const float a0 = sdy[wr + fi][kk + fk];
+        const float a1 = sdy[wr + 16 + fi][kk + fk];
+        const float b0 = sw[kk + fk][wc + fi];
+        const float b1 = sw[kk + fk][wc + 16 + fi];
+        acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
+      }
+      __syncthreads();
     }
-    __syncthreads();
   }
   const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
 #pragma unroll
@@ -171,15 +200,29 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
 }
 
 // ---------------------------------------------------------------------------
-// K8b: dw[N,K] = (dy*mask)^T[N,M] @ x[M,K];  db[N] = sum_m (dy*mask)[m][n]
-// grid: (ceil(N/64), ceil(K/64)); blocks with blockIdx.y==0 also produce db.
+// K8b: dw[g,N,K] = (dy*mask)^T @ x per group, SPLIT-K over the batch:
+// grid.z = G*S; slice s covers batch rows [s*chunk, (s+1)*chunk).  Partials
+// land in ws[g*S+s][N][K] / ws_db[g*S+s][N]; k_reduce_partials folds S.
+// This turns the 49-block (or 7-block, for head layers) serial-1280-batch
+// kernel that dominated the baseline profile (104 us, 46% of GPU time —
+// profiles/r01_baseline_NOTES.md) into a chip-filling grid.
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256) void k_linear_bwd_dwdb(
+__global__ __launch_bounds__(256) void k_linear_bwd_dwdb_splitk(
     const float* __restrict__ dy, const float* __restrict__ x,
-    const float* __restrict__ yout, float* __restrict__ dw,
-    float* __restrict__ db, int M, int N, int K, int act) {
+    const float* __restrict__ yout, float* __restrict__ ws,
+    float* __restrict__ ws_db, int M, int N, int K, int act, int S,
+    int chunk, long xgs) {
   __shared__ float sa[BN][PAD_K];   // dy^T tile: [n][m-slice]
   __shared__ float sb[BK][PAD_N];   // x tile:    [m-slice][k]
+  const int gs = blockIdx.z;        // g*S + s
+  const int g = gs / S, s = gs % S;
+  const float* dyg = dy + (long)g * M * N;
+  const float* yg = yout + (long)g * M * N;
+  const float* xg = x + (long)g * xgs;
+  float* wsp = ws + (long)gs * N * K;
+  float* dbp = ws_db + (long)gs * N;
+  const int m_lo = s * chunk;
+  const int m_hi = min(M, m_lo + chunk);
   const int n0 = blockIdx.x * BM, c0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
@@ -189,16 +232,16 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dwdb(
   float db_acc = 0.f;
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
-  for (int m0 = 0; m0 < M; m0 += BK) {
+  for (int m0 = m_lo; m0 < m_hi; m0 += BK) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {  // dy block 32(m) x 64(n), transposed store
       const int idx = tid * 8 + j;
       const int n = idx & 63, m = idx >> 6;
       const int gm = m0 + m, gn = n0 + n;
       float v = 0.f;
-      if (gm < M && gn < N) {
-        v = dy[(long)gm * N + gn];
-        if (act == 1 && yout[(long)gm * N + gn] <= 0.f) v = 0.f;
+      if (gm < m_hi && gn < N) {
+        v = dyg[(long)gm * N + gn];
+        if (act == 1 && yg[(long)gm * N + gn] <= 0.f) v = 0.f;
       }
       sa[n][m] = v;
     }
@@ -207,7 +250,7 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dwdb(
       const int idx = tid * 8 + j;
       const int m = idx >> 6, c = idx & 63;
       const int gm = m0 + m, gk = c0 + c;
-      sb[m][c] = (gm < M && gk < K) ? x[(long)gm * K + gk] : 0.f;
+      sb[m][c] = (gm < m_hi && gk < K) ? xg[(long)gm * K + gk] : 0.f;
     }
     __syncthreads();
     if (do_db && tid < BN) {
@@ -227,7 +270,7 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dwdb(
     }
     __syncthreads();
   }
-  if (do_db && tid < BN && n0 + tid < N) db[n0 + tid] = db_acc;
+  if (do_db && tid < BN && n0 + tid < N) dbp[n0 + tid] = db_acc;
   const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
 #pragma unroll
   for (int mi = 0; mi < 2; ++mi)
@@ -238,18 +281,56 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dwdb(
       for (int r = 0; r < 4; ++r) {
         const int row = n0 + wr + mi * 16 + fk * 4 + r;
         const int col = c0 + wc + ni * 16 + fi;
-        if (row < N && col < K) dw[(long)row * K + col] = a[r];
+        if (row < N && col < K) wsp[(long)row * K + col] = a[r];
       }
     }
+}
+
+// fold S split-K partials: out[g][i] = sum_s ws[(g*S+s)*stride + i]
+__global__ __launch_bounds__(256) void k_reduce_partials(
+    const float* __restrict__ ws, float* __restrict__ out, long stride,
+    int S, long n_per_g) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int g = blockIdx.y;
+  if (i >= n_per_g) return;
+  const float* base = ws + ((long)g * S) * stride + i;
+  float acc = 0.f;
+  for (int s = 0; s < S; ++s) acc += base[(long)s * stride];
+  out[(long)g * n_per_g + i] = acc;
+}
+
+// ---------------------------------------------------------------------------
+// K12: stratified replay sample — ONE kernel replaces the per-shard
+// index_select/cat chains (~65 torch kernels per update in the baseline
+// profile).  Fields are stacked [T, cap, D]; output row i draws from shard
+// t = i / (B/T) at index floor(rand[i] * size[t]).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_replay_sample(
+    const float* __restrict__ states, const float* __restrict__ actions,
+    const float* __restrict__ rewards, const float* __restrict__ next_states,
+    const float* __restrict__ dones, const float* __restrict__ sizes,
+    const float* __restrict__ rnd, float* __restrict__ o_states,
+    float* __restrict__ o_actions, float* __restrict__ o_rewards,
+    float* __restrict__ o_next_states, float* __restrict__ o_dones,
+    int B, int T, int per, long cap, int Ds, int Da) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B) return;
+  const int t = min(i / per, T - 1);
+  const long idx = (long)(rnd[i] * sizes[t]);
+  const long src = (long)t * cap + idx;
+  for (int j = 0; j < Ds; ++j) {
+    o_states[(long)i * Ds + j] = states[src * Ds + j];
+    o_next_states[(long)i * Ds + j] = next_states[src * Ds + j];
+  }
+  for (int j = 0; j < Da; ++j)
+    o_actions[(long)i * Da + j] = actions[src * Da + j];
+  o_rewards[i] = rewards[src];
+  o_dones[i] = dones[src];
 }
 
 // ---------------------------------------------------------------------------
 // K4: fused tanh-squashed Gaussian sample + log-prob (fwd + bwd).
 // One thread per batch row; A = action_dim <= 32.
-//   ls = clamp(lsr, -20, 2); s = exp(ls); u = mu + s*eps; t = tanh(u)
-//   a = k * t
-//   logp = sum_i [ -0.5 eps_i^2 - ls_i - 0.5 log(2pi)
-//                  - log(k (1 - t_i^2 + 1e-6)) ]
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_squash_fwd(
     const float* __restrict__ mu, const float* __restrict__ lsr,
@@ -275,9 +356,6 @@ __global__ __launch_bounds__(256) void k_squash_fwd(
   logp[i] = lp;
 }
 
-// bwd: dmu_i = ga_i*k*(1-t^2) + gl*2t(1-t^2)/(1-t^2+1e-6)
-//      dlsr_i = mask * [ ga_i*k*(1-t^2)*e*s
-//                        + gl*(-1 + 2t(1-t^2)/(1-t^2+1e-6)*e*s) ]
 __global__ __launch_bounds__(256) void k_squash_bwd(
     const float* __restrict__ ga, const float* __restrict__ gl,
     const float* __restrict__ lsr, const float* __restrict__ ls,
@@ -294,8 +372,7 @@ __global__ __launch_bounds__(256) void k_squash_bwd(
     const float dlp_du = 2.f * t * omt2 / (omt2 + 1e-6f);
     const float s = __expf(ls[idx]);
     const float e = eps[idx];
-    const float gai = ga[idx];
-    const float du = gai * k * omt2 + g * dlp_du;   // dL/du
+    const float du = ga[idx] * k * omt2 + g * dlp_du;   // dL/du
     dmu[idx] = du;
     const float raw = lsr[idx];
     const float mask = (raw >= -20.f && raw <= 2.f) ? 1.f : 0.f;
@@ -304,7 +381,9 @@ __global__ __launch_bounds__(256) void k_squash_bwd(
 }
 
 // ---------------------------------------------------------------------------
-// K5: y = rs*r + gamma*(1-d)*(min(q1,q2) - alpha*lp)
+// K5: y = rs*r + gamma*(1-d)*(min(q1,q2) - alpha*lp).
+// alpha comes per-sample from log_alpha[t_i] (one_hots suffix of mtobs) —
+// no separate gather matmul (reference learner.get_log_alpha).
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_td_target(
     const float* __restrict__ r, const float* __restrict__ d,
@@ -314,6 +393,191 @@ __global__ __launch_bounds__(256) void k_td_target(
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   y[i] = rs * r[i] + gamma * (1.f - d[i]) * (fminf(q1[i], q2[i]) - alpha[i] * lp[i]);
+}
+
+// K5b: MT variant — per-sample alpha computed in-kernel from
+// log_alpha[t_i] (one-hot suffix), removing the gather matmul
+// (reference learner.get_log_alpha, MT10…MTSAC/src/learner.py:213-233).
+__global__ __launch_bounds__(256) void k_td_target_mt(
+    const float* __restrict__ r, const float* __restrict__ d,
+    const float* __restrict__ q1, const float* __restrict__ q2,
+    const float* __restrict__ lp, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, float* __restrict__ y,
+    int n, int T, int oh_stride, float gamma, float rs) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  int t_i = 0;
+  if (T > 1) {
+    float best = -1e30f;
+    for (int t = 0; t < T; ++t) {
+      const float v = onehot[(long)i * oh_stride + t];
+      if (v > best) { best = v; t_i = t; }
+    }
+  }
+  const float alpha = __expf(log_alpha[t_i]);
+  y[i] = rs * r[i] + gamma * (1.f - d[i]) * (fminf(q1[i], q2[i]) - alpha * lp[i]);
+}
+
+// ---------------------------------------------------------------------------
+// K6: fused SAC loss reductions.  Single 256-thread workgroup loops the
+// batch; LDS tree reductions.  T <= 32 tasks.
+//
+// Task machinery (matches reference MT10_Distributed_MTSAC/src/model.py:
+// 99-116 and learner.py:213-233 exactly):
+//   t_i      = argmax over one-hot suffix of states rows
+//   alpha_i  = exp(log_alpha[t_i])           (per-sample temperature)
+//   w_raw_i  = softmax(-exp(log_alpha))[t_i] (weighted-loss path)
+//   coeff_i  = use_w ? (w_raw_i/sum w_raw)/B : 1/B
+//
+// The critic loss runs BEFORE the critic Adam step and the actor/alpha
+// losses AFTER it (reference update_SAC ordering), so they are separate
+// kernel pairs.
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int task_of_row(const float* __restrict__ onehot,
+                                           long i, int oh_stride, int T) {
+  if (T <= 1) return 0;
+  int t_i = 0;
+  float best = -1e30f;
+  for (int t = 0; t < T; ++t) {
+    const float v = onehot[i * oh_stride + t];
+    if (v > best) { best = v; t_i = t; }
+  }
+  return t_i;
+}
+
+// softmax(-exp(log_alpha)) gathered at task t (recomputed per thread; T<=32)
+__device__ __forceinline__ float task_weight(const float* __restrict__ la,
+                                             int T, int t_i) {
+  float mx = -1e30f;
+  for (int t = 0; t < T; ++t) mx = fmaxf(mx, -__expf(la[t]));
+  float den = 0.f;
+  for (int t = 0; t < T; ++t) den += __expf(-__expf(la[t]) - mx);
+  return __expf(-__expf(la[t_i]) - mx) / den;
+}
+
+// out[0]=loss1 out[1]=loss2 out[2]=wsum
+__global__ __launch_bounds__(256) void k_critic_loss_fwd(
+    const float* __restrict__ q1, const float* __restrict__ q2,
+    const float* __restrict__ y, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, float* __restrict__ out,
+    int B, int T, int oh_stride, int use_w) {
+  __shared__ float red[3][256];
+  const int tid = threadIdx.x;
+  float s_l1 = 0.f, s_l2 = 0.f, s_w = 0.f;
+  for (int i = tid; i < B; i += 256) {
+    const int t_i = task_of_row(onehot, i, oh_stride, T);
+    const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+    s_w += w_raw;
+    const float d1 = y[i] - q1[i], d2 = y[i] - q2[i];
+    s_l1 += w_raw * d1 * d1;
+    s_l2 += w_raw * d2 * d2;
+  }
+  red[0][tid] = s_l1; red[1][tid] = s_l2; red[2][tid] = s_w;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (tid < off)
+#pragma unroll
+      for (int r = 0; r < 3; ++r) red[r][tid] += red[r][tid + off];
+    __syncthreads();
+  }
+  if (tid == 0) {
+    const float wsum = use_w ? red[2][0] : 1.f;
+    const float denom = wsum * (float)B;
+    out[0] = red[0][0] / denom;
+    out[1] = red[1][0] / denom;
+    out[2] = wsum;
+  }
+}
+
+// dq1_i = g1 * coeff_i * -2 (y_i - q1_i);  dq2 likewise with g2
+__global__ __launch_bounds__(256) void k_critic_loss_bwd(
+    const float* __restrict__ q1, const float* __restrict__ q2,
+    const float* __restrict__ y, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, const float* __restrict__ saved,
+    const float* __restrict__ gscale, float* __restrict__ dq1,
+    float* __restrict__ dq2, int B, int T, int oh_stride, int use_w) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B) return;
+  const int t_i = task_of_row(onehot, i, oh_stride, T);
+  const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+  const float coeff = (use_w ? w_raw / saved[2] : 1.f) / (float)B;
+  dq1[i] = gscale[0] * coeff * -2.f * (y[i] - q1[i]);
+  dq2[i] = gscale[1] * coeff * -2.f * (y[i] - q2[i]);
+}
+
+// out[0]=actor_loss out[1]=wsum out[2]=alpha_loss out[3]=entropy
+__global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd(
+    const float* __restrict__ aq1, const float* __restrict__ aq2,
+    const float* __restrict__ lp, const float* __restrict__ ls,
+    const float* __restrict__ onehot, const float* __restrict__ log_alpha,
+    float* __restrict__ out, int B, int T, int A, int oh_stride, int use_w,
+    float H_bar) {
+  __shared__ float red[4][256];
+  const int tid = threadIdx.x;
+  constexpr float CE = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
+  float s_pl = 0.f, s_w = 0.f, s_al = 0.f, s_en = 0.f;
+  for (int i = tid; i < B; i += 256) {
+    const int t_i = task_of_row(onehot, i, oh_stride, T);
+    const float la = log_alpha[t_i];
+    const float alpha_i = __expf(la);
+    const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+    s_w += w_raw;
+    const float qmin = fminf(aq1[i], aq2[i]);
+    s_pl += w_raw * -(qmin - alpha_i * lp[i]);
+    s_al += la * (lp[i] + H_bar);
+    float ent = CE * A;
+    for (int a = 0; a < A; ++a) ent += ls[(long)i * A + a];
+    s_en += ent;
+  }
+  red[0][tid] = s_pl; red[1][tid] = s_w; red[2][tid] = s_al;
+  red[3][tid] = s_en;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (tid < off)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
+    __syncthreads();
+  }
+  if (tid == 0) {
+    const float wsum = use_w ? red[1][0] : 1.f;
+    out[0] = red[0][0] / (wsum * (float)B);
+    out[1] = wsum;
+    out[2] = -red[2][0] / (float)B;
+    out[3] = red[3][0] / (float)B;
+  }
+}
+
+// daq1_i = gp*coeff_i*-(aq1<=aq2); daq2_i = gp*coeff_i*-(aq2<aq1)
+// dlp_i  = gp*coeff_i*alpha_i      (alpha detached in actor loss)
+// dla[t] = gal * -(1/B) sum_{i in t}(lp_i + H_bar)   (lp detached)
+__global__ __launch_bounds__(256) void k_actor_alpha_loss_bwd(
+    const float* __restrict__ aq1, const float* __restrict__ aq2,
+    const float* __restrict__ lp, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, const float* __restrict__ saved,
+    const float* __restrict__ gscale, float* __restrict__ daq1,
+    float* __restrict__ daq2, float* __restrict__ dlp,
+    float* __restrict__ dla, int B, int T, int oh_stride, int use_w,
+    float H_bar) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int tid = threadIdx.x;
+  __shared__ float s_dla[32];
+  if (tid < T) s_dla[tid] = 0.f;
+  __syncthreads();
+  if (i < B) {
+    const float gp = gscale[0], gal = gscale[1];
+    const int t_i = task_of_row(onehot, i, oh_stride, T);
+    const float alpha_i = __expf(log_alpha[t_i]);
+    const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+    const float coeff = (use_w ? w_raw / saved[1] : 1.f) / (float)B;
+    const bool first = aq1[i] <= aq2[i];
+    daq1[i] = first ? gp * coeff * -1.f : 0.f;
+    daq2[i] = first ? 0.f : gp * coeff * -1.f;
+    dlp[i] = gp * coeff * alpha_i;
+    atomicAdd(&s_dla[t_i], gal * -(lp[i] + H_bar) / (float)B);
+  }
+  __syncthreads();
+  if (tid < T) atomicAdd(&dla[tid], s_dla[tid]);
 }
 
 // ---------------------------------------------------------------------------
@@ -335,7 +599,6 @@ __global__ __launch_bounds__(256) void k_adam(
 
 // K9b: graph-capturable Adam — the step counter lives on-device so a
 // captured update's bias correction advances across hipGraph replays.
-// k_adam_prolog: step += 1; coeffs = {step_size, inv_sqrt_bc2}.
 __global__ void k_adam_prolog(float* __restrict__ state, float lr, float b1,
                               float b2) {
   const float step = state[0] + 1.f;
@@ -373,53 +636,137 @@ __global__ __launch_bounds__(256) void k_polyak(
 // Host wrappers
 // ===========================================================================
 
-static torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w,
-                                    torch::Tensor b, long act) {
+static torch::Tensor linear_act_fwd_g(torch::Tensor x, torch::Tensor w,
+                                      torch::Tensor b, long act, long G) {
   CHECK_IN(x); CHECK_IN(w); CHECK_IN(b);
   auto xc = x.contiguous(); auto wc = w.contiguous(); auto bc = b.contiguous();
-  const long M = xc.size(0), K = xc.size(1), N = wc.size(0);
-  TORCH_CHECK(wc.size(1) == K, "weight shape mismatch");
-  auto y = torch::empty({M, N}, xc.options());
-  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN);
+  const bool per_group_x = xc.dim() == 3;  // [G,B,K] vs shared [B,K]
+  const long M = per_group_x ? xc.size(1) : xc.size(0);
+  const long K = xc.size(-1);
+  const long N = wc.numel() / (G * K);
+  TORCH_CHECK(wc.numel() == G * N * K && bc.numel() == G * N,
+              "grouped weight shape mismatch");
+  const long xgs = per_group_x ? M * K : 0;
+  auto y = G == 1 ? torch::empty({M, N}, xc.options())
+                  : torch::empty({G, M, N}, xc.options());
+  dim3 grid((M + BM - 1) / BM, (N + BN - 1) / BN, G);
   hipLaunchKernelGGL(k_linear_act_fwd, grid, dim3(256), 0, cur_stream(),
                      xc.data_ptr<float>(), wc.data_ptr<float>(),
                      bc.data_ptr<float>(), y.data_ptr<float>(),
-                     (int)M, (int)N, (int)K, (int)act);
+                     (int)M, (int)N, (int)K, (int)act, xgs);
   return y;
+}
+
+static torch::Tensor linear_act_fwd(torch::Tensor x, torch::Tensor w,
+                                    torch::Tensor b, long act) {
+  return linear_act_fwd_g(x, w, b, act, 1);
+}
+
+static torch::Tensor linear_bwd_dx_g(torch::Tensor dy, torch::Tensor w,
+                                     torch::Tensor yout, long act, long G,
+                                     long sum_over_g) {
+  CHECK_IN(dy); CHECK_IN(w); CHECK_IN(yout);
+  auto dyc = dy.contiguous(); auto wc = w.contiguous();
+  auto yc = yout.contiguous();
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  const long N = G == 1 ? dyc.size(1) : dyc.size(2);
+  const long K = wc.size(-1);
+  // sum mode: one grid, g-loop accumulates (shared-x layer).  per-group
+  // mode: grid.z = G independent dx outputs [G,M,K].
+  const long Gz = (G == 1 || sum_over_g) ? 1 : G;
+  const int Gin = (int)((G > 1 && sum_over_g) ? G : 1);
+  auto dx = Gz == 1 ? torch::empty({M, K}, dyc.options())
+                    : torch::empty({Gz, M, K}, dyc.options());
+  dim3 grid((M + BM - 1) / BM, (K + BN - 1) / BN, Gz);
+  hipLaunchKernelGGL(k_linear_bwd_dx, grid, dim3(256), 0, cur_stream(),
+                     dyc.data_ptr<float>(), wc.data_ptr<float>(),
+                     yc.data_ptr<float>(), dx.data_ptr<float>(),
+                     (int)M, (int)N, (int)K, (int)act, Gin);
+  return dx;
 }
 
 static torch::Tensor linear_bwd_dx(torch::Tensor dy, torch::Tensor w,
                                    torch::Tensor yout, long act) {
-  CHECK_IN(dy); CHECK_IN(w); CHECK_IN(yout);
-  auto dyc = dy.contiguous(); auto wc = w.contiguous();
+  return linear_bwd_dx_g(dy, w, yout, act, 1, 1);
+}
+
+static std::vector<torch::Tensor> linear_bwd_dwdb_g(torch::Tensor dy,
+                                                    torch::Tensor x,
+                                                    torch::Tensor yout,
+                                                    long act, long G) {
+  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(yout);
+  auto dyc = dy.contiguous(); auto xc = x.contiguous();
   auto yc = yout.contiguous();
-  const long M = dyc.size(0), N = dyc.size(1), K = wc.size(1);
-  TORCH_CHECK(wc.size(0) == N, "weight shape mismatch");
-  auto dx = torch::empty({M, K}, dyc.options());
-  dim3 grid((M + BM - 1) / BM, (K + BN - 1) / BN);
-  hipLaunchKernelGGL(k_linear_bwd_dx, grid, dim3(256), 0, cur_stream(),
-                     dyc.data_ptr<float>(), wc.data_ptr<float>(),
-                     yc.data_ptr<float>(), dx.data_ptr<float>(),
-                     (int)M, (int)N, (int)K, (int)act);
-  return dx;
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  const long N = G == 1 ? dyc.size(1) : dyc.size(2);
+  const long K = xc.size(-1);
+  const long xgs = xc.dim() == 3 ? M * K : 0;
+  // split-K: slices of >=128 batch rows, enough blocks to fill 256 CUs
+  long S = (M + 127) / 128;
+  S = std::max<long>(1, std::min<long>(S, 16));
+  const int chunk = (int)(((M + S - 1) / S + BK - 1) / BK * BK);
+  S = (M + chunk - 1) / chunk;
+  auto ws = torch::empty({G * S, N, K}, dyc.options());
+  auto ws_db = torch::empty({G * S, N}, dyc.options());
+  dim3 grid((N + BM - 1) / BM, (K + BN - 1) / BN, G * S);
+  hipLaunchKernelGGL(k_linear_bwd_dwdb_splitk, grid, dim3(256), 0,
+                     cur_stream(), dyc.data_ptr<float>(), xc.data_ptr<float>(),
+                     yc.data_ptr<float>(), ws.data_ptr<float>(),
+                     ws_db.data_ptr<float>(), (int)M, (int)N, (int)K,
+                     (int)act, (int)S, chunk, xgs);
+  auto dw = G == 1 ? torch::empty({N, K}, dyc.options())
+                   : torch::empty({G, N, K}, dyc.options());
+  auto db = G == 1 ? torch::empty({N}, dyc.options())
+                   : torch::empty({G, N}, dyc.options());
+  if (S == 1) {
+    dw.copy_(ws.view_as(dw));
+    db.copy_(ws_db.view_as(db));
+  } else {
+    const long nw = N * K;
+    dim3 g1((nw + 255) / 256, G);
+    hipLaunchKernelGGL(k_reduce_partials, g1, dim3(256), 0, cur_stream(),
+                       ws.data_ptr<float>(), dw.data_ptr<float>(), nw,
+                       (int)S, nw);
+    dim3 g2((N + 255) / 256, G);
+    hipLaunchKernelGGL(k_reduce_partials, g2, dim3(256), 0, cur_stream(),
+                       ws_db.data_ptr<float>(), db.data_ptr<float>(), N,
+                       (int)S, N);
+  }
+  return {dw, db};
 }
 
 static std::vector<torch::Tensor> linear_bwd_dwdb(torch::Tensor dy,
                                                   torch::Tensor x,
                                                   torch::Tensor yout,
                                                   long act) {
-  CHECK_IN(dy); CHECK_IN(x); CHECK_IN(yout);
-  auto dyc = dy.contiguous(); auto xc = x.contiguous();
-  auto yc = yout.contiguous();
-  const long M = dyc.size(0), N = dyc.size(1), K = xc.size(1);
-  auto dw = torch::empty({N, K}, dyc.options());
-  auto db = torch::empty({N}, dyc.options());
-  dim3 grid((N + BM - 1) / BM, (K + BN - 1) / BN);
-  hipLaunchKernelGGL(k_linear_bwd_dwdb, grid, dim3(256), 0, cur_stream(),
-                     dyc.data_ptr<float>(), xc.data_ptr<float>(),
-                     yc.data_ptr<float>(), dw.data_ptr<float>(),
-                     db.data_ptr<float>(), (int)M, (int)N, (int)K, (int)act);
-  return {dw, db};
+  return linear_bwd_dwdb_g(dy, x, yout, act, 1);
+}
+
+static std::vector<torch::Tensor> replay_sample(
+    torch::Tensor states, torch::Tensor actions, torch::Tensor rewards,
+    torch::Tensor next_states, torch::Tensor dones, torch::Tensor sizes,
+    torch::Tensor rnd, long B) {
+  CHECK_IN(states); CHECK_IN(sizes); CHECK_IN(rnd);
+  const long T = states.size(0), cap = states.size(1);
+  const long Ds = states.size(2), Da = actions.size(2);
+  const int per = (int)(B / T);
+  TORCH_CHECK(per * T == B, "batch must divide by num_tasks");
+  auto opt = states.options();
+  auto o_s = torch::empty({B, Ds}, opt);
+  auto o_a = torch::empty({B, Da}, opt);
+  auto o_r = torch::empty({B, 1}, opt);
+  auto o_ns = torch::empty({B, Ds}, opt);
+  auto o_d = torch::empty({B, 1}, opt);
+  hipLaunchKernelGGL(k_replay_sample, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), states.data_ptr<float>(),
+                     actions.data_ptr<float>(), rewards.data_ptr<float>(),
+                     next_states.data_ptr<float>(), dones.data_ptr<float>(),
+                     sizes.data_ptr<float>(), rnd.data_ptr<float>(),
+                     o_s.data_ptr<float>(), o_a.data_ptr<float>(),
+                     o_r.data_ptr<float>(), o_ns.data_ptr<float>(),
+                     o_d.data_ptr<float>(), (int)B, (int)T, per, cap,
+                     (int)Ds, (int)Da);
+  return {o_s, o_a, o_r, o_ns, o_d};
 }
 
 static std::vector<torch::Tensor> squashed_gaussian_fwd(torch::Tensor mu,
@@ -483,6 +830,105 @@ static torch::Tensor td_target(torch::Tensor r, torch::Tensor d,
   return y;
 }
 
+static torch::Tensor td_target_mt(torch::Tensor r, torch::Tensor d,
+                                  torch::Tensor q1, torch::Tensor q2,
+                                  torch::Tensor lp, torch::Tensor states,
+                                  torch::Tensor log_alpha, long T,
+                                  double gamma, double rs) {
+  CHECK_IN(r); CHECK_IN(states); CHECK_IN(log_alpha);
+  const long n = r.numel();
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto y = torch::empty_like(r.contiguous());
+  hipLaunchKernelGGL(k_td_target_mt, dim3((n + 255) / 256), dim3(256), 0,
+                     cur_stream(), r.contiguous().data_ptr<float>(),
+                     d.contiguous().data_ptr<float>(),
+                     q1.contiguous().data_ptr<float>(),
+                     q2.contiguous().data_ptr<float>(),
+                     lp.contiguous().data_ptr<float>(), oh,
+                     log_alpha.data_ptr<float>(), y.data_ptr<float>(),
+                     (int)n, (int)T, (int)oh_stride, (float)gamma,
+                     (float)rs);
+  return y;
+}
+
+static std::vector<torch::Tensor> critic_loss_fwd(
+    torch::Tensor q1, torch::Tensor q2, torch::Tensor y,
+    torch::Tensor states, torch::Tensor log_alpha, long T, long use_w) {
+  CHECK_IN(q1); CHECK_IN(states); CHECK_IN(log_alpha);
+  const long B = q1.size(0);
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto out = torch::empty({3}, q1.options());
+  hipLaunchKernelGGL(k_critic_loss_fwd, dim3(1), dim3(256), 0, cur_stream(),
+                     q1.data_ptr<float>(), q2.data_ptr<float>(),
+                     y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
+                     out.data_ptr<float>(), (int)B, (int)T, (int)oh_stride,
+                     (int)use_w);
+  return {out};
+}
+
+static std::vector<torch::Tensor> critic_loss_bwd(
+    torch::Tensor q1, torch::Tensor q2, torch::Tensor y,
+    torch::Tensor states, torch::Tensor log_alpha, torch::Tensor saved,
+    torch::Tensor gscale, long T, long use_w) {
+  CHECK_IN(q1);
+  const long B = q1.size(0);
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto dq1 = torch::empty_like(q1);
+  auto dq2 = torch::empty_like(q2);
+  hipLaunchKernelGGL(k_critic_loss_bwd, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), q1.data_ptr<float>(), q2.data_ptr<float>(),
+                     y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
+                     saved.data_ptr<float>(), gscale.data_ptr<float>(),
+                     dq1.data_ptr<float>(), dq2.data_ptr<float>(), (int)B,
+                     (int)T, (int)oh_stride, (int)use_w);
+  return {dq1, dq2};
+}
+
+static torch::Tensor actor_alpha_loss_fwd(
+    torch::Tensor aq1, torch::Tensor aq2, torch::Tensor lp, torch::Tensor ls,
+    torch::Tensor states, torch::Tensor log_alpha, long T, long use_w,
+    double H_bar) {
+  CHECK_IN(aq1); CHECK_IN(states); CHECK_IN(log_alpha);
+  const long B = aq1.size(0);
+  const long A = ls.size(1);
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto out = torch::empty({4}, aq1.options());
+  hipLaunchKernelGGL(k_actor_alpha_loss_fwd, dim3(1), dim3(256), 0,
+                     cur_stream(), aq1.data_ptr<float>(),
+                     aq2.data_ptr<float>(), lp.data_ptr<float>(),
+                     ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
+                     out.data_ptr<float>(), (int)B, (int)T, (int)A,
+                     (int)oh_stride, (int)use_w, (float)H_bar);
+  return out;
+}
+
+static std::vector<torch::Tensor> actor_alpha_loss_bwd(
+    torch::Tensor aq1, torch::Tensor aq2, torch::Tensor lp,
+    torch::Tensor states, torch::Tensor log_alpha, torch::Tensor saved,
+    torch::Tensor gscale, long T, long use_w, double H_bar) {
+  CHECK_IN(aq1);
+  const long B = aq1.size(0);
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto daq1 = torch::empty_like(aq1);
+  auto daq2 = torch::empty_like(aq2);
+  auto dlp = torch::empty_like(lp);
+  auto dla = torch::zeros_like(log_alpha);
+  hipLaunchKernelGGL(k_actor_alpha_loss_bwd, dim3((B + 255) / 256), dim3(256),
+                     0, cur_stream(), aq1.data_ptr<float>(),
+                     aq2.data_ptr<float>(), lp.data_ptr<float>(), oh,
+                     log_alpha.data_ptr<float>(), saved.data_ptr<float>(),
+                     gscale.data_ptr<float>(), daq1.data_ptr<float>(),
+                     daq2.data_ptr<float>(), dlp.data_ptr<float>(),
+                     dla.data_ptr<float>(), (int)B, (int)T, (int)oh_stride,
+                     (int)use_w, (float)H_bar);
+  return {daq1, daq2, dlp, dla};
+}
+
 static void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        torch::Tensor v, long step, double lr, double b1,
                        double b2, double eps) {
@@ -526,11 +972,20 @@ static void polyak_(torch::Tensor t, torch::Tensor s, double tau) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("linear_act_fwd", &linear_act_fwd, "fused GEMM+bias+act forward");
+  mod.def("linear_act_fwd_g", &linear_act_fwd_g, "grouped (twin) variant");
   mod.def("linear_bwd_dx", &linear_bwd_dx, "GEMM backward dX (fused mask)");
-  mod.def("linear_bwd_dwdb", &linear_bwd_dwdb, "GEMM backward dW+db");
+  mod.def("linear_bwd_dx_g", &linear_bwd_dx_g, "grouped, sums over G");
+  mod.def("linear_bwd_dwdb", &linear_bwd_dwdb, "split-K backward dW+db");
+  mod.def("linear_bwd_dwdb_g", &linear_bwd_dwdb_g, "grouped split-K dW+db");
+  mod.def("replay_sample", &replay_sample, "stratified replay gather");
   mod.def("squashed_gaussian_fwd", &squashed_gaussian_fwd);
   mod.def("squashed_gaussian_bwd", &squashed_gaussian_bwd);
   mod.def("td_target", &td_target);
+  mod.def("td_target_mt", &td_target_mt);
+  mod.def("critic_loss_fwd", &critic_loss_fwd);
+  mod.def("critic_loss_bwd", &critic_loss_bwd);
+  mod.def("actor_alpha_loss_fwd", &actor_alpha_loss_fwd);
+  mod.def("actor_alpha_loss_bwd", &actor_alpha_loss_bwd);
   mod.def("adam_step_", &adam_step_);
   mod.def("adam_step_dev_", &adam_step_dev_);
   mod.def("polyak_", &polyak_);

@@ -123,6 +123,175 @@ def squashed_gaussian(mu: torch.Tensor, log_std_raw: torch.Tensor,
 
 
 # ---------------------------------------------------------------------------
+# Fused twin-MLP (grouped GEMMs: both Q networks in one launch per layer,
+# stacked weights [2,N,K] — views into the critic's flat parameter buffer)
+# ---------------------------------------------------------------------------
+
+class _FusedTwinMLP(torch.autograd.Function):
+    """Twin-Q MLP: q1,q2 from shared x through stacked weights [2,N,K].
+
+    Layer 0 consumes the shared x (group-stride 0); later layers run on
+    per-group activations [2,B,N].  Backward mirrors this: per-group dX for
+    hidden layers, group-SUMMED dX at layer 0 (both Qs consume the same x —
+    reference Critic.forward computes Q1(x), Q2(x) on one cat([s,a])).
+    """
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, n_layers: int, *wb):
+        ws = wb[:n_layers]      # each [2, N, K]
+        bs = wb[n_layers:]      # each [2, N]
+        ext = native()
+        acts = [x]
+        h = x
+        for i in range(n_layers):
+            act = ACT_RELU if i < n_layers - 1 else ACT_NONE
+            h = ext.linear_act_fwd_g(h, ws[i], bs[i], act, 2)
+            acts.append(h)
+        ctx.save_for_backward(*acts, *ws)
+        ctx.n_layers = n_layers
+        return h[0], h[1]
+
+    @staticmethod
+    def backward(ctx, dq1: torch.Tensor, dq2: torch.Tensor):
+        n = ctx.n_layers
+        saved = ctx.saved_tensors
+        acts = saved[: n + 1]
+        ws = saved[n + 1:]
+        ext = native()
+        dy = torch.stack([dq1, dq2], dim=0).contiguous()  # [2,B,Nout]
+        dws = [None] * n
+        dbs = [None] * n
+        for i in range(n - 1, -1, -1):
+            act = ACT_RELU if i < n - 1 else ACT_NONE
+            dw, db = ext.linear_bwd_dwdb_g(dy, acts[i], acts[i + 1], act, 2)
+            dws[i], dbs[i] = dw, db
+            if i > 0:
+                dy = ext.linear_bwd_dx_g(dy, ws[i], acts[i + 1], act, 2, 0)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.linear_bwd_dx_g(dy, ws[0], acts[1],
+                                     ACT_RELU if n > 1 else ACT_NONE, 2, 1)
+        return (dx, None, *dws, *dbs)
+
+
+def twin_mlp_forward(x: torch.Tensor, stacked_ws, stacked_bs):
+    """Both Q heads in one grouped launch per layer.  stacked_ws[i] is a
+    [2,N,K] view into the critic flat parameter buffer."""
+    return _FusedTwinMLP.apply(x, len(stacked_ws), *stacked_ws, *stacked_bs)
+
+
+# ---------------------------------------------------------------------------
+# Fused SAC loss heads (reference math in kernel K6; one graph node per loss
+# instead of ~15 torch glue kernels each).
+# ---------------------------------------------------------------------------
+
+class _CriticLoss(torch.autograd.Function):
+    """(loss1, loss2) = (optionally task-weighted) MSE vs TD target.
+
+    Matches reference Critic.cal_loss (MT10…MTSAC/src/model.py:157-196):
+    w_i = softmax(-exp(log_alpha))[t_i] renormalized; loss = mean(w*l).
+    """
+
+    @staticmethod
+    def forward(ctx, q1, q2, y, states, log_alpha_det, num_tasks: int,
+                use_weighted: bool):
+        ext = native()
+        out = ext.critic_loss_fwd(q1.contiguous(), q2.contiguous(),
+                                  y.contiguous(), states, log_alpha_det,
+                                  num_tasks, int(use_weighted))[0]
+        ctx.save_for_backward(q1, q2, y, states, log_alpha_det, out)
+        ctx.meta = (num_tasks, use_weighted)
+        return out[0], out[1]
+
+    @staticmethod
+    def backward(ctx, g1, g2):
+        q1, q2, y, states, la, saved = ctx.saved_tensors
+        T, use_w = ctx.meta
+        gscale = torch.stack([g1.reshape(()), g2.reshape(())]).contiguous()
+        dq1, dq2 = native().critic_loss_bwd(
+            q1.contiguous(), q2.contiguous(), y.contiguous(), states, la,
+            saved, gscale, T, int(use_w))
+        return dq1, dq2, None, None, None, None, None
+
+
+def critic_loss(q1, q2, y, states, log_alpha_det, num_tasks, use_weighted):
+    if _use_native(q1):
+        return _CriticLoss.apply(q1, q2, y, states, log_alpha_det,
+                                 num_tasks, use_weighted)
+    l1 = (y - q1) ** 2
+    l2 = (y - q2) ** 2
+    if use_weighted:
+        w = torch_ref.task_weights(states[:, -num_tasks:],
+                                   log_alpha_det.exp()).unsqueeze(-1)
+        l1, l2 = w * l1, w * l2
+    return l1.mean(), l2.mean()
+
+
+class _ActorAlphaLoss(torch.autograd.Function):
+    """(actor_loss, alpha_loss, entropy) fused.
+
+    actor_loss = mean/weighted of -(min(q1,q2) - alpha_i * logp)
+    alpha_loss = -mean(log_alpha[t_i] * (logp.detach() + H_bar))
+    entropy    = mean(0.5*A*(1+log 2pi) + sum log_std)   [diagnostic]
+
+    Gradients: actor_loss -> q1,q2,logp; alpha_loss -> log_alpha.
+    (alpha is detached in the actor loss and logp in the alpha loss —
+    reference learner.py:316, model.py:80-116.)
+    """
+
+    @staticmethod
+    def forward(ctx, q1, q2, logp, log_stds, states, log_alpha,
+                num_tasks: int, use_weighted: bool, H_bar: float):
+        ext = native()
+        out = ext.actor_alpha_loss_fwd(
+            q1.contiguous(), q2.contiguous(), logp.contiguous(),
+            log_stds.contiguous(), states, log_alpha.detach(), num_tasks,
+            int(use_weighted), H_bar)
+        ctx.save_for_backward(q1, q2, logp, states, log_alpha.detach(), out)
+        ctx.meta = (num_tasks, use_weighted, H_bar)
+        actor_l, alpha_l, entropy = out[0], out[2], out[3]
+        ctx.mark_non_differentiable(entropy)
+        return actor_l, alpha_l, entropy
+
+    @staticmethod
+    def backward(ctx, gp, gal, _gent):
+        q1, q2, logp, states, la, saved = ctx.saved_tensors
+        T, use_w, H_bar = ctx.meta
+        gscale = torch.stack([gp.reshape(()), gal.reshape(())]).contiguous()
+        daq1, daq2, dlp, dla = native().actor_alpha_loss_bwd(
+            q1.contiguous(), q2.contiguous(), logp.contiguous(), states, la,
+            saved, gscale, T, int(use_w), H_bar)
+        return daq1, daq2, dlp, None, None, dla, None, None, None
+
+
+def actor_alpha_loss(q1, q2, logp, log_stds, states, log_alpha, num_tasks,
+                     use_weighted, H_bar):
+    """Returns (actor_loss, alpha_loss, entropy)."""
+    if _use_native(q1):
+        return _ActorAlphaLoss.apply(q1, q2, logp, log_stds, states,
+                                     log_alpha, num_tasks, use_weighted,
+                                     float(H_bar))
+    qmin = torch.min(q1, q2)
+    if num_tasks > 1:
+        one_hots = states[:, -num_tasks:]
+        alpha = torch_ref.gather_log_alpha(one_hots,
+                                           log_alpha).exp().detach()
+        la_g = torch_ref.gather_log_alpha(one_hots, log_alpha)
+    else:
+        alpha = log_alpha.exp().detach()
+        la_g = log_alpha
+    pl = -(qmin - alpha * logp)
+    if use_weighted:
+        w = torch_ref.task_weights(states[:, -num_tasks:],
+                                   log_alpha.exp().detach()).unsqueeze(-1)
+        pl = w * pl
+    actor_loss = pl.mean()
+    alpha_loss = -(la_g * (logp.detach() + H_bar)).mean()
+    entropy = torch_ref.entropy_from_log_std(log_stds)
+    return actor_loss, alpha_loss, entropy
+
+
+# ---------------------------------------------------------------------------
 # Non-differentiable / glue ops
 # ---------------------------------------------------------------------------
 

@@ -34,25 +34,36 @@ FIELDS = ("states", "actions", "rewards", "next_states", "dones")
 
 
 class ReplayShard:
-    """One SoA ring buffer (one task shard)."""
+    """One SoA ring buffer (one task shard).
+
+    Field tensors may be externally provided views (ShardedReplay stacks all
+    shards as [T, cap, D] so the fused sampling kernel gathers across tasks
+    in one launch).
+    """
 
     def __init__(self, capacity: int, state_dim: int, action_dim: int,
-                 device: torch.device | str = "cpu"):
+                 device: torch.device | str = "cpu", buffers=None,
+                 size_dev: Optional[torch.Tensor] = None):
         self.capacity = int(capacity)
         self.device = torch.device(device)
         self.state_dim = state_dim
         self.action_dim = action_dim
         dev = self.device
-        self.states = torch.zeros(self.capacity, state_dim, device=dev)
-        self.actions = torch.zeros(self.capacity, action_dim, device=dev)
-        self.rewards = torch.zeros(self.capacity, 1, device=dev)
-        self.next_states = torch.zeros(self.capacity, state_dim, device=dev)
-        self.dones = torch.zeros(self.capacity, 1, device=dev)
+        if buffers is not None:
+            (self.states, self.actions, self.rewards, self.next_states,
+             self.dones) = buffers
+        else:
+            self.states = torch.zeros(self.capacity, state_dim, device=dev)
+            self.actions = torch.zeros(self.capacity, action_dim, device=dev)
+            self.rewards = torch.zeros(self.capacity, 1, device=dev)
+            self.next_states = torch.zeros(self.capacity, state_dim, device=dev)
+            self.dones = torch.zeros(self.capacity, 1, device=dev)
         self.write_ptr = 0
         self.size = 0
         # device-resident size for hipGraph-safe sampling (updated by
         # append OUTSIDE any captured region; read inside the graph)
-        self.size_dev = torch.zeros(1, device=dev)
+        self.size_dev = (size_dev if size_dev is not None
+                         else torch.zeros(1, device=dev))
 
     def __len__(self) -> int:
         return self.size
@@ -123,9 +134,24 @@ class ShardedReplay:
         self.num_tasks = num_tasks
         self.device = torch.device(device)
         per_task = int(buffer_size) // num_tasks  # reference replay_buffers.py:37-41
+        dev = self.device
+        # stacked [T, cap, D] field storage; shards are views into it so the
+        # fused k_replay_sample kernel gathers the whole stratified batch in
+        # one launch.
+        self.f_states = torch.zeros(num_tasks, per_task, state_dim, device=dev)
+        self.f_actions = torch.zeros(num_tasks, per_task, action_dim, device=dev)
+        self.f_rewards = torch.zeros(num_tasks, per_task, 1, device=dev)
+        self.f_next_states = torch.zeros(num_tasks, per_task, state_dim,
+                                         device=dev)
+        self.f_dones = torch.zeros(num_tasks, per_task, 1, device=dev)
+        self.sizes_dev = torch.zeros(num_tasks, device=dev)
         self.shards: List[ReplayShard] = [
-            ReplayShard(per_task, state_dim, action_dim, device)
-            for _ in range(num_tasks)]
+            ReplayShard(per_task, state_dim, action_dim, device,
+                        buffers=(self.f_states[t], self.f_actions[t],
+                                 self.f_rewards[t], self.f_next_states[t],
+                                 self.f_dones[t]),
+                        size_dev=self.sizes_dev[t:t + 1])
+            for t in range(num_tasks)]
         self.generator = None
         if seed is not None:
             self.generator = torch.Generator(device=self.device)
@@ -181,6 +207,16 @@ class ShardedReplay:
         the concat shuffle (every consumer of the batch is
         permutation-invariant: all losses are batch means)."""
         gen = None if graph_safe else self.generator
+        if graph_safe and self.device.type == "cuda":
+            from ..ops import has_native, native, native_enabled
+            if native_enabled() and has_native() \
+                    and batch_size % self.num_tasks == 0:
+                rnd = torch.rand(batch_size, device=self.device)
+                o = native().replay_sample(
+                    self.f_states, self.f_actions, self.f_rewards,
+                    self.f_next_states, self.f_dones, self.sizes_dev, rnd,
+                    batch_size)
+                return dict(zip(FIELDS, o))
         if self.num_tasks == 1:
             return self.shards[0].sample(batch_size, gen, graph_safe)
         per = batch_size // self.num_tasks

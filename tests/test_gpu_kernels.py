@@ -247,3 +247,182 @@ def test_graph_captured_update_matches_eager():
         assert val == val, f"NaN metric {k}"
     for p in (engine.actor_group.flat_data, engine.critic_group.flat_data):
         assert torch.isfinite(p).all()
+
+
+def test_grouped_twin_fwd_matches_single():
+    ext = req_native()
+    torch.manual_seed(10)
+    B, N, K = 1280, 400, 53
+    x = torch.randn(B, K, device="cuda")
+    w = torch.randn(2, N, K, device="cuda") / K ** 0.5
+    b = torch.randn(2, N, device="cuda")
+    y = ext.linear_act_fwd_g(x, w, b, 1, 2)
+    assert y.shape == (2, B, N)
+    for g in range(2):
+        ref = torch.relu(torch.nn.functional.linear(x, w[g], b[g]))
+        assert torch.allclose(y[g], ref, atol=1e-4, rtol=1e-4)
+    # per-group x
+    x2 = torch.randn(2, B, N, device="cuda")
+    w2 = torch.randn(2, 64, N, device="cuda") / N ** 0.5
+    b2 = torch.randn(2, 64, device="cuda")
+    y2 = ext.linear_act_fwd_g(x2, w2, b2, 0, 2)
+    for g in range(2):
+        ref = torch.nn.functional.linear(x2[g], w2[g], b2[g])
+        assert torch.allclose(y2[g], ref, atol=1e-4, rtol=1e-4)
+
+
+def test_twin_mlp_autograd():
+    """Stacked twin-MLP fwd+bwd vs two independent torch MLPs."""
+    torch.manual_seed(11)
+    from distributed_sac_amd.ops.functional import twin_mlp_forward
+    B, K, H = 512, 53, 128
+    dims = [(H, K), (H, H), (1, H)]
+    ws = [torch.randn(2, n, k, device="cuda").div_(k ** 0.5).requires_grad_(True)
+          for n, k in dims]
+    bs = [torch.randn(2, n, device="cuda").mul_(0.1).requires_grad_(True)
+          for n, _ in dims]
+    x = torch.randn(B, K, device="cuda", requires_grad=True)
+    q1, q2 = twin_mlp_forward(x, ws, bs)
+    loss = (q1 ** 2).mean() + (q2 * 3).mean()
+    loss.backward()
+
+    x2 = x.detach().clone().requires_grad_(True)
+    ws2 = [w.detach().clone().requires_grad_(True) for w in ws]
+    bs2 = [b.detach().clone().requires_grad_(True) for b in bs]
+    outs = []
+    for g in range(2):
+        h = x2
+        for i in range(3):
+            h = torch.nn.functional.linear(h, ws2[i][g], bs2[i][g])
+            if i < 2:
+                h = torch.relu(h)
+        outs.append(h)
+    assert torch.allclose(q1, outs[0], atol=1e-4, rtol=1e-4)
+    assert torch.allclose(q2, outs[1], atol=1e-4, rtol=1e-4)
+    loss2 = (outs[0] ** 2).mean() + (outs[1] * 3).mean()
+    loss2.backward()
+    assert torch.allclose(x.grad, x2.grad, atol=1e-4, rtol=1e-3)
+    for i in range(3):
+        assert torch.allclose(ws[i].grad, ws2[i].grad, atol=1e-4, rtol=1e-3), i
+        assert torch.allclose(bs[i].grad, bs2[i].grad, atol=1e-4, rtol=1e-3), i
+
+
+@pytest.mark.parametrize("use_w,T", [(0, 1), (0, 10), (1, 10)])
+def test_fused_critic_loss(use_w, T):
+    torch.manual_seed(12)
+    from distributed_sac_amd.ops.functional import critic_loss
+    from distributed_sac_amd.ops import torch_ref as TR
+    B, D = 1280, 39 + T
+    states = torch.randn(B, D, device="cuda")
+    idx = torch.randint(0, T, (B,), device="cuda")
+    states[:, -T:] = torch.nn.functional.one_hot(idx, T).float()
+    q1 = torch.randn(B, 1, device="cuda", requires_grad=True)
+    q2 = torch.randn(B, 1, device="cuda", requires_grad=True)
+    y = torch.randn(B, 1, device="cuda")
+    la = torch.randn(T, device="cuda") * 0.3
+    l1, l2 = critic_loss(q1, q2, y, states, la, T, bool(use_w))
+    (l1 + l2).backward()
+
+    q1r = q1.detach().clone().requires_grad_(True)
+    q2r = q2.detach().clone().requires_grad_(True)
+    rl1 = (y - q1r) ** 2
+    rl2 = (y - q2r) ** 2
+    if use_w:
+        w = TR.task_weights(states[:, -T:], la.exp()).unsqueeze(-1)
+        rl1, rl2 = w * rl1, w * rl2
+    rl1, rl2 = rl1.mean(), rl2.mean()
+    assert torch.allclose(l1, rl1, atol=1e-5, rtol=1e-4)
+    assert torch.allclose(l2, rl2, atol=1e-5, rtol=1e-4)
+    (rl1 + rl2).backward()
+    assert torch.allclose(q1.grad, q1r.grad, atol=1e-6, rtol=1e-4)
+    assert torch.allclose(q2.grad, q2r.grad, atol=1e-6, rtol=1e-4)
+
+
+@pytest.mark.parametrize("use_w,T", [(0, 1), (1, 10)])
+def test_fused_actor_alpha_loss(use_w, T):
+    torch.manual_seed(13)
+    from distributed_sac_amd.ops.functional import actor_alpha_loss
+    from distributed_sac_amd.ops import torch_ref as TR
+    B, A, D = 1280, 4, 39 + T
+    H_bar = -float(A)
+    states = torch.randn(B, D, device="cuda")
+    idx = torch.randint(0, T, (B,), device="cuda")
+    states[:, -T:] = torch.nn.functional.one_hot(idx, T).float()
+    q1 = torch.randn(B, 1, device="cuda", requires_grad=True)
+    q2 = torch.randn(B, 1, device="cuda", requires_grad=True)
+    lp = torch.randn(B, 1, device="cuda", requires_grad=True)
+    ls = torch.randn(B, A, device="cuda")
+    la = (torch.randn(T, device="cuda") * 0.3).requires_grad_(True)
+
+    al, all_, ent = actor_alpha_loss(q1, q2, lp, ls, states, la, T,
+                                     bool(use_w), H_bar)
+    (al + all_).backward()
+
+    q1r = q1.detach().clone().requires_grad_(True)
+    q2r = q2.detach().clone().requires_grad_(True)
+    lpr = lp.detach().clone().requires_grad_(True)
+    lar = la.detach().clone().requires_grad_(True)
+    if T > 1:
+        oh = states[:, -T:]
+        alpha = (oh @ lar.unsqueeze(0).t()).exp().detach()
+        la_g = oh @ lar.unsqueeze(0).t()
+    else:
+        alpha = lar.exp().detach()
+        la_g = lar
+    pl = -(torch.min(q1r, q2r) - alpha * lpr)
+    if use_w:
+        w = TR.task_weights(states[:, -T:], lar.exp().detach()).unsqueeze(-1)
+        pl = w * pl
+    rel = pl.mean()
+    rela = -(la_g * (lpr.detach() + H_bar)).mean()
+    rent = TR.entropy_from_log_std(ls)
+    assert torch.allclose(al, rel, atol=1e-5, rtol=1e-4)
+    assert torch.allclose(all_, rela, atol=1e-5, rtol=1e-4)
+    assert torch.allclose(ent, rent, atol=1e-4, rtol=1e-4)
+    (rel + rela).backward()
+    assert torch.allclose(q1.grad, q1r.grad, atol=1e-6, rtol=1e-4)
+    assert torch.allclose(q2.grad, q2r.grad, atol=1e-6, rtol=1e-4)
+    assert torch.allclose(lp.grad, lpr.grad, atol=1e-6, rtol=1e-4)
+    assert torch.allclose(la.grad, lar.grad, atol=1e-5, rtol=1e-4)
+
+
+def test_fused_replay_sample():
+    from distributed_sac_amd.replay import ShardedReplay
+    torch.manual_seed(14)
+    T, B = 10, 1280
+    r = ShardedReplay(100000, T, 49, 4, device="cuda")
+    for t in range(T):
+        n = 500 + 37 * t
+        st = torch.randn(n, 49, device="cuda")
+        st[:, 40:] = 0.0
+        st[:, 39 + t] = 1.0  # tag rows by task
+        r.shards[t].append(st, torch.randn(n, 4, device="cuda"),
+                           torch.full((n, 1), float(t), device="cuda"),
+                           st.clone(), torch.zeros(n, 1, device="cuda"))
+    out = r.sample(B, graph_safe=True)
+    assert out["states"].shape == (B, 49)
+    per = B // T
+    for t in range(T):
+        seg = out["rewards"][t * per:(t + 1) * per]
+        assert (seg == float(t)).all(), f"shard {t} mis-sampled"
+        oh = out["states"][t * per:(t + 1) * per, 39:]
+        assert (oh.argmax(dim=1) == t).all()
+
+
+def test_td_target_mt_kernel():
+    ext = req_native()
+    torch.manual_seed(15)
+    B, T = 1280, 10
+    states = torch.randn(B, 49, device="cuda")
+    idx = torch.randint(0, T, (B,), device="cuda")
+    states[:, -T:] = torch.nn.functional.one_hot(idx, T).float()
+    r = torch.randn(B, 1, device="cuda")
+    d = (torch.rand(B, 1, device="cuda") < 0.3).float()
+    q1 = torch.randn(B, 1, device="cuda")
+    q2 = torch.randn(B, 1, device="cuda")
+    lp = torch.randn(B, 1, device="cuda")
+    la = torch.randn(T, device="cuda") * 0.5
+    y = ext.td_target_mt(r, d, q1, q2, lp, states, la, T, 0.99, 1.5)
+    alpha = la.exp()[idx].unsqueeze(-1)
+    ref = 1.5 * r + 0.99 * (1 - d) * (torch.min(q1, q2) - alpha * lp)
+    assert torch.allclose(y, ref, atol=1e-5, rtol=1e-5)
