@@ -98,6 +98,7 @@ class SACEngine:
         else:
             raise ValueError(f"variant {self.variant} not handled here")
         self.H_bar = torch.tensor([-float(cfg.action_dim)], device=dev)
+        self.H_bar_f = -float(cfg.action_dim)  # python scalar: graph-safe
         self.alpha = self.log_alpha.exp().detach()
 
     @staticmethod
@@ -351,7 +352,7 @@ class SACEngine:
         aq1, aq2 = self._critic_q(states, sampled_actions)
         policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(
             aq1, aq2, log_probs, log_stds, states, self.log_alpha, T, use_w,
-            float(self.H_bar))
+            self.H_bar_f)
         (policy_loss + loss_log_alpha).backward()
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.actor_group.flat_grad)
