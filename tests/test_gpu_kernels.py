@@ -394,7 +394,7 @@ def test_fused_replay_sample():
     for t in range(T):
         n = 500 + 37 * t
         st = torch.randn(n, 49, device="cuda")
-        st[:, 40:] = 0.0
+        st[:, 39:] = 0.0
         st[:, 39 + t] = 1.0  # tag rows by task
         r.shards[t].append(st, torch.randn(n, 4, device="cuda"),
                            torch.full((n, 1), float(t), device="cuda"),
