@@ -62,8 +62,11 @@ __global__ __launch_bounds__(256) void k_linear_act_fwd(
     const float* __restrict__ x, const float* __restrict__ w,
     const float* __restrict__ b, float* __restrict__ y,
     int M, int N, int K, int act, long xgs) {
-  __shared__ float sx[BM][PAD_K];
-  __shared__ float sw[BN][PAD_K];
+  // double-buffered: global loads for tile t+1 issue BEFORE the MFMAs of
+  // tile t (HBM latency hides under compute — guide T14); ds_writes into
+  // the other buffer, ONE barrier per K-tile.
+  __shared__ float sx[2][BM][PAD_K];
+  __shared__ float sw[2][BN][PAD_K];
   const long g = blockIdx.z;
   x += g * xgs;            // 0 = activations shared across groups
   w += g * (long)N * K;
@@ -76,28 +79,40 @@ __global__ __launch_bounds__(256) void k_linear_act_fwd(
   const int fi = lane & 15, fk = lane >> 4;  // fragment row / k index
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
+  const int lr = (tid * 8) >> 5, lc = (tid * 8) & 31;  // this thread's slot
+  float rx[8], rw[8];
+#define LOAD_TILE_FWD(k0)                                                   \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
+    const int gk = (k0) + lc + j;                                           \
+    rx[j] = (m0 + lr < M && gk < K) ? x[(long)(m0 + lr) * K + gk] : 0.f;    \
+    rw[j] = (n0 + lr < N && gk < K) ? w[(long)(n0 + lr) * K + gk] : 0.f;    \
+  }
+#define STORE_TILE_FWD(buf)                                                 \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
+    sx[buf][lr][lc + j] = rx[j];                                            \
+    sw[buf][lr][lc + j] = rw[j];                                            \
+  }
+  LOAD_TILE_FWD(0)
+  STORE_TILE_FWD(0)
+  __syncthreads();
+  int cur = 0;
   for (int k0 = 0; k0 < K; k0 += BK) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {  // 2048 elements / 256 threads
-      const int idx = tid * 8 + j;
-      const int r = idx >> 5, c = idx & 31;
-      const int gk = k0 + c;
-      sx[r][c] = (m0 + r < M && gk < K) ? x[(long)(m0 + r) * K + gk] : 0.f;
-      sw[r][c] = (n0 + r < N && gk < K) ? w[(long)(n0 + r) * K + gk] : 0.f;
-    }
-    __syncthreads();
+    const bool more = k0 + BK < K;
+    if (more) LOAD_TILE_FWD(k0 + BK)
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 4) {
-      const float a0 = sx[wr + fi][kk + fk];
-      const float a1 = sx[wr + 16 + fi][kk + fk];
-      const float b0 = sw[wc + fi][kk + fk];
-      const float b1 = sw[wc + 16 + fi][kk + fk];
+      const float a0 = sx[cur][wr + fi][kk + fk];
+      const float a1 = sx[cur][wr + 16 + fi][kk + fk];
+      const float b0 = sw[cur][wc + fi][kk + fk];
+      const float b1 = sw[cur][wc + 16 + fi][kk + fk];
       acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
       acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01, 0, 0, 0);
       acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
       acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
     }
+    if (more) STORE_TILE_FWD(cur ^ 1)
     __syncthreads();
+    cur ^= 1;
   }
   // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + reg
   const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
@@ -129,10 +144,10 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
     const float* __restrict__ dy, const float* __restrict__ w,
     const float* __restrict__ yout, float* __restrict__ dx,
     int M, int N, int K, int act, int G) {
-  __shared__ float sdy[BM][PAD_K];   // [m][n-slice]
-  __shared__ float sw[BK][PAD_N];    // [n-slice][k]
+  __shared__ float sdy[2][BM][PAD_K];   // [m][n-slice]
+  __shared__ float sw[2][BK][PAD_N];     // [n-slice][k]
   // per-group mode: grid.z = G_outer with G==1 inside; summed mode:
-  // grid.z == 1 and the g-loop below accumulates over G groups.
+  // grid.z == 1 and the flattened tile loop accumulates over G groups.
   const long z = blockIdx.z;
   dy += z * (long)M * N;
   w += z * (long)N * K;
@@ -145,44 +160,56 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
   const int fi = lane & 15, fk = lane >> 4;
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
-  for (int g = 0; g < G; ++g) {
-    const float* dyg = dy + (long)g * M * N;
-    const float* wg = w + (long)g * N * K;
-    const float* yg = yout + (long)g * M * N;
-    for (int n0 = 0; n0 < N; n0 += BK) {
+  const int nt_per_g = (N + BK - 1) / BK;
+  const int nt = G * nt_per_g;
+  const int ar = (tid * 8) >> 5, ac = (tid * 8) & 31;  // dy slot (64x32)
+  const int br = (tid * 8) >> 6, bc = (tid * 8) & 63;  // w slot (32x64)
+  float rdy[8], rw[8];
+#define LOAD_TILE_DX(ti)                                                    \
+  {                                                                         \
+    const int g_ = (ti) / nt_per_g;                                         \
+    const int n0_ = ((ti) % nt_per_g) * BK;                                 \
+    const float* dyg = dy + (long)g_ * M * N;                               \
+    const float* wg = w + (long)g_ * N * K;                                 \
+    const float* yg = yout + (long)g_ * M * N;                              \
+    _Pragma("unroll") for (int j = 0; j < 8; ++j) {                         \
+      const int gm = m0 + ar, gn = n0_ + ac + j;                            \
+      float v = 0.f;                                                        \
+      if (gm < M && gn < N) {                                               \
+        v = dyg[(long)gm * N + gn];                                         \
+        if (act == 1 && yg[(long)gm * N + gn] <= 0.f) v = 0.f;              \
+      }                                                                     \
+      rdy[j] = v;                                                           \
+      const int wn = n0_ + br, wk = c0 + bc + j;                            \
+      rw[j] = (wn < N && wk < K) ? wg[(long)wn * K + wk] : 0.f;             \
+    }                                                                       \
+  }
+#define STORE_TILE_DX(buf)                                                  \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
+    sdy[buf][ar][ac + j] = rdy[j];                                          \
+    sw[buf][br][bc + j] = rw[j];                                            \
+  }
+  LOAD_TILE_DX(0)
+  STORE_TILE_DX(0)
+  __syncthreads();
+  int cur = 0;
+  for (int ti = 0; ti < nt; ++ti) {
+    const bool more = ti + 1 < nt;
+    if (more) LOAD_TILE_DX(ti + 1)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {  // dy tile: 64x32
-        const int idx = tid * 8 + j;
-        const int r = idx >> 5, c = idx & 31;
-        const int gm = m0 + r, gn = n0 + c;
-        float v = 0.f;
-        if (gm < M && gn < N) {
-          v = dyg[(long)gm * N + gn];
-          if (act == 1 && yg[(long)gm * N + gn] <= 0.f) v = 0.f;
-        }
-        sdy[r][c] = v;
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {  // w tile: 32x64, k fast
-        const int idx = tid * 8 + j;
-        const int r = idx >> 6, c = idx & 63;
-        const int gn = n0 + r, gk = c0 + c;
-        sw[r][c] = (gn < N && gk < K) ? wg[(long)gn * K + gk] : 0.f;
-      }
-      __syncthreads();
-#pragma unroll
-      for (int kk = 0; kk < BK; kk += 4) {
-        const float a0 = sdy[wr + fi][kk + fk];
-        const float a1 = sdy[wr + 16 + fi][kk + fk];
-        const float b0 = sw[kk + fk][wc + fi];
-        const float b1 = sw[kk + fk][wc + 16 + fi];
-        acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
-        acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01, 0, 0, 0);
-        acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
-        acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
-      }
-      __syncthreads();
+    for (int kk = 0; kk < BK; kk += 4) {
+      const float a0 = sdy[cur][wr + fi][kk + fk];
+      const float a1 = sdy[cur][wr + 16 + fi][kk + fk];
+      const float b0 = sw[cur][kk + fk][wc + fi];
+      const float b1 = sw[cur][kk + fk][wc + 16 + fi];
+      acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
     }
+    if (more) STORE_TILE_DX(cur ^ 1)
+    __syncthreads();
+    cur ^= 1;
   }
   const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
 #pragma unroll
@@ -212,8 +239,8 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dwdb_splitk(
     const float* __restrict__ yout, float* __restrict__ ws,
     float* __restrict__ ws_db, int M, int N, int K, int act, int S,
     int chunk, long xgs) {
-  __shared__ float sa[BN][PAD_K];   // dy^T tile: [n][m-slice]
-  __shared__ float sb[BK][PAD_N];   // x tile:    [m-slice][k]
+  __shared__ float sa[2][BN][PAD_K];   // dy^T tile: [n][m-slice]
+  __shared__ float sb[2][BK][PAD_N];    // x tile:    [m-slice][k]
   const int gs = blockIdx.z;        // g*S + s
   const int g = gs / S, s = gs % S;
   const float* dyg = dy + (long)g * M * N;
@@ -232,43 +259,51 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dwdb_splitk(
   float db_acc = 0.f;
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
+  const int an = (tid * 8) & 63, am = (tid * 8) >> 6;  // dy slot (n fast)
+  const int bm = (tid * 8) >> 6, bc = (tid * 8) & 63;  // x slot (k fast)
+  float rdy[8], rx[8];
+#define LOAD_TILE_DW(m0_)                                                   \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
+    const int gm = (m0_) + am, gn = n0 + an + j;                            \
+    float v = 0.f;                                                          \
+    if (gm < m_hi && gn < N) {                                              \
+      v = dyg[(long)gm * N + gn];                                           \
+      if (act == 1 && yg[(long)gm * N + gn] <= 0.f) v = 0.f;                \
+    }                                                                       \
+    rdy[j] = v;                                                             \
+    const int xm = (m0_) + bm, xk = c0 + bc + j;                            \
+    rx[j] = (xm < m_hi && xk < K) ? xg[(long)xm * K + xk] : 0.f;            \
+  }
+#define STORE_TILE_DW(buf)                                                  \
+  _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
+    sa[buf][an + j][am] = rdy[j];                                           \
+    sb[buf][bm][bc + j] = rx[j];                                            \
+  }
+  LOAD_TILE_DW(m_lo)
+  STORE_TILE_DW(0)
+  __syncthreads();
+  int cur = 0;
   for (int m0 = m_lo; m0 < m_hi; m0 += BK) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {  // dy block 32(m) x 64(n), transposed store
-      const int idx = tid * 8 + j;
-      const int n = idx & 63, m = idx >> 6;
-      const int gm = m0 + m, gn = n0 + n;
-      float v = 0.f;
-      if (gm < m_hi && gn < N) {
-        v = dyg[(long)gm * N + gn];
-        if (act == 1 && yg[(long)gm * N + gn] <= 0.f) v = 0.f;
-      }
-      sa[n][m] = v;
-    }
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {  // x tile 32(m) x 64(k)
-      const int idx = tid * 8 + j;
-      const int m = idx >> 6, c = idx & 63;
-      const int gm = m0 + m, gk = c0 + c;
-      sb[m][c] = (gm < m_hi && gk < K) ? xg[(long)gm * K + gk] : 0.f;
-    }
-    __syncthreads();
+    const bool more = m0 + BK < m_hi;
+    if (more) LOAD_TILE_DW(m0 + BK)
     if (do_db && tid < BN) {
 #pragma unroll
-      for (int m = 0; m < BK; ++m) db_acc += sa[tid][m];
+      for (int m = 0; m < BK; ++m) db_acc += sa[cur][tid][m];
     }
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 4) {
-      const float a0 = sa[wr + fi][kk + fk];
-      const float a1 = sa[wr + 16 + fi][kk + fk];
-      const float b0 = sb[kk + fk][wc + fi];
-      const float b1 = sb[kk + fk][wc + 16 + fi];
+      const float a0 = sa[cur][wr + fi][kk + fk];
+      const float a1 = sa[cur][wr + 16 + fi][kk + fk];
+      const float b0 = sb[cur][kk + fk][wc + fi];
+      const float b1 = sb[cur][kk + fk][wc + 16 + fi];
       acc00 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc00, 0, 0, 0);
       acc01 = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc01, 0, 0, 0);
       acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
       acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
     }
+    if (more) STORE_TILE_DW(cur ^ 1)
     __syncthreads();
+    cur ^= 1;
   }
   if (do_db && tid < BN && n0 + tid < N) dbp[n0 + tid] = db_acc;
   const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
@@ -313,19 +348,24 @@ __global__ __launch_bounds__(256) void k_replay_sample(
     float* __restrict__ o_actions, float* __restrict__ o_rewards,
     float* __restrict__ o_next_states, float* __restrict__ o_dones,
     int B, int T, int per, long cap, int Ds, int Da) {
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  // one wave per batch row: lanes stride the row's columns (coalesced on
+  // both the gathered source row and the packed destination row)
+  const int i = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
   if (i >= B) return;
   const int t = min(i / per, T - 1);
   const long idx = (long)(rnd[i] * sizes[t]);
   const long src = (long)t * cap + idx;
-  for (int j = 0; j < Ds; ++j) {
+  for (int j = lane; j < Ds; j += 64) {
     o_states[(long)i * Ds + j] = states[src * Ds + j];
     o_next_states[(long)i * Ds + j] = next_states[src * Ds + j];
   }
-  for (int j = 0; j < Da; ++j)
+  for (int j = lane; j < Da; j += 64)
     o_actions[(long)i * Da + j] = actions[src * Da + j];
-  o_rewards[i] = rewards[src];
-  o_dones[i] = dones[src];
+  if (lane == 0) {
+    o_rewards[i] = rewards[src];
+    o_dones[i] = dones[src];
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -456,6 +496,23 @@ __device__ __forceinline__ float task_weight(const float* __restrict__ la,
   return __expf(-__expf(la[t_i]) - mx) / den;
 }
 
+
+// thread 0 fills smw[0..T) with softmax(-exp(log_alpha)); call before use,
+// followed by __syncthreads().
+__device__ __forceinline__ void fill_task_weights(
+    float* __restrict__ smw, const float* __restrict__ la, int T) {
+  if (threadIdx.x == 0) {
+    float mx = -1e30f;
+    for (int t = 0; t < T; ++t) mx = fmaxf(mx, -__expf(la[t]));
+    float den = 0.f;
+    for (int t = 0; t < T; ++t) {
+      smw[t] = __expf(-__expf(la[t]) - mx);
+      den += smw[t];
+    }
+    for (int t = 0; t < T; ++t) smw[t] /= den;
+  }
+}
+
 // out[0]=loss1 out[1]=loss2 out[2]=wsum
 __global__ __launch_bounds__(256) void k_critic_loss_fwd(
     const float* __restrict__ q1, const float* __restrict__ q2,
@@ -463,11 +520,14 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd(
     const float* __restrict__ log_alpha, float* __restrict__ out,
     int B, int T, int oh_stride, int use_w) {
   __shared__ float red[3][256];
+  __shared__ float smw[32];
   const int tid = threadIdx.x;
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  __syncthreads();
   float s_l1 = 0.f, s_l2 = 0.f, s_w = 0.f;
   for (int i = tid; i < B; i += 256) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
-    const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+    const float w_raw = use_w ? smw[t_i] : 1.f;
     s_w += w_raw;
     const float d1 = y[i] - q1[i], d2 = y[i] - q2[i];
     s_l1 += w_raw * d1 * d1;
@@ -497,10 +557,13 @@ __global__ __launch_bounds__(256) void k_critic_loss_bwd(
     const float* __restrict__ log_alpha, const float* __restrict__ saved,
     const float* __restrict__ gscale, float* __restrict__ dq1,
     float* __restrict__ dq2, int B, int T, int oh_stride, int use_w) {
+  __shared__ float smw[32];
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  __syncthreads();
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= B) return;
   const int t_i = task_of_row(onehot, i, oh_stride, T);
-  const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+  const float w_raw = use_w ? smw[t_i] : 1.f;
   const float coeff = (use_w ? w_raw / saved[2] : 1.f) / (float)B;
   dq1[i] = gscale[0] * coeff * -2.f * (y[i] - q1[i]);
   dq2[i] = gscale[1] * coeff * -2.f * (y[i] - q2[i]);
@@ -514,14 +577,17 @@ __global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd(
     float* __restrict__ out, int B, int T, int A, int oh_stride, int use_w,
     float H_bar) {
   __shared__ float red[4][256];
+  __shared__ float smw[32];
   const int tid = threadIdx.x;
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  __syncthreads();
   constexpr float CE = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
   float s_pl = 0.f, s_w = 0.f, s_al = 0.f, s_en = 0.f;
   for (int i = tid; i < B; i += 256) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float la = log_alpha[t_i];
     const float alpha_i = __expf(la);
-    const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+    const float w_raw = use_w ? smw[t_i] : 1.f;
     s_w += w_raw;
     const float qmin = fminf(aq1[i], aq2[i]);
     s_pl += w_raw * -(qmin - alpha_i * lp[i]);
@@ -562,13 +628,15 @@ __global__ __launch_bounds__(256) void k_actor_alpha_loss_bwd(
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int tid = threadIdx.x;
   __shared__ float s_dla[32];
+  __shared__ float smw[32];
   if (tid < T) s_dla[tid] = 0.f;
+  if (use_w) fill_task_weights(smw, log_alpha, T);
   __syncthreads();
   if (i < B) {
     const float gp = gscale[0], gal = gscale[1];
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float alpha_i = __expf(log_alpha[t_i]);
-    const float w_raw = use_w ? task_weight(log_alpha, T, t_i) : 1.f;
+    const float w_raw = use_w ? smw[t_i] : 1.f;
     const float coeff = (use_w ? w_raw / saved[1] : 1.f) / (float)B;
     const bool first = aq1[i] <= aq2[i];
     daq1[i] = first ? gp * coeff * -1.f : 0.f;
@@ -757,7 +825,7 @@ static std::vector<torch::Tensor> replay_sample(
   auto o_r = torch::empty({B, 1}, opt);
   auto o_ns = torch::empty({B, Ds}, opt);
   auto o_d = torch::empty({B, 1}, opt);
-  hipLaunchKernelGGL(k_replay_sample, dim3((B + 255) / 256), dim3(256), 0,
+  hipLaunchKernelGGL(k_replay_sample, dim3((B + 3) / 4), dim3(256), 0,
                      cur_stream(), states.data_ptr<float>(),
                      actions.data_ptr<float>(), rewards.data_ptr<float>(),
                      next_states.data_ptr<float>(), dones.data_ptr<float>(),

@@ -65,8 +65,9 @@ class _FusedMLP(torch.autograd.Function):
         for i in range(n - 1, -1, -1):
             # mask==1 only for hidden layers (their saved act is post-ReLU)
             act = ACT_RELU if i < n - 1 else ACT_NONE
-            dw, db = ext.linear_bwd_dwdb(dy, acts[i], acts[i + 1], act)
-            dws[i], dbs[i] = dw, db
+            if ctx.needs_input_grad[2 + i]:
+                dw, db = ext.linear_bwd_dwdb(dy, acts[i], acts[i + 1], act)
+                dws[i], dbs[i] = dw, db
             if i > 0:
                 dy = ext.linear_bwd_dx(dy, ws[i], acts[i + 1], act)
         dx = ext.linear_bwd_dx(dy, ws[0], acts[1],
@@ -163,8 +164,10 @@ class _FusedTwinMLP(torch.autograd.Function):
         dbs = [None] * n
         for i in range(n - 1, -1, -1):
             act = ACT_RELU if i < n - 1 else ACT_NONE
-            dw, db = ext.linear_bwd_dwdb_g(dy, acts[i], acts[i + 1], act, 2)
-            dws[i], dbs[i] = dw, db
+            if ctx.needs_input_grad[2 + i]:
+                dw, db = ext.linear_bwd_dwdb_g(dy, acts[i], acts[i + 1],
+                                               act, 2)
+                dws[i], dbs[i] = dw, db
             if i > 0:
                 dy = ext.linear_bwd_dx_g(dy, ws[i], acts[i + 1], act, 2, 0)
         dx = None
