@@ -1,0 +1,170 @@
+"""Learner process — replay ingest + SAC updates + weight publishing.
+
+Re-implements the reference Learner process (LunarLander_Distributed_SAC/
+src/learner.py:21-316, MT10_Distributed_MTSAC/src/learner.py) without
+Redis: transitions arrive as blocks on a multiprocessing queue and go
+straight into the HBM-resident sharded replay; weights publish through the
+shared-memory seqlock snapshot (one D2H copy per publish, no pickling).
+
+Reference semantics kept:
+- wait until ``len(replay) >= start_memory_len`` before training
+  (learner.wait_until_memoryReady);
+- hard copy critics -> targets at train start (soft_update tau=1.0);
+- ``update_delay`` iteration thinning: the published update_iteration
+  counter advances every loop but a gradient update runs only when
+  ``it % update_delay == 0`` (learner.run:293-295) — so published
+  iteration = grad_steps * update_delay;
+- checkpoint every ``save_period`` real updates.
+"""
+
+from __future__ import annotations
+
+import queue as pyqueue
+import time
+from typing import Dict, Optional
+
+import torch
+
+from ..algo.sac import SACEngine
+from ..checkpoint import save_checkpoint
+from ..config import SACConfig
+from ..replay import ShardedReplay
+from ..utils import MetricLogger, StepTimer
+from .param_server import ParamSnapshot
+
+
+class Learner:
+    def __init__(self, cfg: SACConfig, device: str, snapshot: ParamSnapshot,
+                 sample_queue, log_queue=None,
+                 logger: Optional[MetricLogger] = None,
+                 save_dir: Optional[str] = None, save_period: int = 0,
+                 update_delay: Optional[int] = None, use_graph: bool = True,
+                 ddp=None, seed: int = 0):
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.engine = SACEngine(cfg, device)
+        if ddp is not None:
+            self.engine.attach_ddp(ddp)
+        num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
+        self.replay = ShardedReplay(cfg.buffer_size, num_tasks,
+                                    cfg.mtobs_dim, cfg.action_dim,
+                                    device=device, seed=seed)
+        self.snapshot = snapshot
+        self.sample_queue = sample_queue
+        self.log_queue = log_queue
+        self.logger = logger or MetricLogger(None)
+        self.save_dir = save_dir
+        self.save_period = save_period
+        self.update_delay = (update_delay if update_delay is not None
+                             else max(1, cfg.update_delay))
+        self.use_graph = use_graph and self.device.type == "cuda"
+        self._graph_ready = False
+        self.iteration_counter = 0   # reference update_iteration (thinned)
+        self.grad_steps = 0
+        self.update_timer = StepTimer()
+        self.ingest_count = 0
+
+    # -- ingest --------------------------------------------------------
+    def drain_queue(self, max_blocks: int = 64) -> int:
+        n = 0
+        for _ in range(max_blocks):
+            try:
+                _pid, task, blk = self.sample_queue.get_nowait()
+            except (pyqueue.Empty, OSError):
+                break
+            self.replay.append_numpy(task_idx=task, **blk)
+            got = blk["states"].shape[0]
+            self.engine.total_step += got
+            self.ingest_count += got
+            n += 1
+        return n
+
+    def drain_logs(self) -> None:
+        if self.log_queue is None:
+            return
+        for _ in range(256):
+            try:
+                item = self.log_queue.get_nowait()
+            except (pyqueue.Empty, OSError):
+                return
+            kind, pid, task, step, value = item
+            self.logger.add_scalar(f"{kind}/task_{task}_player_{pid}",
+                                   value, step)
+
+    # -- update --------------------------------------------------------
+    def publish(self) -> None:
+        self.snapshot.publish(self.engine.actor_group.flat_data,
+                              self.iteration_counter)
+
+    def ready(self) -> bool:
+        return len(self.replay) * max(1, self.replay.num_tasks) >= \
+            self.cfg.start_memory_len and len(self.replay) > 0
+
+    def _ensure_graph(self) -> None:
+        if self.use_graph and not self._graph_ready:
+            try:
+                self.engine.capture(self.replay, self.cfg.batch_size)
+                self._graph_ready = True
+            except Exception as e:  # pragma: no cover
+                self.logger.print(f"hipGraph capture failed ({e!r}); eager")
+                self.use_graph = False
+
+    def train_step(self) -> Optional[Dict[str, float]]:
+        """One reference loop iteration: counter++, update when thinned
+        counter fires, publish."""
+        self.iteration_counter += 1
+        metrics = None
+        if self.iteration_counter % self.update_delay == 0:
+            if self.use_graph:
+                self._ensure_graph()
+            if self._graph_ready:
+                out = self.engine.graphed_update()
+                metrics_t = out
+            else:
+                metrics_t = self.engine.update_tensors(
+                    self.replay.sample(self.cfg.batch_size,
+                                       graph_safe=self.device.type == "cuda"))
+                self.engine.update_iteration += 1
+            self.grad_steps += 1
+            self.update_timer.mark()
+            self.publish()
+            if self.save_dir and self.save_period and \
+                    self.grad_steps % self.save_period == 0:
+                save_checkpoint(self.engine, self.save_dir,
+                                update_iteration=self.iteration_counter)
+            if self.grad_steps % 100 == 0:
+                metrics = {k: float(v) for k, v in metrics_t.items()}
+                self.logger.add_scalars("learner", metrics, self.grad_steps)
+        return metrics
+
+    # -- main loop -----------------------------------------------------
+    def run(self, stop_event=None, max_grad_steps: Optional[int] = None,
+            max_seconds: Optional[float] = None) -> Dict[str, float]:
+        self.publish()  # initial weights (reference learner.run start)
+        t0 = time.perf_counter()
+        while not self.ready():
+            if stop_event is not None and stop_event.is_set():
+                return {}
+            self.drain_queue()
+            self.drain_logs()
+            time.sleep(0.01)
+            if max_seconds and time.perf_counter() - t0 > max_seconds:
+                return {}
+        self.engine.hard_copy_targets()
+        self.logger.print("######### Start train #########")
+        while True:
+            if stop_event is not None and stop_event.is_set():
+                break
+            self.drain_queue()
+            self.drain_logs()
+            self.train_step()
+            if max_grad_steps and self.grad_steps >= max_grad_steps:
+                break
+            if max_seconds and time.perf_counter() - t0 > max_seconds:
+                break
+        return {
+            "grad_steps": self.grad_steps,
+            "iterations": self.iteration_counter,
+            "ingested": self.ingest_count,
+            "grad_steps_per_sec": self.update_timer.rate(),
+        }
