@@ -1,0 +1,358 @@
+"""CAREEngine — the CARE / CARE(M) learner update.
+
+Implements the reference CARE learner exactly (MT10_Distributed_CARE/src/
+learner.py:281-404, MT1_Distributed_CARE/src/learner.py):
+
+- z_context computed once per update from the context encoder (:290);
+- critic loss backward carries the ONLY gradients into the context encoder
+  (actor uses z_context.detach(), :326-335), applied by a separate Adam
+  after the SAC update (:399);
+- the actor optimizes only its mu/log_std head (:143-149) and receives the
+  critic's state encoder by hard copy every update (tau=1.0 tie, :402);
+- detach_z_encs=True in the actor step stops gradients into the mixture
+  (:326-335);
+- target Q heads Polyak at tau, target state encoder at state_encoder_tau
+  (:361-367; hardcoded 0.05 in MT1: MT1…/src/learner.py:311);
+- CARE(M) (use_modified_care): frozen raw 768-d embeddings, projection in
+  stateEncoder.mlp_context, weighted losses.
+
+MI355X notes: the critic flat buffer is ordered [state-encoder params |
+interleaved Q-head pairs], so the two Polyak taus are two fused kernels
+over flat slices, the actor-SE tie is one flat copy, and the Q heads run
+as grouped twin GEMMs.  In the GPU path, forward passes whose gradients
+the reference computes but discards (actor-side trunk, critic-side encoder
+in the actor step) run under no_grad — numerically identical updates.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+
+from ..config import SACConfig
+from ..models.care import CAREActor, CARECritic
+from ..models.context_encoder import contextEncoder
+from ..ops import functional as Fops
+from ..ops.flat import FlatParams, FusedAdam
+from .sac import SACEngine
+
+
+class CAREEngine(SACEngine):
+    def __init__(self, cfg: SACConfig, device="cpu"):
+        assert cfg.variant == "care" and cfg.encoder is not None
+        super().__init__(cfg, device)
+
+    # ------------------------------------------------------------------
+    def _build_models(self) -> None:
+        cfg, dev = self.cfg, self.device
+        enc_cfg = dict(cfg.encoder)
+        enc_cfg.setdefault("RoBERTa_embedding_dim", 768)
+        self.enc_cfg = enc_cfg
+        self.use_modified_care = cfg.use_modified_care
+        self.se_tau = float(enc_cfg.get("state_encoder_tau",
+                                        cfg.state_encoder_tau))
+        actor_cfg = {
+            "state_dim": cfg.state_dim, "action_dim": cfg.action_dim,
+            "action_bound": cfg.action_bound,
+            "actor_hidden_dim": cfg.actor_hidden_dim,
+        }
+        critic_cfg = {
+            "state_dim": cfg.state_dim, "action_dim": cfg.action_dim,
+            "critic_hidden_dim": cfg.critic_hidden_dim,
+        }
+        self.context_encoder = contextEncoder(enc_cfg,
+                                              self.use_modified_care).to(dev)
+        self.actor = CAREActor(actor_cfg, enc_cfg, self.use_modified_care).to(dev)
+        self.local_critic = CARECritic(critic_cfg, enc_cfg,
+                                       self.use_modified_care).to(dev)
+        self.target_critic = CARECritic(critic_cfg, enc_cfg,
+                                        self.use_modified_care).to(dev)
+        self.log_alpha = nn.Parameter(torch.full(
+            (cfg.num_tasks,), float(cfg.log_alpha), device=dev))
+        self.H_bar = torch.tensor([-float(cfg.action_dim)], device=dev)
+        self.H_bar_f = -float(cfg.action_dim)
+        self.alpha = self.log_alpha.exp().detach()
+        # CARE uses weighted losses iff modified CARE
+        # (reference learner.update_SAC use_weighted_loss=use_modified_care)
+        self.use_weighted_loss = self.use_modified_care
+
+    def _critic_layer_pairs(self, target: bool = False):
+        c = self.target_critic if target else self.local_critic
+        return self._critic_linears(c)
+
+    def _se_then_heads(self, critic: CARECritic):
+        order = list(critic.state_encoder.parameters())
+        self._se_numel = sum(p.numel() for p in order)
+        for l1, l2 in self._critic_linears(critic):
+            order += [l1.weight, l2.weight, l1.bias, l2.bias]
+        return order
+
+    def _build_optimizers(self) -> None:
+        cfg = self.cfg
+        self.actor_group = FlatParams(self.actor.mu_log_std_layer.parameters())
+        self.actor_se_group = FlatParams(self.actor.state_encoder.parameters(),
+                                         with_grad=False)
+        self.critic_group = FlatParams(self._se_then_heads(self.local_critic))
+        self.target_group = FlatParams(self._se_then_heads(self.target_critic),
+                                       with_grad=False)
+        self.alpha_group = FlatParams([self.log_alpha])
+        ctx_params = [p for p in self.context_encoder.parameters()
+                      if p.requires_grad]
+        self.context_group = FlatParams(ctx_params) if ctx_params else None
+
+        self.actor_optimizer = FusedAdam(self.actor_group, lr=cfg.lr_actor)
+        self.critic_optimizer = FusedAdam(self.critic_group, lr=cfg.lr_critic)
+        self.log_alpha_optimizer = FusedAdam(self.alpha_group, lr=cfg.lr_actor)
+        self.context_encoder_optimizer = (
+            FusedAdam(self.context_group,
+                      lr=float(self.enc_cfg.get("lr_contextEnc", cfg.lr_actor)))
+            if self.context_group is not None else None)
+
+        self._twin_local = self._build_twin_stacks(
+            self.critic_group, self._critic_layer_pairs(), with_grad=True)
+        self._twin_target = self._build_twin_stacks(
+            self.target_group, self._critic_layer_pairs(target=True),
+            with_grad=False)
+        self._twin_local_frozen = ([w.detach() for w in self._twin_local[0]],
+                                   [b.detach() for b in self._twin_local[1]])
+        self.hard_copy_targets()
+        self.tie_actor_state_encoder()
+
+    def attach_ddp(self, ddp) -> None:
+        self.ddp = ddp
+        if ddp is not None and ddp.enabled:
+            for g in (self.actor_group, self.critic_group, self.alpha_group,
+                      self.actor_se_group):
+                ddp.broadcast_params(g.flat_data)
+            if self.context_group is not None:
+                ddp.broadcast_params(self.context_group.flat_data)
+            self.hard_copy_targets()
+
+    @torch.no_grad()
+    def tie_actor_state_encoder(self) -> None:
+        """actor.state_encoder <- critic.state_encoder hard copy (reference
+        soft_update(..., tau=1.0), learner.py:402) as one flat copy."""
+        self.actor_se_group.flat_data.copy_(
+            self.critic_group.flat_data[: self._se_numel])
+
+    def zero_grad(self) -> None:
+        self.actor_group.zero_grad()
+        self.critic_group.zero_grad()
+        self.alpha_group.zero_grad()
+        if self.context_group is not None:
+            self.context_group.zero_grad()
+        self.actor_group.rebind_grads()
+        self.critic_group.rebind_grads()
+        self.alpha_group.rebind_grads()
+        if self.context_group is not None:
+            self.context_group.rebind_grads()
+
+    def _polyak_targets(self) -> None:
+        """Q heads at tau, state encoder at state_encoder_tau — two fused
+        kernels over the flat slices (reference learner.py:361-367)."""
+        from ..ops.flat import flat_polyak_  # noqa: F401
+        se = self._se_numel
+        t, s = self.target_group.flat_data, self.critic_group.flat_data
+        self._polyak_slice(t[:se], s[:se], self.se_tau)
+        self._polyak_slice(t[se:], s[se:], self.tau)
+
+    @staticmethod
+    @torch.no_grad()
+    def _polyak_slice(t: torch.Tensor, s: torch.Tensor, tau: float) -> None:
+        from ..ops import has_native, native, native_enabled
+        if t.is_cuda and native_enabled() and has_native():
+            native().polyak_(t, s, float(tau))
+        else:
+            t.mul_(1.0 - tau).add_(s, alpha=tau)
+
+    # ------------------------------------------------------------------
+    def _sample_care(self, mtobss, z_context, detach_z_encs=False):
+        mu, log_std_raw = self.actor(mtobss, z_context, detach_z_encs)
+        eps = self._next_eps(mu)
+        return Fops.squashed_gaussian(mu, log_std_raw, eps, self.actor.k)
+
+    def update_tensors(self, batch: Dict[str, torch.Tensor]):
+        if self._use_fused(batch["states"]):
+            return self._update_tensors_fused(batch)
+        states, actions = batch["states"], batch["actions"]
+        rewards, next_states, dones = (batch["rewards"], batch["next_states"],
+                                       batch["dones"])
+        T = self.num_tasks
+        alpha = Fops.gather_log_alpha(states[:, -T:],
+                                      self.log_alpha).exp().detach()
+        self.zero_grad()
+
+        # z_context once per update (reference :290)
+        z_context = self.context_encoder(states)
+        with torch.no_grad():
+            na, nlp, _ = self._sample_care(next_states, z_context)
+            q1_t, q2_t = self.target_critic(next_states, z_context, na)
+            y = Fops.td_target(rewards, dones, q1_t, q2_t, nlp, alpha,
+                               self.gamma, self.reward_scale)
+
+        l1, l2 = self.local_critic.cal_loss(
+            states, z_context, actions, y,
+            use_weighted_loss=self.use_weighted_loss, num_tasks=T,
+            alphas=self.log_alpha.exp().detach())
+        q_loss = l1 + l2
+        q_loss.backward(retain_graph=True)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self.critic_optimizer.step()
+
+        zc_d = z_context.detach()
+        sa, lp, ls = self._sample_care(states, zc_d, detach_z_encs=True)
+        q1, q2 = self.local_critic(states, zc_d, sa, detach_z_encs=True)
+        q_min = torch.min(q1, q2)
+        policy_loss = self.actor.cal_loss(
+            lp, q_min, alpha, use_weighted_loss=self.use_weighted_loss,
+            mtobss=states, num_tasks=T,
+            alphas=self.log_alpha.exp().detach())
+        policy_loss.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
+        self.actor_optimizer.step()
+
+        entropy = Fops.entropy_from_log_std(ls)
+        log_alpha_g = Fops.gather_log_alpha(states[:, -T:], self.log_alpha)
+        loss_log_alpha = -(log_alpha_g * (lp.detach() + self.H_bar)).mean()
+        loss_log_alpha.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+        self.log_alpha_optimizer.step()
+        self.alpha = self.log_alpha.exp().detach()
+
+        self._polyak_targets()
+        if self.context_encoder_optimizer is not None:
+            if self.ddp is not None:
+                self.ddp.allreduce_grad_(self.context_group.flat_grad)
+            self.context_encoder_optimizer.step()   # grads from critic bwd
+        self.tie_actor_state_encoder()
+        return {"critic_loss": q_loss.detach(),
+                "actor_loss": policy_loss.detach(),
+                "alpha_loss": loss_log_alpha.detach(),
+                "entropy": entropy.detach()}
+
+    def _update_tensors_fused(self, batch):
+        """GPU path: fused losses + twin-head grouped GEMMs.  Forwards whose
+        grads the reference discards run under no_grad (see module doc)."""
+        from ..ops import native
+        states, actions = batch["states"], batch["actions"]
+        rewards, next_states, dones = (batch["rewards"], batch["next_states"],
+                                       batch["dones"])
+        T = self.num_tasks
+        use_w = self.use_weighted_loss
+        self.zero_grad()
+
+        z_context = self.context_encoder(states)
+        with torch.no_grad():
+            na, nlp, _ = self._sample_care(next_states, z_context)
+            enc_t = self.target_critic.encode(next_states, z_context)
+            xt = torch.cat([enc_t, na], dim=-1)
+            q1_t, q2_t = Fops.twin_mlp_forward(xt, *self._twin_target)
+            y = native().td_target_mt(rewards, dones, q1_t, q2_t, nlp,
+                                      states, self.log_alpha.detach(), T,
+                                      self.gamma, self.reward_scale)
+
+        enc = self.local_critic.encode(states, z_context)
+        x = torch.cat([enc, actions], dim=-1)
+        q1, q2 = Fops.twin_mlp_forward(x, *self._twin_local)
+        l1, l2 = Fops.critic_loss(q1, q2, y, states, self.log_alpha.detach(),
+                                  T, use_w)
+        q_loss = l1 + l2
+        q_loss.backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+            if self.context_group is not None:
+                self.ddp.allreduce_grad_(self.context_group.flat_grad)
+        self.critic_optimizer.step()
+
+        zc_d = z_context.detach()
+        with torch.no_grad():
+            # actor-side encoder: grads into trunk are discarded by the
+            # reference (actor optimizer holds only the head) — skip them
+            enc_a = self.actor.state_encoder(zc_d, states, detach_z_encs=True)
+        mu_lsr = self.actor.mu_log_std_layer(enc_a)
+        mu = mu_lsr[:, : self.actor.action_dim]
+        lsr = mu_lsr[:, self.actor.action_dim:]
+        eps = self._next_eps(mu)
+        sa, lp, ls = Fops.squashed_gaussian(mu, lsr, eps, self.actor.k)
+        with torch.no_grad():
+            enc_c = self.local_critic.encode(states, zc_d, detach_z_encs=True)
+        xa = torch.cat([enc_c, sa], dim=-1)
+        aq1, aq2 = Fops.twin_mlp_forward(xa, *self._twin_local_frozen)
+        policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(
+            aq1, aq2, lp, ls, states, self.log_alpha, T, use_w, self.H_bar_f)
+        (policy_loss + loss_log_alpha).backward()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
+            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+        self.actor_optimizer.step()
+        self.log_alpha_optimizer.step()
+        self.alpha = self.log_alpha.exp().detach()
+
+        self._polyak_targets()
+        if self.context_encoder_optimizer is not None:
+            self.context_encoder_optimizer.step()
+        self.tie_actor_state_encoder()
+        return {"critic_loss": q_loss.detach(),
+                "actor_loss": policy_loss.detach(),
+                "alpha_loss": loss_log_alpha.detach(),
+                "entropy": entropy.detach()}
+
+    @torch.no_grad()
+    def publish_params(self) -> torch.Tensor:
+        """[actor.state_encoder | actor head | trainable context params] —
+        the CARE rollout policy needs the tied encoder and (original CARE)
+        the trained context encoder (reference C3 publishes actor +
+        context_encoder state_dicts, learner.py:412-422)."""
+        parts = [self.actor_se_group.flat_data, self.actor_group.flat_data]
+        if self.context_group is not None:
+            parts.append(self.context_group.flat_data)
+        return torch.cat(parts)
+
+    # ------------------------------------------------------------------
+    def checkpoint_state(self) -> Dict:
+        def cpu_sd(m):
+            return {k: v.cpu() for k, v in m.state_dict().items()}
+        ctx_opt = (self.context_encoder_optimizer.state_dict()
+                   if self.context_encoder_optimizer is not None else
+                   {"state": {}, "param_groups": []})
+        # MT10_Distributed_CARE/src/learner.py:178-198 key layout
+        return {
+            "update_iteration": self.update_iteration,
+            "total_step": self.total_step,
+            "context_encoder": cpu_sd(self.context_encoder),
+            "context_encoder_optimizer": ctx_opt,
+            "local_critic": cpu_sd(self.local_critic),
+            "critic_optimizer": self.critic_optimizer.state_dict(),
+            "target_critic": cpu_sd(self.target_critic),
+            "actor": cpu_sd(self.actor),
+            "actor_optimizer": self.actor_optimizer.state_dict(),
+            "log_alpha": self.log_alpha.detach().cpu(),
+            "log_alpha_optimizer": self.log_alpha_optimizer.state_dict(),
+            "alpha": self.alpha.cpu(),
+        }
+
+    def load_checkpoint_state(self, ckpt: Dict) -> None:
+        self.update_iteration = int(ckpt.get("update_iteration", 0))
+        self.total_step = int(ckpt.get("total_step", 0))
+        self.context_encoder.load_state_dict(ckpt["context_encoder"])
+        self.local_critic.load_state_dict(ckpt["local_critic"])
+        self.target_critic.load_state_dict(ckpt["target_critic"])
+        self.actor.load_state_dict(ckpt["actor"])
+        with torch.no_grad():
+            self.log_alpha.copy_(ckpt["log_alpha"].to(self.device))
+        self.alpha = self.log_alpha.exp().detach()
+        for name, opt in (("critic_optimizer", self.critic_optimizer),
+                          ("actor_optimizer", self.actor_optimizer),
+                          ("log_alpha_optimizer", self.log_alpha_optimizer)):
+            if name in ckpt:
+                opt.load_state_dict(ckpt[name])
+        if (self.context_encoder_optimizer is not None
+                and "context_encoder_optimizer" in ckpt
+                and ckpt["context_encoder_optimizer"].get("param_groups")):
+            self.context_encoder_optimizer.load_state_dict(
+                ckpt["context_encoder_optimizer"])

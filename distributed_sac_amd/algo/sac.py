@@ -192,6 +192,12 @@ class SACEngine:
             self.hard_copy_targets()
 
     @torch.no_grad()
+    def publish_params(self) -> torch.Tensor:
+        """Flat weight vector for the rollout-side snapshot (matches
+        workers.player.policy_params order)."""
+        return self.actor_group.flat_data
+
+    @torch.no_grad()
     def hard_copy_targets(self) -> None:
         """targets <- critics (reference soft_update tau=1.0 at run start)."""
         flat_polyak_(self.target_group, self.critic_group, 1.0)

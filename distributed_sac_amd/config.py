@@ -149,6 +149,8 @@ class SACConfig:
             c.encoder = cfg["encoder"]
         if "state_encoder_tau" in cfg:
             c.state_encoder_tau = float(cfg["state_encoder_tau"])
+        if c.encoder and "state_encoder_tau" in c.encoder:
+            c.state_encoder_tau = float(c.encoder["state_encoder_tau"])
 
         if variant is not None:
             c.variant = variant

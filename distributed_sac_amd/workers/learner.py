@@ -25,7 +25,7 @@ from typing import Dict, Optional
 
 import torch
 
-from ..algo.sac import SACEngine
+from ..algo import create_engine
 from ..checkpoint import save_checkpoint
 from ..config import SACConfig
 from ..replay import ShardedReplay
@@ -42,7 +42,7 @@ class Learner:
                  ddp=None, seed: int = 0):
         self.cfg = cfg
         self.device = torch.device(device)
-        self.engine = SACEngine(cfg, device)
+        self.engine = create_engine(cfg, device)
         if ddp is not None:
             self.engine.attach_ddp(ddp)
         num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
@@ -93,7 +93,7 @@ class Learner:
 
     # -- update --------------------------------------------------------
     def publish(self) -> None:
-        self.snapshot.publish(self.engine.actor_group.flat_data,
+        self.snapshot.publish(self.engine.publish_params(),
                               self.iteration_counter)
 
     def ready(self) -> bool:

@@ -26,11 +26,8 @@ from .trainer import default_env_fn
 
 
 def _actor_numel(cfg: SACConfig) -> int:
-    actor = (LLActor(cfg.state_dim, cfg.action_dim, cfg.actor_hidden_dim)
-             if cfg.variant in ("sac", "vsac") else
-             Actor(cfg.state_dim, cfg.action_dim, cfg.actor_hidden_dim,
-                   num_tasks=cfg.num_tasks))
-    return sum(p.numel() for p in actor.parameters())
+    from .player import build_actor, policy_params
+    return sum(p.numel() for p in policy_params(build_actor(cfg)))
 
 
 def default_task_partition(num_tasks: int, num_players: int) -> List[List[int]]:

@@ -40,15 +40,57 @@ def build_actor(cfg: SACConfig, device="cpu") -> nn.Module:
     if cfg.variant in ("sac", "vsac"):
         return LLActor(cfg.state_dim, cfg.action_dim, cfg.actor_hidden_dim,
                        cfg.action_bound).to(device)
+    if cfg.variant == "care":
+        return CAREPolicy(cfg).to(device)
     return Actor(cfg.state_dim, cfg.action_dim, cfg.actor_hidden_dim,
                  cfg.action_bound, num_tasks=cfg.num_tasks).to(device)
 
 
+class CAREPolicy(nn.Module):
+    """Rollout-side CARE bundle: context encoder + actor with the reference
+    Player's action path (MT10_Distributed_CARE/src/player.py:202-209:
+    z_context = context_encoder(mtobs); actor.get_action(mtobs, z))."""
+
+    def __init__(self, cfg: SACConfig):
+        super().__init__()
+        from ..models.care import CAREActor
+        from ..models.context_encoder import contextEncoder
+        enc_cfg = dict(cfg.encoder)
+        enc_cfg.setdefault("RoBERTa_embedding_dim", 768)
+        self.context_encoder = contextEncoder(enc_cfg, cfg.use_modified_care)
+        self.actor = CAREActor(
+            {"state_dim": cfg.state_dim, "action_dim": cfg.action_dim,
+             "action_bound": cfg.action_bound,
+             "actor_hidden_dim": cfg.actor_hidden_dim},
+            enc_cfg, cfg.use_modified_care)
+        self.k = self.actor.k
+
+    @torch.no_grad()
+    def get_action(self, mtobss: torch.Tensor, stochastic: bool = True):
+        z = self.context_encoder(mtobss)
+        return self.actor.get_action(mtobss, z, stochastic=stochastic)
+
+    def load_state_dict_from_checkpoint(self, ckpt):
+        self.actor.load_state_dict(ckpt["actor"])
+        if "context_encoder" in ckpt:
+            self.context_encoder.load_state_dict(ckpt["context_encoder"])
+
+
+def policy_params(policy: nn.Module):
+    """Ordered trainable params matching SACEngine/CAREEngine
+    publish_params: [actor params | trainable context params]."""
+    if isinstance(policy, CAREPolicy):
+        return (list(policy.actor.parameters())
+                + [p for p in policy.context_encoder.parameters()
+                   if p.requires_grad])
+    return list(policy.parameters())
+
+
 @torch.no_grad()
 def apply_flat_params(actor: nn.Module, flat: torch.Tensor) -> None:
-    """Copy a flat fp32 vector (learner actor_group order = parameters()
-    order) into the local actor."""
-    torch.nn.utils.vector_to_parameters(flat, actor.parameters())
+    """Copy a flat fp32 vector (engine publish_params order) into the
+    local policy."""
+    torch.nn.utils.vector_to_parameters(flat, policy_params(actor))
 
 
 def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
@@ -112,7 +154,13 @@ def evaluate_checkpoint(cfg: SACConfig, checkpoint_path: str, env_fn,
     success-rate/reward over N episodes (main.py else-branch +
     player.calculate_success_rate)."""
     actor = build_actor(cfg)
-    it = load_actor_for_eval(actor, checkpoint_path)
+    if cfg.variant == "care":
+        from ..checkpoint import load_checkpoint
+        ckpt = load_checkpoint(checkpoint_path)
+        actor.load_state_dict_from_checkpoint(ckpt)
+        it = int(ckpt.get("update_iteration", 0))
+    else:
+        it = load_actor_for_eval(actor, checkpoint_path)
     actor.eval()
     env = env_fn(cfg, task_idx, seed)
     rewards, successes = [], 0

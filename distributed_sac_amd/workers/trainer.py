@@ -13,7 +13,7 @@ from typing import Callable, Dict, List, Optional
 
 import torch
 
-from ..algo.sac import SACEngine
+from ..algo import create_engine
 from ..config import SACConfig
 from ..envs import make_synthetic
 from ..replay import ShardedReplay
@@ -29,6 +29,19 @@ def default_env_fn(cfg: SACConfig, task_idx: int, seed: int):
                         action_bound=tuple(cfg.action_bound))
 
 
+class _CAREEnginePolicy:
+    """In-process rollout adapter: engine's own context encoder + actor
+    (single-process Trainer shares weights by construction)."""
+
+    def __init__(self, engine):
+        self.engine = engine
+
+    @torch.no_grad()
+    def get_action(self, mtobss, stochastic=True):
+        z = self.engine.context_encoder(mtobss)
+        return self.engine.actor.get_action(mtobss, z, stochastic=stochastic)
+
+
 class Trainer:
     def __init__(self, cfg: SACConfig, device: str = "cpu",
                  env_fn: Optional[Callable] = None,
@@ -36,7 +49,7 @@ class Trainer:
                  logger: Optional[MetricLogger] = None):
         self.cfg = cfg
         self.device = torch.device(device)
-        self.engine = SACEngine(cfg, device)
+        self.engine = create_engine(cfg, device)
         num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
         self.replay = ShardedReplay(cfg.buffer_size, num_tasks,
                                     cfg.mtobs_dim, cfg.action_dim,
@@ -47,7 +60,10 @@ class Trainer:
             for j in range(envs_per_task):
                 envs.append(env_fn(cfg, t, seed * 10007 + t * 101 + j))
                 tasks.append(t)
-        self.rollout = VecRollout(cfg, envs, tasks, self.engine.actor,
+        rollout_actor = self.engine.actor
+        if cfg.variant == "care":
+            rollout_actor = _CAREEnginePolicy(self.engine)
+        self.rollout = VecRollout(cfg, envs, tasks, rollout_actor,
                                   device=device, seed=seed)
         self.logger = logger or MetricLogger(None)
         self.env_timer = StepTimer()

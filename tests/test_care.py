@@ -1,0 +1,342 @@
+"""CARE variant tests: encoder structure/naming, update equivalence vs the
+reference math (MT10_Distributed_CARE/src/learner.py:281-404), gradient-
+flow rules (tie / detach / context-grads-from-critic-only), checkpoint
+schema, end-to-end trainer."""
+
+import copy
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+from torch.distributions import Normal
+
+from distributed_sac_amd.algo import CAREEngine, create_engine
+from distributed_sac_amd.config import SACConfig, load_variant
+
+
+def care_cfg(tmp_path=None, modified=True, num_tasks=4) -> SACConfig:
+    c = SACConfig()
+    c.variant = "care"
+    c.state_dim, c.action_dim = 12, 3
+    c.num_tasks = num_tasks
+    c.actor_hidden_dim = c.critic_hidden_dim = [48, 48]
+    c.batch_size = 16
+    c.use_modified_care = modified
+    c.use_weighted_loss = modified
+    base = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "cfg", "metadata")
+    # synthesize tiny metadata for num_tasks tasks
+    names = [f"task-{i}" for i in range(num_tasks)]
+    rng = np.random.default_rng(0)
+    emb = {n: rng.standard_normal(32).tolist() for n in names}
+    d = tmp_path if tmp_path else base
+    tn = os.path.join(str(d), f"_test_tasks_{num_tasks}.json")
+    te = os.path.join(str(d), f"_test_emb_{num_tasks}.json")
+    json.dump(names, open(tn, "w"))
+    json.dump(emb, open(te, "w"))
+    c.encoder = {
+        "state_dim": 12, "num_tasks": num_tasks, "num_encoders": 3,
+        "pretrained_embedding_json_path": te, "task_name_json_path": tn,
+        "hidden_dims_contextEnc": [20, 20], "embedding_dim_contextEnc": 20,
+        "output_dim_contextEnc": 20, "RoBERTa_embedding_dim": 32,
+        "lr_contextEnc": 3e-4, "hidden_dims_mixtureEnc": [20],
+        "output_dim_mixtureEnc": 20, "state_encoder_tau": 0.05,
+    }
+    return c
+
+
+def care_batch(cfg, seed=0):
+    from tests.test_engine import make_batch
+    return make_batch(cfg, cfg.batch_size, seed)
+
+
+def test_state_dict_naming(tmp_path):
+    cfg = care_cfg(tmp_path)
+    engine = CAREEngine(cfg, "cpu")
+    a_keys = set(engine.actor.state_dict().keys())
+    assert "state_encoder.mixture_encoders.mixtureEncoders.0.W" in a_keys
+    assert "state_encoder.trunk.0.weight" in a_keys
+    assert "state_encoder.mlp_context.0.weight" in a_keys  # modified CARE
+    assert "mu_log_std_layer.0.weight" in a_keys
+    c_keys = set(engine.local_critic.state_dict().keys())
+    assert "Q_function_1.0.weight" in c_keys
+    assert "state_encoder.mixture_encoders.mixtureEncoders.2.b" in c_keys
+    ctx_keys = set(engine.context_encoder.state_dict().keys())
+    assert "embedding.0.weight" in ctx_keys  # frozen embedding
+
+
+def test_original_care_context_structure(tmp_path):
+    cfg = care_cfg(tmp_path, modified=False, num_tasks=1)
+    engine = CAREEngine(cfg, "cpu")
+    keys = set(engine.context_encoder.state_dict().keys())
+    # original CARE: embedding header + mlp exist and are trainable
+    assert "embedding.2.0.weight" in keys
+    assert "mlp.0.weight" in keys
+    assert engine.context_group is not None
+    assert engine.context_encoder_optimizer is not None
+    # modified CARE has no trainable context params
+    cfg2 = care_cfg(tmp_path, modified=True)
+    e2 = CAREEngine(cfg2, "cpu")
+    assert e2.context_group is None
+
+
+def _reference_care_update(engine_init, batch, eps_pairs, cfg, steps):
+    """Straight re-implementation of the reference CARE update loop on
+    deep-copied modules with torch.optim.Adam."""
+    ctx = copy.deepcopy(engine_init.context_encoder)
+    actor = copy.deepcopy(engine_init.actor)
+    critic = copy.deepcopy(engine_init.local_critic)
+    target = copy.deepcopy(engine_init.target_critic)
+    log_alpha = nn.Parameter(engine_init.log_alpha.detach().clone())
+    T = cfg.num_tasks
+    use_w = cfg.use_modified_care
+    opt_ctx = (torch.optim.Adam([p for p in ctx.parameters()
+                                 if p.requires_grad], lr=3e-4)
+               if any(p.requires_grad for p in ctx.parameters()) else None)
+    opt_a = torch.optim.Adam(actor.mu_log_std_layer.parameters(), lr=cfg.lr_actor)
+    opt_c = torch.optim.Adam(critic.parameters(), lr=cfg.lr_critic)
+    opt_al = torch.optim.Adam([log_alpha], lr=cfg.lr_actor)
+    H_bar = torch.tensor([-float(cfg.action_dim)])
+    k = actor.k
+    se_tau = cfg.encoder["state_encoder_tau"]
+
+    def soft(local, tgt, tau):
+        for tp, lp_ in zip(tgt.parameters(), local.parameters()):
+            tp.data.copy_(tau * lp_.data + (1 - tau) * tp.data)
+
+    def sample(mod, mtobss, z, eps, detach_z_encs=False):
+        mu, lsr = mod(mtobss, z, detach_z_encs)
+        ls = torch.clamp(lsr, -20, 2)
+        std = ls.exp()
+        u = mu + std * eps
+        a = k * torch.tanh(u)
+        lp = (Normal(mu, std).log_prob(u)
+              - torch.log(k * (1 - (a / k) ** 2 + 1e-6))).sum(-1, keepdim=True)
+        return a, lp, ls
+
+    for step in range(steps):
+        b = batch(step)
+        states, actions = b["states"], b["actions"]
+        rewards, next_states, dones = b["rewards"], b["next_states"], b["dones"]
+        oh = states[:, -T:]
+        alpha = (oh @ log_alpha.unsqueeze(0).t()).exp().detach()
+        opt_a.zero_grad(); opt_c.zero_grad(); opt_al.zero_grad()
+        ctx.zero_grad()
+        if opt_ctx:
+            opt_ctx.zero_grad()
+        eps1, eps2 = eps_pairs(step)
+
+        z = ctx(states)
+        with torch.no_grad():
+            na, nlp, _ = sample(actor, next_states, z, eps1)
+            q1t, q2t = target(next_states, z, na)
+            y = cfg.reward_scale * rewards + cfg.gamma * (1 - dones) * (
+                torch.min(q1t, q2t) - alpha * nlp)
+        l1, l2 = critic.cal_loss(states, z, actions, y,
+                                 use_weighted_loss=use_w, num_tasks=T,
+                                 alphas=log_alpha.exp().detach())
+        (l1 + l2).backward(retain_graph=True)
+        opt_c.step()
+
+        zd = z.detach()
+        sa, lp, ls = sample(actor, states, zd, eps2, detach_z_encs=True)
+        q1, q2 = critic(states, zd, sa, detach_z_encs=True)
+        qmin = torch.min(q1, q2)
+        pl = -(qmin - alpha * lp)
+        if use_w:
+            ti = torch.argmax(oh, dim=1)
+            w = torch.softmax(-log_alpha.exp().detach(), 0)[ti].detach()
+            w = (w / w.sum()).unsqueeze(-1)
+            pl = w * pl
+        pl.mean().backward()
+        opt_a.step()
+
+        la_g = oh @ log_alpha.unsqueeze(0).t()
+        (-(la_g * (lp.detach() + H_bar)).mean()).backward()
+        opt_al.step()
+
+        soft(critic.Q_function_1, target.Q_function_1, cfg.tau)
+        soft(critic.Q_function_2, target.Q_function_2, cfg.tau)
+        soft(critic.state_encoder, target.state_encoder, se_tau)
+        if opt_ctx:
+            opt_ctx.step()
+        soft(critic.state_encoder, actor.state_encoder, 1.0)
+    return ctx, actor, critic, target, log_alpha
+
+
+@pytest.mark.parametrize("modified", [True, False])
+def test_care_update_matches_reference_math(tmp_path, modified):
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path, modified=modified)
+    engine = CAREEngine(cfg, "cpu")
+    engine_init = copy.deepcopy(engine)
+
+    def batch(step):
+        return {k: v.clone() for k, v in care_batch(cfg, seed=step).items()}
+
+    eps_store = {}
+
+    def eps_pairs(step):
+        if step not in eps_store:
+            g = torch.Generator().manual_seed(500 + step)
+            eps_store[step] = (
+                torch.randn(cfg.batch_size, cfg.action_dim, generator=g),
+                torch.randn(cfg.batch_size, cfg.action_dim, generator=g))
+        return eps_store[step]
+
+    for step in range(3):
+        e1, e2 = eps_pairs(step)
+        engine._eps_queue = [e1.clone(), e2.clone()]
+        engine.update(batch(step))
+
+    ctx_r, actor_r, critic_r, target_r, la_r = _reference_care_update(
+        engine_init, batch, eps_pairs, cfg, steps=3)
+
+    for (n, p), (_, pr) in zip(engine.actor.named_parameters(),
+                               actor_r.named_parameters()):
+        assert torch.allclose(p, pr, atol=3e-6), f"actor {n}"
+    for (n, p), (_, pr) in zip(engine.local_critic.named_parameters(),
+                               critic_r.named_parameters()):
+        assert torch.allclose(p, pr, atol=3e-6), f"critic {n}"
+    for (n, p), (_, pr) in zip(engine.target_critic.named_parameters(),
+                               target_r.named_parameters()):
+        assert torch.allclose(p, pr, atol=3e-6), f"target {n}"
+    for (n, p), (_, pr) in zip(engine.context_encoder.named_parameters(),
+                               ctx_r.named_parameters()):
+        assert torch.allclose(p, pr, atol=3e-6), f"ctx {n}"
+    assert torch.allclose(engine.log_alpha, la_r, atol=3e-6)
+
+
+def test_actor_se_tied_to_critic_se(tmp_path):
+    cfg = care_cfg(tmp_path)
+    engine = CAREEngine(cfg, "cpu")
+    engine.update(care_batch(cfg, 0))
+    for (n, pa), (_, pc) in zip(
+            engine.actor.state_encoder.named_parameters(),
+            engine.local_critic.state_encoder.named_parameters()):
+        assert torch.equal(pa, pc), n
+
+
+def test_care_checkpoint_schema(tmp_path):
+    from distributed_sac_amd.checkpoint import load_checkpoint, save_checkpoint
+    cfg = care_cfg(tmp_path)
+    engine = CAREEngine(cfg, "cpu")
+    p = save_checkpoint(engine, str(tmp_path), update_iteration=9)
+    ckpt = load_checkpoint(p)
+    # MT10_Distributed_CARE/src/learner.py:178-198 key layout
+    for key in ("context_encoder", "context_encoder_optimizer",
+                "local_critic", "critic_optimizer", "target_critic",
+                "actor", "actor_optimizer", "log_alpha",
+                "log_alpha_optimizer", "alpha"):
+        assert key in ckpt, key
+    e2 = CAREEngine(cfg, "cpu")
+    e2.load_checkpoint_state(ckpt)
+    x = care_batch(cfg, 3)["states"]
+    z = e2.context_encoder(x)
+    z1 = engine.context_encoder(x)
+    assert torch.allclose(z, z1, atol=1e-7)
+    mu2, _ = e2.actor(x, z)
+    mu1, _ = engine.actor(x, z1)
+    assert torch.allclose(mu1, mu2, atol=1e-7)
+
+
+def test_care_trainer_end_to_end(tmp_path):
+    from distributed_sac_amd.workers import Trainer
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path)
+    cfg.buffer_size = 4000
+    cfg.start_memory_len = 64
+    cfg.random_step = 16
+    cfg.max_episode_time = 40
+    tr = Trainer(cfg, device="cpu", seed=0)
+    m = tr.train(env_steps_per_iter=30, updates_per_iter=1, iterations=3)
+    assert m and np.isfinite(m["critic_loss"])
+
+
+def test_canonical_care_cfg_builds():
+    cfg = load_variant("care")
+    engine = create_engine(cfg, "cpu")
+    assert isinstance(engine, CAREEngine)
+    assert engine.use_modified_care
+    assert engine.context_group is None  # frozen embeddings only
+    b_states = torch.randn(8, cfg.mtobs_dim)
+    oh = torch.nn.functional.one_hot(torch.randint(0, 10, (8,)), 10).float()
+    b_states[:, -10:] = oh
+    z = engine.context_encoder(b_states)
+    assert z.shape == (8, 768)
+    mu, lsr = engine.actor(b_states, z)
+    assert mu.shape == (8, 4)
+
+
+def test_mt1_care_cfg_builds():
+    cfg = load_variant("mt1_care")
+    assert not cfg.use_modified_care
+    engine = create_engine(cfg, "cpu")
+    assert engine.context_group is not None  # original CARE trains context
+    b_states = torch.randn(4, cfg.mtobs_dim)
+    b_states[:, -1:] = 1.0
+    z = engine.context_encoder(b_states)
+    assert z.shape == (4, 50)
+
+
+@pytest.mark.gpu
+def test_care_gpu_fused_matches_cpu(tmp_path):
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path)
+    e_cpu = CAREEngine(cfg, "cpu")
+    e_gpu = CAREEngine(cfg, "cuda:0")
+    e_gpu.context_encoder.load_state_dict(e_cpu.context_encoder.state_dict())
+    e_gpu.actor.load_state_dict(e_cpu.actor.state_dict())
+    e_gpu.local_critic.load_state_dict(e_cpu.local_critic.state_dict())
+    e_gpu.hard_copy_targets()
+    e_cpu.hard_copy_targets()
+    e_gpu.tie_actor_state_encoder()
+    e_cpu.tie_actor_state_encoder()
+    for step in range(3):
+        batch = care_batch(cfg, seed=step)
+        g = torch.Generator().manual_seed(77 + step)
+        eps = [torch.randn(cfg.batch_size, cfg.action_dim, generator=g)
+               for _ in range(2)]
+        e_cpu._eps_queue = [e.clone() for e in eps]
+        e_gpu._eps_queue = [e.clone() for e in eps]
+        m_cpu = e_cpu.update({k: v.clone() for k, v in batch.items()})
+        m_gpu = e_gpu.update({k: v.cuda() for k, v in batch.items()})
+    assert abs(m_cpu["critic_loss"] - m_gpu["critic_loss"]) < 2e-3
+    assert abs(m_cpu["actor_loss"] - m_gpu["actor_loss"]) < 2e-3
+    for (n, pc), (_, pg) in zip(e_cpu.actor.named_parameters(),
+                                e_gpu.actor.named_parameters()):
+        assert torch.allclose(pc, pg.cpu(), atol=2e-4, rtol=1e-3), n
+    for (n, pc), (_, pg) in zip(e_cpu.local_critic.named_parameters(),
+                                e_gpu.local_critic.named_parameters()):
+        assert torch.allclose(pc, pg.cpu(), atol=2e-4, rtol=1e-3), n
+
+
+@pytest.mark.gpu
+def test_care_graph_capture(tmp_path):
+    from distributed_sac_amd.replay import ShardedReplay
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path)
+    engine = CAREEngine(cfg, "cuda:0")
+    replay = ShardedReplay(4000, cfg.num_tasks, cfg.mtobs_dim,
+                           cfg.action_dim, device="cuda:0")
+    for t in range(cfg.num_tasks):
+        n = 256
+        st = torch.randn(n, cfg.mtobs_dim, device="cuda:0")
+        oh = torch.zeros(n, cfg.num_tasks, device="cuda:0")
+        oh[:, t] = 1
+        st[:, -cfg.num_tasks:] = oh
+        replay.shards[t].append(
+            st, torch.rand(n, cfg.action_dim, device="cuda:0") * 2 - 1,
+            torch.randn(n, 1, device="cuda:0"), st.clone(),
+            torch.zeros(n, 1, device="cuda:0"))
+    engine.capture(replay, cfg.batch_size)
+    for _ in range(4):
+        m = engine.graphed_update()
+    torch.cuda.synchronize()
+    for k, v in m.items():
+        assert float(v) == float(v), k
+    assert torch.isfinite(engine.critic_group.flat_data).all()
