@@ -97,7 +97,7 @@ def main():
     else:
         device = "cpu"
 
-    from distributed_sac_amd.algo import SACEngine
+    from distributed_sac_amd.algo import create_engine
     from distributed_sac_amd.config import load_variant
     from distributed_sac_amd.parallel import DataParallelGroup
     from distributed_sac_amd.replay import ShardedReplay
@@ -107,7 +107,7 @@ def main():
     torch.manual_seed(1000 + rank)
 
     ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
-    engine = SACEngine(cfg, device)
+    engine = create_engine(cfg, device)
     if ddp is not None:
         engine.attach_ddp(ddp)
 
