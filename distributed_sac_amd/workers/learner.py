@@ -39,7 +39,8 @@ class Learner:
                  logger: Optional[MetricLogger] = None,
                  save_dir: Optional[str] = None, save_period: int = 0,
                  update_delay: Optional[int] = None, use_graph: bool = True,
-                 ddp=None, seed: int = 0):
+                 ddp=None, seed: int = 0, heartbeat=None,
+                 heartbeat_timeout: float = 60.0):
         self.cfg = cfg
         self.device = torch.device(device)
         self.engine = create_engine(cfg, device)
@@ -63,6 +64,34 @@ class Learner:
         self.grad_steps = 0
         self.update_timer = StepTimer()
         self.ingest_count = 0
+        # failure detection (absent in the reference — SURVEY §5.3):
+        # shared wall-clock heartbeats, slot -1 = learner, others = players
+        self.heartbeat = heartbeat
+        self.heartbeat_timeout = heartbeat_timeout
+        self._dead_players = set()
+        self._hb_check = 0
+
+    def check_heartbeats(self) -> None:
+        if self.heartbeat is None:
+            return
+        self._hb_check += 1
+        if self._hb_check % 200 != 0:
+            return
+        import time as _t
+        now = _t.time()
+        self.heartbeat[-1] = now
+        for pid in range(self.heartbeat.numel() - 1):
+            last = float(self.heartbeat[pid])
+            if last > 0 and now - last > self.heartbeat_timeout \
+                    and pid not in self._dead_players:
+                self._dead_players.add(pid)
+                self.logger.print(
+                    f"WARNING: player {pid} heartbeat stale "
+                    f"({now - last:.0f}s) — continuing without it")
+
+    @property
+    def dead_players(self):
+        return set(self._dead_players)
 
     # -- ingest --------------------------------------------------------
     def drain_queue(self, max_blocks: int = 64) -> int:
@@ -157,6 +186,7 @@ class Learner:
                 break
             self.drain_queue()
             self.drain_logs()
+            self.check_heartbeats()
             self.train_step()
             if max_grad_steps and self.grad_steps >= max_grad_steps:
                 break

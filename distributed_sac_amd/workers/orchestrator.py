@@ -58,6 +58,8 @@ class DistributedTrainer:
         self.sample_queue = self.ctx.Queue(maxsize=1024)
         self.log_queue = self.ctx.Queue(maxsize=4096)
         self.stop_event = self.ctx.Event()
+        self.heartbeat = torch.zeros(num_players + 1, dtype=torch.float64)
+        self.heartbeat.share_memory_()
         num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
         self.partitions = default_task_partition(num_tasks, num_players)
         self.players: List[mp.Process] = []
@@ -66,7 +68,8 @@ class DistributedTrainer:
         self.learner = Learner(cfg, device, self.snapshot, self.sample_queue,
                                self.log_queue, logger=logger,
                                save_dir=save_dir, save_period=save_period,
-                               use_graph=use_graph, ddp=ddp, seed=seed)
+                               use_graph=use_graph, ddp=ddp, seed=seed,
+                               heartbeat=self.heartbeat)
 
     def start_players(self) -> None:
         for pid, tasks in enumerate(self.partitions):
@@ -74,7 +77,8 @@ class DistributedTrainer:
                 target=run_player,
                 args=(pid, self.cfg, self.env_fn, tasks, self.snapshot,
                       self.sample_queue, self.log_queue, self.stop_event,
-                      self.chunk_steps, self.seed + 131 * pid, 2, 0, None),
+                      self.chunk_steps, self.seed + 131 * pid, 2, 0, None,
+                      self.heartbeat),
                 daemon=True)
             p.start()
             self.players.append(p)

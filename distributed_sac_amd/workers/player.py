@@ -99,7 +99,8 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
                chunk_steps: int = 64, seed: int = 0,
                print_period_episodes: int = 0,
                eval_every_episodes: int = 0,
-               max_chunks: Optional[int] = None) -> None:
+               max_chunks: Optional[int] = None,
+               heartbeat: Optional[torch.Tensor] = None) -> None:
     """Infinite rollout loop (reference Player.run)."""
     torch.manual_seed(seed)
     actor = build_actor(cfg)
@@ -114,6 +115,8 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
     while stop_event is None or not stop_event.is_set():
         # pull parameters when a newer snapshot exists (reference
         # pull_parameters: apply only when update_iteration changed)
+        if heartbeat is not None:
+            heartbeat[player_id] = time.time()
         it = snapshot.read(flat, last_iteration)
         if it is not None:
             apply_flat_params(actor, flat)

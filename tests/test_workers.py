@@ -63,3 +63,45 @@ def test_evaluate_checkpoint(tmp_path):
     assert out["episodes"] == 2
     assert np.isfinite(out["mean_reward"])
     assert 0.0 <= out["success_rate"] <= 1.0
+
+
+def test_analysis_plot_utils(tmp_path):
+    """CARE analysis module (reference plot_utils parity, but runnable)."""
+    from distributed_sac_amd.algo import CAREEngine
+    from distributed_sac_amd.checkpoint import save_checkpoint
+    from distributed_sac_amd.analysis import (attention_map, cal_z_context,
+                                              plot_attention_map,
+                                              z_context_cosine_similarity)
+    from tests.test_care import care_cfg
+    cfg = care_cfg(tmp_path)
+    engine = CAREEngine(cfg, "cpu")
+    p = save_checkpoint(engine, str(tmp_path), update_iteration=1)
+    z = cal_z_context(cfg, p)
+    assert z.shape == (cfg.num_tasks, 32)  # modified CARE: raw embeddings
+    amap = attention_map(cfg, p)
+    assert amap.shape == (cfg.num_tasks, cfg.encoder["num_encoders"])
+    assert abs(amap.sum(axis=1) - 1.0).max() < 1e-5
+    sim = z_context_cosine_similarity(cfg, p)
+    assert sim.shape == (cfg.num_tasks, cfg.num_tasks)
+    assert abs(np.diag(sim) - 1.0).max() < 1e-5
+    out = plot_attention_map(cfg, p, str(tmp_path / "amap.png"))
+    import os
+    assert os.path.exists(out)
+
+
+def test_heartbeat_watchdog():
+    import time
+    import torch as th
+    from distributed_sac_amd.workers.learner import Learner
+    from distributed_sac_amd.workers.param_server import ParamSnapshot
+    import queue
+    cfg = tiny_cfg("sac")
+    hb = th.zeros(3, dtype=th.float64)
+    hb[0] = time.time() - 120  # player 0 stale
+    hb[1] = time.time()       # player 1 alive
+    lr = Learner(cfg, "cpu", ParamSnapshot(8), queue.Queue(),
+                 heartbeat=hb, heartbeat_timeout=60.0)
+    for _ in range(200):
+        lr.check_heartbeats()
+    assert lr.dead_players == {0}
+    assert float(hb[-1]) > 0  # learner heartbeat written
