@@ -143,16 +143,18 @@ __global__ __launch_bounds__(256) void k_linear_act_fwd(
 __global__ __launch_bounds__(256) void k_linear_bwd_dx(
     const float* __restrict__ dy, const float* __restrict__ w,
     const float* __restrict__ yout, float* __restrict__ dx,
-    int M, int N, int K, int act, int G) {
+    int M, int N, int K, int act, int G, int S) {
   __shared__ float sdy[2][BM][PAD_K];   // [m][n-slice]
   __shared__ float sw[2][BK][PAD_N];     // [n-slice][k]
-  // per-group mode: grid.z = G_outer with G==1 inside; summed mode:
-  // grid.z == 1 and the flattened tile loop accumulates over G groups.
+  // grid.z = Gz*S: Gz per-group slots (1 in summed mode), S slices over
+  // the flattened (g, n0) reduction-tile loop (split-reduce fills the
+  // chip for narrow-K layers; partials folded by k_reduce_partials).
   const long z = blockIdx.z;
-  dy += z * (long)M * N;
-  w += z * (long)N * K;
-  yout += z * (long)M * N;
-  dx += z * (long)M * K;
+  const long gz = z / S, sl = z % S;
+  dy += gz * (long)M * N;
+  w += gz * (long)N * K;
+  yout += gz * (long)M * N;
+  dx += z * (long)M * K;   // one output slab per (gz, slice)
   const int m0 = blockIdx.x * BM, c0 = blockIdx.y * BN;  // c over K
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
@@ -161,7 +163,10 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
   const int nt_per_g = (N + BK - 1) / BK;
-  const int nt = G * nt_per_g;
+  const int nt_all = G * nt_per_g;
+  const int per_s = (nt_all + S - 1) / S;
+  const int t_lo = (int)sl * per_s;
+  const int nt = min(nt_all, t_lo + per_s);
   const int ar = (tid * 8) >> 5, ac = (tid * 8) & 31;  // dy slot (64x32)
   const int br = (tid * 8) >> 6, bc = (tid * 8) & 63;  // w slot (32x64)
   float rdy[8], rw[8];
@@ -189,11 +194,24 @@ __global__ __launch_bounds__(256) void k_linear_bwd_dx(
     sdy[buf][ar][ac + j] = rdy[j];                                          \
     sw[buf][br][bc + j] = rw[j];                                            \
   }
-  LOAD_TILE_DX(0)
+  if (t_lo >= nt) {  // empty slice: still must write zeros
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = m0 + wr + mi * 16 + fk * 4 + r;
+          const int col = c0 + wc + ni * 16 + fi;
+          if (row < M && col < K) dx[(long)row * K + col] = 0.f;
+        }
+    return;
+  }
+  LOAD_TILE_DX(t_lo)
   STORE_TILE_DX(0)
   __syncthreads();
   int cur = 0;
-  for (int ti = 0; ti < nt; ++ti) {
+  for (int ti = t_lo; ti < nt; ++ti) {
     const bool more = ti + 1 < nt;
     if (more) LOAD_TILE_DX(ti + 1)
 #pragma unroll
@@ -743,13 +761,30 @@ static torch::Tensor linear_bwd_dx_g(torch::Tensor dy, torch::Tensor w,
   // mode: grid.z = G independent dx outputs [G,M,K].
   const long Gz = (G == 1 || sum_over_g) ? 1 : G;
   const int Gin = (int)((G > 1 && sum_over_g) ? G : 1);
+  const long nbx = (M + BM - 1) / BM, nby = (K + BN - 1) / BN;
+  const long nt_all = Gin * ((N + BK - 1) / BK);
+  // split-reduce the tile loop when the grid underfills the 256-CU chip
+  long S = 1;
+  const long base_blocks = nbx * nby * Gz;
+  if (base_blocks < 256) {
+    S = std::min<long>(nt_all, std::max<long>(1, 512 / base_blocks));
+  }
   auto dx = Gz == 1 ? torch::empty({M, K}, dyc.options())
                     : torch::empty({Gz, M, K}, dyc.options());
-  dim3 grid((M + BM - 1) / BM, (K + BN - 1) / BN, Gz);
+  auto out = dx;
+  if (S > 1) out = torch::empty({Gz * S, M, K}, dyc.options());
+  dim3 grid(nbx, nby, Gz * S);
   hipLaunchKernelGGL(k_linear_bwd_dx, grid, dim3(256), 0, cur_stream(),
                      dyc.data_ptr<float>(), wc.data_ptr<float>(),
-                     yc.data_ptr<float>(), dx.data_ptr<float>(),
-                     (int)M, (int)N, (int)K, (int)act, Gin);
+                     yc.data_ptr<float>(), out.data_ptr<float>(),
+                     (int)M, (int)N, (int)K, (int)act, Gin, (int)S);
+  if (S > 1) {
+    const long nmk = M * K;
+    dim3 g1((nmk + 255) / 256, Gz);
+    hipLaunchKernelGGL(k_reduce_partials, g1, dim3(256), 0, cur_stream(),
+                       out.data_ptr<float>(), dx.data_ptr<float>(), nmk,
+                       (int)S, nmk);
+  }
   return dx;
 }
 
@@ -769,9 +804,11 @@ static std::vector<torch::Tensor> linear_bwd_dwdb_g(torch::Tensor dy,
   const long N = G == 1 ? dyc.size(1) : dyc.size(2);
   const long K = xc.size(-1);
   const long xgs = xc.dim() == 3 ? M * K : 0;
-  // split-K: slices of >=128 batch rows, enough blocks to fill 256 CUs
-  long S = (M + 127) / 128;
-  S = std::max<long>(1, std::min<long>(S, 16));
+  // split-K over the batch: pick S so total blocks ~ 2x CU count (just
+  // enough to fill the chip; larger S doubles the partial-slab traffic)
+  const long nbx = (N + BM - 1) / BM, nby = (K + BN - 1) / BN;
+  long S = std::max<long>(1, 512 / std::max<long>(1, nbx * nby * G));
+  S = std::min<long>(S, (M + BK - 1) / BK);
   const int chunk = (int)(((M + S - 1) / S + BK - 1) / BK * BK);
   S = (M + chunk - 1) / chunk;
   auto ws = torch::empty({G * S, N, K}, dyc.options());
