@@ -66,20 +66,24 @@ def prefill_replay(replay, cfg, per_shard: int, device, seed: int):
         )
 
 
-def rollout_probe(cfg, device, n_envs: int = 16, steps: int = 100) -> float:
-    """Aux metric: env-steps/sec of one vectorized synthetic-env player."""
-    from distributed_sac_amd.workers.trainer import Trainer
+def rollout_probe(cfg, device, envs_per_task: int = 32,
+                  steps: int = 200) -> float:
+    """Aux metric: env-steps/sec of ONE vectorized synthetic-env player
+    (batched env stepping + batched policy inference)."""
     import copy
+    from distributed_sac_amd.workers.player import build_actor
+    from distributed_sac_amd.workers.rollout import FastSyntheticRollout
     c = copy.deepcopy(cfg)
-    c.random_step = 0  # measure policy-inference path, not warmup sampling
+    c.random_step = 0  # measure the policy-inference path, not warmup
     num_tasks = c.num_tasks if c.variant in ("mtsac", "care") else 1
-    per_task = max(1, n_envs // num_tasks)
-    tr = Trainer(c, device=device, envs_per_task=per_task, seed=1234)
-    tr.collect_steps(10)  # warm
+    actor = build_actor(c)
+    ro = FastSyntheticRollout(c, list(range(num_tasks)), actor,
+                              envs_per_task=envs_per_task, seed=1234)
+    ro.collect(10)  # warm
     t0 = time.perf_counter()
-    pushed = tr.collect_steps(steps)
+    ro.collect(steps)
     dt = time.perf_counter() - t0
-    return pushed / dt
+    return steps * num_tasks * envs_per_task / dt
 
 
 def main():

@@ -97,3 +97,62 @@ def make_synthetic(preset: str = "lunarlander", seed: int = 0,
         return SyntheticEnv(39, 4, max_episode_steps or 500, seed,
                             success_info=True)
     raise ValueError(preset)
+
+
+class BatchedSyntheticEnv:
+    """N synthetic envs stepped as ONE numpy batch op.
+
+    The rollout worker's lockstep loop (VecRollout) spends most of its CPU
+    time in per-env python ``step`` calls; this env family steps all N
+    copies in one vectorized update (states [N, S]), lifting a worker's
+    env-steps/s by an order of magnitude.  Same dynamics family as
+    :class:`SyntheticEnv` (shared A/B per instance, per-env noise).
+    """
+
+    def __init__(self, n_envs: int, state_dim: int, action_dim: int,
+                 max_episode_steps: int = 500, seed: int = 0,
+                 success_info: bool = False, action_bound=(-1.0, 1.0)):
+        self.n_envs = n_envs
+        self.state_dim = state_dim
+        self.action_dim = action_dim
+        self.max_episode_steps = max_episode_steps
+        self.success_info = success_info
+        self._rng = np.random.default_rng(seed)
+        self.A = self._rng.normal(0, 1.0 / np.sqrt(state_dim),
+                                  (state_dim, state_dim)).astype(np.float32)
+        self.B = self._rng.normal(0, 0.3, (state_dim, action_dim)).astype(np.float32)
+        self.low, self.high = action_bound
+        self.states = np.zeros((n_envs, state_dim), dtype=np.float32)
+        self.t = np.zeros(n_envs, dtype=np.int64)
+
+    def reset_all(self) -> np.ndarray:
+        self.t[:] = 0
+        self.states = self._rng.normal(
+            0, 1, (self.n_envs, self.state_dim)).astype(np.float32)
+        return self.states.copy()
+
+    def sample_actions(self) -> np.ndarray:
+        return self._rng.uniform(self.low, self.high,
+                                 (self.n_envs, self.action_dim)).astype(np.float32)
+
+    def step_all(self, actions: np.ndarray):
+        """Returns (next_states, rewards, dones, successes); auto-resets
+        finished envs (dones reflect the pre-reset transition)."""
+        a = np.clip(actions.astype(np.float32), self.low, self.high)
+        noise = self._rng.normal(0, 0.05,
+                                 self.states.shape).astype(np.float32)
+        self.states = 0.98 * (self.states @ self.A.T) + a @ self.B.T + noise
+        np.clip(self.states, -10.0, 10.0, out=self.states)
+        self.t += 1
+        rewards = (-0.1 * np.square(self.states).mean(axis=1)
+                   - 0.01 * np.square(a).mean(axis=1)).astype(np.float32)
+        dones = self.t >= self.max_episode_steps
+        success = (np.square(self.states).mean(axis=1) < 0.5).astype(np.float32) \
+            if self.success_info else np.zeros(self.n_envs, dtype=np.float32)
+        out_states = self.states.copy()
+        if dones.any():
+            idx = np.nonzero(dones)[0]
+            self.states[idx] = self._rng.normal(
+                0, 1, (len(idx), self.state_dim)).astype(np.float32)
+            self.t[idx] = 0
+        return out_states, rewards, dones, success

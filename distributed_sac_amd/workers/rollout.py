@@ -143,3 +143,84 @@ class VecRollout:
         self.ep_steps[i] = 0
         self.ep_rewards[i] = 0.0
         return successes / episodes
+
+
+class FastSyntheticRollout:
+    """High-throughput rollout over BatchedSyntheticEnv: one numpy step +
+    one batched policy inference per lockstep tick across all tasks.
+
+    Same contract as VecRollout.collect (per-task transition blocks with
+    warmup and mtobs semantics); terminal flags are pure time-limits in the
+    synthetic family, so stored dones are 0 (reference time-limit masking).
+    """
+
+    def __init__(self, cfg: SACConfig, task_indices: List[int],
+                 actor: torch.nn.Module, envs_per_task: int = 32,
+                 device: torch.device | str = "cpu", seed: int = 0):
+        from ..envs.synthetic import BatchedSyntheticEnv
+        self.cfg = cfg
+        self.tasks = sorted(set(task_indices))
+        self.actor = actor
+        self.device = torch.device(device)
+        self.num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 0
+        self.envs_per_task = envs_per_task
+        self.envs: Dict[int, object] = {}
+        self.obs: Dict[int, np.ndarray] = {}
+        for t in self.tasks:
+            env = BatchedSyntheticEnv(
+                envs_per_task, cfg.state_dim, cfg.action_dim,
+                max_episode_steps=cfg.max_episode_time,
+                seed=seed * 9173 + t,
+                success_info=cfg.variant in ("vsac", "mtsac", "care"),
+                action_bound=tuple(cfg.action_bound))
+            self.envs[t] = env
+            self.obs[t] = env.reset_all()
+        self.warmup_remaining = {t: cfg.random_step for t in self.tasks}
+        self.total_steps_per_task = {t: 0 for t in self.tasks}
+        # one-hot suffix per task, precomputed
+        self._oh = {t: one_hot(t, self.num_tasks) if self.num_tasks else None
+                    for t in self.tasks}
+
+    def _mtobs_block(self, t: int, states: np.ndarray) -> np.ndarray:
+        if not self.num_tasks:
+            return states
+        oh = np.broadcast_to(self._oh[t], (states.shape[0], self.num_tasks))
+        return np.concatenate([states, oh], axis=1)
+
+    @torch.no_grad()
+    def collect(self, n_steps: int) -> Dict[int, Dict[str, np.ndarray]]:
+        nt, ne = len(self.tasks), self.envs_per_task
+        S = self.cfg.mtobs_dim
+        A = self.cfg.action_dim
+        out = {t: {k: [] for k in FIELDS_ROLLOUT} for t in self.tasks}
+        for _ in range(n_steps):
+            obs_all = np.concatenate(
+                [self._mtobs_block(t, self.obs[t]) for t in self.tasks])
+            # batched policy inference across every env of every task
+            any_policy = any(self.warmup_remaining[t] <= 0 for t in self.tasks)
+            if any_policy:
+                x = torch.from_numpy(obs_all.astype(np.float32)).to(self.device)
+                pol = self.actor.get_action(x, stochastic=True).cpu().numpy()
+            for i, t in enumerate(self.tasks):
+                env = self.envs[t]
+                if self.warmup_remaining[t] > 0:
+                    acts = env.sample_actions()
+                    self.warmup_remaining[t] -= ne
+                else:
+                    acts = pol[i * ne:(i + 1) * ne]
+                states_mt = obs_all[i * ne:(i + 1) * ne]
+                next_states, rewards, dones, _succ = env.step_all(acts)
+                next_mt = self._mtobs_block(t, next_states)
+                out[t]["states"].append(states_mt)
+                out[t]["actions"].append(acts)
+                out[t]["rewards"].append(rewards.reshape(-1, 1))
+                out[t]["next_states"].append(next_mt)
+                # synthetic terminals are pure time-limits -> masked to 0
+                out[t]["dones"].append(np.zeros((ne, 1), dtype=np.float32))
+                self.obs[t] = env.states.copy()
+                self.total_steps_per_task[t] += ne
+        return {t: {k: np.concatenate(v).astype(np.float32)
+                    for k, v in blk.items()} for t, blk in out.items()}
+
+
+FIELDS_ROLLOUT = ("states", "actions", "rewards", "next_states", "dones")

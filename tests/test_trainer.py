@@ -73,3 +73,23 @@ def test_canonical_cfgs_load():
         if v == "care":
             assert cfg.encoder is not None
             assert cfg.encoder["num_encoders"] == 6
+
+
+def test_fast_synthetic_rollout():
+    from distributed_sac_amd.workers.player import build_actor
+    from distributed_sac_amd.workers.rollout import FastSyntheticRollout
+    cfg = tiny_cfg("mtsac")
+    cfg.random_step = 8
+    actor = build_actor(cfg)
+    ro = FastSyntheticRollout(cfg, list(range(cfg.num_tasks)), actor,
+                              envs_per_task=4, seed=0)
+    blocks = ro.collect(6)
+    for t in range(cfg.num_tasks):
+        blk = blocks[t]
+        assert blk["states"].shape == (24, cfg.mtobs_dim)
+        assert blk["actions"].shape == (24, cfg.action_dim)
+        oh = blk["states"][:, -cfg.num_tasks:]
+        assert (oh.argmax(axis=1) == t).all()
+        assert (blk["dones"] == 0).all()
+    # warmup exhausted after 8 steps (4 envs x 2 ticks)
+    assert all(v <= 0 for v in ro.warmup_remaining.values())
