@@ -40,7 +40,7 @@ class Learner:
                  save_dir: Optional[str] = None, save_period: int = 0,
                  update_delay: Optional[int] = None, use_graph: bool = True,
                  ddp=None, seed: int = 0, heartbeat=None,
-                 heartbeat_timeout: float = 60.0):
+                 heartbeat_timeout: float = 60.0, rings=None):
         self.cfg = cfg
         self.device = torch.device(device)
         self.engine = create_engine(cfg, device)
@@ -52,6 +52,10 @@ class Learner:
                                     device=device, seed=seed)
         self.snapshot = snapshot
         self.sample_queue = sample_queue
+        self.rings = rings or []
+        if self.rings and self.device.type == "cuda":
+            for r in self.rings:
+                r.pin()  # hipHostRegister: DMA-able drains
         self.log_queue = log_queue
         self.logger = logger or MetricLogger(None)
         self.save_dir = save_dir
@@ -94,8 +98,29 @@ class Learner:
         return set(self._dead_players)
 
     # -- ingest --------------------------------------------------------
-    def drain_queue(self, max_blocks: int = 64) -> int:
+    def drain_rings(self, max_blocks: int = 64) -> int:
+        """Drain the native shared-memory SPSC rings (one per player)."""
         n = 0
+        for ring in self.rings:
+            for _ in range(max_blocks):
+                out = ring.pop()
+                if not out:
+                    break
+                task = int(out[0].item())
+                self.replay.shards[task].append(
+                    out[1].to(self.device, non_blocking=True),
+                    out[2].to(self.device, non_blocking=True),
+                    out[3].to(self.device, non_blocking=True),
+                    out[4].to(self.device, non_blocking=True),
+                    out[5].to(self.device, non_blocking=True))
+                got = out[1].shape[0]
+                self.engine.total_step += got
+                self.ingest_count += got
+                n += 1
+        return n
+
+    def drain_queue(self, max_blocks: int = 64) -> int:
+        n = self.drain_rings(max_blocks)
         for _ in range(max_blocks):
             try:
                 _pid, task, blk = self.sample_queue.get_nowait()

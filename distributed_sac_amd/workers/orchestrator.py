@@ -50,7 +50,7 @@ class DistributedTrainer:
                  logger: Optional[MetricLogger] = None,
                  save_dir: Optional[str] = None, save_period: int = 0,
                  chunk_steps: int = 64, seed: int = 0, use_graph: bool = True,
-                 ddp=None):
+                 ddp=None, transport: str = "auto"):
         self.cfg = cfg
         self.env_fn = env_fn or default_env_fn
         self.ctx = mp.get_context("spawn")
@@ -65,11 +65,28 @@ class DistributedTrainer:
         self.players: List[mp.Process] = []
         self.chunk_steps = chunk_steps
         self.seed = seed
+        # native shared-memory SPSC rings (one per player) when available;
+        # mp.Queue otherwise ("auto"), or forced via transport=
+        from .. import ops as _ops
+        use_rings = (transport == "shm"
+                     or (transport == "auto" and _ops.has_native()))
+        self.rings = []
+        self.ring_names: List[str] = []
+        if use_rings:
+            import os as _os
+            ext = _ops.native()
+            slot_floats = chunk_steps * (2 * cfg.mtobs_dim
+                                         + cfg.action_dim + 2) + 64
+            for pid in range(num_players):
+                name = f"/dsac_{_os.getpid()}_{seed}_{pid}"
+                self.rings.append(ext.ShmRing(name, 256, slot_floats,
+                                              cfg.mtobs_dim, cfg.action_dim))
+                self.ring_names.append(name)
         self.learner = Learner(cfg, device, self.snapshot, self.sample_queue,
                                self.log_queue, logger=logger,
                                save_dir=save_dir, save_period=save_period,
                                use_graph=use_graph, ddp=ddp, seed=seed,
-                               heartbeat=self.heartbeat)
+                               heartbeat=self.heartbeat, rings=self.rings)
 
     def start_players(self) -> None:
         for pid, tasks in enumerate(self.partitions):
@@ -78,7 +95,8 @@ class DistributedTrainer:
                 args=(pid, self.cfg, self.env_fn, tasks, self.snapshot,
                       self.sample_queue, self.log_queue, self.stop_event,
                       self.chunk_steps, self.seed + 131 * pid, 2, 0, None,
-                      self.heartbeat),
+                      self.heartbeat,
+                      self.ring_names[pid] if self.ring_names else None),
                 daemon=True)
             p.start()
             self.players.append(p)

@@ -100,9 +100,14 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
                print_period_episodes: int = 0,
                eval_every_episodes: int = 0,
                max_chunks: Optional[int] = None,
-               heartbeat: Optional[torch.Tensor] = None) -> None:
+               heartbeat: Optional[torch.Tensor] = None,
+               ring_name: Optional[str] = None) -> None:
     """Infinite rollout loop (reference Player.run)."""
     torch.manual_seed(seed)
+    ring = None
+    if ring_name is not None:
+        from .. import ops
+        ring = ops.native().ShmRing.open(ring_name)
     actor = build_actor(cfg)
     actor.eval()
     envs = [env_fn(cfg, t, seed * 7919 + i) for i, t in enumerate(task_idx_list)]
@@ -123,7 +128,22 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
             last_iteration = it
         blocks = rollout.collect(chunk_steps)
         for t, blk in blocks.items():
-            if blk["states"].shape[0]:
+            if not blk["states"].shape[0]:
+                continue
+            if ring is not None:
+                ok = ring.push(t, torch.from_numpy(blk["states"]),
+                               torch.from_numpy(blk["actions"]),
+                               torch.from_numpy(blk["rewards"].reshape(-1, 1)),
+                               torch.from_numpy(blk["next_states"]),
+                               torch.from_numpy(blk["dones"].reshape(-1, 1)))
+                if not ok:
+                    time.sleep(0.005)  # learner behind; retry once
+                    ring.push(t, torch.from_numpy(blk["states"]),
+                              torch.from_numpy(blk["actions"]),
+                              torch.from_numpy(blk["rewards"].reshape(-1, 1)),
+                              torch.from_numpy(blk["next_states"]),
+                              torch.from_numpy(blk["dones"].reshape(-1, 1)))
+            else:
                 try:
                     sample_queue.put((player_id, t, blk), timeout=5.0)
                 except pyqueue.Full:  # learner stalled: drop oldest work

@@ -22,7 +22,8 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="distributed_sac_amd.ops._hip_ops",
-            sources=["distributed_sac_amd/ops/_hip/dsac_kernels.hip"],
+            sources=["distributed_sac_amd/ops/_hip/dsac_kernels.hip",
+                     "distributed_sac_amd/ops/_hip/shm_ring.cpp"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],

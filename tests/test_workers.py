@@ -105,3 +105,22 @@ def test_heartbeat_watchdog():
         lr.check_heartbeats()
     assert lr.dead_players == {0}
     assert float(hb[-1]) > 0  # learner heartbeat written
+
+
+@pytest.mark.timeout(300)
+def test_distributed_trainer_shm_transport():
+    """End-to-end async run over the native shared-memory rings."""
+    from distributed_sac_amd import ops
+    if not ops.has_native():
+        pytest.skip("native extension not built")
+    cfg = tiny_cfg("mtsac")
+    cfg.start_memory_len = 160
+    cfg.random_step = 32
+    cfg.update_delay = 1
+    dt = DistributedTrainer(cfg, device="cpu", num_players=2,
+                            chunk_steps=16, seed=3, use_graph=False,
+                            transport="shm")
+    assert len(dt.rings) == 2
+    stats = dt.run(max_grad_steps=10, max_seconds=120)
+    assert stats.get("grad_steps", 0) >= 10
+    assert stats["ingested"] >= cfg.start_memory_len
