@@ -52,6 +52,7 @@ class SACEngine:
         self._graph = None
         self._eps_queue: Optional[list] = None  # test hook: deterministic eps
         self.ddp = None  # optional DataParallelGroup (set via attach_ddp)
+        self._side_stream = None
         self._build_models()
         self._build_optimizers()
 
@@ -329,11 +330,30 @@ class SACEngine:
             "entropy": entropy.detach(),
         }
 
+    def _actor_weights(self):
+        if self.variant in ("sac", "vsac"):
+            ws = [m.weight for m in self.actor.layer_intermediate] \
+                + [self.actor.mu_log_std_layer.weight]
+            bs = [m.bias for m in self.actor.layer_intermediate] \
+                + [self.actor.mu_log_std_layer.bias]
+            return ws, bs
+        import torch.nn as nn_
+        lins = [m for m in self.actor.mu_log_std_layer
+                if isinstance(m, nn_.Linear)]
+        return [m.weight for m in lins], [m.bias for m in lins]
+
     def _update_tensors_fused(self, batch):
-        """GPU path: same math/order as update_tensors, with the loss heads,
-        per-sample alpha gather and entropy fused into single kernels
-        (ops.functional.critic_loss / actor_alpha_loss) and the twin-Q
-        critics running as grouped GEMMs."""
+        """GPU path: same math/order as update_tensors, restructured for
+        the hardware (numerically identical):
+
+        - ONE actor forward + squash over cat(next_states, states) — both
+          use pre-actor-step weights (reference samples current actions
+          after the critic step, but actor weights are unchanged until the
+          actor step); backward runs only on the grad-carrying states half;
+        - target-critic TD forward overlaps the local-critic loss forward
+          on a side stream (independent weight sets, both pre-critic-step);
+        - fused loss kernels, grouped twin-Q GEMMs, one fused Adam launch
+          for actor+alpha."""
         from ..ops import native
         states = batch["states"]
         actions = batch["actions"]
@@ -342,16 +362,44 @@ class SACEngine:
         dones = batch["dones"]
         T = self.num_tasks
         use_w = self.use_weighted_loss
+        B = states.shape[0]
+        A = self.cfg.action_dim
 
         self.zero_grad()
-        with torch.no_grad():
-            next_actions, next_log_probs, _ = self._sample(next_states)
-            q1_t, q2_t = self._target_q(next_states, next_actions)
-            y = native().td_target_mt(
-                rewards, dones, q1_t, q2_t, next_log_probs, states,
-                self.log_alpha.detach(), T, self.gamma, self.reward_scale)
+
+        # --- batched actor forward: [next | current] ------------------
+        x_cat = torch.cat([next_states, states], dim=0)
+        ws, bs = self._actor_weights()
+        out = Fops.mlp_forward(x_cat, ws, bs, grad_row_start=B)
+        mu, lsr = out[:, :A], out[:, A:]
+        if self._eps_queue:
+            eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
+        else:
+            eps = torch.randn_like(mu)
+        a_cat, lp_cat, ls_cat = Fops.squashed_gaussian(mu, lsr, eps,
+                                                       self.actor.k)
+        next_actions = a_cat[:B].detach()
+        next_log_probs = lp_cat[:B].detach()
+        sampled_actions = a_cat[B:]
+        log_probs = lp_cat[B:]
+        log_stds = ls_cat[B:]
+
+        # --- TD target on a side stream, overlapped with the critic-loss
+        # forward below (independent: target vs local weights) ----------
+        cur = torch.cuda.current_stream(self.device)
+        if self._side_stream is None:
+            self._side_stream = torch.cuda.Stream(self.device)
+        self._side_stream.wait_stream(cur)
+        with torch.cuda.stream(self._side_stream):
+            with torch.no_grad():
+                q1_t, q2_t = self._target_q(next_states, next_actions)
+                y = native().td_target_mt(
+                    rewards, dones, q1_t, q2_t, next_log_probs, states,
+                    self.log_alpha.detach(), T, self.gamma,
+                    self.reward_scale)
 
         q1, q2 = self._critic_q(states, actions)
+        cur.wait_stream(self._side_stream)
         l1, l2 = Fops.critic_loss(q1, q2, y, states, self.log_alpha.detach(),
                                   T, use_w)
         q_loss = l1 + l2
@@ -360,7 +408,7 @@ class SACEngine:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()
 
-        sampled_actions, log_probs, log_stds = self._sample(states)
+        # --- actor/alpha step (post-critic-step critic, frozen heads) --
         xa = torch.cat([states, sampled_actions], dim=-1)
         aq1, aq2 = Fops.twin_mlp_forward(xa, *self._twin_local_frozen)
         policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(
@@ -370,8 +418,8 @@ class SACEngine:
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.actor_group.flat_grad)
             self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
-        self.actor_optimizer.step()
-        self.log_alpha_optimizer.step()
+        FusedAdam.step_many([self.actor_optimizer,
+                             self.log_alpha_optimizer])
         self.alpha = self.log_alpha.exp().detach()
 
         flat_polyak_(self.target_group, self.critic_group, self.tau)

@@ -708,6 +708,44 @@ __global__ __launch_bounds__(256) void k_adam_dev(
   p[i] -= step_size * mi / (sqrtf(vi) * inv_sqrt_bc2 + eps);
 }
 
+// K9c: all three optimizer groups (critic was stepped earlier; this one
+// covers actor+alpha+optional context in ONE launch) — or any <=3 groups.
+struct AdamGroup {
+  float* p; const float* g; float* m; float* v; const float* st; long n;
+};
+
+__global__ void k_adam_prolog3(float* s0, float* s1, float* s2,
+                               float lr0, float lr1, float lr2,
+                               float b1, float b2) {
+  const int i = threadIdx.x;
+  float* s = i == 0 ? s0 : (i == 1 ? s1 : s2);
+  const float lr = i == 0 ? lr0 : (i == 1 ? lr1 : lr2);
+  if (s == nullptr) return;
+  const float step = s[0] + 1.f;
+  s[0] = step;
+  s[1] = lr / (1.f - __powf(b1, step));
+  s[2] = 1.f / sqrtf(1.f - __powf(b2, step));
+}
+
+__global__ __launch_bounds__(256) void k_adam_multi(
+    float* p0, const float* g0, float* m0, float* v0, const float* st0, long n0,
+    float* p1, const float* g1, float* m1, float* v1, const float* st1, long n1,
+    float* p2, const float* g2, float* m2, float* v2, const float* st2, long n2,
+    float b1, float b2, float eps) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  float* p; const float* g; float* m; float* v; const float* st;
+  if (i < n0) { p = p0; g = g0; m = m0; v = v0; st = st0; }
+  else if ((i -= n0) < n1) { p = p1; g = g1; m = m1; v = v1; st = st1; }
+  else if ((i -= n1) < n2) { p = p2; g = g2; m = m2; v = v2; st = st2; }
+  else return;
+  const float gi = g[i];
+  const float mi = b1 * m[i] + (1.f - b1) * gi;
+  const float vi = b2 * v[i] + (1.f - b2) * gi * gi;
+  m[i] = mi;
+  v[i] = vi;
+  p[i] -= st[1] * mi / (sqrtf(vi) * st[2] + eps);
+}
+
 // ---------------------------------------------------------------------------
 // K10: t = (1-tau)*t + tau*s over flat buffers.
 // ---------------------------------------------------------------------------
@@ -1066,6 +1104,45 @@ static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      (float)eps);
 }
 
+static void adam_step_multi_(std::vector<torch::Tensor> ps,
+                             std::vector<torch::Tensor> gs,
+                             std::vector<torch::Tensor> ms,
+                             std::vector<torch::Tensor> vs,
+                             std::vector<torch::Tensor> states,
+                             std::vector<double> lrs, double b1, double b2,
+                             double eps) {
+  const size_t G = ps.size();
+  TORCH_CHECK(G >= 1 && G <= 3, "1..3 groups");
+  float* P[3] = {nullptr, nullptr, nullptr};
+  const float* Gr[3] = {nullptr, nullptr, nullptr};
+  float* M[3] = {nullptr, nullptr, nullptr};
+  float* V[3] = {nullptr, nullptr, nullptr};
+  float* St[3] = {nullptr, nullptr, nullptr};
+  long N[3] = {0, 0, 0};
+  double LR[3] = {0, 0, 0};
+  long total = 0;
+  for (size_t g = 0; g < G; ++g) {
+    CHECK_IN(ps[g]);
+    P[g] = ps[g].data_ptr<float>();
+    Gr[g] = gs[g].data_ptr<float>();
+    M[g] = ms[g].data_ptr<float>();
+    V[g] = vs[g].data_ptr<float>();
+    St[g] = states[g].data_ptr<float>();
+    N[g] = ps[g].numel();
+    LR[g] = lrs[g];
+    total += N[g];
+  }
+  hipLaunchKernelGGL(k_adam_prolog3, dim3(1), dim3(3), 0, cur_stream(),
+                     St[0], St[1], St[2], (float)LR[0], (float)LR[1],
+                     (float)LR[2], (float)b1, (float)b2);
+  hipLaunchKernelGGL(k_adam_multi, dim3((total + 255) / 256), dim3(256), 0,
+                     cur_stream(),
+                     P[0], Gr[0], M[0], V[0], St[0], N[0],
+                     P[1], Gr[1], M[1], V[1], St[1], N[1],
+                     P[2], Gr[2], M[2], V[2], St[2], N[2],
+                     (float)b1, (float)b2, (float)eps);
+}
+
 static void polyak_(torch::Tensor t, torch::Tensor s, double tau) {
   CHECK_IN(t);
   const long n = t.numel();
@@ -1096,5 +1173,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("actor_alpha_loss_bwd", &actor_alpha_loss_bwd);
   mod.def("adam_step_", &adam_step_);
   mod.def("adam_step_dev_", &adam_step_dev_);
+  mod.def("adam_step_multi_", &adam_step_multi_);
   mod.def("polyak_", &polyak_);
 }

@@ -155,6 +155,28 @@ class FusedAdam:
         denom = (v.sqrt() / math.sqrt(bc2)).add_(self.eps)
         p.addcdiv_(m, denom, value=-self.lr / bc1)
 
+    @staticmethod
+    @torch.no_grad()
+    def step_many(opts: List["FusedAdam"]) -> None:
+        """Step up to 3 optimizers in ONE fused kernel launch (same betas/
+        eps; device-resident step state required)."""
+        live = [o for o in opts if o is not None]
+        if (1 <= len(live) <= 3
+                and all(o._dev_state is not None for o in live)
+                and all(o.betas == live[0].betas and o.eps == live[0].eps
+                        for o in live)):
+            native().adam_step_multi_(
+                [o.group.flat_data for o in live],
+                [o.group.flat_grad for o in live],
+                [o.exp_avg for o in live],
+                [o.exp_avg_sq for o in live],
+                [o._dev_state for o in live],
+                [o.lr for o in live],
+                live[0].betas[0], live[0].betas[1], live[0].eps)
+            return
+        for o in live:
+            o.step()
+
     # -- torch.optim.Adam-compatible (de)serialization ---------------------
 
     def state_dict(self) -> Dict:

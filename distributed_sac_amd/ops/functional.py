@@ -38,7 +38,8 @@ class _FusedMLP(torch.autograd.Function):
     """
 
     @staticmethod
-    def forward(ctx, x: torch.Tensor, n_layers: int, *wb):
+    def forward(ctx, x: torch.Tensor, n_layers: int, grad_row_start: int,
+                *wb):
         ws = wb[:n_layers]
         bs = wb[n_layers:]
         ext = native()
@@ -50,6 +51,7 @@ class _FusedMLP(torch.autograd.Function):
             acts.append(h)
         ctx.save_for_backward(*acts, *ws)
         ctx.n_layers = n_layers
+        ctx.grad_row_start = grad_row_start
         return h
 
     @staticmethod
@@ -59,13 +61,20 @@ class _FusedMLP(torch.autograd.Function):
         acts = saved[: n + 1]
         ws = saved[n + 1:]
         ext = native()
+        # rows below grad_row_start carry no gradient by construction (the
+        # batched TD-side forward): slice them out of the backward chain
+        # (row slices of row-major tensors stay contiguous)
+        r0 = ctx.grad_row_start
+        if r0:
+            acts = [a[r0:] for a in acts]
+            grad_out = grad_out[r0:]
         dy = grad_out.contiguous()
         dws: List[Optional[torch.Tensor]] = [None] * n
         dbs: List[Optional[torch.Tensor]] = [None] * n
         for i in range(n - 1, -1, -1):
             # mask==1 only for hidden layers (their saved act is post-ReLU)
             act = ACT_RELU if i < n - 1 else ACT_NONE
-            if ctx.needs_input_grad[2 + i]:
+            if ctx.needs_input_grad[3 + i]:
                 dw, db = ext.linear_bwd_dwdb(dy, acts[i], acts[i + 1], act)
                 dws[i], dbs[i] = dw, db
             if i > 0:
@@ -73,15 +82,21 @@ class _FusedMLP(torch.autograd.Function):
         dx = ext.linear_bwd_dx(dy, ws[0], acts[1],
                                ACT_RELU if n > 1 else ACT_NONE) \
             if ctx.needs_input_grad[0] else None
-        return (dx, None, *dws, *dbs)
+        return (dx, None, None, *dws, *dbs)
 
 
 def mlp_forward(x: torch.Tensor,
                 weights: Sequence[torch.Tensor],
-                biases: Sequence[torch.Tensor]) -> torch.Tensor:
-    """ReLU-hidden MLP with linear output (reference build_mlp semantics)."""
+                biases: Sequence[torch.Tensor],
+                grad_row_start: int = 0) -> torch.Tensor:
+    """ReLU-hidden MLP with linear output (reference build_mlp semantics).
+
+    grad_row_start>0: rows [0, start) are forward-only (their output grads
+    are structurally zero — e.g. the TD half of a batched actor pass), so
+    backward runs on the remaining rows only."""
     if _use_native(x):
-        return _FusedMLP.apply(x, len(weights), *weights, *biases)
+        return _FusedMLP.apply(x, len(weights), grad_row_start,
+                               *weights, *biases)
     return torch_ref.mlp_forward(x, weights, biases)
 
 
