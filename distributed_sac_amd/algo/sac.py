@@ -367,56 +367,33 @@ class SACEngine:
 
         self.zero_grad()
 
-        # --- batched actor forward: [next | current] ------------------
-        import os as _os
-        batched = _os.environ.get("DSAC_NO_BATCHED_ACTOR", "0") != "1"
-        overlap = _os.environ.get("DSAC_NO_OVERLAP", "0") != "1"
+        # --- batched actor forward: [next | current] halves (both use
+        # pre-actor-step weights; backward slices to the states half).
+        # A side-stream overlap of the TD forward was A/B-tested and LOST
+        # (cross-stream event sync in captured graphs cost ~5-25% —
+        # profiles/r04_NOTES.md), so the TD block stays on the main stream.
+        x_cat = torch.cat([next_states, states], dim=0)
         ws, bs = self._actor_weights()
-        if batched:
-            x_cat = torch.cat([next_states, states], dim=0)
-            out = Fops.mlp_forward(x_cat, ws, bs, grad_row_start=B)
-            mu, lsr = out[:, :A], out[:, A:]
-            if self._eps_queue:
-                eps = torch.cat([self._next_eps(mu[:B]),
-                                 self._next_eps(mu[:B])])
-            else:
-                eps = torch.randn_like(mu)
-            a_cat, lp_cat, ls_cat = Fops.squashed_gaussian(mu, lsr, eps,
-                                                           self.actor.k)
-            next_actions = a_cat[:B].detach()
-            next_log_probs = lp_cat[:B].detach()
-            sampled_actions = a_cat[B:]
-            log_probs = lp_cat[B:]
-            log_stds = ls_cat[B:]
+        out = Fops.mlp_forward(x_cat, ws, bs, grad_row_start=B)
+        mu, lsr = out[:, :A], out[:, A:]
+        if self._eps_queue:
+            eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
         else:
-            with torch.no_grad():
-                next_actions, next_log_probs, _ = self._sample(next_states)
-            sampled_actions, log_probs, log_stds = self._sample(states)
+            eps = torch.randn_like(mu)
+        a_cat, lp_cat, ls_cat = Fops.squashed_gaussian(mu, lsr, eps,
+                                                       self.actor.k)
+        next_actions = a_cat[:B].detach()
+        next_log_probs = lp_cat[:B].detach()
+        sampled_actions = a_cat[B:]
+        log_probs = lp_cat[B:]
+        log_stds = ls_cat[B:]
 
-        # --- TD target, optionally on a side stream overlapping the
-        # critic-loss forward (independent: target vs local weights) ----
-        cur = torch.cuda.current_stream(self.device)
-        if overlap:
-            if self._side_stream is None:
-                self._side_stream = torch.cuda.Stream(self.device)
-            self._side_stream.wait_stream(cur)
-            with torch.cuda.stream(self._side_stream):
-                with torch.no_grad():
-                    q1_t, q2_t = self._target_q(next_states, next_actions)
-                    y = native().td_target_mt(
-                        rewards, dones, q1_t, q2_t, next_log_probs, states,
-                        self.log_alpha.detach(), T, self.gamma,
-                        self.reward_scale)
-            q1, q2 = self._critic_q(states, actions)
-            cur.wait_stream(self._side_stream)
-        else:
-            with torch.no_grad():
-                q1_t, q2_t = self._target_q(next_states, next_actions)
-                y = native().td_target_mt(
-                    rewards, dones, q1_t, q2_t, next_log_probs, states,
-                    self.log_alpha.detach(), T, self.gamma,
-                    self.reward_scale)
-            q1, q2 = self._critic_q(states, actions)
+        with torch.no_grad():
+            q1_t, q2_t = self._target_q(next_states, next_actions)
+            y = native().td_target_mt(
+                rewards, dones, q1_t, q2_t, next_log_probs, states,
+                self.log_alpha.detach(), T, self.gamma, self.reward_scale)
+        q1, q2 = self._critic_q(states, actions)
         l1, l2 = Fops.critic_loss(q1, q2, y, states, self.log_alpha.detach(),
                                   T, use_w)
         q_loss = l1 + l2
