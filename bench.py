@@ -41,6 +41,8 @@ def parse_args():
     p.add_argument("--skip-rollout-probe", action="store_true")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph capture of the update step")
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                   help="GEMM compute dtype (fp32 masters either way)")
     return p.parse_args()
 
 
@@ -112,6 +114,9 @@ def main():
 
     ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
     engine = create_engine(cfg, device)
+    if args.dtype == "bf16" and cfg.variant != "care":
+        engine.precision = "bf16"
+        engine._init_bf16_mirrors()
     if ddp is not None:
         engine.attach_ddp(ddp)
 
@@ -182,7 +187,7 @@ def main():
             # derivable): nearest derived reference rate is MT1-CARE
             # 5.1 grad-steps/s @ B=1024 on a GTX 1080.
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "env_steps_per_sec": (round(env_rate, 1) if env_rate else None),
             "config": {

@@ -38,7 +38,11 @@ from ..ops.flat import FlatParams, FusedAdam, flat_polyak_
 class SACEngine:
     """Holds models/optimizers and performs one SAC gradient update."""
 
-    def __init__(self, cfg: SACConfig, device: torch.device | str = "cpu"):
+    def __init__(self, cfg: SACConfig, device: torch.device | str = "cpu",
+                 precision: Optional[str] = None):
+        import os as _os
+        self.precision = (precision or
+                          _os.environ.get("DSAC_PRECISION", "fp32"))
         self.cfg = cfg
         self.variant = cfg.variant
         self.device = torch.device(device)
@@ -180,7 +184,59 @@ class SACEngine:
         # next update) — skipping their computation saves 4 dW GEMMs/step.
         self._twin_local_frozen = ([w.detach() for w in self._twin_local[0]],
                                    [b.detach() for b in self._twin_local[1]])
+        self._init_bf16_mirrors()
         self.hard_copy_targets()
+
+    # -- bf16 mixed precision: fp32 masters + bf16 compute mirrors --------
+    def _init_bf16_mirrors(self) -> None:
+        self._bf16 = (self.precision == "bf16"
+                      and self.device.type == "cuda")
+        if not self._bf16:
+            return
+        dev = self.device
+        self._critic_bf16 = torch.empty(self.critic_group.numel,
+                                        dtype=torch.bfloat16, device=dev)
+        self._target_bf16 = torch.empty(self.target_group.numel,
+                                        dtype=torch.bfloat16, device=dev)
+        self._actor_bf16 = torch.empty(self.actor_group.numel,
+                                       dtype=torch.bfloat16, device=dev)
+        self._twin_local_bf16 = self._stack_views(self.critic_group,
+                                                  self._critic_bf16,
+                                                  self._critic_layer_pairs())
+        self._twin_target_bf16 = self._stack_views(
+            self.target_group, self._target_bf16,
+            self._critic_layer_pairs(target=True))
+        ws, _ = self._actor_weights()
+        self._actor_ws_bf16 = []
+        for w in ws:
+            i = next(j for j, q in enumerate(self.actor_group.params)
+                     if q is w)
+            off = self.actor_group.offsets[i]
+            self._actor_ws_bf16.append(
+                self._actor_bf16[off:off + w.numel()].view_as(w))
+        self.refresh_bf16()
+
+    @staticmethod
+    def _stack_views(group, mirror, pairs):
+        ws = []
+        for l1, l2 in pairs:
+            ia = next(i for i, q in enumerate(group.params)
+                      if q is l1.weight)
+            off = group.offsets[ia]
+            n = l1.weight.numel()
+            ws.append(mirror[off:off + 2 * n].view(2, *l1.weight.shape))
+        return ws
+
+    @torch.no_grad()
+    def refresh_bf16(self, which: str = "all") -> None:
+        if not getattr(self, "_bf16", False):
+            return
+        if which in ("all", "critic"):
+            self._critic_bf16.copy_(self.critic_group.flat_data)
+        if which in ("all", "target"):
+            self._target_bf16.copy_(self.target_group.flat_data)
+        if which in ("all", "actor"):
+            self._actor_bf16.copy_(self.actor_group.flat_data)
 
     def attach_ddp(self, ddp) -> None:
         """Join a data-parallel group: sync replicas, then every update
@@ -191,6 +247,7 @@ class SACEngine:
             ddp.broadcast_params(self.critic_group.flat_data)
             ddp.broadcast_params(self.alpha_group.flat_data)
             self.hard_copy_targets()
+            self.refresh_bf16()
 
     @torch.no_grad()
     def publish_params(self) -> torch.Tensor:
@@ -202,6 +259,7 @@ class SACEngine:
     def hard_copy_targets(self) -> None:
         """targets <- critics (reference soft_update tau=1.0 at run start)."""
         flat_polyak_(self.target_group, self.critic_group, 1.0)
+        self.refresh_bf16("target")
 
     def zero_grad(self) -> None:
         self.actor_group.zero_grad()
@@ -220,6 +278,9 @@ class SACEngine:
     def _critic_q(self, states, actions):
         if self._use_fused(states):
             x = torch.cat([states, actions], dim=-1)
+            if getattr(self, "_bf16", False):
+                return Fops.twin_mlp_forward_bf16(x, *self._twin_local,
+                                                  self._twin_local_bf16)
             return Fops.twin_mlp_forward(x, *self._twin_local)
         if self.variant in ("sac", "vsac"):
             return (self.local_critic_1(states, actions),
@@ -229,6 +290,9 @@ class SACEngine:
     def _target_q(self, states, actions):
         if self._use_fused(states):
             x = torch.cat([states, actions], dim=-1)
+            if getattr(self, "_bf16", False):
+                return Fops.twin_mlp_forward_bf16(x, *self._twin_target,
+                                                  self._twin_target_bf16)
             return Fops.twin_mlp_forward(x, *self._twin_target)
         if self.variant in ("sac", "vsac"):
             return (self.target_critic_1(states, actions),
@@ -374,7 +438,11 @@ class SACEngine:
         # profiles/r04_NOTES.md), so the TD block stays on the main stream.
         x_cat = torch.cat([next_states, states], dim=0)
         ws, bs = self._actor_weights()
-        out = Fops.mlp_forward(x_cat, ws, bs, grad_row_start=B)
+        if getattr(self, "_bf16", False):
+            out = Fops.mlp_forward_bf16(x_cat, ws, bs, self._actor_ws_bf16,
+                                        grad_row_start=B)
+        else:
+            out = Fops.mlp_forward(x_cat, ws, bs, grad_row_start=B)
         mu, lsr = out[:, :A], out[:, A:]
         if self._eps_queue:
             eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
@@ -401,10 +469,16 @@ class SACEngine:
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()
+        self.refresh_bf16("critic")
 
         # --- actor/alpha step (post-critic-step critic, frozen heads) --
         xa = torch.cat([states, sampled_actions], dim=-1)
-        aq1, aq2 = Fops.twin_mlp_forward(xa, *self._twin_local_frozen)
+        if getattr(self, "_bf16", False):
+            aq1, aq2 = Fops.twin_mlp_forward_bf16(xa,
+                                                  *self._twin_local_frozen,
+                                                  self._twin_local_bf16)
+        else:
+            aq1, aq2 = Fops.twin_mlp_forward(xa, *self._twin_local_frozen)
         policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(
             aq1, aq2, log_probs, log_stds, states, self.log_alpha, T, use_w,
             self.H_bar_f)
@@ -414,9 +488,11 @@ class SACEngine:
             self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
+        self.refresh_bf16("actor")
         self.alpha = self.log_alpha.exp().detach()
 
         flat_polyak_(self.target_group, self.critic_group, self.tau)
+        self.refresh_bf16("target")
         return {
             "critic_loss": q_loss.detach(),
             "actor_loss": policy_loss.detach(),

@@ -1,0 +1,439 @@
+// bf16_gemm.hip — bf16 GEMM family for the SAC hot path (gfx950).
+//
+// Mixed precision: fp32 master weights (Adam) with bf16 mirrors; GEMMs run
+// on v_mfma_f32_16x16x32_bf16 (K=32 per instruction — 8x the fp32 MFMA's
+// K=4 at similar issue cost) with fp32 accumulation; weight grads (dW/db)
+// accumulate and store fp32.  Halves LDS/HBM traffic vs the fp32 path.
+//
+// Fragment maps (verified by the identity-A/asymmetric-B GPU test):
+//   A: lane l -> row i = l&15,  k = (l>>4)*8 .. +8   (8 bf16 = 4 VGPRs)
+//   B: lane l -> col j = l&15,  k = (l>>4)*8 .. +8
+//   C/D: col = l&15, row = (l>>4)*4 + reg (4 fp32) — same as fp32 MFMA.
+//
+// LDS images are arranged so every fragment read is ONE 16-byte short8
+// read of 8 consecutive bf16: operands whose fragment k-run crosses the
+// global minor axis are stored transposed (scatter ds_write, like the
+// fp32 dW kernel's dy^T tile).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using u16 = unsigned short;
+
+static constexpr int TBM = 64;   // rows per tile
+static constexpr int TBN = 64;   // cols per tile
+static constexpr int TBK = 64;   // reduction per tile (2 mfma k-steps)
+// bf16 row pad: +8 elements (16B) keeps the 16-lane b128 groups spread
+static constexpr int TPAD = TBK + 8;   // 72 bf16 = 144 B rows
+
+#define CHECK_BF16(t) TORCH_CHECK((t).is_cuda() && (t).scalar_type() == torch::kBFloat16, \
+                                  #t " must be a bf16 HIP tensor")
+#define CHECK_F32(t) TORCH_CHECK((t).is_cuda() && (t).scalar_type() == torch::kFloat32, \
+                                 #t " must be a fp32 HIP tensor")
+
+static inline hipStream_t cur_stream2() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+__device__ __forceinline__ u16 f32_to_bf16_rne(float f) {
+  union { float f; unsigned u; } v{f};
+  unsigned u = v.u;
+  u += 0x7FFFu + ((u >> 16) & 1u);   // round-to-nearest-even
+  return (u16)(u >> 16);
+}
+
+__global__ __launch_bounds__(256) void k_cast_f32_bf16(
+    const float* __restrict__ src, u16* __restrict__ dst, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dst[i] = f32_to_bf16_rne(src[i]);
+}
+
+__global__ __launch_bounds__(256) void k_cast_bf16_f32(
+    const u16* __restrict__ src, float* __restrict__ dst, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  union { float f; unsigned u; } v;
+  v.u = ((unsigned)src[i]) << 16;
+  dst[i] = v.f;
+}
+
+// ---------------------------------------------------------------------------
+// fwd: y[g,M,N] = act(x[M,K]_bf16 @ w[g,N,K]_bf16^T + b[g,N]_f32)
+// Both LDS tiles are k-minor (row-major [row][k]) — fragment k-runs are
+// contiguous for A (x rows) and B (w rows).  grid (M/64, N/64, G).
+// out_f32: 1 -> write fp32 y; 0 -> write bf16 y.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_bf16_fwd(
+    const u16* __restrict__ x, const u16* __restrict__ w,
+    const float* __restrict__ b, void* __restrict__ y,
+    int M, int N, int K, int act, long xgs, int out_f32) {
+  __shared__ u16 sx[2][TBM][TPAD];
+  __shared__ u16 sw[2][TBN][TPAD];
+  const long g = blockIdx.z;
+  x += g * xgs;
+  w += g * (long)N * K;
+  b += g * (long)N;
+  const int m0 = blockIdx.x * TBM, n0 = blockIdx.y * TBN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
+  const int fi = lane & 15, fk = lane >> 4;
+  f32x4 acc00{}, acc01{}, acc10{}, acc11{};
+
+  // loader: 64x64 u16 tile = 4096 elems / 256 thr = 16 per thread
+  const int lr = tid >> 2, lc0 = (tid & 3) * 16;
+#define LOAD16(dst_row, src, rowlim, gk0)                                   \
+  {                                                                         \
+    const int r_ = lr;                                                      \
+    _Pragma("unroll") for (int j = 0; j < 16; ++j) {                        \
+      const int gk = (gk0) + lc0 + j;                                       \
+      dst_row[lc0 + j] = (r_ < (rowlim) && gk < K)                          \
+          ? src[(long)r_ * K + gk] : (u16)0;                                \
+    }                                                                       \
+  }
+
+  for (int k0 = 0, buf = 0; k0 < K; k0 += TBK, buf ^= 1) {
+    {
+      const u16* xs = x + (long)m0 * K;
+      const u16* wsrc = w + (long)n0 * K;
+      LOAD16(sx[buf][lr], xs, (M - m0 < TBM ? M - m0 : TBM), k0)
+      LOAD16(sw[buf][lr], wsrc, (N - n0 < TBN ? N - n0 : TBN), k0)
+    }
+    __syncthreads();
+#pragma unroll
+    for (int kk = 0; kk < TBK; kk += 32) {
+      const bf16x8 a0 = *(const bf16x8*)&sx[buf][wr + fi][kk + fk * 8];
+      const bf16x8 a1 = *(const bf16x8*)&sx[buf][wr + 16 + fi][kk + fk * 8];
+      const bf16x8 b0 = *(const bf16x8*)&sw[buf][wc + fi][kk + fk * 8];
+      const bf16x8 b1 = *(const bf16x8*)&sw[buf][wc + 16 + fi][kk + fk * 8];
+      acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
+  float* yf = (float*)y + g * (long)M * N;
+  u16* yh = (u16*)y + g * (long)M * N;
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const f32x4 a = *accs[mi][ni];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wr + mi * 16 + fk * 4 + r;
+        const int col = n0 + wc + ni * 16 + fi;
+        if (row < M && col < N) {
+          float v = a[r] + b[col];
+          if (act == 1) v = fmaxf(v, 0.f);
+          if (out_f32) yf[(long)row * N + col] = v;
+          else yh[(long)row * N + col] = f32_to_bf16_rne(v);
+        }
+      }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// dx: dx[M,K] = sum_g (dy*mask)[g,M,N] @ w[g,N,K]
+// A = dy (m-major, n-minor: k-run over n contiguous) — natural layout.
+// B = w: fragment k-run is over n at fixed k column -> store w TRANSPOSED
+// [k][n] (scatter write).  grid (M/64, K/64, Gz); per-group via grid.z
+// (sum_over_g=0) or summed inner loop (=1).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_bf16_dx(
+    const u16* __restrict__ dy, const u16* __restrict__ w,
+    const u16* __restrict__ yout, u16* __restrict__ dx,
+    int M, int N, int K, int act, int G) {
+  __shared__ u16 sdy[2][TBM][TPAD];     // [m][n-run]
+  __shared__ u16 swT[2][TBN][TPAD];     // [k][n-run]  (transposed w)
+  const long z = blockIdx.z;
+  dy += z * (long)M * N;
+  w += z * (long)N * K;
+  yout += z * (long)M * N;
+  dx += z * (long)M * K;
+  const int m0 = blockIdx.x * TBM, c0 = blockIdx.y * TBN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
+  const int fi = lane & 15, fk = lane >> 4;
+  f32x4 acc00{}, acc01{}, acc10{}, acc11{};
+
+  const int lr = tid >> 2, lc0 = (tid & 3) * 16;
+  for (int g = 0; g < G; ++g) {
+    const u16* dyg = dy + (long)g * M * N;
+    const u16* wg = w + (long)g * N * K;
+    const u16* yg = yout + (long)g * M * N;
+    for (int n0 = 0, buf = 0; n0 < N; n0 += TBK, buf ^= 1) {
+      // dy tile [64 m][64 n] with fused relu mask
+      {
+        const int gm = m0 + lr;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int gn = n0 + lc0 + j;
+          u16 v = 0;
+          if (gm < M && gn < N) {
+            v = dyg[(long)gm * N + gn];
+            if (act == 1 && yg[(long)gm * N + gn] == 0) v = 0;
+          }
+          sdy[buf][lr][lc0 + j] = v;
+        }
+      }
+      // w tile transposed: read w[n0+lr][c0+lc0+j] -> swT[lc0+j][lr]
+      {
+        const int gn = n0 + lr;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int gk = c0 + lc0 + j;
+          swT[buf][lc0 + j][lr] =
+              (gn < N && gk < K) ? wg[(long)gn * K + gk] : (u16)0;
+        }
+      }
+      __syncthreads();
+#pragma unroll
+      for (int kk = 0; kk < TBK; kk += 32) {
+        const bf16x8 a0 = *(const bf16x8*)&sdy[buf][wr + fi][kk + fk * 8];
+        const bf16x8 a1 = *(const bf16x8*)&sdy[buf][wr + 16 + fi][kk + fk * 8];
+        const bf16x8 b0 = *(const bf16x8*)&swT[buf][wc + fi][kk + fk * 8];
+        const bf16x8 b1 = *(const bf16x8*)&swT[buf][wc + 16 + fi][kk + fk * 8];
+        acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+  const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const f32x4 a = *accs[mi][ni];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wr + mi * 16 + fk * 4 + r;
+        const int col = c0 + wc + ni * 16 + fi;
+        if (row < M && col < K) dx[(long)row * K + col] = f32_to_bf16_rne(a[r]);
+      }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// dwdb: dw[g,N,K]_f32 = (dy*mask)^T @ x; db fused; split-K over the batch
+// into fp32 partials (reuses the fp32 k_reduce_partials for the fold).
+// A = dy^T: [n][m-run] (transposed store); B = x: fragment k-run over m at
+// fixed k -> x transposed [k][m-run].  grid (N/64, K/64, G*S).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
+    const u16* __restrict__ dy, const u16* __restrict__ x,
+    const u16* __restrict__ yout, float* __restrict__ ws,
+    float* __restrict__ ws_db, int M, int N, int K, int act, int S,
+    int chunk, long xgs) {
+  __shared__ u16 sa[2][TBN][TPAD];    // dy^T: [n][m-run]
+  __shared__ u16 sbT[2][TBN][TPAD];   // x^T:  [k][m-run]
+  const int gs = blockIdx.z;
+  const int g = gs / S, s = gs % S;
+  const u16* dyg = dy + (long)g * M * N;
+  const u16* yg = yout + (long)g * M * N;
+  const u16* xg = x + (long)g * xgs;
+  float* wsp = ws + (long)gs * N * K;
+  float* dbp = ws_db + (long)gs * N;
+  const int m_lo = s * chunk;
+  const int m_hi = min(M, m_lo + chunk);
+  const int n0 = blockIdx.x * TBM, c0 = blockIdx.y * TBN;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
+  const int fi = lane & 15, fk = lane >> 4;
+  const bool do_db = (blockIdx.y == 0);
+  float db_acc = 0.f;
+  f32x4 acc00{}, acc01{}, acc10{}, acc11{};
+
+  const int lr = tid >> 2, lc0 = (tid & 3) * 16;
+  for (int m0 = m_lo, buf = 0; m0 < m_hi; m0 += TBK, buf ^= 1) {
+    {
+      const int gm = m0 + lr;   // this thread's batch row
+      // dy block [64 m][64 n] -> sa[n][m] (scatter), masked
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int gn = n0 + lc0 + j;
+        u16 v = 0;
+        if (gm < m_hi && gn < N) {
+          v = dyg[(long)gm * N + gn];
+          if (act == 1 && yg[(long)gm * N + gn] == 0) v = 0;
+        }
+        sa[buf][lc0 + j][lr] = v;
+      }
+      // x block [64 m][64 k] -> sbT[k][m] (scatter)
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int gk = c0 + lc0 + j;
+        sbT[buf][lc0 + j][lr] =
+            (gm < m_hi && gk < K) ? xg[(long)gm * K + gk] : (u16)0;
+      }
+    }
+    __syncthreads();
+    if (do_db && tid < TBN) {
+#pragma unroll
+      for (int m = 0; m < TBK; ++m) {
+        union { float f; unsigned u; } v;
+        v.u = ((unsigned)sa[buf][tid][m]) << 16;
+        db_acc += v.f;
+      }
+    }
+#pragma unroll
+    for (int kk = 0; kk < TBK; kk += 32) {
+      const bf16x8 a0 = *(const bf16x8*)&sa[buf][wr + fi][kk + fk * 8];
+      const bf16x8 a1 = *(const bf16x8*)&sa[buf][wr + 16 + fi][kk + fk * 8];
+      const bf16x8 b0 = *(const bf16x8*)&sbT[buf][wc + fi][kk + fk * 8];
+      const bf16x8 b1 = *(const bf16x8*)&sbT[buf][wc + 16 + fi][kk + fk * 8];
+      acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  if (do_db && tid < TBN && n0 + tid < N) dbp[n0 + tid] = db_acc;
+  const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) {
+      const f32x4 a = *accs[mi][ni];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = n0 + wr + mi * 16 + fk * 4 + r;
+        const int col = c0 + wc + ni * 16 + fi;
+        if (row < N && col < K) wsp[(long)row * K + col] = a[r];
+      }
+    }
+}
+
+// fp32 partial fold (duplicate of the fp32 TU's reducer; no RDC linking)
+__global__ __launch_bounds__(256) void k_reduce_partials2(
+    const float* __restrict__ ws, float* __restrict__ out, long stride,
+    int S, long n_per_g) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int g = blockIdx.y;
+  if (i >= n_per_g) return;
+  const float* base = ws + ((long)g * S) * stride + i;
+  float acc = 0.f;
+  for (int s = 0; s < S; ++s) acc += base[(long)s * stride];
+  out[(long)g * n_per_g + i] = acc;
+}
+
+// ===========================================================================
+// Host wrappers
+// ===========================================================================
+
+static void f32_to_bf16_(torch::Tensor src, torch::Tensor dst) {
+  CHECK_F32(src); CHECK_BF16(dst);
+  const long n = src.numel();
+  TORCH_CHECK(dst.numel() == n);
+  hipLaunchKernelGGL(k_cast_f32_bf16, dim3((n + 255) / 256), dim3(256), 0,
+                     cur_stream2(), src.data_ptr<float>(),
+                     (u16*)dst.data_ptr(), n);
+}
+
+static torch::Tensor linear_act_fwd_bf16(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, long act, long G,
+                                         long out_f32) {
+  CHECK_BF16(x); CHECK_BF16(w); CHECK_F32(b);
+  auto xc = x.contiguous(); auto wc = w.contiguous(); auto bc = b.contiguous();
+  const bool per_group_x = xc.dim() == 3;
+  const long M = per_group_x ? xc.size(1) : xc.size(0);
+  const long K = xc.size(-1);
+  const long N = wc.numel() / (G * K);
+  const long xgs = per_group_x ? M * K : 0;
+  auto opts = xc.options().dtype(out_f32 ? torch::kFloat32 : torch::kBFloat16);
+  auto y = G == 1 ? torch::empty({M, N}, opts)
+                  : torch::empty({G, M, N}, opts);
+  dim3 grid((M + TBM - 1) / TBM, (N + TBN - 1) / TBN, G);
+  hipLaunchKernelGGL(k_bf16_fwd, grid, dim3(256), 0, cur_stream2(),
+                     (const u16*)xc.data_ptr(), (const u16*)wc.data_ptr(),
+                     bc.data_ptr<float>(), y.data_ptr(),
+                     (int)M, (int)N, (int)K, (int)act, xgs, (int)out_f32);
+  return y;
+}
+
+static torch::Tensor linear_bwd_dx_bf16(torch::Tensor dy, torch::Tensor w,
+                                        torch::Tensor yout, long act,
+                                        long G, long sum_over_g) {
+  CHECK_BF16(dy); CHECK_BF16(w); CHECK_BF16(yout);
+  auto dyc = dy.contiguous(); auto wc = w.contiguous();
+  auto yc = yout.contiguous();
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  const long N = G == 1 ? dyc.size(1) : dyc.size(2);
+  const long K = wc.size(-1);
+  const long Gz = (G == 1 || sum_over_g) ? 1 : G;
+  const int Gin = (int)((G > 1 && sum_over_g) ? G : 1);
+  auto dx = Gz == 1 ? torch::empty({M, K}, dyc.options())
+                    : torch::empty({Gz, M, K}, dyc.options());
+  dim3 grid((M + TBM - 1) / TBM, (K + TBN - 1) / TBN, Gz);
+  hipLaunchKernelGGL(k_bf16_dx, grid, dim3(256), 0, cur_stream2(),
+                     (const u16*)dyc.data_ptr(), (const u16*)wc.data_ptr(),
+                     (const u16*)yc.data_ptr(), (u16*)dx.data_ptr(),
+                     (int)M, (int)N, (int)K, (int)act, Gin);
+  return dx;
+}
+
+static std::vector<torch::Tensor> linear_bwd_dwdb_bf16(
+    torch::Tensor dy, torch::Tensor x, torch::Tensor yout, long act,
+    long G) {
+  CHECK_BF16(dy); CHECK_BF16(x); CHECK_BF16(yout);
+  auto dyc = dy.contiguous(); auto xc = x.contiguous();
+  auto yc = yout.contiguous();
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  const long N = G == 1 ? dyc.size(1) : dyc.size(2);
+  const long K = xc.size(-1);
+  const long xgs = xc.dim() == 3 ? M * K : 0;
+  const long nbx = (N + TBM - 1) / TBM, nby = (K + TBN - 1) / TBN;
+  long S = std::max<long>(1, 512 / std::max<long>(1, nbx * nby * G));
+  S = std::min<long>(S, (M + TBK - 1) / TBK);
+  const int chunk = (int)(((M + S - 1) / S + TBK - 1) / TBK * TBK);
+  S = (M + chunk - 1) / chunk;
+  auto fopts = dyc.options().dtype(torch::kFloat32);
+  auto ws = torch::empty({G * S, N, K}, fopts);
+  auto ws_db = torch::empty({G * S, N}, fopts);
+  dim3 grid(nbx, nby, G * S);
+  hipLaunchKernelGGL(k_bf16_dwdb_splitk, grid, dim3(256), 0, cur_stream2(),
+                     (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
+                     (const u16*)yc.data_ptr(), ws.data_ptr<float>(),
+                     ws_db.data_ptr<float>(), (int)M, (int)N, (int)K,
+                     (int)act, (int)S, chunk, xgs);
+  auto dw = G == 1 ? torch::empty({N, K}, fopts)
+                   : torch::empty({G, N, K}, fopts);
+  auto db = G == 1 ? torch::empty({N}, fopts)
+                   : torch::empty({G, N}, fopts);
+  if (S == 1) {
+    dw.copy_(ws.view_as(dw));
+    db.copy_(ws_db.view_as(db));
+  } else {
+    const long nw = N * K;
+    dim3 g1((nw + 255) / 256, G);
+    hipLaunchKernelGGL(k_reduce_partials2, g1, dim3(256), 0, cur_stream2(),
+                       ws.data_ptr<float>(), dw.data_ptr<float>(), nw,
+                       (int)S, nw);
+    dim3 g2((N + 255) / 256, G);
+    hipLaunchKernelGGL(k_reduce_partials2, g2, dim3(256), 0, cur_stream2(),
+                       ws_db.data_ptr<float>(), db.data_ptr<float>(), N,
+                       (int)S, N);
+  }
+  return {dw, db};
+}
+
+void register_bf16(pybind11::module_& m) {
+  m.def("f32_to_bf16_", &f32_to_bf16_);
+  m.def("linear_act_fwd_bf16", &linear_act_fwd_bf16);
+  m.def("linear_bwd_dx_bf16", &linear_bwd_dx_bf16);
+  m.def("linear_bwd_dwdb_bf16", &linear_bwd_dwdb_bf16);
+}

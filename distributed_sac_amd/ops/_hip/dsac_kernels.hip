@@ -80,25 +80,34 @@ __global__ __launch_bounds__(256) void k_linear_act_fwd(
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
   const int lr = (tid * 8) >> 5, lc = (tid * 8) & 31;  // this thread's slot
-  float rx[8], rw[8];
-#define LOAD_TILE_FWD(k0)                                                   \
+  // 2-tiles-ahead prefetch: two register sets; loads for tile t+2 issue
+  // while tile t computes, so each load gets ~2 MFMA phases of latency
+  // cover (the 1-deep version stalled: HBM latency > one MFMA phase).
+  float rxa[8], rwa[8], rxb[8], rwb[8];
+#define LOAD_TILE_FWD(k0, rx, rw)                                           \
   _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
     const int gk = (k0) + lc + j;                                           \
     rx[j] = (m0 + lr < M && gk < K) ? x[(long)(m0 + lr) * K + gk] : 0.f;    \
     rw[j] = (n0 + lr < N && gk < K) ? w[(long)(n0 + lr) * K + gk] : 0.f;    \
   }
-#define STORE_TILE_FWD(buf)                                                 \
+#define STORE_TILE_FWD(buf, rx, rw)                                         \
   _Pragma("unroll") for (int j = 0; j < 8; ++j) {                           \
     sx[buf][lr][lc + j] = rx[j];                                            \
     sw[buf][lr][lc + j] = rw[j];                                            \
   }
-  LOAD_TILE_FWD(0)
-  STORE_TILE_FWD(0)
+  LOAD_TILE_FWD(0, rxa, rwa)
+  STORE_TILE_FWD(0, rxa, rwa)
+  if (BK < K) LOAD_TILE_FWD(BK, rxa, rwa)
   __syncthreads();
   int cur = 0;
   for (int k0 = 0; k0 < K; k0 += BK) {
-    const bool more = k0 + BK < K;
-    if (more) LOAD_TILE_FWD(k0 + BK)
+    // issue loads for tile t+2 into the register set whose data for tile
+    // t+1 has NOT yet been written to LDS?  No: set A holds t+1 (issued
+    // last iter / prologue); set B receives t+2 now; at loop end we write
+    // t+1 (set A) into the other LDS buffer and swap the sets.
+    const bool have_next = k0 + BK < K;
+    const bool have_next2 = k0 + 2 * BK < K;
+    if (have_next2) LOAD_TILE_FWD(k0 + 2 * BK, rxb, rwb)
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 4) {
       const float a0 = sx[cur][wr + fi][kk + fk];
@@ -110,7 +119,15 @@ __global__ __launch_bounds__(256) void k_linear_act_fwd(
       acc10 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc10, 0, 0, 0);
       acc11 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc11, 0, 0, 0);
     }
-    if (more) STORE_TILE_FWD(cur ^ 1)
+    if (have_next) {
+      STORE_TILE_FWD(cur ^ 1, rxa, rwa)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {  // swap register sets (compiled away)
+        const float tx = rxa[j], tw = rwa[j];
+        rxa[j] = rxb[j]; rwa[j] = rwb[j];
+        rxb[j] = tx; rwb[j] = tw;
+      }
+    }
     __syncthreads();
     cur ^= 1;
   }
@@ -1153,9 +1170,11 @@ static void polyak_(torch::Tensor t, torch::Tensor s, double tau) {
 }
 
 void register_shm_ring(pybind11::module_& m);
+void register_bf16(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   register_shm_ring(mod);
+  register_bf16(mod);
   mod.def("linear_act_fwd", &linear_act_fwd, "fused GEMM+bias+act forward");
   mod.def("linear_act_fwd_g", &linear_act_fwd_g, "grouped (twin) variant");
   mod.def("linear_bwd_dx", &linear_bwd_dx, "GEMM backward dX (fused mask)");

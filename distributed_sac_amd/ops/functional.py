@@ -340,3 +340,113 @@ gather_log_alpha = torch_ref.gather_log_alpha
 entropy_from_log_std = torch_ref.entropy_from_log_std
 batched_linear = torch_ref.batched_linear
 attention_pool = torch_ref.attention_pool
+
+
+# ---------------------------------------------------------------------------
+# bf16 mixed-precision GEMM paths (fp32 master weights + bf16 compute
+# mirrors; fp32 accumulation and fp32 weight grads).  The bf16 mirrors ride
+# along as non-tracked args; the fp32 parameters are the autograd inputs so
+# dW/db accumulate into the flat fp32 gradient buffers.
+# ---------------------------------------------------------------------------
+
+class _FusedMLPBF16(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, n_layers: int, grad_row_start: int,
+                ws_bf16, *wb_f32):
+        ext = native()
+        bs = wb_f32[n_layers:]
+        xh = x.to(torch.bfloat16)
+        acts = [xh]
+        h = xh
+        for i in range(n_layers):
+            last = i == n_layers - 1
+            act = ACT_NONE if last else ACT_RELU
+            h = ext.linear_act_fwd_bf16(h, ws_bf16[i], bs[i].contiguous(),
+                                        act, 1, 1 if last else 0)
+            acts.append(h)
+        ctx.save_for_backward(*acts[:-1])
+        ctx.ws_bf16 = ws_bf16
+        ctx.n_layers = n_layers
+        ctx.grad_row_start = grad_row_start
+        return h  # fp32
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        n = ctx.n_layers
+        acts = list(ctx.saved_tensors)  # bf16: input + hidden outputs
+        ws = ctx.ws_bf16
+        ext = native()
+        r0 = ctx.grad_row_start
+        if r0:
+            acts = [a[r0:] for a in acts]
+            grad_out = grad_out[r0:]
+        dy = grad_out.contiguous().to(torch.bfloat16)
+        dws = [None] * n
+        dbs = [None] * n
+        for i in range(n - 1, -1, -1):
+            act = ACT_RELU if i < n - 1 else ACT_NONE
+            yout = acts[i + 1] if i < n - 1 else acts[i]  # dummy when act=0
+            dw, db = ext.linear_bwd_dwdb_bf16(dy, acts[i], yout, act, 1)
+            dws[i], dbs[i] = dw, db
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, ws[i], yout, act, 1, 1)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.linear_bwd_dx_bf16(
+                dy, ws[0], acts[1] if n > 1 else acts[0],
+                ACT_RELU if n > 1 else ACT_NONE, 1, 1).to(torch.float32)
+        return (dx, None, None, None, *dws, *dbs)
+
+
+def mlp_forward_bf16(x, ws_f32, bs_f32, ws_bf16, grad_row_start: int = 0):
+    return _FusedMLPBF16.apply(x, len(ws_f32), grad_row_start,
+                               tuple(ws_bf16), *ws_f32, *bs_f32)
+
+
+class _FusedTwinMLPBF16(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, n_layers: int, ws_bf16, *wb_f32):
+        ext = native()
+        bs = wb_f32[n_layers:]
+        xh = x.to(torch.bfloat16)
+        acts = [xh]
+        h = xh
+        for i in range(n_layers):
+            last = i == n_layers - 1
+            act = ACT_NONE if last else ACT_RELU
+            h = ext.linear_act_fwd_bf16(h, ws_bf16[i], bs[i].contiguous(),
+                                        act, 2, 1 if last else 0)
+            acts.append(h)
+        ctx.save_for_backward(*acts[:-1])
+        ctx.ws_bf16 = ws_bf16
+        ctx.n_layers = n_layers
+        return h[0], h[1]  # fp32 [B, 1] each
+
+    @staticmethod
+    def backward(ctx, dq1, dq2):
+        n = ctx.n_layers
+        acts = list(ctx.saved_tensors)
+        ws = ctx.ws_bf16
+        ext = native()
+        dy = torch.stack([dq1, dq2], dim=0).contiguous().to(torch.bfloat16)
+        dws = [None] * n
+        dbs = [None] * n
+        for i in range(n - 1, -1, -1):
+            act = ACT_RELU if i < n - 1 else ACT_NONE
+            yout = acts[i + 1] if i < n - 1 else acts[i]
+            if ctx.needs_input_grad[2 + i]:
+                dw, db = ext.linear_bwd_dwdb_bf16(dy, acts[i], yout, act, 2)
+                dws[i], dbs[i] = dw, db
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, ws[i], yout, act, 2, 0)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.linear_bwd_dx_bf16(
+                dy, ws[0], acts[1] if n > 1 else acts[0],
+                ACT_RELU if n > 1 else ACT_NONE, 2, 1).to(torch.float32)
+        return (dx, None, None, *dws, *dbs)
+
+
+def twin_mlp_forward_bf16(x, stacked_ws_f32, stacked_bs_f32, ws_bf16):
+    return _FusedTwinMLPBF16.apply(x, len(ws_bf16), tuple(ws_bf16),
+                                   *stacked_ws_f32, *stacked_bs_f32)

@@ -426,3 +426,130 @@ def test_td_target_mt_kernel():
     alpha = la.exp()[idx].unsqueeze(-1)
     ref = 1.5 * r + 0.99 * (1 - d) * (torch.min(q1, q2) - alpha * lp)
     assert torch.allclose(y, ref, atol=1e-5, rtol=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# bf16 mixed-precision path
+# ---------------------------------------------------------------------------
+
+def test_bf16_fragment_layout_identity():
+    """A=I with asymmetric B: catches any transposed/miswired fragment map
+    (guide rule: symmetric B would silently pass a row<->col swap)."""
+    ext = req_native()
+    M = N = K = 64
+    A = torch.eye(64, device="cuda")
+    B = (torch.arange(64.0, device="cuda").view(64, 1) * 100
+         + torch.arange(64.0, device="cuda").view(1, 64))  # B[i,j]=100i+j
+    # y = A @ W^T with W = B --> y[i,j] = B[j,i]
+    y = ext.linear_act_fwd_bf16(A.to(torch.bfloat16),
+                                B.to(torch.bfloat16),
+                                torch.zeros(64, device="cuda"), 0, 1, 1)
+    ref = B.t()
+    assert torch.allclose(y, ref, atol=16.0, rtol=1e-2), \
+        (y - ref).abs().max().item()  # bf16 quantizes 100i+j to ~1% rel
+
+
+@pytest.mark.parametrize("M,N,K", [(1280, 400, 400), (1280, 400, 49),
+                                   (2560, 16, 400), (100, 7, 3)])
+def test_bf16_fwd(M, N, K):
+    ext = req_native()
+    torch.manual_seed(20)
+    x = torch.randn(M, K, device="cuda")
+    w = torch.randn(N, K, device="cuda") / K ** 0.5
+    b = torch.randn(N, device="cuda")
+    xh, wh = x.to(torch.bfloat16), w.to(torch.bfloat16)
+    y = ext.linear_act_fwd_bf16(xh, wh, b, 1, 1, 1)  # fp32 out
+    ref = torch.relu(torch.nn.functional.linear(
+        xh.float(), wh.float(), b))
+    assert torch.allclose(y, ref, atol=5e-2, rtol=2e-2), \
+        (y - ref).abs().max().item()
+    yh = ext.linear_act_fwd_bf16(xh, wh, b, 1, 1, 0)  # bf16 out
+    assert torch.allclose(yh.float(), ref, atol=2e-1, rtol=2e-2)
+
+
+def test_bf16_grouped_and_bwd():
+    ext = req_native()
+    torch.manual_seed(21)
+    B, N, K, G = 1280, 400, 53, 2
+    x = torch.randn(B, K, device="cuda").to(torch.bfloat16)
+    w = (torch.randn(G, N, K, device="cuda") / K ** 0.5).to(torch.bfloat16)
+    b = torch.randn(G, N, device="cuda")
+    y = ext.linear_act_fwd_bf16(x, w, b, 1, G, 0)
+    assert y.shape == (G, B, N) and y.dtype == torch.bfloat16
+    for g in range(G):
+        ref = torch.relu(torch.nn.functional.linear(x.float(), w[g].float(),
+                                                    b[g]))
+        assert torch.allclose(y[g].float(), ref, atol=2e-1, rtol=2e-2)
+
+    dy = torch.randn(G, B, N, device="cuda").to(torch.bfloat16)
+    # dx summed over groups, relu-masked by y
+    dx = ext.linear_bwd_dx_bf16(dy, w, y, 1, G, 1)
+    ref_dx = sum((dy[g].float() * (y[g].float() > 0)) @ w[g].float()
+                 for g in range(G))
+    assert torch.allclose(dx.float(), ref_dx, atol=1.5, rtol=3e-2), \
+        (dx.float() - ref_dx).abs().max().item()
+    # per-group dx
+    dxp = ext.linear_bwd_dx_bf16(dy, w, y, 1, G, 0)
+    for g in range(G):
+        ref_g = (dy[g].float() * (y[g].float() > 0)) @ w[g].float()
+        assert torch.allclose(dxp[g].float(), ref_g, atol=1.0, rtol=3e-2)
+    # dwdb fp32 out
+    dw, db = ext.linear_bwd_dwdb_bf16(dy, x, y, 1, G)
+    assert dw.dtype == torch.float32
+    for g in range(G):
+        dy_m = dy[g].float() * (y[g].float() > 0)
+        ref_dw = dy_m.t() @ x.float()
+        ref_db = dy_m.sum(0)
+        assert torch.allclose(dw[g], ref_dw, atol=2.0, rtol=3e-2), \
+            (dw[g] - ref_dw).abs().max().item()
+        assert torch.allclose(db[g], ref_db, atol=1.0, rtol=3e-2)
+
+
+def test_bf16_engine_update_and_graph():
+    """bf16 engine update: finite, tracks the fp32 engine loosely, graph-
+    capturable, and the critic learns on a fixed batch."""
+    from distributed_sac_amd.algo import SACEngine
+    from distributed_sac_amd.replay import ShardedReplay
+    from tests.test_engine import make_batch, small_cfg
+    torch.manual_seed(0)
+    cfg = small_cfg("mtsac")
+    e32 = SACEngine(cfg, "cuda:0")
+    e16 = SACEngine(cfg, "cuda:0", precision="bf16")
+    e16.actor.load_state_dict(e32.actor.state_dict())
+    e16.local_critic.load_state_dict(e32.local_critic.state_dict())
+    e16.hard_copy_targets()
+    e16.refresh_bf16()
+    e32.hard_copy_targets()
+    for step in range(3):
+        batch = {k: v.cuda() for k, v in make_batch(cfg, seed=step).items()}
+        eps = [torch.randn(cfg.batch_size, cfg.action_dim, device="cuda")
+               for _ in range(2)]
+        e32._eps_queue = [e.clone() for e in eps]
+        e16._eps_queue = [e.clone() for e in eps]
+        m32 = e32.update({k: v.clone() for k, v in batch.items()})
+        m16 = e16.update({k: v.clone() for k, v in batch.items()})
+    assert abs(m32["critic_loss"] - m16["critic_loss"]) < 0.25 + \
+        0.25 * abs(m32["critic_loss"])
+    for v in m16.values():
+        assert v == v
+
+    # graph capture with bf16 kernels
+    replay = ShardedReplay(4000, cfg.num_tasks, cfg.mtobs_dim,
+                           cfg.action_dim, device="cuda:0")
+    for t in range(cfg.num_tasks):
+        n = 256
+        st = torch.randn(n, cfg.mtobs_dim, device="cuda")
+        oh = torch.zeros(n, cfg.num_tasks, device="cuda")
+        oh[:, t] = 1
+        st[:, -cfg.num_tasks:] = oh
+        replay.shards[t].append(
+            st, torch.rand(n, cfg.action_dim, device="cuda") * 2 - 1,
+            torch.randn(n, 1, device="cuda"), st.clone(),
+            torch.zeros(n, 1, device="cuda"))
+    e16.capture(replay, cfg.batch_size)
+    losses = []
+    for _ in range(30):
+        m = e16.graphed_update()
+        losses.append(float(m["critic_loss"]))
+    assert all(v == v for v in losses)
+    assert losses[-1] < losses[0] * 1.5  # not diverging
