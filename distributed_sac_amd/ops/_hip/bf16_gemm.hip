@@ -86,15 +86,27 @@ __global__ __launch_bounds__(256) void k_bf16_fwd(
   const int fi = lane & 15, fk = lane >> 4;
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
-  // loader: 64x64 u16 tile = 4096 elems / 256 thr = 16 per thread
+  // loader: 64x64 u16 tile = 4096 elems / 256 thr = 16 per thread.
+  // Fast path: two 16-byte vector loads + stores when the 16-element run
+  // is fully in range and 16B-aligned (K % 8 == 0 rows) — the scalar
+  // fallback was the dominant stall in the first bf16 cut (PMC: 80%
+  // SQ_WAIT_ANY; profiles/r05_NOTES.md).
   const int lr = tid >> 2, lc0 = (tid & 3) * 16;
+  const bool krows_aligned = (K % 8) == 0;
 #define LOAD16(dst_row, src, rowlim, gk0)                                   \
   {                                                                         \
     const int r_ = lr;                                                      \
-    _Pragma("unroll") for (int j = 0; j < 16; ++j) {                        \
-      const int gk = (gk0) + lc0 + j;                                       \
-      dst_row[lc0 + j] = (r_ < (rowlim) && gk < K)                          \
-          ? src[(long)r_ * K + gk] : (u16)0;                                \
+    if (krows_aligned && r_ < (rowlim) && (gk0) + lc0 + 16 <= K) {          \
+      const uint4* sp = (const uint4*)&src[(long)r_ * K + (gk0) + lc0];     \
+      uint4* dp = (uint4*)&dst_row[lc0];                                    \
+      dp[0] = sp[0];                                                        \
+      dp[1] = sp[1];                                                        \
+    } else {                                                                \
+      _Pragma("unroll") for (int j = 0; j < 16; ++j) {                      \
+        const int gk = (gk0) + lc0 + j;                                     \
+        dst_row[lc0 + j] = (r_ < (rowlim) && gk < K)                        \
+            ? src[(long)r_ * K + gk] : (u16)0;                              \
+      }                                                                     \
     }                                                                       \
   }
 
@@ -175,15 +187,36 @@ __global__ __launch_bounds__(256) void k_bf16_dx(
       // dy tile [64 m][64 n] with fused relu mask
       {
         const int gm = m0 + lr;
+        if ((N % 8) == 0 && gm < M && n0 + lc0 + 16 <= N) {
+          const uint4* dp_ = (const uint4*)&dyg[(long)gm * N + n0 + lc0];
+          uint4 v0 = dp_[0], v1 = dp_[1];
+          if (act == 1) {
+            const uint4* yp = (const uint4*)&yg[(long)gm * N + n0 + lc0];
+            uint4 m0v = yp[0], m1v = yp[1];
+            // zero dy lanes where the (bf16) activation is exactly 0
+            const u16* ym0 = (const u16*)&m0v;
+            const u16* ym1 = (const u16*)&m1v;
+            u16* d0 = (u16*)&v0;
+            u16* d1 = (u16*)&v1;
 #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int gn = n0 + lc0 + j;
-          u16 v = 0;
-          if (gm < M && gn < N) {
-            v = dyg[(long)gm * N + gn];
-            if (act == 1 && yg[(long)gm * N + gn] == 0) v = 0;
+            for (int j = 0; j < 8; ++j) {
+              if (ym0[j] == 0) d0[j] = 0;
+              if (ym1[j] == 0) d1[j] = 0;
+            }
           }
-          sdy[buf][lr][lc0 + j] = v;
+          ((uint4*)&sdy[buf][lr][lc0])[0] = v0;
+          ((uint4*)&sdy[buf][lr][lc0])[1] = v1;
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) {
+            const int gn = n0 + lc0 + j;
+            u16 v = 0;
+            if (gm < M && gn < N) {
+              v = dyg[(long)gm * N + gn];
+              if (act == 1 && yg[(long)gm * N + gn] == 0) v = 0;
+            }
+            sdy[buf][lr][lc0 + j] = v;
+          }
         }
       }
       // w tile transposed: read w[n0+lr][c0+lc0+j] -> swT[lc0+j][lr]

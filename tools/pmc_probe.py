@@ -20,3 +20,20 @@ for _ in range(50):
     y16 = ext.linear_act_fwd_bf16(xh, wh, b2, 1, 2, 0)
 torch.cuda.synchronize()
 print("done")
+# extended A/B: isolated timings
+import time
+def t(fn, iters=200):
+    for _ in range(20): fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters): fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1e6
+print("fp32 fwd G2 400x400:", round(t(lambda: ext.linear_act_fwd_g(x, w, b2, 1, 2)), 2), "us")
+print("bf16 fwd G2 400x400:", round(t(lambda: ext.linear_act_fwd_bf16(xh, wh, b2, 1, 2, 0)), 2), "us")
+dy32 = torch.randn(2, B, H, device="cuda"); y32m = torch.relu(torch.randn(2, B, H, device="cuda"))
+dyh = dy32.to(torch.bfloat16); yhm = y32m.to(torch.bfloat16)
+print("fp32 dx pg:", round(t(lambda: ext.linear_bwd_dx_g(dy32, w, y32m, 1, 2, 0)), 2), "us")
+print("bf16 dx pg:", round(t(lambda: ext.linear_bwd_dx_bf16(dyh, wh, yhm, 1, 2, 0)), 2), "us")
+print("fp32 dwdb:", round(t(lambda: ext.linear_bwd_dwdb_g(dy32, x, y32m, 1, 2)), 2), "us")
+print("bf16 dwdb:", round(t(lambda: ext.linear_bwd_dwdb_bf16(dyh, xh, yhm, 1, 2)), 2), "us")
