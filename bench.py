@@ -41,7 +41,7 @@ def parse_args():
     p.add_argument("--skip-rollout-probe", action="store_true")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph capture of the update step")
-    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="bf16",
                    help="GEMM compute dtype (fp32 masters either way)")
     return p.parse_args()
 
@@ -114,7 +114,9 @@ def main():
 
     ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
     engine = create_engine(cfg, device)
-    if args.dtype == "bf16" and cfg.variant != "care":
+    if cfg.variant == "care":
+        args.dtype = "fp32"  # CARE path runs fp32 (encoder in torch ops)
+    elif args.dtype == "bf16" and device.startswith("cuda"):
         engine.precision = "bf16"
         engine._init_bf16_mirrors()
     if ddp is not None:

@@ -222,11 +222,23 @@ __global__ __launch_bounds__(256) void k_bf16_dx(
       // w tile transposed: read w[n0+lr][c0+lc0+j] -> swT[lc0+j][lr]
       {
         const int gn = n0 + lr;
+        if ((K % 8) == 0 && gn < N && c0 + lc0 + 16 <= K) {
+          const uint4* sp = (const uint4*)&wg[(long)gn * K + c0 + lc0];
+          uint4 v0 = sp[0], v1 = sp[1];
+          const u16* e0 = (const u16*)&v0;
+          const u16* e1 = (const u16*)&v1;
 #pragma unroll
-        for (int j = 0; j < 16; ++j) {
-          const int gk = c0 + lc0 + j;
-          swT[buf][lc0 + j][lr] =
-              (gn < N && gk < K) ? wg[(long)gn * K + gk] : (u16)0;
+          for (int j = 0; j < 8; ++j) {
+            swT[buf][lc0 + j][lr] = e0[j];
+            swT[buf][lc0 + 8 + j][lr] = e1[j];
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) {
+            const int gk = c0 + lc0 + j;
+            swT[buf][lc0 + j][lr] =
+                (gn < N && gk < K) ? wg[(long)gn * K + gk] : (u16)0;
+          }
         }
       }
       __syncthreads();
@@ -295,22 +307,57 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
     {
       const int gm = m0 + lr;   // this thread's batch row
       // dy block [64 m][64 n] -> sa[n][m] (scatter), masked
+      if ((N % 8) == 0 && gm < m_hi && n0 + lc0 + 16 <= N) {
+        const uint4* dp_ = (const uint4*)&dyg[(long)gm * N + n0 + lc0];
+        uint4 v0 = dp_[0], v1 = dp_[1];
+        u16* d0 = (u16*)&v0;
+        u16* d1 = (u16*)&v1;
+        if (act == 1) {
+          const uint4* yp = (const uint4*)&yg[(long)gm * N + n0 + lc0];
+          uint4 y0 = yp[0], y1 = yp[1];
+          const u16* m0_ = (const u16*)&y0;
+          const u16* m1_ = (const u16*)&y1;
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int gn = n0 + lc0 + j;
-        u16 v = 0;
-        if (gm < m_hi && gn < N) {
-          v = dyg[(long)gm * N + gn];
-          if (act == 1 && yg[(long)gm * N + gn] == 0) v = 0;
+          for (int j = 0; j < 8; ++j) {
+            if (m0_[j] == 0) d0[j] = 0;
+            if (m1_[j] == 0) d1[j] = 0;
+          }
         }
-        sa[buf][lc0 + j][lr] = v;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          sa[buf][lc0 + j][lr] = d0[j];
+          sa[buf][lc0 + 8 + j][lr] = d1[j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int gn = n0 + lc0 + j;
+          u16 v = 0;
+          if (gm < m_hi && gn < N) {
+            v = dyg[(long)gm * N + gn];
+            if (act == 1 && yg[(long)gm * N + gn] == 0) v = 0;
+          }
+          sa[buf][lc0 + j][lr] = v;
+        }
       }
       // x block [64 m][64 k] -> sbT[k][m] (scatter)
+      if ((K % 8) == 0 && gm < m_hi && c0 + lc0 + 16 <= K) {
+        const uint4* sp = (const uint4*)&xg[(long)gm * K + c0 + lc0];
+        uint4 v0 = sp[0], v1 = sp[1];
+        const u16* e0 = (const u16*)&v0;
+        const u16* e1 = (const u16*)&v1;
 #pragma unroll
-      for (int j = 0; j < 16; ++j) {
-        const int gk = c0 + lc0 + j;
-        sbT[buf][lc0 + j][lr] =
-            (gm < m_hi && gk < K) ? xg[(long)gm * K + gk] : (u16)0;
+        for (int j = 0; j < 8; ++j) {
+          sbT[buf][lc0 + j][lr] = e0[j];
+          sbT[buf][lc0 + 8 + j][lr] = e1[j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int gk = c0 + lc0 + j;
+          sbT[buf][lc0 + j][lr] =
+              (gm < m_hi && gk < K) ? xg[(long)gm * K + gk] : (u16)0;
+        }
       }
     }
     __syncthreads();
