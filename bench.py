@@ -114,9 +114,7 @@ def main():
 
     ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
     engine = create_engine(cfg, device)
-    if cfg.variant == "care":
-        args.dtype = "fp32"  # CARE path runs fp32 (encoder in torch ops)
-    elif args.dtype == "bf16" and device.startswith("cuda"):
+    if args.dtype == "bf16" and device.startswith("cuda"):
         engine.precision = "bf16"
         engine._init_bf16_mirrors()
     if ddp is not None:

@@ -117,6 +117,7 @@ class CAREEngine(SACEngine):
             with_grad=False)
         self._twin_local_frozen = ([w.detach() for w in self._twin_local[0]],
                                    [b.detach() for b in self._twin_local[1]])
+        self._init_bf16_mirrors()
         self.hard_copy_targets()
         self.tie_actor_state_encoder()
 
@@ -235,30 +236,78 @@ class CAREEngine(SACEngine):
                 "alpha_loss": loss_log_alpha.detach(),
                 "entropy": entropy.detach()}
 
+    def _twin_fwd(self, x, which: str):
+        if getattr(self, "_bf16", False):
+            if which == "target":
+                return Fops.twin_mlp_forward_bf16(x, *self._twin_target,
+                                                  self._twin_target_bf16)
+            if which == "frozen":
+                return Fops.twin_mlp_forward_bf16(x,
+                                                  *self._twin_local_frozen,
+                                                  self._twin_local_bf16)
+            return Fops.twin_mlp_forward_bf16(x, *self._twin_local,
+                                              self._twin_local_bf16)
+        if which == "target":
+            return Fops.twin_mlp_forward(x, *self._twin_target)
+        if which == "frozen":
+            return Fops.twin_mlp_forward(x, *self._twin_local_frozen)
+        return Fops.twin_mlp_forward(x, *self._twin_local)
+
     def _update_tensors_fused(self, batch):
-        """GPU path: fused losses + twin-head grouped GEMMs.  Forwards whose
-        grads the reference discards run under no_grad (see module doc)."""
+        """GPU path: fused losses + twin-head grouped GEMMs (bf16 mirrors
+        when enabled).  Forwards whose grads the reference discards run
+        under no_grad (see module doc); the actor head runs ONE batched
+        pass over cat(next, states) like the SAC engine (the actor-side
+        state encoder is no-grad in both halves)."""
         from ..ops import native
         states, actions = batch["states"], batch["actions"]
         rewards, next_states, dones = (batch["rewards"], batch["next_states"],
                                        batch["dones"])
         T = self.num_tasks
         use_w = self.use_weighted_loss
+        B = states.shape[0]
+        A = self.actor.action_dim
         self.zero_grad()
 
         z_context = self.context_encoder(states)
+        zc_d = z_context.detach()
+
+        # batched actor pass: [next | current], SE under no_grad (grads
+        # into trunk/mixture are discarded by the reference), head grads
+        # only from the states half
         with torch.no_grad():
-            na, nlp, _ = self._sample_care(next_states, z_context)
+            x_cat = torch.cat([next_states, states], dim=0)
+            z2 = torch.cat([zc_d, zc_d], dim=0)
+            enc_cat = self.actor.state_encoder(z2, x_cat, detach_z_encs=True)
+        ws, bs = self._actor_weights()
+        if getattr(self, "_bf16", False):
+            mu_lsr = Fops.mlp_forward_bf16(enc_cat, ws, bs,
+                                           self._actor_ws_bf16,
+                                           grad_row_start=B)
+        else:
+            mu_lsr = Fops.mlp_forward(enc_cat, ws, bs, grad_row_start=B)
+        mu = mu_lsr[:, :A]
+        lsr = mu_lsr[:, A:]
+        if self._eps_queue:
+            eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
+        else:
+            eps = torch.randn_like(mu)
+        a_cat, lp_cat, ls_cat = Fops.squashed_gaussian(mu, lsr, eps,
+                                                       self.actor.k)
+        na, nlp = a_cat[:B].detach(), lp_cat[:B].detach()
+        sa, lp, ls = a_cat[B:], lp_cat[B:], ls_cat[B:]
+
+        with torch.no_grad():
             enc_t = self.target_critic.encode(next_states, z_context)
             xt = torch.cat([enc_t, na], dim=-1)
-            q1_t, q2_t = Fops.twin_mlp_forward(xt, *self._twin_target)
+            q1_t, q2_t = self._twin_fwd(xt, "target")
             y = native().td_target_mt(rewards, dones, q1_t, q2_t, nlp,
                                       states, self.log_alpha.detach(), T,
                                       self.gamma, self.reward_scale)
 
         enc = self.local_critic.encode(states, z_context)
         x = torch.cat([enc, actions], dim=-1)
-        q1, q2 = Fops.twin_mlp_forward(x, *self._twin_local)
+        q1, q2 = self._twin_fwd(x, "local")
         l1, l2 = Fops.critic_loss(q1, q2, y, states, self.log_alpha.detach(),
                                   T, use_w)
         q_loss = l1 + l2
@@ -268,32 +317,25 @@ class CAREEngine(SACEngine):
             if self.context_group is not None:
                 self.ddp.allreduce_grad_(self.context_group.flat_grad)
         self.critic_optimizer.step()
+        self.refresh_bf16("critic")
 
-        zc_d = z_context.detach()
-        with torch.no_grad():
-            # actor-side encoder: grads into trunk are discarded by the
-            # reference (actor optimizer holds only the head) — skip them
-            enc_a = self.actor.state_encoder(zc_d, states, detach_z_encs=True)
-        mu_lsr = self.actor.mu_log_std_layer(enc_a)
-        mu = mu_lsr[:, : self.actor.action_dim]
-        lsr = mu_lsr[:, self.actor.action_dim:]
-        eps = self._next_eps(mu)
-        sa, lp, ls = Fops.squashed_gaussian(mu, lsr, eps, self.actor.k)
         with torch.no_grad():
             enc_c = self.local_critic.encode(states, zc_d, detach_z_encs=True)
         xa = torch.cat([enc_c, sa], dim=-1)
-        aq1, aq2 = Fops.twin_mlp_forward(xa, *self._twin_local_frozen)
+        aq1, aq2 = self._twin_fwd(xa, "frozen")
         policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(
             aq1, aq2, lp, ls, states, self.log_alpha, T, use_w, self.H_bar_f)
         (policy_loss + loss_log_alpha).backward()
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.actor_group.flat_grad)
             self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
-        self.actor_optimizer.step()
-        self.log_alpha_optimizer.step()
+        from ..ops.flat import FusedAdam as _FA
+        _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
+        self.refresh_bf16("actor")
         self.alpha = self.log_alpha.exp().detach()
 
         self._polyak_targets()
+        self.refresh_bf16("target")
         if self.context_encoder_optimizer is not None:
             self.context_encoder_optimizer.step()
         self.tie_actor_state_encoder()
