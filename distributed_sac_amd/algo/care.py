@@ -103,7 +103,9 @@ class CAREEngine(SACEngine):
         self.context_group = FlatParams(ctx_params) if ctx_params else None
 
         self.actor_optimizer = FusedAdam(self.actor_group, lr=cfg.lr_actor)
-        self.critic_optimizer = FusedAdam(self.critic_group, lr=cfg.lr_critic)
+        self.critic_optimizer = FusedAdam(
+            self.critic_group, lr=cfg.lr_critic,
+            ref_params=list(self.local_critic.parameters()))
         self.log_alpha_optimizer = FusedAdam(self.alpha_group, lr=cfg.lr_actor)
         self.context_encoder_optimizer = (
             FusedAdam(self.context_group,

@@ -170,7 +170,17 @@ class SACEngine:
         self.target_group = FlatParams(self._target_params(), with_grad=False)
         self.alpha_group = FlatParams([self.log_alpha])
         self.actor_optimizer = FusedAdam(self.actor_group, lr=cfg.lr_actor)
-        self.critic_optimizer = FusedAdam(self.critic_group, lr=cfg.lr_critic)
+        # checkpoint state is indexed in the REFERENCE optimizer order:
+        # chain(critic1, critic2) for LL/VSAC (learner.build_optimizer),
+        # local_critic.parameters() for MT — not our interleaved layout
+        if self.variant in ("sac", "vsac"):
+            ref_order = (list(self.local_critic_1.parameters())
+                         + list(self.local_critic_2.parameters()))
+        else:
+            ref_order = list(self.local_critic.parameters())
+        self.critic_optimizer = FusedAdam(self.critic_group,
+                                          lr=cfg.lr_critic,
+                                          ref_params=ref_order)
         # reference uses lr_actor for log_alpha (learner.py build_optimizer)
         self.log_alpha_optimizer = FusedAdam(self.alpha_group, lr=cfg.lr_actor)
         self._twin_local = self._build_twin_stacks(
@@ -565,7 +575,10 @@ class SACEngine:
         """Inverse of checkpoint_state — also fixes the reference's broken
         learner resume (it referenced a nonexistent ``actor.optimizer``,
         LunarLander…/src/learner.py:178)."""
-        self.update_iteration = int(ckpt.get("update_iteration", 0))
+        # LL variant saves 'episode_idx' where MT variants save
+        # 'update_iteration' (LunarLander…/src/learner.py:144-163)
+        self.update_iteration = int(ckpt.get("update_iteration",
+                                             ckpt.get("episode_idx", 0)))
         self.total_step = int(ckpt.get("total_step", 0))
         self.actor.load_state_dict(ckpt["actor"])
         if self.variant in ("sac", "vsac"):

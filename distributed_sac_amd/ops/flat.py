@@ -107,8 +107,18 @@ class FusedAdam:
     """
 
     def __init__(self, group: FlatParams, lr: float,
-                 betas=(0.9, 0.999), eps: float = 1e-8):
+                 betas=(0.9, 0.999), eps: float = 1e-8, ref_params=None):
         self.group = group
+        # (de)serialization order: the reference checkpoints index optimizer
+        # state by the REFERENCE's parameter order (e.g. chain(critic1,
+        # critic2)), which may differ from the flat group's layout (we
+        # interleave layer pairs for the stacked twin-GEMM views).
+        self.ref_params = (list(ref_params) if ref_params is not None
+                           else list(group.params))
+        self._ref_offsets = []
+        for rp in self.ref_params:
+            i = next(j for j, q in enumerate(group.params) if q is rp)
+            self._ref_offsets.append(group.offsets[i])
         self.lr = float(lr)
         self.betas = (float(betas[0]), float(betas[1]))
         self.eps = float(eps)
@@ -181,7 +191,7 @@ class FusedAdam:
 
     def state_dict(self) -> Dict:
         state = {}
-        for i, (p, off) in enumerate(zip(self.group.params, self.group.offsets)):
+        for i, (p, off) in enumerate(zip(self.ref_params, self._ref_offsets)):
             n = p.numel()
             state[i] = {
                 "step": torch.tensor(float(self.step_count)),
@@ -192,7 +202,7 @@ class FusedAdam:
             "lr": self.lr, "betas": self.betas, "eps": self.eps,
             "weight_decay": 0, "amsgrad": False, "maximize": False,
             "foreach": None, "capturable": False, "differentiable": False,
-            "fused": None, "params": list(range(len(self.group.params))),
+            "fused": None, "params": list(range(len(self.ref_params))),
         }
         return {"state": state, "param_groups": [group]}
 
@@ -208,7 +218,7 @@ class FusedAdam:
         if not state:
             return
         steps = []
-        for i, (p, off) in enumerate(zip(self.group.params, self.group.offsets)):
+        for i, (p, off) in enumerate(zip(self.ref_params, self._ref_offsets)):
             key = i if i in state else str(i)
             if key not in state:
                 continue

@@ -2,6 +2,7 @@
 
 import os
 
+import pytest
 import torch
 
 from distributed_sac_amd.algo import SACEngine
@@ -92,3 +93,49 @@ def test_player_side_eval_load(tmp_path):
     assert it == 5
     x = torch.randn(3, cfg.state_dim)
     assert torch.allclose(actor(x)[0], engine.actor(x)[0], atol=1e-6)
+
+
+REF_LL_CKPT = ("/root/reference/saved_models/LunarLander_Distributed_SAC/"
+               "checkpoint_165000.tar")
+
+
+@pytest.mark.skipif(not os.path.exists(REF_LL_CKPT),
+                    reason="reference checkpoint not present on this host")
+def test_load_actual_reference_checkpoint():
+    """The REAL reference artifact (trained by the reference repo on a GTX
+    1080, torch 1.x) must load into our engine and policy unchanged."""
+    import numpy as np
+    from distributed_sac_amd.config import load_variant
+    from distributed_sac_amd.models import LLActor
+    from distributed_sac_amd.workers.player import evaluate_checkpoint
+    from distributed_sac_amd.workers.trainer import default_env_fn
+
+    cfg = load_variant("sac")
+    engine = SACEngine(cfg, "cpu")
+    load_into_engine(engine, REF_LL_CKPT)
+    assert engine.update_iteration == 165000
+    assert float(engine.log_alpha) < -2.0  # trained temperature
+    # optimizer state restored into the fused Adam
+    assert engine.critic_optimizer.step_count > 0
+
+    actor = LLActor(cfg.state_dim, cfg.action_dim, cfg.actor_hidden_dim)
+    it = load_actor_for_eval(actor, REF_LL_CKPT)
+    assert it == 0 or it == 165000  # LL stores episode_idx, not update_it
+    x = torch.randn(7, 8)
+    mu_e, _ = engine.actor(x)
+    mu_a, _ = actor(x)
+    assert torch.allclose(mu_e, mu_a, atol=1e-7)
+    # engine can continue training from the reference state
+    from tests.test_engine import make_batch
+    cfg_small = cfg
+    batch = {
+        "states": torch.randn(32, 8), "actions": torch.rand(32, 2) * 2 - 1,
+        "rewards": torch.randn(32, 1), "next_states": torch.randn(32, 8),
+        "dones": torch.zeros(32, 1),
+    }
+    m = engine.update(batch)
+    assert np.isfinite(m["critic_loss"])
+    # and the eval-mode player runs it on an env
+    out = evaluate_checkpoint(cfg, REF_LL_CKPT, default_env_fn, episodes=2)
+    assert np.isfinite(out["mean_reward"])
+
