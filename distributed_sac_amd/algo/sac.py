@@ -416,6 +416,156 @@ class SACEngine:
                 if isinstance(m, nn_.Linear)]
         return [m.weight for m in lins], [m.bias for m in lins]
 
+    @torch.no_grad()
+    def _mlp_fwd_manual(self, x_f32, ws_bf16, bs_f32):
+        from ..ops import native
+        ext = native()
+        xh = x_f32.to(torch.bfloat16)
+        acts = [xh]
+        h = xh
+        n = len(ws_bf16)
+        for i in range(n):
+            last = i == n - 1
+            h = ext.linear_act_fwd_bf16(h, ws_bf16[i], bs_f32[i].contiguous(),
+                                        0 if last else 1, 1,
+                                        1 if last else 0)
+            acts.append(h)
+        return h, acts
+
+    @torch.no_grad()
+    def _twin_fwd_manual(self, x_f32, ws_bf16, bs_f32):
+        from ..ops import native
+        ext = native()
+        xh = x_f32.to(torch.bfloat16)
+        acts = [xh]
+        h = xh
+        n = len(ws_bf16)
+        for i in range(n):
+            last = i == n - 1
+            h = ext.linear_act_fwd_bf16(h, ws_bf16[i], bs_f32[i].contiguous(),
+                                        0 if last else 1, 2,
+                                        1 if last else 0)
+            acts.append(h)
+        return h[0], h[1], acts
+
+    @torch.no_grad()
+    def _update_tensors_manual(self, batch):
+        """bf16 path with HAND-ROLLED backward: every gradient-producing
+        kernel writes straight into the flat fp32 gradient buffers (no
+        autograd bookkeeping, no AccumulateGrad adds, no zero_grad except
+        the tiny atomically-accumulated alpha grad).  Numerically identical
+        to the autograd bf16 path (same kernels, same order) — verified by
+        the manual-vs-autograd GPU test."""
+        from ..ops import native
+        ext = native()
+        states = batch["states"]
+        actions = batch["actions"]
+        rewards = batch["rewards"]
+        next_states = batch["next_states"]
+        dones = batch["dones"]
+        T = self.num_tasks
+        use_w = self.use_weighted_loss
+        B = states.shape[0]
+        A = self.cfg.action_dim
+        la_det = self.log_alpha.detach()
+        nl_a = len(self._actor_ws_bf16)
+        nl_c = len(self._twin_local_bf16)
+
+        # ---- batched actor forward + squash --------------------------
+        x_cat = torch.cat([next_states, states], dim=0)
+        ws_f32, bs_f32 = self._actor_weights()
+        out, acts_a = self._mlp_fwd_manual(x_cat, self._actor_ws_bf16, bs_f32)
+        mu, lsr = out[:, :A], out[:, A:]
+        if self._eps_queue:
+            eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
+        else:
+            eps = torch.randn_like(mu)
+        a_cat, lp_cat, tanh_u, ls_cat = ext.squashed_gaussian_fwd(
+            mu, lsr, eps, float(self.actor.k))
+        na, nlp = a_cat[:B], lp_cat[:B]
+        sa, lp, ls = a_cat[B:], lp_cat[B:], ls_cat[B:]
+
+        # ---- TD target ------------------------------------------------
+        xt = torch.cat([next_states, na], dim=-1)
+        q1_t, q2_t, _ = self._twin_fwd_manual(
+            xt, self._twin_target_bf16, self._twin_target[1])
+        y = ext.td_target_mt(rewards, dones, q1_t, q2_t, nlp, states,
+                             la_det, T, self.gamma, self.reward_scale)
+
+        # ---- critic loss + manual backward ---------------------------
+        x = torch.cat([states, actions], dim=-1)
+        q1, q2, acts_c = self._twin_fwd_manual(
+            x, self._twin_local_bf16, self._twin_local[1])
+        closs = ext.critic_loss_fwd(q1, q2, y, states, la_det, T,
+                                    int(use_w))[0]
+        dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
+                                  int(use_w))
+        wsg = [w.grad for w in self._twin_local[0]]
+        bsg = [b.grad for b in self._twin_local[1]]
+        for i in range(nl_c - 1, -1, -1):
+            act = 1 if i < nl_c - 1 else 0
+            yout = acts_c[i + 1] if i < nl_c - 1 else acts_c[i]
+            ext.linear_bwd_dwdb_bf16_out(dy, acts_c[i], yout, act, 2,
+                                         wsg[i], bsg[i])
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                            yout, act, 2, 0)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self.critic_optimizer.step()
+        self.refresh_bf16("critic")
+
+        # ---- actor/alpha loss + manual backward ----------------------
+        xa = torch.cat([states, sa], dim=-1)
+        aq1, aq2, acts_f = self._twin_fwd_manual(
+            xa, self._twin_local_bf16, self._twin_local[1])
+        al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls, states,
+                                      la_det, T, int(use_w), self.H_bar_f)
+        self.alpha_group.flat_grad.zero_()
+        daq, dlp = ext.actor_alpha_loss_bwd2(
+            aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
+            T, int(use_w), self.H_bar_f)
+        dy = daq
+        for i in range(nl_c - 1, 0, -1):
+            act = 1 if i < nl_c - 1 else 0
+            yout = acts_f[i + 1] if i < nl_c - 1 else acts_f[i]
+            dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                        yout, act, 2, 0)
+        dxa = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
+                                     acts_f[1] if nl_c > 1 else acts_f[0],
+                                     1 if nl_c > 1 else 0, 2, 1)
+        dsa = dxa[:, states.shape[1]:].float()
+        dhead = ext.squashed_gaussian_bwd2(
+            dsa, dlp, lsr[B:], ls_cat[B:], eps[B:], tanh_u[B:],
+            float(self.actor.k))
+        dy = dhead
+        wag = [w.grad for w in ws_f32]
+        bag = [b.grad for b in bs_f32]
+        for i in range(nl_a - 1, -1, -1):
+            act = 1 if i < nl_a - 1 else 0
+            yout = (acts_a[i + 1][B:] if i < nl_a - 1 else acts_a[i][B:])
+            ext.linear_bwd_dwdb_bf16_out(dy, acts_a[i][B:], yout, act, 1,
+                                         wag[i], bag[i])
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
+                                            yout, act, 1, 1)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
+            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+        FusedAdam.step_many([self.actor_optimizer,
+                             self.log_alpha_optimizer])
+        self.refresh_bf16("actor")
+        self.alpha = self.log_alpha.exp().detach()
+
+        flat_polyak_(self.target_group, self.critic_group, self.tau)
+        self.refresh_bf16("target")
+        return {
+            "critic_loss": closs[0] + closs[1],
+            "actor_loss": al[0],
+            "alpha_loss": al[2],
+            "entropy": al[3],
+        }
+
     def _update_tensors_fused(self, batch):
         """GPU path: same math/order as update_tensors, restructured for
         the hardware (numerically identical):
@@ -439,6 +589,10 @@ class SACEngine:
         B = states.shape[0]
         A = self.cfg.action_dim
 
+        import os as _os
+        if getattr(self, "_bf16", False) \
+                and _os.environ.get("DSAC_NO_MANUAL", "0") != "1":
+            return self._update_tensors_manual(batch)
         self.zero_grad()
 
         # --- batched actor forward: [next | current] halves (both use

@@ -585,6 +585,31 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd(
   }
 }
 
+// bf16 stacked-output variant for the manual backward: dq[2,B] bf16
+__global__ __launch_bounds__(256) void k_critic_loss_bwd2(
+    const float* __restrict__ q1, const float* __restrict__ q2,
+    const float* __restrict__ y, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, const float* __restrict__ saved,
+    unsigned short* __restrict__ dq, int B, int T, int oh_stride,
+    int use_w) {
+  __shared__ float smw[32];
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  __syncthreads();
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B) return;
+  const int t_i = task_of_row(onehot, i, oh_stride, T);
+  const float w_raw = use_w ? smw[t_i] : 1.f;
+  const float coeff = (use_w ? w_raw / saved[2] : 1.f) / (float)B;
+  auto cvt = [](float f) -> unsigned short {
+    union { float f; unsigned u; } v{f};
+    unsigned u = v.u;
+    u += 0x7FFFu + ((u >> 16) & 1u);
+    return (unsigned short)(u >> 16);
+  };
+  dq[i] = cvt(coeff * -2.f * (y[i] - q1[i]));
+  dq[B + i] = cvt(coeff * -2.f * (y[i] - q2[i]));
+}
+
 // dq1_i = g1 * coeff_i * -2 (y_i - q1_i);  dq2 likewise with g2
 __global__ __launch_bounds__(256) void k_critic_loss_bwd(
     const float* __restrict__ q1, const float* __restrict__ q2,
@@ -681,6 +706,74 @@ __global__ __launch_bounds__(256) void k_actor_alpha_loss_bwd(
   }
   __syncthreads();
   if (tid < T) atomicAdd(&dla[tid], s_dla[tid]);
+}
+
+// manual-backward actor/alpha variant: daq[2,B] bf16 stacked, dlp fp32,
+// dla accumulated into the caller buffer (zeroed by caller).
+__global__ __launch_bounds__(256) void k_actor_alpha_loss_bwd2(
+    const float* __restrict__ aq1, const float* __restrict__ aq2,
+    const float* __restrict__ lp, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, const float* __restrict__ saved,
+    unsigned short* __restrict__ daq, float* __restrict__ dlp,
+    float* __restrict__ dla, int B, int T, int oh_stride, int use_w,
+    float H_bar) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int tid = threadIdx.x;
+  __shared__ float s_dla[32];
+  __shared__ float smw[32];
+  if (tid < T) s_dla[tid] = 0.f;
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  __syncthreads();
+  if (i < B) {
+    const int t_i = task_of_row(onehot, i, oh_stride, T);
+    const float alpha_i = __expf(log_alpha[t_i]);
+    const float w_raw = use_w ? smw[t_i] : 1.f;
+    const float coeff = (use_w ? w_raw / saved[1] : 1.f) / (float)B;
+    const bool first = aq1[i] <= aq2[i];
+    auto cvt = [](float f) -> unsigned short {
+      union { float f; unsigned u; } v{f};
+      unsigned u = v.u;
+      u += 0x7FFFu + ((u >> 16) & 1u);
+      return (unsigned short)(u >> 16);
+    };
+    daq[i] = cvt(first ? coeff * -1.f : 0.f);
+    daq[B + i] = cvt(first ? 0.f : coeff * -1.f);
+    dlp[i] = coeff * alpha_i;
+    atomicAdd(&s_dla[t_i], -(lp[i] + H_bar) / (float)B);
+  }
+  __syncthreads();
+  if (tid < T) atomicAdd(&dla[tid], s_dla[tid]);
+}
+
+// squash backward emitting the joined actor-head gradient [B, 2A] bf16
+// ([dmu | dlog_std_raw] — the mu_log_std_layer output layout)
+__global__ __launch_bounds__(256) void k_squash_bwd2(
+    const float* __restrict__ ga, const float* __restrict__ gl,
+    const float* __restrict__ lsr, const float* __restrict__ ls,
+    const float* __restrict__ eps, const float* __restrict__ tanh_u,
+    unsigned short* __restrict__ dhead, int B, int A, float k) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B) return;
+  const float g = gl[i];
+  auto cvt = [](float f) -> unsigned short {
+    union { float f; unsigned u; } v{f};
+    unsigned u = v.u;
+    u += 0x7FFFu + ((u >> 16) & 1u);
+    return (unsigned short)(u >> 16);
+  };
+  for (int a = 0; a < A; ++a) {
+    const long idx = (long)i * A + a;
+    const float t = tanh_u[idx];
+    const float omt2 = 1.f - t * t;
+    const float dlp_du = 2.f * t * omt2 / (omt2 + 1e-6f);
+    const float s = __expf(ls[idx]);
+    const float e = eps[idx];
+    const float du = ga[idx] * k * omt2 + g * dlp_du;
+    const float raw = lsr[idx];
+    const float mask = (raw >= -20.f && raw <= 2.f) ? 1.f : 0.f;
+    dhead[(long)i * 2 * A + a] = cvt(du);
+    dhead[(long)i * 2 * A + A + a] = cvt(mask * (du * e * s - g));
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -1089,6 +1182,64 @@ static std::vector<torch::Tensor> actor_alpha_loss_bwd(
   return {daq1, daq2, dlp, dla};
 }
 
+static torch::Tensor critic_loss_bwd2(
+    torch::Tensor q1, torch::Tensor q2, torch::Tensor y,
+    torch::Tensor states, torch::Tensor log_alpha, torch::Tensor saved,
+    long T, long use_w) {
+  CHECK_IN(q1);
+  const long B = q1.size(0);
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto dq = torch::empty({2, B, 1}, q1.options().dtype(torch::kBFloat16));
+  hipLaunchKernelGGL(k_critic_loss_bwd2, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), q1.data_ptr<float>(), q2.data_ptr<float>(),
+                     y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
+                     saved.data_ptr<float>(),
+                     (unsigned short*)dq.data_ptr(), (int)B, (int)T,
+                     (int)oh_stride, (int)use_w);
+  return dq;
+}
+
+static std::vector<torch::Tensor> actor_alpha_loss_bwd2(
+    torch::Tensor aq1, torch::Tensor aq2, torch::Tensor lp,
+    torch::Tensor states, torch::Tensor log_alpha, torch::Tensor saved,
+    torch::Tensor dla_out, long T, long use_w, double H_bar) {
+  CHECK_IN(aq1); CHECK_IN(dla_out);
+  const long B = aq1.size(0);
+  const long oh_stride = states.size(1);
+  const float* oh = states.data_ptr<float>() + (oh_stride - T);
+  auto daq = torch::empty({2, B, 1}, aq1.options().dtype(torch::kBFloat16));
+  auto dlp = torch::empty_like(lp);
+  hipLaunchKernelGGL(k_actor_alpha_loss_bwd2, dim3((B + 255) / 256),
+                     dim3(256), 0, cur_stream(), aq1.data_ptr<float>(),
+                     aq2.data_ptr<float>(), lp.data_ptr<float>(), oh,
+                     log_alpha.data_ptr<float>(), saved.data_ptr<float>(),
+                     (unsigned short*)daq.data_ptr(), dlp.data_ptr<float>(),
+                     dla_out.data_ptr<float>(), (int)B, (int)T,
+                     (int)oh_stride, (int)use_w, (float)H_bar);
+  return {daq, dlp};
+}
+
+static torch::Tensor squashed_gaussian_bwd2(
+    torch::Tensor ga, torch::Tensor gl, torch::Tensor lsr, torch::Tensor ls,
+    torch::Tensor eps, torch::Tensor tanh_u, double k) {
+  CHECK_IN(ga);
+  auto gac = ga.contiguous(); auto glc = gl.contiguous();
+  auto lsrc = lsr.contiguous(); auto lsc = ls.contiguous();
+  auto ec = eps.contiguous(); auto tc = tanh_u.contiguous();
+  const long B = gac.size(0), A = gac.size(1);
+  auto dhead = torch::empty({B, 2 * A},
+                            gac.options().dtype(torch::kBFloat16));
+  hipLaunchKernelGGL(k_squash_bwd2, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), gac.data_ptr<float>(),
+                     glc.data_ptr<float>(), lsrc.data_ptr<float>(),
+                     lsc.data_ptr<float>(), ec.data_ptr<float>(),
+                     tc.data_ptr<float>(),
+                     (unsigned short*)dhead.data_ptr(), (int)B, (int)A,
+                     (float)k);
+  return dhead;
+}
+
 static void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        torch::Tensor v, long step, double lr, double b1,
                        double b2, double eps) {
@@ -1190,6 +1341,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("critic_loss_bwd", &critic_loss_bwd);
   mod.def("actor_alpha_loss_fwd", &actor_alpha_loss_fwd);
   mod.def("actor_alpha_loss_bwd", &actor_alpha_loss_bwd);
+  mod.def("critic_loss_bwd2", &critic_loss_bwd2);
+  mod.def("actor_alpha_loss_bwd2", &actor_alpha_loss_bwd2);
+  mod.def("squashed_gaussian_bwd2", &squashed_gaussian_bwd2);
   mod.def("adam_step_", &adam_step_);
   mod.def("adam_step_dev_", &adam_step_dev_);
   mod.def("adam_step_multi_", &adam_step_multi_);

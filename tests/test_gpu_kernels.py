@@ -553,3 +553,46 @@ def test_bf16_engine_update_and_graph():
         losses.append(float(m["critic_loss"]))
     assert all(v == v for v in losses)
     assert losses[-1] < losses[0] * 1.5  # not diverging
+
+
+def test_manual_backward_matches_autograd_bf16():
+    """The hand-rolled bf16 backward must track the autograd bf16 path
+    (same kernels, same order — only the gradient bookkeeping differs)."""
+    import os
+    from distributed_sac_amd.algo import SACEngine
+    from tests.test_engine import make_batch, small_cfg
+    torch.manual_seed(0)
+    cfg = small_cfg("mtsac")
+    e_man = SACEngine(cfg, "cuda:0", precision="bf16")
+    e_aut = SACEngine(cfg, "cuda:0", precision="bf16")
+    e_aut.actor.load_state_dict(e_man.actor.state_dict())
+    e_aut.local_critic.load_state_dict(e_man.local_critic.state_dict())
+    e_aut.hard_copy_targets()
+    e_aut.refresh_bf16()
+    e_man.hard_copy_targets()
+    e_man.refresh_bf16()
+    os.environ["DSAC_NO_MANUAL"] = "1"
+    try:
+        for step in range(3):
+            batch = {k: v.cuda() for k, v in
+                     make_batch(cfg, seed=step).items()}
+            eps = [torch.randn(cfg.batch_size, cfg.action_dim,
+                               device="cuda") for _ in range(2)]
+            e_aut._eps_queue = [e.clone() for e in eps]
+            m_aut = e_aut.update({k: v.clone() for k, v in batch.items()})
+            os.environ["DSAC_NO_MANUAL"] = "0"
+            e_man._eps_queue = [e.clone() for e in eps]
+            m_man = e_man.update({k: v.clone() for k, v in batch.items()})
+            os.environ["DSAC_NO_MANUAL"] = "1"
+    finally:
+        os.environ.pop("DSAC_NO_MANUAL", None)
+    assert abs(m_aut["critic_loss"] - m_man["critic_loss"]) < 1e-4 + \
+        1e-3 * abs(m_aut["critic_loss"])
+    assert abs(m_aut["actor_loss"] - m_man["actor_loss"]) < 1e-3
+    for (n, pa), (_, pm) in zip(e_aut.actor.named_parameters(),
+                                e_man.actor.named_parameters()):
+        assert torch.allclose(pa, pm, atol=5e-5, rtol=1e-3), n
+    for (n, pa), (_, pm) in zip(e_aut.local_critic.named_parameters(),
+                                e_man.local_critic.named_parameters()):
+        assert torch.allclose(pa, pm, atol=5e-5, rtol=1e-3), n
+    assert torch.allclose(e_aut.log_alpha, e_man.log_alpha, atol=1e-5)
