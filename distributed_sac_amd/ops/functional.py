@@ -434,7 +434,11 @@ class _FusedTwinMLPBF16(torch.autograd.Function):
         for i in range(n - 1, -1, -1):
             act = ACT_RELU if i < n - 1 else ACT_NONE
             yout = acts[i + 1] if i < n - 1 else acts[i]
-            if ctx.needs_input_grad[2 + i]:
+            # forward args: (x, n_layers, ws_bf16_tuple, *ws_f32, *bs_f32)
+            # -> ws_f32[i] is input index 3+i (the non-tensor tuple at 2
+            # always reports False — indexing 2+i silently skipped layer
+            # 0's dW)
+            if ctx.needs_input_grad[3 + i]:
                 dw, db = ext.linear_bwd_dwdb_bf16(dy, acts[i], yout, act, 2)
                 dws[i], dbs[i] = dw, db
             if i > 0:
