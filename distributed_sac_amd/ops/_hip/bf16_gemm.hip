@@ -411,6 +411,34 @@ __global__ __launch_bounds__(256) void k_reduce_partials2(
   out[(long)g * n_per_g + i] = acc;
 }
 
+// one launch folds BOTH the dw slabs and the db slabs (flat index over
+// G*(N*K) then G*N)
+__global__ __launch_bounds__(256) void k_reduce_dwdb(
+    const float* __restrict__ ws, const float* __restrict__ ws_db,
+    float* __restrict__ dw, float* __restrict__ db, long nw_per_g,
+    long nb_per_g, int S, int G) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long tot_w = (long)G * nw_per_g;
+  if (i < tot_w) {
+    const int g = (int)(i / nw_per_g);
+    const long e = i % nw_per_g;
+    const float* base = ws + ((long)g * S) * nw_per_g + e;
+    float acc = 0.f;
+    for (int s = 0; s < S; ++s) acc += base[(long)s * nw_per_g];
+    dw[i] = acc;
+    return;
+  }
+  i -= tot_w;
+  if (i < (long)G * nb_per_g) {
+    const int g = (int)(i / nb_per_g);
+    const long e = i % nb_per_g;
+    const float* base = ws_db + ((long)g * S) * nb_per_g + e;
+    float acc = 0.f;
+    for (int s = 0; s < S; ++s) acc += base[(long)s * nb_per_g];
+    db[i] = acc;
+  }
+}
+
 // ===========================================================================
 // Host wrappers
 // ===========================================================================
@@ -498,15 +526,11 @@ static std::vector<torch::Tensor> linear_bwd_dwdb_bf16(
     dw.copy_(ws.view_as(dw));
     db.copy_(ws_db.view_as(db));
   } else {
-    const long nw = N * K;
-    dim3 g1((nw + 255) / 256, G);
-    hipLaunchKernelGGL(k_reduce_partials2, g1, dim3(256), 0, cur_stream2(),
-                       ws.data_ptr<float>(), dw.data_ptr<float>(), nw,
-                       (int)S, nw);
-    dim3 g2((N + 255) / 256, G);
-    hipLaunchKernelGGL(k_reduce_partials2, g2, dim3(256), 0, cur_stream2(),
-                       ws_db.data_ptr<float>(), db.data_ptr<float>(), N,
-                       (int)S, N);
+    const long tot = G * (N * K + N);
+    hipLaunchKernelGGL(k_reduce_dwdb, dim3((tot + 255) / 256), dim3(256), 0,
+                       cur_stream2(), ws.data_ptr<float>(),
+                       ws_db.data_ptr<float>(), dw.data_ptr<float>(),
+                       db.data_ptr<float>(), N * K, N, (int)S, (int)G);
   }
   return {dw, db};
 }
@@ -547,13 +571,10 @@ static void linear_bwd_dwdb_bf16_out(torch::Tensor dy, torch::Tensor x,
                      (const u16*)yc.data_ptr(), ws_p, db_p, (int)M, (int)N,
                      (int)K, (int)act, (int)S, chunk, xgs);
   if (S > 1) {
-    const long nw = N * K;
-    dim3 g1((nw + 255) / 256, G);
-    hipLaunchKernelGGL(k_reduce_partials2, g1, dim3(256), 0, cur_stream2(),
-                       ws_p, dw_out.data_ptr<float>(), nw, (int)S, nw);
-    dim3 g2((N + 255) / 256, G);
-    hipLaunchKernelGGL(k_reduce_partials2, g2, dim3(256), 0, cur_stream2(),
-                       db_p, db_out.data_ptr<float>(), N, (int)S, N);
+    const long tot = G * (N * K + N);
+    hipLaunchKernelGGL(k_reduce_dwdb, dim3((tot + 255) / 256), dim3(256), 0,
+                       cur_stream2(), ws_p, db_p, dw_out.data_ptr<float>(),
+                       db_out.data_ptr<float>(), N * K, N, (int)S, (int)G);
   }
 }
 

@@ -548,7 +548,13 @@ __device__ __forceinline__ void fill_task_weights(
   }
 }
 
-// out[0]=loss1 out[1]=loss2 out[2]=wsum
+__global__ void k_loss_zero(float* __restrict__ raw, int n) {
+  if ((int)threadIdx.x < n) raw[threadIdx.x] = 0.f;
+}
+
+// out: [0]=loss1 [1]=loss2 [2]=wsum, [3..5]=atomic raw sums (zeroed by
+// k_loss_zero); multi-block grid-stride + 1-thread finalize (the single-
+// workgroup version left 255/256 CUs idle for ~20us per loss)
 __global__ __launch_bounds__(256) void k_critic_loss_fwd(
     const float* __restrict__ q1, const float* __restrict__ q2,
     const float* __restrict__ y, const float* __restrict__ onehot,
@@ -560,7 +566,7 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd(
   if (use_w) fill_task_weights(smw, log_alpha, T);
   __syncthreads();
   float s_l1 = 0.f, s_l2 = 0.f, s_w = 0.f;
-  for (int i = tid; i < B; i += 256) {
+  for (int i = blockIdx.x * 256 + tid; i < B; i += 256 * gridDim.x) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float w_raw = use_w ? smw[t_i] : 1.f;
     s_w += w_raw;
@@ -577,12 +583,19 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd(
     __syncthreads();
   }
   if (tid == 0) {
-    const float wsum = use_w ? red[2][0] : 1.f;
-    const float denom = wsum * (float)B;
-    out[0] = red[0][0] / denom;
-    out[1] = red[1][0] / denom;
-    out[2] = wsum;
+    atomicAdd(&out[3], red[0][0]);
+    atomicAdd(&out[4], red[1][0]);
+    atomicAdd(&out[5], red[2][0]);
   }
+}
+
+__global__ void k_critic_loss_finalize(float* __restrict__ out, int B,
+                                       int use_w) {
+  const float wsum = use_w ? out[5] : 1.f;
+  const float denom = wsum * (float)B;
+  out[0] = out[3] / denom;
+  out[1] = out[4] / denom;
+  out[2] = wsum;
 }
 
 // bf16 stacked-output variant for the manual backward: dq[2,B] bf16
@@ -643,7 +656,7 @@ __global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd(
   __syncthreads();
   constexpr float CE = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
   float s_pl = 0.f, s_w = 0.f, s_al = 0.f, s_en = 0.f;
-  for (int i = tid; i < B; i += 256) {
+  for (int i = blockIdx.x * 256 + tid; i < B; i += 256 * gridDim.x) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float la = log_alpha[t_i];
     const float alpha_i = __expf(la);
@@ -666,12 +679,20 @@ __global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd(
     __syncthreads();
   }
   if (tid == 0) {
-    const float wsum = use_w ? red[1][0] : 1.f;
-    out[0] = red[0][0] / (wsum * (float)B);
-    out[1] = wsum;
-    out[2] = -red[2][0] / (float)B;
-    out[3] = red[3][0] / (float)B;
+    atomicAdd(&out[4], red[0][0]);
+    atomicAdd(&out[5], red[1][0]);
+    atomicAdd(&out[6], red[2][0]);
+    atomicAdd(&out[7], red[3][0]);
   }
+}
+
+__global__ void k_actor_alpha_loss_finalize(float* __restrict__ out, int B,
+                                            int use_w) {
+  const float wsum = use_w ? out[5] : 1.f;
+  out[0] = out[4] / (wsum * (float)B);
+  out[1] = wsum;
+  out[2] = -out[6] / (float)B;
+  out[3] = out[7] / (float)B;
 }
 
 // daq1_i = gp*coeff_i*-(aq1<=aq2); daq2_i = gp*coeff_i*-(aq2<aq1)
@@ -1112,11 +1133,17 @@ static std::vector<torch::Tensor> critic_loss_fwd(
   const long B = q1.size(0);
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
-  auto out = torch::empty({3}, q1.options());
-  hipLaunchKernelGGL(k_critic_loss_fwd, dim3(1), dim3(256), 0, cur_stream(),
-                     q1.data_ptr<float>(), q2.data_ptr<float>(),
+  auto out = torch::empty({6}, q1.options());
+  hipLaunchKernelGGL(k_loss_zero, dim3(1), dim3(8), 0, cur_stream(),
+                     out.data_ptr<float>(), 6);
+  const int nblk = (int)std::min<long>((B + 255) / 256, 32);
+  hipLaunchKernelGGL(k_critic_loss_fwd, dim3(nblk), dim3(256), 0,
+                     cur_stream(), q1.data_ptr<float>(), q2.data_ptr<float>(),
                      y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
                      out.data_ptr<float>(), (int)B, (int)T, (int)oh_stride,
+                     (int)use_w);
+  hipLaunchKernelGGL(k_critic_loss_finalize, dim3(1), dim3(1), 0,
+                     cur_stream(), out.data_ptr<float>(), (int)B,
                      (int)use_w);
   return {out};
 }
@@ -1149,13 +1176,19 @@ static torch::Tensor actor_alpha_loss_fwd(
   const long A = ls.size(1);
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
-  auto out = torch::empty({4}, aq1.options());
-  hipLaunchKernelGGL(k_actor_alpha_loss_fwd, dim3(1), dim3(256), 0,
+  auto out = torch::empty({8}, aq1.options());
+  hipLaunchKernelGGL(k_loss_zero, dim3(1), dim3(8), 0, cur_stream(),
+                     out.data_ptr<float>(), 8);
+  const int nblk = (int)std::min<long>((B + 255) / 256, 32);
+  hipLaunchKernelGGL(k_actor_alpha_loss_fwd, dim3(nblk), dim3(256), 0,
                      cur_stream(), aq1.data_ptr<float>(),
                      aq2.data_ptr<float>(), lp.data_ptr<float>(),
                      ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
                      out.data_ptr<float>(), (int)B, (int)T, (int)A,
                      (int)oh_stride, (int)use_w, (float)H_bar);
+  hipLaunchKernelGGL(k_actor_alpha_loss_finalize, dim3(1), dim3(1), 0,
+                     cur_stream(), out.data_ptr<float>(), (int)B,
+                     (int)use_w);
   return out;
 }
 
