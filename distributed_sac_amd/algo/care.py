@@ -120,6 +120,7 @@ class CAREEngine(SACEngine):
         self._twin_local_frozen = ([w.detach() for w in self._twin_local[0]],
                                    [b.detach() for b in self._twin_local[1]])
         self._init_bf16_mirrors()
+        self._init_care_bf16()
         self.hard_copy_targets()
         self.tie_actor_state_encoder()
 
@@ -238,6 +239,87 @@ class CAREEngine(SACEngine):
                 "alpha_loss": loss_log_alpha.detach(),
                 "entropy": entropy.detach()}
 
+    # -- bf16 fast state-encoder machinery ------------------------------
+    def _init_care_bf16(self) -> None:
+        if not getattr(self, "_bf16", False):
+            self._se_fast = False
+            return
+        self._se_fast = True
+        dev = self.device
+
+        def mlp_linears(seq):
+            import torch.nn as nn_
+            return [m for m in seq if isinstance(m, nn_.Linear)]
+
+        def build(critic, group, mirror):
+            se = critic.state_encoder
+            mix = [m for m in se.mixture_encoders.mixtureEncoders
+                   if hasattr(m, "W")]
+            k = mix[0].num_encoders
+            info = {"k": k,
+                    "mixW": [m.W for m in mix],
+                    "mixB": [m.b for m in mix],
+                    "mixT": [torch.empty(k, m.W.shape[2], m.W.shape[1],
+                                         dtype=torch.bfloat16, device=dev)
+                             for m in mix]}
+            for name, seq in (("trunk", se.trunk),
+                              ("mlpctx", getattr(se, "mlp_context", None))):
+                if seq is None:
+                    info[name] = None
+                    continue
+                lins = mlp_linears(seq)
+                ws, bs, wsh = [], [], []
+                for l in lins:
+                    i = next(j for j, q in enumerate(group.params)
+                             if q is l.weight)
+                    off = group.offsets[i]
+                    wsh.append(mirror[off:off + l.weight.numel()]
+                               .view_as(l.weight))
+                    ws.append(l.weight)
+                    bs.append(l.bias)
+                info[name] = (ws, bs, wsh)
+            return info
+
+        self._se_local = build(self.local_critic, self.critic_group,
+                               self._critic_bf16)
+        self._se_target = build(self.target_critic, self.target_group,
+                                self._target_bf16)
+        self._refresh_mixT()
+
+    @torch.no_grad()
+    def _refresh_mixT(self, which: str = "all") -> None:
+        if not getattr(self, "_se_fast", False):
+            return
+        if which in ("all", "critic"):
+            for m, W in zip(self._se_local["mixT"], self._se_local["mixW"]):
+                m.copy_(W.detach().permute(0, 2, 1))
+        if which in ("all", "target"):
+            for m, W in zip(self._se_target["mixT"],
+                            self._se_target["mixW"]):
+                m.copy_(W.detach().permute(0, 2, 1))
+
+    def refresh_bf16(self, which: str = "all") -> None:
+        super().refresh_bf16(which)
+        self._refresh_mixT(which)
+
+    def _se_fwd_fast(self, info, states_2d, z_context):
+        """stateEncoder.forward via bf16 kernels (value-identical to the
+        module path up to bf16 rounding; skips the reference's divide by
+        alpha.sum(dim=1) which is exactly softmax-sum==1)."""
+        z_encs = Fops.grouped_mlp_bf16(states_2d, info["mixW"],
+                                       info["mixB"], info["mixT"],
+                                       info["k"])              # [k,B,50] f32
+        tws, tbs, twsh = info["trunk"]
+        logits = Fops.mlp_forward_bf16(z_context.detach(), tws, tbs, twsh)
+        alpha = torch.softmax(logits, dim=-1)                  # [B,k]
+        z_enc = (z_encs * alpha.t().unsqueeze(-1)).sum(0)      # [B,50]
+        if info["mlpctx"] is not None:
+            cws, cbs, cwsh = info["mlpctx"]
+            zc = Fops.mlp_forward_bf16(z_context, cws, cbs, cwsh)
+        else:
+            zc = z_context
+        return torch.cat([zc, z_enc], dim=1)
+
     def _twin_fwd(self, x, which: str):
         if getattr(self, "_bf16", False):
             if which == "target":
@@ -274,20 +356,29 @@ class CAREEngine(SACEngine):
         z_context = self.context_encoder(states)
         zc_d = z_context.detach()
 
-        # batched actor pass: [next | current], SE under no_grad (grads
-        # into trunk/mixture are discarded by the reference), head grads
-        # only from the states half
-        with torch.no_grad():
-            x_cat = torch.cat([next_states, states], dim=0)
-            z2 = torch.cat([zc_d, zc_d], dim=0)
-            enc_cat = self.actor.state_encoder(z2, x_cat, detach_z_encs=True)
+        # ONE batched local-SE forward over [next | current]: the actor's
+        # tied encoder equals the pre-step critic encoder (hard copy at the
+        # end of every update), so its encodings are the same VALUES — the
+        # states half doubles as the critic-loss encoding (with grad), and
+        # a detached copy feeds the actor head.
+        x_cat = torch.cat([next_states, states], dim=0)
+        if getattr(self, "_se_fast", False):
+            z2 = torch.cat([z_context, z_context], dim=0)
+            enc_cat = self._se_fwd_fast(self._se_local, x_cat, z2)
+        else:
+            with torch.no_grad():
+                z2 = torch.cat([zc_d, zc_d], dim=0)
+                enc_cat = self.actor.state_encoder(z2, x_cat,
+                                                   detach_z_encs=True)
         ws, bs = self._actor_weights()
+        enc_actor_in = (enc_cat.detach() if getattr(self, "_se_fast", False)
+                        else enc_cat)
         if getattr(self, "_bf16", False):
-            mu_lsr = Fops.mlp_forward_bf16(enc_cat, ws, bs,
+            mu_lsr = Fops.mlp_forward_bf16(enc_actor_in, ws, bs,
                                            self._actor_ws_bf16,
                                            grad_row_start=B)
         else:
-            mu_lsr = Fops.mlp_forward(enc_cat, ws, bs, grad_row_start=B)
+            mu_lsr = Fops.mlp_forward(enc_actor_in, ws, bs, grad_row_start=B)
         mu = mu_lsr[:, :A]
         lsr = mu_lsr[:, A:]
         if self._eps_queue:
@@ -300,14 +391,21 @@ class CAREEngine(SACEngine):
         sa, lp, ls = a_cat[B:], lp_cat[B:], ls_cat[B:]
 
         with torch.no_grad():
-            enc_t = self.target_critic.encode(next_states, z_context)
+            if getattr(self, "_se_fast", False):
+                enc_t = self._se_fwd_fast(self._se_target, next_states,
+                                          z_context)
+            else:
+                enc_t = self.target_critic.encode(next_states, z_context)
             xt = torch.cat([enc_t, na], dim=-1)
             q1_t, q2_t = self._twin_fwd(xt, "target")
             y = native().td_target_mt(rewards, dones, q1_t, q2_t, nlp,
                                       states, self.log_alpha.detach(), T,
                                       self.gamma, self.reward_scale)
 
-        enc = self.local_critic.encode(states, z_context)
+        if getattr(self, "_se_fast", False):
+            enc = enc_cat[B:]
+        else:
+            enc = self.local_critic.encode(states, z_context)
         x = torch.cat([enc, actions], dim=-1)
         q1, q2 = self._twin_fwd(x, "local")
         l1, l2 = Fops.critic_loss(q1, q2, y, states, self.log_alpha.detach(),
@@ -322,7 +420,11 @@ class CAREEngine(SACEngine):
         self.refresh_bf16("critic")
 
         with torch.no_grad():
-            enc_c = self.local_critic.encode(states, zc_d, detach_z_encs=True)
+            if getattr(self, "_se_fast", False):
+                enc_c = self._se_fwd_fast(self._se_local, states, zc_d)
+            else:
+                enc_c = self.local_critic.encode(states, zc_d,
+                                                 detach_z_encs=True)
         xa = torch.cat([enc_c, sa], dim=-1)
         aq1, aq2 = self._twin_fwd(xa, "frozen")
         policy_loss, loss_log_alpha, entropy = Fops.actor_alpha_loss(

@@ -454,3 +454,63 @@ class _FusedTwinMLPBF16(torch.autograd.Function):
 def twin_mlp_forward_bf16(x, stacked_ws_f32, stacked_bs_f32, ws_bf16):
     return _FusedTwinMLPBF16.apply(x, len(ws_bf16), tuple(ws_bf16),
                                    *stacked_ws_f32, *stacked_bs_f32)
+
+
+class _GroupedMLPBF16(torch.autograd.Function):
+    """CARE mixture-of-k-encoders as grouped bf16 GEMMs (grid.z = k).
+
+    Weights are the reference's batched-Linear parameters W (k,in,out) /
+    b (k,1,out) (state_encoder.Linear); compute uses TRANSPOSED bf16
+    mirrors (k,out,in) matching the GEMM kernel's [G,N,K] layout.  Layer 0
+    consumes shared x; later layers per-group [k,B,*].  Output [k,B,out]
+    fp32.  Backward runs full-batch (upstream grads are zero outside the
+    grad-carrying rows; the mixture layers are 50-wide — cheaper than
+    slicing per-group activations contiguous).
+    """
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, n_layers: int, k: int, ws_bf16T,
+                *wb_f32):
+        ext = native()
+        bs = wb_f32[n_layers:]
+        xh = x.to(torch.bfloat16)
+        acts = [xh]
+        h = xh
+        for i in range(n_layers):
+            last = i == n_layers - 1
+            h = ext.linear_act_fwd_bf16(
+                h, ws_bf16T[i], bs[i].view(k, -1).contiguous(),
+                0 if last else 1, k, 1 if last else 0)
+            acts.append(h)
+        ctx.save_for_backward(*acts[:-1])
+        ctx.ws_bf16T = ws_bf16T
+        ctx.n_layers = n_layers
+        ctx.k = k
+        return h
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        n, k = ctx.n_layers, ctx.k
+        acts = list(ctx.saved_tensors)
+        ws = ctx.ws_bf16T
+        ext = native()
+        dy = grad_out.contiguous().to(torch.bfloat16)
+        dws = [None] * n
+        dbs = [None] * n
+        for i in range(n - 1, -1, -1):
+            act = ACT_RELU if i < n - 1 else ACT_NONE
+            yout = acts[i + 1] if i < n - 1 else acts[i]
+            # fwd args: (x, n_layers, k, ws_bf16T, *W, *b) -> W[i] at 4+i
+            if ctx.needs_input_grad[4 + i]:
+                dwT, db = ext.linear_bwd_dwdb_bf16(dy, acts[i], yout, act, k)
+                dws[i] = dwT.permute(0, 2, 1)     # -> (k, in, out)
+                dbs[i] = db.view(k, 1, -1)
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, ws[i], yout, act, k, 0)
+        return (None, None, None, None, *dws, *dbs)
+
+
+def grouped_mlp_bf16(x, W_f32_list, b_f32_list, ws_bf16T_list, k: int):
+    return _GroupedMLPBF16.apply(x, len(W_f32_list), k,
+                                 tuple(ws_bf16T_list), *W_f32_list,
+                                 *b_f32_list)

@@ -340,3 +340,58 @@ def test_care_graph_capture(tmp_path):
     for k, v in m.items():
         assert float(v) == float(v), k
     assert torch.isfinite(engine.critic_group.flat_data).all()
+
+
+@pytest.mark.gpu
+def test_care_bf16_fast_se(tmp_path):
+    """bf16 CARE (fast grouped-GEMM state encoder) must track the fp32
+    engine loosely and train stably under graph capture."""
+    from distributed_sac_amd.replay import ShardedReplay
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path)
+    e32 = CAREEngine(cfg, "cuda:0")
+    e16 = CAREEngine(cfg, "cuda:0", precision="bf16")
+    e16.context_encoder.load_state_dict(e32.context_encoder.state_dict())
+    e16.actor.load_state_dict(e32.actor.state_dict())
+    e16.local_critic.load_state_dict(e32.local_critic.state_dict())
+    for e in (e32, e16):
+        e.hard_copy_targets()
+        e.tie_actor_state_encoder()
+    e16.refresh_bf16()
+    for step in range(3):
+        batch = {k: v.cuda() for k, v in care_batch(cfg, seed=step).items()}
+        g = torch.Generator().manual_seed(11 + step)
+        eps = [torch.randn(cfg.batch_size, cfg.action_dim, generator=g)
+               .cuda() for _ in range(2)]
+        e32._eps_queue = [e.clone() for e in eps]
+        e16._eps_queue = [e.clone() for e in eps]
+        m32 = e32.update({k: v.clone() for k, v in batch.items()})
+        m16 = e16.update({k: v.clone() for k, v in batch.items()})
+    assert abs(m32["critic_loss"] - m16["critic_loss"]) < \
+        0.05 + 0.1 * abs(m32["critic_loss"])
+    assert abs(m32["actor_loss"] - m16["actor_loss"]) < 0.05
+
+    replay = ShardedReplay(4000, cfg.num_tasks, cfg.mtobs_dim,
+                           cfg.action_dim, device="cuda:0")
+    for t in range(cfg.num_tasks):
+        n = 256
+        st = torch.randn(n, cfg.mtobs_dim, device="cuda:0")
+        oh = torch.zeros(n, cfg.num_tasks, device="cuda:0")
+        oh[:, t] = 1
+        st[:, -cfg.num_tasks:] = oh
+        replay.shards[t].append(
+            st, torch.rand(n, cfg.action_dim, device="cuda:0") * 2 - 1,
+            torch.randn(n, 1, device="cuda:0"), st.clone(),
+            torch.zeros(n, 1, device="cuda:0"))
+    e16.capture(replay, cfg.batch_size)
+    losses = []
+    for _ in range(30):
+        m = e16.graphed_update()
+        losses.append(float(m["critic_loss"]))
+    assert all(v == v for v in losses)
+    assert losses[-1] < losses[0] * 1.5
+    # the tie keeps actor SE == critic SE in bf16 mode too
+    for (n_, pa), (_, pc) in zip(
+            e16.actor.state_encoder.named_parameters(),
+            e16.local_critic.state_encoder.named_parameters()):
+        assert torch.equal(pa, pc), n_
