@@ -113,10 +113,9 @@ def main():
     torch.manual_seed(1000 + rank)
 
     ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
-    engine = create_engine(cfg, device)
-    if args.dtype == "bf16" and device.startswith("cuda"):
-        engine.precision = "bf16"
-        engine._init_bf16_mirrors()
+    precision = ("bf16" if args.dtype == "bf16"
+                 and device.startswith("cuda") else "fp32")
+    engine = create_engine(cfg, device, precision=precision)
     if ddp is not None:
         engine.attach_ddp(ddp)
 

@@ -40,9 +40,10 @@ from .sac import SACEngine
 
 
 class CAREEngine(SACEngine):
-    def __init__(self, cfg: SACConfig, device="cpu"):
+    def __init__(self, cfg: SACConfig, device="cpu",
+                 precision: Optional[str] = None):
         assert cfg.variant == "care" and cfg.encoder is not None
-        super().__init__(cfg, device)
+        super().__init__(cfg, device, precision=precision)
 
     # ------------------------------------------------------------------
     def _build_models(self) -> None:
