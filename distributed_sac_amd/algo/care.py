@@ -303,10 +303,13 @@ class CAREEngine(SACEngine):
         super().refresh_bf16(which)
         self._refresh_mixT(which)
 
-    def _se_fwd_fast(self, info, states_2d, z_context):
+    def _se_fwd_fast(self, info, mtobss_2d, z_context):
         """stateEncoder.forward via bf16 kernels (value-identical to the
         module path up to bf16 rounding; skips the reference's divide by
         alpha.sum(dim=1) which is exactly softmax-sum==1)."""
+        # mixture encoders consume the RAW state — strip the one-hot
+        # suffix (reference stateEncoder.mtobss2states_taskIndices)
+        states_2d = mtobss_2d[:, : mtobss_2d.shape[1] - self.num_tasks]
         z_encs = Fops.grouped_mlp_bf16(states_2d, info["mixW"],
                                        info["mixB"], info["mixT"],
                                        info["k"])              # [k,B,50] f32
