@@ -68,6 +68,9 @@ class Learner:
         self.grad_steps = 0
         self.update_timer = StepTimer()
         self.ingest_count = 0
+        # per-phase wall-time accounting (SURVEY §5.1 — absent in reference)
+        self.phase_seconds = {"drain": 0.0, "logs": 0.0, "update": 0.0,
+                              "publish": 0.0, "other": 0.0}
         # failure detection (absent in the reference — SURVEY §5.3):
         # shared wall-clock heartbeats, slot -1 = learner, others = players
         self.heartbeat = heartbeat
@@ -169,6 +172,7 @@ class Learner:
         self.iteration_counter += 1
         metrics = None
         if self.iteration_counter % self.update_delay == 0:
+            t0 = time.perf_counter()
             if self.use_graph:
                 self._ensure_graph()
             if self._graph_ready:
@@ -181,7 +185,10 @@ class Learner:
                 self.engine.update_iteration += 1
             self.grad_steps += 1
             self.update_timer.mark()
+            t1 = time.perf_counter()
+            self.phase_seconds["update"] += t1 - t0
             self.publish()
+            self.phase_seconds["publish"] += time.perf_counter() - t1
             if self.save_dir and self.save_period and \
                     self.grad_steps % self.save_period == 0:
                 save_checkpoint(self.engine, self.save_dir,
@@ -209,8 +216,13 @@ class Learner:
         while True:
             if stop_event is not None and stop_event.is_set():
                 break
+            t0 = time.perf_counter()
             self.drain_queue()
+            t1 = time.perf_counter()
+            self.phase_seconds["drain"] += t1 - t0
             self.drain_logs()
+            t2 = time.perf_counter()
+            self.phase_seconds["logs"] += t2 - t1
             self.check_heartbeats()
             self.train_step()
             if max_grad_steps and self.grad_steps >= max_grad_steps:
@@ -222,4 +234,6 @@ class Learner:
             "iterations": self.iteration_counter,
             "ingested": self.ingest_count,
             "grad_steps_per_sec": self.update_timer.rate(),
+            "phase_seconds": {k: round(v, 2)
+                              for k, v in self.phase_seconds.items()},
         }
