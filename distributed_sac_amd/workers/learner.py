@@ -160,8 +160,12 @@ class Learner:
     def _ensure_graph(self) -> None:
         if self.use_graph and not self._graph_ready:
             try:
+                self.logger.print("capturing update graph...")
+                t0 = time.perf_counter()
                 self.engine.capture(self.replay, self.cfg.batch_size)
                 self._graph_ready = True
+                self.logger.print(
+                    f"graph captured in {time.perf_counter() - t0:.1f}s")
             except Exception as e:  # pragma: no cover
                 self.logger.print(f"hipGraph capture failed ({e!r}); eager")
                 self.use_graph = False
@@ -213,6 +217,7 @@ class Learner:
                 return {}
         self.engine.hard_copy_targets()
         self.logger.print("######### Start train #########")
+        last_report = time.perf_counter()
         while True:
             if stop_event is not None and stop_event.is_set():
                 break
@@ -225,6 +230,13 @@ class Learner:
             self.phase_seconds["logs"] += t2 - t1
             self.check_heartbeats()
             self.train_step()
+            if time.perf_counter() - last_report > 10.0:
+                last_report = time.perf_counter()
+                self.logger.print(
+                    f"it={self.iteration_counter} grad={self.grad_steps} "
+                    f"rate={self.update_timer.rate():.1f}/s "
+                    f"ingested={self.ingest_count} "
+                    f"phases={ {k: round(v, 1) for k, v in self.phase_seconds.items()} }")
             if max_grad_steps and self.grad_steps >= max_grad_steps:
                 break
             if max_seconds and time.perf_counter() - t0 > max_seconds:
