@@ -70,7 +70,8 @@ class Learner:
         self.ingest_count = 0
         # per-phase wall-time accounting (SURVEY §5.1 — absent in reference)
         self.phase_seconds = {"drain": 0.0, "logs": 0.0, "update": 0.0,
-                              "publish": 0.0, "other": 0.0}
+                              "sync": 0.0, "publish": 0.0, "other": 0.0}
+        self._pub_pinned = None
         # failure detection (absent in the reference — SURVEY §5.3):
         # shared wall-clock heartbeats, slot -1 = learner, others = players
         self.heartbeat = heartbeat
@@ -150,8 +151,15 @@ class Learner:
 
     # -- update --------------------------------------------------------
     def publish(self) -> None:
-        self.snapshot.publish(self.engine.publish_params(),
-                              self.iteration_counter)
+        flat = self.engine.publish_params()
+        if flat.is_cuda:
+            if self._pub_pinned is None or \
+                    self._pub_pinned.numel() != flat.numel():
+                self._pub_pinned = torch.empty(flat.numel(),
+                                               pin_memory=True)
+            self._pub_pinned.copy_(flat.reshape(-1), non_blocking=False)
+            flat = self._pub_pinned
+        self.snapshot.publish(flat, self.iteration_counter)
 
     def ready(self) -> bool:
         return len(self.replay) * max(1, self.replay.num_tasks) >= \
@@ -191,8 +199,12 @@ class Learner:
             self.update_timer.mark()
             t1 = time.perf_counter()
             self.phase_seconds["update"] += t1 - t0
+            if self.device.type == "cuda":
+                torch.cuda.synchronize(self.device)
+            t2 = time.perf_counter()
+            self.phase_seconds["sync"] += t2 - t1
             self.publish()
-            self.phase_seconds["publish"] += time.perf_counter() - t1
+            self.phase_seconds["publish"] += time.perf_counter() - t2
             if self.save_dir and self.save_period and \
                     self.grad_steps % self.save_period == 0:
                 save_checkpoint(self.engine, self.save_dir,
@@ -221,13 +233,13 @@ class Learner:
         while True:
             if stop_event is not None and stop_event.is_set():
                 break
-            t0 = time.perf_counter()
+            p0 = time.perf_counter()
             self.drain_queue()
-            t1 = time.perf_counter()
-            self.phase_seconds["drain"] += t1 - t0
+            p1 = time.perf_counter()
+            self.phase_seconds["drain"] += p1 - p0
             self.drain_logs()
-            t2 = time.perf_counter()
-            self.phase_seconds["logs"] += t2 - t1
+            p2 = time.perf_counter()
+            self.phase_seconds["logs"] += p2 - p1
             self.check_heartbeats()
             self.train_step()
             if time.perf_counter() - last_report > 10.0:
