@@ -37,6 +37,7 @@ def main():
     ap.add_argument("--logdir", default=None)
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--no-graph", action="store_true")
+    ap.add_argument("--precision", choices=["fp32", "bf16"], default=None)
     ap.add_argument("--env", default=None,
                     help="env name for make_env (default: synthetic)")
     # eval mode
@@ -93,7 +94,8 @@ def main():
                             env_fn=env_fn, logger=logger,
                             save_dir=args.save_dir,
                             save_period=args.save_period, seed=args.seed,
-                            use_graph=not args.no_graph, ddp=ddp)
+                            use_graph=not args.no_graph, ddp=ddp,
+                            precision=args.precision)
     stats = dt.run(max_grad_steps=args.max_grad_steps,
                    max_seconds=args.max_seconds)
     print(json.dumps(stats))

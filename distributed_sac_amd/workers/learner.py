@@ -40,10 +40,14 @@ class Learner:
                  save_dir: Optional[str] = None, save_period: int = 0,
                  update_delay: Optional[int] = None, use_graph: bool = True,
                  ddp=None, seed: int = 0, heartbeat=None,
-                 heartbeat_timeout: float = 60.0, rings=None):
+                 heartbeat_timeout: float = 60.0, rings=None,
+                 precision: str = None, publish_interval_s: float = 0.02):
         self.cfg = cfg
         self.device = torch.device(device)
-        self.engine = create_engine(cfg, device)
+        if precision is None:
+            precision = "bf16" if torch.device(device).type == "cuda" \
+                else "fp32"
+        self.engine = create_engine(cfg, device, precision=precision)
         if ddp is not None:
             self.engine.attach_ddp(ddp)
         num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
@@ -72,6 +76,13 @@ class Learner:
         self.phase_seconds = {"drain": 0.0, "logs": 0.0, "update": 0.0,
                               "sync": 0.0, "publish": 0.0, "other": 0.0}
         self._pub_pinned = None
+        # publish throttle: the learner updates orders of magnitude faster
+        # than players poll; the reference's version-gated pull (apply only
+        # when update_iteration changed) makes intermediate publishes
+        # unobservable, so we publish at most every publish_interval_s
+        # (docs/DESIGN_NOTES.md) instead of syncing D2H every update.
+        self.publish_interval_s = publish_interval_s
+        self._last_publish = 0.0
         # failure detection (absent in the reference — SURVEY §5.3):
         # shared wall-clock heartbeats, slot -1 = learner, others = players
         self.heartbeat = heartbeat
@@ -199,12 +210,10 @@ class Learner:
             self.update_timer.mark()
             t1 = time.perf_counter()
             self.phase_seconds["update"] += t1 - t0
-            if self.device.type == "cuda":
-                torch.cuda.synchronize(self.device)
-            t2 = time.perf_counter()
-            self.phase_seconds["sync"] += t2 - t1
-            self.publish()
-            self.phase_seconds["publish"] += time.perf_counter() - t2
+            if t1 - self._last_publish >= self.publish_interval_s:
+                self._last_publish = t1
+                self.publish()   # pinned D2H copy syncs the queued updates
+                self.phase_seconds["publish"] += time.perf_counter() - t1
             if self.save_dir and self.save_period and \
                     self.grad_steps % self.save_period == 0:
                 save_checkpoint(self.engine, self.save_dir,

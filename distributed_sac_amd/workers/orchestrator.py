@@ -50,7 +50,7 @@ class DistributedTrainer:
                  logger: Optional[MetricLogger] = None,
                  save_dir: Optional[str] = None, save_period: int = 0,
                  chunk_steps: int = 64, seed: int = 0, use_graph: bool = True,
-                 ddp=None, transport: str = "auto"):
+                 ddp=None, transport: str = "auto", precision: str = None):
         self.cfg = cfg
         self.env_fn = env_fn or default_env_fn
         self.ctx = mp.get_context("spawn")
@@ -86,7 +86,8 @@ class DistributedTrainer:
                                self.log_queue, logger=logger,
                                save_dir=save_dir, save_period=save_period,
                                use_graph=use_graph, ddp=ddp, seed=seed,
-                               heartbeat=self.heartbeat, rings=self.rings)
+                               heartbeat=self.heartbeat, rings=self.rings,
+                               precision=precision)
 
     def start_players(self) -> None:
         for pid, tasks in enumerate(self.partitions):
