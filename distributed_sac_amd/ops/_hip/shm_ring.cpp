@@ -150,8 +150,11 @@ class ShmRing {
     return true;
   }
 
-  // consumer: pops one block as CPU tensors (empty list when drained)
-  std::vector<torch::Tensor> pop() {
+  // consumer: pops one block as CPU tensors (empty list when drained).
+  // pinned=true allocates page-locked tensors so the follow-up H2D copies
+  // are true async DMA (pageable copies measured ~0.5-2 ms each on the
+  // learner hot loop).
+  std::vector<torch::Tensor> pop(bool pinned = false) {
     const uint64_t tail = hdr_->tail.load(std::memory_order_relaxed);
     const uint64_t head = hdr_->head.load(std::memory_order_acquire);
     if (tail == head) return {};
@@ -159,7 +162,8 @@ class ShmRing {
     auto* slot = (Slot*)(slots_ + slot_bytes_ * (tail % hdr_->n_slots));
     const int64_t n = slot->n;
     const int64_t task = slot->task;
-    auto opts = torch::TensorOptions().dtype(torch::kFloat32);
+    auto opts = torch::TensorOptions().dtype(torch::kFloat32)
+        .pinned_memory(pinned);
     auto s = torch::empty({n, (int64_t)Ds}, opts);
     auto a = torch::empty({n, (int64_t)Da}, opts);
     auto r = torch::empty({n, 1}, opts);
@@ -209,7 +213,7 @@ void register_shm_ring(pybind11::module_& m) {
         return new ShmRing(name, 0, 0, 0, 0, /*create=*/false);
       })
       .def("push", &ShmRing::push)
-      .def("pop", &ShmRing::pop)
+      .def("pop", &ShmRing::pop, pybind11::arg("pinned") = false)
       .def("pin", &ShmRing::pin)
       .def("pending", &ShmRing::pending)
       .def("capacity", &ShmRing::capacity)

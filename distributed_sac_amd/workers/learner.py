@@ -116,9 +116,10 @@ class Learner:
     def drain_rings(self, max_blocks: int = 64) -> int:
         """Drain the native shared-memory SPSC rings (one per player)."""
         n = 0
+        pinned = self.device.type == "cuda"
         for ring in self.rings:
             for _ in range(max_blocks):
-                out = ring.pop()
+                out = ring.pop(pinned)
                 if not out:
                     break
                 task = int(out[0].item())
