@@ -83,6 +83,9 @@ class Learner:
         # (docs/DESIGN_NOTES.md) instead of syncing D2H every update.
         self.publish_interval_s = publish_interval_s
         self._last_publish = 0.0
+        self._pub_stream = None
+        self._pub_event = None
+        self._pub_iteration = 0
         # failure detection (absent in the reference — SURVEY §5.3):
         # shared wall-clock heartbeats, slot -1 = learner, others = players
         self.heartbeat = heartbeat
@@ -163,15 +166,33 @@ class Learner:
 
     # -- update --------------------------------------------------------
     def publish(self) -> None:
+        """Two-phase asynchronous publish: the D2H copy runs on a side
+        stream (never blocking the update pipeline); the shared-memory
+        snapshot is written from the pinned staging buffer once the copy's
+        event has completed (checked on the next publish tick).  Players
+        see weights at most one publish interval stale — well inside the
+        reference's own staleness (its players applied weights whenever
+        the pickle download finished)."""
         flat = self.engine.publish_params()
-        if flat.is_cuda:
-            if self._pub_pinned is None or \
-                    self._pub_pinned.numel() != flat.numel():
-                self._pub_pinned = torch.empty(flat.numel(),
-                                               pin_memory=True)
-            self._pub_pinned.copy_(flat.reshape(-1), non_blocking=False)
-            flat = self._pub_pinned
-        self.snapshot.publish(flat, self.iteration_counter)
+        if not flat.is_cuda:
+            self.snapshot.publish(flat, self.iteration_counter)
+            return
+        if self._pub_stream is None:
+            self._pub_stream = torch.cuda.Stream(self.device)
+            self._pub_event = torch.cuda.Event()
+            self._pub_pinned = torch.empty(flat.numel(), pin_memory=True)
+        # phase 1: flush the previous copy if it finished
+        if self._pub_iteration and self._pub_event.query():
+            self.snapshot.publish(self._pub_pinned, self._pub_iteration)
+            self._pub_iteration = 0
+        # phase 2: start a fresh async copy (skip if one is in flight)
+        if self._pub_iteration == 0:
+            self._pub_stream.wait_stream(
+                torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(self._pub_stream):
+                self._pub_pinned.copy_(flat.reshape(-1), non_blocking=True)
+            self._pub_event.record(self._pub_stream)
+            self._pub_iteration = self.iteration_counter
 
     def ready(self) -> bool:
         return len(self.replay) * max(1, self.replay.num_tasks) >= \
