@@ -103,6 +103,11 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
                heartbeat: Optional[torch.Tensor] = None,
                ring_name: Optional[str] = None) -> None:
     """Infinite rollout loop (reference Player.run)."""
+    # rollout inference is tiny (a handful of envs): one intra-op thread.
+    # torch's default pool (~cores/2 PER PROCESS) oversubscribes the host
+    # and starves the learner's HIP runtime threads (measured: host-side
+    # GPU calls stretched from us to ms with 2 default-pool players).
+    torch.set_num_threads(1)
     torch.manual_seed(seed)
     ring = None
     if ring_name is not None:
