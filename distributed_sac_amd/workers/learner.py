@@ -284,11 +284,17 @@ class Learner:
                 break
             if max_seconds and time.perf_counter() - t0 > max_seconds:
                 break
+        if self.device.type == "cuda":
+            torch.cuda.synchronize(self.device)  # completion-accurate rate
+        wall = time.perf_counter() - t0
         return {
             "grad_steps": self.grad_steps,
             "iterations": self.iteration_counter,
             "ingested": self.ingest_count,
-            "grad_steps_per_sec": self.update_timer.rate(),
+            "wall_seconds": round(wall, 2),
+            "grad_steps_per_sec": round(self.grad_steps / max(wall, 1e-9), 1),
+            "env_steps_per_sec": round(self.ingest_count / max(wall, 1e-9), 1),
+            "ring_drops": sum(int(r.dropped()) for r in self.rings),
             "phase_seconds": {k: round(v, 2)
                               for k, v in self.phase_seconds.items()},
         }
