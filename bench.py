@@ -134,7 +134,10 @@ def main():
             torch.cuda.synchronize()
 
     graphed = False
-    if device.startswith("cuda") and not args.no_graph:
+    # RCCL collectives are NOT capturable on this stack (the capture probe
+    # aborts the process via the NCCL watchdog, not a catchable error) —
+    # DP ranks run the eager manual-backward path instead.
+    if device.startswith("cuda") and not args.no_graph and ddp is None:
         try:
             engine.capture(replay, cfg.batch_size)
             graphed = True

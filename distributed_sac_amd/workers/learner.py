@@ -66,7 +66,8 @@ class Learner:
         self.save_period = save_period
         self.update_delay = (update_delay if update_delay is not None
                              else max(1, cfg.update_delay))
-        self.use_graph = use_graph and self.device.type == "cuda"
+        self.use_graph = (use_graph and self.device.type == "cuda"
+                          and (ddp is None or not ddp.enabled))
         self._graph_ready = False
         self.iteration_counter = 0   # reference update_iteration (thinned)
         self.grad_steps = 0
