@@ -120,6 +120,11 @@ class CAREEngine(SACEngine):
             with_grad=False)
         self._twin_local_frozen = ([w.detach() for w in self._twin_local[0]],
                                    [b.detach() for b in self._twin_local[1]])
+        self._aa_arena = torch.zeros(
+            self.actor_group.numel + self.alpha_group.numel,
+            device=self.device)
+        off = self.actor_group.adopt_grad_arena(self._aa_arena, 0)
+        self.alpha_group.adopt_grad_arena(self._aa_arena, off)
         self._init_bf16_mirrors()
         self._init_care_bf16()
         self.hard_copy_targets()
@@ -435,8 +440,7 @@ class CAREEngine(SACEngine):
             aq1, aq2, lp, ls, states, self.log_alpha, T, use_w, self.H_bar_f)
         (policy_loss + loss_log_alpha).backward()
         if self.ddp is not None:
-            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
-            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+            self.ddp.allreduce_grad_(self._aa_arena)
         from ..ops.flat import FusedAdam as _FA
         _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
         self.refresh_bf16("actor")

@@ -194,6 +194,13 @@ class SACEngine:
         # next update) — skipping their computation saves 4 dW GEMMs/step.
         self._twin_local_frozen = ([w.detach() for w in self._twin_local[0]],
                                    [b.detach() for b in self._twin_local[1]])
+        # actor+alpha share one gradient arena: their all-reduce (both due
+        # between the same backward and the fused Adam) is ONE RCCL message
+        self._aa_arena = torch.zeros(
+            self.actor_group.numel + self.alpha_group.numel,
+            device=self.device)
+        off = self.actor_group.adopt_grad_arena(self._aa_arena, 0)
+        self.alpha_group.adopt_grad_arena(self._aa_arena, off)
         self._init_bf16_mirrors()
         self.hard_copy_targets()
 
@@ -550,8 +557,7 @@ class SACEngine:
                 dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
                                             yout, act, 1, 1)
         if self.ddp is not None:
-            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
-            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+            self.ddp.allreduce_grad_(self._aa_arena)
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
         self.refresh_bf16("actor")
@@ -648,8 +654,7 @@ class SACEngine:
             self.H_bar_f)
         (policy_loss + loss_log_alpha).backward()
         if self.ddp is not None:
-            self.ddp.allreduce_grad_(self.actor_group.flat_grad)
-            self.ddp.allreduce_grad_(self.alpha_group.flat_grad)
+            self.ddp.allreduce_grad_(self._aa_arena)
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
         self.refresh_bf16("actor")

@@ -70,6 +70,18 @@ class FlatParams:
         if self.flat_grad is not None:
             self.flat_grad.zero_()
 
+    def adopt_grad_arena(self, arena: torch.Tensor, offset: int) -> int:
+        """Re-home this group's flat gradient into a shared arena slice so
+        several groups can all-reduce as ONE message.  Returns the next
+        free arena offset."""
+        n = self.numel
+        new = arena[offset:offset + n]
+        if self.flat_grad is not None:
+            new.copy_(self.flat_grad)
+        self.flat_grad = new
+        self.rebind_grads()
+        return offset + n
+
     def rebind_grads(self) -> None:
         """Re-point .grad views (autograd can replace .grad if it was None)."""
         for p, off in zip(self.params, self.offsets):
