@@ -23,7 +23,6 @@ MI355X-first mechanics under the reference math:
 
 from __future__ import annotations
 
-import itertools
 from typing import Dict, Optional
 
 import torch

@@ -14,7 +14,7 @@ Works from a CARE checkpoint ``.tar`` + its cfg.
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Optional
 
 import numpy as np
 import torch

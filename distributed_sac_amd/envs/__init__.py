@@ -8,8 +8,6 @@ import succeeds and the synthetic stand-in otherwise (CI has neither).
 
 from __future__ import annotations
 
-from typing import Any, List, Optional, Tuple
-
 from .synthetic import BoxSpace, SyntheticEnv, make_synthetic  # noqa: F401
 
 

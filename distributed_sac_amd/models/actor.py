@@ -22,7 +22,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import functional as Fops
-from .mlp import FusedMLP, build_mlp, weights_init
+from .mlp import build_mlp, weights_init
 
 
 class Actor(nn.Module):

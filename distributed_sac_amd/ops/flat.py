@@ -17,7 +17,7 @@ reference checkpoints (learner.save_checkpoint) round-trip.
 from __future__ import annotations
 
 import math
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, Iterable, List
 
 import torch
 import torch.nn as nn

@@ -26,7 +26,7 @@ bf16-appropriate tolerances.
 from __future__ import annotations
 
 import math
-from typing import Iterable, List, Optional, Sequence
+from typing import Iterable, Optional, Sequence
 
 import torch
 import torch.nn.functional as F

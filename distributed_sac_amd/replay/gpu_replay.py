@@ -25,7 +25,7 @@ replacement draw — at batch 1280 from >=5000 entries the collision rate is
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch

@@ -17,7 +17,6 @@ import torch
 import torch.multiprocessing as mp
 
 from ..config import SACConfig
-from ..models import Actor, LLActor
 from ..utils import MetricLogger
 from .learner import Learner
 from .param_server import ParamSnapshot

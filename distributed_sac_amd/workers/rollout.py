@@ -18,13 +18,12 @@ Semantics kept from the reference:
 
 from __future__ import annotations
 
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch
 
 from ..config import SACConfig
-from ..models import Actor, LLActor
 
 
 def one_hot(idx: int, n: int) -> np.ndarray:

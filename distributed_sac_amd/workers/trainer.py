@@ -9,13 +9,12 @@ fully asynchronous multi-process topology lives in
 
 from __future__ import annotations
 
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, Optional
 
 import torch
 
 from ..algo import create_engine
 from ..config import SACConfig
-from ..envs import make_synthetic
 from ..replay import ShardedReplay
 from ..utils import MetricLogger, StepTimer
 from .rollout import VecRollout
