@@ -714,7 +714,9 @@ class SACEngine:
             "alpha": self.alpha.cpu(),
         }
         if self.variant in ("sac", "vsac"):
-            # LunarLander…/src/learner.py:144-163 key layout
+            # LunarLander…/src/learner.py:144-163 key layout (LL names the
+            # counter 'episode_idx'; keep both for exact-schema readers)
+            state["episode_idx"] = self.update_iteration
             state.update({
                 "local_critic_1": cpu_sd(self.local_critic_1),
                 "local_critic_2": cpu_sd(self.local_critic_2),

@@ -25,6 +25,7 @@ def test_schema_ll(tmp_path):
                 "log_alpha_optimizer", "alpha"):
         assert key in ckpt, key
     assert "local_critic" not in ckpt
+    assert ckpt["episode_idx"] == 123  # LL counter key (learner.py:144-163)
     # actor state_dict keys match the reference LL actor module
     assert "layer_intermediate.0.weight" in ckpt["actor"]
     assert "mu_log_std_layer.weight" in ckpt["actor"]
