@@ -27,6 +27,8 @@ def save_checkpoint(engine, save_dir: str, update_iteration: Optional[int] = Non
     it = engine.update_iteration if update_iteration is None else update_iteration
     state = engine.checkpoint_state()
     state["update_iteration"] = it
+    if "episode_idx" in state:  # LL variant counter name
+        state["episode_idx"] = it
     path = os.path.join(save_dir, f"{prefix}_{it}.tar")
     torch.save(state, path)
     return path
