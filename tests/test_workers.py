@@ -1,6 +1,8 @@
 """Async multi-process topology tests (CPU): players -> queue -> learner
 -> snapshot -> players, with clean shutdown."""
 
+import os
+
 import numpy as np
 import pytest
 import torch
@@ -124,3 +126,36 @@ def test_distributed_trainer_shm_transport():
     stats = dt.run(max_grad_steps=10, max_seconds=120)
     assert stats.get("grad_steps", 0) >= 10
     assert stats["ingested"] >= cfg.start_memory_len
+
+
+def test_shm_ring_stress_and_wrap():
+    """Producer/consumer correctness across ring wrap (sequence-tagged)."""
+    from distributed_sac_amd import ops
+    if not ops.has_native():
+        pytest.skip("native extension not built")
+    import threading
+    ext = ops.native()
+    ring = ext.ShmRing(f"/dsac_stress_{os.getpid()}", 8, 4096, 5, 2)
+    N = 500
+    got = []
+
+    def producer():
+        i = 0
+        while i < N:
+            s = torch.full((4, 5), float(i))
+            ok = ring.push(i % 7, s, torch.zeros(4, 2), torch.zeros(4, 1),
+                           s.clone(), torch.zeros(4, 1))
+            if ok:
+                i += 1
+
+    t = threading.Thread(target=producer)
+    t.start()
+    while len(got) < N:
+        out = ring.pop()
+        if out:
+            got.append((int(out[0].item()), float(out[1][0, 0])))
+    t.join()
+    assert ring.pending() == 0
+    for i, (task, val) in enumerate(got):
+        assert task == i % 7 and val == float(i), (i, task, val)
+
