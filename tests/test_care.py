@@ -395,3 +395,26 @@ def test_care_bf16_fast_se(tmp_path):
             e16.actor.state_encoder.named_parameters(),
             e16.local_critic.state_encoder.named_parameters()):
         assert torch.equal(pa, pc), n_
+
+
+@pytest.mark.gpu
+def test_original_care_bf16_gpu(tmp_path):
+    """Original CARE (trainable context encoder, no mlp_context) on the
+    bf16 fast-SE path: context grads flow, engine trains."""
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path, modified=False)
+    engine = CAREEngine(cfg, "cuda:0", precision="bf16")
+    assert engine.context_group is not None
+    ctx0 = engine.context_group.flat_data.clone()
+    for step in range(5):
+        batch = {k: v.cuda() for k, v in care_batch(cfg, seed=step).items()}
+        m = engine.update(batch)
+    for v in m.values():
+        assert v == v
+    # the context encoder moved (grads from the critic loss only)
+    assert not torch.allclose(ctx0, engine.context_group.flat_data)
+    # actor SE still tied
+    for (n_, pa), (_, pc) in zip(
+            engine.actor.state_encoder.named_parameters(),
+            engine.local_critic.state_encoder.named_parameters()):
+        assert torch.equal(pa, pc), n_

@@ -37,3 +37,6 @@ print("fp32 dx pg:", round(t(lambda: ext.linear_bwd_dx_g(dy32, w, y32m, 1, 2, 0)
 print("bf16 dx pg:", round(t(lambda: ext.linear_bwd_dx_bf16(dyh, wh, yhm, 1, 2, 0)), 2), "us")
 print("fp32 dwdb:", round(t(lambda: ext.linear_bwd_dwdb_g(dy32, x, y32m, 1, 2)), 2), "us")
 print("bf16 dwdb:", round(t(lambda: ext.linear_bwd_dwdb_bf16(dyh, xh, yhm, 1, 2)), 2), "us")
+# (a 2-wave 64x32-tile fwd variant was A/B'd and measured SLOWER —
+# 12.5 vs 12.0 us at N=400, 9.0 vs 7.6 at K=54: the doubled x-tile
+# re-reads outweigh the occupancy gain; variant removed.)
