@@ -510,3 +510,4 @@ class CAREEngine(SACEngine):
                 and ckpt["context_encoder_optimizer"].get("param_groups")):
             self.context_encoder_optimizer.load_state_dict(
                 ckpt["context_encoder_optimizer"])
+        self.refresh_bf16()  # bf16 mirrors (incl. mixT) track the masters

@@ -758,4 +758,5 @@ class SACEngine:
             self.critic_optimizer.load_state_dict(ckpt["critic_optimizer"])
         if "log_alpha_optimizer" in ckpt:
             self.log_alpha_optimizer.load_state_dict(ckpt["log_alpha_optimizer"])
+        self.refresh_bf16()  # bf16 compute mirrors must track loaded masters
 
