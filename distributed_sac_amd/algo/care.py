@@ -352,6 +352,10 @@ class CAREEngine(SACEngine):
         under no_grad (see module doc); the actor head runs ONE batched
         pass over cat(next, states) like the SAC engine (the actor-side
         state encoder is no-grad in both halves)."""
+        import os as _os
+        if (getattr(self, "_se_fast", False) and self.use_modified_care
+                and _os.environ.get("DSAC_NO_MANUAL", "0") != "1"):
+            return self._update_tensors_manual(batch)
         from ..ops import native
         states, actions = batch["states"], batch["actions"]
         rewards, next_states, dones = (batch["rewards"], batch["next_states"],
@@ -455,6 +459,223 @@ class CAREEngine(SACEngine):
                 "actor_loss": policy_loss.detach(),
                 "alpha_loss": loss_log_alpha.detach(),
                 "entropy": entropy.detach()}
+
+
+    # -- hand-rolled backward (modified CARE, bf16) ----------------------
+    @torch.no_grad()
+    def _se_fwd_manual(self, info, states_bf16, zc16, save=False):
+        """stateEncoder forward on raw kernels, optionally saving the
+        activations the manual backward needs.  Value-identical to
+        :meth:`_se_fwd_fast` up to bf16 rounding (the attention pool runs
+        in one fused kernel, fp32 accumulate)."""
+        from ..ops import native
+        ext = native()
+        k = info["k"]
+        acts_m = [states_bf16]
+        h = states_bf16
+        nm = len(info["mixT"])
+        z_encs = None
+        for i in range(nm):
+            last = i == nm - 1
+            h = ext.linear_act_fwd_bf16(h, info["mixT"][i], info["mixB"][i],
+                                        0 if last else 1, k,
+                                        1 if last else 0)
+            if last:
+                z_encs = h                       # f32 [k,M,D]
+            else:
+                acts_m.append(h)
+        tws, tbs, twsh = info["trunk"]
+        acts_t = [zc16]
+        h = zc16
+        nt = len(twsh)
+        for i in range(nt):
+            last = i == nt - 1
+            h = ext.linear_act_fwd_bf16(h, twsh[i], tbs[i].contiguous(),
+                                        0 if last else 1, 1,
+                                        1 if last else 0)
+            if not last:
+                acts_t.append(h)
+        alpha, z_enc16 = ext.attn_pool_fwd(h, z_encs)
+        cws, cbs, cwsh = info["mlpctx"]
+        acts_c = [zc16]
+        hc = zc16
+        nc = len(cwsh)
+        for i in range(nc):
+            last = i == nc - 1
+            hc = ext.linear_act_fwd_bf16(hc, cwsh[i], cbs[i].contiguous(),
+                                         0 if last else 1, 1, 0)
+            if not last:
+                acts_c.append(hc)
+        enc = torch.cat([hc, z_enc16], dim=1)    # bf16 [M, zc+D]
+        if save:
+            return enc, dict(acts_m=acts_m, z_encs=z_encs, acts_t=acts_t,
+                             alpha=alpha, acts_c=acts_c)
+        return enc, None
+
+    @staticmethod
+    def _mlp_bwd_into_grads(ext, dy, acts, wsh, ws, bs, G=1, dx_at_0=False,
+                            transpose_w=0):
+        """Backward an act-fwd chain, writing dW/db straight into the flat
+        fp32 grad views (same loop shape as the SAC manual path)."""
+        n = len(wsh)
+        for i in range(n - 1, -1, -1):
+            act = 1 if i < n - 1 else 0
+            yout = acts[i + 1] if i < n - 1 else acts[i]
+            ext.linear_bwd_dwdb_bf16_out(dy, acts[i], yout, act, G,
+                                         ws[i].grad, bs[i].grad,
+                                         transpose_w)
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, wsh[i], yout, act, G, 0)
+        return dy
+
+    @torch.no_grad()
+    def _update_tensors_manual(self, batch):
+        """Modified-CARE bf16 update with a HAND-ROLLED backward: critic
+        loss gradients flow through the twin heads into the state encoder
+        (attention pool -> mlp_context / trunk / mixture) with every dW/db
+        kernel writing directly into the flat critic-group gradient buffer
+        (reference gradient-flow rules: MT10_Distributed_CARE/src/
+        learner.py:281-404 — trunk input detached, z_context frozen
+        embedding, actor optimizes only its head)."""
+        from ..ops import native
+        from ..ops.flat import FusedAdam as _FA
+        ext = native()
+        states, actions = batch["states"], batch["actions"]
+        rewards, next_states, dones = (batch["rewards"],
+                                       batch["next_states"], batch["dones"])
+        T = self.num_tasks
+        use_w = self.use_weighted_loss
+        B = states.shape[0]
+        A = self.actor.action_dim
+        la_det = self.log_alpha.detach()
+        sd = states.shape[1] - T                 # raw state dim
+        zc_dim = self._se_local["mlpctx"][0][-1].shape[0]
+        D = self._se_local["mixW"][-1].shape[2]
+        info = self._se_local
+        nl_c = len(self._twin_local_bf16)
+        nl_a = len(self._actor_ws_bf16)
+
+        z_context = self.context_encoder(states)         # frozen embedding
+        zc16 = z_context.to(torch.bfloat16)
+        z2 = torch.cat([zc16, zc16], dim=0)
+
+        # ---- ONE batched local-SE forward over [next | current] -------
+        x_cat = torch.cat([next_states[:, :sd], states[:, :sd]], dim=0)
+        enc_cat, se_saved = self._se_fwd_manual(info, x_cat.to(torch.bfloat16),
+                                                z2, save=True)
+
+        # ---- batched actor head + squash ------------------------------
+        ws_f32, bs_f32 = self._actor_weights()
+        out, acts_a = self._mlp_fwd_manual(enc_cat, self._actor_ws_bf16,
+                                           bs_f32)
+        mu, lsr = out[:, :A], out[:, A:]
+        if self._eps_queue:
+            eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
+        else:
+            eps = torch.randn_like(mu)
+        a_cat, lp_cat, tanh_u, ls_cat = ext.squashed_gaussian_fwd(
+            mu, lsr, eps, float(self.actor.k))
+        na, nlp = a_cat[:B], lp_cat[:B]
+        sa, lp = a_cat[B:], lp_cat[B:]
+
+        # ---- TD target (target SE + target heads, no grad) ------------
+        enc_t, _ = self._se_fwd_manual(self._se_target,
+                                       next_states[:, :sd].to(torch.bfloat16),
+                                       zc16)
+        xt = torch.cat([enc_t, na.to(torch.bfloat16)], dim=-1)
+        q1_t, q2_t, _ = self._twin_fwd_manual(xt, self._twin_target_bf16,
+                                              self._twin_target[1])
+        y = ext.td_target_mt(rewards, dones, q1_t, q2_t, nlp, states,
+                             la_det, T, self.gamma, self.reward_scale)
+
+        # ---- critic loss + manual backward ----------------------------
+        x = torch.cat([enc_cat[B:], actions.to(torch.bfloat16)], dim=-1)
+        q1, q2, acts_q = self._twin_fwd_manual(x, self._twin_local_bf16,
+                                               self._twin_local[1])
+        closs = ext.critic_loss_fwd(q1, q2, y, states, la_det, T,
+                                    int(use_w))[0]
+        dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
+                                  int(use_w))
+        wsg, bsg = self._twin_local
+        for i in range(nl_c - 1, -1, -1):
+            act = 1 if i < nl_c - 1 else 0
+            yout = acts_q[i + 1] if i < nl_c - 1 else acts_q[i]
+            ext.linear_bwd_dwdb_bf16_out(dy, acts_q[i], yout, act, 2,
+                                         wsg[i].grad, bsg[i].grad)
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                            yout, act, 2, 0)
+        dx0 = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
+                                     acts_q[1] if nl_c > 1 else acts_q[0],
+                                     1 if nl_c > 1 else 0, 2, 1)  # [B,se+A]
+        # state-encoder backward (states half of the batched acts)
+        zen = se_saved["z_encs"][:, B:].contiguous()     # f32 [k,B,D]
+        dzencs, dlogits = ext.attn_pool_bwd(zen, se_saved["alpha"][B:],
+                                            dx0, dx0.shape[1], zc_dim)
+        cws, cbs, cwsh = info["mlpctx"]
+        self._mlp_bwd_into_grads(ext, dx0[:, :zc_dim].contiguous(),
+                                 [a[B:] for a in se_saved["acts_c"]],
+                                 cwsh, cws, cbs)
+        tws, tbs, twsh = info["trunk"]
+        self._mlp_bwd_into_grads(ext, dlogits,
+                                 [a[B:] for a in se_saved["acts_t"]],
+                                 twsh, tws, tbs)
+        acts_m = [se_saved["acts_m"][0][B:]] +                  [a[:, B:].contiguous() for a in se_saved["acts_m"][1:]]
+        self._mlp_bwd_into_grads(ext, dzencs, acts_m, info["mixT"],
+                                 info["mixW"], info["mixB"],
+                                 G=info["k"], transpose_w=1)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self.critic_optimizer.step()
+        self.refresh_bf16("critic")
+
+        # ---- actor/alpha loss + manual backward -----------------------
+        enc_c, _ = self._se_fwd_manual(info, states[:, :sd].to(torch.bfloat16),
+                                       zc16)               # post-step SE
+        xa = torch.cat([enc_c, sa.to(torch.bfloat16)], dim=-1)
+        aq1, aq2, acts_f = self._twin_fwd_manual(xa, self._twin_local_bf16,
+                                                 self._twin_local[1])
+        al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls_cat[B:], states,
+                                      la_det, T, int(use_w), self.H_bar_f)
+        self.alpha_group.flat_grad.zero_()
+        daq, dlp = ext.actor_alpha_loss_bwd2(
+            aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
+            T, int(use_w), self.H_bar_f)
+        dy = daq
+        for i in range(nl_c - 1, 0, -1):
+            act = 1 if i < nl_c - 1 else 0
+            yout = acts_f[i + 1] if i < nl_c - 1 else acts_f[i]
+            dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                        yout, act, 2, 0)
+        dxa = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
+                                     acts_f[1] if nl_c > 1 else acts_f[0],
+                                     1 if nl_c > 1 else 0, 2, 1)
+        dsa = dxa[:, enc_c.shape[1]:].float()
+        dhead = ext.squashed_gaussian_bwd2(
+            dsa, dlp, lsr[B:], ls_cat[B:], eps[B:], tanh_u[B:],
+            float(self.actor.k))
+        dy = dhead
+        for i in range(nl_a - 1, -1, -1):
+            act = 1 if i < nl_a - 1 else 0
+            yout = (acts_a[i + 1][B:] if i < nl_a - 1 else acts_a[i][B:])
+            ext.linear_bwd_dwdb_bf16_out(dy, acts_a[i][B:], yout, act, 1,
+                                         ws_f32[i].grad, bs_f32[i].grad)
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
+                                            yout, act, 1, 1)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self._aa_arena)
+        _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
+        self.refresh_bf16("actor")
+        self.alpha = self.log_alpha.exp().detach()
+
+        self._polyak_targets()
+        self.refresh_bf16("target")
+        self.tie_actor_state_encoder()
+        return {"critic_loss": closs[0] + closs[1],
+                "actor_loss": al[0],
+                "alpha_loss": al[2],
+                "entropy": al[3]}
 
     @torch.no_grad()
     def publish_params(self) -> torch.Tensor:

@@ -281,7 +281,7 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
     const u16* __restrict__ dy, const u16* __restrict__ x,
     const u16* __restrict__ yout, float* __restrict__ ws,
     float* __restrict__ ws_db, int M, int N, int K, int act, int S,
-    int chunk, long xgs) {
+    int chunk, long xgs, int transpose_w) {
   __shared__ u16 sa[2][TBN][TPAD];    // dy^T: [n][m-run]
   __shared__ u16 sbT[2][TBN][TPAD];   // x^T:  [k][m-run]
   const int gs = blockIdx.z;
@@ -393,9 +393,82 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
       for (int r = 0; r < 4; ++r) {
         const int row = n0 + wr + mi * 16 + fk * 4 + r;
         const int col = c0 + wc + ni * 16 + fi;
-        if (row < N && col < K) wsp[(long)row * K + col] = a[r];
+        // transpose_w: destination is the (G,K,N) master layout (the
+        // mixture Linear holds W as (k,in,out)) — store dW^T directly so
+        // no permute kernel is needed on the manual path.
+        if (row < N && col < K)
+          wsp[transpose_w ? (long)col * N + row : (long)row * K + col] = a[r];
       }
     }
+}
+
+// ---------------------------------------------------------------------------
+// CARE attention pool, fused (reference state_encoder.py:85-94).
+// fwd: alpha = softmax(logits[M,E]); z_enc[M,D] = sum_e alpha_e * z_encs[e,M,D]
+//      (the reference's divide by alpha.sum() is a softmax no-op, skipped).
+// bwd: given dz (grad wrt z_enc, a strided bf16 row-slice of the head-input
+//      gradient), emit  d_z_encs[e] = alpha_e * dz   (bf16, feeds the grouped
+//      mixture backward) and  dlogits = alpha*(s - <alpha,s>), s_e = <dz,z_e>
+//      (bf16, feeds the trunk backward) — one launch instead of the ~8
+//      mul/sum/softmax-backward kernels autograd records.
+// One wavefront per batch row; requires D <= 64 and E <= 16 (cfg: 50, 6).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_attn_pool_fwd(
+    const float* __restrict__ logits, const float* __restrict__ z_encs,
+    float* __restrict__ alpha_out, u16* __restrict__ z_enc_out,
+    int M, int E, int D) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (row >= M) return;
+  float a[16];
+  float mx = -1e30f;
+  for (int e = 0; e < E; ++e) { a[e] = logits[(long)row * E + e]; mx = fmaxf(mx, a[e]); }
+  float den = 0.f;
+  for (int e = 0; e < E; ++e) { a[e] = __expf(a[e] - mx); den += a[e]; }
+  const float inv = 1.f / den;
+  for (int e = 0; e < E; ++e) a[e] *= inv;
+  if (lane < E) alpha_out[(long)row * E + lane] = a[lane];
+  for (int d = lane; d < D; d += 64) {
+    float acc = 0.f;
+    for (int e = 0; e < E; ++e)
+      acc += a[e] * z_encs[((long)e * M + row) * D + d];
+    z_enc_out[(long)row * D + d] = f32_to_bf16_rne(acc);
+  }
+}
+
+__device__ __forceinline__ float wave_sum64(float v) {
+  for (int o = 32; o; o >>= 1) v += __shfl_down(v, o, 64);
+  return __shfl(v, 0, 64);
+}
+
+__global__ __launch_bounds__(256) void k_attn_pool_bwd(
+    const float* __restrict__ z_encs, const float* __restrict__ alpha,
+    const u16* __restrict__ dz, long dz_ld, long dz_off,
+    u16* __restrict__ d_zencs, u16* __restrict__ dlogits,
+    int M, int E, int D) {
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (row >= M) return;
+  float dzv = 0.f;
+  if (lane < D) {
+    union { float f; unsigned u; } v;
+    v.u = ((unsigned)dz[(long)row * dz_ld + dz_off + lane]) << 16;
+    dzv = v.f;
+  }
+  float s[16];
+  float t = 0.f;
+  for (int e = 0; e < E; ++e) {
+    const float ae = alpha[(long)row * E + e];
+    const float ze = lane < D ? z_encs[((long)e * M + row) * D + lane] : 0.f;
+    s[e] = wave_sum64(ze * dzv);
+    t += ae * s[e];
+    if (lane < D)
+      d_zencs[((long)e * M + row) * D + lane] = f32_to_bf16_rne(ae * dzv);
+  }
+  if (lane < E) {
+    const float ae = alpha[(long)row * E + lane];
+    dlogits[(long)row * E + lane] = f32_to_bf16_rne(ae * (s[lane] - t));
+  }
 }
 
 // fp32 partial fold (duplicate of the fp32 TU's reducer; no RDC linking)
@@ -517,7 +590,7 @@ static std::vector<torch::Tensor> linear_bwd_dwdb_bf16(
                      (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
                      (const u16*)yc.data_ptr(), ws.data_ptr<float>(),
                      ws_db.data_ptr<float>(), (int)M, (int)N, (int)K,
-                     (int)act, (int)S, chunk, xgs);
+                     (int)act, (int)S, chunk, xgs, 0);
   auto dw = G == 1 ? torch::empty({N, K}, fopts)
                    : torch::empty({G, N, K}, fopts);
   auto db = G == 1 ? torch::empty({N}, fopts)
@@ -538,7 +611,7 @@ static std::vector<torch::Tensor> linear_bwd_dwdb_bf16(
 static void linear_bwd_dwdb_bf16_out(torch::Tensor dy, torch::Tensor x,
                                      torch::Tensor yout, long act, long G,
                                      torch::Tensor dw_out,
-                                     torch::Tensor db_out) {
+                                     torch::Tensor db_out, long transpose_w) {
   CHECK_BF16(dy); CHECK_BF16(x); CHECK_BF16(yout);
   CHECK_F32(dw_out); CHECK_F32(db_out);
   auto dyc = dy.contiguous(); auto xc = x.contiguous();
@@ -569,7 +642,7 @@ static void linear_bwd_dwdb_bf16_out(torch::Tensor dy, torch::Tensor x,
   hipLaunchKernelGGL(k_bf16_dwdb_splitk, grid, dim3(256), 0, cur_stream2(),
                      (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
                      (const u16*)yc.data_ptr(), ws_p, db_p, (int)M, (int)N,
-                     (int)K, (int)act, (int)S, chunk, xgs);
+                     (int)K, (int)act, (int)S, chunk, xgs, (int)transpose_w);
   if (S > 1) {
     const long tot = G * (N * K + N);
     hipLaunchKernelGGL(k_reduce_dwdb, dim3((tot + 255) / 256), dim3(256), 0,
@@ -578,10 +651,50 @@ static void linear_bwd_dwdb_bf16_out(torch::Tensor dy, torch::Tensor x,
   }
 }
 
+static std::vector<torch::Tensor> attn_pool_fwd(torch::Tensor logits,
+                                                torch::Tensor z_encs) {
+  CHECK_F32(logits); CHECK_F32(z_encs);
+  auto lc = logits.contiguous(); auto zc = z_encs.contiguous();
+  const long M = lc.size(0), E = lc.size(1), D = zc.size(2);
+  TORCH_CHECK(zc.size(0) == E && zc.size(1) == M && E <= 16 && D <= 64,
+              "attn_pool_fwd: need z_encs (E,M,D), E<=16, D<=64");
+  auto alpha = torch::empty({M, E}, lc.options());
+  auto z_enc = torch::empty({M, D}, lc.options().dtype(torch::kBFloat16));
+  hipLaunchKernelGGL(k_attn_pool_fwd, dim3((M + 3) / 4), dim3(256), 0,
+                     cur_stream2(), lc.data_ptr<float>(), zc.data_ptr<float>(),
+                     alpha.data_ptr<float>(), (u16*)z_enc.data_ptr(),
+                     (int)M, (int)E, (int)D);
+  return {alpha, z_enc};
+}
+
+static std::vector<torch::Tensor> attn_pool_bwd(torch::Tensor z_encs,
+                                                torch::Tensor alpha,
+                                                torch::Tensor dz, long dz_ld,
+                                                long dz_off) {
+  CHECK_F32(z_encs); CHECK_F32(alpha); CHECK_BF16(dz);
+  TORCH_CHECK(dz.is_contiguous(), "attn_pool_bwd: dz base must be contiguous");
+  auto ac = alpha.contiguous(); auto zc = z_encs.contiguous();
+  const long E = zc.size(0), M = zc.size(1), D = zc.size(2);
+  TORCH_CHECK(E <= 16 && D <= 64);
+  auto dzencs = torch::empty({E, M, D}, dz.options());
+  auto dlogits = torch::empty({M, E}, dz.options());
+  hipLaunchKernelGGL(k_attn_pool_bwd, dim3((M + 3) / 4), dim3(256), 0,
+                     cur_stream2(), zc.data_ptr<float>(), ac.data_ptr<float>(),
+                     (const u16*)dz.data_ptr(), dz_ld, dz_off,
+                     (u16*)dzencs.data_ptr(), (u16*)dlogits.data_ptr(),
+                     (int)M, (int)E, (int)D);
+  return {dzencs, dlogits};
+}
+
 void register_bf16(pybind11::module_& m) {
   m.def("f32_to_bf16_", &f32_to_bf16_);
+  m.def("attn_pool_fwd", &attn_pool_fwd);
+  m.def("attn_pool_bwd", &attn_pool_bwd);
   m.def("linear_act_fwd_bf16", &linear_act_fwd_bf16);
   m.def("linear_bwd_dx_bf16", &linear_bwd_dx_bf16);
   m.def("linear_bwd_dwdb_bf16", &linear_bwd_dwdb_bf16);
-  m.def("linear_bwd_dwdb_bf16_out", &linear_bwd_dwdb_bf16_out);
+  m.def("linear_bwd_dwdb_bf16_out", &linear_bwd_dwdb_bf16_out,
+        pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("yout"),
+        pybind11::arg("act"), pybind11::arg("G"), pybind11::arg("dw_out"),
+        pybind11::arg("db_out"), pybind11::arg("transpose_w") = 0);
 }
