@@ -284,6 +284,18 @@ class CAREEngine(SACEngine):
                     ws.append(l.weight)
                     bs.append(l.bias)
                 info[name] = (ws, bs, wsh)
+            # fused narrow-chain eligibility: every layer <= 64 wide,
+            # <= 6 layers (k_bf16_mlp_narrow constraints)
+            def narrow_ok(widths, L):
+                return L <= 6 and all(w <= 64 for w in widths)
+            info["mix_narrow"] = narrow_ok(
+                [m.W.shape[2] for m in mix], len(mix))
+            info["trunk_narrow"] = narrow_ok(
+                [w.shape[0] for w in info["trunk"][0]],
+                len(info["trunk"][0]))
+            info["ctx_narrow"] = (info["mlpctx"] is not None and narrow_ok(
+                [w.shape[0] for w in info["mlpctx"][0]],
+                len(info["mlpctx"][0])))
             return info
 
         self._se_local = build(self.local_critic, self.critic_group,
@@ -471,41 +483,60 @@ class CAREEngine(SACEngine):
         from ..ops import native
         ext = native()
         k = info["k"]
-        acts_m = [states_bf16]
-        h = states_bf16
-        nm = len(info["mixT"])
-        z_encs = None
-        for i in range(nm):
-            last = i == nm - 1
-            h = ext.linear_act_fwd_bf16(h, info["mixT"][i], info["mixB"][i],
-                                        0 if last else 1, k,
-                                        1 if last else 0)
-            if last:
-                z_encs = h                       # f32 [k,M,D]
-            else:
-                acts_m.append(h)
+        sv = 1 if save else 0
+        if info["mix_narrow"]:
+            res = ext.mlp_narrow_fwd_bf16(states_bf16, info["mixT"],
+                                          list(info["mixB"]), k, 0, 1, sv)
+            z_encs = res[0]                      # f32 [k,M,D]
+            acts_m = [states_bf16] + res[1:]
+        else:
+            acts_m = [states_bf16]
+            h = states_bf16
+            nm = len(info["mixT"])
+            z_encs = None
+            for i in range(nm):
+                last = i == nm - 1
+                h = ext.linear_act_fwd_bf16(h, info["mixT"][i],
+                                            info["mixB"][i],
+                                            0 if last else 1, k,
+                                            1 if last else 0)
+                if last:
+                    z_encs = h
+                else:
+                    acts_m.append(h)
         tws, tbs, twsh = info["trunk"]
-        acts_t = [zc16]
-        h = zc16
-        nt = len(twsh)
-        for i in range(nt):
-            last = i == nt - 1
-            h = ext.linear_act_fwd_bf16(h, twsh[i], tbs[i].contiguous(),
-                                        0 if last else 1, 1,
-                                        1 if last else 0)
-            if not last:
-                acts_t.append(h)
+        if info["trunk_narrow"]:
+            res = ext.mlp_narrow_fwd_bf16(zc16, twsh, list(tbs), 1, 0, 1, sv)
+            h = res[0]                           # logits f32
+            acts_t = [zc16] + res[1:]
+        else:
+            acts_t = [zc16]
+            h = zc16
+            nt = len(twsh)
+            for i in range(nt):
+                last = i == nt - 1
+                h = ext.linear_act_fwd_bf16(h, twsh[i], tbs[i].contiguous(),
+                                            0 if last else 1, 1,
+                                            1 if last else 0)
+                if not last:
+                    acts_t.append(h)
         alpha, z_enc16 = ext.attn_pool_fwd(h, z_encs)
         cws, cbs, cwsh = info["mlpctx"]
-        acts_c = [zc16]
-        hc = zc16
-        nc = len(cwsh)
-        for i in range(nc):
-            last = i == nc - 1
-            hc = ext.linear_act_fwd_bf16(hc, cwsh[i], cbs[i].contiguous(),
-                                         0 if last else 1, 1, 0)
-            if not last:
-                acts_c.append(hc)
+        if info["ctx_narrow"]:
+            res = ext.mlp_narrow_fwd_bf16(zc16, cwsh, list(cbs), 1, 0, 0, sv)
+            hc = res[0]                          # zc projection, bf16
+            acts_c = [zc16] + res[1:]
+        else:
+            acts_c = [zc16]
+            hc = zc16
+            nc = len(cwsh)
+            for i in range(nc):
+                last = i == nc - 1
+                hc = ext.linear_act_fwd_bf16(hc, cwsh[i],
+                                             cbs[i].contiguous(),
+                                             0 if last else 1, 1, 0)
+                if not last:
+                    acts_c.append(hc)
         enc = torch.cat([hc, z_enc16], dim=1)    # bf16 [M, zc+D]
         if save:
             return enc, dict(acts_m=acts_m, z_encs=z_encs, acts_t=acts_t,

@@ -403,6 +403,164 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
 }
 
 // ---------------------------------------------------------------------------
+// Fused narrow MLP chain (CARE state-encoder trunks): every layer is at
+// most 64 wide, so ONE workgroup can push a 64-row tile of the batch
+// through the WHOLE chain, staging each layer's activation in LDS — one
+// kernel launch instead of one per layer.  Layer 0 streams x/W0 through
+// the usual double-buffered k-loop (K0 arbitrary, e.g. the 768-d RoBERTa
+// embedding); layers >=1 read the staged activation (k = prev width <= 64,
+// a single MFMA k-tile) against an LDS-resident weight.  When `acts` slots
+// are set, intermediate activations are also written to global memory for
+// the manual backward (same compact [G,M,N] layout the per-layer kernels
+// produce).  grid (ceil(M/64), 1, G).
+// ---------------------------------------------------------------------------
+struct NarrowDesc {
+  const u16* w[6];
+  const float* b[6];
+  u16* acts[6];       // saved post-act outputs for layers 0..L-2 (or null)
+  int N[6];
+  int L;
+};
+
+__global__ __launch_bounds__(256) void k_bf16_mlp_narrow(
+    const u16* __restrict__ x, NarrowDesc d, void* __restrict__ y,
+    int M, int K0, long xgs, int act_last, int out_f32) {
+  __shared__ u16 sx[2][TBM][TPAD];
+  __shared__ u16 sw[2][TBM][TPAD];
+  __shared__ u16 sy[TBM][TPAD];
+  const long g = blockIdx.z;
+  const int m0 = blockIdx.x * TBM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
+  const int fi = lane & 15, fk = lane >> 4;
+  const int lr = tid >> 2, lc0 = (tid & 3) * 16;
+  const u16* xg = x + g * xgs;
+  const int rowlim = (M - m0 < TBM ? M - m0 : TBM);
+
+  f32x4 acc00{}, acc01{}, acc10{}, acc11{};
+  {  // ---- layer 0 ----
+    const u16* w0 = d.w[0] + g * (long)d.N[0] * K0;
+    const bool ka = (K0 % 8) == 0;
+    for (int k0 = 0, buf = 0; k0 < K0; k0 += TBK, buf ^= 1) {
+      if (ka && lr < rowlim && k0 + lc0 + 16 <= K0) {
+        const uint4* sp = (const uint4*)&xg[((long)(m0 + lr)) * K0 + k0 + lc0];
+        uint4* dp = (uint4*)&sx[buf][lr][lc0];
+        dp[0] = sp[0]; dp[1] = sp[1];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int gk = k0 + lc0 + j;
+          sx[buf][lr][lc0 + j] = (lr < rowlim && gk < K0)
+              ? xg[((long)(m0 + lr)) * K0 + gk] : (u16)0;
+        }
+      }
+      if (ka && lr < d.N[0] && k0 + lc0 + 16 <= K0) {
+        const uint4* sp = (const uint4*)&w0[(long)lr * K0 + k0 + lc0];
+        uint4* dp = (uint4*)&sw[buf][lr][lc0];
+        dp[0] = sp[0]; dp[1] = sp[1];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int gk = k0 + lc0 + j;
+          sw[buf][lr][lc0 + j] = (lr < d.N[0] && gk < K0)
+              ? w0[(long)lr * K0 + gk] : (u16)0;
+        }
+      }
+      __syncthreads();
+#pragma unroll
+      for (int kk = 0; kk < TBK; kk += 32) {
+        const bf16x8 a0 = *(const bf16x8*)&sx[buf][wr + fi][kk + fk * 8];
+        const bf16x8 a1 = *(const bf16x8*)&sx[buf][wr + 16 + fi][kk + fk * 8];
+        const bf16x8 b0 = *(const bf16x8*)&sw[buf][wc + fi][kk + fk * 8];
+        const bf16x8 b1 = *(const bf16x8*)&sw[buf][wc + 16 + fi][kk + fk * 8];
+        acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+
+  for (int li = 0;; ++li) {
+    const bool last = li == d.L - 1;
+    const int N = d.N[li];
+    const float* bb = d.b[li] + g * (long)N;
+    const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
+    if (last) {
+      float* yf = (float*)y + g * (long)M * N;
+      u16* yh = (u16*)y + g * (long)M * N;
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const f32x4 a = *accs[mi][ni];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = m0 + wr + mi * 16 + fk * 4 + r;
+            const int col = wc + ni * 16 + fi;
+            if (row < M && col < N) {
+              float v = a[r] + bb[col];
+              if (act_last == 1) v = fmaxf(v, 0.f);
+              if (out_f32) yf[(long)row * N + col] = v;
+              else yh[(long)row * N + col] = f32_to_bf16_rne(v);
+            }
+          }
+        }
+      return;
+    }
+    __syncthreads();   // sy free from previous layer's reads
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const f32x4 a = *accs[mi][ni];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = wr + mi * 16 + fk * 4 + r;
+          const int col = wc + ni * 16 + fi;
+          float v = col < N ? fmaxf(a[r] + bb[col], 0.f) : 0.f;
+          sy[row][col] = f32_to_bf16_rne(v);
+        }
+      }
+    __syncthreads();
+    if (d.acts[li] != nullptr) {
+      u16* ag = d.acts[li] + g * (long)M * N;
+      if (lr < rowlim) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int col = lc0 + j;
+          if (col < N) ag[((long)(m0 + lr)) * N + col] = sy[lr][col];
+        }
+      }
+    }
+    // next layer's weight tile: [Nn, K=N<=64] into sw[0]
+    const int Nn = d.N[li + 1];
+    const u16* wn = d.w[li + 1] + g * (long)Nn * N;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int gk = lc0 + j;
+      sw[0][lr][lc0 + j] = (lr < Nn && gk < N) ? wn[(long)lr * N + gk] : (u16)0;
+    }
+    __syncthreads();
+    acc00 = f32x4{}; acc01 = f32x4{}; acc10 = f32x4{}; acc11 = f32x4{};
+#pragma unroll
+    for (int kk = 0; kk < TBK; kk += 32) {
+      const bf16x8 a0 = *(const bf16x8*)&sy[wr + fi][kk + fk * 8];
+      const bf16x8 a1 = *(const bf16x8*)&sy[wr + 16 + fi][kk + fk * 8];
+      const bf16x8 b0 = *(const bf16x8*)&sw[0][wc + fi][kk + fk * 8];
+      const bf16x8 b1 = *(const bf16x8*)&sw[0][wc + 16 + fi][kk + fk * 8];
+      acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+      acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+      acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+      acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
 // CARE attention pool, fused (reference state_encoder.py:85-94).
 // fwd: alpha = softmax(logits[M,E]); z_enc[M,D] = sum_e alpha_e * z_encs[e,M,D]
 //      (the reference's divide by alpha.sum() is a softmax no-op, skipped).
@@ -686,9 +844,59 @@ static std::vector<torch::Tensor> attn_pool_bwd(torch::Tensor z_encs,
   return {dzencs, dlogits};
 }
 
+static std::vector<torch::Tensor> mlp_narrow_fwd_bf16(
+    torch::Tensor x, std::vector<torch::Tensor> ws,
+    std::vector<torch::Tensor> bs, long G, long act_last, long out_f32,
+    long save) {
+  CHECK_BF16(x);
+  const int L = (int)ws.size();
+  TORCH_CHECK(L >= 2 && L <= 6 && bs.size() == ws.size());
+  auto xc = x.contiguous();
+  const bool per_group_x = xc.dim() == 3;
+  const long M = per_group_x ? xc.size(1) : xc.size(0);
+  const long K0 = xc.size(-1);
+  const long xgs = per_group_x ? M * K0 : 0;
+  NarrowDesc d{};
+  d.L = L;
+  std::vector<torch::Tensor> keep;   // contiguity holders
+  long K = K0;
+  std::vector<torch::Tensor> out;
+  out.resize(1);
+  for (int i = 0; i < L; ++i) {
+    CHECK_BF16(ws[i]); CHECK_F32(bs[i]);
+    auto wc = ws[i].contiguous(); auto bc = bs[i].contiguous();
+    keep.push_back(wc); keep.push_back(bc);
+    const long N = wc.numel() / (G * K);
+    TORCH_CHECK(N <= 64, "narrow MLP: layer width must be <= 64");
+    TORCH_CHECK(i == 0 || K <= 64);
+    d.w[i] = (const u16*)wc.data_ptr();
+    d.b[i] = bc.data_ptr<float>();
+    d.N[i] = (int)N;
+    d.acts[i] = nullptr;
+    if (save && i < L - 1) {
+      auto a = G == 1 ? torch::empty({M, N}, xc.options())
+                      : torch::empty({G, M, N}, xc.options());
+      d.acts[i] = (u16*)a.data_ptr();
+      out.push_back(a);
+    }
+    K = N;
+  }
+  auto yopts = xc.options().dtype(out_f32 ? torch::kFloat32
+                                          : torch::kBFloat16);
+  auto y = G == 1 ? torch::empty({M, (long)d.N[L - 1]}, yopts)
+                  : torch::empty({G, M, (long)d.N[L - 1]}, yopts);
+  out[0] = y;
+  dim3 grid((M + TBM - 1) / TBM, 1, G);
+  hipLaunchKernelGGL(k_bf16_mlp_narrow, grid, dim3(256), 0, cur_stream2(),
+                     (const u16*)xc.data_ptr(), d, y.data_ptr(), (int)M,
+                     (int)K0, xgs, (int)act_last, (int)out_f32);
+  return out;
+}
+
 void register_bf16(pybind11::module_& m) {
   m.def("f32_to_bf16_", &f32_to_bf16_);
   m.def("attn_pool_fwd", &attn_pool_fwd);
+  m.def("mlp_narrow_fwd_bf16", &mlp_narrow_fwd_bf16);
   m.def("attn_pool_bwd", &attn_pool_bwd);
   m.def("linear_act_fwd_bf16", &linear_act_fwd_bf16);
   m.def("linear_bwd_dx_bf16", &linear_bwd_dx_bf16);

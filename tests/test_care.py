@@ -476,3 +476,33 @@ def test_care_manual_vs_autograd_gpu(tmp_path, monkeypatch):
                    (e1.alpha_group, e2.alpha_group)):
         d = (g1.flat_data - g2.flat_data).abs().max().item()
         assert d < 3e-3, f"param drift {d}"
+
+
+@pytest.mark.gpu
+def test_mlp_narrow_fused_gpu():
+    """One-launch fused narrow MLP chain vs the per-layer kernels: same
+    loaders/MFMA/rounding -> bitwise-equal activations and outputs."""
+    from distributed_sac_amd.ops import native
+    ext = native()
+    torch.manual_seed(1)
+    for G, K0, M in ((6, 39, 200), (1, 768, 130)):
+        dims = [K0, 50, 50, 50]
+        ws = [torch.randn(max(G, 1), dims[i + 1], dims[i],
+                          device="cuda").to(torch.bfloat16).squeeze(0)
+              if G == 1 else
+              torch.randn(G, dims[i + 1], dims[i],
+                          device="cuda").to(torch.bfloat16)
+              for i in range(3)]
+        bs = [torch.randn(G * dims[i + 1], device="cuda") for i in range(3)]
+        x = torch.randn(M, K0, device="cuda").to(torch.bfloat16)
+        y, a1, a2 = ext.mlp_narrow_fwd_bf16(x, ws, bs, G, 0, 1, 1)
+        h, acts = x, []
+        for i in range(3):
+            last = i == 2
+            h = ext.linear_act_fwd_bf16(h, ws[i], bs[i], 0 if last else 1,
+                                        G, 1 if last else 0)
+            if not last:
+                acts.append(h)
+        assert torch.equal(a1, acts[0])
+        assert torch.equal(a2, acts[1])
+        assert torch.equal(y, h)
