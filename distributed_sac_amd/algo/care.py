@@ -543,22 +543,6 @@ class CAREEngine(SACEngine):
                              alpha=alpha, acts_c=acts_c)
         return enc, None
 
-    @staticmethod
-    def _mlp_bwd_into_grads(ext, dy, acts, wsh, ws, bs, G=1, dx_at_0=False,
-                            transpose_w=0):
-        """Backward an act-fwd chain, writing dW/db straight into the flat
-        fp32 grad views (same loop shape as the SAC manual path)."""
-        n = len(wsh)
-        for i in range(n - 1, -1, -1):
-            act = 1 if i < n - 1 else 0
-            yout = acts[i + 1] if i < n - 1 else acts[i]
-            ext.linear_bwd_dwdb_bf16_out(dy, acts[i], yout, act, G,
-                                         ws[i].grad, bs[i].grad,
-                                         transpose_w)
-            if i > 0:
-                dy = ext.linear_bwd_dx_bf16(dy, wsh[i], yout, act, G, 0)
-        return dy
-
     @torch.no_grad()
     def _update_tensors_manual(self, batch):
         """Modified-CARE bf16 update with a HAND-ROLLED backward: critic
@@ -628,11 +612,17 @@ class CAREEngine(SACEngine):
         dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
                                   int(use_w))
         wsg, bsg = self._twin_local
+        fg_c = self.critic_group.flat_grad
+        arena_c, S_c, ch_c = self._dw_arena("critic",
+                                            self.critic_group.numel, B)
+        base_c = fg_c.data_ptr()
         for i in range(nl_c - 1, -1, -1):
             act = 1 if i < nl_c - 1 else 0
             yout = acts_q[i + 1] if i < nl_c - 1 else acts_q[i]
-            ext.linear_bwd_dwdb_bf16_out(dy, acts_q[i], yout, act, 2,
-                                         wsg[i].grad, bsg[i].grad)
+            ext.linear_bwd_dwdb_arena(
+                dy, acts_q[i], yout, act, 2, arena_c,
+                (wsg[i].grad.data_ptr() - base_c) // 4,
+                (bsg[i].grad.data_ptr() - base_c) // 4, S_c, ch_c, 0)
             if i > 0:
                 dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
                                             yout, act, 2, 0)
@@ -644,17 +634,23 @@ class CAREEngine(SACEngine):
         dzencs, dlogits = ext.attn_pool_bwd(zen, se_saved["alpha"][B:],
                                             dx0, dx0.shape[1], zc_dim)
         cws, cbs, cwsh = info["mlpctx"]
-        self._mlp_bwd_into_grads(ext, dx0[:, :zc_dim].contiguous(),
-                                 [a[B:] for a in se_saved["acts_c"]],
-                                 cwsh, cws, cbs)
+        self._mlp_bwd_arena(ext, dx0[:, :zc_dim].contiguous(),
+                            [a[B:] for a in se_saved["acts_c"]], cwsh,
+                            [w.grad for w in cws], [b.grad for b in cbs],
+                            fg_c, arena_c, S_c, ch_c)
         tws, tbs, twsh = info["trunk"]
-        self._mlp_bwd_into_grads(ext, dlogits,
-                                 [a[B:] for a in se_saved["acts_t"]],
-                                 twsh, tws, tbs)
-        acts_m = [se_saved["acts_m"][0][B:]] +                  [a[:, B:].contiguous() for a in se_saved["acts_m"][1:]]
-        self._mlp_bwd_into_grads(ext, dzencs, acts_m, info["mixT"],
-                                 info["mixW"], info["mixB"],
-                                 G=info["k"], transpose_w=1)
+        self._mlp_bwd_arena(ext, dlogits,
+                            [a[B:] for a in se_saved["acts_t"]], twsh,
+                            [w.grad for w in tws], [b.grad for b in tbs],
+                            fg_c, arena_c, S_c, ch_c)
+        acts_m = ([se_saved["acts_m"][0][B:]]
+                  + [a[:, B:].contiguous() for a in se_saved["acts_m"][1:]])
+        self._mlp_bwd_arena(ext, dzencs, acts_m, info["mixT"],
+                            [w.grad for w in info["mixW"]],
+                            [b.grad for b in info["mixB"]],
+                            fg_c, arena_c, S_c, ch_c,
+                            G=info["k"], transpose_w=1)
+        ext.reduce_arena(arena_c, fg_c, S_c)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()
@@ -686,14 +682,21 @@ class CAREEngine(SACEngine):
             dsa, dlp, lsr[B:], ls_cat[B:], eps[B:], tanh_u[B:],
             float(self.actor.k))
         dy = dhead
+        fg_a = self.actor_group.flat_grad
+        arena_a, S_a, ch_a = self._dw_arena("actor",
+                                            self.actor_group.numel, B)
+        base_a = fg_a.data_ptr()
         for i in range(nl_a - 1, -1, -1):
             act = 1 if i < nl_a - 1 else 0
             yout = (acts_a[i + 1][B:] if i < nl_a - 1 else acts_a[i][B:])
-            ext.linear_bwd_dwdb_bf16_out(dy, acts_a[i][B:], yout, act, 1,
-                                         ws_f32[i].grad, bs_f32[i].grad)
+            ext.linear_bwd_dwdb_arena(
+                dy, acts_a[i][B:], yout, act, 1, arena_a,
+                (ws_f32[i].grad.data_ptr() - base_a) // 4,
+                (bs_f32[i].grad.data_ptr() - base_a) // 4, S_a, ch_a, 0)
             if i > 0:
                 dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
                                             yout, act, 1, 1)
+        ext.reduce_arena(arena_a, fg_a, S_a)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self._aa_arena)
         _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])

@@ -454,6 +454,41 @@ class SACEngine:
             acts.append(h)
         return h[0], h[1], acts
 
+    def _dw_arena(self, name: str, numel: int, B: int):
+        """[S, group_numel] fp32 split-K arena for one backward phase:
+        every layer's partials land at its flat-gradient offsets, and ONE
+        k_reduce_arena launch folds the whole phase's gradient (instead of
+        one reduce per layer).  chunk targets ~8 batch splits, rounded to
+        the 64-row GEMM tile."""
+        store = getattr(self, "_dw_arenas", None)
+        if store is None:
+            store = self._dw_arenas = {}
+        key = (name, numel, B)
+        if key not in store:
+            chunk = ((B + 7) // 8 + 63) // 64 * 64
+            S = (B + chunk - 1) // chunk
+            store[key] = (torch.empty(S, numel, device=self.device),
+                          S, chunk)
+        return store[key]
+
+    @staticmethod
+    def _mlp_bwd_arena(ext, dy, acts, wsh, ws_grad, bs_grad, flat_grad,
+                       arena, S, chunk, G=1, transpose_w=0):
+        """Backward an act-fwd chain; dW/db partials go to the phase arena
+        at the layer's flat-grad offsets (fold happens once per phase)."""
+        n = len(wsh)
+        base = flat_grad.data_ptr()
+        for i in range(n - 1, -1, -1):
+            act = 1 if i < n - 1 else 0
+            yout = acts[i + 1] if i < n - 1 else acts[i]
+            ext.linear_bwd_dwdb_arena(
+                dy, acts[i], yout, act, G, arena,
+                (ws_grad[i].data_ptr() - base) // 4,
+                (bs_grad[i].data_ptr() - base) // 4, S, chunk, transpose_w)
+            if i > 0:
+                dy = ext.linear_bwd_dx_bf16(dy, wsh[i], yout, act, G, 0)
+        return dy
+
     @torch.no_grad()
     def _update_tensors_manual(self, batch):
         """bf16 path with HAND-ROLLED backward: every gradient-producing
@@ -508,14 +543,12 @@ class SACEngine:
                                   int(use_w))
         wsg = [w.grad for w in self._twin_local[0]]
         bsg = [b.grad for b in self._twin_local[1]]
-        for i in range(nl_c - 1, -1, -1):
-            act = 1 if i < nl_c - 1 else 0
-            yout = acts_c[i + 1] if i < nl_c - 1 else acts_c[i]
-            ext.linear_bwd_dwdb_bf16_out(dy, acts_c[i], yout, act, 2,
-                                         wsg[i], bsg[i])
-            if i > 0:
-                dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
-                                            yout, act, 2, 0)
+        fg_c = self.critic_group.flat_grad
+        arena_c, S_c, ch_c = self._dw_arena("critic",
+                                            self.critic_group.numel, B)
+        self._mlp_bwd_arena(ext, dy, acts_c, self._twin_local_bf16,
+                            wsg, bsg, fg_c, arena_c, S_c, ch_c, G=2)
+        ext.reduce_arena(arena_c, fg_c, S_c)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()
@@ -547,14 +580,21 @@ class SACEngine:
         dy = dhead
         wag = [w.grad for w in ws_f32]
         bag = [b.grad for b in bs_f32]
+        fg_a = self.actor_group.flat_grad
+        arena_a, S_a, ch_a = self._dw_arena("actor",
+                                            self.actor_group.numel, B)
+        base_a = fg_a.data_ptr()
         for i in range(nl_a - 1, -1, -1):
             act = 1 if i < nl_a - 1 else 0
             yout = (acts_a[i + 1][B:] if i < nl_a - 1 else acts_a[i][B:])
-            ext.linear_bwd_dwdb_bf16_out(dy, acts_a[i][B:], yout, act, 1,
-                                         wag[i], bag[i])
+            ext.linear_bwd_dwdb_arena(
+                dy, acts_a[i][B:], yout, act, 1, arena_a,
+                (wag[i].data_ptr() - base_a) // 4,
+                (bag[i].data_ptr() - base_a) // 4, S_a, ch_a, 0)
             if i > 0:
                 dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
                                             yout, act, 1, 1)
+        ext.reduce_arena(arena_a, fg_a, S_a)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self._aa_arena)
         FusedAdam.step_many([self.actor_optimizer,

@@ -281,7 +281,8 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
     const u16* __restrict__ dy, const u16* __restrict__ x,
     const u16* __restrict__ yout, float* __restrict__ ws,
     float* __restrict__ ws_db, int M, int N, int K, int act, int S,
-    int chunk, long xgs, int transpose_w) {
+    int chunk, long xgs, int transpose_w, long s_stride, long w_off,
+    long b_off) {
   __shared__ u16 sa[2][TBN][TPAD];    // dy^T: [n][m-run]
   __shared__ u16 sbT[2][TBN][TPAD];   // x^T:  [k][m-run]
   const int gs = blockIdx.z;
@@ -289,8 +290,16 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
   const u16* dyg = dy + (long)g * M * N;
   const u16* yg = yout + (long)g * M * N;
   const u16* xg = x + (long)g * xgs;
-  float* wsp = ws + (long)gs * N * K;
-  float* dbp = ws_db + (long)gs * N;
+  // s_stride>0: phase-arena mode — partials for split s of EVERY layer
+  // land in one [S, group_numel] arena laid out in flat-gradient order
+  // (w_off/b_off = the layer's offsets in the flat buffer), so ONE
+  // reduce per phase folds the whole gradient.
+  float* wsp = s_stride > 0
+      ? ws + (long)s * s_stride + w_off + (long)g * N * K
+      : ws + (long)gs * N * K;
+  float* dbp = s_stride > 0
+      ? ws + (long)s * s_stride + b_off + (long)g * N
+      : ws_db + (long)gs * N;
   const int m_lo = s * chunk;
   const int m_hi = min(M, m_lo + chunk);
   const int n0 = blockIdx.x * TBM, c0 = blockIdx.y * TBN;
@@ -670,6 +679,16 @@ __global__ __launch_bounds__(256) void k_reduce_dwdb(
   }
 }
 
+__global__ __launch_bounds__(256) void k_reduce_arena(
+    const float* __restrict__ arena, float* __restrict__ out, long stride,
+    long n, int S) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  float acc = 0.f;
+  for (int s = 0; s < S; ++s) acc += arena[(long)s * stride + i];
+  out[i] = acc;
+}
+
 // ===========================================================================
 // Host wrappers
 // ===========================================================================
@@ -748,7 +767,7 @@ static std::vector<torch::Tensor> linear_bwd_dwdb_bf16(
                      (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
                      (const u16*)yc.data_ptr(), ws.data_ptr<float>(),
                      ws_db.data_ptr<float>(), (int)M, (int)N, (int)K,
-                     (int)act, (int)S, chunk, xgs, 0);
+                     (int)act, (int)S, chunk, xgs, 0, 0, 0, 0);
   auto dw = G == 1 ? torch::empty({N, K}, fopts)
                    : torch::empty({G, N, K}, fopts);
   auto db = G == 1 ? torch::empty({N}, fopts)
@@ -800,7 +819,7 @@ static void linear_bwd_dwdb_bf16_out(torch::Tensor dy, torch::Tensor x,
   hipLaunchKernelGGL(k_bf16_dwdb_splitk, grid, dim3(256), 0, cur_stream2(),
                      (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
                      (const u16*)yc.data_ptr(), ws_p, db_p, (int)M, (int)N,
-                     (int)K, (int)act, (int)S, chunk, xgs, (int)transpose_w);
+                     (int)K, (int)act, (int)S, chunk, xgs, (int)transpose_w, 0, 0, 0);
   if (S > 1) {
     const long tot = G * (N * K + N);
     hipLaunchKernelGGL(k_reduce_dwdb, dim3((tot + 255) / 256), dim3(256), 0,
@@ -893,10 +912,47 @@ static std::vector<torch::Tensor> mlp_narrow_fwd_bf16(
   return out;
 }
 
+static void linear_bwd_dwdb_arena(torch::Tensor dy, torch::Tensor x,
+                                  torch::Tensor yout, long act, long G,
+                                  torch::Tensor arena, long w_off,
+                                  long b_off, long S, long chunk,
+                                  long transpose_w) {
+  CHECK_BF16(dy); CHECK_BF16(x); CHECK_BF16(yout); CHECK_F32(arena);
+  auto dyc = dy.contiguous(); auto xc = x.contiguous();
+  auto yc = yout.contiguous();
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  const long N = G == 1 ? dyc.size(1) : dyc.size(2);
+  const long K = xc.size(-1);
+  const long xgs = xc.dim() == 3 ? M * K : 0;
+  const long stride = arena.size(1);
+  TORCH_CHECK(arena.dim() == 2 && arena.size(0) >= S && arena.is_contiguous());
+  TORCH_CHECK(w_off + G * N * K <= stride && b_off + G * N <= stride);
+  const long nbx = (N + TBM - 1) / TBM, nby = (K + TBN - 1) / TBN;
+  dim3 grid(nbx, nby, G * S);
+  hipLaunchKernelGGL(k_bf16_dwdb_splitk, grid, dim3(256), 0, cur_stream2(),
+                     (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
+                     (const u16*)yc.data_ptr(), arena.data_ptr<float>(),
+                     nullptr, (int)M, (int)N, (int)K, (int)act, (int)S,
+                     (int)chunk, xgs, (int)transpose_w, stride, w_off,
+                     b_off);
+}
+
+static void reduce_arena(torch::Tensor arena, torch::Tensor out, long S) {
+  CHECK_F32(arena); CHECK_F32(out);
+  TORCH_CHECK(arena.dim() == 2 && arena.is_contiguous()
+              && out.is_contiguous() && arena.size(1) == out.numel());
+  const long n = out.numel();
+  hipLaunchKernelGGL(k_reduce_arena, dim3((n + 255) / 256), dim3(256), 0,
+                     cur_stream2(), arena.data_ptr<float>(),
+                     out.data_ptr<float>(), arena.size(1), n, (int)S);
+}
+
 void register_bf16(pybind11::module_& m) {
   m.def("f32_to_bf16_", &f32_to_bf16_);
   m.def("attn_pool_fwd", &attn_pool_fwd);
   m.def("mlp_narrow_fwd_bf16", &mlp_narrow_fwd_bf16);
+  m.def("linear_bwd_dwdb_arena", &linear_bwd_dwdb_arena);
+  m.def("reduce_arena", &reduce_arena);
   m.def("attn_pool_bwd", &attn_pool_bwd);
   m.def("linear_act_fwd_bf16", &linear_act_fwd_bf16);
   m.def("linear_bwd_dx_bf16", &linear_bwd_dx_bf16);
