@@ -159,23 +159,31 @@ class CAREEngine(SACEngine):
         if self.context_group is not None:
             self.context_group.rebind_grads()
 
-    def _polyak_targets(self) -> None:
+    def _polyak_targets(self, mirror=None) -> None:
         """Q heads at tau, state encoder at state_encoder_tau — two fused
-        kernels over the flat slices (reference learner.py:361-367)."""
-        from ..ops.flat import flat_polyak_  # noqa: F401
+        kernels over the flat slices (reference learner.py:361-367); with
+        ``mirror`` the target bf16 mirror slices refresh in-kernel."""
         se = self._se_numel
         t, s = self.target_group.flat_data, self.critic_group.flat_data
-        self._polyak_slice(t[:se], s[:se], self.se_tau)
-        self._polyak_slice(t[se:], s[se:], self.tau)
+        m0 = mirror[:se] if mirror is not None else None
+        m1 = mirror[se:] if mirror is not None else None
+        self._polyak_slice(t[:se], s[:se], self.se_tau, m0)
+        self._polyak_slice(t[se:], s[se:], self.tau, m1)
 
     @staticmethod
     @torch.no_grad()
-    def _polyak_slice(t: torch.Tensor, s: torch.Tensor, tau: float) -> None:
+    def _polyak_slice(t: torch.Tensor, s: torch.Tensor, tau: float,
+                      mirror=None) -> None:
         from ..ops import has_native, native, native_enabled
         if t.is_cuda and native_enabled() and has_native():
-            native().polyak_(t, s, float(tau))
-        else:
-            t.mul_(1.0 - tau).add_(s, alpha=tau)
+            if mirror is not None:
+                native().polyak_(t, s, float(tau), mirror)
+            else:
+                native().polyak_(t, s, float(tau))
+            return
+        t.mul_(1.0 - tau).add_(s, alpha=tau)
+        if mirror is not None:
+            mirror.copy_(t)
 
     # ------------------------------------------------------------------
     def _sample_care(self, mtobss, z_context, detach_z_encs=False):
@@ -653,8 +661,8 @@ class CAREEngine(SACEngine):
         ext.reduce_arena(arena_c, fg_c, S_c)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
-        self.critic_optimizer.step()
-        self.refresh_bf16("critic")
+        self.critic_optimizer.step()   # adam kernel refreshes flat mirror
+        self._refresh_mixT("critic")   # transposed mixture views still need it
 
         # ---- actor/alpha loss + manual backward -----------------------
         enc_c, _ = self._se_fwd_manual(info, states[:, :sd].to(torch.bfloat16),
@@ -700,11 +708,10 @@ class CAREEngine(SACEngine):
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self._aa_arena)
         _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
-        self.refresh_bf16("actor")
         self.alpha = self.log_alpha.exp().detach()
 
-        self._polyak_targets()
-        self.refresh_bf16("target")
+        self._polyak_targets(mirror=self._target_bf16)
+        self._refresh_mixT("target")
         self.tie_actor_state_encoder()
         return {"critic_loss": closs[0] + closs[1],
                 "actor_loss": al[0],

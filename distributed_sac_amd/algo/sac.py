@@ -222,6 +222,8 @@ class SACEngine:
         self._twin_target_bf16 = self._stack_views(
             self.target_group, self._target_bf16,
             self._critic_layer_pairs(target=True))
+        self.critic_optimizer.bf16_mirror = self._critic_bf16
+        self.actor_optimizer.bf16_mirror = self._actor_bf16
         ws, _ = self._actor_weights()
         self._actor_ws_bf16 = []
         for w in ws:
@@ -551,8 +553,7 @@ class SACEngine:
         ext.reduce_arena(arena_c, fg_c, S_c)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
-        self.critic_optimizer.step()
-        self.refresh_bf16("critic")
+        self.critic_optimizer.step()   # adam kernel also refreshes mirror
 
         # ---- actor/alpha loss + manual backward ----------------------
         xa = torch.cat([states, sa], dim=-1)
@@ -599,11 +600,10 @@ class SACEngine:
             self.ddp.allreduce_grad_(self._aa_arena)
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
-        self.refresh_bf16("actor")
         self.alpha = self.log_alpha.exp().detach()
 
-        flat_polyak_(self.target_group, self.critic_group, self.tau)
-        self.refresh_bf16("target")
+        flat_polyak_(self.target_group, self.critic_group, self.tau,
+                     mirror=getattr(self, "_target_bf16", None))
         return {
             "critic_loss": closs[0] + closs[1],
             "actor_loss": al[0],
@@ -696,11 +696,10 @@ class SACEngine:
             self.ddp.allreduce_grad_(self._aa_arena)
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
-        self.refresh_bf16("actor")
         self.alpha = self.log_alpha.exp().detach()
 
-        flat_polyak_(self.target_group, self.critic_group, self.tau)
-        self.refresh_bf16("target")
+        flat_polyak_(self.target_group, self.critic_group, self.tau,
+                     mirror=getattr(self, "_target_bf16", None))
         return {
             "critic_loss": q_loss.detach(),
             "actor_loss": policy_loss.detach(),

@@ -824,10 +824,20 @@ __global__ void k_adam_prolog(float* __restrict__ state, float lr, float b1,
   state[2] = 1.f / sqrtf(1.f - __powf(b2, step));    // inv_sqrt_bc2
 }
 
+__device__ __forceinline__ unsigned short f32_bf16_rne_d(float f) {
+  union { float f; unsigned u; } v{f};
+  unsigned u = v.u;
+  u += 0x7FFFu + ((u >> 16) & 1u);
+  return (unsigned short)(u >> 16);
+}
+
+// mir (optional): the group's bf16 compute mirror — emitting the refreshed
+// mirror from the SAME kernel kills the separate full-buffer cast launch.
 __global__ __launch_bounds__(256) void k_adam_dev(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ m, float* __restrict__ v,
-    const float* __restrict__ state, long n, float b1, float b2, float eps) {
+    const float* __restrict__ state, long n, float b1, float b2, float eps,
+    unsigned short* __restrict__ mir) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const float step_size = state[1], inv_sqrt_bc2 = state[2];
@@ -836,7 +846,9 @@ __global__ __launch_bounds__(256) void k_adam_dev(
   const float vi = b2 * v[i] + (1.f - b2) * gi * gi;
   m[i] = mi;
   v[i] = vi;
-  p[i] -= step_size * mi / (sqrtf(vi) * inv_sqrt_bc2 + eps);
+  const float pn = p[i] - step_size * mi / (sqrtf(vi) * inv_sqrt_bc2 + eps);
+  p[i] = pn;
+  if (mir != nullptr) mir[i] = f32_bf16_rne_d(pn);
 }
 
 // K9c: all three optimizer groups (critic was stepped earlier; this one
@@ -862,29 +874,36 @@ __global__ __launch_bounds__(256) void k_adam_multi(
     float* p0, const float* g0, float* m0, float* v0, const float* st0, long n0,
     float* p1, const float* g1, float* m1, float* v1, const float* st1, long n1,
     float* p2, const float* g2, float* m2, float* v2, const float* st2, long n2,
-    float b1, float b2, float eps) {
+    float b1, float b2, float eps, unsigned short* mr0, unsigned short* mr1,
+    unsigned short* mr2) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   float* p; const float* g; float* m; float* v; const float* st;
-  if (i < n0) { p = p0; g = g0; m = m0; v = v0; st = st0; }
-  else if ((i -= n0) < n1) { p = p1; g = g1; m = m1; v = v1; st = st1; }
-  else if ((i -= n1) < n2) { p = p2; g = g2; m = m2; v = v2; st = st2; }
+  unsigned short* mr;
+  if (i < n0) { p = p0; g = g0; m = m0; v = v0; st = st0; mr = mr0; }
+  else if ((i -= n0) < n1) { p = p1; g = g1; m = m1; v = v1; st = st1; mr = mr1; }
+  else if ((i -= n1) < n2) { p = p2; g = g2; m = m2; v = v2; st = st2; mr = mr2; }
   else return;
   const float gi = g[i];
   const float mi = b1 * m[i] + (1.f - b1) * gi;
   const float vi = b2 * v[i] + (1.f - b2) * gi * gi;
   m[i] = mi;
   v[i] = vi;
-  p[i] -= st[1] * mi / (sqrtf(vi) * st[2] + eps);
+  const float pn = p[i] - st[1] * mi / (sqrtf(vi) * st[2] + eps);
+  p[i] = pn;
+  if (mr != nullptr) mr[i] = f32_bf16_rne_d(pn);
 }
 
 // ---------------------------------------------------------------------------
 // K10: t = (1-tau)*t + tau*s over flat buffers.
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_polyak(
-    float* __restrict__ t, const float* __restrict__ s, long n, float tau) {
+    float* __restrict__ t, const float* __restrict__ s, long n, float tau,
+    unsigned short* __restrict__ mir) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  t[i] = (1.f - tau) * t[i] + tau * s[i];
+  const float tn = (1.f - tau) * t[i] + tau * s[i];
+  t[i] = tn;
+  if (mir != nullptr) mir[i] = f32_bf16_rne_d(tn);
 }
 
 // ===========================================================================
@@ -1291,10 +1310,17 @@ static void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 
 static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                            torch::Tensor v, torch::Tensor state, double lr,
-                           double b1, double b2, double eps) {
+                           double b1, double b2, double eps,
+                           c10::optional<torch::Tensor> mir_opt) {
   CHECK_IN(p); CHECK_IN(state);
   TORCH_CHECK(state.numel() >= 3, "state = {step, step_size, inv_sqrt_bc2}");
   const long n = p.numel();
+  unsigned short* mp = nullptr;
+  if (mir_opt.has_value() && mir_opt->defined() && mir_opt->numel() > 0) {
+    TORCH_CHECK(mir_opt->numel() == n
+                && mir_opt->scalar_type() == torch::kBFloat16);
+    mp = (unsigned short*)mir_opt->data_ptr();
+  }
   hipLaunchKernelGGL(k_adam_prolog, dim3(1), dim3(1), 0, cur_stream(),
                      state.data_ptr<float>(), (float)lr, (float)b1,
                      (float)b2);
@@ -1302,7 +1328,7 @@ static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      state.data_ptr<float>(), n, (float)b1, (float)b2,
-                     (float)eps);
+                     (float)eps, mp);
 }
 
 static void adam_step_multi_(std::vector<torch::Tensor> ps,
@@ -1311,7 +1337,7 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
                              std::vector<torch::Tensor> vs,
                              std::vector<torch::Tensor> states,
                              std::vector<double> lrs, double b1, double b2,
-                             double eps) {
+                             double eps, std::vector<torch::Tensor> mirs) {
   const size_t G = ps.size();
   TORCH_CHECK(G >= 1 && G <= 3, "1..3 groups");
   float* P[3] = {nullptr, nullptr, nullptr};
@@ -1321,6 +1347,7 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
   float* St[3] = {nullptr, nullptr, nullptr};
   long N[3] = {0, 0, 0};
   double LR[3] = {0, 0, 0};
+  unsigned short* MR[3] = {nullptr, nullptr, nullptr};
   long total = 0;
   for (size_t g = 0; g < G; ++g) {
     CHECK_IN(ps[g]);
@@ -1331,6 +1358,11 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
     St[g] = states[g].data_ptr<float>();
     N[g] = ps[g].numel();
     LR[g] = lrs[g];
+    if (g < mirs.size() && mirs[g].defined() && mirs[g].numel() > 0) {
+      TORCH_CHECK(mirs[g].numel() == N[g]
+                  && mirs[g].scalar_type() == torch::kBFloat16);
+      MR[g] = (unsigned short*)mirs[g].data_ptr();
+    }
     total += N[g];
   }
   hipLaunchKernelGGL(k_adam_prolog3, dim3(1), dim3(3), 0, cur_stream(),
@@ -1341,16 +1373,23 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
                      P[0], Gr[0], M[0], V[0], St[0], N[0],
                      P[1], Gr[1], M[1], V[1], St[1], N[1],
                      P[2], Gr[2], M[2], V[2], St[2], N[2],
-                     (float)b1, (float)b2, (float)eps);
+                     (float)b1, (float)b2, (float)eps, MR[0], MR[1], MR[2]);
 }
 
-static void polyak_(torch::Tensor t, torch::Tensor s, double tau) {
+static void polyak_(torch::Tensor t, torch::Tensor s, double tau,
+                    c10::optional<torch::Tensor> mir_opt) {
   CHECK_IN(t);
   const long n = t.numel();
   TORCH_CHECK(s.numel() == n, "polyak buffers must match");
+  unsigned short* mp = nullptr;
+  if (mir_opt.has_value() && mir_opt->defined() && mir_opt->numel() > 0) {
+    TORCH_CHECK(mir_opt->numel() == n
+                && mir_opt->scalar_type() == torch::kBFloat16);
+    mp = (unsigned short*)mir_opt->data_ptr();
+  }
   hipLaunchKernelGGL(k_polyak, dim3((n + 255) / 256), dim3(256), 0,
                      cur_stream(), t.data_ptr<float>(), s.data_ptr<float>(),
-                     n, (float)tau);
+                     n, (float)tau, mp);
 }
 
 void register_shm_ring(pybind11::module_& m);
@@ -1378,7 +1417,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("actor_alpha_loss_bwd2", &actor_alpha_loss_bwd2);
   mod.def("squashed_gaussian_bwd2", &squashed_gaussian_bwd2);
   mod.def("adam_step_", &adam_step_);
-  mod.def("adam_step_dev_", &adam_step_dev_);
-  mod.def("adam_step_multi_", &adam_step_multi_);
-  mod.def("polyak_", &polyak_);
+  mod.def("adam_step_dev_", &adam_step_dev_,
+          pybind11::arg("p"), pybind11::arg("g"), pybind11::arg("m"),
+          pybind11::arg("v"), pybind11::arg("state"), pybind11::arg("lr"),
+          pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
+          pybind11::arg("mir") = pybind11::none());
+  mod.def("adam_step_multi_", &adam_step_multi_,
+          pybind11::arg("ps"), pybind11::arg("gs"), pybind11::arg("ms"),
+          pybind11::arg("vs"), pybind11::arg("states"), pybind11::arg("lrs"),
+          pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
+          pybind11::arg("mirs") = std::vector<torch::Tensor>());
+  mod.def("polyak_", &polyak_,
+          pybind11::arg("t"), pybind11::arg("s"), pybind11::arg("tau"),
+          pybind11::arg("mir") = pybind11::none());
 }

@@ -17,7 +17,7 @@ reference checkpoints (learner.save_checkpoint) round-trip.
 from __future__ import annotations
 
 import math
-from typing import Dict, Iterable, List
+from typing import Dict, Iterable, List, Optional
 
 import torch
 import torch.nn as nn
@@ -95,17 +95,25 @@ class FlatParams:
         return self.flat_data[off:off + p.numel()].view_as(p.data)
 
 
-def flat_polyak_(target: FlatParams, source: FlatParams, tau: float) -> None:
-    """target <- tau*source + (1-tau)*target over whole flat buffers.
+def flat_polyak_(target: FlatParams, source: FlatParams, tau: float,
+                 mirror: Optional[torch.Tensor] = None) -> None:
+    """target <- tau*source + (1-tau)*target over whole flat buffers; with
+    ``mirror`` set, the target's bf16 compute mirror is refreshed in the
+    same kernel (no separate cast launch).
 
     Requires identical parameter ordering (enforced by matching numel).
     """
     assert target.numel == source.numel
     t, s = target.flat_data, source.flat_data
     if t.is_cuda and native_enabled() and has_native():
-        native().polyak_(t, s, float(tau))
-    else:
-        t.mul_(1.0 - tau).add_(s, alpha=tau)
+        if mirror is not None:
+            native().polyak_(t, s, float(tau), mirror)
+        else:
+            native().polyak_(t, s, float(tau))
+        return
+    t.mul_(1.0 - tau).add_(s, alpha=tau)
+    if mirror is not None:
+        mirror.copy_(t)
 
 
 class FusedAdam:
@@ -141,6 +149,9 @@ class FusedAdam:
         # coefficients live on-device ({step, step_size, inv_sqrt_bc2}) so a
         # captured update keeps correct Adam bias correction across replays.
         self._dev_state = None
+        # optional bf16 compute mirror refreshed in the SAME adam kernel
+        # (set by the engine when mixed precision is on)
+        self.bf16_mirror: Optional[torch.Tensor] = None
         if group.flat_data.is_cuda and native_enabled() and has_native():
             self._dev_state = torch.zeros(3, device=group.flat_data.device)
 
@@ -166,8 +177,13 @@ class FusedAdam:
         m, v = self.exp_avg, self.exp_avg_sq
         b1, b2 = self.betas
         if self._dev_state is not None:
-            native().adam_step_dev_(p, g, m, v, self._dev_state,
-                                    self.lr, b1, b2, self.eps)
+            if self.bf16_mirror is not None:
+                native().adam_step_dev_(p, g, m, v, self._dev_state,
+                                        self.lr, b1, b2, self.eps,
+                                        self.bf16_mirror)
+            else:
+                native().adam_step_dev_(p, g, m, v, self._dev_state,
+                                        self.lr, b1, b2, self.eps)
             return
         self._step_count += 1
         bc1 = 1 - b1 ** self._step_count
@@ -194,7 +210,9 @@ class FusedAdam:
                 [o.exp_avg_sq for o in live],
                 [o._dev_state for o in live],
                 [o.lr for o in live],
-                live[0].betas[0], live[0].betas[1], live[0].eps)
+                live[0].betas[0], live[0].betas[1], live[0].eps,
+                [o.bf16_mirror if o.bf16_mirror is not None
+                 else torch.Tensor() for o in live])
             return
         for o in live:
             o.step()
