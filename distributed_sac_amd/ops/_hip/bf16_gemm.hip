@@ -785,49 +785,6 @@ static std::vector<torch::Tensor> linear_bwd_dwdb_bf16(
   return {dw, db};
 }
 
-static void linear_bwd_dwdb_bf16_out(torch::Tensor dy, torch::Tensor x,
-                                     torch::Tensor yout, long act, long G,
-                                     torch::Tensor dw_out,
-                                     torch::Tensor db_out, long transpose_w) {
-  CHECK_BF16(dy); CHECK_BF16(x); CHECK_BF16(yout);
-  CHECK_F32(dw_out); CHECK_F32(db_out);
-  auto dyc = dy.contiguous(); auto xc = x.contiguous();
-  auto yc = yout.contiguous();
-  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
-  const long N = G == 1 ? dyc.size(1) : dyc.size(2);
-  const long K = xc.size(-1);
-  const long xgs = xc.dim() == 3 ? M * K : 0;
-  const long nbx = (N + TBM - 1) / TBM, nby = (K + TBN - 1) / TBN;
-  long S = std::max<long>(1, 512 / std::max<long>(1, nbx * nby * G));
-  S = std::min<long>(S, (M + TBK - 1) / TBK);
-  const int chunk = (int)(((M + S - 1) / S + TBK - 1) / TBK * TBK);
-  S = (M + chunk - 1) / chunk;
-  auto fopts = dw_out.options();
-  torch::Tensor ws, ws_db;
-  float* ws_p;
-  float* db_p;
-  if (S == 1) {   // partials land directly in the grad views
-    ws_p = dw_out.data_ptr<float>();
-    db_p = db_out.data_ptr<float>();
-  } else {
-    ws = torch::empty({G * S, N, K}, fopts);
-    ws_db = torch::empty({G * S, N}, fopts);
-    ws_p = ws.data_ptr<float>();
-    db_p = ws_db.data_ptr<float>();
-  }
-  dim3 grid(nbx, nby, G * S);
-  hipLaunchKernelGGL(k_bf16_dwdb_splitk, grid, dim3(256), 0, cur_stream2(),
-                     (const u16*)dyc.data_ptr(), (const u16*)xc.data_ptr(),
-                     (const u16*)yc.data_ptr(), ws_p, db_p, (int)M, (int)N,
-                     (int)K, (int)act, (int)S, chunk, xgs, (int)transpose_w, 0, 0, 0);
-  if (S > 1) {
-    const long tot = G * (N * K + N);
-    hipLaunchKernelGGL(k_reduce_dwdb, dim3((tot + 255) / 256), dim3(256), 0,
-                       cur_stream2(), ws_p, db_p, dw_out.data_ptr<float>(),
-                       db_out.data_ptr<float>(), N * K, N, (int)S, (int)G);
-  }
-}
-
 static std::vector<torch::Tensor> attn_pool_fwd(torch::Tensor logits,
                                                 torch::Tensor z_encs) {
   CHECK_F32(logits); CHECK_F32(z_encs);
@@ -957,8 +914,4 @@ void register_bf16(pybind11::module_& m) {
   m.def("linear_act_fwd_bf16", &linear_act_fwd_bf16);
   m.def("linear_bwd_dx_bf16", &linear_bwd_dx_bf16);
   m.def("linear_bwd_dwdb_bf16", &linear_bwd_dwdb_bf16);
-  m.def("linear_bwd_dwdb_bf16_out", &linear_bwd_dwdb_bf16_out,
-        pybind11::arg("dy"), pybind11::arg("x"), pybind11::arg("yout"),
-        pybind11::arg("act"), pybind11::arg("G"), pybind11::arg("dw_out"),
-        pybind11::arg("db_out"), pybind11::arg("transpose_w") = 0);
 }
