@@ -130,3 +130,15 @@ def test_batched_linear_and_attention_pool():
     a = torch.softmax(logits, -1)
     ref = (z * a.unsqueeze(-1)).sum(1)
     assert torch.allclose(pooled, ref, atol=1e-6)
+
+
+def test_dw_arena_split_math():
+    """S/chunk selection for the phase arena: uniform 64-row-aligned batch
+    splits, ~8 target splits, exact cover of B rows."""
+    from distributed_sac_amd.algo.sac import SACEngine
+    for B in (16, 64, 192, 256, 1024, 1280, 2560, 1000):
+        chunk = ((B + 7) // 8 + 63) // 64 * 64
+        S = (B + chunk - 1) // chunk
+        assert chunk % 64 == 0
+        assert S * chunk >= B > (S - 1) * chunk
+        assert S <= 8
