@@ -25,6 +25,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -93,7 +94,14 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    n_gpus = max(args.gpus, world)
+    # the whole-job aggregate must reflect the ranks that actually ran:
+    # under the driver's torchrun launch world == --gpus; a bare
+    # single-process run reports 1 even if --gpus was passed larger.
+    n_gpus = world
+    if args.gpus > world and rank == 0:
+        print(f"[bench] note: --gpus {args.gpus} requested but "
+              f"WORLD_SIZE={world}; reporting n_gpus={world}",
+              file=sys.stderr)
 
     if args.device:
         device = args.device
@@ -189,7 +197,7 @@ def main():
             # derivable): nearest derived reference rate is MT1-CARE
             # 5.1 grad-steps/s @ B=1024 on a GTX 1080.
             "vs_baseline": None,
-            "dtype": args.dtype,
+            "dtype": precision,
             "data": "synthetic",
             "env_steps_per_sec": (round(env_rate, 1) if env_rate else None),
             "config": {
