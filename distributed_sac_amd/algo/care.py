@@ -139,6 +139,7 @@ class CAREEngine(SACEngine):
             if self.context_group is not None:
                 ddp.broadcast_params(self.context_group.flat_data)
             self.hard_copy_targets()
+            self.refresh_bf16()  # mirrors must track broadcast params
 
     @torch.no_grad()
     def tie_actor_state_encoder(self) -> None:
