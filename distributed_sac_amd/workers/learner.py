@@ -287,6 +287,15 @@ class Learner:
                 break
         if self.device.type == "cuda":
             torch.cuda.synchronize(self.device)  # completion-accurate rate
+            # flush any in-flight async publish so players/evaluators see
+            # the final weights
+            if self._pub_iteration and self._pub_event is not None:
+                self._pub_event.synchronize()
+                self.snapshot.publish(self._pub_pinned, self._pub_iteration)
+                self._pub_iteration = 0
+        if self.save_dir and self.grad_steps:
+            save_checkpoint(self.engine, self.save_dir,
+                            update_iteration=self.iteration_counter)
         wall = time.perf_counter() - t0
         return {
             "grad_steps": self.grad_steps,
