@@ -30,6 +30,9 @@ namespace {
 struct RingHeader {
   std::atomic<uint64_t> head;   // next slot to write (producer)
   std::atomic<uint64_t> tail;   // next slot to read (consumer)
+  std::atomic<uint64_t> drops;  // full-ring push failures (any producer) —
+                                // lives in the header so the CONSUMER's
+                                // ring object reports them too
   uint32_t n_slots;
   uint32_t slot_floats;         // payload capacity per slot
   uint32_t state_dim;
@@ -37,7 +40,7 @@ struct RingHeader {
   uint32_t magic;
 };
 
-constexpr uint32_t MAGIC = 0xD5AC0001;
+constexpr uint32_t MAGIC = 0xD5AC0002;
 
 struct Slot {
   uint32_t task;
@@ -85,6 +88,7 @@ class ShmRing {
     if (create) {
       hdr_->head.store(0);
       hdr_->tail.store(0);
+      hdr_->drops.store(0);
       hdr_->n_slots = (uint32_t)n_slots;
       hdr_->slot_floats = (uint32_t)slot_floats;
       hdr_->state_dim = (uint32_t)state_dim;
@@ -113,7 +117,9 @@ class ShmRing {
   }
   int64_t state_dim() const { return hdr_->state_dim; }
   int64_t action_dim() const { return hdr_->action_dim; }
-  int64_t dropped() const { return dropped_; }
+  int64_t dropped() const {
+    return (int64_t)hdr_->drops.load(std::memory_order_relaxed);
+  }
 
   // producer: one transition block -> one slot. Returns false (and counts
   // a drop) when the ring is full — the producer never blocks.
@@ -130,7 +136,7 @@ class ShmRing {
     const uint64_t head = hdr_->head.load(std::memory_order_relaxed);
     const uint64_t tail = hdr_->tail.load(std::memory_order_acquire);
     if (head - tail >= hdr_->n_slots) {
-      ++dropped_;
+      hdr_->drops.fetch_add(1, std::memory_order_relaxed);
       return false;
     }
     auto* slot = (Slot*)(slots_ + slot_bytes_ * (head % hdr_->n_slots));
@@ -193,7 +199,6 @@ class ShmRing {
   void* base_ = nullptr;
   RingHeader* hdr_ = nullptr;
   char* slots_ = nullptr;
-  int64_t dropped_ = 0;
 };
 
 }  // namespace

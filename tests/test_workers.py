@@ -159,3 +159,27 @@ def test_shm_ring_stress_and_wrap():
     for i, (task, val) in enumerate(got):
         assert task == i % 7 and val == float(i), (i, task, val)
 
+
+
+def test_ring_drops_visible_to_consumer():
+    """Full-ring push failures count in the SHARED header: the learner's
+    (consumer) ring object must see drops caused by the producer object
+    (they were producer-process-local before, always reporting 0)."""
+    from distributed_sac_amd import ops
+    if not ops.has_native():
+        pytest.skip("native extension not built")
+    ext = ops.native()
+    name = f"/dsac_test_drops_{os.getpid()}"
+    Ds, Da, cap = 4, 2, 2
+    prod = ext.ShmRing(name, cap, 64 * (2 * Ds + Da + 2), Ds, Da)
+    cons = ext.ShmRing.open(name)
+    blk = [torch.zeros(8, Ds), torch.zeros(8, Da), torch.zeros(8, 1),
+           torch.zeros(8, Ds), torch.zeros(8, 1)]
+    assert prod.push(0, *blk) and prod.push(0, *blk)
+    assert not prod.push(0, *blk)          # full -> dropped
+    assert not prod.push(0, *blk)
+    assert cons.dropped() == 2             # visible on the consumer side
+    assert prod.dropped() == 2
+    assert cons.pop(False)                 # drain one
+    assert prod.push(0, *blk)              # space again
+    assert cons.dropped() == 2
