@@ -156,8 +156,6 @@ class ShardedReplay:
         if seed is not None:
             self.generator = torch.Generator(device=self.device)
             self.generator.manual_seed(seed)
-        self._pinned: Dict[str, torch.Tensor] = {}
-
     def __len__(self) -> int:
         """min over shards (reference replay_buffers.__len__:102-107)."""
         return min(len(s) for s in self.shards)
@@ -167,17 +165,15 @@ class ShardedReplay:
         return sum(len(s) for s in self.shards)
 
     def _stage(self, name: str, arr: np.ndarray) -> torch.Tensor:
-        """numpy -> pinned host tensor -> async device copy."""
+        """numpy -> pinned host tensor -> async device copy.
+
+        Uses torch's caching host allocator (``pin_memory()``): it records
+        the H2D copy's event and defers buffer reuse until it completes —
+        a hand-cached pinned buffer here would let the NEXT append's host
+        write race an in-flight copy."""
         t = torch.from_numpy(np.ascontiguousarray(arr, dtype=np.float32))
         if self.device.type == "cuda":
-            key = name
-            buf = self._pinned.get(key)
-            if buf is None or buf.numel() < t.numel():
-                buf = torch.empty(max(t.numel(), 4096), pin_memory=True)
-                self._pinned[key] = buf
-            staged = buf[: t.numel()].view_as(t.reshape(-1)).view(t.shape)
-            staged.copy_(t)
-            return staged.to(self.device, non_blocking=True)
+            return t.pin_memory().to(self.device, non_blocking=True)
         return t
 
     @torch.no_grad()
