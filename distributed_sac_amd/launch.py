@@ -38,6 +38,9 @@ def main():
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--no-graph", action="store_true")
     ap.add_argument("--precision", choices=["fp32", "bf16"], default=None)
+    ap.add_argument("--eval-every", type=int, default=None,
+                    help="player success-eval period in episodes "
+                         "(default: 40 for Meta-World variants, 0 for LL)")
     ap.add_argument("--env", default=None,
                     help="env name for make_env (default: synthetic)")
     # eval mode
@@ -90,12 +93,18 @@ def main():
     logger = MetricLogger(args.logdir, stdout=True) if args.logdir else \
         MetricLogger(None, stdout=True)
     logger.write_hyperparameters(cfg.raw)
+    # reference MT mains pass eval_episode_idx=40 in training; success
+    # semantics exist for the Meta-World variants only
+    eval_every = 40 if cfg.variant in ("vsac", "mtsac", "care") else 0
+    if args.eval_every is None:
+        args.eval_every = eval_every
     dt = DistributedTrainer(cfg, device=device, num_players=args.players,
                             env_fn=env_fn, logger=logger,
                             save_dir=args.save_dir,
                             save_period=args.save_period, seed=args.seed,
                             use_graph=not args.no_graph, ddp=ddp,
-                            precision=args.precision)
+                            precision=args.precision,
+                            eval_every_episodes=args.eval_every)
     stats = dt.run(max_grad_steps=args.max_grad_steps,
                    max_seconds=args.max_seconds)
     print(json.dumps(stats))

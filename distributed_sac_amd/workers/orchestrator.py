@@ -49,7 +49,8 @@ class DistributedTrainer:
                  logger: Optional[MetricLogger] = None,
                  save_dir: Optional[str] = None, save_period: int = 0,
                  chunk_steps: int = 64, seed: int = 0, use_graph: bool = True,
-                 ddp=None, transport: str = "auto", precision: str = None):
+                 ddp=None, transport: str = "auto", precision: str = None,
+                 eval_every_episodes: int = 0):
         self.cfg = cfg
         self.env_fn = env_fn or default_env_fn
         self.ctx = mp.get_context("spawn")
@@ -64,6 +65,9 @@ class DistributedTrainer:
         self.players: List[mp.Process] = []
         self.chunk_steps = chunk_steps
         self.seed = seed
+        # reference Player(eval_episode_idx=40): periodic deterministic
+        # success-rate evaluation in the players (0 = off)
+        self.eval_every_episodes = eval_every_episodes
         # native shared-memory SPSC rings (one per player) when available;
         # mp.Queue otherwise ("auto"), or forced via transport=
         from .. import ops as _ops
@@ -94,8 +98,8 @@ class DistributedTrainer:
                 target=run_player,
                 args=(pid, self.cfg, self.env_fn, tasks, self.snapshot,
                       self.sample_queue, self.log_queue, self.stop_event,
-                      self.chunk_steps, self.seed + 131 * pid, 2, 0, None,
-                      self.heartbeat,
+                      self.chunk_steps, self.seed + 131 * pid, 2,
+                      self.eval_every_episodes, None, self.heartbeat,
                       self.ring_names[pid] if self.ring_names else None),
                 daemon=True)
             p.start()
