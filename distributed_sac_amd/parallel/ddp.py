@@ -2,9 +2,11 @@
 
 Net-new vs the single-GPU reference: N learner processes (one
 per MI355X, ``torch.distributed`` backend "nccl" = RCCL over xGMI) hold
-bitwise-identical replicas; each update all-reduces the THREE flat gradient
-buffers (critic, actor, log_alpha — one message each, since FlatParams
-already concatenates every tensor of a group).
+bitwise-identical replicas; each update all-reduces TWO flat messages:
+the critic group's gradient and the shared actor+log_alpha arena
+(FlatParams already concatenates every tensor of a group, and the
+actor/alpha groups share one contiguous arena precisely so they ride one
+all-reduce).
 
 Topology note: gradients are ≤6 MB fp32, so the all-reduce is
 latency-bound on xGMI's 7 p2p links — one fused message per group (not
