@@ -196,7 +196,11 @@ def main():
             # (MT10-MTSAC rate logs stripped; LL rate only
             # derivable): nearest derived reference rate is MT1-CARE
             # 5.1 grad-steps/s @ B=1024 on a GTX 1080.
-            "vs_baseline": None,
+            # BASELINE.md gives a derivable reference rate only for the
+            # LunarLander config (55,290 grad steps over the 0.45 h logged
+            # run = ~34.1/s on a GTX 1080); the MT rates are unpublished.
+            "vs_baseline": (round(value / 34.1, 1)
+                            if args.config == "sac" else None),
             "dtype": precision,
             "data": "synthetic",
             "env_steps_per_sec": (round(env_rate, 1) if env_rate else None),
