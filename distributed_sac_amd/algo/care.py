@@ -643,23 +643,55 @@ class CAREEngine(SACEngine):
         dzencs, dlogits = ext.attn_pool_bwd(zen, se_saved["alpha"][B:],
                                             dx0, dx0.shape[1], zc_dim)
         cws, cbs, cwsh = info["mlpctx"]
-        self._mlp_bwd_arena(ext, dx0[:, :zc_dim].contiguous(),
-                            [a[B:] for a in se_saved["acts_c"]], cwsh,
-                            [w.grad for w in cws], [b.grad for b in cbs],
-                            fg_c, arena_c, S_c, ch_c)
         tws, tbs, twsh = info["trunk"]
-        self._mlp_bwd_arena(ext, dlogits,
-                            [a[B:] for a in se_saved["acts_t"]], twsh,
-                            [w.grad for w in tws], [b.grad for b in tbs],
-                            fg_c, arena_c, S_c, ch_c)
         acts_m = ([se_saved["acts_m"][0][B:]]
                   + [a[:, B:].contiguous() for a in se_saved["acts_m"][1:]])
-        self._mlp_bwd_arena(ext, dzencs, acts_m, info["mixT"],
-                            [w.grad for w in info["mixW"]],
-                            [b.grad for b in info["mixB"]],
-                            fg_c, arena_c, S_c, ch_c,
-                            G=info["k"], transpose_w=1)
-        ext.reduce_arena(arena_c, fg_c, S_c)
+        import os as _os
+        use_fused_se_bwd = (
+            _os.environ.get("DSAC_NARROW_BWD", "0") == "1"
+            and info["mix_narrow"] and info["trunk_narrow"]
+            and info["ctx_narrow"])
+        if use_fused_se_bwd:
+            # round-2 staging (docs/ROADMAP_r2.md item 3): whole-chain
+            # fused backward, one launch per chain, partials per 64-row
+            # block; OFF by default until GPU-validated.
+            base = fg_c.data_ptr()
+
+            def offs(ps):
+                return [(p_.grad.data_ptr() - base) // 4 for p_ in ps]
+            arena2, S2, _ = self._dw_arena("se@rowblocks",
+                                           self.critic_group.numel, B)
+            ext.mlp_narrow_bwd_bf16(dx0[:, :zc_dim].contiguous(),
+                                    [a[B:] for a in se_saved["acts_c"]],
+                                    cwsh, arena2, offs(cws), offs(cbs),
+                                    1, 0)
+            ext.mlp_narrow_bwd_bf16(dlogits,
+                                    [a[B:] for a in se_saved["acts_t"]],
+                                    twsh, arena2, offs(tws), offs(tbs),
+                                    1, 0)
+            ext.mlp_narrow_bwd_bf16(dzencs, acts_m, info["mixT"], arena2,
+                                    offs(info["mixW"]),
+                                    offs(info["mixB"]), info["k"], 1)
+            se_n = self._se_numel
+            ext.reduce_arena(arena2, fg_c, S2, 0, se_n)
+            ext.reduce_arena(arena_c, fg_c, S_c, se_n, -1)
+        else:
+            self._mlp_bwd_arena(ext, dx0[:, :zc_dim].contiguous(),
+                                [a[B:] for a in se_saved["acts_c"]], cwsh,
+                                [w.grad for w in cws],
+                                [b.grad for b in cbs],
+                                fg_c, arena_c, S_c, ch_c)
+            self._mlp_bwd_arena(ext, dlogits,
+                                [a[B:] for a in se_saved["acts_t"]], twsh,
+                                [w.grad for w in tws],
+                                [b.grad for b in tbs],
+                                fg_c, arena_c, S_c, ch_c)
+            self._mlp_bwd_arena(ext, dzencs, acts_m, info["mixT"],
+                                [w.grad for w in info["mixW"]],
+                                [b.grad for b in info["mixB"]],
+                                fg_c, arena_c, S_c, ch_c,
+                                G=info["k"], transpose_w=1)
+            ext.reduce_arena(arena_c, fg_c, S_c)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()   # adam kernel refreshes flat mirror

@@ -467,8 +467,13 @@ class SACEngine:
             store = self._dw_arenas = {}
         key = (name, numel, B)
         if key not in store:
-            chunk = ((B + 7) // 8 + 63) // 64 * 64
-            S = (B + chunk - 1) // chunk
+            if name.endswith("@rowblocks"):
+                # one arena row per 64-row block (fused narrow backward)
+                S = (B + 63) // 64
+                chunk = 64
+            else:
+                chunk = ((B + 7) // 8 + 63) // 64 * 64
+                S = (B + chunk - 1) // chunk
             store[key] = (torch.empty(S, numel, device=self.device),
                           S, chunk)
         return store[key]

@@ -570,6 +570,166 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow(
 }
 
 // ---------------------------------------------------------------------------
+// Fused narrow MLP chain BACKWARD (round-2 staging — compiled and bound
+// but NOT called by default: no GPU budget remained this round to
+// validate it; enable via CAREEngine env gate DSAC_NARROW_BWD=1 after
+// the bitwise check against the per-layer path).
+//
+// One workgroup owns 64 batch rows of group g and walks the whole chain
+// backward: per layer, (a) mask dy by the saved post-ReLU activation,
+// (b) dW/db partial = dy^T @ a_i via MFMA into the caller's arena at
+// arena[s=blockIdx.x][w_off_i] (flat-gradient layout, one reduce for the
+// whole chain set), (c) dy_{i-1} = (dy_i @ W_i) * relu'(a_i).  The chain
+// input is detached in every use (reference gradient-flow rules), so no
+// dx0 is produced.  All layers <= 64 wide; L <= 6.
+// grid (ceil(M/64), 1, G); arena S = gridDim.x.
+// ---------------------------------------------------------------------------
+struct NarrowBwdDesc {
+  const u16* w[6];      // kernel-layout weights [G,N_i,K_i] (bf16 mirrors)
+  const u16* acts[6];   // acts[0]=chain input x; acts[i]=post-act out of i-1
+  long w_off[6];        // flat-gradient offsets for dW_i
+  long b_off[6];        //   "                 "     db_i
+  int N[6];             // layer widths; K_i = N_{i-1} (K_0 = Kin)
+  int L;
+  int transpose_w;      // store dW^T (the mixture's (k,in,out) masters)
+};
+
+__global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
+    const u16* __restrict__ dy_last, NarrowBwdDesc d,
+    float* __restrict__ arena, long s_stride, int M, int Kin) {
+  __shared__ u16 s_dy[TBM][TPAD];    // dy_i   [m][n]
+  __shared__ u16 s_dyT[TBM][TPAD];   // dy_i^T [n][m]
+  __shared__ u16 s_a[TBM][TPAD];     // a_i    [m][k]
+  __shared__ u16 s_aT[TBM][TPAD];    // a_i^T  [k][m]
+  __shared__ u16 s_wT[TBM][TPAD];    // W_i^T  [k][n]
+  const long g = blockIdx.z;
+  const int s = blockIdx.x;
+  const int m0 = s * TBM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
+  const int fi = lane & 15, fk = lane >> 4;
+  const int lr = tid >> 2, lc0 = (tid & 3) * 16;
+  const int rowlim = (M - m0 < TBM ? M - m0 : TBM);
+
+  // ---- stage dy_{L-1} (no activation on the last layer) --------------
+  {
+    const int N = d.N[d.L - 1];
+    const u16* dyg = dy_last + g * (long)M * N;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      const int col = lc0 + j;
+      const u16 v = (lr < rowlim && col < N)
+          ? dyg[((long)(m0 + lr)) * N + col] : (u16)0;
+      s_dy[lr][col] = v;
+      s_dyT[col][lr] = v;
+    }
+  }
+
+  for (int i = d.L - 1; i >= 0; --i) {
+    const int N = d.N[i];
+    const int K = i == 0 ? Kin : d.N[i - 1];
+    // ---- stage a_i (natural + transposed); zero-pad ------------------
+    {
+      const u16* ag = d.acts[i] + g * (long)M * K;
+      __syncthreads();
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const int col = lc0 + j;
+        const u16 v = (lr < rowlim && col < K)
+            ? ag[((long)(m0 + lr)) * K + col] : (u16)0;
+        s_a[lr][col] = v;
+        s_aT[col][lr] = v;
+      }
+    }
+    __syncthreads();
+    // ---- dW_i = dy^T @ a_i ; db_i = col-sums of dy -------------------
+    {
+      f32x4 acc00{}, acc01{}, acc10{}, acc11{};
+#pragma unroll
+      for (int kk = 0; kk < TBK; kk += 32) {
+        const bf16x8 a0 = *(const bf16x8*)&s_dyT[wr + fi][kk + fk * 8];
+        const bf16x8 a1 = *(const bf16x8*)&s_dyT[wr + 16 + fi][kk + fk * 8];
+        const bf16x8 b0 = *(const bf16x8*)&s_aT[wc + fi][kk + fk * 8];
+        const bf16x8 b1 = *(const bf16x8*)&s_aT[wc + 16 + fi][kk + fk * 8];
+        acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+      }
+      float* wsp = arena + (long)s * s_stride + d.w_off[i] + g * (long)N * K;
+      const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const f32x4 a = *accs[mi][ni];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = wr + mi * 16 + fk * 4 + r;   // n
+            const int col = wc + ni * 16 + fi;           // k
+            if (row < N && col < K)
+              wsp[d.transpose_w ? (long)col * N + row
+                                : (long)row * K + col] = a[r];
+          }
+        }
+      if (tid < TBN && tid < N) {
+        float db = 0.f;
+#pragma unroll
+        for (int m = 0; m < TBM; ++m) {
+          union { float f; unsigned u; } v;
+          v.u = ((unsigned)s_dyT[tid][m]) << 16;
+          db += v.f;
+        }
+        arena[(long)s * s_stride + d.b_off[i] + g * (long)N + tid] = db;
+      }
+    }
+    if (i == 0) return;
+    // ---- dy_{i-1} = (dy_i @ W_i) * relu-mask(a_i) --------------------
+    {
+      const u16* wg = d.w[i] + g * (long)N * K;
+      __syncthreads();
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {           // W^T: [k][n]
+        const int gk = lc0 + j;
+        s_wT[gk][lr] = (lr < N && gk < K)
+            ? wg[(long)lr * K + gk] : (u16)0;
+      }
+      __syncthreads();
+      f32x4 acc00{}, acc01{}, acc10{}, acc11{};
+#pragma unroll
+      for (int kk = 0; kk < TBK; kk += 32) {
+        const bf16x8 a0 = *(const bf16x8*)&s_dy[wr + fi][kk + fk * 8];
+        const bf16x8 a1 = *(const bf16x8*)&s_dy[wr + 16 + fi][kk + fk * 8];
+        const bf16x8 b0 = *(const bf16x8*)&s_wT[wc + fi][kk + fk * 8];
+        const bf16x8 b1 = *(const bf16x8*)&s_wT[wc + 16 + fi][kk + fk * 8];
+        acc00 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b0, acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b1, acc01, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
+      }
+      __syncthreads();   // s_dy free for the next layer's image
+      const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const f32x4 a = *accs[mi][ni];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = wr + mi * 16 + fk * 4 + r;   // m
+            const int col = wc + ni * 16 + fi;           // k (< 64)
+            u16 v = 0;
+            if (s_a[row][col] != 0) v = f32_to_bf16_rne(a[r]);
+            s_dy[row][col] = v;
+            s_dyT[col][row] = v;
+          }
+        }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // CARE attention pool, fused (reference state_encoder.py:85-94).
 // fwd: alpha = softmax(logits[M,E]); z_enc[M,D] = sum_e alpha_e * z_encs[e,M,D]
 //      (the reference's divide by alpha.sum() is a softmax no-op, skipped).
@@ -681,12 +841,12 @@ __global__ __launch_bounds__(256) void k_reduce_dwdb(
 
 __global__ __launch_bounds__(256) void k_reduce_arena(
     const float* __restrict__ arena, float* __restrict__ out, long stride,
-    long n, int S) {
+    long n, int S, long lo) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   float acc = 0.f;
-  for (int s = 0; s < S; ++s) acc += arena[(long)s * stride + i];
-  out[i] = acc;
+  for (int s = 0; s < S; ++s) acc += arena[(long)s * stride + lo + i];
+  out[lo + i] = acc;
 }
 
 // ===========================================================================
@@ -894,14 +1054,63 @@ static void linear_bwd_dwdb_arena(torch::Tensor dy, torch::Tensor x,
                      b_off);
 }
 
-static void reduce_arena(torch::Tensor arena, torch::Tensor out, long S) {
+static void reduce_arena(torch::Tensor arena, torch::Tensor out, long S,
+                         long lo, long hi) {
   CHECK_F32(arena); CHECK_F32(out);
   TORCH_CHECK(arena.dim() == 2 && arena.is_contiguous()
               && out.is_contiguous() && arena.size(1) == out.numel());
-  const long n = out.numel();
+  if (hi < 0) hi = out.numel();
+  const long n = hi - lo;
+  TORCH_CHECK(lo >= 0 && n >= 0 && hi <= out.numel());
+  if (n == 0) return;
   hipLaunchKernelGGL(k_reduce_arena, dim3((n + 255) / 256), dim3(256), 0,
                      cur_stream2(), arena.data_ptr<float>(),
-                     out.data_ptr<float>(), arena.size(1), n, (int)S);
+                     out.data_ptr<float>(), arena.size(1), n, (int)S, lo);
+}
+
+static void mlp_narrow_bwd_bf16(torch::Tensor dy_last,
+                                std::vector<torch::Tensor> acts,
+                                std::vector<torch::Tensor> ws,
+                                torch::Tensor arena,
+                                std::vector<long> w_offs,
+                                std::vector<long> b_offs, long G,
+                                long transpose_w) {
+  CHECK_BF16(dy_last); CHECK_F32(arena);
+  const int L = (int)ws.size();
+  TORCH_CHECK(L >= 2 && L <= 6 && (int)acts.size() == L
+              && (int)w_offs.size() == L && (int)b_offs.size() == L);
+  TORCH_CHECK(arena.dim() == 2 && arena.is_contiguous());
+  auto dyc = dy_last.contiguous();
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  NarrowBwdDesc d{};
+  d.L = L;
+  d.transpose_w = (int)transpose_w;
+  std::vector<torch::Tensor> keep;
+  long Kin = acts[0].size(-1);
+  long K = Kin;
+  for (int i = 0; i < L; ++i) {
+    CHECK_BF16(ws[i]); CHECK_BF16(acts[i]);
+    auto wc = ws[i].contiguous(); auto ac = acts[i].contiguous();
+    keep.push_back(wc); keep.push_back(ac);
+    const long N = wc.numel() / (G * K);
+    TORCH_CHECK(N <= 64 && (i == 0 || K <= 64) && Kin <= 64);
+    d.w[i] = (const u16*)wc.data_ptr();
+    d.acts[i] = (const u16*)ac.data_ptr();
+    d.w_off[i] = w_offs[i];
+    d.b_off[i] = b_offs[i];
+    d.N[i] = (int)N;
+    TORCH_CHECK(w_offs[i] + G * N * K <= arena.size(1)
+                && b_offs[i] + G * N <= arena.size(1));
+    K = N;
+  }
+  TORCH_CHECK(dyc.numel() == G * M * d.N[L - 1]);
+  const long S = (M + TBM - 1) / TBM;
+  TORCH_CHECK(arena.size(0) >= S, "arena rows must cover ceil(M/64)");
+  dim3 grid(S, 1, G);
+  hipLaunchKernelGGL(k_bf16_mlp_narrow_bwd, grid, dim3(256), 0,
+                     cur_stream2(), (const u16*)dyc.data_ptr(), d,
+                     arena.data_ptr<float>(), arena.size(1), (int)M,
+                     (int)Kin);
 }
 
 void register_bf16(pybind11::module_& m) {
@@ -909,7 +1118,10 @@ void register_bf16(pybind11::module_& m) {
   m.def("attn_pool_fwd", &attn_pool_fwd);
   m.def("mlp_narrow_fwd_bf16", &mlp_narrow_fwd_bf16);
   m.def("linear_bwd_dwdb_arena", &linear_bwd_dwdb_arena);
-  m.def("reduce_arena", &reduce_arena);
+  m.def("reduce_arena", &reduce_arena, pybind11::arg("arena"),
+        pybind11::arg("out"), pybind11::arg("S"),
+        pybind11::arg("lo") = 0, pybind11::arg("hi") = -1);
+  m.def("mlp_narrow_bwd_bf16", &mlp_narrow_bwd_bf16);
   m.def("attn_pool_bwd", &attn_pool_bwd);
   m.def("linear_act_fwd_bf16", &linear_act_fwd_bf16);
   m.def("linear_bwd_dx_bf16", &linear_bwd_dx_bf16);
