@@ -629,22 +629,23 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
   for (int i = d.L - 1; i >= 0; --i) {
     const int N = d.N[i];
     const int K = i == 0 ? Kin : d.N[i - 1];
-    // ---- stage a_i (natural + transposed); zero-pad ------------------
-    {
-      const u16* ag = d.acts[i] + g * (long)M * K;
-      __syncthreads();
+    const u16* ag = d.acts[i] + g * (long)M * K;
+    float* wsp = arena + (long)s * s_stride + d.w_off[i] + g * (long)N * K;
+    // ---- dW_i = dy^T @ a_i, looping a_i's columns in 64-wide tiles
+    // (layer 0 of the trunk/mlp_context chains has K = 768) -----------
+    for (int c0 = 0; c0 < K; c0 += TBN) {
+      __syncthreads();   // prior readers of s_a/s_aT done
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
-        const int col = lc0 + j;
+        const int col = c0 + lc0 + j;
         const u16 v = (lr < rowlim && col < K)
             ? ag[((long)(m0 + lr)) * K + col] : (u16)0;
-        s_a[lr][col] = v;
-        s_aT[col][lr] = v;
+        s_aT[lc0 + j][lr] = v;
+        // natural image only needed for the dx relu mask (i>0, where
+        // K <= 64 so this is the one and only tile)
+        s_a[lr][lc0 + j] = v;
       }
-    }
-    __syncthreads();
-    // ---- dW_i = dy^T @ a_i ; db_i = col-sums of dy -------------------
-    {
+      __syncthreads();
       f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 #pragma unroll
       for (int kk = 0; kk < TBK; kk += 32) {
@@ -657,7 +658,6 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
         acc10 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b0, acc10, 0, 0, 0);
         acc11 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b1, acc11, 0, 0, 0);
       }
-      float* wsp = arena + (long)s * s_stride + d.w_off[i] + g * (long)N * K;
       const f32x4* accs[2][2] = {{&acc00, &acc01}, {&acc10, &acc11}};
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi)
@@ -667,22 +667,22 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const int row = wr + mi * 16 + fk * 4 + r;   // n
-            const int col = wc + ni * 16 + fi;           // k
+            const int col = c0 + wc + ni * 16 + fi;      // k
             if (row < N && col < K)
               wsp[d.transpose_w ? (long)col * N + row
                                 : (long)row * K + col] = a[r];
           }
         }
-      if (tid < TBN && tid < N) {
-        float db = 0.f;
+    }
+    if (tid < TBN && tid < N) {   // db_i = col-sums of dy
+      float db = 0.f;
 #pragma unroll
-        for (int m = 0; m < TBM; ++m) {
-          union { float f; unsigned u; } v;
-          v.u = ((unsigned)s_dyT[tid][m]) << 16;
-          db += v.f;
-        }
-        arena[(long)s * s_stride + d.b_off[i] + g * (long)N + tid] = db;
+      for (int m = 0; m < TBM; ++m) {
+        union { float f; unsigned u; } v;
+        v.u = ((unsigned)s_dyT[tid][m]) << 16;
+        db += v.f;
       }
+      arena[(long)s * s_stride + d.b_off[i] + g * (long)N + tid] = db;
     }
     if (i == 0) return;
     // ---- dy_{i-1} = (dy_i @ W_i) * relu-mask(a_i) --------------------
@@ -1093,7 +1093,7 @@ static void mlp_narrow_bwd_bf16(torch::Tensor dy_last,
     auto wc = ws[i].contiguous(); auto ac = acts[i].contiguous();
     keep.push_back(wc); keep.push_back(ac);
     const long N = wc.numel() / (G * K);
-    TORCH_CHECK(N <= 64 && (i == 0 || K <= 64) && Kin <= 64);
+    TORCH_CHECK(N <= 64 && (i == 0 || K <= 64));
     d.w[i] = (const u16*)wc.data_ptr();
     d.acts[i] = (const u16*)ac.data_ptr();
     d.w_off[i] = w_offs[i];
