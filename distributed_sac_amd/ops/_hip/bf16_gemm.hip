@@ -587,6 +587,8 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow(
 struct NarrowBwdDesc {
   const u16* w[6];      // kernel-layout weights [G,N_i,K_i] (bf16 mirrors)
   const u16* acts[6];   // acts[0]=chain input x; acts[i]=post-act out of i-1
+  long a_gs[6];         // per-group stride of acts[i] (0 = shared across
+                        // groups, e.g. the mixture's raw-state input)
   long w_off[6];        // flat-gradient offsets for dW_i
   long b_off[6];        //   "                 "     db_i
   int N[6];             // layer widths; K_i = N_{i-1} (K_0 = Kin)
@@ -629,7 +631,7 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
   for (int i = d.L - 1; i >= 0; --i) {
     const int N = d.N[i];
     const int K = i == 0 ? Kin : d.N[i - 1];
-    const u16* ag = d.acts[i] + g * (long)M * K;
+    const u16* ag = d.acts[i] + g * d.a_gs[i];
     float* wsp = arena + (long)s * s_stride + d.w_off[i] + g * (long)N * K;
     // ---- dW_i = dy^T @ a_i, looping a_i's columns in 64-wide tiles
     // (layer 0 of the trunk/mlp_context chains has K = 768) -----------
@@ -1096,6 +1098,8 @@ static void mlp_narrow_bwd_bf16(torch::Tensor dy_last,
     TORCH_CHECK(N <= 64 && (i == 0 || K <= 64));
     d.w[i] = (const u16*)wc.data_ptr();
     d.acts[i] = (const u16*)ac.data_ptr();
+    d.a_gs[i] = ac.dim() == 3 ? M * K : 0;   // 2-D act = shared across G
+    TORCH_CHECK(ac.numel() == (ac.dim() == 3 ? G : 1) * M * K);
     d.w_off[i] = w_offs[i];
     d.b_off[i] = b_offs[i];
     d.N[i] = (int)N;
