@@ -48,6 +48,8 @@ class Trainer:
                  logger: Optional[MetricLogger] = None):
         self.cfg = cfg
         self.device = torch.device(device)
+        if self.device.type != "cuda":
+            torch.set_num_threads(1)   # tiny ops thrash the intra-op pool
         self.engine = create_engine(cfg, device)
         num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
         self.replay = ShardedReplay(cfg.buffer_size, num_tasks,

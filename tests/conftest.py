@@ -19,3 +19,9 @@ def pytest_collection_modifyitems(config, items):
     for item in items:
         if "gpu" in item.keywords:
             item.add_marker(skip)
+
+
+# tiny-op workloads thrash torch's default intra-op pool (measured 100x
+# slowdown on an 8-core runner); one thread keeps the CPU suite fast
+import torch as _torch  # noqa: E402
+_torch.set_num_threads(1)
