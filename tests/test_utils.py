@@ -118,3 +118,18 @@ def test_cfg_value_parity_with_reference():
     for n in names:
         check(load(os.path.join(here, n)),
               load(os.path.join(ref_root, n)))
+
+
+def test_tfevents_reader_roundtrip(tmp_path):
+    """Writer -> reader roundtrip: scalars come back tag/step/value-exact
+    (the reader hand-decodes the protobuf; no tensorboard installed)."""
+    from distributed_sac_amd.analysis.read_tfevents import read_scalars_dir
+
+    w = TFEventWriter(str(tmp_path))
+    w.add_scalar("a/loss", 1.25, 3)
+    w.add_scalar("a/loss", 0.75, 7)
+    w.add_scalar("b/rate", 42.0, 1)
+    w.close()
+    series = read_scalars_dir(str(tmp_path))
+    assert series["a/loss"] == [(3, 1.25), (7, 0.75)]
+    assert series["b/rate"] == [(1, 42.0)]
