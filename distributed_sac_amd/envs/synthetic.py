@@ -35,16 +35,23 @@ class SyntheticEnv:
 
     def __init__(self, state_dim: int, action_dim: int,
                  max_episode_steps: int = 500, seed: int = 0,
-                 success_info: bool = False, action_bound=(-1.0, 1.0)):
+                 success_info: bool = False, action_bound=(-1.0, 1.0),
+                 dynamics_seed: Optional[int] = None):
         self.state_dim = state_dim
         self.action_dim = action_dim
         self.max_episode_steps = max_episode_steps
         self.success_info = success_info
         self._rng = np.random.default_rng(seed)
-        # fixed, well-conditioned dynamics: s' = 0.98*A s + B a + noise
-        self.A = self._rng.normal(0, 1.0 / np.sqrt(state_dim),
-                                  (state_dim, state_dim)).astype(np.float32)
-        self.B = self._rng.normal(0, 0.3, (state_dim, action_dim)).astype(np.float32)
+        # fixed, well-conditioned dynamics: s' = 0.98*A s + B a + noise.
+        # dynamics_seed decouples the SYSTEM from the episode randomness:
+        # every worker playing "task t" must face the SAME dynamics (a
+        # per-worker seed only varies resets/noise) — otherwise one task's
+        # replay shard mixes transitions from different systems.
+        rng_dyn = np.random.default_rng(
+            seed if dynamics_seed is None else dynamics_seed)
+        self.A = rng_dyn.normal(0, 1.0 / np.sqrt(state_dim),
+                                (state_dim, state_dim)).astype(np.float32)
+        self.B = rng_dyn.normal(0, 0.3, (state_dim, action_dim)).astype(np.float32)
         self.action_space = BoxSpace(action_bound[0], action_bound[1],
                                      (action_dim,), self._rng)
         self.observation_space = BoxSpace(-np.inf, np.inf, (state_dim,), self._rng)
@@ -90,12 +97,14 @@ class SyntheticEnv:
 
 
 def make_synthetic(preset: str = "lunarlander", seed: int = 0,
-                   max_episode_steps: Optional[int] = None) -> SyntheticEnv:
+                   max_episode_steps: Optional[int] = None,
+                   dynamics_seed: Optional[int] = None) -> SyntheticEnv:
     if preset == "lunarlander":
-        return SyntheticEnv(8, 2, max_episode_steps or 500, seed)
+        return SyntheticEnv(8, 2, max_episode_steps or 500, seed,
+                            dynamics_seed=dynamics_seed)
     if preset == "metaworld":
         return SyntheticEnv(39, 4, max_episode_steps or 500, seed,
-                            success_info=True)
+                            success_info=True, dynamics_seed=dynamics_seed)
     raise ValueError(preset)
 
 
@@ -111,16 +120,19 @@ class BatchedSyntheticEnv:
 
     def __init__(self, n_envs: int, state_dim: int, action_dim: int,
                  max_episode_steps: int = 500, seed: int = 0,
-                 success_info: bool = False, action_bound=(-1.0, 1.0)):
+                 success_info: bool = False, action_bound=(-1.0, 1.0),
+                 dynamics_seed: Optional[int] = None):
         self.n_envs = n_envs
         self.state_dim = state_dim
         self.action_dim = action_dim
         self.max_episode_steps = max_episode_steps
         self.success_info = success_info
         self._rng = np.random.default_rng(seed)
-        self.A = self._rng.normal(0, 1.0 / np.sqrt(state_dim),
-                                  (state_dim, state_dim)).astype(np.float32)
-        self.B = self._rng.normal(0, 0.3, (state_dim, action_dim)).astype(np.float32)
+        rng_dyn = np.random.default_rng(
+            seed if dynamics_seed is None else dynamics_seed)
+        self.A = rng_dyn.normal(0, 1.0 / np.sqrt(state_dim),
+                                (state_dim, state_dim)).astype(np.float32)
+        self.B = rng_dyn.normal(0, 0.3, (state_dim, action_dim)).astype(np.float32)
         self.low, self.high = action_bound
         self.states = np.zeros((n_envs, state_dim), dtype=np.float32)
         self.t = np.zeros(n_envs, dtype=np.int64)

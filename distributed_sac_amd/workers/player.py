@@ -121,6 +121,7 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
     last_iteration = -1
     flat = torch.zeros(snapshot.buf.numel())
     episodes_seen = {t: 0 for t in set(task_idx_list)}
+    eval_done_at = {t: 0 for t in set(task_idx_list)}
     n_chunks = 0
     while stop_event is None or not stop_event.is_set():
         # pull parameters when a newer snapshot exists (reference
@@ -164,8 +165,9 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
                     log_queue.put(("reward", player_id, t,
                                    rollout.total_steps_per_task[t],
                                    float(np.mean(recent))))
-                if eval_every_episodes and done_eps and \
-                        done_eps % eval_every_episodes == 0:
+                if eval_every_episodes and \
+                        done_eps >= eval_done_at[t] + eval_every_episodes:
+                    eval_done_at[t] = done_eps
                     rate = rollout.evaluate_success_rate(t, episodes=10)
                     log_queue.put(("success_rate", player_id, t,
                                    rollout.total_steps_per_task[t], rate))

@@ -171,7 +171,8 @@ class FastSyntheticRollout:
                 max_episode_steps=cfg.max_episode_time,
                 seed=seed * 9173 + t,
                 success_info=cfg.variant in ("vsac", "mtsac", "care"),
-                action_bound=tuple(cfg.action_bound))
+                action_bound=tuple(cfg.action_bound),
+                dynamics_seed=t)   # same task = same system
             self.envs[t] = env
             self.obs[t] = env.reset_all()
         self.warmup_remaining = {t: cfg.random_step for t in self.tasks}

@@ -25,7 +25,8 @@ def default_env_fn(cfg: SACConfig, task_idx: int, seed: int):
     return SyntheticEnv(cfg.state_dim, cfg.action_dim,
                         max_episode_steps=cfg.max_episode_time, seed=seed,
                         success_info=cfg.variant in ("vsac", "mtsac", "care"),
-                        action_bound=tuple(cfg.action_bound))
+                        action_bound=tuple(cfg.action_bound),
+                        dynamics_seed=task_idx)   # same task = same system
 
 
 class _CAREEnginePolicy:

@@ -154,3 +154,18 @@ def test_trainer_determinism_same_seed():
     a3, c3 = run(8)
     assert torch.equal(a1, a2) and torch.equal(c1, c2)
     assert not torch.equal(a1, a3)
+
+
+def test_same_task_same_dynamics_across_workers():
+    """Two workers playing the same task face the SAME dynamical system
+    (dynamics keyed by task, episode randomness by worker seed) — one
+    task's replay shard must not mix different systems."""
+    from distributed_sac_amd.workers.trainer import default_env_fn
+    cfg = tiny_cfg("mtsac")
+    e1 = default_env_fn(cfg, 3, seed=100)
+    e2 = default_env_fn(cfg, 3, seed=999)
+    e_other = default_env_fn(cfg, 4, seed=100)
+    assert np.array_equal(e1.A, e2.A) and np.array_equal(e1.B, e2.B)
+    assert not np.array_equal(e1.A, e_other.A)
+    # but resets differ by worker seed
+    assert not np.array_equal(e1.reset(), e2.reset())
