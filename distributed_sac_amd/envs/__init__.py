@@ -50,15 +50,34 @@ def make_env(name: str, seed: int = 0, allow_synthetic: bool = True):
     raise ValueError(f"unknown env {name}")
 
 
+class _MT1EpisodeTaskEnv:  # pragma: no cover - needs metaworld
+    """Reference MT1 players call ``set_task(random.choice(train_tasks))``
+    before every episode (MT1_Distributed_VSAC/src/player.py:93-100) —
+    re-randomize the goal on each reset so rollouts see the full task
+    distribution, not one frozen goal."""
+
+    def __init__(self, env, train_tasks, seed: int):
+        import random
+        self._env = env
+        self._tasks = list(train_tasks)
+        self._rng = random.Random(seed)
+
+    def reset(self):
+        self._env.set_task(self._rng.choice(self._tasks))
+        return self._env.reset()
+
+    def __getattr__(self, name):
+        return getattr(self._env, name)
+
+
 def _make_metaworld(name: str, seed: int):  # pragma: no cover - needs metaworld
     """Meta-World construction mirroring reference main.py (MT1: env from
-    train_classes + set_task over train_tasks; MT10: dict of classes)."""
+    train_classes + per-episode set_task over train_tasks; MT10: dict of
+    classes)."""
     import metaworld
-    import random
     if name == "mt10":
         return metaworld.MT10()
     task_name = name.split("-", 1)[1] if "-" in name else "pick-place-v2"
     mt1 = metaworld.MT1(task_name)
     env = mt1.train_classes[task_name]()
-    env.set_task(random.choice(mt1.train_tasks))
-    return env
+    return _MT1EpisodeTaskEnv(env, mt1.train_tasks, seed)
