@@ -140,3 +140,23 @@ def test_load_actual_reference_checkpoint():
     out = evaluate_checkpoint(cfg, REF_LL_CKPT, default_env_fn, episodes=2)
     assert np.isfinite(out["mean_reward"])
 
+
+
+def test_shipped_trained_checkpoint_evaluates():
+    """The trained synthetic-VSAC artifact under saved_models/ loads and
+    achieves near-optimal deterministic reward (r11 learning run)."""
+    import json
+
+    from distributed_sac_amd.config import Decoder, SACConfig
+    from distributed_sac_amd.workers.player import evaluate_checkpoint
+    from distributed_sac_amd.workers.trainer import default_env_fn
+
+    root = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "saved_models", "MT1_VSAC_synthetic_tiny")
+    cfg = SACConfig.from_dict(
+        json.load(open(os.path.join(root, "cfg.json")), cls=Decoder), "vsac")
+    out = evaluate_checkpoint(cfg, os.path.join(root, "checkpoint_553159.tar"),
+                              default_env_fn, episodes=3, seed=123)
+    assert out["update_iteration"] == 553159
+    assert out["mean_reward"] > -5.0          # random policy is ~-70
+    assert out["success_rate"] == 1.0
