@@ -160,3 +160,24 @@ def test_shipped_trained_checkpoint_evaluates():
     assert out["update_iteration"] == 553159
     assert out["mean_reward"] > -5.0          # random policy is ~-70
     assert out["success_rate"] == 1.0
+
+
+def test_shipped_mtsac_checkpoint_evaluates():
+    """The trained MT10-MTSAC artifact (reference 3x400 architecture, MT
+    .tar schema) evaluates with success 1.0 on an arbitrary task."""
+    import json
+
+    from distributed_sac_amd.config import Decoder, SACConfig
+    from distributed_sac_amd.workers.player import evaluate_checkpoint
+    from distributed_sac_amd.workers.trainer import default_env_fn
+
+    root = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "saved_models", "MT10_MTSAC_synthetic")
+    cfg = SACConfig.from_dict(
+        json.load(open(os.path.join(root, "cfg.json")), cls=Decoder), "mtsac")
+    out = evaluate_checkpoint(cfg, os.path.join(root, "checkpoint_70718.tar"),
+                              default_env_fn, task_idx=7, episodes=2,
+                              seed=321)
+    assert out["update_iteration"] == 70718
+    assert out["mean_reward"] > -10.0
+    assert out["success_rate"] == 1.0
