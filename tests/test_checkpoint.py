@@ -181,3 +181,23 @@ def test_shipped_mtsac_checkpoint_evaluates():
     assert out["update_iteration"] == 70718
     assert out["mean_reward"] > -10.0
     assert out["success_rate"] == 1.0
+
+
+def test_shipped_ll_checkpoint_evaluates():
+    """The trained LunarLander-SAC artifact (reference LL .tar schema,
+    update_delay-thinned counters) evaluates near-optimally."""
+    import json
+
+    from distributed_sac_amd.config import Decoder, SACConfig
+    from distributed_sac_amd.workers.player import evaluate_checkpoint
+    from distributed_sac_amd.workers.trainer import default_env_fn
+
+    root = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "saved_models", "LunarLander_synthetic")
+    cfg = SACConfig.from_dict(
+        json.load(open(os.path.join(root, "cfg.json")), cls=Decoder), "sac")
+    out = evaluate_checkpoint(
+        cfg, os.path.join(root, "checkpoint_292428.tar"),
+        default_env_fn, episodes=2, seed=77)
+    assert out["update_iteration"] == 292428
+    assert out["mean_reward"] > -5.0   # random policy ~-70
