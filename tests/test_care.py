@@ -506,3 +506,23 @@ def test_mlp_narrow_fused_gpu():
         assert torch.equal(a1, acts[0])
         assert torch.equal(a2, acts[1])
         assert torch.equal(y, h)
+
+
+def test_care_evaluate_checkpoint_roundtrip(tmp_path):
+    """CARE eval mode: a saved engine checkpoint loads into the
+    rollout-side CAREPolicy bundle (actor + context encoder) and runs
+    deterministic episodes (reference main.py is_train=False branch)."""
+    from distributed_sac_amd.checkpoint import save_checkpoint
+    from distributed_sac_amd.workers.player import evaluate_checkpoint
+    from distributed_sac_amd.workers.trainer import default_env_fn
+
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path, modified=True)
+    cfg.max_episode_time = 40
+    engine = CAREEngine(cfg, "cpu")
+    path = save_checkpoint(engine, str(tmp_path / "ck"), update_iteration=7)
+    out = evaluate_checkpoint(cfg, path, default_env_fn, task_idx=1,
+                              episodes=2, seed=5)
+    assert out["update_iteration"] == 7
+    assert out["episodes"] == 2
+    assert np.isfinite(out["mean_reward"])
