@@ -78,6 +78,9 @@ class CAREEngine(SACEngine):
         # CARE uses weighted losses iff modified CARE
         # (reference learner.update_SAC use_weighted_loss=use_modified_care)
         self.use_weighted_loss = self.use_modified_care
+        # strict-parity degenerate-broadcast mode (docs/PARITY.md)
+        self.degenerate_w = \
+            getattr(cfg, "weighted_loss_mode", "corrected") == "reference"
 
     def _critic_layer_pairs(self, target: bool = False):
         c = self.target_critic if target else self.local_critic
@@ -194,6 +197,11 @@ class CAREEngine(SACEngine):
 
     def update_tensors(self, batch: Dict[str, torch.Tensor]):
         if self._use_fused(batch["states"]):
+            if self.degenerate_w and self.use_weighted_loss:
+                raise RuntimeError(
+                    "weighted_loss_mode='reference' is a CPU strict-parity "
+                    "mode; the fused HIP kernels implement corrected "
+                    "weighting (docs/PARITY.md)")
             return self._update_tensors_fused(batch)
         states, actions = batch["states"], batch["actions"]
         rewards, next_states, dones = (batch["rewards"], batch["next_states"],
@@ -214,7 +222,8 @@ class CAREEngine(SACEngine):
         l1, l2 = self.local_critic.cal_loss(
             states, z_context, actions, y,
             use_weighted_loss=self.use_weighted_loss, num_tasks=T,
-            alphas=self.log_alpha.exp().detach())
+            alphas=self.log_alpha.exp().detach(),
+            degenerate=self.degenerate_w)
         q_loss = l1 + l2
         q_loss.backward(retain_graph=True)
         if self.ddp is not None:
@@ -228,7 +237,8 @@ class CAREEngine(SACEngine):
         policy_loss = self.actor.cal_loss(
             lp, q_min, alpha, use_weighted_loss=self.use_weighted_loss,
             mtobss=states, num_tasks=T,
-            alphas=self.log_alpha.exp().detach())
+            alphas=self.log_alpha.exp().detach(),
+            degenerate=self.degenerate_w)
         policy_loss.backward()
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.actor_group.flat_grad)

@@ -49,6 +49,12 @@ class SACEngine:
         self.tau = cfg.tau
         self.reward_scale = cfg.reward_scale
         self.use_weighted_loss = cfg.use_weighted_loss and cfg.variant == "mtsac"
+        # strict-parity mode reproducing the reference's degenerate
+        # (B,)x(B,1)->(B,B) weighted-loss broadcast (docs/PARITY.md);
+        # supported on the torch path only — the fused HIP kernels
+        # implement the corrected per-sample weighting
+        self.degenerate_w = \
+            getattr(cfg, "weighted_loss_mode", "corrected") == "reference"
         self.num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
         self.update_iteration = 0
         self.total_step = 0
@@ -338,6 +344,11 @@ class SACEngine:
         critic step → actor step → alpha step → Polyak.
         """
         if self._use_fused(batch["states"]):
+            if self.degenerate_w and self.use_weighted_loss:
+                raise RuntimeError(
+                    "weighted_loss_mode='reference' (degenerate broadcast) "
+                    "is a CPU strict-parity mode; the fused HIP kernels "
+                    "implement corrected weighting (docs/PARITY.md)")
             return self._update_tensors_fused(batch)
         states = batch["states"]
         actions = batch["actions"]
@@ -363,7 +374,8 @@ class SACEngine:
             l1, l2 = self.local_critic.cal_loss(
                 states, actions, y,
                 use_weighted_loss=self.use_weighted_loss,
-                alphas=self.log_alpha.exp().detach())
+                alphas=self.log_alpha.exp().detach(),
+                degenerate=self.degenerate_w)
             q_loss = l1 + l2
         q_loss.backward()
         if self.ddp is not None:
@@ -378,7 +390,8 @@ class SACEngine:
             policy_loss = self.actor.cal_loss(
                 log_probs, q_min, alpha,
                 use_weighted_loss=self.use_weighted_loss, mtobss=states,
-                alphas=self.log_alpha.exp().detach())
+                alphas=self.log_alpha.exp().detach(),
+                degenerate=self.degenerate_w)
         else:
             policy_loss = self.actor.cal_loss(log_probs, q_min, alpha)
         policy_loss.backward()

@@ -89,6 +89,11 @@ class SACConfig:
     max_episode_time: int = 500
 
     use_weighted_loss: bool = False
+    # "corrected" (default): true per-sample task weighting.
+    # "reference": reproduce the reference's degenerate (B,)x(B,1)->(B,B)
+    # broadcast where the weights cancel (= mean(loss)/B) — strict-parity
+    # mode, CPU torch path only.  See docs/PARITY.md "weighted loss".
+    weighted_loss_mode: str = "corrected"
     # CARE-only block (reference cfg "encoder"):
     encoder: Optional[Dict[str, Any]] = None
     use_modified_care: bool = False
@@ -143,6 +148,8 @@ class SACConfig:
                 setattr(c, key, float(cfg[key]))
         if "use_weighted_loss" in cfg:
             c.use_weighted_loss = bool(cfg["use_weighted_loss"])
+        if "weighted_loss_mode" in cfg:
+            c.weighted_loss_mode = str(cfg["weighted_loss_mode"])
         if "use_modified_care" in cfg:
             c.use_modified_care = bool(cfg["use_modified_care"])
         if "encoder" in cfg:

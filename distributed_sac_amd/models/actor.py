@@ -82,13 +82,18 @@ class Actor(nn.Module):
         return a
 
     def cal_loss(self, log_probs, Q_min, alpha, use_weighted_loss=False,
-                 mtobss=None, alphas=None):
+                 mtobss=None, alphas=None, degenerate=False):
         """Policy loss -(Qmin - alpha*logpi), optionally task-weighted
-        (reference MT10…MTSAC/src/model.py:80-116)."""
+        (reference MT10…MTSAC/src/model.py:80-116).
+
+        DELIBERATE DEVIATION (docs/PARITY.md "weighted loss"): the
+        reference's (B,)x(B,1) broadcast makes the weights cancel
+        (= mean(loss)/B); default here is true per-sample weighting,
+        ``degenerate=True`` reproduces the reference numerics."""
         loss = -(Q_min - alpha * log_probs)
         if use_weighted_loss and alphas is not None and mtobss is not None:
             w = Fops.task_weights(mtobss[:, -self.num_tasks:], alphas)
-            loss = w.unsqueeze(-1) * loss
+            loss = (w * loss) if degenerate else w.unsqueeze(-1) * loss
         return loss.mean()
 
 

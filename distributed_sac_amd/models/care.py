@@ -63,12 +63,15 @@ class CAREActor(nn.Module):
         return a
 
     def cal_loss(self, log_probs, Q_min, alpha, use_weighted_loss=False,
-                 mtobss=None, num_tasks=None, alphas=None):
+                 mtobss=None, num_tasks=None, alphas=None,
+                 degenerate=False):
+        # degenerate=True reproduces the reference's (B,)x(B,1)->(B,B)
+        # broadcast where the weights cancel (docs/PARITY.md)
         loss = -(Q_min - alpha * log_probs)
         if use_weighted_loss and num_tasks and alphas is not None \
                 and mtobss is not None:
             w = Fops.task_weights(mtobss[:, -num_tasks:], alphas)
-            loss = w.unsqueeze(-1) * loss
+            loss = (w * loss) if degenerate else w.unsqueeze(-1) * loss
         return loss.mean()
 
 
@@ -101,11 +104,13 @@ class CARECritic(nn.Module):
 
     def cal_loss(self, mtobss, z_context, action, td_target_values,
                  detach_z_encs=False, use_weighted_loss=False,
-                 num_tasks=None, alphas=None):
+                 num_tasks=None, alphas=None, degenerate=False):
         q1, q2 = self.forward(mtobss, z_context, action, detach_z_encs)
         l1 = (td_target_values - q1) ** 2
         l2 = (td_target_values - q2) ** 2
         if use_weighted_loss and num_tasks and alphas is not None:
-            w = Fops.task_weights(mtobss[:, -num_tasks:], alphas).unsqueeze(-1)
+            w = Fops.task_weights(mtobss[:, -num_tasks:], alphas)
+            if not degenerate:  # corrected default (docs/PARITY.md)
+                w = w.unsqueeze(-1)
             l1, l2 = w * l1, w * l2
         return l1.mean(), l2.mean()

@@ -39,15 +39,28 @@ class Critic(nn.Module):
         return self.Q_function_1(x), self.Q_function_2(x)
 
     def cal_loss(self, mtobss, action, td_target_values,
-                 use_weighted_loss: bool = False, alphas: Optional[torch.Tensor] = None):
+                 use_weighted_loss: bool = False,
+                 alphas: Optional[torch.Tensor] = None,
+                 degenerate: bool = False):
         """Twin MSE vs TD target, optionally task-weighted (reference
-        MT10…MTSAC/src/model.py:157-196)."""
+        MT10…MTSAC/src/model.py:157-196).
+
+        DELIBERATE DEVIATION (docs/PARITY.md "weighted loss"): the
+        reference multiplies weights (B,) by loss (B,1), broadcasting to
+        (B,B), so its mean equals mean(loss)/B and the weights cancel —
+        it effectively trains UNWEIGHTED at 1/B scale.  Default here is
+        TRUE per-sample weighting (``w.unsqueeze(-1) * loss``); pass
+        ``degenerate=True`` (cfg ``weighted_loss_mode="reference"``) to
+        reproduce the reference's actual numerics for strict parity."""
         q1, q2 = self.forward(mtobss, action)
         l1 = (td_target_values - q1) ** 2
         l2 = (td_target_values - q2) ** 2
         if use_weighted_loss and alphas is not None:
-            w = Fops.task_weights(mtobss[:, -self.num_tasks:], alphas).unsqueeze(-1)
-            l1, l2 = w * l1, w * l2
+            w = Fops.task_weights(mtobss[:, -self.num_tasks:], alphas)
+            if degenerate:  # reference (B,)*(B,1)->(B,B) broadcast
+                l1, l2 = w * l1, w * l2
+            else:
+                l1, l2 = w.unsqueeze(-1) * l1, w.unsqueeze(-1) * l2
         return l1.mean(), l2.mean()
 
 

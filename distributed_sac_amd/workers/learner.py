@@ -185,23 +185,41 @@ class Learner:
         if self._pub_stream is None:
             self._pub_stream = torch.cuda.Stream(self.device)
             self._pub_event = torch.cuda.Event()
+            self._pub_stage_event = torch.cuda.Event()
+            self._pub_stage = torch.empty_like(flat.reshape(-1))
             self._pub_pinned = torch.empty(flat.numel(), pin_memory=True)
         # phase 1: flush the previous copy if it finished
         if self._pub_iteration and self._pub_event.query():
             self.snapshot.publish(self._pub_pinned, self._pub_iteration)
             self._pub_iteration = 0
-        # phase 2: start a fresh async copy (skip if one is in flight)
+        # phase 2: start a fresh async copy (skip if one is in flight).
+        # Tear-safety: flat_data is snapshotted by a D2D copy into a
+        # device staging buffer first, and the MAIN stream waits on that
+        # tiny copy (µs) before it may enqueue the next Adam step that
+        # would overwrite flat_data; the slow D2H then reads only the
+        # immutable staging buffer, so the pinned snapshot can never mix
+        # parameters from two adjacent updates (ADVICE.md round-1 finding).
         if self._pub_iteration == 0:
-            self._pub_stream.wait_stream(
-                torch.cuda.current_stream(self.device))
+            main = torch.cuda.current_stream(self.device)
+            self._pub_stream.wait_stream(main)
             with torch.cuda.stream(self._pub_stream):
-                self._pub_pinned.copy_(flat.reshape(-1), non_blocking=True)
+                self._pub_stage.copy_(flat.reshape(-1), non_blocking=True)
+                self._pub_stage_event.record(self._pub_stream)
+                self._pub_pinned.copy_(self._pub_stage, non_blocking=True)
             self._pub_event.record(self._pub_stream)
+            main.wait_event(self._pub_stage_event)
             self._pub_iteration = self.iteration_counter
 
     def ready(self) -> bool:
-        return len(self.replay) * max(1, self.replay.num_tasks) >= \
-            self.cfg.start_memory_len and len(self.replay) > 0
+        """Reference start gate: train only once the MIN per-task shard
+        holds ``start_memory_len`` transitions (MT10_Distributed_MTSAC/src/
+        learner.py:354-358 with replay_buffers.__len__ = min over shards).
+        ``len(self.replay)`` is already the min-shard count; capping by the
+        shard capacity avoids a deadlock when a test-sized buffer is
+        smaller than the gate.  Matches Trainer.ready."""
+        return len(self.replay) > 0 and \
+            len(self.replay) >= min(self.cfg.start_memory_len,
+                                    self.replay.shards[0].capacity)
 
     def _ensure_graph(self) -> None:
         if self.use_graph and not self._graph_ready:

@@ -119,6 +119,7 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
     rollout = VecRollout(cfg, envs, task_idx_list, actor, device="cpu",
                          seed=seed)
     last_iteration = -1
+    dropped_blocks = 0
     flat = torch.zeros(snapshot.buf.numel())
     episodes_seen = {t: 0 for t in set(task_idx_list)}
     eval_done_at = {t: 0 for t in set(task_idx_list)}
@@ -137,18 +138,30 @@ def run_player(player_id: int, cfg: SACConfig, env_fn: Callable,
             if not blk["states"].shape[0]:
                 continue
             if ring is not None:
-                ok = ring.push(t, torch.from_numpy(blk["states"]),
-                               torch.from_numpy(blk["actions"]),
-                               torch.from_numpy(blk["rewards"].reshape(-1, 1)),
-                               torch.from_numpy(blk["next_states"]),
-                               torch.from_numpy(blk["dones"].reshape(-1, 1)))
+                tensors = (torch.from_numpy(blk["states"]),
+                           torch.from_numpy(blk["actions"]),
+                           torch.from_numpy(blk["rewards"].reshape(-1, 1)),
+                           torch.from_numpy(blk["next_states"]),
+                           torch.from_numpy(blk["dones"].reshape(-1, 1)))
+                # bounded backoff instead of fire-and-forget: the reference
+                # (Redis list) never dropped data; we retry ~100 ms before
+                # conceding, and LOG the drop player-side (the shared ring
+                # header also counts it for the learner)
+                ok = ring.push(t, *tensors)
+                attempts = 0
+                while not ok and attempts < 20:
+                    if stop_event is not None and stop_event.is_set():
+                        break
+                    time.sleep(0.005)  # learner behind
+                    ok = ring.push(t, *tensors)
+                    attempts += 1
                 if not ok:
-                    time.sleep(0.005)  # learner behind; retry once
-                    ring.push(t, torch.from_numpy(blk["states"]),
-                              torch.from_numpy(blk["actions"]),
-                              torch.from_numpy(blk["rewards"].reshape(-1, 1)),
-                              torch.from_numpy(blk["next_states"]),
-                              torch.from_numpy(blk["dones"].reshape(-1, 1)))
+                    dropped_blocks += 1
+                    if dropped_blocks <= 10 or dropped_blocks % 100 == 0:
+                        print(f"[player {player_id}] ring full after "
+                              f"{attempts} retries — dropped block "
+                              f"(task {t}, {blk['states'].shape[0]} steps, "
+                              f"{dropped_blocks} total drops)", flush=True)
             else:
                 try:
                     sample_queue.put((player_id, t, blk), timeout=5.0)

@@ -91,6 +91,51 @@ def test_analysis_plot_utils(tmp_path):
     assert os.path.exists(out)
 
 
+def test_learner_ready_min_shard_semantics():
+    """Reference start gate (MT10…MTSAC/src/learner.py:354-358 +
+    replay_buffers.__len__ = min over shards): training may begin only
+    once the MIN per-task shard holds start_memory_len transitions.
+    Round-1 bug: `min_shard * num_tasks >= start_memory_len` started MT10
+    training ~10x early; this pins the corrected semantics and its
+    agreement with Trainer.ready."""
+    import queue
+    from distributed_sac_amd.workers.learner import Learner
+    from distributed_sac_amd.workers.param_server import ParamSnapshot
+    from distributed_sac_amd.workers.trainer import Trainer
+
+    cfg = tiny_cfg("mtsac")
+    cfg.start_memory_len = 100
+    lr = Learner(cfg, "cpu", ParamSnapshot(8), queue.Queue())
+    T = cfg.num_tasks
+
+    def fill(replay, task, n):
+        replay.append_numpy(
+            states=np.zeros((n, cfg.mtobs_dim), dtype=np.float32),
+            actions=np.zeros((n, cfg.action_dim), dtype=np.float32),
+            rewards=np.zeros(n, dtype=np.float32),
+            next_states=np.zeros((n, cfg.mtobs_dim), dtype=np.float32),
+            dones=np.zeros(n, dtype=np.float32), task_idx=task)
+
+    assert not lr.ready()
+    # total count >= start_memory_len spread over shards: NOT ready
+    for t in range(T):
+        fill(lr.replay, t, 100 // T + 1)
+    assert len(lr.replay) * T >= cfg.start_memory_len
+    assert not lr.ready()
+    # every shard except one at the gate: still not ready
+    for t in range(T - 1):
+        fill(lr.replay, t, 100)
+    assert not lr.ready()
+    fill(lr.replay, T - 1, 100)
+    assert lr.ready()
+
+    # agreement with the single-process Trainer gate
+    tr = Trainer(cfg, device="cpu", seed=0)
+    for t in range(T):
+        fill(tr.replay, t, 100)
+    assert tr.ready()
+
+
 def test_heartbeat_watchdog():
     import time
     import torch as th
