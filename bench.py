@@ -71,6 +71,10 @@ def parse_args():
                         "rings -> learner (single rank)")
     p.add_argument("--async-seconds", type=float, default=30.0)
     p.add_argument("--players", type=int, default=4)
+    p.add_argument("--force-ddp", action="store_true",
+                   help="initialize the process group + run the DP code "
+                        "path even at world_size 1 (single-GPU RCCL "
+                        "validation / launch-gap profiling)")
     return p.parse_args()
 
 
@@ -143,20 +147,32 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
             torch.cuda.synchronize()
 
     graphed = False
+    dp_graphed = False
     # RCCL collectives are NOT capturable on this stack (the capture probe
     # aborts the process via the NCCL watchdog, not a catchable error) —
-    # DP ranks run the eager manual-backward path instead.
-    if device.startswith("cuda") and not args.no_graph and ddp is None:
-        try:
-            engine.capture(replay, cfg.batch_size)
-            graphed = True
-        except Exception as e:  # pragma: no cover
-            print(f"[bench] hipGraph capture failed, eager fallback: {e!r}",
-                  file=sys.stderr)
+    # DP ranks use the SEGMENTED capture (three hipGraphs with the two
+    # eager all-reduces between replays, engine.capture_dp).
+    if device.startswith("cuda") and not args.no_graph:
+        if ddp is None:
+            try:
+                engine.capture(replay, cfg.batch_size)
+                graphed = True
+            except Exception as e:  # pragma: no cover
+                print(f"[bench] hipGraph capture failed, eager fallback: "
+                      f"{e!r}", file=sys.stderr)
+        elif getattr(engine, "_bf16", False):
+            try:
+                engine.capture_dp(replay, cfg.batch_size)
+                dp_graphed = True
+            except Exception as e:  # pragma: no cover
+                print(f"[bench] segmented DP capture failed, eager "
+                      f"fallback: {e!r}", file=sys.stderr)
 
     def one_step():
         if graphed:
             engine.graphed_update()
+        elif dp_graphed:
+            engine.dp_graphed_update()
         else:
             engine.update_tensors(replay.sample(cfg.batch_size,
                                                 graph_safe=True))
@@ -165,6 +181,10 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
     for _ in range(args.warmup):
         one_step()
 
+    if rank == 0 and (graphed or dp_graphed):
+        print(f"[bench {label}] capture: "
+              f"{'full hipGraph' if graphed else 'segmented DP hipGraphs'}",
+              file=sys.stderr)
     windows = []
     while True:
         if ddp is not None:
@@ -279,7 +299,9 @@ def main():
         print(json.dumps(result))
         return
 
-    ddp = DataParallelGroup(device=torch.device(device)) if world > 1 else None
+    ddp = (DataParallelGroup(device=torch.device(device),
+                             force=args.force_ddp)
+           if (world > 1 or args.force_ddp) else None)
     engine, replay = build_engine_and_replay(cfg, device, precision, rank,
                                              ddp=ddp)
 

@@ -516,7 +516,29 @@ class SACEngine:
         autograd bookkeeping, no AccumulateGrad adds, no zero_grad except
         the tiny atomically-accumulated alpha grad).  Numerically identical
         to the autograd bf16 path (same kernels, same order) — verified by
-        the manual-vs-autograd GPU test."""
+        the manual-vs-autograd GPU test.
+
+        Structured as three SEGMENTS split at the two DP all-reduce
+        boundaries so the data-parallel path can hipGraph-capture each
+        segment and run the (non-capturable) RCCL collectives eagerly
+        between replays — see :meth:`capture_dp`:
+          seg1: forwards + TD target + critic backward  -> critic grad
+          seg2: critic Adam + actor/alpha forwards+backwards -> aa grads
+          seg3: actor+alpha fused Adam + Polyak
+        """
+        self._manual_seg1(batch)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self._manual_seg2()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self._aa_arena)
+        return self._manual_seg3()
+
+    @torch.no_grad()
+    def _manual_seg1(self, batch):
+        """Segment 1: batched actor forward + squash, TD target, critic
+        loss forward + backward.  Ends with the critic flat gradient ready
+        for all-reduce."""
         from ..ops import native
         ext = native()
         states = batch["states"]
@@ -529,8 +551,6 @@ class SACEngine:
         B = states.shape[0]
         A = self.cfg.action_dim
         la_det = self.log_alpha.detach()
-        nl_a = len(self._actor_ws_bf16)
-        nl_c = len(self._twin_local_bf16)
 
         # ---- batched actor forward + squash --------------------------
         x_cat = torch.cat([next_states, states], dim=0)
@@ -544,7 +564,6 @@ class SACEngine:
         a_cat, lp_cat, tanh_u, ls_cat = ext.squashed_gaussian_fwd(
             mu, lsr, eps, float(self.actor.k))
         na, nlp = a_cat[:B], lp_cat[:B]
-        sa, lp, ls = a_cat[B:], lp_cat[B:], ls_cat[B:]
 
         # ---- TD target ------------------------------------------------
         xt = torch.cat([next_states, na], dim=-1)
@@ -569,11 +588,34 @@ class SACEngine:
         self._mlp_bwd_arena(ext, dy, acts_c, self._twin_local_bf16,
                             wsg, bsg, fg_c, arena_c, S_c, ch_c, G=2)
         ext.reduce_arena(arena_c, fg_c, S_c)
-        if self.ddp is not None:
-            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self._dp_st = {
+            "states": states, "sa": a_cat[B:], "lp": lp_cat[B:],
+            "lsr": lsr, "ls_cat": ls_cat, "eps": eps, "tanh_u": tanh_u,
+            "acts_a": acts_a, "ws_f32": ws_f32, "bs_f32": bs_f32,
+            "la_det": la_det, "closs": closs, "B": B,
+        }
+
+    @torch.no_grad()
+    def _manual_seg2(self):
+        """Segment 2: critic Adam (mirror refreshed in-kernel), actor-side
+        twin forward on the post-step critic, actor/alpha losses and full
+        manual backward.  Ends with the fused actor+alpha gradient arena
+        ready for all-reduce."""
+        from ..ops import native
+        ext = native()
+        st = self._dp_st
         self.critic_optimizer.step()   # adam kernel also refreshes mirror
 
-        # ---- actor/alpha loss + manual backward ----------------------
+        states, sa, lp = st["states"], st["sa"], st["lp"]
+        B = st["B"]
+        T = self.num_tasks
+        use_w = self.use_weighted_loss
+        la_det = st["la_det"]
+        ls = st["ls_cat"][B:]
+        nl_a = len(self._actor_ws_bf16)
+        nl_c = len(self._twin_local_bf16)
+        acts_a = st["acts_a"]
+
         xa = torch.cat([states, sa], dim=-1)
         aq1, aq2, acts_f = self._twin_fwd_manual(
             xa, self._twin_local_bf16, self._twin_local[1])
@@ -594,11 +636,11 @@ class SACEngine:
                                      1 if nl_c > 1 else 0, 2, 1)
         dsa = dxa[:, states.shape[1]:].float()
         dhead = ext.squashed_gaussian_bwd2(
-            dsa, dlp, lsr[B:], ls_cat[B:], eps[B:], tanh_u[B:],
-            float(self.actor.k))
+            dsa, dlp, st["lsr"][B:], st["ls_cat"][B:], st["eps"][B:],
+            st["tanh_u"][B:], float(self.actor.k))
         dy = dhead
-        wag = [w.grad for w in ws_f32]
-        bag = [b.grad for b in bs_f32]
+        wag = [w.grad for w in st["ws_f32"]]
+        bag = [b.grad for b in st["bs_f32"]]
         fg_a = self.actor_group.flat_grad
         arena_a, S_a, ch_a = self._dw_arena("actor",
                                             self.actor_group.numel, B)
@@ -614,14 +656,18 @@ class SACEngine:
                 dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
                                             yout, act, 1, 1)
         ext.reduce_arena(arena_a, fg_a, S_a)
-        if self.ddp is not None:
-            self.ddp.allreduce_grad_(self._aa_arena)
+        st["al"] = al
+
+    @torch.no_grad()
+    def _manual_seg3(self):
+        """Segment 3: fused actor+alpha Adam, Polyak target update."""
+        st = self._dp_st
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
         self.alpha = self.log_alpha.exp().detach()
-
         flat_polyak_(self.target_group, self.critic_group, self.tau,
                      mirror=getattr(self, "_target_bf16", None))
+        closs, al = st["closs"], st["al"]
         return {
             "critic_loss": closs[0] + closs[1],
             "actor_loss": al[0],
@@ -752,6 +798,57 @@ class SACEngine:
         self._graph.replay()
         self.update_iteration += 1
         return self._graph_metrics
+
+    # ------------------------------------------------------------------
+    # Data-parallel segmented capture: RCCL collectives are NOT
+    # hipGraph-capturable on this ROCm stack (tools/probe_rccl_graph.py —
+    # the capture probe aborts the process via the NCCL watchdog), so a DP
+    # update is captured as THREE graphs split at the all-reduce
+    # boundaries; replay runs g1 -> eager AR(critic grad) -> g2 -> eager
+    # AR(actor+alpha arena) -> g3.  This recovers hipGraph launch
+    # efficiency for ~all compute kernels while keeping the collectives
+    # eager.  All graphs share one memory pool so cross-segment
+    # intermediates (activations, squash state) keep stable addresses.
+    # ------------------------------------------------------------------
+    def capture_dp(self, replay, batch_size: int, warmup_iters: int = 3):
+        assert self.device.type == "cuda", "capture needs a GPU"
+        assert getattr(self, "_bf16", False), \
+            "segmented DP capture runs the bf16 manual-backward path"
+        torch.cuda.synchronize(self.device)
+        side = torch.cuda.Stream(self.device)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                # full eager DP update (including real all-reduces: warms
+                # RCCL's communicator and keeps rank launch order aligned)
+                self._update_tensors_manual(
+                    replay.sample(batch_size, graph_safe=True))
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        torch.cuda.synchronize(self.device)
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            self._manual_seg1(replay.sample(batch_size, graph_safe=True))
+        g2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g2, pool=g1.pool()):
+            self._manual_seg2()
+        g3 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g3, pool=g1.pool()):
+            self._dp_graph_metrics = self._manual_seg3()
+        self._dp_graphs = (g1, g2, g3)
+        return self._dp_graphs
+
+    def dp_graphed_update(self) -> Dict[str, torch.Tensor]:
+        """Replay the segmented DP update with eager RCCL between."""
+        g1, g2, g3 = self._dp_graphs
+        g1.replay()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        g2.replay()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self._aa_arena)
+        g3.replay()
+        self.update_iteration += 1
+        return self._dp_graph_metrics
 
     # ------------------------------------------------------------------
     # Checkpointing — reference .tar schema (learner.save_checkpoint).

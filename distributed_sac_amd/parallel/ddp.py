@@ -30,25 +30,39 @@ class DataParallelGroup:
 
     def __init__(self, backend: Optional[str] = None,
                  device: Optional[torch.device] = None,
-                 timeout_s: int = 300):
+                 timeout_s: int = 300, force: bool = False):
+        """``force=True`` initializes the process group even at
+        world_size==1 — exercises the full RCCL init + collective call
+        path on a single GPU (validation/profiling of the eager DP
+        plumbing without an 8-GPU node)."""
         self.world_size = int(os.environ.get("WORLD_SIZE", "1"))
         self.rank = int(os.environ.get("RANK", "0"))
         self.local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-        self.enabled = self.world_size > 1
+        self.enabled = self.world_size > 1 or force
         self.device = device
         if self.enabled and not dist.is_initialized():
             backend = backend or ("nccl" if torch.cuda.is_available() else "gloo")
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29531")
+            os.environ.setdefault("RANK", "0")
+            os.environ.setdefault("WORLD_SIZE", "1")
             dist.init_process_group(backend=backend,
                                     timeout=timedelta(seconds=timeout_s))
         self.backend = dist.get_backend() if self.enabled else None
+        # RCCL supports in-collective averaging (one launch, no separate
+        # div kernel); gloo does not
+        self._avg = self.enabled and self.backend == "nccl"
 
     @torch.no_grad()
     def allreduce_grad_(self, flat_grad: torch.Tensor) -> None:
         """In-place gradient averaging: one fused message per group."""
         if not self.enabled:
             return
-        dist.all_reduce(flat_grad, op=dist.ReduceOp.SUM)
-        flat_grad.div_(self.world_size)
+        if self._avg:
+            dist.all_reduce(flat_grad, op=dist.ReduceOp.AVG)
+        else:
+            dist.all_reduce(flat_grad, op=dist.ReduceOp.SUM)
+            flat_grad.div_(self.world_size)
 
     def barrier(self) -> None:
         if self.enabled:

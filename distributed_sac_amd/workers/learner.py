@@ -70,9 +70,16 @@ class Learner:
         self.save_period = save_period
         self.update_delay = (update_delay if update_delay is not None
                              else max(1, cfg.update_delay))
+        dp_active = ddp is not None and ddp.enabled
         self.use_graph = (use_graph and self.device.type == "cuda"
-                          and (ddp is None or not ddp.enabled))
+                          and not dp_active)
+        # DP learners use the segmented capture (three hipGraphs, eager
+        # RCCL all-reduces between replays — SACEngine.capture_dp)
+        self.use_dp_graph = (use_graph and self.device.type == "cuda"
+                             and dp_active
+                             and getattr(self.engine, "_bf16", False))
         self._graph_ready = False
+        self._dp_graph_ready = False
         self.iteration_counter = 0   # reference update_iteration (thinned)
         self.grad_steps = 0
         self.update_timer = StepTimer()
@@ -233,6 +240,17 @@ class Learner:
             except Exception as e:  # pragma: no cover
                 self.logger.print(f"hipGraph capture failed ({e!r}); eager")
                 self.use_graph = False
+        if self.use_dp_graph and not self._dp_graph_ready:
+            try:  # pragma: no cover - requires multi-GPU RCCL
+                self.logger.print("capturing segmented DP graphs...")
+                t0 = time.perf_counter()
+                self.engine.capture_dp(self.replay, self.cfg.batch_size)
+                self._dp_graph_ready = True
+                self.logger.print(
+                    f"DP graphs captured in {time.perf_counter() - t0:.1f}s")
+            except Exception as e:  # pragma: no cover
+                self.logger.print(f"DP capture failed ({e!r}); eager")
+                self.use_dp_graph = False
 
     def train_step(self) -> Optional[Dict[str, float]]:
         """One reference loop iteration: counter++, update when thinned
@@ -241,11 +259,13 @@ class Learner:
         metrics = None
         if self.iteration_counter % self.update_delay == 0:
             t0 = time.perf_counter()
-            if self.use_graph:
+            if self.use_graph or self.use_dp_graph:
                 self._ensure_graph()
             if self._graph_ready:
                 out = self.engine.graphed_update()
                 metrics_t = out
+            elif self._dp_graph_ready:  # pragma: no cover - multi-GPU
+                metrics_t = self.engine.dp_graphed_update()
             else:
                 metrics_t = self.engine.update_tensors(
                     self.replay.sample(self.cfg.batch_size,
