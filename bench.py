@@ -211,7 +211,9 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
         print(f"[bench {label}] {len(windows)} windows x {args.steps} steps: "
               f"median {med * 1e3:.2f} ms (min {lo * 1e3:.2f}, "
               f"max {hi * 1e3:.2f})", file=sys.stderr)
-    return med, len(windows), graphed
+    mode = ("graph" if graphed
+            else "dp-seg-graph" if dp_graphed else "eager")
+    return med, len(windows), mode
 
 
 def run_async_bench(cfg, args, device, precision):
@@ -309,8 +311,8 @@ def main():
     if rank == 0 and not args.skip_rollout_probe:
         env_rate = rollout_probe(cfg, "cpu")
 
-    med, n_windows, graphed = measure(engine, replay, cfg, args, device,
-                                      ddp, rank)
+    med, n_windows, graph_mode = measure(engine, replay, cfg, args, device,
+                                         ddp, rank)
     steps_per_sec = args.steps / med
     value = n_gpus * steps_per_sec
 
@@ -359,7 +361,10 @@ def main():
                 "batch_per_gpu": cfg.batch_size,
                 "update": "full SAC step: TD target + critic/actor/alpha "
                           "fwd+bwd+fusedAdam + Polyak",
-                "hipgraph": graphed,
+                # "graph" = one captured hipGraph; "dp-seg-graph" = three
+                # captured segments with eager RCCL all-reduces between
+                "hipgraph": graph_mode != "eager",
+                "launch_mode": graph_mode,
             },
         }
         print(json.dumps(result))

@@ -658,13 +658,16 @@ class CAREEngine(SACEngine):
                   + [a[:, B:].contiguous() for a in se_saved["acts_m"][1:]])
         import os as _os
         use_fused_se_bwd = (
-            _os.environ.get("DSAC_NARROW_BWD", "0") == "1"
+            _os.environ.get("DSAC_NARROW_BWD", "1") == "1"
             and info["mix_narrow"] and info["trunk_narrow"]
             and info["ctx_narrow"])
         if use_fused_se_bwd:
-            # round-2 staging (docs/ROADMAP_r2.md item 3): whole-chain
-            # fused backward, one launch per chain, partials per 64-row
-            # block; OFF by default until GPU-validated.
+            # DEFAULT ON since round 2: GPU-validated (exact parameter
+            # agreement vs the per-layer arena path, +3.8% eager rate —
+            # gpurun_out r2 call1 / tests/test_gpu_kernels.py narrow gate
+            # test); whole-chain fused backward, one launch per chain,
+            # partials per 64-row block.  DSAC_NARROW_BWD=0 restores the
+            # per-layer path.
             base = fg_c.data_ptr()
 
             def offs(ps):

@@ -605,3 +605,91 @@ def test_manual_backward_matches_autograd_bf16():
                                 e_man.local_critic.named_parameters()):
         assert torch.allclose(pa, pm, atol=1e-3, rtol=2e-2), n
     assert torch.allclose(e_aut.log_alpha, e_man.log_alpha, atol=1e-3)
+
+
+def test_narrow_bwd_gate_equivalence(tmp_path):
+    """The fused narrow-chain SE backward (default since round 2) must
+    produce the same parameters as the per-layer arena path — only the
+    partial-sum association differs (measured exact on MI355X)."""
+    import os
+    from distributed_sac_amd.algo import CAREEngine
+    from tests.test_care import care_batch, care_cfg
+
+    def make(gate):
+        os.environ["DSAC_NARROW_BWD"] = gate
+        torch.manual_seed(0)
+        cfg = care_cfg(tmp_path, modified=True)
+        return CAREEngine(cfg, "cuda:0", precision="bf16"), cfg
+
+    try:
+        e_off, cfg = make("0")
+        e_on, _ = make("1")
+        e_on.load_checkpoint_state(e_off.checkpoint_state())
+        e_on.hard_copy_targets()
+        e_off.hard_copy_targets()
+        for step in range(3):
+            batch = {k: v.cuda() for k, v in
+                     care_batch(cfg, seed=step).items()}
+            eps = [torch.randn(cfg.batch_size, cfg.action_dim,
+                               device="cuda") for _ in range(2)]
+            os.environ["DSAC_NARROW_BWD"] = "0"
+            e_off._eps_queue = [e.clone() for e in eps]
+            e_off.update({k: v.clone() for k, v in batch.items()})
+            os.environ["DSAC_NARROW_BWD"] = "1"
+            e_on._eps_queue = [e.clone() for e in eps]
+            e_on.update({k: v.clone() for k, v in batch.items()})
+    finally:
+        os.environ.pop("DSAC_NARROW_BWD", None)
+    for (n, pa), (_, pb) in zip(e_off.local_critic.named_parameters(),
+                                e_on.local_critic.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-5), n
+    for (n, pa), (_, pb) in zip(e_off.actor.named_parameters(),
+                                e_on.actor.named_parameters()):
+        assert torch.allclose(pa, pb, atol=1e-5), n
+    assert torch.allclose(e_off.log_alpha, e_on.log_alpha, atol=1e-6)
+
+
+def test_dp_segmented_graphs_world1():
+    """Single-GPU validation of the data-parallel launch path: a forced
+    world-1 RCCL group (AVG all-reduce == identity), segmented capture
+    (3 hipGraphs with eager collectives between), 30 replays stable."""
+    import torch.distributed as dist
+    from distributed_sac_amd.algo import SACEngine
+    from distributed_sac_amd.parallel import DataParallelGroup
+    from distributed_sac_amd.replay import ShardedReplay
+    from tests.test_engine import small_cfg
+
+    cfg = small_cfg("mtsac")
+    ddp = DataParallelGroup(device=torch.device("cuda:0"), force=True)
+    assert ddp.enabled and ddp.world_size == 1
+    try:
+        # world-1 AVG all-reduce must be an identity (replica semantics)
+        t = torch.randn(1000, device="cuda")
+        t0 = t.clone()
+        ddp.allreduce_grad_(t)
+        assert torch.equal(t, t0)
+
+        torch.manual_seed(0)
+        engine = SACEngine(cfg, "cuda:0", precision="bf16")
+        engine.attach_ddp(ddp)
+        replay = ShardedReplay(4096, cfg.num_tasks, cfg.mtobs_dim,
+                               cfg.action_dim, device="cuda:0", seed=3)
+        for tsk in range(cfg.num_tasks):
+            n = 256
+            s = torch.randn(n, cfg.mtobs_dim, device="cuda")
+            s[:, -cfg.num_tasks:] = 0
+            s[:, -cfg.num_tasks + tsk] = 1
+            replay.shards[tsk].append(
+                s, torch.rand(n, cfg.action_dim, device="cuda") * 2 - 1,
+                torch.randn(n, 1, device="cuda"), s.clone(),
+                torch.zeros(n, 1, device="cuda"))
+        engine.capture_dp(replay, cfg.batch_size)
+        losses = []
+        for _ in range(30):
+            m = engine.dp_graphed_update()
+            losses.append(float(m["critic_loss"]))
+        assert all(v == v for v in losses)          # no NaN
+        assert losses[-1] < losses[0] * 1.5          # not diverging
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
