@@ -10,7 +10,7 @@ cd /tmp
   for cfg in "mtsac:" "care:--config care" "dp1:--force-ddp"; do
     name="${cfg%%:*}"; extra="${cfg#*:}"
     echo "== rocprof $name =="
-    timeout 400 rocprofv3 --kernel-trace --stats -d /tmp/prof_$name -- \
+    timeout 400 rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof_$name -- \
       python /root/repo/bench.py --steps 200 --warmup 30 \
         --min-timed-seconds 0.1 --max-windows 2 \
         --skip-rollout-probe --skip-fp32-probe $extra \
