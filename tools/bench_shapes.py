@@ -82,8 +82,8 @@ def main():
     for lbl, M, N, K, G, sumg in dx_shapes:
         dy = bf(M, N) if G == 1 else bf(G, M, N)
         w = bf(G, N, K) if G > 1 else bf(N, K)
-        yo = bf(M, K) if (G == 1 or sumg) else bf(G, M, K)
-        us = timeit(lambda: ext.linear_bwd_dx_bf16(dy, w, yo, 1, G, sumg))
+        yo = dy  # relu mask tensor is dy-shaped; act=0 leaves it unused
+        us = timeit(lambda: ext.linear_bwd_dx_bf16(dy, w, yo, 0, G, sumg))
         flops = 2.0 * M * K * N * G
         print(f"  {lbl} M{M:5d} N{N:4d} K{K:4d} G{G}  {us:7.2f} us  "
               f"{flops/us/1e6:7.1f} GFLOP/s  x{counts_dx[lbl]}")
