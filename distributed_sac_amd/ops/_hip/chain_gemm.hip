@@ -1,0 +1,286 @@
+// chain_gemm.hip — fused wide MLP-chain kernels for the SAC hot path
+// (gfx950).  Round-2 addition.
+//
+// Motivation (profiles/r16_round2_call1-3.md): every per-layer GEMM
+// launch costs ~10-17 us at these tiny shapes REGARDLESS of FLOPs — the
+// whole family is launch/latency-bound (e.g. k_bf16_fwd on the twin head
+// M1280 N1 G2 runs 0.1 GFLOP/s).  An MLP forward is row-parallel, so one
+// workgroup can own a 16-row batch slice and run the ENTIRE chain:
+// activations stay in LDS, weights stream from L2 (every WG reads the
+// same ~320 KB per 400x400 layer), and the 4-6 launches per chain
+// collapse to ONE.  The first layer reads the raw fp32 inputs (optionally
+// two tensors column-concatenated, e.g. [states || actions]) and converts
+// to bf16 in the loader — this also eliminates the separate torch::cat +
+// bf16-cast launches (VERDICT round-1 item 5).
+//
+// Geometry: 256 threads = 4 waves; each wave computes 16x16 MFMA output
+// tiles (v_mfma_f32_16x16x32_bf16), striding the layer's N dimension in
+// PAIRS of tiles (two independent accumulators interleave the MFMA and
+// global-B-fragment load streams).  A-fragments come from the LDS
+// activation buffer: row stride 520 u16 = 260 dwords = 4 banks mod 64, so
+// the 16-lane b128 read groups touch all 64 banks exactly once —
+// conflict-free without padding tricks (see cdna_hip_programming.md §2).
+//
+// grid (ceil(M/16), 1, G).  G>1 = twin critics: shared x, per-group
+// weights/biases/outputs (same convention as k_bf16_fwd).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+namespace chain {
+
+using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+using u16 = unsigned short;
+
+static constexpr int TM = 16;     // batch rows per workgroup
+static constexpr int CMAX = 512;  // max layer width / input width
+static constexpr int CPAD = 520;  // LDS row stride (u16): 260 dwords ≡ 4 mod 64
+
+#define CHAIN_CHECK_BF16(t) \
+  TORCH_CHECK((t).is_cuda() && (t).scalar_type() == torch::kBFloat16, \
+              #t " must be a bf16 HIP tensor")
+#define CHAIN_CHECK_F32(t) \
+  TORCH_CHECK((t).is_cuda() && (t).scalar_type() == torch::kFloat32, \
+              #t " must be a fp32 HIP tensor")
+
+static inline hipStream_t cur_stream3() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+__device__ __forceinline__ u16 f32_to_bf16_rne3(float f) {
+  union { float f; unsigned u; } v{f};
+  unsigned u = v.u;
+  u += 0x7FFFu + ((u >> 16) & 1u);
+  return (u16)(u >> 16);
+}
+
+struct ChainFwdDesc {
+  const u16* w[6];    // [G, N_l, K_l] bf16 (k-minor)
+  const float* b[6];  // [G, N_l]
+  u16* acts[6];       // post-act output of layer l (null = don't save)
+  int N[6];
+  int act[6];         // 1 = ReLU
+  int L;
+};
+
+// B-operand fragment: 8 consecutive bf16 of one weight row from GLOBAL.
+__device__ __forceinline__ bf16x8 load_bfrag(const u16* __restrict__ w,
+                                             int row, int N, int K,
+                                             int k8) {
+  if (row < N) {
+    const long base = (long)row * K + k8;
+    if (k8 + 8 <= K && (base & 7) == 0)
+      return *(const bf16x8*)&w[base];
+    bf16x8 v{};
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      if (k8 + j < K) v[j] = ((const __bf16*)w)[base + j];
+    return v;
+  }
+  return bf16x8{};
+}
+
+__global__ __launch_bounds__(256) void k_bf16_chain_fwd(
+    const void* __restrict__ x1, const void* __restrict__ x2,
+    int C1, int C2, int x_f32, u16* __restrict__ xsave,
+    ChainFwdDesc d, void* __restrict__ y, int M, int out_f32) {
+  __shared__ u16 sa[2][TM][CPAD];
+  const int g = blockIdx.z;
+  const int m0 = blockIdx.x * TM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int fi = lane & 15, fk = lane >> 4;
+  const int rowlim = (M - m0 < TM ? M - m0 : TM);
+  const int K0 = C1 + C2;
+
+  // zero both activation buffers (pads must read 0 in the MFMA A-frags)
+  {
+    unsigned* p = (unsigned*)sa;
+    for (int i = tid; i < 2 * TM * CPAD / 2; i += 256) p[i] = 0u;
+  }
+  __syncthreads();
+
+  // ---- input tile -> sa[0] (converted to bf16; optional col-concat) ----
+  {
+    const int r = tid >> 4;      // 16 threads per row
+    const int lc = tid & 15;
+    if (r < rowlim) {
+      const long row = m0 + r;
+      for (int c = lc; c < K0; c += 16) {
+        float v;
+        if (c < C1)
+          v = x_f32 ? ((const float*)x1)[row * C1 + c]
+                    : (float)((const __bf16*)x1)[row * C1 + c];
+        else
+          v = x_f32 ? ((const float*)x2)[row * C2 + (c - C1)]
+                    : (float)((const __bf16*)x2)[row * C2 + (c - C1)];
+        const u16 h = f32_to_bf16_rne3(v);
+        sa[0][r][c] = h;
+        if (xsave != nullptr && g == 0) xsave[row * K0 + c] = h;
+      }
+    }
+  }
+  __syncthreads();
+
+  int cur = 0;
+  int prevw0 = K0, prevw1 = CMAX;  // written widths of buffers 0/1
+  int K = K0;
+  for (int li = 0; li < d.L; ++li) {
+    const int N = d.N[li];
+    const bool last = li == d.L - 1;
+    const int relu = d.act[li];
+    const u16* wg = d.w[li] + (long)g * N * K;
+    const float* bb = d.b[li] + (long)g * N;
+    u16* ag = (!last && d.acts[li] != nullptr)
+                  ? d.acts[li] + (long)g * M * N : nullptr;
+    float* yf = (float*)y + (long)g * M * N;
+    u16* yh = (u16*)y + (long)g * M * N;
+    const int ntiles = (N + 15) >> 4;
+    const int nb = cur ^ 1;
+
+    for (int t0 = wid * 2; t0 < ntiles; t0 += 8) {
+      const bool has1 = (t0 + 1) < ntiles;
+      f32x4 acc0{}, acc1{};
+      const int row0 = t0 * 16 + fi;
+      const int row1 = row0 + 16;
+      for (int k = 0; k < K; k += 32) {
+        const int k8 = k + fk * 8;
+        const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
+        const bf16x8 b0 = load_bfrag(wg, row0, N, K, k8);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
+        if (has1) {
+          const bf16x8 b1 = load_bfrag(wg, row1, N, K, k8);
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+        }
+      }
+      // epilogue: C/D map col = lane&15, row = (lane>>4)*4 + r
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        if (half == 1 && !has1) break;
+        const f32x4 a = half ? acc1 : acc0;
+        const int col = (half ? t0 + 1 : t0) * 16 + fi;
+        if (col < N) {
+          const float bias = bb[col];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = fk * 4 + r;
+            float v = a[r] + bias;
+            if (relu) v = fmaxf(v, 0.f);
+            if (last) {
+              if (row < rowlim) {
+                if (out_f32) yf[(long)(m0 + row) * N + col] = v;
+                else yh[(long)(m0 + row) * N + col] = f32_to_bf16_rne3(v);
+              }
+            } else {
+              const u16 h = f32_to_bf16_rne3(v);
+              sa[nb][row][col] = h;
+              if (ag != nullptr && row < rowlim)
+                ag[(long)(m0 + row) * N + col] = h;
+            }
+          }
+        }
+      }
+    }
+    if (!last) {
+      // zero the stale tail of the next buffer: cols [N, prevw(nb))
+      const int pw = nb ? prevw1 : prevw0;
+      for (int i = tid; i < TM * (pw > N ? pw - N : 0); i += 256) {
+        const int r = i / (pw - N), c = N + i % (pw - N);
+        sa[nb][r][c] = 0;
+      }
+      if (nb) prevw1 = N; else prevw0 = N;
+      __syncthreads();
+      cur = nb;
+      K = N;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host binding: y(, acts...) = chain_fwd(x1[, x2], ws, bs, acts_save,
+//                                        act_last, G, out_f32)
+// ws[i]: bf16 [G*N_i, K_i] (or [N,K] / [G,N,K]); bs[i]: f32 [G*N_i].
+// Inputs fp32 or bf16, 2-D [M, C]; x2 may be an empty tensor.
+// Returns [y, x_bf16, act_0, ..., act_{L-2}] — x_bf16 is the converted
+// (concatenated) input, so the backward consumes the same activation list
+// the per-layer path produced.
+// ---------------------------------------------------------------------------
+static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
+    torch::Tensor x1, torch::Tensor x2, std::vector<torch::Tensor> ws,
+    std::vector<torch::Tensor> bs, long act_last, long G, long out_f32) {
+  const int L = (int)ws.size();
+  TORCH_CHECK(L >= 1 && L <= 6 && (int)bs.size() == L);
+  TORCH_CHECK(x1.is_cuda() && x1.dim() == 2);
+  const bool has2 = x2.numel() > 0;
+  const bool xf32 = x1.scalar_type() == torch::kFloat32;
+  TORCH_CHECK(x1.scalar_type() == torch::kFloat32
+              || x1.scalar_type() == torch::kBFloat16);
+  auto x1c = x1.contiguous();
+  torch::Tensor x2c = x2;
+  if (has2) {
+    TORCH_CHECK(x2.is_cuda() && x2.dim() == 2
+                && x2.scalar_type() == x1.scalar_type()
+                && x2.size(0) == x1.size(0));
+    x2c = x2.contiguous();
+  }
+  const long M = x1c.size(0);
+  const long C1 = x1c.size(1), C2 = has2 ? x2c.size(1) : 0;
+  const long K0 = C1 + C2;
+  TORCH_CHECK(K0 <= CMAX, "chain fwd: input width must be <= ", CMAX);
+
+  ChainFwdDesc d{};
+  d.L = L;
+  std::vector<torch::Tensor> keep, out;
+  auto bopts = x1c.options().dtype(torch::kBFloat16);
+  auto xsave = torch::empty({M, K0}, bopts);
+  out.resize(2);
+  out[1] = xsave;
+  long K = K0;
+  for (int i = 0; i < L; ++i) {
+    CHAIN_CHECK_BF16(ws[i]);
+    CHAIN_CHECK_F32(bs[i]);
+    auto wc = ws[i].contiguous();
+    auto bc = bs[i].contiguous();
+    keep.push_back(wc);
+    keep.push_back(bc);
+    const long N = wc.numel() / (G * K);
+    TORCH_CHECK(N * G * K == wc.numel(), "weight shape mismatch at layer ",
+                i, " (K=", K, ")");
+    TORCH_CHECK(N <= CMAX, "chain fwd: layer width must be <= ", CMAX);
+    d.w[i] = (const u16*)wc.data_ptr();
+    d.b[i] = bc.data_ptr<float>();
+    d.N[i] = (int)N;
+    d.act[i] = (i == L - 1) ? (int)act_last : 1;
+    d.acts[i] = nullptr;
+    if (i < L - 1) {
+      auto a = G == 1 ? torch::empty({M, N}, bopts)
+                      : torch::empty({G, M, N}, bopts);
+      d.acts[i] = (u16*)a.data_ptr();
+      out.push_back(a);
+    }
+    K = N;
+  }
+  auto yopts = x1c.options().dtype(out_f32 ? torch::kFloat32
+                                           : torch::kBFloat16);
+  auto y = G == 1 ? torch::empty({M, (long)d.N[L - 1]}, yopts)
+                  : torch::empty({G, M, (long)d.N[L - 1]}, yopts);
+  out[0] = y;
+  dim3 grid((M + TM - 1) / TM, 1, G);
+  hipLaunchKernelGGL(k_bf16_chain_fwd, grid, dim3(256), 0, cur_stream3(),
+                     x1c.data_ptr(), has2 ? x2c.data_ptr() : nullptr,
+                     (int)C1, (int)C2, (int)(xf32 ? 1 : 0),
+                     (u16*)xsave.data_ptr(), d, y.data_ptr(), (int)M,
+                     (int)out_f32);
+  return out;
+}
+
+}  // namespace chain
+
+void register_chain(pybind11::module_& m) {
+  m.def("mlp_chain_fwd_bf16", &chain::mlp_chain_fwd_bf16);
+}

@@ -1394,10 +1394,12 @@ static void polyak_(torch::Tensor t, torch::Tensor s, double tau,
 
 void register_shm_ring(pybind11::module_& m);
 void register_bf16(pybind11::module_& m);
+void register_chain(pybind11::module_& m);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   register_shm_ring(mod);
   register_bf16(mod);
+  register_chain(mod);
   mod.def("linear_act_fwd", &linear_act_fwd, "fused GEMM+bias+act forward");
   mod.def("linear_act_fwd_g", &linear_act_fwd_g, "grouped (twin) variant");
   mod.def("linear_bwd_dx", &linear_bwd_dx, "GEMM backward dX (fused mask)");

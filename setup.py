@@ -24,7 +24,8 @@ setup(
             name="distributed_sac_amd.ops._hip_ops",
             sources=["distributed_sac_amd/ops/_hip/dsac_kernels.hip",
                      "distributed_sac_amd/ops/_hip/shm_ring.cpp",
-                     "distributed_sac_amd/ops/_hip/bf16_gemm.hip"],
+                     "distributed_sac_amd/ops/_hip/bf16_gemm.hip",
+                     "distributed_sac_amd/ops/_hip/chain_gemm.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
