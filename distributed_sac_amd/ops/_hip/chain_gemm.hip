@@ -143,27 +143,77 @@ __global__ __launch_bounds__(256) void k_bf16_chain_fwd(
     const int ntiles = (N + 15) >> 4;
     const int nb = cur ^ 1;
 
-    for (int t0 = wid * 2; t0 < ntiles; t0 += 8) {
-      const bool has1 = (t0 + 1) < ntiles;
-      f32x4 acc0{}, acc1{};
-      const int row0 = t0 * 16 + fi;
-      const int row1 = row0 + 16;
-      for (int k = 0; k < K; k += 32) {
-        const int k8 = k + fk * 8;
-        const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
-        const bf16x8 b0 = load_bfrag(wg, row0, N, K, k8);
-        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
-        if (has1) {
-          const bf16x8 b1 = load_bfrag(wg, row1, N, K, k8);
+    // K-body where every lane's 8-run is in-range AND 16B-aligned:
+    // requires K % 8 == 0 for row alignment; the tail (K % 32) and
+    // odd-K layers (layer 0's 49/53-d inputs) take the guarded path.
+    const bool k_aligned = (K % 8) == 0;
+    const int kbody = k_aligned ? (K & ~31) : 0;
+
+    // QUAD-tile interleave: 4 independent accumulator chains per wave
+    // keep 4 B-fragment loads in flight per k-step with NO per-load
+    // branches (a per-element guard makes hipcc branch around each load
+    // and drain vmcnt — the round-1 guarded loop ran 0.74x per-layer).
+    for (int t0 = wid * 4; t0 < ntiles; t0 += 16) {
+      const int nq = (ntiles - t0 < 4) ? (ntiles - t0) : 4;
+      f32x4 acc0{}, acc1{}, acc2{}, acc3{};
+      const long r0 = (long)(t0 * 16 + fi) * K + fk * 8;
+      const long rs = (long)16 * K;
+      if (nq == 4 && (t0 + 4) * 16 <= N) {
+        // ---- fast path: all rows valid, unconditional 16B loads ----
+#pragma unroll 4
+        for (int k = 0; k < kbody; k += 32) {
+          const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k + fk * 8];
+          const bf16x8 b0 = *(const bf16x8*)&wg[r0 + k];
+          const bf16x8 b1 = *(const bf16x8*)&wg[r0 + rs + k];
+          const bf16x8 b2 = *(const bf16x8*)&wg[r0 + 2 * rs + k];
+          const bf16x8 b3 = *(const bf16x8*)&wg[r0 + 3 * rs + k];
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
           acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+          acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
+          acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
+        }
+        for (int k = kbody; k < K; k += 32) {  // masked tail
+          const int k8 = k + fk * 8;
+          const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
+          const int row = t0 * 16 + fi;
+          const bf16x8 b0 = load_bfrag(wg, row, N, K, k8);
+          const bf16x8 b1 = load_bfrag(wg, row + 16, N, K, k8);
+          const bf16x8 b2 = load_bfrag(wg, row + 32, N, K, k8);
+          const bf16x8 b3 = load_bfrag(wg, row + 48, N, K, k8);
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+          acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
+          acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
+        }
+      } else {
+        // ---- boundary tile-group / odd-K layer: guarded loads ----
+        const int row = t0 * 16 + fi;
+        for (int k = 0; k < K; k += 32) {
+          const int k8 = k + fk * 8;
+          const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
+          const bf16x8 b0 = load_bfrag(wg, row, N, K, k8);
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
+          if (nq > 1) {
+            const bf16x8 b1 = load_bfrag(wg, row + 16, N, K, k8);
+            acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+          }
+          if (nq > 2) {
+            const bf16x8 b2 = load_bfrag(wg, row + 32, N, K, k8);
+            acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
+          }
+          if (nq > 3) {
+            const bf16x8 b3 = load_bfrag(wg, row + 48, N, K, k8);
+            acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
+          }
         }
       }
       // epilogue: C/D map col = lane&15, row = (lane>>4)*4 + r
+      const f32x4 accs[4] = {acc0, acc1, acc2, acc3};
 #pragma unroll
-      for (int half = 0; half < 2; ++half) {
-        if (half == 1 && !has1) break;
-        const f32x4 a = half ? acc1 : acc0;
-        const int col = (half ? t0 + 1 : t0) * 16 + fi;
+      for (int q = 0; q < 4; ++q) {
+        if (q >= nq) break;
+        const f32x4 a = accs[q];
+        const int col = (t0 + q) * 16 + fi;
         if (col < N) {
           const float bias = bb[col];
 #pragma unroll
