@@ -85,7 +85,9 @@ __device__ __forceinline__ bf16x8 load_bfrag(const u16* __restrict__ w,
   return bf16x8{};
 }
 
-__global__ __launch_bounds__(256) void k_bf16_chain_fwd(
+static constexpr int NTHR = 512;   // 8 waves: 2/SIMD latency overlap
+
+__global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const void* __restrict__ x1, const void* __restrict__ x2,
     int C1, int C2, int x_f32, u16* __restrict__ xsave,
     ChainFwdDesc d, void* __restrict__ y, int M, int out_f32) {
@@ -101,17 +103,17 @@ __global__ __launch_bounds__(256) void k_bf16_chain_fwd(
   // zero both activation buffers (pads must read 0 in the MFMA A-frags)
   {
     unsigned* p = (unsigned*)sa;
-    for (int i = tid; i < 2 * TM * CPAD / 2; i += 256) p[i] = 0u;
+    for (int i = tid; i < 2 * TM * CPAD / 2; i += NTHR) p[i] = 0u;
   }
   __syncthreads();
 
   // ---- input tile -> sa[0] (converted to bf16; optional col-concat) ----
   {
-    const int r = tid >> 4;      // 16 threads per row
-    const int lc = tid & 15;
+    const int r = tid >> 5;      // 32 threads per row
+    const int lc = tid & 31;
     if (r < rowlim) {
       const long row = m0 + r;
-      for (int c = lc; c < K0; c += 16) {
+      for (int c = lc; c < K0; c += 32) {
         float v;
         if (c < C1)
           v = x_f32 ? ((const float*)x1)[row * C1 + c]
@@ -153,25 +155,48 @@ __global__ __launch_bounds__(256) void k_bf16_chain_fwd(
     // keep 4 B-fragment loads in flight per k-step with NO per-load
     // branches (a per-element guard makes hipcc branch around each load
     // and drain vmcnt — the round-1 guarded loop ran 0.74x per-layer).
-    for (int t0 = wid * 4; t0 < ntiles; t0 += 16) {
+    for (int t0 = wid * 4; t0 < ntiles; t0 += 4 * (NTHR / 64)) {
       const int nq = (ntiles - t0 < 4) ? (ntiles - t0) : 4;
       f32x4 acc0{}, acc1{}, acc2{}, acc3{};
       const long r0 = (long)(t0 * 16 + fi) * K + fk * 8;
       const long rs = (long)16 * K;
       if (nq == 4 && (t0 + 4) * 16 <= N) {
-        // ---- fast path: all rows valid, unconditional 16B loads ----
-#pragma unroll 4
-        for (int k = 0; k < kbody; k += 32) {
-          const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k + fk * 8];
-          const bf16x8 b0 = *(const bf16x8*)&wg[r0 + k];
-          const bf16x8 b1 = *(const bf16x8*)&wg[r0 + rs + k];
-          const bf16x8 b2 = *(const bf16x8*)&wg[r0 + 2 * rs + k];
-          const bf16x8 b3 = *(const bf16x8*)&wg[r0 + 3 * rs + k];
-          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
-          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
-          acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
-          acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
+        // ---- fast path: all rows valid, unconditional 16B loads,
+        // manual unroll-2 software pipeline: set P computes while set Q
+        // loads (and vice versa) so the 4 B-fragment loads of step k+1
+        // stay in flight across step k's MFMAs (hipcc emits counted
+        // vmcnt for its own loads when the consumer is a different
+        // register set — no cross-iteration register copies).
+#define LOADQ(v0, v1, v2, v3, kk)                                  \
+        v0 = *(const bf16x8*)&wg[r0 + (kk)];                       \
+        v1 = *(const bf16x8*)&wg[r0 + rs + (kk)];                  \
+        v2 = *(const bf16x8*)&wg[r0 + 2 * rs + (kk)];              \
+        v3 = *(const bf16x8*)&wg[r0 + 3 * rs + (kk)];
+#define MF4(v0, v1, v2, v3, kk)                                            \
+        {                                                                  \
+          const bf16x8 a_ = *(const bf16x8*)&sa[cur][fi][(kk) + fk * 8];   \
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v0, acc0, 0, 0, 0); \
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v1, acc1, 0, 0, 0); \
+          acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v2, acc2, 0, 0, 0); \
+          acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v3, acc3, 0, 0, 0); \
         }
+        if (kbody >= 32) {
+          bf16x8 p0, p1, p2, p3, q0, q1, q2, q3;
+          LOADQ(p0, p1, p2, p3, 0)
+          int k = 0;
+          while (k + 64 <= kbody) {
+            LOADQ(q0, q1, q2, q3, k + 32)
+            MF4(p0, p1, p2, p3, k)
+            if (k + 64 < kbody) {
+              LOADQ(p0, p1, p2, p3, k + 64)
+            }
+            MF4(q0, q1, q2, q3, k + 32)
+            k += 64;
+          }
+          if (k < kbody) MF4(p0, p1, p2, p3, k)
+        }
+#undef LOADQ
+#undef MF4
         for (int k = kbody; k < K; k += 32) {  // masked tail
           const int k8 = k + fk * 8;
           const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
@@ -239,7 +264,7 @@ __global__ __launch_bounds__(256) void k_bf16_chain_fwd(
     if (!last) {
       // zero the stale tail of the next buffer: cols [N, prevw(nb))
       const int pw = nb ? prevw1 : prevw0;
-      for (int i = tid; i < TM * (pw > N ? pw - N : 0); i += 256) {
+      for (int i = tid; i < TM * (pw > N ? pw - N : 0); i += NTHR) {
         const int r = i / (pw - N), c = N + i % (pw - N);
         sa[nb][r][c] = 0;
       }
@@ -321,7 +346,7 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
                   : torch::empty({G, M, (long)d.N[L - 1]}, yopts);
   out[0] = y;
   dim3 grid((M + TM - 1) / TM, 1, G);
-  hipLaunchKernelGGL(k_bf16_chain_fwd, grid, dim3(256), 0, cur_stream3(),
+  hipLaunchKernelGGL(k_bf16_chain_fwd, grid, dim3(chain::NTHR), 0, cur_stream3(),
                      x1c.data_ptr(), has2 ? x2c.data_ptr() : nullptr,
                      (int)C1, (int)C2, (int)(xf32 ? 1 : 0),
                      (u16*)xsave.data_ptr(), d, y.data_ptr(), (int)M,
