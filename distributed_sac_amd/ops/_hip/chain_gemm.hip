@@ -5,7 +5,7 @@
 // launch costs ~10-17 us at these tiny shapes REGARDLESS of FLOPs — the
 // whole family is launch/latency-bound (e.g. k_bf16_fwd on the twin head
 // M1280 N1 G2 runs 0.1 GFLOP/s).  An MLP forward is row-parallel, so one
-// workgroup can own a 16-row batch slice and run the ENTIRE chain:
+// workgroup can own a batch row-slice and run the ENTIRE chain:
 // activations stay in LDS, weights stream from L2 (every WG reads the
 // same ~320 KB per 400x400 layer), and the 4-6 launches per chain
 // collapse to ONE.  The first layer reads the raw fp32 inputs (optionally
@@ -13,15 +13,20 @@
 // to bf16 in the loader — this also eliminates the separate torch::cat +
 // bf16-cast launches (VERDICT round-1 item 5).
 //
-// Geometry: 256 threads = 4 waves; each wave computes 16x16 MFMA output
-// tiles (v_mfma_f32_16x16x32_bf16), striding the layer's N dimension in
-// PAIRS of tiles (two independent accumulators interleave the MFMA and
-// global-B-fragment load streams).  A-fragments come from the LDS
-// activation buffer: row stride 520 u16 = 260 dwords = 4 banks mod 64, so
-// the 16-lane b128 read groups touch all 64 banks exactly once —
+// Geometry: 512 threads = 8 waves (2/SIMD so loads of one wave overlap
+// MFMAs of the other); each wave computes 16x16 MFMA output tiles
+// (v_mfma_f32_16x16x32_bf16) in QUADS of column tiles x RM row-fragments
+// — 4*RM independent accumulator chains keep 4 B-fragment loads in
+// flight per k-step, and the k-loop is a manual unroll-2 software
+// pipeline (set P computes while set Q loads) so the loads span a full
+// MFMA step.  RM=2 (32 rows/WG) halves the total L2 weight re-read
+// (traffic = M/TM * layer bytes) at the cost of half the workgroups —
+// the host picks RM by M.  A-fragments come from the LDS activation
+// buffer: row stride 520 u16 = 260 dwords = 4 banks mod 64, so the
+// 16-lane b128 read groups touch all 64 banks exactly once —
 // conflict-free without padding tricks (see cdna_hip_programming.md §2).
 //
-// grid (ceil(M/16), 1, G).  G>1 = twin critics: shared x, per-group
+// grid (ceil(M/TM), 1, G).  G>1 = twin critics: shared x, per-group
 // weights/biases/outputs (same convention as k_bf16_fwd).
 
 #include <hip/hip_runtime.h>
@@ -37,9 +42,9 @@ using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using u16 = unsigned short;
 
-static constexpr int TM = 16;     // batch rows per workgroup
 static constexpr int CMAX = 512;  // max layer width / input width
 static constexpr int CPAD = 520;  // LDS row stride (u16): 260 dwords ≡ 4 mod 64
+static constexpr int NTHR = 512;  // 8 waves
 
 #define CHAIN_CHECK_BF16(t) \
   TORCH_CHECK((t).is_cuda() && (t).scalar_type() == torch::kBFloat16, \
@@ -85,34 +90,34 @@ __device__ __forceinline__ bf16x8 load_bfrag(const u16* __restrict__ w,
   return bf16x8{};
 }
 
-static constexpr int NTHR = 512;   // 8 waves: 2/SIMD latency overlap
-
+template <int RM>
 __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const void* __restrict__ x1, const void* __restrict__ x2,
     int C1, int C2, int x_f32, u16* __restrict__ xsave,
     ChainFwdDesc d, void* __restrict__ y, int M, int out_f32) {
-  __shared__ u16 sa[2][TM][CPAD];
+  constexpr int TMv = 16 * RM;
+  __shared__ u16 sa[2][TMv][CPAD];
   const int g = blockIdx.z;
-  const int m0 = blockIdx.x * TM;
+  const int m0 = blockIdx.x * TMv;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
   const int fi = lane & 15, fk = lane >> 4;
-  const int rowlim = (M - m0 < TM ? M - m0 : TM);
+  const int rowlim = (M - m0 < TMv ? M - m0 : TMv);
   const int K0 = C1 + C2;
 
   // zero both activation buffers (pads must read 0 in the MFMA A-frags)
   {
     unsigned* p = (unsigned*)sa;
-    for (int i = tid; i < 2 * TM * CPAD / 2; i += NTHR) p[i] = 0u;
+    for (int i = tid; i < 2 * TMv * CPAD / 2; i += NTHR) p[i] = 0u;
   }
   __syncthreads();
 
   // ---- input tile -> sa[0] (converted to bf16; optional col-concat) ----
   {
-    const int r = tid >> 5;      // 32 threads per row
+    const int r = tid >> 5;      // 32 threads per row covers 16 rows/pass
     const int lc = tid & 31;
-    if (r < rowlim) {
-      const long row = m0 + r;
+    for (int rr = r; rr < rowlim; rr += NTHR / 32) {
+      const long row = m0 + rr;
       for (int c = lc; c < K0; c += 32) {
         float v;
         if (c < C1)
@@ -122,7 +127,7 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
           v = x_f32 ? ((const float*)x2)[row * C2 + (c - C1)]
                     : (float)((const __bf16*)x2)[row * C2 + (c - C1)];
         const u16 h = f32_to_bf16_rne3(v);
-        sa[0][r][c] = h;
+        sa[0][rr][c] = h;
         if (xsave != nullptr && g == 0) xsave[row * K0 + c] = h;
       }
     }
@@ -151,111 +156,96 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const bool k_aligned = (K % 8) == 0;
     const int kbody = k_aligned ? (K & ~31) : 0;
 
-    // QUAD-tile interleave: 4 independent accumulator chains per wave
-    // keep 4 B-fragment loads in flight per k-step with NO per-load
-    // branches (a per-element guard makes hipcc branch around each load
-    // and drain vmcnt — the round-1 guarded loop ran 0.74x per-layer).
     for (int t0 = wid * 4; t0 < ntiles; t0 += 4 * (NTHR / 64)) {
       const int nq = (ntiles - t0 < 4) ? (ntiles - t0) : 4;
-      f32x4 acc0{}, acc1{}, acc2{}, acc3{};
+      f32x4 acc[RM][4];
+#pragma unroll
+      for (int m = 0; m < RM; ++m)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[m][j] = f32x4{};
       const long r0 = (long)(t0 * 16 + fi) * K + fk * 8;
       const long rs = (long)16 * K;
+
+#define LOADQ(dst, kk)                                                    \
+      _Pragma("unroll") for (int j = 0; j < 4; ++j)                       \
+        dst[j] = *(const bf16x8*)&wg[r0 + j * rs + (kk)];
+#define MF4(bset, kk)                                                     \
+      {                                                                   \
+        _Pragma("unroll") for (int m = 0; m < RM; ++m) {                  \
+          const bf16x8 a_ =                                               \
+              *(const bf16x8*)&sa[cur][m * 16 + fi][(kk) + fk * 8];       \
+          _Pragma("unroll") for (int j = 0; j < 4; ++j)                   \
+            acc[m][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
+                a_, bset[j], acc[m][j], 0, 0, 0);                         \
+        }                                                                 \
+      }
+
       if (nq == 4 && (t0 + 4) * 16 <= N) {
-        // ---- fast path: all rows valid, unconditional 16B loads,
-        // manual unroll-2 software pipeline: set P computes while set Q
-        // loads (and vice versa) so the 4 B-fragment loads of step k+1
-        // stay in flight across step k's MFMAs (hipcc emits counted
-        // vmcnt for its own loads when the consumer is a different
-        // register set — no cross-iteration register copies).
-#define LOADQ(v0, v1, v2, v3, kk)                                  \
-        v0 = *(const bf16x8*)&wg[r0 + (kk)];                       \
-        v1 = *(const bf16x8*)&wg[r0 + rs + (kk)];                  \
-        v2 = *(const bf16x8*)&wg[r0 + 2 * rs + (kk)];              \
-        v3 = *(const bf16x8*)&wg[r0 + 3 * rs + (kk)];
-#define MF4(v0, v1, v2, v3, kk)                                            \
-        {                                                                  \
-          const bf16x8 a_ = *(const bf16x8*)&sa[cur][fi][(kk) + fk * 8];   \
-          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v0, acc0, 0, 0, 0); \
-          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v1, acc1, 0, 0, 0); \
-          acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v2, acc2, 0, 0, 0); \
-          acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_, v3, acc3, 0, 0, 0); \
-        }
+        // fast path: unconditional loads, unroll-2 software pipeline
         if (kbody >= 32) {
-          bf16x8 p0, p1, p2, p3, q0, q1, q2, q3;
-          LOADQ(p0, p1, p2, p3, 0)
+          bf16x8 P[4], Q[4];
+          LOADQ(P, 0)
           int k = 0;
           while (k + 64 <= kbody) {
-            LOADQ(q0, q1, q2, q3, k + 32)
-            MF4(p0, p1, p2, p3, k)
+            LOADQ(Q, k + 32)
+            MF4(P, k)
             if (k + 64 < kbody) {
-              LOADQ(p0, p1, p2, p3, k + 64)
+              LOADQ(P, k + 64)
             }
-            MF4(q0, q1, q2, q3, k + 32)
+            MF4(Q, k + 32)
             k += 64;
           }
-          if (k < kbody) MF4(p0, p1, p2, p3, k)
+          if (k < kbody) MF4(P, k)
         }
-#undef LOADQ
-#undef MF4
         for (int k = kbody; k < K; k += 32) {  // masked tail
           const int k8 = k + fk * 8;
-          const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
-          const int row = t0 * 16 + fi;
-          const bf16x8 b0 = load_bfrag(wg, row, N, K, k8);
-          const bf16x8 b1 = load_bfrag(wg, row + 16, N, K, k8);
-          const bf16x8 b2 = load_bfrag(wg, row + 32, N, K, k8);
-          const bf16x8 b3 = load_bfrag(wg, row + 48, N, K, k8);
-          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
-          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
-          acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
-          acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
+          bf16x8 B[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            B[j] = load_bfrag(wg, t0 * 16 + j * 16 + fi, N, K, k8);
+          MF4(B, k)
         }
       } else {
-        // ---- boundary tile-group / odd-K layer: guarded loads ----
-        const int row = t0 * 16 + fi;
+        // boundary tile-group / odd-K layer: guarded loads
         for (int k = 0; k < K; k += 32) {
           const int k8 = k + fk * 8;
-          const bf16x8 a = *(const bf16x8*)&sa[cur][fi][k8];
-          const bf16x8 b0 = load_bfrag(wg, row, N, K, k8);
-          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
-          if (nq > 1) {
-            const bf16x8 b1 = load_bfrag(wg, row + 16, N, K, k8);
-            acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
-          }
-          if (nq > 2) {
-            const bf16x8 b2 = load_bfrag(wg, row + 32, N, K, k8);
-            acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
-          }
-          if (nq > 3) {
-            const bf16x8 b3 = load_bfrag(wg, row + 48, N, K, k8);
-            acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
-          }
+          bf16x8 B[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            B[j] = (j < nq) ? load_bfrag(wg, t0 * 16 + j * 16 + fi, N, K, k8)
+                            : bf16x8{};
+          MF4(B, k)
         }
       }
+#undef LOADQ
+#undef MF4
+
       // epilogue: C/D map col = lane&15, row = (lane>>4)*4 + r
-      const f32x4 accs[4] = {acc0, acc1, acc2, acc3};
 #pragma unroll
-      for (int q = 0; q < 4; ++q) {
-        if (q >= nq) break;
-        const f32x4 a = accs[q];
-        const int col = (t0 + q) * 16 + fi;
-        if (col < N) {
-          const float bias = bb[col];
+      for (int m = 0; m < RM; ++m) {
 #pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int row = fk * 4 + r;
-            float v = a[r] + bias;
-            if (relu) v = fmaxf(v, 0.f);
-            if (last) {
-              if (row < rowlim) {
-                if (out_f32) yf[(long)(m0 + row) * N + col] = v;
-                else yh[(long)(m0 + row) * N + col] = f32_to_bf16_rne3(v);
+        for (int q = 0; q < 4; ++q) {
+          if (q >= nq) break;
+          const f32x4 a = acc[m][q];
+          const int col = (t0 + q) * 16 + fi;
+          if (col < N) {
+            const float bias = bb[col];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int row = m * 16 + fk * 4 + r;
+              float v = a[r] + bias;
+              if (relu) v = fmaxf(v, 0.f);
+              if (last) {
+                if (row < rowlim) {
+                  if (out_f32) yf[(long)(m0 + row) * N + col] = v;
+                  else yh[(long)(m0 + row) * N + col] = f32_to_bf16_rne3(v);
+                }
+              } else {
+                const u16 h = f32_to_bf16_rne3(v);
+                sa[nb][row][col] = h;
+                if (ag != nullptr && row < rowlim)
+                  ag[(long)(m0 + row) * N + col] = h;
               }
-            } else {
-              const u16 h = f32_to_bf16_rne3(v);
-              sa[nb][row][col] = h;
-              if (ag != nullptr && row < rowlim)
-                ag[(long)(m0 + row) * N + col] = h;
             }
           }
         }
@@ -264,7 +254,7 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     if (!last) {
       // zero the stale tail of the next buffer: cols [N, prevw(nb))
       const int pw = nb ? prevw1 : prevw0;
-      for (int i = tid; i < TM * (pw > N ? pw - N : 0); i += NTHR) {
+      for (int i = tid; i < TMv * (pw > N ? pw - N : 0); i += NTHR) {
         const int r = i / (pw - N), c = N + i % (pw - N);
         sa[nb][r][c] = 0;
       }
@@ -277,17 +267,18 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
 }
 
 // ---------------------------------------------------------------------------
-// Host binding: y(, acts...) = chain_fwd(x1[, x2], ws, bs, acts_save,
-//                                        act_last, G, out_f32)
+// Host binding: [y, x_bf16, act_0, ..., act_{L-2}] = mlp_chain_fwd_bf16(
+//     x1, x2_or_empty, ws, bs, act_last, G, out_f32, rm=0)
 // ws[i]: bf16 [G*N_i, K_i] (or [N,K] / [G,N,K]); bs[i]: f32 [G*N_i].
-// Inputs fp32 or bf16, 2-D [M, C]; x2 may be an empty tensor.
-// Returns [y, x_bf16, act_0, ..., act_{L-2}] — x_bf16 is the converted
-// (concatenated) input, so the backward consumes the same activation list
-// the per-layer path produced.
+// Inputs fp32 or bf16, 2-D [M, C]; x2 may be an empty tensor.  x_bf16 is
+// the converted (concatenated) input, so the backward consumes the same
+// activation list the per-layer path produced.  rm: 0 = auto (by M),
+// 1/2 = force 16/32 rows per workgroup.
 // ---------------------------------------------------------------------------
 static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
     torch::Tensor x1, torch::Tensor x2, std::vector<torch::Tensor> ws,
-    std::vector<torch::Tensor> bs, long act_last, long G, long out_f32) {
+    std::vector<torch::Tensor> bs, long act_last, long G, long out_f32,
+    long rm) {
   const int L = (int)ws.size();
   TORCH_CHECK(L >= 1 && L <= 6 && (int)bs.size() == L);
   TORCH_CHECK(x1.is_cuda() && x1.dim() == 2);
@@ -345,17 +336,32 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
   auto y = G == 1 ? torch::empty({M, (long)d.N[L - 1]}, yopts)
                   : torch::empty({G, M, (long)d.N[L - 1]}, yopts);
   out[0] = y;
-  dim3 grid((M + TM - 1) / TM, 1, G);
-  hipLaunchKernelGGL(k_bf16_chain_fwd, grid, dim3(chain::NTHR), 0, cur_stream3(),
-                     x1c.data_ptr(), has2 ? x2c.data_ptr() : nullptr,
-                     (int)C1, (int)C2, (int)(xf32 ? 1 : 0),
-                     (u16*)xsave.data_ptr(), d, y.data_ptr(), (int)M,
-                     (int)out_f32);
+  // RM=2 halves the L2 weight re-read; keep >= ~64 workgroups for
+  // chip fill, else fall back to RM=1.
+  int use_rm = (int)rm;
+  if (use_rm == 0) use_rm = ((M / 32) * G >= 64) ? 2 : 1;
+  const int TMv = 16 * use_rm;
+  dim3 grid((M + TMv - 1) / TMv, 1, G);
+  if (use_rm == 2)
+    hipLaunchKernelGGL(k_bf16_chain_fwd<2>, grid, dim3(NTHR), 0,
+                       cur_stream3(), x1c.data_ptr(),
+                       has2 ? x2c.data_ptr() : nullptr, (int)C1, (int)C2,
+                       (int)(xf32 ? 1 : 0), (u16*)xsave.data_ptr(), d,
+                       y.data_ptr(), (int)M, (int)out_f32);
+  else
+    hipLaunchKernelGGL(k_bf16_chain_fwd<1>, grid, dim3(NTHR), 0,
+                       cur_stream3(), x1c.data_ptr(),
+                       has2 ? x2c.data_ptr() : nullptr, (int)C1, (int)C2,
+                       (int)(xf32 ? 1 : 0), (u16*)xsave.data_ptr(), d,
+                       y.data_ptr(), (int)M, (int)out_f32);
   return out;
 }
 
 }  // namespace chain
 
 void register_chain(pybind11::module_& m) {
-  m.def("mlp_chain_fwd_bf16", &chain::mlp_chain_fwd_bf16);
+  m.def("mlp_chain_fwd_bf16", &chain::mlp_chain_fwd_bf16,
+        pybind11::arg("x1"), pybind11::arg("x2"), pybind11::arg("ws"),
+        pybind11::arg("bs"), pybind11::arg("act_last"), pybind11::arg("G"),
+        pybind11::arg("out_f32"), pybind11::arg("rm") = 0);
 }

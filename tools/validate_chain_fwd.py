@@ -56,7 +56,7 @@ def per_layer_ref(x1, x2, ws, bs, act_last, G, out_f32):
 
 
 def run_case(name, M, dims, G, two_src=False, act_last=0, out_f32=1,
-             bench=False, in_bf16=False):
+             bench=False, in_bf16=False, rm=0):
     torch.manual_seed(0)
     K0 = dims[0]
     if two_src:
@@ -82,7 +82,7 @@ def run_case(name, M, dims, G, two_src=False, act_last=0, out_f32=1,
     y_ref, acts_ref = per_layer_ref(x1, x2, ws, bs, act_last, G, out_f32)
     out = ext.mlp_chain_fwd_bf16(
         x1, x2 if x2 is not None else torch.empty(0, device=dev),
-        ws, bs, act_last, G, out_f32)
+        ws, bs, act_last, G, out_f32, rm)
     y, xsave = out[0], out[1]
     ok = True
     d0 = (xsave.float() - acts_ref[0].float()).abs().max().item()
@@ -98,13 +98,15 @@ def run_case(name, M, dims, G, two_src=False, act_last=0, out_f32=1,
     print(f"  {name}: xsave_d={d0:.1e} y_d={dy:.3e} acts_d="
           f"{[f'{v:.1e}' for v in dacts]} -> {'OK' if ok else 'FAIL'}")
     if bench:
+        x2e = x2 if x2 is not None else torch.empty(0, device=dev)
         t_ref = timeit(lambda: per_layer_ref(x1, x2, ws, bs, act_last, G,
                                              out_f32))
-        t_new = timeit(lambda: ext.mlp_chain_fwd_bf16(
-            x1, x2 if x2 is not None else torch.empty(0, device=dev),
-            ws, bs, act_last, G, out_f32))
-        print(f"      per-layer {t_ref:8.2f} us   chain {t_new:8.2f} us   "
-              f"({t_ref / t_new:.2f}x)")
+        t1 = timeit(lambda: ext.mlp_chain_fwd_bf16(
+            x1, x2e, ws, bs, act_last, G, out_f32, 1))
+        t2 = timeit(lambda: ext.mlp_chain_fwd_bf16(
+            x1, x2e, ws, bs, act_last, G, out_f32, 2))
+        print(f"      per-layer {t_ref:8.2f} us   chain rm1 {t1:8.2f} us "
+              f"({t_ref / t1:.2f}x)   rm2 {t2:8.2f} us ({t_ref / t2:.2f}x)")
     return ok
 
 
@@ -122,6 +124,11 @@ def main():
                    two_src=True, bench=True)
     ok &= run_case("bf16-in        ", 512, [104, 400, 1], 2, in_bf16=True)
     ok &= run_case("width512       ", 512, [512, 512, 16], 1)
+    ok &= run_case("rm1-forced     ", 2560, [49, 400, 400, 8], 1, rm=1)
+    ok &= run_case("rm2-forced     ", 1280, [53, 400, 400, 1], 2,
+                   two_src=True, rm=2)
+    ok &= run_case("rm2-oddM       ", 1304, [53, 400, 400, 1], 2,
+                   two_src=True, rm=2)
     print("PASS" if ok else "FAIL")
     return 0 if ok else 1
 
