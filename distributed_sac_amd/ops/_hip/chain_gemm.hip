@@ -163,12 +163,27 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
       for (int m = 0; m < RM; ++m)
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[m][j] = f32x4{};
-      const long r0 = (long)(t0 * 16 + fi) * K + fk * 8;
-      const long rs = (long)16 * K;
+      // Per-lane CLAMPED row bases: surplus quad slots (j >= nq) and
+      // out-of-range rows of a partial-N tile load a VALID row instead
+      // of branching — their MFMA lanes produce garbage that the
+      // epilogue's col<N / q<nq masks never store, and out-of-range K
+      // contributions are killed by the zero-padded LDS A operand.
+      // This keeps EVERY K%8==0 layer (including the N=1/N=8 heads and
+      // the 25th tile of a 400-wide layer) on the branch-free pipelined
+      // path — the guarded loop (a per-load branch + vmcnt drain) made
+      // one straggler wave serialize the whole chain at the layer
+      // barrier.
+      long rb[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = (t0 + (j < nq ? j : nq - 1)) * 16 + fi;
+        if (row >= N) row = N - 1;
+        rb[j] = (long)row * K;
+      }
 
 #define LOADQ(dst, kk)                                                    \
       _Pragma("unroll") for (int j = 0; j < 4; ++j)                       \
-        dst[j] = *(const bf16x8*)&wg[r0 + j * rs + (kk)];
+        dst[j] = *(const bf16x8*)&wg[rb[j] + fk * 8 + (kk)];
 #define MF4(bset, kk)                                                     \
       {                                                                   \
         _Pragma("unroll") for (int m = 0; m < RM; ++m) {                  \
@@ -180,8 +195,8 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
         }                                                                 \
       }
 
-      if (nq == 4 && (t0 + 4) * 16 <= N) {
-        // fast path: unconditional loads, unroll-2 software pipeline
+      if (k_aligned) {
+        // unconditional loads, unroll-2 software pipeline
         if (kbody >= 32) {
           bf16x8 P[4], Q[4];
           LOADQ(P, 0)
@@ -197,16 +212,20 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
           }
           if (k < kbody) MF4(P, k)
         }
-        for (int k = kbody; k < K; k += 32) {  // masked tail
-          const int k8 = k + fk * 8;
+        if (kbody < K) {
+          // K%32 tail: clamp each lane's k-run into range — lanes whose
+          // true k-run starts past K read duplicate (valid) bytes that
+          // multiply the zero LDS pad.
+          const int k8 = kbody + fk * 8;
+          const int k8c = (k8 + 8 <= K) ? k8 : (K - 8);
           bf16x8 B[4];
 #pragma unroll
           for (int j = 0; j < 4; ++j)
-            B[j] = load_bfrag(wg, t0 * 16 + j * 16 + fi, N, K, k8);
-          MF4(B, k)
+            B[j] = *(const bf16x8*)&wg[rb[j] + k8c];
+          MF4(B, kbody)
         }
       } else {
-        // boundary tile-group / odd-K layer: guarded loads
+        // odd-K layer (the 49/53-d layer-0 inputs): guarded loads
         for (int k = 0; k < K; k += 32) {
           const int k8 = k + fk * 8;
           bf16x8 B[4];
