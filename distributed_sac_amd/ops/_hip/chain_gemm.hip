@@ -196,21 +196,34 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
       }
 
       if (k_aligned) {
-        // unconditional loads, unroll-2 software pipeline
+        // unconditional loads, 3-set rotating software pipeline: loads
+        // run TWO k-steps ahead of their MFMAs (12 B-fragments in
+        // flight), hiding cold-L2/L3 weight-miss latency that a 1-deep
+        // pipeline left exposed.
         if (kbody >= 32) {
-          bf16x8 P[4], Q[4];
-          LOADQ(P, 0)
+          bf16x8 S0[4], S1[4], S2[4];
+          LOADQ(S0, 0)
+          if (32 < kbody) LOADQ(S1, 32)
           int k = 0;
-          while (k + 64 <= kbody) {
-            LOADQ(Q, k + 32)
-            MF4(P, k)
-            if (k + 64 < kbody) {
-              LOADQ(P, k + 64)
+          while (k + 96 <= kbody) {
+            LOADQ(S2, k + 64)
+            MF4(S0, k)
+            if (k + 96 < kbody) {
+              LOADQ(S0, k + 96)
             }
-            MF4(Q, k + 32)
-            k += 64;
+            MF4(S1, k + 32)
+            if (k + 128 < kbody) {
+              LOADQ(S1, k + 128)
+            }
+            MF4(S2, k + 64)
+            k += 96;
           }
-          if (k < kbody) MF4(P, k)
+          if (kbody - k == 32) {
+            MF4(S0, k)
+          } else if (kbody - k == 64) {
+            MF4(S0, k)
+            MF4(S1, k + 32)
+          }
         }
         if (kbody < K) {
           // K%32 tail: clamp each lane's k-run into range — lanes whose
