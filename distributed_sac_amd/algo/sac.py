@@ -534,6 +534,26 @@ class SACEngine:
             self.ddp.allreduce_grad_(self._aa_arena)
         return self._manual_seg3()
 
+    @property
+    def _use_chain(self) -> bool:
+        """Fused MLP-chain forward kernels (one launch per chain,
+        in-kernel f32 concat+cast input — kills the per-layer launch
+        latency plus the cat/cast launches; round-2).  DSAC_CHAIN=0
+        restores the per-layer path."""
+        import os as _os
+        return _os.environ.get("DSAC_CHAIN", "1") == "1"
+
+    @torch.no_grad()
+    def _chain_fwd(self, x1, x2, ws_bf16, bs_f32, G, act_last=0,
+                   out_f32=True, rowcat=False, save_acts=True):
+        from ..ops import native
+        out = native().mlp_chain_fwd_bf16(
+            x1, x2 if x2 is not None else x1.new_empty(0),
+            list(ws_bf16), [b.contiguous() for b in bs_f32],
+            int(act_last), int(G), 1 if out_f32 else 0, 0,
+            1 if rowcat else 0, 1 if save_acts else 0)
+        return out[0], [out[1]] + list(out[2:])
+
     @torch.no_grad()
     def _manual_seg1(self, batch):
         """Segment 1: batched actor forward + squash, TD target, critic
@@ -551,11 +571,18 @@ class SACEngine:
         B = states.shape[0]
         A = self.cfg.action_dim
         la_det = self.log_alpha.detach()
+        chain = self._use_chain
 
         # ---- batched actor forward + squash --------------------------
-        x_cat = torch.cat([next_states, states], dim=0)
         ws_f32, bs_f32 = self._actor_weights()
-        out, acts_a = self._mlp_fwd_manual(x_cat, self._actor_ws_bf16, bs_f32)
+        if chain:
+            out, acts_a = self._chain_fwd(next_states, states,
+                                          self._actor_ws_bf16, bs_f32,
+                                          G=1, rowcat=True)
+        else:
+            x_cat = torch.cat([next_states, states], dim=0)
+            out, acts_a = self._mlp_fwd_manual(x_cat, self._actor_ws_bf16,
+                                               bs_f32)
         mu, lsr = out[:, :A], out[:, A:]
         if self._eps_queue:
             eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
@@ -566,16 +593,29 @@ class SACEngine:
         na, nlp = a_cat[:B], lp_cat[:B]
 
         # ---- TD target ------------------------------------------------
-        xt = torch.cat([next_states, na], dim=-1)
-        q1_t, q2_t, _ = self._twin_fwd_manual(
-            xt, self._twin_target_bf16, self._twin_target[1])
+        if chain:
+            yt, _ = self._chain_fwd(next_states, na,
+                                    self._twin_target_bf16,
+                                    self._twin_target[1], G=2,
+                                    save_acts=False)
+            q1_t, q2_t = yt[0], yt[1]
+        else:
+            xt = torch.cat([next_states, na], dim=-1)
+            q1_t, q2_t, _ = self._twin_fwd_manual(
+                xt, self._twin_target_bf16, self._twin_target[1])
         y = ext.td_target_mt(rewards, dones, q1_t, q2_t, nlp, states,
                              la_det, T, self.gamma, self.reward_scale)
 
         # ---- critic loss + manual backward ---------------------------
-        x = torch.cat([states, actions], dim=-1)
-        q1, q2, acts_c = self._twin_fwd_manual(
-            x, self._twin_local_bf16, self._twin_local[1])
+        if chain:
+            yq, acts_c = self._chain_fwd(states, actions,
+                                         self._twin_local_bf16,
+                                         self._twin_local[1], G=2)
+            q1, q2 = yq[0], yq[1]
+        else:
+            x = torch.cat([states, actions], dim=-1)
+            q1, q2, acts_c = self._twin_fwd_manual(
+                x, self._twin_local_bf16, self._twin_local[1])
         closs = ext.critic_loss_fwd(q1, q2, y, states, la_det, T,
                                     int(use_w))[0]
         dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
@@ -616,9 +656,14 @@ class SACEngine:
         nl_c = len(self._twin_local_bf16)
         acts_a = st["acts_a"]
 
-        xa = torch.cat([states, sa], dim=-1)
-        aq1, aq2, acts_f = self._twin_fwd_manual(
-            xa, self._twin_local_bf16, self._twin_local[1])
+        if self._use_chain:
+            ya, acts_f = self._chain_fwd(states, sa, self._twin_local_bf16,
+                                         self._twin_local[1], G=2)
+            aq1, aq2 = ya[0], ya[1]
+        else:
+            xa = torch.cat([states, sa], dim=-1)
+            aq1, aq2, acts_f = self._twin_fwd_manual(
+                xa, self._twin_local_bf16, self._twin_local[1])
         al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls, states,
                                       la_det, T, int(use_w), self.H_bar_f)
         self.alpha_group.flat_grad.zero_()

@@ -93,7 +93,8 @@ __device__ __forceinline__ bf16x8 load_bfrag(const u16* __restrict__ w,
 template <int RM>
 __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const void* __restrict__ x1, const void* __restrict__ x2,
-    int C1, int C2, int x_f32, u16* __restrict__ xsave,
+    int C1, int C2, int x_f32, int rowcat, int M1,
+    u16* __restrict__ xsave,
     ChainFwdDesc d, void* __restrict__ y, int M, int out_f32) {
   constexpr int TMv = 16 * RM;
   __shared__ u16 sa[2][TMv][CPAD];
@@ -103,7 +104,7 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
   const int lane = tid & 63, wid = tid >> 6;
   const int fi = lane & 15, fk = lane >> 4;
   const int rowlim = (M - m0 < TMv ? M - m0 : TMv);
-  const int K0 = C1 + C2;
+  const int K0 = rowcat ? C1 : C1 + C2;
 
   // zero both activation buffers (pads must read 0 in the MFMA A-frags)
   {
@@ -112,7 +113,9 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
   }
   __syncthreads();
 
-  // ---- input tile -> sa[0] (converted to bf16; optional col-concat) ----
+  // ---- input tile -> sa[0] (converted to bf16) --------------------
+  // rowcat: x = [x1 ; x2] stacked over rows (equal widths C1);
+  // else:   x = [x1 || x2] concatenated over columns (C1 + C2).
   {
     const int r = tid >> 5;      // 32 threads per row covers 16 rows/pass
     const int lc = tid & 31;
@@ -120,12 +123,18 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
       const long row = m0 + rr;
       for (int c = lc; c < K0; c += 32) {
         float v;
-        if (c < C1)
+        if (rowcat) {
+          const bool top = row < M1;
+          const long rr2 = top ? row : row - M1;
+          v = x_f32 ? ((const float*)(top ? x1 : x2))[rr2 * C1 + c]
+                    : (float)((const __bf16*)(top ? x1 : x2))[rr2 * C1 + c];
+        } else if (c < C1) {
           v = x_f32 ? ((const float*)x1)[row * C1 + c]
                     : (float)((const __bf16*)x1)[row * C1 + c];
-        else
+        } else {
           v = x_f32 ? ((const float*)x2)[row * C2 + (c - C1)]
                     : (float)((const __bf16*)x2)[row * C2 + (c - C1)];
+        }
         const u16 h = f32_to_bf16_rne3(v);
         sa[0][rr][c] = h;
         if (xsave != nullptr && g == 0) xsave[row * K0 + c] = h;
@@ -300,17 +309,20 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
 
 // ---------------------------------------------------------------------------
 // Host binding: [y, x_bf16, act_0, ..., act_{L-2}] = mlp_chain_fwd_bf16(
-//     x1, x2_or_empty, ws, bs, act_last, G, out_f32, rm=0)
+//     x1, x2_or_empty, ws, bs, act_last, G, out_f32, rm=0, rowcat=0,
+//     save_acts=1)
 // ws[i]: bf16 [G*N_i, K_i] (or [N,K] / [G,N,K]); bs[i]: f32 [G*N_i].
 // Inputs fp32 or bf16, 2-D [M, C]; x2 may be an empty tensor.  x_bf16 is
 // the converted (concatenated) input, so the backward consumes the same
 // activation list the per-layer path produced.  rm: 0 = auto (by M),
-// 1/2 = force 16/32 rows per workgroup.
+// 1/2 = force 16/32 rows per workgroup.  rowcat=1 stacks x1/x2 over ROWS
+// (equal widths; M = M1+M2) instead of columns.  save_acts=0 skips the
+// intermediate-activation writes (no-grad forwards, e.g. TD target).
 // ---------------------------------------------------------------------------
 static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
     torch::Tensor x1, torch::Tensor x2, std::vector<torch::Tensor> ws,
     std::vector<torch::Tensor> bs, long act_last, long G, long out_f32,
-    long rm) {
+    long rm, long rowcat, long save_acts) {
   const int L = (int)ws.size();
   TORCH_CHECK(L >= 1 && L <= 6 && (int)bs.size() == L);
   TORCH_CHECK(x1.is_cuda() && x1.dim() == 2);
@@ -322,13 +334,19 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
   torch::Tensor x2c = x2;
   if (has2) {
     TORCH_CHECK(x2.is_cuda() && x2.dim() == 2
-                && x2.scalar_type() == x1.scalar_type()
-                && x2.size(0) == x1.size(0));
+                && x2.scalar_type() == x1.scalar_type());
+    if (rowcat) {
+      TORCH_CHECK(x2.size(1) == x1.size(1), "rowcat needs equal widths");
+    } else {
+      TORCH_CHECK(x2.size(0) == x1.size(0));
+    }
     x2c = x2.contiguous();
   }
-  const long M = x1c.size(0);
+  TORCH_CHECK(!rowcat || has2, "rowcat needs two inputs");
+  const long M1 = x1c.size(0);
+  const long M = rowcat ? M1 + x2c.size(0) : M1;
   const long C1 = x1c.size(1), C2 = has2 ? x2c.size(1) : 0;
-  const long K0 = C1 + C2;
+  const long K0 = rowcat ? C1 : C1 + C2;
   TORCH_CHECK(K0 <= CMAX, "chain fwd: input width must be <= ", CMAX);
 
   ChainFwdDesc d{};
@@ -355,7 +373,7 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
     d.N[i] = (int)N;
     d.act[i] = (i == L - 1) ? (int)act_last : 1;
     d.acts[i] = nullptr;
-    if (i < L - 1) {
+    if (save_acts && i < L - 1) {
       auto a = G == 1 ? torch::empty({M, N}, bopts)
                       : torch::empty({G, M, N}, bopts);
       d.acts[i] = (u16*)a.data_ptr();
@@ -378,13 +396,15 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
     hipLaunchKernelGGL(k_bf16_chain_fwd<2>, grid, dim3(NTHR), 0,
                        cur_stream3(), x1c.data_ptr(),
                        has2 ? x2c.data_ptr() : nullptr, (int)C1, (int)C2,
-                       (int)(xf32 ? 1 : 0), (u16*)xsave.data_ptr(), d,
+                       (int)(xf32 ? 1 : 0), (int)rowcat, (int)M1,
+                       (u16*)xsave.data_ptr(), d,
                        y.data_ptr(), (int)M, (int)out_f32);
   else
     hipLaunchKernelGGL(k_bf16_chain_fwd<1>, grid, dim3(NTHR), 0,
                        cur_stream3(), x1c.data_ptr(),
                        has2 ? x2c.data_ptr() : nullptr, (int)C1, (int)C2,
-                       (int)(xf32 ? 1 : 0), (u16*)xsave.data_ptr(), d,
+                       (int)(xf32 ? 1 : 0), (int)rowcat, (int)M1,
+                       (u16*)xsave.data_ptr(), d,
                        y.data_ptr(), (int)M, (int)out_f32);
   return out;
 }
@@ -395,5 +415,6 @@ void register_chain(pybind11::module_& m) {
   m.def("mlp_chain_fwd_bf16", &chain::mlp_chain_fwd_bf16,
         pybind11::arg("x1"), pybind11::arg("x2"), pybind11::arg("ws"),
         pybind11::arg("bs"), pybind11::arg("act_last"), pybind11::arg("G"),
-        pybind11::arg("out_f32"), pybind11::arg("rm") = 0);
+        pybind11::arg("out_f32"), pybind11::arg("rm") = 0,
+        pybind11::arg("rowcat") = 0, pybind11::arg("save_acts") = 1);
 }
