@@ -386,10 +386,13 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
   auto y = G == 1 ? torch::empty({M, (long)d.N[L - 1]}, yopts)
                   : torch::empty({G, M, (long)d.N[L - 1]}, yopts);
   out[0] = y;
-  // RM=2 halves the L2 weight re-read; keep >= ~64 workgroups for
-  // chip fill, else fall back to RM=1.
+  // RM=1 measured faster at every bench shape (RM=2 halves the L2
+  // weight re-read but also halves the workgroups, and these chains are
+  // occupancy/latency-bound, not L2-bound — tools/validate_chain_fwd.py
+  // A/B and the in-graph profile both agree).  rm=2 stays available for
+  // experiments.
   int use_rm = (int)rm;
-  if (use_rm == 0) use_rm = ((M / 32) * G >= 64) ? 2 : 1;
+  if (use_rm == 0) use_rm = 1;
   const int TMv = 16 * use_rm;
   dim3 grid((M + TMv - 1) / TMv, 1, G);
   if (use_rm == 2)
