@@ -277,37 +277,16 @@ __global__ __launch_bounds__(256) void k_bf16_dx(
 // A = dy^T: [n][m-run] (transposed store); B = x: fragment k-run over m at
 // fixed k -> x transposed [k][m-run].  grid (N/64, K/64, G*S).
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
-    const u16* __restrict__ dy, const u16* __restrict__ x,
-    const u16* __restrict__ yout, float* __restrict__ ws,
-    float* __restrict__ ws_db, int M, int N, int K, int act, int S,
-    int chunk, long xgs, int transpose_w, long s_stride, long w_off,
-    long b_off) {
-  __shared__ u16 sa[2][TBN][TPAD];    // dy^T: [n][m-run]
-  __shared__ u16 sbT[2][TBN][TPAD];   // x^T:  [k][m-run]
-  const int gs = blockIdx.z;
-  const int g = gs / S, s = gs % S;
-  const u16* dyg = dy + (long)g * M * N;
-  const u16* yg = yout + (long)g * M * N;
-  const u16* xg = x + (long)g * xgs;
-  // s_stride>0: phase-arena mode — partials for split s of EVERY layer
-  // land in one [S, group_numel] arena laid out in flat-gradient order
-  // (w_off/b_off = the layer's offsets in the flat buffer), so ONE
-  // reduce per phase folds the whole gradient.
-  float* wsp = s_stride > 0
-      ? ws + (long)s * s_stride + w_off + (long)g * N * K
-      : ws + (long)gs * N * K;
-  float* dbp = s_stride > 0
-      ? ws + (long)s * s_stride + b_off + (long)g * N
-      : ws_db + (long)gs * N;
-  const int m_lo = s * chunk;
-  const int m_hi = min(M, m_lo + chunk);
-  const int n0 = blockIdx.x * TBM, c0 = blockIdx.y * TBN;
+__device__ __forceinline__ void dwdb_body(
+    u16 (*sa)[TBN][TPAD], u16 (*sbT)[TBN][TPAD],
+    const u16* __restrict__ dyg, const u16* __restrict__ xg,
+    const u16* __restrict__ yg, float* __restrict__ wsp,
+    float* __restrict__ dbp, int N, int K, int act, int m_lo, int m_hi,
+    int n0, int c0, bool do_db, int transpose_w) {
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
   const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
   const int fi = lane & 15, fk = lane >> 4;
-  const bool do_db = (blockIdx.y == 0);
   float db_acc = 0.f;
   f32x4 acc00{}, acc01{}, acc10{}, acc11{};
 
@@ -409,6 +388,72 @@ __global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
           wsp[transpose_w ? (long)col * N + row : (long)row * K + col] = a[r];
       }
     }
+}
+
+__global__ __launch_bounds__(256) void k_bf16_dwdb_splitk(
+    const u16* __restrict__ dy, const u16* __restrict__ x,
+    const u16* __restrict__ yout, float* __restrict__ ws,
+    float* __restrict__ ws_db, int M, int N, int K, int act, int S,
+    int chunk, long xgs, int transpose_w, long s_stride, long w_off,
+    long b_off) {
+  __shared__ u16 sa[2][TBN][TPAD];    // dy^T: [n][m-run]
+  __shared__ u16 sbT[2][TBN][TPAD];   // x^T:  [k][m-run]
+  const int gs = blockIdx.z;
+  const int g = gs / S, s = gs % S;
+  const u16* dyg = dy + (long)g * M * N;
+  const u16* yg = yout + (long)g * M * N;
+  const u16* xg = x + (long)g * xgs;
+  // s_stride>0: phase-arena mode — partials for split s of EVERY layer
+  // land in one [S, group_numel] arena laid out in flat-gradient order
+  // (w_off/b_off = the layer's offsets in the flat buffer), so ONE
+  // reduce per phase folds the whole gradient.
+  float* wsp = s_stride > 0
+      ? ws + (long)s * s_stride + w_off + (long)g * N * K
+      : ws + (long)gs * N * K;
+  float* dbp = s_stride > 0
+      ? ws + (long)s * s_stride + b_off + (long)g * N
+      : ws_db + (long)gs * N;
+  dwdb_body(sa, sbT, dyg, xg, yg, wsp, dbp, N, K, act, s * chunk,
+            min(M, s * chunk + chunk), blockIdx.x * TBM,
+            blockIdx.y * TBN, blockIdx.y == 0, transpose_w);
+}
+
+// ---------------------------------------------------------------------------
+// Grouped dwdb: ONE launch computes the dW/db split-K partials of up to
+// six layers (a whole chain's weight gradients) straight into the phase
+// arena.  The per-layer dy inputs are the PRE-MASKED dys the fused
+// dx-chain kernel saved (act=0 here).  grid (sum_l nbx_l*nby_l, 1, G*S).
+// ---------------------------------------------------------------------------
+struct DwGroupDesc {
+  const u16* dy[6];
+  const u16* x[6];
+  long xgs[6];
+  long w_off[6], b_off[6];
+  int N[6], K[6], nbx[6];
+  int cum[7];
+  int L, S, chunk, M;
+};
+
+__global__ __launch_bounds__(256) void k_bf16_dwdb_grouped(
+    DwGroupDesc d, float* __restrict__ arena, long s_stride) {
+  __shared__ u16 sa[2][TBN][TPAD];
+  __shared__ u16 sbT[2][TBN][TPAD];
+  const int bx = blockIdx.x;
+  int l = 0;
+  while (l + 1 < d.L && bx >= d.cum[l + 1]) ++l;
+  const int local = bx - d.cum[l];
+  const int n0 = (local % d.nbx[l]) * TBM;
+  const int c0 = (local / d.nbx[l]) * TBN;
+  const int gs = blockIdx.z;
+  const int g = gs / d.S, s = gs % d.S;
+  const int N = d.N[l], K = d.K[l];
+  const u16* dyg = d.dy[l] + (long)g * d.M * N;
+  const u16* xg = d.x[l] + (long)g * d.xgs[l];
+  float* wsp = arena + (long)s * s_stride + d.w_off[l] + (long)g * N * K;
+  float* dbp = arena + (long)s * s_stride + d.b_off[l] + (long)g * N;
+  dwdb_body(sa, sbT, dyg, xg, dyg, wsp, dbp, N, K, /*act=*/0,
+            s * d.chunk, min(d.M, s * d.chunk + d.chunk), n0, c0,
+            c0 == 0, /*transpose_w=*/0);
 }
 
 // ---------------------------------------------------------------------------
@@ -1117,7 +1162,61 @@ static void mlp_narrow_bwd_bf16(torch::Tensor dy_last,
                      (int)Kin);
 }
 
+static void dwdb_grouped_arena(std::vector<torch::Tensor> dys,
+                               std::vector<torch::Tensor> xs,
+                               torch::Tensor arena,
+                               std::vector<long> w_offs,
+                               std::vector<long> b_offs, long G, long S,
+                               long chunk) {
+  CHECK_F32(arena);
+  const int L = (int)dys.size();
+  TORCH_CHECK(L >= 1 && L <= 6 && (int)xs.size() == L
+              && (int)w_offs.size() == L && (int)b_offs.size() == L);
+  TORCH_CHECK(arena.dim() == 2 && arena.is_contiguous()
+              && arena.size(0) >= S);
+  const long stride = arena.size(1);
+  DwGroupDesc d{};
+  d.L = L;
+  d.S = (int)S;
+  d.chunk = (int)chunk;
+  std::vector<torch::Tensor> keep;
+  long M = -1;
+  int cum = 0;
+  for (int i = 0; i < L; ++i) {
+    CHECK_BF16(dys[i]);
+    CHECK_BF16(xs[i]);
+    auto dc = dys[i].contiguous();
+    auto xc = xs[i].contiguous();
+    keep.push_back(dc);
+    keep.push_back(xc);
+    TORCH_CHECK(dc.dim() == 3 && dc.size(0) == G);
+    const long Mi = dc.size(1), N = dc.size(2);
+    const long K = xc.size(-1);
+    if (M < 0) M = Mi;
+    TORCH_CHECK(Mi == M, "all layers must share the batch dim");
+    TORCH_CHECK(xc.numel() == (xc.dim() == 3 ? G : 1) * M * K);
+    d.dy[i] = (const u16*)dc.data_ptr();
+    d.x[i] = (const u16*)xc.data_ptr();
+    d.xgs[i] = xc.dim() == 3 ? M * K : 0;
+    d.w_off[i] = w_offs[i];
+    d.b_off[i] = b_offs[i];
+    TORCH_CHECK(w_offs[i] + G * N * K <= stride
+                && b_offs[i] + G * N <= stride);
+    d.N[i] = (int)N;
+    d.K[i] = (int)K;
+    d.nbx[i] = (int)((N + TBM - 1) / TBM);
+    d.cum[i] = cum;
+    cum += d.nbx[i] * (int)((K + TBN - 1) / TBN);
+  }
+  d.cum[L] = cum;
+  d.M = (int)M;
+  dim3 grid(cum, 1, (unsigned)(G * S));
+  hipLaunchKernelGGL(k_bf16_dwdb_grouped, grid, dim3(256), 0,
+                     cur_stream2(), d, arena.data_ptr<float>(), stride);
+}
+
 void register_bf16(pybind11::module_& m) {
+  m.def("dwdb_grouped_arena", &dwdb_grouped_arena);
   m.def("f32_to_bf16_", &f32_to_bf16_);
   m.def("attn_pool_fwd", &attn_pool_fwd);
   m.def("mlp_narrow_fwd_bf16", &mlp_narrow_fwd_bf16);

@@ -308,6 +308,371 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
 }
 
 // ---------------------------------------------------------------------------
+// Multi-tensor 64x64 tiled weight transpose: wt[g][k][n] = w[g][n][k]
+// (bf16).  ONE launch re-materializes every transposed weight mirror the
+// fused dx-chain consumes — the mirrors change every Adam step, and the
+// dx GEMM's B-fragment needs n-minor (k-run over n) layout; reading W^T
+// from global with the fwd chain's fast contiguous loader beats per-tile
+// LDS transpose staging inside the chain (which re-stages W per WG).
+// ---------------------------------------------------------------------------
+struct TransDesc {
+  const u16* w[8];
+  u16* wt[8];
+  int N[8], K[8];
+  int cum[9];   // cumulative tiles per layer (tiles = G*ceil(N/64)*ceil(K/64))
+  int ntx[8];   // ceil(N/64)
+  int nty[8];   // ceil(K/64)
+  int L;
+  int G;
+};
+
+__global__ __launch_bounds__(256) void k_bf16_transpose_multi(TransDesc d) {
+  __shared__ u16 t[64][72];   // +8 u16 pad: conflict-free column reads
+  int bx = blockIdx.x;
+  int l = 0;
+  while (l + 1 < d.L && bx >= d.cum[l + 1]) ++l;
+  int local = bx - d.cum[l];
+  const int per_g = d.ntx[l] * d.nty[l];
+  const int g = local / per_g;
+  local -= g * per_g;
+  const int n0 = (local % d.ntx[l]) * 64;
+  const int k0 = (local / d.ntx[l]) * 64;
+  const int N = d.N[l], K = d.K[l];
+  const u16* w = d.w[l] + (long)g * N * K;
+  u16* wt = d.wt[l] + (long)g * N * K;
+  const int tid = threadIdx.x;
+  const int r = tid >> 3, c0 = (tid & 7) * 8;   // 32 rows/pass, 8 cols/thread
+#pragma unroll
+  for (int rr = 0; rr < 64; rr += 32) {
+    const int row = n0 + r + rr;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int col = k0 + c0 + j;
+      t[r + rr][c0 + j] = (row < N && col < K)
+          ? w[(long)row * K + col] : (u16)0;
+    }
+  }
+  __syncthreads();
+#pragma unroll
+  for (int rr = 0; rr < 64; rr += 32) {
+    const int row = k0 + r + rr;        // wt row = k
+    if (row < K) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int col = n0 + c0 + j;    // wt col = n
+        if (col < N) wt[(long)row * N + col] = t[c0 + j][r + rr];
+      }
+    }
+  }
+}
+
+static void transpose_weights_bf16(std::vector<torch::Tensor> ws,
+                                   std::vector<torch::Tensor> wts,
+                                   long G) {
+  const int L = (int)ws.size();
+  TORCH_CHECK(L >= 1 && L <= 8 && (int)wts.size() == L);
+  TransDesc d{};
+  d.L = L;
+  d.G = (int)G;
+  int cum = 0;
+  std::vector<torch::Tensor> keep;
+  for (int i = 0; i < L; ++i) {
+    CHAIN_CHECK_BF16(ws[i]);
+    CHAIN_CHECK_BF16(wts[i]);
+    auto wc = ws[i].contiguous();
+    keep.push_back(wc);
+    TORCH_CHECK(wts[i].is_contiguous());
+    const long NK = wc.numel() / G;
+    const long K = wc.size(-1);
+    const long N = NK / K;
+    TORCH_CHECK(wts[i].numel() == wc.numel());
+    d.w[i] = (const u16*)wc.data_ptr();
+    d.wt[i] = (u16*)wts[i].data_ptr();
+    d.N[i] = (int)N;
+    d.K[i] = (int)K;
+    d.ntx[i] = (int)((N + 63) / 64);
+    d.nty[i] = (int)((K + 63) / 64);
+    d.cum[i] = cum;
+    cum += (int)G * d.ntx[i] * d.nty[i];
+  }
+  d.cum[L] = cum;
+  hipLaunchKernelGGL(k_bf16_transpose_multi, dim3(cum), dim3(256), 0,
+                     cur_stream3(), d);
+}
+
+// ---------------------------------------------------------------------------
+// Fused MLP-chain BACKWARD-dx: walk the chain top-down, keeping dy in
+// LDS; per layer (a) mask dy by the saved post-ReLU activation and write
+// the masked dy to global (the grouped dwdb launch consumes it with no
+// re-masking), (b) dy_{l-1} = dy_masked @ W_l via the TRANSPOSED weight
+// mirror (same branch-free quad/pipelined loader as the forward chain).
+// Optionally emits the chain-input gradient restricted to columns
+// >= dx0_lo (the SAC actor step needs only the ACTION columns of
+// d(critic input)) as fp32 per group.  grid (ceil(M/16), 1, G).
+// ---------------------------------------------------------------------------
+struct ChainDxDesc {
+  const u16* wt[6];     // [G, K_l, N_l] n-minor transposed weights
+  const u16* yout[6];   // post-act output of layer l (mask); null if act=0
+  long yo_gs[6];        // per-group element stride of yout
+  u16* dysave[6];       // masked dy_l [G, M, N_l] (null = skip)
+  int N[6];
+  int act[6];
+  int L;
+  int K0;
+  int dx0_lo;           // -1 = no input gradient
+};
+
+__global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
+    const u16* __restrict__ dy_last, ChainDxDesc d,
+    float* __restrict__ dx0, int M) {
+  constexpr int TMv = 16;
+  __shared__ u16 sd[2][TMv][CPAD];
+  const int g = blockIdx.z;
+  const int m0 = blockIdx.x * TMv;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wid = tid >> 6;
+  const int fi = lane & 15, fk = lane >> 4;
+  const int rowlim = (M - m0 < TMv ? M - m0 : TMv);
+
+  {
+    unsigned* p = (unsigned*)sd;
+    for (int i = tid; i < 2 * TMv * CPAD / 2; i += NTHR) p[i] = 0u;
+  }
+  __syncthreads();
+
+  // load dy_last tile
+  const int NL = d.N[d.L - 1];
+  {
+    const int r = tid >> 5;
+    const int lc = tid & 31;
+    const u16* src = dy_last + (long)g * M * NL;
+    for (int rr = r; rr < rowlim; rr += NTHR / 32)
+      for (int c = lc; c < NL; c += 32)
+        sd[0][rr][c] = src[(long)(m0 + rr) * NL + c];
+  }
+  __syncthreads();
+
+  int cur = 0;
+  int prevw0 = NL, prevw1 = CMAX;
+  for (int li = d.L - 1; li >= 0; --li) {
+    const int N = d.N[li];
+    const int K = li > 0 ? d.N[li - 1] : d.K0;
+
+    // ---- mask by relu' + save masked dy --------------------------------
+    {
+      const int r = tid >> 5;
+      const int lc = tid & 31;
+      const u16* yo = d.act[li] && d.yout[li] != nullptr
+          ? d.yout[li] + (long)g * d.yo_gs[li] : nullptr;
+      u16* sv = d.dysave[li] != nullptr
+          ? d.dysave[li] + (long)g * M * N : nullptr;
+      for (int rr = r; rr < rowlim; rr += NTHR / 32) {
+        const long row = m0 + rr;
+        for (int c = lc; c < N; c += 32) {
+          u16 v = sd[cur][rr][c];
+          if (yo != nullptr && yo[row * N + c] == 0) v = 0;
+          sd[cur][rr][c] = v;
+          if (sv != nullptr) sv[row * N + c] = v;
+        }
+      }
+    }
+    __syncthreads();
+
+    const bool want_dx0 = li == 0 && d.dx0_lo >= 0;
+    if (li == 0 && !want_dx0) break;
+
+    // ---- GEMM: out[16, K] = dy_pre[16, N] @ Wt[K, N] -------------------
+    const u16* wtg = d.wt[li] + (long)g * N * K;   // [K][N]
+    const int col_lo = want_dx0 ? d.dx0_lo : 0;
+    const int t_lo = col_lo >> 4;
+    const int ntiles = (K + 15) >> 4;
+    const int nb = cur ^ 1;
+    const bool n_aligned = (N % 8) == 0;
+    const int nbody = n_aligned ? (N & ~31) : 0;
+    float* dx0g = want_dx0
+        ? dx0 + (long)g * M * (K - col_lo) : nullptr;
+
+    for (int t0 = t_lo + wid * 4; t0 < ntiles; t0 += 4 * (NTHR / 64)) {
+      const int nq = (ntiles - t0 < 4) ? (ntiles - t0) : 4;
+      f32x4 acc[1][4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[0][j] = f32x4{};
+      long rb[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = (t0 + (j < nq ? j : nq - 1)) * 16 + fi;   // k index
+        if (row >= K) row = K - 1;
+        rb[j] = (long)row * N;
+      }
+
+#define LOADQ(dst, kk)                                                    \
+      _Pragma("unroll") for (int j = 0; j < 4; ++j)                       \
+        dst[j] = *(const bf16x8*)&wtg[rb[j] + fk * 8 + (kk)];
+#define MF4(bset, kk)                                                     \
+      {                                                                   \
+        const bf16x8 a_ =                                                 \
+            *(const bf16x8*)&sd[cur][fi][(kk) + fk * 8];                  \
+        _Pragma("unroll") for (int j = 0; j < 4; ++j)                     \
+          acc[0][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(            \
+              a_, bset[j], acc[0][j], 0, 0, 0);                           \
+      }
+
+      if (n_aligned) {
+        if (nbody >= 32) {
+          bf16x8 S0[4], S1[4], S2[4];
+          LOADQ(S0, 0)
+          if (32 < nbody) LOADQ(S1, 32)
+          int k = 0;
+          while (k + 96 <= nbody) {
+            LOADQ(S2, k + 64)
+            MF4(S0, k)
+            if (k + 96 < nbody) {
+              LOADQ(S0, k + 96)
+            }
+            MF4(S1, k + 32)
+            if (k + 128 < nbody) {
+              LOADQ(S1, k + 128)
+            }
+            MF4(S2, k + 64)
+            k += 96;
+          }
+          if (nbody - k == 32) {
+            MF4(S0, k)
+          } else if (nbody - k == 64) {
+            MF4(S0, k)
+            MF4(S1, k + 32)
+          }
+        }
+        if (nbody < N) {
+          const int k8 = nbody + fk * 8;
+          const int k8c = (k8 + 8 <= N) ? k8 : (N - 8);
+          bf16x8 B[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            B[j] = *(const bf16x8*)&wtg[rb[j] + k8c];
+          MF4(B, nbody)
+        }
+      } else {
+        // tiny-N head layers (N = 1 or 8): guarded single pass
+        for (int k = 0; k < N; k += 32) {
+          const int k8 = k + fk * 8;
+          bf16x8 B[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            B[j] = (j < nq)
+                ? load_bfrag(wtg, t0 * 16 + j * 16 + fi, K, N, k8)
+                : bf16x8{};
+          MF4(B, k)
+        }
+      }
+#undef LOADQ
+#undef MF4
+
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (q >= nq) break;
+        const f32x4 a = acc[0][q];
+        const int col = (t0 + q) * 16 + fi;   // k index
+        if (col < K) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int row = fk * 4 + r;
+            if (want_dx0) {
+              if (row < rowlim && col >= col_lo)
+                dx0g[(long)(m0 + row) * (K - col_lo) + (col - col_lo)] =
+                    a[r];
+            } else {
+              sd[nb][row][col] = f32_to_bf16_rne3(a[r]);
+            }
+          }
+        }
+      }
+    }
+    if (want_dx0) break;
+    {
+      const int pw = nb ? prevw1 : prevw0;
+      for (int i = tid; i < TMv * (pw > K ? pw - K : 0); i += NTHR) {
+        const int r = i / (pw - K), c = K + i % (pw - K);
+        sd[nb][r][c] = 0;
+      }
+      if (nb) prevw1 = K; else prevw0 = K;
+    }
+    __syncthreads();
+    cur = nb;
+  }
+}
+
+// Host binding: [dy_0 .. dy_{L-1} [, dx0]] = mlp_chain_dx_bf16(
+//     dy_last, wts, youts, K0, acts_flags, G, save_dys, dx0_lo)
+// wts[i]: TRANSPOSED bf16 weights [G*K_i, N_i]; youts[i]: mask tensors
+// (post-act outputs, same [G?,M,N_i] layout) or an empty tensor where
+// act=0.  Returns the per-layer MASKED dy tensors (empty placeholders
+// when save_dys=0) and, when dx0_lo >= 0, the fp32 input gradient
+// restricted to columns >= dx0_lo, per group [G, M, K0-dx0_lo].
+static std::vector<torch::Tensor> mlp_chain_dx_bf16(
+    torch::Tensor dy_last, std::vector<torch::Tensor> wts,
+    std::vector<torch::Tensor> youts, long K0,
+    std::vector<long> acts, long G, long save_dys, long dx0_lo) {
+  CHAIN_CHECK_BF16(dy_last);
+  const int L = (int)wts.size();
+  TORCH_CHECK(L >= 1 && L <= 6 && (int)youts.size() == L
+              && (int)acts.size() == L);
+  auto dyc = dy_last.contiguous();
+  const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+  ChainDxDesc d{};
+  d.L = L;
+  d.K0 = (int)K0;
+  d.dx0_lo = (int)dx0_lo;
+  std::vector<torch::Tensor> keep, out;
+  auto bopts = dyc.options();
+  long K = K0;
+  for (int i = 0; i < L; ++i) {
+    CHAIN_CHECK_BF16(wts[i]);
+    auto wc = wts[i].contiguous();
+    keep.push_back(wc);
+    const long N = wc.numel() / (G * K);
+    TORCH_CHECK(N * G * K == wc.numel(), "wt shape mismatch at layer ", i);
+    TORCH_CHECK(N <= CMAX && K <= CMAX);
+    d.wt[i] = (const u16*)wc.data_ptr();
+    d.N[i] = (int)N;
+    d.act[i] = (int)acts[i];
+    d.yout[i] = nullptr;
+    d.yo_gs[i] = 0;
+    if (acts[i] && youts[i].numel() > 0) {
+      CHAIN_CHECK_BF16(youts[i]);
+      auto yc = youts[i].contiguous();
+      keep.push_back(yc);
+      TORCH_CHECK(yc.numel() == (yc.dim() == 3 ? G : 1) * M * N);
+      d.yout[i] = (const u16*)yc.data_ptr();
+      d.yo_gs[i] = yc.dim() == 3 ? M * N : 0;
+    }
+    d.dysave[i] = nullptr;
+    if (save_dys) {
+      auto t = torch::empty({G, M, N}, bopts);
+      d.dysave[i] = (u16*)t.data_ptr();
+      out.push_back(t);
+    } else {
+      out.push_back(torch::Tensor());
+    }
+    K = N;
+  }
+  TORCH_CHECK(dyc.numel() == G * M * d.N[L - 1]);
+  torch::Tensor dx0;
+  if (dx0_lo >= 0) {
+    TORCH_CHECK(dx0_lo < K0);
+    dx0 = torch::empty({G, M, K0 - dx0_lo},
+                       bopts.dtype(torch::kFloat32));
+    out.push_back(dx0);
+  }
+  dim3 grid((M + 15) / 16, 1, G);
+  hipLaunchKernelGGL(k_bf16_chain_dx, grid, dim3(NTHR), 0, cur_stream3(),
+                     (const u16*)dyc.data_ptr(), d,
+                     dx0_lo >= 0 ? dx0.data_ptr<float>() : nullptr,
+                     (int)M);
+  return out;
+}
+
+// ---------------------------------------------------------------------------
 // Host binding: [y, x_bf16, act_0, ..., act_{L-2}] = mlp_chain_fwd_bf16(
 //     x1, x2_or_empty, ws, bs, act_last, G, out_f32, rm=0, rowcat=0,
 //     save_acts=1)
@@ -415,6 +780,8 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
 }  // namespace chain
 
 void register_chain(pybind11::module_& m) {
+  m.def("transpose_weights_bf16", &chain::transpose_weights_bf16);
+  m.def("mlp_chain_dx_bf16", &chain::mlp_chain_dx_bf16);
   m.def("mlp_chain_fwd_bf16", &chain::mlp_chain_fwd_bf16,
         pybind11::arg("x1"), pybind11::arg("x2"), pybind11::arg("ws"),
         pybind11::arg("bs"), pybind11::arg("act_last"), pybind11::arg("G"),
