@@ -238,6 +238,17 @@ class SACEngine:
             off = self.actor_group.offsets[i]
             self._actor_ws_bf16.append(
                 self._actor_bf16[off:off + w.numel()].view_as(w))
+        # TRANSPOSED weight mirrors for the fused dx-chain kernels
+        # (re-materialized per update by transpose_weights_bf16: critic
+        # pre-step at seg1 for the critic backward, critic post-step in
+        # seg2 for the actor-side dx, actor pre-step at seg1)
+        self._twin_local_wt = [
+            torch.empty(w.shape[0], w.shape[2], w.shape[1],
+                        dtype=torch.bfloat16, device=dev)
+            for w in self._twin_local_bf16]
+        self._actor_wt = [
+            torch.empty(w.shape[1], w.shape[0], dtype=torch.bfloat16,
+                        device=dev) for w in self._actor_ws_bf16]
         self.refresh_bf16()
 
     @staticmethod
@@ -625,8 +636,29 @@ class SACEngine:
         fg_c = self.critic_group.flat_grad
         arena_c, S_c, ch_c = self._dw_arena("critic",
                                             self.critic_group.numel, B)
-        self._mlp_bwd_arena(ext, dy, acts_c, self._twin_local_bf16,
-                            wsg, bsg, fg_c, arena_c, S_c, ch_c, G=2)
+        if chain:
+            nl_c = len(self._twin_local_bf16)
+            # refresh W^T mirrors: critic pre-step (this backward) and
+            # actor pre-step (seg2's actor dW dx walk) — ONE launch
+            ext.transpose_weights_bf16(
+                list(self._twin_local_bf16) + list(self._actor_ws_bf16),
+                list(self._twin_local_wt) + list(self._actor_wt),
+                [2] * nl_c + [1] * len(self._actor_ws_bf16))
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts = [acts_c[i + 1] for i in range(nl_c - 1)] + [empty_h]
+            aflags = [1] * (nl_c - 1) + [0]
+            dys = ext.mlp_chain_dx_bf16(dy, list(self._twin_local_wt),
+                                        youts, acts_c[0].shape[-1],
+                                        aflags, 2, 1, -1)
+            base_c = fg_c.data_ptr()
+            ext.dwdb_grouped_arena(
+                list(dys), [acts_c[i] for i in range(nl_c)], arena_c,
+                [(w.data_ptr() - base_c) // 4 for w in wsg],
+                [(b.data_ptr() - base_c) // 4 for b in bsg],
+                2, S_c, ch_c)
+        else:
+            self._mlp_bwd_arena(ext, dy, acts_c, self._twin_local_bf16,
+                                wsg, bsg, fg_c, arena_c, S_c, ch_c, G=2)
         ext.reduce_arena(arena_c, fg_c, S_c)
         self._dp_st = {
             "states": states, "sa": a_cat[B:], "lp": lp_cat[B:],
@@ -670,36 +702,66 @@ class SACEngine:
         daq, dlp = ext.actor_alpha_loss_bwd2(
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)
-        dy = daq
-        for i in range(nl_c - 1, 0, -1):
-            act = 1 if i < nl_c - 1 else 0
-            yout = acts_f[i + 1] if i < nl_c - 1 else acts_f[i]
-            dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
-                                        yout, act, 2, 0)
-        dxa = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
-                                     acts_f[1] if nl_c > 1 else acts_f[0],
-                                     1 if nl_c > 1 else 0, 2, 1)
-        dsa = dxa[:, states.shape[1]:].float()
+        if self._use_chain:
+            # critic W^T mirrors must reflect the POST-step critic
+            ext.transpose_weights_bf16(list(self._twin_local_bf16),
+                                       list(self._twin_local_wt),
+                                       [2] * nl_c)
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts_f = [acts_f[i + 1] for i in range(nl_c - 1)] + [empty_h]
+            outs = ext.mlp_chain_dx_bf16(
+                daq, list(self._twin_local_wt), youts_f,
+                acts_f[0].shape[-1], [1] * (nl_c - 1) + [0], 2, 0,
+                states.shape[1])
+            dx0 = outs[-1]          # [2, B, A] fp32 (action columns only)
+            dsa = dx0[0] + dx0[1]   # sum over the twin Q heads
+        else:
+            dy = daq
+            for i in range(nl_c - 1, 0, -1):
+                act = 1 if i < nl_c - 1 else 0
+                yout = acts_f[i + 1] if i < nl_c - 1 else acts_f[i]
+                dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                            yout, act, 2, 0)
+            dxa = ext.linear_bwd_dx_bf16(
+                dy, self._twin_local_bf16[0],
+                acts_f[1] if nl_c > 1 else acts_f[0],
+                1 if nl_c > 1 else 0, 2, 1)
+            dsa = dxa[:, states.shape[1]:].float()
         dhead = ext.squashed_gaussian_bwd2(
             dsa, dlp, st["lsr"][B:], st["ls_cat"][B:], st["eps"][B:],
             st["tanh_u"][B:], float(self.actor.k))
-        dy = dhead
         wag = [w.grad for w in st["ws_f32"]]
         bag = [b.grad for b in st["bs_f32"]]
         fg_a = self.actor_group.flat_grad
         arena_a, S_a, ch_a = self._dw_arena("actor",
                                             self.actor_group.numel, B)
         base_a = fg_a.data_ptr()
-        for i in range(nl_a - 1, -1, -1):
-            act = 1 if i < nl_a - 1 else 0
-            yout = (acts_a[i + 1][B:] if i < nl_a - 1 else acts_a[i][B:])
-            ext.linear_bwd_dwdb_arena(
-                dy, acts_a[i][B:], yout, act, 1, arena_a,
-                (wag[i].data_ptr() - base_a) // 4,
-                (bag[i].data_ptr() - base_a) // 4, S_a, ch_a, 0)
-            if i > 0:
-                dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
-                                            yout, act, 1, 1)
+        if self._use_chain:
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts_a = [acts_a[i + 1][B:] for i in range(nl_a - 1)] \
+                + [empty_h]
+            dys_a = ext.mlp_chain_dx_bf16(
+                dhead, list(self._actor_wt), youts_a,
+                acts_a[0].shape[-1], [1] * (nl_a - 1) + [0], 1, 1, -1)
+            ext.dwdb_grouped_arena(
+                list(dys_a), [acts_a[i][B:] for i in range(nl_a)],
+                arena_a,
+                [(w.data_ptr() - base_a) // 4 for w in wag],
+                [(b.data_ptr() - base_a) // 4 for b in bag],
+                1, S_a, ch_a)
+        else:
+            dy = dhead
+            for i in range(nl_a - 1, -1, -1):
+                act = 1 if i < nl_a - 1 else 0
+                yout = (acts_a[i + 1][B:] if i < nl_a - 1
+                        else acts_a[i][B:])
+                ext.linear_bwd_dwdb_arena(
+                    dy, acts_a[i][B:], yout, act, 1, arena_a,
+                    (wag[i].data_ptr() - base_a) // 4,
+                    (bag[i].data_ptr() - base_a) // 4, S_a, ch_a, 0)
+                if i > 0:
+                    dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
+                                                yout, act, 1, 1)
         ext.reduce_arena(arena_a, fg_a, S_a)
         st["al"] = al
 

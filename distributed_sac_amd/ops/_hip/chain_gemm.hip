@@ -322,8 +322,8 @@ struct TransDesc {
   int cum[9];   // cumulative tiles per layer (tiles = G*ceil(N/64)*ceil(K/64))
   int ntx[8];   // ceil(N/64)
   int nty[8];   // ceil(K/64)
+  int Gl[8];    // per-tensor group count
   int L;
-  int G;
 };
 
 __global__ __launch_bounds__(256) void k_bf16_transpose_multi(TransDesc d) {
@@ -335,6 +335,7 @@ __global__ __launch_bounds__(256) void k_bf16_transpose_multi(TransDesc d) {
   const int per_g = d.ntx[l] * d.nty[l];
   const int g = local / per_g;
   local -= g * per_g;
+  (void)d.Gl;
   const int n0 = (local % d.ntx[l]) * 64;
   const int k0 = (local / d.ntx[l]) * 64;
   const int N = d.N[l], K = d.K[l];
@@ -368,12 +369,12 @@ __global__ __launch_bounds__(256) void k_bf16_transpose_multi(TransDesc d) {
 
 static void transpose_weights_bf16(std::vector<torch::Tensor> ws,
                                    std::vector<torch::Tensor> wts,
-                                   long G) {
+                                   std::vector<long> Gs) {
   const int L = (int)ws.size();
-  TORCH_CHECK(L >= 1 && L <= 8 && (int)wts.size() == L);
+  TORCH_CHECK(L >= 1 && L <= 8 && (int)wts.size() == L
+              && (int)Gs.size() == L);
   TransDesc d{};
   d.L = L;
-  d.G = (int)G;
   int cum = 0;
   std::vector<torch::Tensor> keep;
   for (int i = 0; i < L; ++i) {
@@ -382,6 +383,7 @@ static void transpose_weights_bf16(std::vector<torch::Tensor> ws,
     auto wc = ws[i].contiguous();
     keep.push_back(wc);
     TORCH_CHECK(wts[i].is_contiguous());
+    const long G = Gs[i];
     const long NK = wc.numel() / G;
     const long K = wc.size(-1);
     const long N = NK / K;
@@ -390,6 +392,7 @@ static void transpose_weights_bf16(std::vector<torch::Tensor> ws,
     d.wt[i] = (u16*)wts[i].data_ptr();
     d.N[i] = (int)N;
     d.K[i] = (int)K;
+    d.Gl[i] = (int)G;
     d.ntx[i] = (int)((N + 63) / 64);
     d.nty[i] = (int)((K + 63) / 64);
     d.cum[i] = cum;

@@ -58,7 +58,7 @@ def test_transpose():
     ws2 = [bf(400, 53), bf(400, 400), bf(1, 400), bf(8, 400)]
     wt2 = [torch.empty(w.shape[1], w.shape[0], device=dev,
                        dtype=torch.bfloat16) for w in ws2]
-    ext.transpose_weights_bf16(ws2, wt2, 1)
+    ext.transpose_weights_bf16(ws2, wt2, [1] * len(ws2))
     d = max((a.t().contiguous() - b).abs().max().item()
             for a, b in zip(ws2, wt2))
     check(f"transpose 2-D (d={d})", d == 0)
@@ -66,7 +66,7 @@ def test_transpose():
     ws3 = [bf(2, 400, 53), bf(2, 1, 400), bf(2, 400, 400)]
     wt3 = [torch.empty(w.shape[0], w.shape[2], w.shape[1], device=dev,
                        dtype=torch.bfloat16) for w in ws3]
-    ext.transpose_weights_bf16(ws3, wt3, 2)
+    ext.transpose_weights_bf16(ws3, wt3, [2] * len(ws3))
     d = max((a.transpose(1, 2).contiguous() - b).abs().max().item()
             for a, b in zip(ws3, wt3))
     check(f"transpose G=2 (d={d})", d == 0)
@@ -84,7 +84,7 @@ def test_chain_dx():
         w = bf(G, N, K)
         ws.append(w)
         wt = torch.empty(G, K, N, device=dev, dtype=torch.bfloat16)
-        ext.transpose_weights_bf16([w], [wt], G)
+        ext.transpose_weights_bf16([w], [wt], [G])
         wts.append(wt)
         last = i == L - 1
         acts_flags.append(0 if last else 1)
