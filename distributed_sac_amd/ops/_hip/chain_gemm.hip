@@ -90,15 +90,6 @@ __device__ __forceinline__ bf16x8 load_bfrag(const u16* __restrict__ w,
   return bf16x8{};
 }
 
-// Cooperative glds stage of a 32-wide operand slice into LDS rows of
-// 64 B: dest is lane-linear (global_load_lds requirement), the 16-B
-// chunk index is XOR-swizzled on the SOURCE address (chunk ^ (row>>2)&3)
-// so the matching swizzled ds_read_b128 fragments are bank-conflict-free
-// (16 consecutive rows x 4 chunks land on all 16 bank quads).  Source
-// row and k-run are CLAMPED into range: junk lands in unread LDS rows /
-// multiplies the zero-padded A operand.
-#define CHAIN_STAGE32(ldsbase, src, NROWS, ROWLEN, kk, NPASS)               { _Pragma("unroll") for (int p_ = 0; p_ < 4; ++p_) {                          if (p_ < (NPASS)) {                                                         const int idx_ = p_ * NTHR + tid;                                         int srow_ = idx_ >> 2;                                                    if (srow_ >= (NROWS)) srow_ = (NROWS) - 1;                                const int sw_ = (idx_ & 3) ^ ((srow_ >> 2) & 3);                          long gk_ = (kk) + sw_ * 8;                                                if (gk_ + 8 > (ROWLEN)) gk_ = (ROWLEN) - 8;                               auto* gsrc_ = (const __attribute__((address_space(1))) void*)                 ((src) + (long)srow_ * (ROWLEN) + gk_);                               auto* ldst_ = (__attribute__((address_space(3))) void*)                       ((ldsbase) + idx_ * 8);                                               __builtin_amdgcn_global_load_lds(gsrc_, ldst_, 16, 0, 0);               }                                                                     } }
-
 template <int RM>
 __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const void* __restrict__ x1, const void* __restrict__ x2,
@@ -106,13 +97,7 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     u16* __restrict__ xsave,
     ChainFwdDesc d, void* __restrict__ y, int M, int out_f32) {
   constexpr int TMv = 16 * RM;
-  // ONE __shared__ object (a second one makes hipcc drain vmcnt(0)
-  // before every ds_read while a glds is in flight — guide §5 trap 4a):
-  // [0, 2*TMv*CPAD)              activation ping/pong buffers
-  // [2*TMv*CPAD, +2*CMAX*32)     weight k-slice ping/pong (glds dest)
-  __shared__ u16 ldsraw[2 * TMv * CPAD + 2 * CMAX * 32];
-  u16 (*sa)[TMv][CPAD] = (u16(*)[TMv][CPAD])ldsraw;
-  u16* const swb = ldsraw + 2 * TMv * CPAD;   // + buf*CMAX*32
+  __shared__ u16 sa[2][TMv][CPAD];
   const int g = blockIdx.z;
   const int m0 = blockIdx.x * TMv;
   const int tid = threadIdx.x;
@@ -180,71 +165,101 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const bool k_aligned = (K % 8) == 0;
     const int kbody = k_aligned ? (K & ~31) : 0;
 
-    // single quad pass per wave: ntiles <= 32 == 8 waves * 4 (CMAX=512)
-    {
-      const int t0 = wid * 4;
+    for (int t0 = wid * 4; t0 < ntiles; t0 += 4 * (NTHR / 64)) {
       const int nq = (ntiles - t0 < 4) ? (ntiles - t0) : 4;
       f32x4 acc[RM][4];
 #pragma unroll
       for (int m = 0; m < RM; ++m)
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[m][j] = f32x4{};
-      // clamped LDS weight rows per quad slot (junk lanes masked in the
-      // epilogue; out-of-range k killed by the zero-padded A operand)
-      int rwl[4];
+      // Per-lane CLAMPED row bases: surplus quad slots (j >= nq) and
+      // out-of-range rows of a partial-N tile load a VALID row instead
+      // of branching — their MFMA lanes produce garbage that the
+      // epilogue's col<N / q<nq masks never store, and out-of-range K
+      // contributions are killed by the zero-padded LDS A operand.
+      // This keeps EVERY K%8==0 layer (including the N=1/N=8 heads and
+      // the 25th tile of a 400-wide layer) on the branch-free pipelined
+      // path — the guarded loop (a per-load branch + vmcnt drain) made
+      // one straggler wave serialize the whole chain at the layer
+      // barrier.
+      long rb[4];
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        int row = (t0 + (j < (nq > 0 ? nq : 1) ? j
-                         : (nq > 0 ? nq - 1 : 0))) * 16 + fi;
+        int row = (t0 + (j < nq ? j : nq - 1)) * 16 + fi;
         if (row >= N) row = N - 1;
-        if (row < 0) row = 0;
-        rwl[j] = row * 32 + (fk ^ ((row >> 2) & 3)) * 8;
+        rb[j] = (long)row * K;
       }
 
-#define MF4L(bufi, kk)                                                          if (nq > 0) {                                                               _Pragma("unroll") for (int m = 0; m < RM; ++m) {                            const bf16x8 a_ =                                                             *(const bf16x8*)&sa[cur][m * 16 + fi][(kk) + fk * 8];                 _Pragma("unroll") for (int j = 0; j < 4; ++j) {                             const bf16x8 b_ = *(const bf16x8*)&swb[(bufi) * CMAX * 32 + rwl[j]];                     acc[m][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                          a_, b_, acc[m][j], 0, 0, 0);                                        }                                                                       }                                                                       }
+#define LOADQ(dst, kk)                                                    \
+      _Pragma("unroll") for (int j = 0; j < 4; ++j)                       \
+        dst[j] = *(const bf16x8*)&wg[rb[j] + fk * 8 + (kk)];
+#define MF4(bset, kk)                                                     \
+      {                                                                   \
+        _Pragma("unroll") for (int m = 0; m < RM; ++m) {                  \
+          const bf16x8 a_ =                                               \
+              *(const bf16x8*)&sa[cur][m * 16 + fi][(kk) + fk * 8];       \
+          _Pragma("unroll") for (int j = 0; j < 4; ++j)                   \
+            acc[m][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(          \
+                a_, bset[j], acc[m][j], 0, 0, 0);                         \
+        }                                                                 \
+      }
 
       if (k_aligned) {
-        // glds-staged weight k-slices, 2 LDS buffers, one barrier per
-        // step: the NEXT slice's DMA is in flight during this step's
-        // MFMAs and drains at the __syncthreads() (which emits vmcnt(0)
-        // while an LDS-DMA is outstanding).
-        const int npass = (N * 4 + NTHR - 1) / NTHR;
-        const int steps = (K + 31) / 32;
-        int buf = 0;
-        CHAIN_STAGE32(swb, wg, N, K, 0, npass)
-        __syncthreads();
-        for (int st = 0; st < steps; ++st) {
-          if (st + 1 < steps)
-            CHAIN_STAGE32(swb + (buf ^ 1) * CMAX * 32, wg, N, K, (st + 1) * 32, npass)
-          MF4L(buf, st * 32)
-          __syncthreads();
-          buf ^= 1;
+        // unconditional loads, 3-set rotating software pipeline: loads
+        // run TWO k-steps ahead of their MFMAs (12 B-fragments in
+        // flight), hiding cold-L2/L3 weight-miss latency that a 1-deep
+        // pipeline left exposed.
+        if (kbody >= 32) {
+          bf16x8 S0[4], S1[4], S2[4];
+          LOADQ(S0, 0)
+          if (32 < kbody) LOADQ(S1, 32)
+          int k = 0;
+          while (k + 96 <= kbody) {
+            LOADQ(S2, k + 64)
+            MF4(S0, k)
+            if (k + 96 < kbody) {
+              LOADQ(S0, k + 96)
+            }
+            MF4(S1, k + 32)
+            if (k + 128 < kbody) {
+              LOADQ(S1, k + 128)
+            }
+            MF4(S2, k + 64)
+            k += 96;
+          }
+          if (kbody - k == 32) {
+            MF4(S0, k)
+          } else if (kbody - k == 64) {
+            MF4(S0, k)
+            MF4(S1, k + 32)
+          }
+        }
+        if (kbody < K) {
+          // K%32 tail: clamp each lane's k-run into range — lanes whose
+          // true k-run starts past K read duplicate (valid) bytes that
+          // multiply the zero LDS pad.
+          const int k8 = kbody + fk * 8;
+          const int k8c = (k8 + 8 <= K) ? k8 : (K - 8);
+          bf16x8 B[4];
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            B[j] = *(const bf16x8*)&wg[rb[j] + k8c];
+          MF4(B, kbody)
         }
       } else {
-        // odd-K layer (the 49/53-d layer-0 inputs): guarded per-lane
-        // loads — tiny K, one or two steps
-        const int row = t0 * 16 + fi;
+        // odd-K layer (the 49/53-d layer-0 inputs): guarded loads
         for (int k = 0; k < K; k += 32) {
           const int k8 = k + fk * 8;
           bf16x8 B[4];
 #pragma unroll
           for (int j = 0; j < 4; ++j)
-            B[j] = (j < nq) ? load_bfrag(wg, row + j * 16, N, K, k8)
+            B[j] = (j < nq) ? load_bfrag(wg, t0 * 16 + j * 16 + fi, N, K, k8)
                             : bf16x8{};
-          if (nq > 0) {
-#pragma unroll
-            for (int m = 0; m < RM; ++m) {
-              const bf16x8 a_ =
-                  *(const bf16x8*)&sa[cur][m * 16 + fi][k8];
-#pragma unroll
-              for (int j = 0; j < 4; ++j)
-                acc[m][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a_, B[j], acc[m][j], 0, 0, 0);
-            }
-          }
+          MF4(B, k)
         }
       }
-#undef MF4L
+#undef LOADQ
+#undef MF4
 
       // epilogue: C/D map col = lane&15, row = (lane>>4)*4 + r
 #pragma unroll
