@@ -600,8 +600,13 @@ class CAREEngine(SACEngine):
 
         # ---- batched actor head + squash ------------------------------
         ws_f32, bs_f32 = self._actor_weights()
-        out, acts_a = self._mlp_fwd_manual(enc_cat, self._actor_ws_bf16,
-                                           bs_f32)
+        chain = self._use_chain
+        if chain:
+            out, acts_a = self._chain_fwd(enc_cat, None,
+                                          self._actor_ws_bf16, bs_f32, G=1)
+        else:
+            out, acts_a = self._mlp_fwd_manual(enc_cat, self._actor_ws_bf16,
+                                               bs_f32)
         mu, lsr = out[:, :A], out[:, A:]
         if self._eps_queue:
             eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
@@ -616,16 +621,31 @@ class CAREEngine(SACEngine):
         enc_t, _ = self._se_fwd_manual(self._se_target,
                                        next_states[:, :sd].to(torch.bfloat16),
                                        zc16)
-        xt = torch.cat([enc_t, na.to(torch.bfloat16)], dim=-1)
-        q1_t, q2_t, _ = self._twin_fwd_manual(xt, self._twin_target_bf16,
-                                              self._twin_target[1])
+        if chain:
+            yt, _ = self._chain_fwd(enc_t, na, self._twin_target_bf16,
+                                    self._twin_target[1], G=2,
+                                    save_acts=False)
+            q1_t, q2_t = yt[0], yt[1]
+        else:
+            xt = torch.cat([enc_t, na.to(torch.bfloat16)], dim=-1)
+            q1_t, q2_t, _ = self._twin_fwd_manual(xt,
+                                                  self._twin_target_bf16,
+                                                  self._twin_target[1])
         y = ext.td_target_mt(rewards, dones, q1_t, q2_t, nlp, states,
                              la_det, T, self.gamma, self.reward_scale)
 
         # ---- critic loss + manual backward ----------------------------
-        x = torch.cat([enc_cat[B:], actions.to(torch.bfloat16)], dim=-1)
-        q1, q2, acts_q = self._twin_fwd_manual(x, self._twin_local_bf16,
-                                               self._twin_local[1])
+        if chain:
+            yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
+                                         self._twin_local_bf16,
+                                         self._twin_local[1], G=2)
+            q1, q2 = yq[0], yq[1]
+            head_in_dim = acts_q[0].shape[-1]
+        else:
+            x = torch.cat([enc_cat[B:], actions.to(torch.bfloat16)], dim=-1)
+            q1, q2, acts_q = self._twin_fwd_manual(x, self._twin_local_bf16,
+                                                   self._twin_local[1])
+            head_in_dim = x.shape[1]
         closs = ext.critic_loss_fwd(q1, q2, y, states, la_det, T,
                                     int(use_w))[0]
         dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
@@ -635,19 +655,38 @@ class CAREEngine(SACEngine):
         arena_c, S_c, ch_c = self._dw_arena("critic",
                                             self.critic_group.numel, B)
         base_c = fg_c.data_ptr()
-        for i in range(nl_c - 1, -1, -1):
-            act = 1 if i < nl_c - 1 else 0
-            yout = acts_q[i + 1] if i < nl_c - 1 else acts_q[i]
-            ext.linear_bwd_dwdb_arena(
-                dy, acts_q[i], yout, act, 2, arena_c,
-                (wsg[i].grad.data_ptr() - base_c) // 4,
-                (bsg[i].grad.data_ptr() - base_c) // 4, S_c, ch_c, 0)
-            if i > 0:
-                dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
-                                            yout, act, 2, 0)
-        dx0 = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
-                                     acts_q[1] if nl_c > 1 else acts_q[0],
-                                     1 if nl_c > 1 else 0, 2, 1)  # [B,se+A]
+        if chain:
+            ext.transpose_weights_bf16(
+                list(self._twin_local_bf16) + list(self._actor_ws_bf16),
+                list(self._twin_local_wt) + list(self._actor_wt),
+                [2] * nl_c + [1] * nl_a)
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts_q = [acts_q[i + 1] for i in range(nl_c - 1)] + [empty_h]
+            aflags_q = [1] * (nl_c - 1) + [0]
+            outs = ext.mlp_chain_dx_bf16(dy, list(self._twin_local_wt),
+                                         youts_q, head_in_dim, aflags_q,
+                                         2, 1, 0)
+            dys_q, dx0f = outs[:-1], outs[-1]
+            dx0 = (dx0f[0] + dx0f[1]).to(torch.bfloat16)  # [B, se+A]
+            ext.dwdb_grouped_arena(
+                list(dys_q), [acts_q[i] for i in range(nl_c)], arena_c,
+                [(w.grad.data_ptr() - base_c) // 4 for w in wsg],
+                [(b.grad.data_ptr() - base_c) // 4 for b in bsg],
+                2, S_c, ch_c)
+        else:
+            for i in range(nl_c - 1, -1, -1):
+                act = 1 if i < nl_c - 1 else 0
+                yout = acts_q[i + 1] if i < nl_c - 1 else acts_q[i]
+                ext.linear_bwd_dwdb_arena(
+                    dy, acts_q[i], yout, act, 2, arena_c,
+                    (wsg[i].grad.data_ptr() - base_c) // 4,
+                    (bsg[i].grad.data_ptr() - base_c) // 4, S_c, ch_c, 0)
+                if i > 0:
+                    dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                                yout, act, 2, 0)
+            dx0 = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
+                                         acts_q[1] if nl_c > 1 else acts_q[0],
+                                         1 if nl_c > 1 else 0, 2, 1)  # [B,se+A]
         # state-encoder backward (states half of the batched acts)
         zen = se_saved["z_encs"][:, B:].contiguous()     # f32 [k,B,D]
         dzencs, dlogits = ext.attn_pool_bwd(zen, se_saved["alpha"][B:],
@@ -713,25 +752,44 @@ class CAREEngine(SACEngine):
         # ---- actor/alpha loss + manual backward -----------------------
         enc_c, _ = self._se_fwd_manual(info, states[:, :sd].to(torch.bfloat16),
                                        zc16)               # post-step SE
-        xa = torch.cat([enc_c, sa.to(torch.bfloat16)], dim=-1)
-        aq1, aq2, acts_f = self._twin_fwd_manual(xa, self._twin_local_bf16,
-                                                 self._twin_local[1])
+        if chain:
+            ya, acts_f = self._chain_fwd(enc_c, sa, self._twin_local_bf16,
+                                         self._twin_local[1], G=2)
+            aq1, aq2 = ya[0], ya[1]
+        else:
+            xa = torch.cat([enc_c, sa.to(torch.bfloat16)], dim=-1)
+            aq1, aq2, acts_f = self._twin_fwd_manual(xa,
+                                                     self._twin_local_bf16,
+                                                     self._twin_local[1])
         al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls_cat[B:], states,
                                       la_det, T, int(use_w), self.H_bar_f)
         self.alpha_group.flat_grad.zero_()
         daq, dlp = ext.actor_alpha_loss_bwd2(
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)
-        dy = daq
-        for i in range(nl_c - 1, 0, -1):
-            act = 1 if i < nl_c - 1 else 0
-            yout = acts_f[i + 1] if i < nl_c - 1 else acts_f[i]
-            dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
-                                        yout, act, 2, 0)
-        dxa = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
-                                     acts_f[1] if nl_c > 1 else acts_f[0],
-                                     1 if nl_c > 1 else 0, 2, 1)
-        dsa = dxa[:, enc_c.shape[1]:].float()
+        if chain:
+            ext.transpose_weights_bf16(list(self._twin_local_bf16),
+                                       list(self._twin_local_wt),
+                                       [2] * nl_c)
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts_f = [acts_f[i + 1] for i in range(nl_c - 1)] + [empty_h]
+            outs = ext.mlp_chain_dx_bf16(
+                daq, list(self._twin_local_wt), youts_f,
+                acts_f[0].shape[-1], [1] * (nl_c - 1) + [0], 2, 0,
+                enc_c.shape[1])
+            dx0a = outs[-1]
+            dsa = dx0a[0] + dx0a[1]
+        else:
+            dy = daq
+            for i in range(nl_c - 1, 0, -1):
+                act = 1 if i < nl_c - 1 else 0
+                yout = acts_f[i + 1] if i < nl_c - 1 else acts_f[i]
+                dy = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[i],
+                                            yout, act, 2, 0)
+            dxa = ext.linear_bwd_dx_bf16(dy, self._twin_local_bf16[0],
+                                         acts_f[1] if nl_c > 1 else acts_f[0],
+                                         1 if nl_c > 1 else 0, 2, 1)
+            dsa = dxa[:, enc_c.shape[1]:].float()
         dhead = ext.squashed_gaussian_bwd2(
             dsa, dlp, lsr[B:], ls_cat[B:], eps[B:], tanh_u[B:],
             float(self.actor.k))
@@ -740,16 +798,32 @@ class CAREEngine(SACEngine):
         arena_a, S_a, ch_a = self._dw_arena("actor",
                                             self.actor_group.numel, B)
         base_a = fg_a.data_ptr()
-        for i in range(nl_a - 1, -1, -1):
-            act = 1 if i < nl_a - 1 else 0
-            yout = (acts_a[i + 1][B:] if i < nl_a - 1 else acts_a[i][B:])
-            ext.linear_bwd_dwdb_arena(
-                dy, acts_a[i][B:], yout, act, 1, arena_a,
-                (ws_f32[i].grad.data_ptr() - base_a) // 4,
-                (bs_f32[i].grad.data_ptr() - base_a) // 4, S_a, ch_a, 0)
-            if i > 0:
-                dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
-                                            yout, act, 1, 1)
+        if chain:
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts_a = [acts_a[i + 1][B:] for i in range(nl_a - 1)] \
+                + [empty_h]
+            dys_a = ext.mlp_chain_dx_bf16(
+                dhead, list(self._actor_wt), youts_a,
+                acts_a[0].shape[-1], [1] * (nl_a - 1) + [0], 1, 1, -1)
+            ext.dwdb_grouped_arena(
+                list(dys_a), [acts_a[i][B:] for i in range(nl_a)],
+                arena_a,
+                [(w.grad.data_ptr() - base_a) // 4 for w in ws_f32],
+                [(b.grad.data_ptr() - base_a) // 4 for b in bs_f32],
+                1, S_a, ch_a)
+        else:
+            for i in range(nl_a - 1, -1, -1):
+                act = 1 if i < nl_a - 1 else 0
+                yout = (acts_a[i + 1][B:] if i < nl_a - 1
+                        else acts_a[i][B:])
+                ext.linear_bwd_dwdb_arena(
+                    dy, acts_a[i][B:], yout, act, 1, arena_a,
+                    (ws_f32[i].grad.data_ptr() - base_a) // 4,
+                    (bs_f32[i].grad.data_ptr() - base_a) // 4, S_a, ch_a,
+                    0)
+                if i > 0:
+                    dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
+                                                yout, act, 1, 1)
         ext.reduce_arena(arena_a, fg_a, S_a)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self._aa_arena)

@@ -93,7 +93,7 @@ __device__ __forceinline__ bf16x8 load_bfrag(const u16* __restrict__ w,
 template <int RM>
 __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
     const void* __restrict__ x1, const void* __restrict__ x2,
-    int C1, int C2, int x_f32, int rowcat, int M1,
+    int C1, int C2, int x1f, int x2f, int rowcat, int M1,
     u16* __restrict__ xsave,
     ChainFwdDesc d, void* __restrict__ y, int M, int out_f32) {
   constexpr int TMv = 16 * RM;
@@ -126,14 +126,15 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
         if (rowcat) {
           const bool top = row < M1;
           const long rr2 = top ? row : row - M1;
-          v = x_f32 ? ((const float*)(top ? x1 : x2))[rr2 * C1 + c]
-                    : (float)((const __bf16*)(top ? x1 : x2))[rr2 * C1 + c];
+          const int f_ = top ? x1f : x2f;
+          v = f_ ? ((const float*)(top ? x1 : x2))[rr2 * C1 + c]
+                 : (float)((const __bf16*)(top ? x1 : x2))[rr2 * C1 + c];
         } else if (c < C1) {
-          v = x_f32 ? ((const float*)x1)[row * C1 + c]
-                    : (float)((const __bf16*)x1)[row * C1 + c];
+          v = x1f ? ((const float*)x1)[row * C1 + c]
+                  : (float)((const __bf16*)x1)[row * C1 + c];
         } else {
-          v = x_f32 ? ((const float*)x2)[row * C2 + (c - C1)]
-                    : (float)((const __bf16*)x2)[row * C2 + (c - C1)];
+          v = x2f ? ((const float*)x2)[row * C2 + (c - C1)]
+                  : (float)((const __bf16*)x2)[row * C2 + (c - C1)];
         }
         const u16 h = f32_to_bf16_rne3(v);
         sa[0][rr][c] = h;
@@ -695,14 +696,17 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
   TORCH_CHECK(L >= 1 && L <= 6 && (int)bs.size() == L);
   TORCH_CHECK(x1.is_cuda() && x1.dim() == 2);
   const bool has2 = x2.numel() > 0;
-  const bool xf32 = x1.scalar_type() == torch::kFloat32;
+  const bool x1f = x1.scalar_type() == torch::kFloat32;
   TORCH_CHECK(x1.scalar_type() == torch::kFloat32
               || x1.scalar_type() == torch::kBFloat16);
   auto x1c = x1.contiguous();
   torch::Tensor x2c = x2;
+  bool x2f = false;
   if (has2) {
+    x2f = x2.scalar_type() == torch::kFloat32;
     TORCH_CHECK(x2.is_cuda() && x2.dim() == 2
-                && x2.scalar_type() == x1.scalar_type());
+                && (x2.scalar_type() == torch::kFloat32
+                    || x2.scalar_type() == torch::kBFloat16));
     if (rowcat) {
       TORCH_CHECK(x2.size(1) == x1.size(1), "rowcat needs equal widths");
     } else {
@@ -767,14 +771,16 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
     hipLaunchKernelGGL(k_bf16_chain_fwd<2>, grid, dim3(NTHR), 0,
                        cur_stream3(), x1c.data_ptr(),
                        has2 ? x2c.data_ptr() : nullptr, (int)C1, (int)C2,
-                       (int)(xf32 ? 1 : 0), (int)rowcat, (int)M1,
+                       (int)(x1f ? 1 : 0), (int)(x2f ? 1 : 0),
+                       (int)rowcat, (int)M1,
                        (u16*)xsave.data_ptr(), d,
                        y.data_ptr(), (int)M, (int)out_f32);
   else
     hipLaunchKernelGGL(k_bf16_chain_fwd<1>, grid, dim3(NTHR), 0,
                        cur_stream3(), x1c.data_ptr(),
                        has2 ? x2c.data_ptr() : nullptr, (int)C1, (int)C2,
-                       (int)(xf32 ? 1 : 0), (int)rowcat, (int)M1,
+                       (int)(x1f ? 1 : 0), (int)(x2f ? 1 : 0),
+                       (int)rowcat, (int)M1,
                        (u16*)xsave.data_ptr(), d,
                        y.data_ptr(), (int)M, (int)out_f32);
   return out;
