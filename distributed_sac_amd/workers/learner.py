@@ -104,6 +104,11 @@ class Learner:
         self.heartbeat_timeout = heartbeat_timeout
         self._dead_players = set()
         self._hb_check = 0
+        # recovery hook (round 2): the orchestrator owns
+        # the process objects and ring names, so it installs a callback
+        # that respawns a heartbeat-dead player on the SAME ring;
+        # returning True clears the player's dead mark
+        self.on_dead_player = None
 
     def check_heartbeats(self) -> None:
         if self.heartbeat is None:
@@ -121,7 +126,19 @@ class Learner:
                 self._dead_players.add(pid)
                 self.logger.print(
                     f"WARNING: player {pid} heartbeat stale "
-                    f"({now - last:.0f}s) — continuing without it")
+                    f"({now - last:.0f}s)")
+                if self.on_dead_player is not None:
+                    try:
+                        if self.on_dead_player(pid):
+                            self.heartbeat[pid] = now
+                            self._dead_players.discard(pid)
+                            self.logger.print(f"player {pid} respawned")
+                        else:
+                            self.logger.print(
+                                f"player {pid} NOT respawned (limit) — "
+                                "continuing without it")
+                    except Exception as e:  # pragma: no cover
+                        self.logger.print(f"respawn of {pid} failed: {e!r}")
 
     @property
     def dead_players(self):
@@ -286,6 +303,18 @@ class Learner:
             if self.grad_steps % 100 == 0:
                 metrics = {k: float(v) for k, v in metrics_t.items()}
                 self.logger.add_scalars("learner", metrics, self.grad_steps)
+                # per-task temperature curve (reference Logger writes the
+                # full alpha array per update — MT10_Distributed_CARE/src/
+                # logger.py:45-132, learner.py:445-464); one D2H of a
+                # <=num_tasks vector per report tick
+                la = self.engine.log_alpha.detach()
+                if la.numel() > 1:
+                    alphas = la.exp().cpu().tolist()
+                    self.logger.add_scalars(
+                        "alpha",
+                        {f"task_{i}": float(v)
+                         for i, v in enumerate(alphas)},
+                        self.grad_steps)
         return metrics
 
     # -- main loop -----------------------------------------------------

@@ -63,6 +63,7 @@ class DistributedTrainer:
         num_tasks = cfg.num_tasks if cfg.variant in ("mtsac", "care") else 1
         self.partitions = default_task_partition(num_tasks, num_players)
         self.players: List[mp.Process] = []
+        self._respawns = {}
         self.chunk_steps = chunk_steps
         self.seed = seed
         # reference Player(eval_episode_idx=40): periodic deterministic
@@ -92,18 +93,41 @@ class DistributedTrainer:
                                heartbeat=self.heartbeat, rings=self.rings,
                                precision=precision)
 
+    def _spawn_player(self, pid: int, seed_bump: int = 0) -> "mp.Process":
+        p = self.ctx.Process(
+            target=run_player,
+            args=(pid, self.cfg, self.env_fn, self.partitions[pid],
+                  self.snapshot, self.sample_queue, self.log_queue,
+                  self.stop_event, self.chunk_steps,
+                  self.seed + 131 * pid + 7907 * seed_bump, 2,
+                  self.eval_every_episodes, None, self.heartbeat,
+                  self.ring_names[pid] if self.ring_names else None),
+            daemon=True)
+        p.start()
+        return p
+
     def start_players(self) -> None:
-        for pid, tasks in enumerate(self.partitions):
-            p = self.ctx.Process(
-                target=run_player,
-                args=(pid, self.cfg, self.env_fn, tasks, self.snapshot,
-                      self.sample_queue, self.log_queue, self.stop_event,
-                      self.chunk_steps, self.seed + 131 * pid, 2,
-                      self.eval_every_episodes, None, self.heartbeat,
-                      self.ring_names[pid] if self.ring_names else None),
-                daemon=True)
-            p.start()
-            self.players.append(p)
+        for pid in range(len(self.partitions)):
+            self.players.append(self._spawn_player(pid))
+        # recovery (VERDICT round-1 item 8): respawn heartbeat-dead
+        # players on the SAME ring, bounded per player
+        self.learner.on_dead_player = self.respawn_player
+
+    def respawn_player(self, pid: int, max_respawns: int = 5) -> bool:
+        """Replace a dead player process, reusing its ring/partition.
+        Returns True when a fresh process is running."""
+        if pid >= len(self.players):
+            return False
+        n = self._respawns.get(pid, 0)
+        if n >= max_respawns:
+            return False
+        old = self.players[pid]
+        if old.is_alive():
+            old.terminate()
+        old.join(timeout=5.0)
+        self._respawns[pid] = n + 1
+        self.players[pid] = self._spawn_player(pid, seed_bump=n + 1)
+        return True
 
     def run(self, max_grad_steps: Optional[int] = None,
             max_seconds: Optional[float] = None) -> Dict[str, float]:

@@ -136,6 +136,43 @@ def test_learner_ready_min_shard_semantics():
     assert tr.ready()
 
 
+def test_per_task_alpha_logging():
+    """Reference Logger writes the full per-task alpha array
+    (MT10_Distributed_CARE/src/logger.py:45-132); the learner must emit
+    alpha/task_i scalars at each report tick."""
+    import queue
+    from distributed_sac_amd.utils import MetricLogger
+    from distributed_sac_amd.workers.learner import Learner
+    from distributed_sac_amd.workers.param_server import ParamSnapshot
+
+    class Recorder(MetricLogger):
+        def __init__(self):
+            super().__init__(None)
+            self.tags = []
+
+        def add_scalar(self, tag, value, step):
+            self.tags.append(tag)
+
+    from distributed_sac_amd.workers.orchestrator import _actor_numel
+    cfg = tiny_cfg("mtsac")
+    cfg.start_memory_len = 32
+    rec = Recorder()
+    lr = Learner(cfg, "cpu", ParamSnapshot(_actor_numel(cfg)),
+                 queue.Queue(), logger=rec, update_delay=1)
+    for t in range(cfg.num_tasks):
+        lr.replay.append_numpy(
+            states=np.random.randn(64, cfg.mtobs_dim).astype(np.float32),
+            actions=np.zeros((64, cfg.action_dim), dtype=np.float32),
+            rewards=np.zeros(64, dtype=np.float32),
+            next_states=np.random.randn(64, cfg.mtobs_dim).astype(np.float32),
+            dones=np.zeros(64, dtype=np.float32), task_idx=t)
+    for _ in range(100):
+        lr.train_step()
+    assert f"alpha/task_0" in rec.tags
+    assert f"alpha/task_{cfg.num_tasks - 1}" in rec.tags
+    assert "learner/critic_loss" in rec.tags
+
+
 def test_heartbeat_watchdog():
     import time
     import torch as th
@@ -228,3 +265,43 @@ def test_ring_drops_visible_to_consumer():
     assert cons.pop(False)                 # drain one
     assert prod.push(0, *blk)              # space again
     assert cons.dropped() == 2
+
+
+@pytest.mark.timeout(300)
+def test_player_respawn_after_heartbeat_death():
+    """Recovery (VERDICT round-1 item 8): a heartbeat-dead player is
+    respawned by the orchestrator on the same ring and ingest continues."""
+    import time
+
+    cfg = tiny_cfg("sac")
+    cfg.start_memory_len = 64
+    cfg.random_step = 32
+    dt = DistributedTrainer(cfg, device="cpu", num_players=1,
+                            chunk_steps=16, seed=5, use_graph=False)
+    dt.start_players()
+    try:
+        lr = dt.learner
+        # wait for first-generation ingest
+        deadline = time.time() + 60
+        while lr.ingest_count == 0 and time.time() < deadline:
+            lr.drain_queue()
+            time.sleep(0.02)
+        assert lr.ingest_count > 0
+        old_proc = dt.players[0]
+        old_proc.terminate()
+        old_proc.join(timeout=10)
+        # mark the heartbeat stale and let the watchdog fire
+        dt.heartbeat[0] = time.time() - 9999
+        for _ in range(400):
+            lr.check_heartbeats()
+        assert dt.players[0] is not old_proc
+        assert dt._respawns[0] == 1
+        deadline = time.time() + 60
+        before = lr.ingest_count
+        while lr.ingest_count <= before and time.time() < deadline:
+            lr.drain_queue()
+            time.sleep(0.02)
+        assert lr.ingest_count > before, "ingest did not resume"
+        assert dt.players[0].is_alive()
+    finally:
+        dt.shutdown()
