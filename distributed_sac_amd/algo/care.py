@@ -323,6 +323,32 @@ class CAREEngine(SACEngine):
                                 self._target_bf16)
         self._refresh_mixT()
 
+        # original-CARE trainable context encoder (round 2): bf16 mirror
+        # + W^T buffers so the manual path can run the context chain on
+        # the fused chain kernels (fwd + dx + grouped dwdb, third arena)
+        self._ctx_chain = None
+        if self.context_group is not None:
+            import torch.nn as nn_
+            lins = [m for m in self.context_encoder.modules()
+                    if isinstance(m, nn_.Linear)]
+            grp = self.context_group
+            mir = torch.empty(grp.numel, dtype=torch.bfloat16, device=dev)
+            ws16, wts, bss = [], [], []
+            for l in lins:
+                i = next(j for j, q in enumerate(grp.params)
+                         if q is l.weight)
+                off = grp.offsets[i]
+                ws16.append(mir[off:off + l.weight.numel()]
+                            .view_as(l.weight))
+                wts.append(torch.empty(l.weight.shape[1],
+                                       l.weight.shape[0],
+                                       dtype=torch.bfloat16, device=dev))
+                bss.append(l.bias)
+            mir.copy_(grp.flat_data)
+            self.context_encoder_optimizer.bf16_mirror = mir
+            self._ctx_chain = {"mir": mir, "ws16": ws16, "wt": wts,
+                               "bs": bss, "lins": lins}
+
     @torch.no_grad()
     def _refresh_mixT(self, which: str = "all") -> None:
         if not getattr(self, "_se_fast", False):
@@ -338,6 +364,8 @@ class CAREEngine(SACEngine):
     def refresh_bf16(self, which: str = "all") -> None:
         super().refresh_bf16(which)
         self._refresh_mixT(which)
+        if which == "all" and getattr(self, "_ctx_chain", None) is not None:
+            self._ctx_chain["mir"].copy_(self.context_group.flat_data)
 
     def _se_fwd_fast(self, info, mtobss_2d, z_context):
         """stateEncoder.forward via bf16 kernels (value-identical to the
@@ -384,8 +412,11 @@ class CAREEngine(SACEngine):
         pass over cat(next, states) like the SAC engine (the actor-side
         state encoder is no-grad in both halves)."""
         import os as _os
-        if (getattr(self, "_se_fast", False) and self.use_modified_care
-                and _os.environ.get("DSAC_NO_MANUAL", "0") != "1"):
+        manual_ok = getattr(self, "_se_fast", False) and (
+            self.use_modified_care
+            or (getattr(self, "_ctx_chain", None) is not None
+                and self._use_chain))
+        if manual_ok and _os.environ.get("DSAC_NO_MANUAL", "0") != "1":
             return self._update_tensors_manual(batch)
         from ..ops import native
         states, actions = batch["states"], batch["actions"]
@@ -540,6 +571,13 @@ class CAREEngine(SACEngine):
                 if not last:
                     acts_t.append(h)
         alpha, z_enc16 = ext.attn_pool_fwd(h, z_encs)
+        if info["mlpctx"] is None:
+            # original CARE: no mlp_context — enc = cat([z_context, z_enc])
+            enc = torch.cat([zc16, z_enc16], dim=1)
+            if save:
+                return enc, dict(acts_m=acts_m, z_encs=z_encs,
+                                 acts_t=acts_t, alpha=alpha, acts_c=None)
+            return enc, None
         cws, cbs, cwsh = info["mlpctx"]
         if info["ctx_narrow"]:
             res = ext.mlp_narrow_fwd_bf16(zc16, cwsh, list(cbs), 1, 0, 0, sv)
@@ -583,13 +621,30 @@ class CAREEngine(SACEngine):
         A = self.actor.action_dim
         la_det = self.log_alpha.detach()
         sd = states.shape[1] - T                 # raw state dim
-        zc_dim = self._se_local["mlpctx"][0][-1].shape[0]
         D = self._se_local["mixW"][-1].shape[2]
         info = self._se_local
         nl_c = len(self._twin_local_bf16)
         nl_a = len(self._actor_ws_bf16)
+        orig = not self.use_modified_care
 
-        z_context = self.context_encoder(states)         # frozen embedding
+        if orig:
+            # original CARE: TRAINABLE context encoder (embedding header +
+            # mlp) — forward on the fused chain kernel over the gathered
+            # frozen embeddings; activations saved for the third-arena
+            # backward below (reference MT1…/learner.py: context grads
+            # come from the critic loss only)
+            cc = self._ctx_chain
+            idx = states[:, -T:].argmax(dim=1)
+            emb = torch.relu(self.context_encoder.embedding[0](idx))
+            z_context, acts_ctx = self._chain_fwd(emb, None, cc["ws16"],
+                                                  cc["bs"], G=1,
+                                                  act_last=0, out_f32=True)
+        else:
+            z_context = self.context_encoder(states)     # frozen embedding
+            acts_ctx = None
+        zc_dim = (self._se_local["mlpctx"][0][-1].shape[0]
+                  if self._se_local["mlpctx"] is not None
+                  else z_context.shape[1])
         zc16 = z_context.to(torch.bfloat16)
         z2 = torch.cat([zc16, zc16], dim=0)
 
@@ -691,7 +746,9 @@ class CAREEngine(SACEngine):
         zen = se_saved["z_encs"][:, B:].contiguous()     # f32 [k,B,D]
         dzencs, dlogits = ext.attn_pool_bwd(zen, se_saved["alpha"][B:],
                                             dx0, dx0.shape[1], zc_dim)
-        cws, cbs, cwsh = info["mlpctx"]
+        mlpctx = info["mlpctx"]
+        if mlpctx is not None:
+            cws, cbs, cwsh = mlpctx
         tws, tbs, twsh = info["trunk"]
         acts_m = ([se_saved["acts_m"][0][B:]]
                   + [a[:, B:].contiguous() for a in se_saved["acts_m"][1:]])
@@ -699,7 +756,7 @@ class CAREEngine(SACEngine):
         use_fused_se_bwd = (
             _os.environ.get("DSAC_NARROW_BWD", "1") == "1"
             and info["mix_narrow"] and info["trunk_narrow"]
-            and info["ctx_narrow"])
+            and (mlpctx is None or info["ctx_narrow"]))
         if use_fused_se_bwd:
             # DEFAULT ON since round 2: GPU-validated (exact parameter
             # agreement vs the per-layer arena path, +3.8% eager rate —
@@ -713,10 +770,11 @@ class CAREEngine(SACEngine):
                 return [(p_.grad.data_ptr() - base) // 4 for p_ in ps]
             arena2, S2, _ = self._dw_arena("se@rowblocks",
                                            self.critic_group.numel, B)
-            ext.mlp_narrow_bwd_bf16(dx0[:, :zc_dim].contiguous(),
-                                    [a[B:] for a in se_saved["acts_c"]],
-                                    cwsh, arena2, offs(cws), offs(cbs),
-                                    1, 0)
+            if mlpctx is not None:
+                ext.mlp_narrow_bwd_bf16(dx0[:, :zc_dim].contiguous(),
+                                        [a[B:] for a in se_saved["acts_c"]],
+                                        cwsh, arena2, offs(cws), offs(cbs),
+                                        1, 0)
             ext.mlp_narrow_bwd_bf16(dlogits,
                                     [a[B:] for a in se_saved["acts_t"]],
                                     twsh, arena2, offs(tws), offs(tbs),
@@ -728,11 +786,13 @@ class CAREEngine(SACEngine):
             ext.reduce_arena(arena2, fg_c, S2, 0, se_n)
             ext.reduce_arena(arena_c, fg_c, S_c, se_n, -1)
         else:
-            self._mlp_bwd_arena(ext, dx0[:, :zc_dim].contiguous(),
-                                [a[B:] for a in se_saved["acts_c"]], cwsh,
-                                [w.grad for w in cws],
-                                [b.grad for b in cbs],
-                                fg_c, arena_c, S_c, ch_c)
+            if mlpctx is not None:
+                self._mlp_bwd_arena(ext, dx0[:, :zc_dim].contiguous(),
+                                    [a[B:] for a in se_saved["acts_c"]],
+                                    cwsh,
+                                    [w.grad for w in cws],
+                                    [b.grad for b in cbs],
+                                    fg_c, arena_c, S_c, ch_c)
             self._mlp_bwd_arena(ext, dlogits,
                                 [a[B:] for a in se_saved["acts_t"]], twsh,
                                 [w.grad for w in tws],
@@ -744,6 +804,34 @@ class CAREEngine(SACEngine):
                                 fg_c, arena_c, S_c, ch_c,
                                 G=info["k"], transpose_w=1)
             ext.reduce_arena(arena_c, fg_c, S_c)
+        if orig:
+            # context-encoder backward: the only z_context gradient source
+            # is the critic-loss concat half of dx0 (reference
+            # retain_graph rule) — fused dx chain + grouped dwdb into the
+            # context arena, Adam stepped at the end of the update
+            cc = self._ctx_chain
+            nlx = len(cc["ws16"])
+            ext.transpose_weights_bf16(list(cc["ws16"]), list(cc["wt"]),
+                                       [1] * nlx)
+            dzc = dx0[:, :zc_dim].contiguous()
+            empty_h = states.new_empty(0, dtype=torch.bfloat16)
+            youts_x = [acts_ctx[i + 1] for i in range(nlx - 1)] + [empty_h]
+            dys_x = ext.mlp_chain_dx_bf16(
+                dzc, list(cc["wt"]), youts_x, acts_ctx[0].shape[-1],
+                [1] * (nlx - 1) + [0], 1, 1, -1)
+            fgx = self.context_group.flat_grad
+            self.context_group.rebind_grads()
+            basex = fgx.data_ptr()
+            arena_x, S_x, ch_x = self._dw_arena(
+                "ctx", self.context_group.numel, B)
+            ext.dwdb_grouped_arena(
+                list(dys_x), [acts_ctx[i] for i in range(nlx)], arena_x,
+                [(l.weight.grad.data_ptr() - basex) // 4
+                 for l in cc["lins"]],
+                [(l.bias.grad.data_ptr() - basex) // 4
+                 for l in cc["lins"]],
+                1, S_x, ch_x)
+            ext.reduce_arena(arena_x, fgx, S_x)
         if self.ddp is not None:
             self.ddp.allreduce_grad_(self.critic_group.flat_grad)
         self.critic_optimizer.step()   # adam kernel refreshes flat mirror
@@ -832,6 +920,10 @@ class CAREEngine(SACEngine):
 
         self._polyak_targets(mirror=self._target_bf16)
         self._refresh_mixT("target")
+        if orig:
+            if self.ddp is not None:
+                self.ddp.allreduce_grad_(self.context_group.flat_grad)
+            self.context_encoder_optimizer.step()   # refreshes ctx mirror
         self.tie_actor_state_encoder()
         return {"critic_loss": closs[0] + closs[1],
                 "actor_loss": al[0],

@@ -22,7 +22,7 @@
 // MFMA step.  RM=2 (32 rows/WG) halves the total L2 weight re-read
 // (traffic = M/TM * layer bytes) at the cost of half the workgroups —
 // the host picks RM by M.  A-fragments come from the LDS activation
-// buffer: row stride 520 u16 = 260 dwords = 4 banks mod 64, so the
+// buffer: row stride 776 u16 = 388 dwords = 4 banks mod 64, so the
 // 16-lane b128 read groups touch all 64 banks exactly once —
 // conflict-free without padding tricks (see cdna_hip_programming.md §2).
 //
@@ -42,8 +42,11 @@ using f32x4 = __attribute__((__vector_size__(4 * sizeof(float)))) float;
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using u16 = unsigned short;
 
-static constexpr int CMAX = 512;  // max layer width / input width
-static constexpr int CPAD = 520;  // LDS row stride (u16): 260 dwords ≡ 4 mod 64
+static constexpr int CMAX = 768;  // max layer width / input width
+                                  // (768 = the RoBERTa context width)
+static constexpr int CPAD = 776;  // LDS row stride (u16): 388 dwords ≡ 4 mod 64
+                                  // -> 16-lane b128 A-fragment groups hit
+                                  // all 64 banks exactly once
 static constexpr int NTHR = 512;  // 8 waves
 
 #define CHAIN_CHECK_BF16(t) \

@@ -526,3 +526,36 @@ def test_care_evaluate_checkpoint_roundtrip(tmp_path):
     assert out["update_iteration"] == 7
     assert out["episodes"] == 2
     assert np.isfinite(out["mean_reward"])
+
+
+@pytest.mark.gpu
+def test_original_care_manual_vs_autograd_gpu(tmp_path, monkeypatch):
+    """Round-2: the original-CARE manual path (trainable context encoder
+    through the fused chain kernels, third arena) must agree with the
+    bf16 autograd path — including the context-encoder parameters."""
+    torch.manual_seed(0)
+    cfg = care_cfg(tmp_path, modified=False)
+    e1 = CAREEngine(cfg, "cuda:0", precision="bf16")   # autograd
+    e2 = CAREEngine(cfg, "cuda:0", precision="bf16")   # manual
+    e2.load_checkpoint_state(e1.checkpoint_state())
+    assert e1.context_group is not None
+    B, A = cfg.batch_size, cfg.action_dim
+    for step in range(3):
+        batch = {k: v.cuda() for k, v in care_batch(cfg, seed=step).items()}
+        eps = [torch.randn(B, A, device="cuda") for _ in range(2)]
+        for e in (e1, e2):
+            e._eps_queue = [t.clone() for t in eps]
+        monkeypatch.setenv("DSAC_NO_MANUAL", "1")
+        m1 = e1.update({k: v.clone() for k, v in batch.items()})
+        monkeypatch.setenv("DSAC_NO_MANUAL", "0")
+        m2 = e2.update(batch)
+        torch.cuda.synchronize()
+        assert abs(float(m1["critic_loss"]) - float(m2["critic_loss"])) < 5e-3
+    for name, g1, g2 in (("critic", e1.critic_group, e2.critic_group),
+                         ("actor", e1.actor_group, e2.actor_group),
+                         ("alpha", e1.alpha_group, e2.alpha_group),
+                         ("context", e1.context_group, e2.context_group)):
+        d = (g1.flat_data - g2.flat_data).abs().max().item()
+        assert d < 3e-3, f"{name} param drift {d}"
+    # context params actually moved on the manual path
+    assert e2.context_group.flat_data.abs().max() > 0
