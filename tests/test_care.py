@@ -550,7 +550,8 @@ def test_original_care_manual_vs_autograd_gpu(tmp_path, monkeypatch):
         monkeypatch.setenv("DSAC_NO_MANUAL", "0")
         m2 = e2.update(batch)
         torch.cuda.synchronize()
-        assert abs(float(m1["critic_loss"]) - float(m2["critic_loss"])) < 5e-3
+        c1, c2 = float(m1["critic_loss"]), float(m2["critic_loss"])
+        assert abs(c1 - c2) < 5e-3 + 1e-4 * abs(c1)
     for name, g1, g2 in (("critic", e1.critic_group, e2.critic_group),
                          ("actor", e1.actor_group, e2.actor_group),
                          ("alpha", e1.alpha_group, e2.alpha_group),
