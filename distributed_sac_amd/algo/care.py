@@ -525,11 +525,17 @@ class CAREEngine(SACEngine):
 
     # -- hand-rolled backward (modified CARE, bf16) ----------------------
     @torch.no_grad()
-    def _se_fwd_manual(self, info, states_bf16, zc16, save=False):
+    def _se_fwd_manual(self, info, states_bf16, zc16, save=False, rep=1):
         """stateEncoder forward on raw kernels, optionally saving the
         activations the manual backward needs.  Value-identical to
         :meth:`_se_fwd_fast` up to bf16 rounding (the attention pool runs
-        in one fused kernel, fp32 accumulate)."""
+        in one fused kernel, fp32 accumulate).
+
+        ``rep=2``: the batched [next | current] call — ``zc16`` carries
+        ONE copy of z_context (B rows) and the trunk / context
+        projections run once on it (round-1 computed them on duplicated
+        2B rows); the fused pool indexes alpha/hc by ``row % B`` and
+        writes the CONCATENATED head input directly (no torch.cat)."""
         from ..ops import native
         ext = native()
         k = info["k"]
@@ -570,13 +576,13 @@ class CAREEngine(SACEngine):
                                             1 if last else 0)
                 if not last:
                     acts_t.append(h)
-        alpha, z_enc16 = ext.attn_pool_fwd(h, z_encs)
         if info["mlpctx"] is None:
-            # original CARE: no mlp_context — enc = cat([z_context, z_enc])
-            enc = torch.cat([zc16, z_enc16], dim=1)
+            # original CARE: no mlp_context — enc = [z_context | z_enc]
+            alpha, enc = ext.attn_pool_fwd_enc(h, z_encs, zc16, rep)
             if save:
                 return enc, dict(acts_m=acts_m, z_encs=z_encs,
-                                 acts_t=acts_t, alpha=alpha, acts_c=None)
+                                 acts_t=acts_t, alpha=alpha, acts_c=None,
+                                 rep=rep)
             return enc, None
         cws, cbs, cwsh = info["mlpctx"]
         if info["ctx_narrow"]:
@@ -594,10 +600,10 @@ class CAREEngine(SACEngine):
                                              0 if last else 1, 1, 0)
                 if not last:
                     acts_c.append(hc)
-        enc = torch.cat([hc, z_enc16], dim=1)    # bf16 [M, zc+D]
+        alpha, enc = ext.attn_pool_fwd_enc(h, z_encs, hc, rep)
         if save:
             return enc, dict(acts_m=acts_m, z_encs=z_encs, acts_t=acts_t,
-                             alpha=alpha, acts_c=acts_c)
+                             alpha=alpha, acts_c=acts_c, rep=rep)
         return enc, None
 
     @torch.no_grad()
@@ -646,12 +652,12 @@ class CAREEngine(SACEngine):
                   if self._se_local["mlpctx"] is not None
                   else z_context.shape[1])
         zc16 = z_context.to(torch.bfloat16)
-        z2 = torch.cat([zc16, zc16], dim=0)
 
-        # ---- ONE batched local-SE forward over [next | current] -------
+        # ---- ONE batched local-SE forward over [next | current]; the
+        # trunk / context projections run ONCE on the (identical) z rows
         x_cat = torch.cat([next_states[:, :sd], states[:, :sd]], dim=0)
         enc_cat, se_saved = self._se_fwd_manual(info, x_cat.to(torch.bfloat16),
-                                                z2, save=True)
+                                                zc16, save=True, rep=2)
 
         # ---- batched actor head + squash ------------------------------
         ws_f32, bs_f32 = self._actor_weights()
@@ -744,7 +750,10 @@ class CAREEngine(SACEngine):
                                          1 if nl_c > 1 else 0, 2, 1)  # [B,se+A]
         # state-encoder backward (states half of the batched acts)
         zen = se_saved["z_encs"][:, B:].contiguous()     # f32 [k,B,D]
-        dzencs, dlogits = ext.attn_pool_bwd(zen, se_saved["alpha"][B:],
+        dedup = se_saved.get("rep", 1) == 2
+        alpha_cur = (se_saved["alpha"] if dedup
+                     else se_saved["alpha"][B:])
+        dzencs, dlogits = ext.attn_pool_bwd(zen, alpha_cur,
                                             dx0, dx0.shape[1], zc_dim)
         mlpctx = info["mlpctx"]
         if mlpctx is not None:
@@ -770,13 +779,15 @@ class CAREEngine(SACEngine):
                 return [(p_.grad.data_ptr() - base) // 4 for p_ in ps]
             arena2, S2, _ = self._dw_arena("se@rowblocks",
                                            self.critic_group.numel, B)
+            def cur_half(acts):
+                return acts if dedup else [a[B:] for a in acts]
             if mlpctx is not None:
                 ext.mlp_narrow_bwd_bf16(dx0[:, :zc_dim].contiguous(),
-                                        [a[B:] for a in se_saved["acts_c"]],
+                                        cur_half(se_saved["acts_c"]),
                                         cwsh, arena2, offs(cws), offs(cbs),
                                         1, 0)
             ext.mlp_narrow_bwd_bf16(dlogits,
-                                    [a[B:] for a in se_saved["acts_t"]],
+                                    cur_half(se_saved["acts_t"]),
                                     twsh, arena2, offs(tws), offs(tbs),
                                     1, 0)
             ext.mlp_narrow_bwd_bf16(dzencs, acts_m, info["mixT"], arena2,
@@ -786,15 +797,17 @@ class CAREEngine(SACEngine):
             ext.reduce_arena(arena2, fg_c, S2, 0, se_n)
             ext.reduce_arena(arena_c, fg_c, S_c, se_n, -1)
         else:
+            def cur_half(acts):
+                return acts if dedup else [a[B:] for a in acts]
             if mlpctx is not None:
                 self._mlp_bwd_arena(ext, dx0[:, :zc_dim].contiguous(),
-                                    [a[B:] for a in se_saved["acts_c"]],
+                                    cur_half(se_saved["acts_c"]),
                                     cwsh,
                                     [w.grad for w in cws],
                                     [b.grad for b in cbs],
                                     fg_c, arena_c, S_c, ch_c)
             self._mlp_bwd_arena(ext, dlogits,
-                                [a[B:] for a in se_saved["acts_t"]], twsh,
+                                cur_half(se_saved["acts_t"]), twsh,
                                 [w.grad for w in tws],
                                 [b.grad for b in tbs],
                                 fg_c, arena_c, S_c, ch_c)

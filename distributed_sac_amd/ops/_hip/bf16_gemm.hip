@@ -787,21 +787,45 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
 //      mul/sum/softmax-backward kernels autograd records.
 // One wavefront per batch row; requires D <= 64 and E <= 16 (cfg: 50, 6).
 // ---------------------------------------------------------------------------
+// enc_out mode (round 2): logits/alpha/hc have Mz = M/rep rows (the
+// trunk and context projections run ONCE on deduplicated z_context rows
+// — the batched [next|current] SE forward duplicated them); the pooled
+// output row r uses alpha[r % Mz] and writes the CONCATENATED head input
+// enc[r] = [hc[r % Mz] | pool(r)] directly (the torch.cat launch is
+// gone).  Legacy mode (enc_out == null): rep == 1, emit z_enc only.
 __global__ __launch_bounds__(256) void k_attn_pool_fwd(
     const float* __restrict__ logits, const float* __restrict__ z_encs,
     float* __restrict__ alpha_out, u16* __restrict__ z_enc_out,
-    int M, int E, int D) {
+    int M, int E, int D, int rep, const u16* __restrict__ hc, int zc,
+    u16* __restrict__ enc_out) {
   const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
   const int lane = threadIdx.x & 63;
   if (row >= M) return;
+  const int Mz = M / rep;
+  const int zrow = row % Mz;
   float a[16];
   float mx = -1e30f;
-  for (int e = 0; e < E; ++e) { a[e] = logits[(long)row * E + e]; mx = fmaxf(mx, a[e]); }
+  for (int e = 0; e < E; ++e) { a[e] = logits[(long)zrow * E + e]; mx = fmaxf(mx, a[e]); }
   float den = 0.f;
   for (int e = 0; e < E; ++e) { a[e] = __expf(a[e] - mx); den += a[e]; }
   const float inv = 1.f / den;
   for (int e = 0; e < E; ++e) a[e] *= inv;
-  if (lane < E) alpha_out[(long)row * E + lane] = a[lane];
+  if (lane < E && row == zrow) alpha_out[(long)zrow * E + lane] = a[lane];
+  if (enc_out != nullptr) {
+    const int W = zc + D;
+    for (int c = lane; c < W; c += 64) {
+      if (c < zc) {
+        enc_out[(long)row * W + c] = hc[(long)zrow * zc + c];
+      } else {
+        const int d = c - zc;
+        float acc = 0.f;
+        for (int e = 0; e < E; ++e)
+          acc += a[e] * z_encs[((long)e * M + row) * D + d];
+        enc_out[(long)row * W + c] = f32_to_bf16_rne(acc);
+      }
+    }
+    return;
+  }
   for (int d = lane; d < D; d += 64) {
     float acc = 0.f;
     for (int e = 0; e < E; ++e)
@@ -1004,8 +1028,37 @@ static std::vector<torch::Tensor> attn_pool_fwd(torch::Tensor logits,
   hipLaunchKernelGGL(k_attn_pool_fwd, dim3((M + 3) / 4), dim3(256), 0,
                      cur_stream2(), lc.data_ptr<float>(), zc.data_ptr<float>(),
                      alpha.data_ptr<float>(), (u16*)z_enc.data_ptr(),
-                     (int)M, (int)E, (int)D);
+                     (int)M, (int)E, (int)D, 1, nullptr, 0, nullptr);
   return {alpha, z_enc};
+}
+
+// pool + concat fused, with the trunk logits / context projection hc
+// computed ONCE on deduplicated rows (rep = M / logits_rows):
+//   enc[r] = [ hc[r % Mz] | sum_e alpha[r % Mz][e] * z_encs[e][r] ]
+static std::vector<torch::Tensor> attn_pool_fwd_enc(torch::Tensor logits,
+                                                    torch::Tensor z_encs,
+                                                    torch::Tensor hc,
+                                                    long rep) {
+  CHECK_F32(logits); CHECK_F32(z_encs); CHECK_BF16(hc);
+  auto lc = logits.contiguous();
+  auto zz = z_encs.contiguous();
+  auto hcc = hc.contiguous();
+  const long Mz = lc.size(0), E = lc.size(1), D = zz.size(2);
+  const long M = Mz * rep;
+  const long zcw = hcc.size(1);
+  TORCH_CHECK(zz.size(0) == E && zz.size(1) == M && E <= 16 && D <= 64,
+              "attn_pool_fwd_enc: need z_encs (E, Mz*rep, D), E<=16, D<=64");
+  TORCH_CHECK(hcc.size(0) == Mz);
+  auto alpha = torch::empty({Mz, E}, lc.options());
+  auto enc = torch::empty({M, zcw + D},
+                          lc.options().dtype(torch::kBFloat16));
+  hipLaunchKernelGGL(k_attn_pool_fwd, dim3((M + 3) / 4), dim3(256), 0,
+                     cur_stream2(), lc.data_ptr<float>(),
+                     zz.data_ptr<float>(), alpha.data_ptr<float>(),
+                     nullptr, (int)M, (int)E, (int)D, (int)rep,
+                     (const u16*)hcc.data_ptr(), (int)zcw,
+                     (u16*)enc.data_ptr());
+  return {alpha, enc};
 }
 
 static std::vector<torch::Tensor> attn_pool_bwd(torch::Tensor z_encs,
@@ -1219,6 +1272,7 @@ void register_bf16(pybind11::module_& m) {
   m.def("dwdb_grouped_arena", &dwdb_grouped_arena);
   m.def("f32_to_bf16_", &f32_to_bf16_);
   m.def("attn_pool_fwd", &attn_pool_fwd);
+  m.def("attn_pool_fwd_enc", &attn_pool_fwd_enc);
   m.def("mlp_narrow_fwd_bf16", &mlp_narrow_fwd_bf16);
   m.def("linear_bwd_dwdb_arena", &linear_bwd_dwdb_arena);
   m.def("reduce_arena", &reduce_arena, pybind11::arg("arena"),
