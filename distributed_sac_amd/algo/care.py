@@ -540,6 +540,44 @@ class CAREEngine(SACEngine):
         ext = native()
         k = info["k"]
         sv = 1 if save else 0
+        tws, tbs, twsh = info["trunk"]
+        all_narrow = (info["mix_narrow"] and info["trunk_narrow"]
+                      and (info["mlpctx"] is None or info["ctx_narrow"]))
+        if all_narrow:
+            # round 2: the whole SE forward's independent chains
+            # (mixture G=k, trunk, mlp_context) run in ONE launch
+            xs = [states_bf16, zc16]
+            wss = [list(info["mixT"]), list(twsh)]
+            bss = [list(info["mixB"]), [b.contiguous() for b in tbs]]
+            Gs, als, ofs = [k, 1], [0, 0], [1, 1]
+            if info["mlpctx"] is not None:
+                cws_, cbs_, cwsh_ = info["mlpctx"]
+                xs.append(zc16)
+                wss.append(list(cwsh_))
+                bss.append([b.contiguous() for b in cbs_])
+                Gs.append(1)
+                als.append(0)
+                ofs.append(0)
+            res = ext.mlp_narrow_fwd_multi(xs, wss, bss, Gs, als, ofs, sv)
+            z_encs = res[0][0]
+            acts_m = [states_bf16] + list(res[0][1:])
+            h = res[1][0]
+            acts_t = [zc16] + list(res[1][1:])
+            if info["mlpctx"] is not None:
+                hc = res[2][0]
+                acts_c = [zc16] + list(res[2][1:])
+                alpha, enc = ext.attn_pool_fwd_enc(h, z_encs, hc, rep)
+                if save:
+                    return enc, dict(acts_m=acts_m, z_encs=z_encs,
+                                     acts_t=acts_t, alpha=alpha,
+                                     acts_c=acts_c, rep=rep)
+                return enc, None
+            alpha, enc = ext.attn_pool_fwd_enc(h, z_encs, zc16, rep)
+            if save:
+                return enc, dict(acts_m=acts_m, z_encs=z_encs,
+                                 acts_t=acts_t, alpha=alpha, acts_c=None,
+                                 rep=rep)
+            return enc, None
         if info["mix_narrow"]:
             res = ext.mlp_narrow_fwd_bf16(states_bf16, info["mixT"],
                                           list(info["mixB"]), k, 0, 1, sv)
@@ -560,7 +598,6 @@ class CAREEngine(SACEngine):
                     z_encs = h
                 else:
                     acts_m.append(h)
-        tws, tbs, twsh = info["trunk"]
         if info["trunk_narrow"]:
             res = ext.mlp_narrow_fwd_bf16(zc16, twsh, list(tbs), 1, 0, 1, sv)
             h = res[0]                           # logits f32

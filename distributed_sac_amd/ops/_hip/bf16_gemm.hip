@@ -476,14 +476,10 @@ struct NarrowDesc {
   int L;
 };
 
-__global__ __launch_bounds__(256) void k_bf16_mlp_narrow(
-    const u16* __restrict__ x, NarrowDesc d, void* __restrict__ y,
-    int M, int K0, long xgs, int act_last, int out_f32) {
-  __shared__ u16 sx[2][TBM][TPAD];
-  __shared__ u16 sw[2][TBM][TPAD];
-  __shared__ u16 sy[TBM][TPAD];
-  const long g = blockIdx.z;
-  const int m0 = blockIdx.x * TBM;
+__device__ __forceinline__ void narrow_body(
+    u16 (*sx)[TBM][TPAD], u16 (*sw)[TBM][TPAD], u16 (*sy)[TPAD],
+    const u16* __restrict__ x, const NarrowDesc& d, void* __restrict__ y,
+    int M, int K0, long xgs, int act_last, int out_f32, int m0, long g) {
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
   const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;
@@ -612,6 +608,47 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow(
     }
     __syncthreads();
   }
+}
+
+__global__ __launch_bounds__(256) void k_bf16_mlp_narrow(
+    const u16* __restrict__ x, NarrowDesc d, void* __restrict__ y,
+    int M, int K0, long xgs, int act_last, int out_f32) {
+  __shared__ u16 sx[2][TBM][TPAD];
+  __shared__ u16 sw[2][TBM][TPAD];
+  __shared__ u16 sy[TBM][TPAD];
+  narrow_body(sx, sw, sy, x, d, y, M, K0, xgs, act_last, out_f32,
+              blockIdx.x * TBM, blockIdx.z);
+}
+
+// ---------------------------------------------------------------------------
+// Multi-chain narrow forward (round 2): ONE launch runs up to 4
+// independent narrow chains (CARE state-encoder = mixture G=k + trunk +
+// mlp_context per forward) — block ranges partition the grid by chain.
+// Each per-chain block count = ceil(M_c/64) * G_c.
+// ---------------------------------------------------------------------------
+struct NarrowMultiDesc {
+  NarrowDesc d[4];
+  const u16* x[4];
+  void* y[4];
+  int M[4], K0[4], act_last[4], out_f32[4], nbm[4];
+  long xgs[4];
+  int cum[5];
+  int C;
+};
+
+__global__ __launch_bounds__(256) void k_bf16_mlp_narrow_multi(
+    NarrowMultiDesc nm) {
+  __shared__ u16 sx[2][TBM][TPAD];
+  __shared__ u16 sw[2][TBM][TPAD];
+  __shared__ u16 sy[TBM][TPAD];
+  const int bx = blockIdx.x;
+  int c = 0;
+  while (c + 1 < nm.C && bx >= nm.cum[c + 1]) ++c;
+  const int local = bx - nm.cum[c];
+  const int g = local / nm.nbm[c];
+  const int m0 = (local % nm.nbm[c]) * TBM;
+  narrow_body(sx, sw, sy, nm.x[c], nm.d[c], nm.y[c], nm.M[c], nm.K0[c],
+              nm.xgs[c], nm.act_last[c], nm.out_f32[c], m0, g);
 }
 
 // ---------------------------------------------------------------------------
@@ -1129,6 +1166,84 @@ static std::vector<torch::Tensor> mlp_narrow_fwd_bf16(
   return out;
 }
 
+// Multi-chain narrow forward: per chain returns [y, act_0, ...] like
+// mlp_narrow_fwd_bf16, all chains in ONE kernel launch.
+static std::vector<std::vector<torch::Tensor>> mlp_narrow_fwd_multi(
+    std::vector<torch::Tensor> xs,
+    std::vector<std::vector<torch::Tensor>> wss,
+    std::vector<std::vector<torch::Tensor>> bss,
+    std::vector<long> Gs, std::vector<long> act_lasts,
+    std::vector<long> out_f32s, long save) {
+  const int C = (int)xs.size();
+  TORCH_CHECK(C >= 1 && C <= 4 && (int)wss.size() == C
+              && (int)bss.size() == C && (int)Gs.size() == C
+              && (int)act_lasts.size() == C && (int)out_f32s.size() == C);
+  NarrowMultiDesc nm{};
+  nm.C = C;
+  std::vector<torch::Tensor> keep;
+  std::vector<std::vector<torch::Tensor>> outs(C);
+  int cum = 0;
+  for (int ci = 0; ci < C; ++ci) {
+    auto& ws = wss[ci];
+    auto& bs = bss[ci];
+    const long G = Gs[ci];
+    const int L = (int)ws.size();
+    TORCH_CHECK(L >= 1 && L <= 6 && (int)bs.size() == L);
+    auto xc = xs[ci].contiguous();
+    CHECK_BF16(xc);
+    keep.push_back(xc);
+    const bool per_group_x = xc.dim() == 3;
+    const long M = per_group_x ? xc.size(1) : xc.size(0);
+    const long K0 = xc.size(-1);
+    NarrowDesc d{};
+    d.L = L;
+    long K = K0;
+    outs[ci].resize(1);
+    for (int i = 0; i < L; ++i) {
+      CHECK_BF16(ws[i]);
+      CHECK_F32(bs[i]);
+      auto wc = ws[i].contiguous();
+      auto bc = bs[i].contiguous();
+      keep.push_back(wc);
+      keep.push_back(bc);
+      const long N = wc.numel() / (G * K);
+      TORCH_CHECK(N <= 64, "narrow multi: layer width must be <= 64");
+      TORCH_CHECK(i == 0 || K <= 64);
+      d.w[i] = (const u16*)wc.data_ptr();
+      d.b[i] = bc.data_ptr<float>();
+      d.N[i] = (int)N;
+      d.acts[i] = nullptr;
+      if (save && i < L - 1) {
+        auto a = G == 1 ? torch::empty({M, N}, xc.options())
+                        : torch::empty({G, M, N}, xc.options());
+        d.acts[i] = (u16*)a.data_ptr();
+        outs[ci].push_back(a);
+      }
+      K = N;
+    }
+    auto yopts = xc.options().dtype(out_f32s[ci] ? torch::kFloat32
+                                                 : torch::kBFloat16);
+    auto y = G == 1 ? torch::empty({M, (long)d.N[L - 1]}, yopts)
+                    : torch::empty({G, M, (long)d.N[L - 1]}, yopts);
+    outs[ci][0] = y;
+    nm.d[ci] = d;
+    nm.x[ci] = (const u16*)xc.data_ptr();
+    nm.y[ci] = y.data_ptr();
+    nm.M[ci] = (int)M;
+    nm.K0[ci] = (int)K0;
+    nm.act_last[ci] = (int)act_lasts[ci];
+    nm.out_f32[ci] = (int)out_f32s[ci];
+    nm.xgs[ci] = per_group_x ? M * K0 : 0;
+    nm.nbm[ci] = (int)((M + TBM - 1) / TBM);
+    nm.cum[ci] = cum;
+    cum += nm.nbm[ci] * (int)G;
+  }
+  nm.cum[C] = cum;
+  hipLaunchKernelGGL(k_bf16_mlp_narrow_multi, dim3(cum), dim3(256), 0,
+                     cur_stream2(), nm);
+  return outs;
+}
+
 static void linear_bwd_dwdb_arena(torch::Tensor dy, torch::Tensor x,
                                   torch::Tensor yout, long act, long G,
                                   torch::Tensor arena, long w_off,
@@ -1274,6 +1389,7 @@ void register_bf16(pybind11::module_& m) {
   m.def("attn_pool_fwd", &attn_pool_fwd);
   m.def("attn_pool_fwd_enc", &attn_pool_fwd_enc);
   m.def("mlp_narrow_fwd_bf16", &mlp_narrow_fwd_bf16);
+  m.def("mlp_narrow_fwd_multi", &mlp_narrow_fwd_multi);
   m.def("linear_bwd_dwdb_arena", &linear_bwd_dwdb_arena);
   m.def("reduce_arena", &reduce_arena, pybind11::arg("arena"),
         pybind11::arg("out"), pybind11::arg("S"),
