@@ -818,18 +818,33 @@ class CAREEngine(SACEngine):
                                            self.critic_group.numel, B)
             def cur_half(acts):
                 return acts if dedup else [a[B:] for a in acts]
+            # all three SE backward chains in ONE launch (round 2)
+            dys_l, actss, wss_l = [], [], []
+            w_os, b_os, Gl, trl = [], [], [], []
             if mlpctx is not None:
-                ext.mlp_narrow_bwd_bf16(dx0[:, :zc_dim].contiguous(),
-                                        cur_half(se_saved["acts_c"]),
-                                        cwsh, arena2, offs(cws), offs(cbs),
-                                        1, 0)
-            ext.mlp_narrow_bwd_bf16(dlogits,
-                                    cur_half(se_saved["acts_t"]),
-                                    twsh, arena2, offs(tws), offs(tbs),
-                                    1, 0)
-            ext.mlp_narrow_bwd_bf16(dzencs, acts_m, info["mixT"], arena2,
-                                    offs(info["mixW"]),
-                                    offs(info["mixB"]), info["k"], 1)
+                dys_l.append(dx0[:, :zc_dim].contiguous())
+                actss.append(cur_half(se_saved["acts_c"]))
+                wss_l.append(list(cwsh))
+                w_os.append(offs(cws))
+                b_os.append(offs(cbs))
+                Gl.append(1)
+                trl.append(0)
+            dys_l.append(dlogits)
+            actss.append(cur_half(se_saved["acts_t"]))
+            wss_l.append(list(twsh))
+            w_os.append(offs(tws))
+            b_os.append(offs(tbs))
+            Gl.append(1)
+            trl.append(0)
+            dys_l.append(dzencs)
+            actss.append(acts_m)
+            wss_l.append(list(info["mixT"]))
+            w_os.append(offs(info["mixW"]))
+            b_os.append(offs(info["mixB"]))
+            Gl.append(info["k"])
+            trl.append(1)
+            ext.mlp_narrow_bwd_multi(dys_l, actss, wss_l, arena2,
+                                     w_os, b_os, Gl, trl)
             se_n = self._se_numel
             ext.reduce_arena(arena2, fg_c, S2, 0, se_n)
             ext.reduce_arena(arena_c, fg_c, S_c, se_n, -1)

@@ -678,16 +678,12 @@ struct NarrowBwdDesc {
   int transpose_w;      // store dW^T (the mixture's (k,in,out) masters)
 };
 
-__global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
-    const u16* __restrict__ dy_last, NarrowBwdDesc d,
-    float* __restrict__ arena, long s_stride, int M, int Kin) {
-  __shared__ u16 s_dy[TBM][TPAD];    // dy_i   [m][n]
-  __shared__ u16 s_dyT[TBM][TPAD];   // dy_i^T [n][m]
-  __shared__ u16 s_a[TBM][TPAD];     // a_i    [m][k]
-  __shared__ u16 s_aT[TBM][TPAD];    // a_i^T  [k][m]
-  __shared__ u16 s_wT[TBM][TPAD];    // W_i^T  [k][n]
-  const long g = blockIdx.z;
-  const int s = blockIdx.x;
+__device__ __forceinline__ void narrow_bwd_body(
+    u16 (*s_dy)[TPAD], u16 (*s_dyT)[TPAD], u16 (*s_a)[TPAD],
+    u16 (*s_aT)[TPAD], u16 (*s_wT)[TPAD],
+    const u16* __restrict__ dy_last, const NarrowBwdDesc& d,
+    float* __restrict__ arena, long s_stride, int M, int Kin,
+    int s, long g) {
   const int m0 = s * TBM;
   const int tid = threadIdx.x;
   const int lane = tid & 63, wid = tid >> 6;
@@ -824,6 +820,46 @@ __global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
 //      mul/sum/softmax-backward kernels autograd records.
 // One wavefront per batch row; requires D <= 64 and E <= 16 (cfg: 50, 6).
 // ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd(
+    const u16* __restrict__ dy_last, NarrowBwdDesc d,
+    float* __restrict__ arena, long s_stride, int M, int Kin) {
+  __shared__ u16 s_dy[TBM][TPAD];    // dy_i   [m][n]
+  __shared__ u16 s_dyT[TBM][TPAD];   // dy_i^T [n][m]
+  __shared__ u16 s_a[TBM][TPAD];     // a_i    [m][k]
+  __shared__ u16 s_aT[TBM][TPAD];    // a_i^T  [k][m]
+  __shared__ u16 s_wT[TBM][TPAD];    // W_i^T  [k][n]
+  narrow_bwd_body(s_dy, s_dyT, s_a, s_aT, s_wT, dy_last, d, arena,
+                  s_stride, M, Kin, blockIdx.x, blockIdx.z);
+}
+
+// Multi-chain narrow backward (round 2): the SE backward's three chains
+// (mlp_context, trunk, mixture) in ONE launch, all writing the same
+// phase arena.  grid (sum_c S_c * G_c).
+struct NarrowBwdMultiDesc {
+  NarrowBwdDesc d[4];
+  const u16* dy[4];
+  int M[4], Kin[4], S[4];
+  int cum[5];
+  int C;
+};
+
+__global__ __launch_bounds__(256) void k_bf16_mlp_narrow_bwd_multi(
+    NarrowBwdMultiDesc nm, float* __restrict__ arena, long s_stride) {
+  __shared__ u16 s_dy[TBM][TPAD];
+  __shared__ u16 s_dyT[TBM][TPAD];
+  __shared__ u16 s_a[TBM][TPAD];
+  __shared__ u16 s_aT[TBM][TPAD];
+  __shared__ u16 s_wT[TBM][TPAD];
+  const int bx = blockIdx.x;
+  int c = 0;
+  while (c + 1 < nm.C && bx >= nm.cum[c + 1]) ++c;
+  const int local = bx - nm.cum[c];
+  const long g = local / nm.S[c];
+  const int sblk = local % nm.S[c];
+  narrow_bwd_body(s_dy, s_dyT, s_a, s_aT, s_wT, nm.dy[c], nm.d[c], arena,
+                  s_stride, nm.M[c], nm.Kin[c], sblk, g);
+}
+
 // enc_out mode (round 2): logits/alpha/hc have Mz = M/rep rows (the
 // trunk and context projections run ONCE on deduplicated z_context rows
 // — the batched [next|current] SE forward duplicated them); the pooled
@@ -1383,7 +1419,81 @@ static void dwdb_grouped_arena(std::vector<torch::Tensor> dys,
                      cur_stream2(), d, arena.data_ptr<float>(), stride);
 }
 
+static void mlp_narrow_bwd_multi(
+    std::vector<torch::Tensor> dy_lasts,
+    std::vector<std::vector<torch::Tensor>> actss,
+    std::vector<std::vector<torch::Tensor>> wss,
+    torch::Tensor arena,
+    std::vector<std::vector<long>> w_offss,
+    std::vector<std::vector<long>> b_offss,
+    std::vector<long> Gs, std::vector<long> transpose_ws) {
+  CHECK_F32(arena);
+  const int C = (int)dy_lasts.size();
+  TORCH_CHECK(C >= 1 && C <= 4 && (int)actss.size() == C
+              && (int)wss.size() == C && (int)w_offss.size() == C
+              && (int)b_offss.size() == C && (int)Gs.size() == C
+              && (int)transpose_ws.size() == C);
+  TORCH_CHECK(arena.dim() == 2 && arena.is_contiguous());
+  NarrowBwdMultiDesc nm{};
+  nm.C = C;
+  std::vector<torch::Tensor> keep;
+  int cum = 0;
+  for (int ci = 0; ci < C; ++ci) {
+    auto dyc = dy_lasts[ci].contiguous();
+    CHECK_BF16(dyc);
+    keep.push_back(dyc);
+    const long G = Gs[ci];
+    auto& ws = wss[ci];
+    auto& acts = actss[ci];
+    const int L = (int)ws.size();
+    TORCH_CHECK(L >= 2 && L <= 6 && (int)acts.size() == L
+                && (int)w_offss[ci].size() == L
+                && (int)b_offss[ci].size() == L);
+    const long M = G == 1 ? dyc.size(0) : dyc.size(1);
+    NarrowBwdDesc d{};
+    d.L = L;
+    d.transpose_w = (int)transpose_ws[ci];
+    long Kin = acts[0].size(-1);
+    long K = Kin;
+    for (int i = 0; i < L; ++i) {
+      CHECK_BF16(ws[i]);
+      CHECK_BF16(acts[i]);
+      auto wc = ws[i].contiguous();
+      auto ac = acts[i].contiguous();
+      keep.push_back(wc);
+      keep.push_back(ac);
+      const long N = wc.numel() / (G * K);
+      TORCH_CHECK(N <= 64 && (i == 0 || K <= 64));
+      d.w[i] = (const u16*)wc.data_ptr();
+      d.acts[i] = (const u16*)ac.data_ptr();
+      d.a_gs[i] = ac.dim() == 3 ? M * K : 0;
+      TORCH_CHECK(ac.numel() == (ac.dim() == 3 ? G : 1) * M * K);
+      d.w_off[i] = w_offss[ci][i];
+      d.b_off[i] = b_offss[ci][i];
+      d.N[i] = (int)N;
+      TORCH_CHECK(w_offss[ci][i] + G * N * K <= arena.size(1)
+                  && b_offss[ci][i] + G * N <= arena.size(1));
+      K = N;
+    }
+    TORCH_CHECK(dyc.numel() == G * M * d.N[L - 1]);
+    const long S = (M + TBM - 1) / TBM;
+    TORCH_CHECK(arena.size(0) >= S, "arena rows must cover ceil(M/64)");
+    nm.d[ci] = d;
+    nm.dy[ci] = (const u16*)dyc.data_ptr();
+    nm.M[ci] = (int)M;
+    nm.Kin[ci] = (int)Kin;
+    nm.S[ci] = (int)S;
+    nm.cum[ci] = cum;
+    cum += (int)(S * G);
+  }
+  nm.cum[C] = cum;
+  hipLaunchKernelGGL(k_bf16_mlp_narrow_bwd_multi, dim3(cum), dim3(256), 0,
+                     cur_stream2(), nm, arena.data_ptr<float>(),
+                     arena.size(1));
+}
+
 void register_bf16(pybind11::module_& m) {
+  m.def("mlp_narrow_bwd_multi", &mlp_narrow_bwd_multi);
   m.def("dwdb_grouped_arena", &dwdb_grouped_arena);
   m.def("f32_to_bf16_", &f32_to_bf16_);
   m.def("attn_pool_fwd", &attn_pool_fwd);
