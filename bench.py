@@ -53,7 +53,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=300)
     p.add_argument("--warmup", type=int, default=30)
     p.add_argument("--config", type=str, default="mtsac",
-                   choices=["mtsac", "sac", "vsac", "care"])
+                   choices=["mtsac", "sac", "vsac", "care", "mt1_care"])
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--skip-rollout-probe", action="store_true")
     p.add_argument("--no-graph", action="store_true",
