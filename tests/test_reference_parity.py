@@ -10,6 +10,7 @@ parameter-level agreement:
 - LunarLander SAC     (LunarLander_Distributed_SAC/src/learner.py:203-239)
 - MT10 MT-SAC         (MT10_Distributed_MTSAC/src/learner.py:253-325)
 - MT10 CARE(M)        (MT10_Distributed_CARE/src/learner.py:281-404)
+- MT1 original CARE   (MT1_Distributed_CARE/src/learner.py:247-314)
 
 RNG alignment: the reference samples actions via Normal(mu, std).rsample()
 (model.get_action_log_prob*), our torch path via torch.randn_like — both
@@ -424,4 +425,109 @@ def test_care_modified_vs_actual_reference_code():
                          L.local_critic.parameters(), "critic", atol=amp)
     _assert_params_close(engine2.target_critic.parameters(),
                          L.target_critic.parameters(), "target", atol=amp)
+    _assert_params_close([engine2.log_alpha], [L.log_alpha], "log_alpha")
+
+
+# ---------------------------------------------------------------------------
+# MT1 original CARE (trainable context encoder; reference hardcodes the
+# state-encoder tau 0.05 and steps the context optimizer after update_SAC)
+# ---------------------------------------------------------------------------
+
+def test_mt1_original_care_vs_actual_reference_code():
+    from distributed_sac_amd.algo import CAREEngine
+    from distributed_sac_amd.config import SACConfig
+
+    with ref_src("MT1_Distributed_CARE"):
+        um = importlib.import_module("utils")
+        lm = importlib.import_module("learner")
+        cfg = um.cfg_read(f"{REF}/cfg/MT1_Distributed_CARE_cfg.json")
+        enc = dict(cfg["encoder"])
+        enc["pretrained_embedding_json_path"] = \
+            f"{REF}/" + enc["pretrained_embedding_json_path"]
+        enc["task_name_json_path"] = f"{REF}/" + enc["task_name_json_path"]
+        enc["device"] = "cpu"   # the MT1 encoder cfg hardcodes cuda:0
+        L = lm.Learner.__new__(lm.Learner)
+        L.cfg = cfg
+        L.actor_cfg = cfg["actor"]
+        L.critic_cfg = cfg["critic"]
+        L.encoder_cfg = enc
+        L.device = torch.device("cpu")
+        L.num_tasks = int(cfg["num_tasks"])
+        L.gamma = cfg["gamma"]
+        L.tau = cfg["tau"]
+        L.reward_scale = cfg["reward_scale"]
+        L.lr_actor = cfg["actor"]["lr_actor"]
+        L.lr_critic = cfg["critic"]["lr_critic"]
+        torch.manual_seed(4)
+        L.build_model()
+        L.build_optimizer()
+
+    mycfg = SACConfig.from_dict(cfg, variant="care")
+    mycfg.encoder = enc
+    assert not mycfg.use_modified_care   # MT1 = original CARE
+    engine = CAREEngine(mycfg, "cpu")
+    engine.context_encoder.load_state_dict(L.context_encoder.state_dict())
+    engine.actor.load_state_dict(L.actor.state_dict())
+    engine.local_critic.load_state_dict(L.local_critic.state_dict())
+    engine.target_critic.load_state_dict(L.target_critic.state_dict())
+    with torch.no_grad():
+        engine.log_alpha.copy_(L.log_alpha)
+    engine.alpha = engine.log_alpha.exp().detach()
+    engine.tie_actor_state_encoder()
+
+    B = 48
+    S = int(cfg["actor"]["state_dim"])
+    A = int(cfg["actor"]["action_dim"])
+    # gradient-level: one update with every lr zeroed
+    for opt in (L.actor_optimizer, L.critic_optimizer,
+                L.log_alpha_optimizer, L.context_encoder_optimizer):
+        for pg in opt.param_groups:
+            pg["lr"] = 0.0
+    for name in ("actor_optimizer", "critic_optimizer",
+                 "log_alpha_optimizer", "context_encoder_optimizer"):
+        getattr(engine, name).lr = 0.0
+    b = _mt_batch(B, S, A, 1, seed=321)
+    torch.manual_seed(9100)
+    alpha = L.get_log_alpha(b["states"]).exp().detach()
+    L.optimizer_zero_grad()
+    L.update_SAC(b["states"], b["actions"], b["rewards"],
+                 b["next_states"], b["dones"], alpha, retain_graph=True)
+    torch.manual_seed(9100)
+    engine.update(dict(b))
+    _assert_grads_close(engine.actor, L.actor, "actor")
+    _assert_grads_close(engine.local_critic, L.local_critic, "critic")
+    _assert_grads_close(engine.context_encoder, L.context_encoder, "ctx")
+    assert (engine.log_alpha.grad - L.log_alpha.grad).abs().max() < 2e-4
+
+    # parameter-level after 3 real Adam updates (fresh optimizer state)
+    L.build_optimizer()
+    engine2 = CAREEngine(mycfg, "cpu")
+    engine2.context_encoder.load_state_dict(L.context_encoder.state_dict())
+    engine2.actor.load_state_dict(L.actor.state_dict())
+    engine2.local_critic.load_state_dict(L.local_critic.state_dict())
+    engine2.target_critic.load_state_dict(L.target_critic.state_dict())
+    with torch.no_grad():
+        engine2.log_alpha.copy_(L.log_alpha)
+    engine2.alpha = engine2.log_alpha.exp().detach()
+    engine2.tie_actor_state_encoder()
+    for k in range(3):
+        b = _mt_batch(B, S, A, 1, seed=322 + k)
+        torch.manual_seed(9200 + k)
+        alpha = L.get_log_alpha(b["states"]).exp().detach()
+        L.optimizer_zero_grad()
+        L.update_SAC(b["states"], b["actions"], b["rewards"],
+                     b["next_states"], b["dones"], alpha,
+                     retain_graph=True)
+        L.context_encoder_optimizer.step()
+        L.soft_update(L.local_critic.state_encoder,
+                      L.actor.state_encoder, tau=1.0)
+        torch.manual_seed(9200 + k)
+        engine2.update(dict(b))
+    amp = 3 * 2 * float(cfg["actor"]["lr_actor"])
+    _assert_params_close(engine2.actor.parameters(), L.actor.parameters(),
+                         "actor", atol=amp)
+    _assert_params_close(engine2.local_critic.parameters(),
+                         L.local_critic.parameters(), "critic", atol=amp)
+    _assert_params_close(engine2.context_encoder.parameters(),
+                         L.context_encoder.parameters(), "ctx", atol=amp)
     _assert_params_close([engine2.log_alpha], [L.log_alpha], "log_alpha")
