@@ -71,6 +71,7 @@ def parse_args():
                         "rings -> learner (single rank)")
     p.add_argument("--async-seconds", type=float, default=30.0)
     p.add_argument("--players", type=int, default=4)
+    p.add_argument("--chunk-steps", type=int, default=64)
     p.add_argument("--force-ddp", action="store_true",
                    help="initialize the process group + run the DP code "
                         "path even at world_size 1 (single-GPU RCCL "
@@ -224,7 +225,7 @@ def run_async_bench(cfg, args, device, precision):
     cfg2.start_memory_len = min(cfg.start_memory_len, 2000)
     dt = DistributedTrainer(cfg2, device=device,
                             num_players=args.players,
-                            chunk_steps=64, seed=7,
+                            chunk_steps=args.chunk_steps, seed=7,
                             use_graph=not args.no_graph,
                             precision=precision)
     stats = dt.run(max_seconds=args.async_seconds)
@@ -289,6 +290,7 @@ def main():
             "data": "synthetic",
             "env_steps_per_sec": stats.get("env_steps_per_sec"),
             "ring_drops": stats.get("ring_drops"),
+            "phase_seconds": stats.get("phase_seconds"),
             "wall_seconds": round(wall, 2),
             "config": {"model": model_desc,
                        "global_batch": cfg.batch_size,
