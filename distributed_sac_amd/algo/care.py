@@ -645,15 +645,47 @@ class CAREEngine(SACEngine):
 
     @torch.no_grad()
     def _update_tensors_manual(self, batch):
-        """Modified-CARE bf16 update with a HAND-ROLLED backward: critic
-        loss gradients flow through the twin heads into the state encoder
-        (attention pool -> mlp_context / trunk / mixture) with every dW/db
-        kernel writing directly into the flat critic-group gradient buffer
+        """CARE bf16 update with a HAND-ROLLED backward (both variants:
+        CARE(M) frozen context, original CARE trainable context encoder),
+        structured as three SEGMENTS split at the DP all-reduce
+        boundaries so capture_dp (inherited from SACEngine) can hipGraph
+        each segment with eager RCCL between replays:
+          seg1: z_context + SE/actor forwards + TD + critic(+context)
+                backward -> critic (+ context) grads
+          seg2: critic Adam + actor-side forward/backward -> aa grads
+          seg3: actor/alpha Adam + Polyak + context Adam + SE tie
         (reference gradient-flow rules: MT10_Distributed_CARE/src/
-        learner.py:281-404 — trunk input detached, z_context frozen
-        embedding, actor optimizes only its head)."""
+        learner.py:281-404 — trunk input detached, actor optimizes only
+        its head, context grads from the critic loss only)."""
+        self._manual_seg1(batch)
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+            if self._dp_st.get("orig") and self.context_group is not None:
+                self.ddp.allreduce_grad_(self.context_group.flat_grad)
+        self._manual_seg2()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self._aa_arena)
+        return self._manual_seg3()
+
+    def dp_graphed_update(self):
+        """Segmented DP replay with CARE's extra context all-reduce."""
+        g1, g2, g3 = self._dp_graphs
+        g1.replay()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+            if (not self.use_modified_care
+                    and self.context_group is not None):
+                self.ddp.allreduce_grad_(self.context_group.flat_grad)
+        g2.replay()
+        if self.ddp is not None:
+            self.ddp.allreduce_grad_(self._aa_arena)
+        g3.replay()
+        self.update_iteration += 1
+        return self._dp_graph_metrics
+
+    @torch.no_grad()
+    def _manual_seg1(self, batch):
         from ..ops import native
-        from ..ops.flat import FusedAdam as _FA
         ext = native()
         states, actions = batch["states"], batch["actions"]
         rewards, next_states, dones = (batch["rewards"],
@@ -897,8 +929,26 @@ class CAREEngine(SACEngine):
                  for l in cc["lins"]],
                 1, S_x, ch_x)
             ext.reduce_arena(arena_x, fgx, S_x)
-        if self.ddp is not None:
-            self.ddp.allreduce_grad_(self.critic_group.flat_grad)
+        self._dp_st = dict(
+            chain=chain, states=states, sd=sd, zc16=zc16, sa=sa, lp=lp,
+            ls_cat=ls_cat, B=B, eps=eps, tanh_u=tanh_u, lsr=lsr,
+            acts_a=acts_a, ws_f32=ws_f32, bs_f32=bs_f32, la_det=la_det,
+            T=T, use_w=use_w, closs=closs, orig=orig)
+
+    @torch.no_grad()
+    def _manual_seg2(self):
+        from ..ops import native
+        ext = native()
+        st = self._dp_st
+        chain = st["chain"]
+        states, sd, zc16 = st["states"], st["sd"], st["zc16"]
+        sa, lp, ls_cat, B = st["sa"], st["lp"], st["ls_cat"], st["B"]
+        eps, tanh_u, lsr = st["eps"], st["tanh_u"], st["lsr"]
+        acts_a, ws_f32, bs_f32 = st["acts_a"], st["ws_f32"], st["bs_f32"]
+        la_det, T, use_w = st["la_det"], st["T"], st["use_w"]
+        info = self._se_local
+        nl_c = len(self._twin_local_bf16)
+        nl_a = len(self._actor_ws_bf16)
         self.critic_optimizer.step()   # adam kernel refreshes flat mirror
         self._refresh_mixT("critic")   # transposed mixture views still need it
 
@@ -978,18 +1028,21 @@ class CAREEngine(SACEngine):
                     dy = ext.linear_bwd_dx_bf16(dy, self._actor_ws_bf16[i],
                                                 yout, act, 1, 1)
         ext.reduce_arena(arena_a, fg_a, S_a)
-        if self.ddp is not None:
-            self.ddp.allreduce_grad_(self._aa_arena)
+        st["al"] = al
+
+    @torch.no_grad()
+    def _manual_seg3(self):
+        from ..ops.flat import FusedAdam as _FA
+        st = self._dp_st
         _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
         self.alpha = self.log_alpha.exp().detach()
 
         self._polyak_targets(mirror=self._target_bf16)
         self._refresh_mixT("target")
-        if orig:
-            if self.ddp is not None:
-                self.ddp.allreduce_grad_(self.context_group.flat_grad)
+        if st["orig"]:
             self.context_encoder_optimizer.step()   # refreshes ctx mirror
         self.tie_actor_state_encoder()
+        closs, al = st["closs"], st["al"]
         return {"critic_loss": closs[0] + closs[1],
                 "actor_loss": al[0],
                 "alpha_loss": al[2],
