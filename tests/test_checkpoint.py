@@ -201,3 +201,23 @@ def test_shipped_ll_checkpoint_evaluates():
         default_env_fn, episodes=2, seed=77)
     assert out["update_iteration"] == 292428
     assert out["mean_reward"] > -5.0   # random policy ~-70
+
+
+def test_gpu_trained_checkpoint_evaluates():
+    """The MI355X-trained MT10-MTSAC artifact (saved_models/mtsac_gpu,
+    858k grad steps in 8 min through the async stack) loads in the
+    reference schema and acts successfully on the synthetic suite."""
+    import numpy as np
+    from distributed_sac_amd.config import load_variant
+    from distributed_sac_amd.workers.player import evaluate_checkpoint
+    from distributed_sac_amd.workers.trainer import default_env_fn
+    path = os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "saved_models", "mtsac_gpu",
+        "checkpoint_5148606.tar")
+    assert os.path.exists(path)
+    cfg = load_variant("mtsac")
+    out = evaluate_checkpoint(cfg, path, default_env_fn, task_idx=2,
+                              episodes=5, seed=3)
+    assert out["update_iteration"] == 5148606
+    assert out["success_rate"] >= 0.8
+    assert np.isfinite(out["mean_reward"])
