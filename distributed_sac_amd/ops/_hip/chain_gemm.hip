@@ -69,6 +69,7 @@ __device__ __forceinline__ u16 f32_to_bf16_rne3(float f) {
 
 struct ChainFwdDesc {
   const u16* w[6];    // [G, N_l, K_l] bf16 (k-minor)
+  const u16* wp[6];   // fragment-PACKED mirror (k_bf16_pack_frag) or null
   const float* b[6];  // [G, N_l]
   u16* acts[6];       // post-act output of layer l (null = don't save)
   int N[6];
@@ -208,7 +209,56 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
         }                                                                 \
       }
 
-      if (k_aligned) {
+      if (d.wp[li] != nullptr) {
+        // FRAGMENT-PACKED path: every load is wp + base + ks*512 +
+        // lane*8 — one contiguous 1 KiB line per wave (TA cost 1/instr
+        // instead of 16), zero-padding baked in (no guards, covers odd
+        // K and partial-N tiles uniformly).
+        const int KS = (K + 31) >> 5;
+        const int NTp = (N + 15) >> 4;
+        const u16* wpl = d.wp[li];
+        long bp[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          int tt = t0 + (j < (nq > 0 ? nq : 1) ? j
+                         : (nq > 0 ? nq - 1 : 0));
+          if (tt >= NTp) tt = NTp - 1;
+          if (tt < 0) tt = 0;
+          bp[j] = (((long)g * NTp + tt) * KS) * 512 + lane * 8;
+        }
+#define LOADP(dst, kk)                                                    \
+        _Pragma("unroll") for (int j = 0; j < 4; ++j)                     \
+          dst[j] = *(const bf16x8*)&wpl[bp[j] + (long)(kk) * 16];
+        // (kk is the k element offset; one 32-k step = 512 u16 = kk*16
+        //  since kk advances by 32)
+        {
+          bf16x8 S0[4], S1[4], S2[4];
+          const int pbody = KS * 32;
+          LOADP(S0, 0)
+          if (32 < pbody) LOADP(S1, 32)
+          int k = 0;
+          while (k + 96 <= pbody) {
+            LOADP(S2, k + 64)
+            MF4(S0, k)
+            if (k + 96 < pbody) {
+              LOADP(S0, k + 96)
+            }
+            MF4(S1, k + 32)
+            if (k + 128 < pbody) {
+              LOADP(S1, k + 128)
+            }
+            MF4(S2, k + 64)
+            k += 96;
+          }
+          if (pbody - k == 32) {
+            MF4(S0, k)
+          } else if (pbody - k == 64) {
+            MF4(S0, k)
+            MF4(S1, k + 32)
+          }
+        }
+#undef LOADP
+      } else if (k_aligned) {
         // unconditional loads, 3-set rotating software pipeline: loads
         // run TWO k-steps ahead of their MFMAs (12 B-fragments in
         // flight), hiding cold-L2/L3 weight-miss latency that a 1-deep
@@ -309,6 +359,103 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
       K = N;
     }
   }
+}
+
+// ---------------------------------------------------------------------------
+// FRAGMENT-PACKED weight mirrors (round 2, the TA fix): the chain
+// kernels' B-operand loads are fragment-shaped — lane l of a wave reads
+// 16 B of row tile*16+(l&15) at k-offset (l>>4)*8, i.e. 16 scattered
+// cache lines per instruction, which kept GRBM_TA_BUSY at 66% of the
+// kernel.  This kernel re-packs a weight matrix ONCE per Adam step into
+// exactly that per-lane layout:
+//     fwd:  P[((g*NT + t)*KS + ks)*64 + l][0..8) = W[g][t*16 + (l&15)]
+//                                           [ks*32 + (l>>4)*8 + j]
+//     dx:   same with W TRANSPOSED (rows = k, cols = n)
+// (zero-filled out of range) so every chain load is base + lane*16 —
+// one contiguous 1 KiB line per wave, and the odd-K / partial-tile
+// guard paths disappear.  One launch packs up to 24 matrices.
+// ---------------------------------------------------------------------------
+struct PackDesc {
+  const u16* w[24];
+  u16* p[24];
+  int N[24], K[24], G[24], dx[24];
+  int cum[25];   // cumulative output chunks (G*NT*KS*64 per matrix)
+  int C;
+};
+
+__global__ __launch_bounds__(256) void k_bf16_pack_frag(PackDesc d) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  int c = 0;
+  while (c + 1 < d.C && i >= d.cum[c + 1]) ++c;
+  if (i >= d.cum[d.C]) return;
+  long local = i - d.cum[c];
+  const int lane = (int)(local & 63);
+  local >>= 6;
+  const int N = d.N[c], K = d.K[c];
+  const bool dx = d.dx[c] != 0;
+  // logical tile dims of the PACKED operand
+  const int rows = dx ? K : N;      // fragment row dimension
+  const int cols = dx ? N : K;      // fragment k dimension
+  const int KS = (cols + 31) >> 5;
+  const int NT = (rows + 15) >> 4;
+  const int ks = (int)(local % KS);
+  local /= KS;
+  const int t = (int)(local % NT);
+  const long g = local / NT;
+  const int row = t * 16 + (lane & 15);
+  const int k0 = ks * 32 + (lane >> 4) * 8;
+  const u16* w = d.w[c] + g * (long)N * K;
+  u16* out = d.p[c] + (i - d.cum[c]) * 8;
+  u16 v[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int kk = k0 + j;
+    if (row < rows && kk < cols)
+      v[j] = dx ? w[(long)kk * K + row]    // W^T: [k-row=row][n=kk] -> W[kk][row]
+                : w[(long)row * K + kk];
+    else
+      v[j] = 0;
+  }
+  *(uint4*)out = *(uint4*)v;
+}
+
+static void pack_weights_frag(std::vector<torch::Tensor> ws,
+                              std::vector<torch::Tensor> ps,
+                              std::vector<long> Gs,
+                              std::vector<long> dxs) {
+  const int C = (int)ws.size();
+  TORCH_CHECK(C >= 1 && C <= 24 && (int)ps.size() == C
+              && (int)Gs.size() == C && (int)dxs.size() == C);
+  PackDesc d{};
+  d.C = C;
+  std::vector<torch::Tensor> keep;
+  long cum = 0;
+  for (int i = 0; i < C; ++i) {
+    CHAIN_CHECK_BF16(ws[i]);
+    CHAIN_CHECK_BF16(ps[i]);
+    auto wc = ws[i].contiguous();
+    keep.push_back(wc);
+    TORCH_CHECK(ps[i].is_contiguous());
+    const long G = Gs[i];
+    const long K = wc.size(-1);
+    const long N = wc.numel() / (G * K);
+    const long rows = dxs[i] ? K : N;
+    const long cols = dxs[i] ? N : K;
+    const long KS = (cols + 31) / 32, NT = (rows + 15) / 16;
+    TORCH_CHECK(ps[i].numel() == G * NT * KS * 512,
+                "packed buffer size mismatch at ", i);
+    d.w[i] = (const u16*)wc.data_ptr();
+    d.p[i] = (u16*)ps[i].data_ptr();
+    d.N[i] = (int)N;
+    d.K[i] = (int)K;
+    d.G[i] = (int)G;
+    d.dx[i] = (int)dxs[i];
+    d.cum[i] = (int)cum;
+    cum += G * NT * KS * 64;
+  }
+  d.cum[C] = (int)cum;
+  hipLaunchKernelGGL(k_bf16_pack_frag, dim3((cum + 255) / 256), dim3(256),
+                     0, cur_stream3(), d);
 }
 
 // ---------------------------------------------------------------------------
@@ -418,6 +565,7 @@ static void transpose_weights_bf16(std::vector<torch::Tensor> ws,
 // d(critic input)) as fp32 per group.  grid (ceil(M/16), 1, G).
 // ---------------------------------------------------------------------------
 struct ChainDxDesc {
+  const u16* wp[6];     // fragment-PACKED W^T mirror (pack dx mode) or null
   const u16* wt[6];     // [G, K_l, N_l] n-minor transposed weights
   const u16* yout[6];   // post-act output of layer l (mask); null if act=0
   long yo_gs[6];        // per-group element stride of yout
@@ -524,7 +672,51 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
               a_, bset[j], acc[0][j], 0, 0, 0);                           \
       }
 
-      if (n_aligned) {
+      if (d.wp[li] != nullptr) {
+        // fragment-packed W^T: unit-stride loads, guards baked in
+        const int KSp = (N + 31) >> 5;
+        const int NTp = (K + 15) >> 4;
+        const u16* wpl = d.wp[li];
+        long bp[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          int tt = t0 + (j < (nq > 0 ? nq : 1) ? j
+                         : (nq > 0 ? nq - 1 : 0));
+          if (tt >= NTp) tt = NTp - 1;
+          if (tt < 0) tt = 0;
+          bp[j] = (((long)g * NTp + tt) * KSp) * 512 + lane * 8;
+        }
+#define LOADP(dst, kk)                                                    \
+        _Pragma("unroll") for (int j = 0; j < 4; ++j)                     \
+          dst[j] = *(const bf16x8*)&wpl[bp[j] + (long)(kk) * 16];
+        {
+          bf16x8 S0[4], S1[4], S2[4];
+          const int pbody = KSp * 32;
+          LOADP(S0, 0)
+          if (32 < pbody) LOADP(S1, 32)
+          int k = 0;
+          while (k + 96 <= pbody) {
+            LOADP(S2, k + 64)
+            MF4(S0, k)
+            if (k + 96 < pbody) {
+              LOADP(S0, k + 96)
+            }
+            MF4(S1, k + 32)
+            if (k + 128 < pbody) {
+              LOADP(S1, k + 128)
+            }
+            MF4(S2, k + 64)
+            k += 96;
+          }
+          if (pbody - k == 32) {
+            MF4(S0, k)
+          } else if (pbody - k == 64) {
+            MF4(S0, k)
+            MF4(S1, k + 32)
+          }
+        }
+#undef LOADP
+      } else if (n_aligned) {
         if (nbody >= 32) {
           bf16x8 S0[4], S1[4], S2[4];
           LOADQ(S0, 0)
@@ -619,11 +811,13 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
 static std::vector<torch::Tensor> mlp_chain_dx_bf16(
     torch::Tensor dy_last, std::vector<torch::Tensor> wts,
     std::vector<torch::Tensor> youts, long K0,
-    std::vector<long> acts, long G, long save_dys, long dx0_lo) {
+    std::vector<long> acts, long G, long save_dys, long dx0_lo,
+    std::vector<torch::Tensor> wps) {
   CHAIN_CHECK_BF16(dy_last);
   const int L = (int)wts.size();
   TORCH_CHECK(L >= 1 && L <= 6 && (int)youts.size() == L
               && (int)acts.size() == L);
+  TORCH_CHECK(wps.empty() || (int)wps.size() == L);
   auto dyc = dy_last.contiguous();
   const long M = G == 1 ? dyc.size(0) : dyc.size(1);
   ChainDxDesc d{};
@@ -641,6 +835,15 @@ static std::vector<torch::Tensor> mlp_chain_dx_bf16(
     TORCH_CHECK(N * G * K == wc.numel(), "wt shape mismatch at layer ", i);
     TORCH_CHECK(N <= CMAX && K <= CMAX);
     d.wt[i] = (const u16*)wc.data_ptr();
+    d.wp[i] = nullptr;
+    if (!wps.empty() && wps[i].numel() > 0) {
+      CHAIN_CHECK_BF16(wps[i]);
+      TORCH_CHECK(wps[i].is_contiguous()
+                  && wps[i].numel() ==
+                         G * ((K + 15) / 16) * ((N + 31) / 32) * 512);
+      d.wp[i] = (const u16*)wps[i].data_ptr();
+      keep.push_back(wps[i]);
+    }
     d.N[i] = (int)N;
     d.act[i] = (int)acts[i];
     d.yout[i] = nullptr;
@@ -694,9 +897,11 @@ static std::vector<torch::Tensor> mlp_chain_dx_bf16(
 static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
     torch::Tensor x1, torch::Tensor x2, std::vector<torch::Tensor> ws,
     std::vector<torch::Tensor> bs, long act_last, long G, long out_f32,
-    long rm, long rowcat, long save_acts) {
+    long rm, long rowcat, long save_acts,
+    std::vector<torch::Tensor> wps) {
   const int L = (int)ws.size();
   TORCH_CHECK(L >= 1 && L <= 6 && (int)bs.size() == L);
+  TORCH_CHECK(wps.empty() || (int)wps.size() == L);
   TORCH_CHECK(x1.is_cuda() && x1.dim() == 2);
   const bool has2 = x2.numel() > 0;
   const bool x1f = x1.scalar_type() == torch::kFloat32;
@@ -744,6 +949,15 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
                 i, " (K=", K, ")");
     TORCH_CHECK(N <= CMAX, "chain fwd: layer width must be <= ", CMAX);
     d.w[i] = (const u16*)wc.data_ptr();
+    d.wp[i] = nullptr;
+    if (!wps.empty() && wps[i].numel() > 0) {
+      CHAIN_CHECK_BF16(wps[i]);
+      TORCH_CHECK(wps[i].is_contiguous()
+                  && wps[i].numel() ==
+                         G * ((N + 15) / 16) * ((K + 31) / 32) * 512);
+      d.wp[i] = (const u16*)wps[i].data_ptr();
+      keep.push_back(wps[i]);
+    }
     d.b[i] = bc.data_ptr<float>();
     d.N[i] = (int)N;
     d.act[i] = (i == L - 1) ? (int)act_last : 1;
@@ -793,10 +1007,17 @@ static std::vector<torch::Tensor> mlp_chain_fwd_bf16(
 
 void register_chain(pybind11::module_& m) {
   m.def("transpose_weights_bf16", &chain::transpose_weights_bf16);
-  m.def("mlp_chain_dx_bf16", &chain::mlp_chain_dx_bf16);
+  m.def("pack_weights_frag", &chain::pack_weights_frag);
+  m.def("mlp_chain_dx_bf16", &chain::mlp_chain_dx_bf16,
+        pybind11::arg("dy_last"), pybind11::arg("wts"),
+        pybind11::arg("youts"), pybind11::arg("K0"), pybind11::arg("acts"),
+        pybind11::arg("G"), pybind11::arg("save_dys"),
+        pybind11::arg("dx0_lo"),
+        pybind11::arg("wps") = std::vector<torch::Tensor>{});
   m.def("mlp_chain_fwd_bf16", &chain::mlp_chain_fwd_bf16,
         pybind11::arg("x1"), pybind11::arg("x2"), pybind11::arg("ws"),
         pybind11::arg("bs"), pybind11::arg("act_last"), pybind11::arg("G"),
         pybind11::arg("out_f32"), pybind11::arg("rm") = 0,
-        pybind11::arg("rowcat") = 0, pybind11::arg("save_acts") = 1);
+        pybind11::arg("rowcat") = 0, pybind11::arg("save_acts") = 1,
+        pybind11::arg("wps") = std::vector<torch::Tensor>{});
 }
