@@ -104,12 +104,13 @@ def run_dx(name, M, dims, G, bench=False):
         wts.append(wt)
         wps.append(packed_of(w, G, True))
         flags.append(0 if last else 1)
+        ysh = (G, M, N) if G > 1 else (M, N)
         youts.append(torch.empty(0, device=dev, dtype=torch.bfloat16)
                      if last else
-                     (torch.randn(G, M, N, device=dev).relu())
+                     (torch.randn(*ysh, device=dev).relu())
                      .to(torch.bfloat16).contiguous())
         K = N
-    dy = bf(G, M, dims[-1])
+    dy = bf(G, M, dims[-1]) if G > 1 else bf(M, dims[-1])
     ref = ext.mlp_chain_dx_bf16(dy, wts, youts, K0, flags, G, 1, 2)
     out = ext.mlp_chain_dx_bf16(dy, wts, youts, K0, flags, G, 1, 2, wps)
     d = max((a.float() - b.float()).abs().max().item()
