@@ -346,8 +346,18 @@ class CAREEngine(SACEngine):
                 bss.append(l.bias)
             mir.copy_(grp.flat_data)
             self.context_encoder_optimizer.bf16_mirror = mir
+
+            def _fpx(w, dx):
+                K = w.shape[-1]
+                N = w.numel() // K
+                r, c = (K, N) if dx else (N, K)
+                return torch.empty(
+                    ((r + 15) // 16) * ((c + 31) // 32) * 512,
+                    dtype=torch.bfloat16, device=dev)
             self._ctx_chain = {"mir": mir, "ws16": ws16, "wt": wts,
-                               "bs": bss, "lins": lins}
+                               "bs": bss, "lins": lins,
+                               "fp": [_fpx(w, False) for w in ws16],
+                               "dxp": [_fpx(w, True) for w in ws16]}
 
     @torch.no_grad()
     def _refresh_mixT(self, which: str = "all") -> None:
@@ -701,6 +711,26 @@ class CAREEngine(SACEngine):
         nl_c = len(self._twin_local_bf16)
         nl_a = len(self._actor_ws_bf16)
         orig = not self.use_modified_care
+        chain0 = self._use_chain
+        if chain0:
+            # ONE launch re-packs every pre-step weight set
+            ws_pack = (list(self._twin_local_bf16) * 2
+                       + list(self._twin_target_bf16)
+                       + list(self._actor_ws_bf16) * 2)
+            ps_pack = (list(self._twin_local_fp)
+                       + list(self._twin_local_dxp)
+                       + list(self._twin_target_fp)
+                       + list(self._actor_fp) + list(self._actor_dxp))
+            gs_pack = [2] * (3 * nl_c) + [1] * (2 * nl_a)
+            dx_pack = ([0] * nl_c + [1] * nl_c + [0] * nl_c
+                       + [0] * nl_a + [1] * nl_a)
+            if orig:
+                cc0 = self._ctx_chain
+                ws_pack += list(cc0["ws16"]) * 2
+                ps_pack += list(cc0["fp"]) + list(cc0["dxp"])
+                gs_pack += [1] * (2 * len(cc0["ws16"]))
+                dx_pack += [0] * len(cc0["ws16"]) + [1] * len(cc0["ws16"])
+            ext.pack_weights_frag(ws_pack, ps_pack, gs_pack, dx_pack)
 
         if orig:
             # original CARE: TRAINABLE context encoder (embedding header +
@@ -711,9 +741,10 @@ class CAREEngine(SACEngine):
             cc = self._ctx_chain
             idx = states[:, -T:].argmax(dim=1)
             emb = torch.relu(self.context_encoder.embedding[0](idx))
-            z_context, acts_ctx = self._chain_fwd(emb, None, cc["ws16"],
-                                                  cc["bs"], G=1,
-                                                  act_last=0, out_f32=True)
+            z_context, acts_ctx = self._chain_fwd(
+                emb, None, cc["ws16"], cc["bs"], G=1, act_last=0,
+                out_f32=True,
+                wps=cc["fp"] if self._use_chain else None)
         else:
             z_context = self.context_encoder(states)     # frozen embedding
             acts_ctx = None
@@ -733,7 +764,8 @@ class CAREEngine(SACEngine):
         chain = self._use_chain
         if chain:
             out, acts_a = self._chain_fwd(enc_cat, None,
-                                          self._actor_ws_bf16, bs_f32, G=1)
+                                          self._actor_ws_bf16, bs_f32,
+                                          G=1, wps=self._actor_fp)
         else:
             out, acts_a = self._mlp_fwd_manual(enc_cat, self._actor_ws_bf16,
                                                bs_f32)
@@ -754,7 +786,8 @@ class CAREEngine(SACEngine):
         if chain:
             yt, _ = self._chain_fwd(enc_t, na, self._twin_target_bf16,
                                     self._twin_target[1], G=2,
-                                    save_acts=False)
+                                    save_acts=False,
+                                    wps=self._twin_target_fp)
             q1_t, q2_t = yt[0], yt[1]
         else:
             xt = torch.cat([enc_t, na.to(torch.bfloat16)], dim=-1)
@@ -768,7 +801,8 @@ class CAREEngine(SACEngine):
         if chain:
             yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
                                          self._twin_local_bf16,
-                                         self._twin_local[1], G=2)
+                                         self._twin_local[1], G=2,
+                                         wps=self._twin_local_fp)
             q1, q2 = yq[0], yq[1]
             head_in_dim = acts_q[0].shape[-1]
         else:
@@ -786,16 +820,13 @@ class CAREEngine(SACEngine):
                                             self.critic_group.numel, B)
         base_c = fg_c.data_ptr()
         if chain:
-            ext.transpose_weights_bf16(
-                list(self._twin_local_bf16) + list(self._actor_ws_bf16),
-                list(self._twin_local_wt) + list(self._actor_wt),
-                [2] * nl_c + [1] * nl_a)
             empty_h = states.new_empty(0, dtype=torch.bfloat16)
             youts_q = [acts_q[i + 1] for i in range(nl_c - 1)] + [empty_h]
             aflags_q = [1] * (nl_c - 1) + [0]
             outs = ext.mlp_chain_dx_bf16(dy, list(self._twin_local_wt),
                                          youts_q, head_in_dim, aflags_q,
-                                         2, 1, 0)
+                                         2, 1, 0,
+                                         list(self._twin_local_dxp))
             dys_q, dx0f = outs[:-1], outs[-1]
             dx0 = (dx0f[0] + dx0f[1]).to(torch.bfloat16)  # [B, se+A]
             ext.dwdb_grouped_arena(
@@ -908,14 +939,12 @@ class CAREEngine(SACEngine):
             # context arena, Adam stepped at the end of the update
             cc = self._ctx_chain
             nlx = len(cc["ws16"])
-            ext.transpose_weights_bf16(list(cc["ws16"]), list(cc["wt"]),
-                                       [1] * nlx)
             dzc = dx0[:, :zc_dim].contiguous()
             empty_h = states.new_empty(0, dtype=torch.bfloat16)
             youts_x = [acts_ctx[i + 1] for i in range(nlx - 1)] + [empty_h]
             dys_x = ext.mlp_chain_dx_bf16(
                 dzc, list(cc["wt"]), youts_x, acts_ctx[0].shape[-1],
-                [1] * (nlx - 1) + [0], 1, 1, -1)
+                [1] * (nlx - 1) + [0], 1, 1, -1, list(cc["dxp"]))
             fgx = self.context_group.flat_grad
             self.context_group.rebind_grads()
             basex = fgx.data_ptr()
@@ -956,8 +985,14 @@ class CAREEngine(SACEngine):
         enc_c, _ = self._se_fwd_manual(info, states[:, :sd].to(torch.bfloat16),
                                        zc16)               # post-step SE
         if chain:
+            # re-pack the POST-Adam critic (fwd + dx) in one launch
+            ext.pack_weights_frag(
+                list(self._twin_local_bf16) * 2,
+                list(self._twin_local_fp) + list(self._twin_local_dxp),
+                [2] * (2 * nl_c), [0] * nl_c + [1] * nl_c)
             ya, acts_f = self._chain_fwd(enc_c, sa, self._twin_local_bf16,
-                                         self._twin_local[1], G=2)
+                                         self._twin_local[1], G=2,
+                                         wps=self._twin_local_fp)
             aq1, aq2 = ya[0], ya[1]
         else:
             xa = torch.cat([enc_c, sa.to(torch.bfloat16)], dim=-1)
@@ -971,15 +1006,12 @@ class CAREEngine(SACEngine):
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)
         if chain:
-            ext.transpose_weights_bf16(list(self._twin_local_bf16),
-                                       list(self._twin_local_wt),
-                                       [2] * nl_c)
             empty_h = states.new_empty(0, dtype=torch.bfloat16)
             youts_f = [acts_f[i + 1] for i in range(nl_c - 1)] + [empty_h]
             outs = ext.mlp_chain_dx_bf16(
                 daq, list(self._twin_local_wt), youts_f,
                 acts_f[0].shape[-1], [1] * (nl_c - 1) + [0], 2, 0,
-                enc_c.shape[1])
+                enc_c.shape[1], list(self._twin_local_dxp))
             dx0a = outs[-1]
             dsa = dx0a[0] + dx0a[1]
         else:
@@ -1007,7 +1039,8 @@ class CAREEngine(SACEngine):
                 + [empty_h]
             dys_a = ext.mlp_chain_dx_bf16(
                 dhead, list(self._actor_wt), youts_a,
-                acts_a[0].shape[-1], [1] * (nl_a - 1) + [0], 1, 1, -1)
+                acts_a[0].shape[-1], [1] * (nl_a - 1) + [0], 1, 1, -1,
+                list(self._actor_dxp))
             ext.dwdb_grouped_arena(
                 list(dys_a), [acts_a[i][B:] for i in range(nl_a)],
                 arena_a,

@@ -238,10 +238,9 @@ class SACEngine:
             off = self.actor_group.offsets[i]
             self._actor_ws_bf16.append(
                 self._actor_bf16[off:off + w.numel()].view_as(w))
-        # TRANSPOSED weight mirrors for the fused dx-chain kernels
-        # (re-materialized per update by transpose_weights_bf16: critic
-        # pre-step at seg1 for the critic backward, critic post-step in
-        # seg2 for the actor-side dx, actor pre-step at seg1)
+        # TRANSPOSED weight mirrors (shape carriers for the dx-chain
+        # binding; the VALUES now come from the fragment-packed mirrors
+        # below, so these are never refreshed on the packed path)
         self._twin_local_wt = [
             torch.empty(w.shape[0], w.shape[2], w.shape[1],
                         dtype=torch.bfloat16, device=dev)
@@ -249,6 +248,31 @@ class SACEngine:
         self._actor_wt = [
             torch.empty(w.shape[1], w.shape[0], dtype=torch.bfloat16,
                         device=dev) for w in self._actor_ws_bf16]
+
+        # FRAGMENT-PACKED weight mirrors (k_bf16_pack_frag): the chain
+        # kernels' B-operand loads become base + lane*16 unit-stride
+        # (one contiguous 1 KiB line per wave) instead of 16 scattered
+        # cache lines — re-packed per update (pre-step set at seg1,
+        # post-critic-Adam set in seg2)
+        def _fp(w, G):
+            K = w.shape[-1]
+            N = w.numel() // (G * K)
+            return torch.empty(
+                G * ((N + 15) // 16) * ((K + 31) // 32) * 512,
+                dtype=torch.bfloat16, device=dev)
+
+        def _dxp(w, G):
+            K = w.shape[-1]
+            N = w.numel() // (G * K)
+            return torch.empty(
+                G * ((K + 15) // 16) * ((N + 31) // 32) * 512,
+                dtype=torch.bfloat16, device=dev)
+
+        self._twin_local_fp = [_fp(w, 2) for w in self._twin_local_bf16]
+        self._twin_target_fp = [_fp(w, 2) for w in self._twin_target_bf16]
+        self._actor_fp = [_fp(w, 1) for w in self._actor_ws_bf16]
+        self._twin_local_dxp = [_dxp(w, 2) for w in self._twin_local_bf16]
+        self._actor_dxp = [_dxp(w, 1) for w in self._actor_ws_bf16]
         self.refresh_bf16()
 
     @staticmethod
@@ -556,13 +580,14 @@ class SACEngine:
 
     @torch.no_grad()
     def _chain_fwd(self, x1, x2, ws_bf16, bs_f32, G, act_last=0,
-                   out_f32=True, rowcat=False, save_acts=True):
+                   out_f32=True, rowcat=False, save_acts=True, wps=None):
         from ..ops import native
         out = native().mlp_chain_fwd_bf16(
             x1, x2 if x2 is not None else x1.new_empty(0),
             list(ws_bf16), [b.contiguous() for b in bs_f32],
             int(act_last), int(G), 1 if out_f32 else 0, 0,
-            1 if rowcat else 0, 1 if save_acts else 0)
+            1 if rowcat else 0, 1 if save_acts else 0,
+            list(wps) if wps is not None else [])
         return out[0], [out[1]] + list(out[2:])
 
     @torch.no_grad()
@@ -583,13 +608,29 @@ class SACEngine:
         A = self.cfg.action_dim
         la_det = self.log_alpha.detach()
         chain = self._use_chain
+        if chain:
+            # ONE launch re-packs every pre-step weight set into the
+            # fragment layout (critic fwd+dx, target fwd, actor fwd+dx)
+            nl_c0 = len(self._twin_local_bf16)
+            nl_a0 = len(self._actor_ws_bf16)
+            ext.pack_weights_frag(
+                list(self._twin_local_bf16) * 2
+                + list(self._twin_target_bf16)
+                + list(self._actor_ws_bf16) * 2,
+                list(self._twin_local_fp) + list(self._twin_local_dxp)
+                + list(self._twin_target_fp)
+                + list(self._actor_fp) + list(self._actor_dxp),
+                [2] * (3 * nl_c0) + [1] * (2 * nl_a0),
+                [0] * nl_c0 + [1] * nl_c0 + [0] * nl_c0
+                + [0] * nl_a0 + [1] * nl_a0)
 
         # ---- batched actor forward + squash --------------------------
         ws_f32, bs_f32 = self._actor_weights()
         if chain:
             out, acts_a = self._chain_fwd(next_states, states,
                                           self._actor_ws_bf16, bs_f32,
-                                          G=1, rowcat=True)
+                                          G=1, rowcat=True,
+                                          wps=self._actor_fp)
         else:
             x_cat = torch.cat([next_states, states], dim=0)
             out, acts_a = self._mlp_fwd_manual(x_cat, self._actor_ws_bf16,
@@ -608,7 +649,8 @@ class SACEngine:
             yt, _ = self._chain_fwd(next_states, na,
                                     self._twin_target_bf16,
                                     self._twin_target[1], G=2,
-                                    save_acts=False)
+                                    save_acts=False,
+                                    wps=self._twin_target_fp)
             q1_t, q2_t = yt[0], yt[1]
         else:
             xt = torch.cat([next_states, na], dim=-1)
@@ -621,7 +663,8 @@ class SACEngine:
         if chain:
             yq, acts_c = self._chain_fwd(states, actions,
                                          self._twin_local_bf16,
-                                         self._twin_local[1], G=2)
+                                         self._twin_local[1], G=2,
+                                         wps=self._twin_local_fp)
             q1, q2 = yq[0], yq[1]
         else:
             x = torch.cat([states, actions], dim=-1)
@@ -638,18 +681,13 @@ class SACEngine:
                                             self.critic_group.numel, B)
         if chain:
             nl_c = len(self._twin_local_bf16)
-            # refresh W^T mirrors: critic pre-step (this backward) and
-            # actor pre-step (seg2's actor dW dx walk) — ONE launch
-            ext.transpose_weights_bf16(
-                list(self._twin_local_bf16) + list(self._actor_ws_bf16),
-                list(self._twin_local_wt) + list(self._actor_wt),
-                [2] * nl_c + [1] * len(self._actor_ws_bf16))
             empty_h = states.new_empty(0, dtype=torch.bfloat16)
             youts = [acts_c[i + 1] for i in range(nl_c - 1)] + [empty_h]
             aflags = [1] * (nl_c - 1) + [0]
             dys = ext.mlp_chain_dx_bf16(dy, list(self._twin_local_wt),
                                         youts, acts_c[0].shape[-1],
-                                        aflags, 2, 1, -1)
+                                        aflags, 2, 1, -1,
+                                        list(self._twin_local_dxp))
             base_c = fg_c.data_ptr()
             ext.dwdb_grouped_arena(
                 list(dys), [acts_c[i] for i in range(nl_c)], arena_c,
@@ -689,8 +727,14 @@ class SACEngine:
         acts_a = st["acts_a"]
 
         if self._use_chain:
+            # re-pack the POST-Adam critic (fwd + dx) in one launch
+            ext.pack_weights_frag(
+                list(self._twin_local_bf16) * 2,
+                list(self._twin_local_fp) + list(self._twin_local_dxp),
+                [2] * (2 * nl_c), [0] * nl_c + [1] * nl_c)
             ya, acts_f = self._chain_fwd(states, sa, self._twin_local_bf16,
-                                         self._twin_local[1], G=2)
+                                         self._twin_local[1], G=2,
+                                         wps=self._twin_local_fp)
             aq1, aq2 = ya[0], ya[1]
         else:
             xa = torch.cat([states, sa], dim=-1)
@@ -703,16 +747,12 @@ class SACEngine:
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)
         if self._use_chain:
-            # critic W^T mirrors must reflect the POST-step critic
-            ext.transpose_weights_bf16(list(self._twin_local_bf16),
-                                       list(self._twin_local_wt),
-                                       [2] * nl_c)
             empty_h = states.new_empty(0, dtype=torch.bfloat16)
             youts_f = [acts_f[i + 1] for i in range(nl_c - 1)] + [empty_h]
             outs = ext.mlp_chain_dx_bf16(
                 daq, list(self._twin_local_wt), youts_f,
                 acts_f[0].shape[-1], [1] * (nl_c - 1) + [0], 2, 0,
-                states.shape[1])
+                states.shape[1], list(self._twin_local_dxp))
             dx0 = outs[-1]          # [2, B, A] fp32 (action columns only)
             dsa = dx0[0] + dx0[1]   # sum over the twin Q heads
         else:
@@ -742,7 +782,8 @@ class SACEngine:
                 + [empty_h]
             dys_a = ext.mlp_chain_dx_bf16(
                 dhead, list(self._actor_wt), youts_a,
-                acts_a[0].shape[-1], [1] * (nl_a - 1) + [0], 1, 1, -1)
+                acts_a[0].shape[-1], [1] * (nl_a - 1) + [0], 1, 1, -1,
+                list(self._actor_dxp))
             ext.dwdb_grouped_arena(
                 list(dys_a), [acts_a[i][B:] for i in range(nl_a)],
                 arena_a,

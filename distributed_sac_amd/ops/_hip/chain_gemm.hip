@@ -376,10 +376,10 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
 // guard paths disappear.  One launch packs up to 24 matrices.
 // ---------------------------------------------------------------------------
 struct PackDesc {
-  const u16* w[24];
-  u16* p[24];
-  int N[24], K[24], G[24], dx[24];
-  int cum[25];   // cumulative output chunks (G*NT*KS*64 per matrix)
+  const u16* w[32];
+  u16* p[32];
+  int N[32], K[32], G[32], dx[32];
+  int cum[33];   // cumulative output chunks (G*NT*KS*64 per matrix)
   int C;
 };
 
@@ -424,7 +424,7 @@ static void pack_weights_frag(std::vector<torch::Tensor> ws,
                               std::vector<long> Gs,
                               std::vector<long> dxs) {
   const int C = (int)ws.size();
-  TORCH_CHECK(C >= 1 && C <= 24 && (int)ps.size() == C
+  TORCH_CHECK(C >= 1 && C <= 32 && (int)ps.size() == C
               && (int)Gs.size() == C && (int)dxs.size() == C);
   PackDesc d{};
   d.C = C;
