@@ -614,7 +614,9 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
     const int K = li > 0 ? d.N[li - 1] : d.K0;
 
     // ---- mask by relu' + save masked dy --------------------------------
-    {
+    // only the ENTRY layer needs this standalone pass: deeper layers'
+    // masks are fused into the producing GEMM's epilogue below
+    if (li == d.L - 1) {
       const int r = tid >> 5;
       const int lc = tid & 31;
       const u16* yo = d.act[li] && d.yout[li] != nullptr
@@ -630,8 +632,8 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
           if (sv != nullptr) sv[row * N + c] = v;
         }
       }
+      __syncthreads();
     }
-    __syncthreads();
 
     const bool want_dx0 = li == 0 && d.dx0_lo >= 0;
     if (li == 0 && !want_dx0) break;
@@ -767,6 +769,15 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
 #undef LOADQ
 #undef MF4
 
+      // epilogue: when producing dy_{li-1}, apply ITS relu' mask and
+      // write ITS dysave right here (saves the standalone pass + two
+      // barriers per layer)
+      const int ln = li - 1;
+      const u16* yo_n = (!want_dx0 && ln >= 0 && d.act[ln]
+                         && d.yout[ln] != nullptr)
+          ? d.yout[ln] + (long)g * d.yo_gs[ln] : nullptr;
+      u16* sv_n = (!want_dx0 && ln >= 0 && d.dysave[ln] != nullptr)
+          ? d.dysave[ln] + (long)g * M * K : nullptr;
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (q >= nq) break;
@@ -781,7 +792,13 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
                 dx0g[(long)(m0 + row) * (K - col_lo) + (col - col_lo)] =
                     a[r];
             } else {
-              sd[nb][row][col] = f32_to_bf16_rne3(a[r]);
+              u16 v = f32_to_bf16_rne3(a[r]);
+              if (yo_n != nullptr && row < rowlim
+                  && yo_n[(long)(m0 + row) * K + col] == 0)
+                v = 0;
+              sd[nb][row][col] = v;
+              if (sv_n != nullptr && row < rowlim)
+                sv_n[(long)(m0 + row) * K + col] = v;
             }
           }
         }
