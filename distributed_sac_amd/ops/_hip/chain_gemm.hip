@@ -140,13 +140,27 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
           v = x2f ? ((const float*)x2)[row * C2 + (c - C1)]
                   : (float)((const __bf16*)x2)[row * C2 + (c - C1)];
         }
-        const u16 h = f32_to_bf16_rne3(v);
-        sa[0][rr][c] = h;
-        if (xsave != nullptr && g == 0) xsave[row * K0 + c] = h;
+        sa[0][rr][c] = f32_to_bf16_rne3(v);
       }
     }
   }
   __syncthreads();
+  // coalesced bf16 input save (backward consumes it): 16-B chunks from
+  // LDS instead of the loader's scalar stores (TA relief)
+  if (xsave != nullptr && g == 0) {
+    const int r = tid >> 5;
+    const int lc = tid & 31;
+    for (int rr = r; rr < rowlim; rr += NTHR / 32) {
+      u16* dst = xsave + (long)(m0 + rr) * K0;
+      const u16* src = &sa[0][rr][0];
+      for (int c = lc * 8; c + 8 <= K0; c += (NTHR / 32) * 8)
+        *(uint4*)&dst[c] = *(const uint4*)&src[c];
+    }
+    if ((K0 & 7) && tid < TMv && tid < rowlim) {
+      for (int c = K0 & ~7; c < K0; ++c)
+        xsave[(long)(m0 + tid) * K0 + c] = sa[0][tid][c];
+    }
+  }
 
   int cur = 0;
   int prevw0 = K0, prevw1 = CMAX;  // written widths of buffers 0/1
@@ -355,6 +369,22 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_fwd(
       }
       if (nb) prevw1 = N; else prevw0 = N;
       __syncthreads();
+      // coalesced activation save from LDS (the epilogue only wrote
+      // LDS; a 16-B-chunk copy replaces 4x as many scalar 2-B stores)
+      if (ag != nullptr) {
+        const int r2 = tid >> 5;
+        const int lc2 = tid & 31;
+        for (int rr = r2; rr < rowlim; rr += NTHR / 32) {
+          u16* dst = ag + (long)(m0 + rr) * N;
+          const u16* src = &sa[nb][rr][0];
+          for (int c = lc2 * 8; c + 8 <= N; c += (NTHR / 32) * 8)
+            *(uint4*)&dst[c] = *(const uint4*)&src[c];
+        }
+        if ((N & 7) && tid < TMv && tid < rowlim) {
+          for (int c = N & ~7; c < N; ++c)
+            ag[(long)(m0 + tid) * N + c] = sa[nb][tid][c];
+        }
+      }
       cur = nb;
       K = N;
     }
@@ -640,6 +670,12 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
 
     // ---- GEMM: out[16, K] = dy_pre[16, N] @ Wt[K, N] -------------------
     const u16* wtg = d.wt[li] + (long)g * N * K;   // [K][N]
+    const int ln = li - 1;
+    const u16* yo_n = (!want_dx0 && ln >= 0 && d.act[ln]
+                       && d.yout[ln] != nullptr)
+        ? d.yout[ln] + (long)g * d.yo_gs[ln] : nullptr;
+    u16* sv_n = (!want_dx0 && ln >= 0 && d.dysave[ln] != nullptr)
+        ? d.dysave[ln] + (long)g * M * K : nullptr;
     const int col_lo = want_dx0 ? d.dx0_lo : 0;
     const int t_lo = col_lo >> 4;
     const int ntiles = (K + 15) >> 4;
@@ -769,15 +805,8 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
 #undef LOADQ
 #undef MF4
 
-      // epilogue: when producing dy_{li-1}, apply ITS relu' mask and
-      // write ITS dysave right here (saves the standalone pass + two
-      // barriers per layer)
-      const int ln = li - 1;
-      const u16* yo_n = (!want_dx0 && ln >= 0 && d.act[ln]
-                         && d.yout[ln] != nullptr)
-          ? d.yout[ln] + (long)g * d.yo_gs[ln] : nullptr;
-      u16* sv_n = (!want_dx0 && ln >= 0 && d.dysave[ln] != nullptr)
-          ? d.dysave[ln] + (long)g * M * K : nullptr;
+      // epilogue: when producing dy_{li-1}, apply ITS relu' mask here
+      // (its dysave is written by the coalesced copy after the barrier)
 #pragma unroll
       for (int q = 0; q < 4; ++q) {
         if (q >= nq) break;
@@ -797,8 +826,6 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
                   && yo_n[(long)(m0 + row) * K + col] == 0)
                 v = 0;
               sd[nb][row][col] = v;
-              if (sv_n != nullptr && row < rowlim)
-                sv_n[(long)(m0 + row) * K + col] = v;
             }
           }
         }
@@ -814,6 +841,21 @@ __global__ __launch_bounds__(NTHR) void k_bf16_chain_dx(
       if (nb) prevw1 = K; else prevw0 = K;
     }
     __syncthreads();
+    // coalesced masked-dy save from LDS for the layer just produced
+    if (sv_n != nullptr) {
+      const int r2 = tid >> 5;
+      const int lc2 = tid & 31;
+      for (int rr = r2; rr < rowlim; rr += NTHR / 32) {
+        u16* dst = sv_n + (long)(m0 + rr) * K;
+        const u16* src = &sd[nb][rr][0];
+        for (int c = lc2 * 8; c + 8 <= K; c += (NTHR / 32) * 8)
+          *(uint4*)&dst[c] = *(const uint4*)&src[c];
+      }
+      if ((K & 7) && tid < TMv && tid < rowlim) {
+        for (int c = K & ~7; c < K; ++c)
+          sv_n[(long)(m0 + tid) * K + c] = sd[nb][tid][c];
+      }
+    }
     cur = nb;
   }
 }
