@@ -759,9 +759,24 @@ class CAREEngine(SACEngine):
         enc_cat, se_saved = self._se_fwd_manual(info, x_cat.to(torch.bfloat16),
                                                 zc16, save=True, rep=2)
 
+        # ---- critic-twin head on a SIDE stream: it needs only enc_cat
+        # (just produced) + actions, so it records as a parallel graph
+        # branch under the actor head / squash / target-SE / TD work
+        chain = self._use_chain
+        yq = acts_q = None
+        if chain:
+            if self._side_stream is None:
+                self._side_stream = torch.cuda.Stream(self.device)
+            s2 = self._side_stream
+            s2.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(s2):
+                yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
+                                             self._twin_local_bf16,
+                                             self._twin_local[1], G=2,
+                                             wps=self._twin_local_fp)
+
         # ---- batched actor head + squash ------------------------------
         ws_f32, bs_f32 = self._actor_weights()
-        chain = self._use_chain
         if chain:
             out, acts_a = self._chain_fwd(enc_cat, None,
                                           self._actor_ws_bf16, bs_f32,
@@ -799,10 +814,12 @@ class CAREEngine(SACEngine):
 
         # ---- critic loss + manual backward ----------------------------
         if chain:
-            yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
-                                         self._twin_local_bf16,
-                                         self._twin_local[1], G=2,
-                                         wps=self._twin_local_fp)
+            cur = torch.cuda.current_stream(self.device)
+            cur.wait_stream(self._side_stream)
+            if not torch.cuda.is_current_stream_capturing():
+                yq.record_stream(cur)
+                for t in acts_q:
+                    t.record_stream(cur)
             q1, q2 = yq[0], yq[1]
             head_in_dim = acts_q[0].shape[-1]
         else:

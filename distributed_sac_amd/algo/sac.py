@@ -624,6 +624,22 @@ class SACEngine:
                 [0] * nl_c0 + [1] * nl_c0 + [0] * nl_c0
                 + [0] * nl_a0 + [1] * nl_a0)
 
+        # ---- critic-twin forward on a SIDE stream (independent of the
+        # actor/target path: it needs only states/actions + pre-step
+        # critic) — inside a captured graph this records a parallel
+        # branch, overlapping ~a third of seg1's serial kernel time
+        yq = acts_c = None
+        if chain:
+            if self._side_stream is None:
+                self._side_stream = torch.cuda.Stream(self.device)
+            s2 = self._side_stream
+            s2.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(s2):
+                yq, acts_c = self._chain_fwd(states, actions,
+                                             self._twin_local_bf16,
+                                             self._twin_local[1], G=2,
+                                             wps=self._twin_local_fp)
+
         # ---- batched actor forward + squash --------------------------
         ws_f32, bs_f32 = self._actor_weights()
         if chain:
@@ -661,10 +677,12 @@ class SACEngine:
 
         # ---- critic loss + manual backward ---------------------------
         if chain:
-            yq, acts_c = self._chain_fwd(states, actions,
-                                         self._twin_local_bf16,
-                                         self._twin_local[1], G=2,
-                                         wps=self._twin_local_fp)
+            cur = torch.cuda.current_stream(self.device)
+            cur.wait_stream(self._side_stream)
+            if not torch.cuda.is_current_stream_capturing():
+                yq.record_stream(cur)
+                for t in acts_c:
+                    t.record_stream(cur)
             q1, q2 = yq[0], yq[1]
         else:
             x = torch.cat([states, actions], dim=-1)
