@@ -26,6 +26,8 @@ in the actor step) run under no_grad — numerically identical updates.
 
 from __future__ import annotations
 
+import os
+
 from typing import Dict, Optional
 
 import torch
@@ -764,7 +766,8 @@ class CAREEngine(SACEngine):
         # branch under the actor head / squash / target-SE / TD work
         chain = self._use_chain
         yq = acts_q = None
-        if chain:
+        overlap = chain and os.environ.get("DSAC_SEG1_OVERLAP", "1") != "0"
+        if overlap:
             if self._side_stream is None:
                 self._side_stream = torch.cuda.Stream(self.device)
             s2 = self._side_stream
@@ -814,12 +817,18 @@ class CAREEngine(SACEngine):
 
         # ---- critic loss + manual backward ----------------------------
         if chain:
-            cur = torch.cuda.current_stream(self.device)
-            cur.wait_stream(self._side_stream)
-            if not torch.cuda.is_current_stream_capturing():
-                yq.record_stream(cur)
-                for t in acts_q:
-                    t.record_stream(cur)
+            if overlap:
+                cur = torch.cuda.current_stream(self.device)
+                cur.wait_stream(self._side_stream)
+                if not torch.cuda.is_current_stream_capturing():
+                    yq.record_stream(cur)
+                    for t in acts_q:
+                        t.record_stream(cur)
+            else:
+                yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
+                                             self._twin_local_bf16,
+                                             self._twin_local[1], G=2,
+                                             wps=self._twin_local_fp)
             q1, q2 = yq[0], yq[1]
             head_in_dim = acts_q[0].shape[-1]
         else:

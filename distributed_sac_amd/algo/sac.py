@@ -23,6 +23,8 @@ MI355X-first mechanics under the reference math:
 
 from __future__ import annotations
 
+import os
+
 from typing import Dict, Optional
 
 import torch
@@ -629,7 +631,8 @@ class SACEngine:
         # critic) — inside a captured graph this records a parallel
         # branch, overlapping ~a third of seg1's serial kernel time
         yq = acts_c = None
-        if chain:
+        overlap = chain and os.environ.get("DSAC_SEG1_OVERLAP", "1") != "0"
+        if overlap:
             if self._side_stream is None:
                 self._side_stream = torch.cuda.Stream(self.device)
             s2 = self._side_stream
@@ -677,12 +680,18 @@ class SACEngine:
 
         # ---- critic loss + manual backward ---------------------------
         if chain:
-            cur = torch.cuda.current_stream(self.device)
-            cur.wait_stream(self._side_stream)
-            if not torch.cuda.is_current_stream_capturing():
-                yq.record_stream(cur)
-                for t in acts_c:
-                    t.record_stream(cur)
+            if overlap:
+                cur = torch.cuda.current_stream(self.device)
+                cur.wait_stream(self._side_stream)
+                if not torch.cuda.is_current_stream_capturing():
+                    yq.record_stream(cur)
+                    for t in acts_c:
+                        t.record_stream(cur)
+            else:
+                yq, acts_c = self._chain_fwd(states, actions,
+                                             self._twin_local_bf16,
+                                             self._twin_local[1], G=2,
+                                             wps=self._twin_local_fp)
             q1, q2 = yq[0], yq[1]
         else:
             x = torch.cat([states, actions], dim=-1)
