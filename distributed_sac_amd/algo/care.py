@@ -26,8 +26,6 @@ in the actor step) run under no_grad — numerically identical updates.
 
 from __future__ import annotations
 
-import os
-
 from typing import Dict, Optional
 
 import torch
@@ -761,25 +759,9 @@ class CAREEngine(SACEngine):
         enc_cat, se_saved = self._se_fwd_manual(info, x_cat.to(torch.bfloat16),
                                                 zc16, save=True, rep=2)
 
-        # ---- critic-twin head on a SIDE stream: it needs only enc_cat
-        # (just produced) + actions, so it records as a parallel graph
-        # branch under the actor head / squash / target-SE / TD work
-        chain = self._use_chain
-        yq = acts_q = None
-        overlap = chain and os.environ.get("DSAC_SEG1_OVERLAP", "1") != "0"
-        if overlap:
-            if self._side_stream is None:
-                self._side_stream = torch.cuda.Stream(self.device)
-            s2 = self._side_stream
-            s2.wait_stream(torch.cuda.current_stream(self.device))
-            with torch.cuda.stream(s2):
-                yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
-                                             self._twin_local_bf16,
-                                             self._twin_local[1], G=2,
-                                             wps=self._twin_local_fp)
-
         # ---- batched actor head + squash ------------------------------
         ws_f32, bs_f32 = self._actor_weights()
+        chain = self._use_chain
         if chain:
             out, acts_a = self._chain_fwd(enc_cat, None,
                                           self._actor_ws_bf16, bs_f32,
@@ -817,18 +799,10 @@ class CAREEngine(SACEngine):
 
         # ---- critic loss + manual backward ----------------------------
         if chain:
-            if overlap:
-                cur = torch.cuda.current_stream(self.device)
-                cur.wait_stream(self._side_stream)
-                if not torch.cuda.is_current_stream_capturing():
-                    yq.record_stream(cur)
-                    for t in acts_q:
-                        t.record_stream(cur)
-            else:
-                yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
-                                             self._twin_local_bf16,
-                                             self._twin_local[1], G=2,
-                                             wps=self._twin_local_fp)
+            yq, acts_q = self._chain_fwd(enc_cat[B:], actions,
+                                         self._twin_local_bf16,
+                                         self._twin_local[1], G=2,
+                                         wps=self._twin_local_fp)
             q1, q2 = yq[0], yq[1]
             head_in_dim = acts_q[0].shape[-1]
         else:

@@ -23,8 +23,6 @@ MI355X-first mechanics under the reference math:
 
 from __future__ import annotations
 
-import os
-
 from typing import Dict, Optional
 
 import torch
@@ -626,23 +624,6 @@ class SACEngine:
                 [0] * nl_c0 + [1] * nl_c0 + [0] * nl_c0
                 + [0] * nl_a0 + [1] * nl_a0)
 
-        # ---- critic-twin forward on a SIDE stream (independent of the
-        # actor/target path: it needs only states/actions + pre-step
-        # critic) — inside a captured graph this records a parallel
-        # branch, overlapping ~a third of seg1's serial kernel time
-        yq = acts_c = None
-        overlap = chain and os.environ.get("DSAC_SEG1_OVERLAP", "1") != "0"
-        if overlap:
-            if self._side_stream is None:
-                self._side_stream = torch.cuda.Stream(self.device)
-            s2 = self._side_stream
-            s2.wait_stream(torch.cuda.current_stream(self.device))
-            with torch.cuda.stream(s2):
-                yq, acts_c = self._chain_fwd(states, actions,
-                                             self._twin_local_bf16,
-                                             self._twin_local[1], G=2,
-                                             wps=self._twin_local_fp)
-
         # ---- batched actor forward + squash --------------------------
         ws_f32, bs_f32 = self._actor_weights()
         if chain:
@@ -680,18 +661,10 @@ class SACEngine:
 
         # ---- critic loss + manual backward ---------------------------
         if chain:
-            if overlap:
-                cur = torch.cuda.current_stream(self.device)
-                cur.wait_stream(self._side_stream)
-                if not torch.cuda.is_current_stream_capturing():
-                    yq.record_stream(cur)
-                    for t in acts_c:
-                        t.record_stream(cur)
-            else:
-                yq, acts_c = self._chain_fwd(states, actions,
-                                             self._twin_local_bf16,
-                                             self._twin_local[1], G=2,
-                                             wps=self._twin_local_fp)
+            yq, acts_c = self._chain_fwd(states, actions,
+                                         self._twin_local_bf16,
+                                         self._twin_local[1], G=2,
+                                         wps=self._twin_local_fp)
             q1, q2 = yq[0], yq[1]
         else:
             x = torch.cat([states, actions], dim=-1)
