@@ -999,9 +999,11 @@ class CAREEngine(SACEngine):
             aq1, aq2, acts_f = self._twin_fwd_manual(xa,
                                                      self._twin_local_bf16,
                                                      self._twin_local[1])
+        # fwd kernel zeroes the alpha flat grad in passing (bwd atomics
+        # target) — no separate fill launch
         al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls_cat[B:], states,
-                                      la_det, T, int(use_w), self.H_bar_f)
-        self.alpha_group.flat_grad.zero_()
+                                      la_det, T, int(use_w), self.H_bar_f,
+                                      self.alpha_group.flat_grad)
         daq, dlp = ext.actor_alpha_loss_bwd2(
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)
@@ -1068,8 +1070,7 @@ class CAREEngine(SACEngine):
         from ..ops.flat import FusedAdam as _FA
         st = self._dp_st
         _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
-        self.alpha = self.log_alpha.exp().detach()
-
+        # self.alpha refreshed lazily outside the graph (see SACEngine)
         self._polyak_targets(mirror=self._target_bf16)
         self._refresh_mixT("target")
         if st["orig"]:
@@ -1112,7 +1113,7 @@ class CAREEngine(SACEngine):
             "actor_optimizer": self.actor_optimizer.state_dict(),
             "log_alpha": self.log_alpha.detach().cpu(),
             "log_alpha_optimizer": self.log_alpha_optimizer.state_dict(),
-            "alpha": self.alpha.cpu(),
+            "alpha": self.log_alpha.detach().exp().cpu(),
         }
 
     def load_checkpoint_state(self, ckpt: Dict) -> None:

@@ -363,7 +363,7 @@ class SACEngine:
         if self.variant == "mtsac":
             one_hots = mtobss[:, -self.num_tasks:]
             return Fops.gather_log_alpha(one_hots, self.log_alpha).exp().detach()
-        return self.alpha
+        return self.log_alpha.exp().detach()
 
     def update(self, batch: Dict[str, torch.Tensor]) -> Dict[str, float]:
         """One SAC gradient update; returns scalar metrics (syncs)."""
@@ -740,9 +740,11 @@ class SACEngine:
             xa = torch.cat([states, sa], dim=-1)
             aq1, aq2, acts_f = self._twin_fwd_manual(
                 xa, self._twin_local_bf16, self._twin_local[1])
+        # the fwd kernel zeroes the alpha flat grad in passing (the bwd's
+        # atomics target) — no separate fill launch
         al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls, states,
-                                      la_det, T, int(use_w), self.H_bar_f)
-        self.alpha_group.flat_grad.zero_()
+                                      la_det, T, int(use_w), self.H_bar_f,
+                                      self.alpha_group.flat_grad)
         daq, dlp = ext.actor_alpha_loss_bwd2(
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)
@@ -812,7 +814,9 @@ class SACEngine:
         st = self._dp_st
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer])
-        self.alpha = self.log_alpha.exp().detach()
+        # NOTE: self.alpha is refreshed lazily (checkpoint/_per_sample_alpha
+        # recompute from log_alpha) — an exp() here would replay as a
+        # ~5 µs kernel every captured step just for bookkeeping
         flat_polyak_(self.target_group, self.critic_group, self.tau,
                      mirror=getattr(self, "_target_bf16", None))
         closs, al = st["closs"], st["al"]
@@ -1013,7 +1017,7 @@ class SACEngine:
             "critic_optimizer": self.critic_optimizer.state_dict(),
             "log_alpha": self.log_alpha.detach().cpu(),
             "log_alpha_optimizer": self.log_alpha_optimizer.state_dict(),
-            "alpha": self.alpha.cpu(),
+            "alpha": self.log_alpha.detach().exp().cpu(),
         }
         if self.variant in ("sac", "vsac"):
             # LunarLander…/src/learner.py:144-163 key layout (LL names the

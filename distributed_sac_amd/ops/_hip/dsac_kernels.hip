@@ -548,25 +548,25 @@ __device__ __forceinline__ void fill_task_weights(
   }
 }
 
-__global__ void k_loss_zero(float* __restrict__ raw, int n) {
-  if ((int)threadIdx.x < n) raw[threadIdx.x] = 0.f;
-}
 
-// out: [0]=loss1 [1]=loss2 [2]=wsum, [3..5]=atomic raw sums (zeroed by
-// k_loss_zero); multi-block grid-stride + 1-thread finalize (the single-
-// workgroup version left 255/256 CUs idle for ~20us per loss)
-__global__ __launch_bounds__(256) void k_critic_loss_fwd(
+
+
+// single-workgroup fused variant: 1024 threads cover B<=few-thousand rows
+// in <=4 strides, LDS tree reduce, thread 0 writes the FINALIZED values —
+// one launch replaces zero + multi-block-atomic fwd + finalize (the
+// multi-block version was only ever launch-bound at these batch sizes)
+__global__ __launch_bounds__(1024) void k_critic_loss_fwd_1wg(
     const float* __restrict__ q1, const float* __restrict__ q2,
     const float* __restrict__ y, const float* __restrict__ onehot,
     const float* __restrict__ log_alpha, float* __restrict__ out,
     int B, int T, int oh_stride, int use_w) {
-  __shared__ float red[3][256];
+  __shared__ float red[3][1024];
   __shared__ float smw[32];
   const int tid = threadIdx.x;
   if (use_w) fill_task_weights(smw, log_alpha, T);
   __syncthreads();
   float s_l1 = 0.f, s_l2 = 0.f, s_w = 0.f;
-  for (int i = blockIdx.x * 256 + tid; i < B; i += 256 * gridDim.x) {
+  for (int i = tid; i < B; i += 1024) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float w_raw = use_w ? smw[t_i] : 1.f;
     s_w += w_raw;
@@ -576,26 +576,20 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd(
   }
   red[0][tid] = s_l1; red[1][tid] = s_l2; red[2][tid] = s_w;
   __syncthreads();
-  for (int off = 128; off > 0; off >>= 1) {
+  for (int off = 512; off > 0; off >>= 1) {
     if (tid < off)
 #pragma unroll
       for (int r = 0; r < 3; ++r) red[r][tid] += red[r][tid + off];
     __syncthreads();
   }
   if (tid == 0) {
-    atomicAdd(&out[3], red[0][0]);
-    atomicAdd(&out[4], red[1][0]);
-    atomicAdd(&out[5], red[2][0]);
+    const float wsum = use_w ? red[2][0] : 1.f;
+    const float denom = wsum * (float)B;
+    out[3] = red[0][0]; out[4] = red[1][0]; out[5] = red[2][0];
+    out[0] = red[0][0] / denom;
+    out[1] = red[1][0] / denom;
+    out[2] = wsum;
   }
-}
-
-__global__ void k_critic_loss_finalize(float* __restrict__ out, int B,
-                                       int use_w) {
-  const float wsum = use_w ? out[5] : 1.f;
-  const float denom = wsum * (float)B;
-  out[0] = out[3] / denom;
-  out[1] = out[4] / denom;
-  out[2] = wsum;
 }
 
 // bf16 stacked-output variant for the manual backward: dq[2,B] bf16
@@ -642,21 +636,28 @@ __global__ __launch_bounds__(256) void k_critic_loss_bwd(
   dq2[i] = gscale[1] * coeff * -2.f * (y[i] - q2[i]);
 }
 
-// out[0]=actor_loss out[1]=wsum out[2]=alpha_loss out[3]=entropy
-__global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd(
+
+
+// single-workgroup fused variant (see k_critic_loss_fwd_1wg); writes
+// finalized out[0..3] + raw sums out[4..7] in one launch.  If dla is
+// non-null its first dla_n floats are zeroed here — the alpha gradient
+// buffer the subsequent k_actor_alpha_loss_bwd* atomics target — so the
+// engine needs no separate fill launch either.
+__global__ __launch_bounds__(1024) void k_actor_alpha_loss_fwd_1wg(
     const float* __restrict__ aq1, const float* __restrict__ aq2,
     const float* __restrict__ lp, const float* __restrict__ ls,
     const float* __restrict__ onehot, const float* __restrict__ log_alpha,
-    float* __restrict__ out, int B, int T, int A, int oh_stride, int use_w,
-    float H_bar) {
-  __shared__ float red[4][256];
+    float* __restrict__ out, float* __restrict__ dla, int dla_n,
+    int B, int T, int A, int oh_stride, int use_w, float H_bar) {
+  __shared__ float red[4][1024];
   __shared__ float smw[32];
   const int tid = threadIdx.x;
+  if (dla != nullptr && tid < dla_n) dla[tid] = 0.f;
   if (use_w) fill_task_weights(smw, log_alpha, T);
   __syncthreads();
   constexpr float CE = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
   float s_pl = 0.f, s_w = 0.f, s_al = 0.f, s_en = 0.f;
-  for (int i = blockIdx.x * 256 + tid; i < B; i += 256 * gridDim.x) {
+  for (int i = tid; i < B; i += 1024) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float la = log_alpha[t_i];
     const float alpha_i = __expf(la);
@@ -672,27 +673,21 @@ __global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd(
   red[0][tid] = s_pl; red[1][tid] = s_w; red[2][tid] = s_al;
   red[3][tid] = s_en;
   __syncthreads();
-  for (int off = 128; off > 0; off >>= 1) {
+  for (int off = 512; off > 0; off >>= 1) {
     if (tid < off)
 #pragma unroll
       for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
     __syncthreads();
   }
   if (tid == 0) {
-    atomicAdd(&out[4], red[0][0]);
-    atomicAdd(&out[5], red[1][0]);
-    atomicAdd(&out[6], red[2][0]);
-    atomicAdd(&out[7], red[3][0]);
+    const float wsum = use_w ? red[1][0] : 1.f;
+    out[4] = red[0][0]; out[5] = red[1][0];
+    out[6] = red[2][0]; out[7] = red[3][0];
+    out[0] = red[0][0] / (wsum * (float)B);
+    out[1] = wsum;
+    out[2] = -red[2][0] / (float)B;
+    out[3] = red[3][0] / (float)B;
   }
-}
-
-__global__ void k_actor_alpha_loss_finalize(float* __restrict__ out, int B,
-                                            int use_w) {
-  const float wsum = use_w ? out[5] : 1.f;
-  out[0] = out[4] / (wsum * (float)B);
-  out[1] = wsum;
-  out[2] = -out[6] / (float)B;
-  out[3] = out[7] / (float)B;
 }
 
 // daq1_i = gp*coeff_i*-(aq1<=aq2); daq2_i = gp*coeff_i*-(aq2<aq1)
@@ -1153,16 +1148,10 @@ static std::vector<torch::Tensor> critic_loss_fwd(
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
   auto out = torch::empty({6}, q1.options());
-  hipLaunchKernelGGL(k_loss_zero, dim3(1), dim3(8), 0, cur_stream(),
-                     out.data_ptr<float>(), 6);
-  const int nblk = (int)std::min<long>((B + 255) / 256, 32);
-  hipLaunchKernelGGL(k_critic_loss_fwd, dim3(nblk), dim3(256), 0,
+  hipLaunchKernelGGL(k_critic_loss_fwd_1wg, dim3(1), dim3(1024), 0,
                      cur_stream(), q1.data_ptr<float>(), q2.data_ptr<float>(),
                      y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
                      out.data_ptr<float>(), (int)B, (int)T, (int)oh_stride,
-                     (int)use_w);
-  hipLaunchKernelGGL(k_critic_loss_finalize, dim3(1), dim3(1), 0,
-                     cur_stream(), out.data_ptr<float>(), (int)B,
                      (int)use_w);
   return {out};
 }
@@ -1189,25 +1178,27 @@ static std::vector<torch::Tensor> critic_loss_bwd(
 static torch::Tensor actor_alpha_loss_fwd(
     torch::Tensor aq1, torch::Tensor aq2, torch::Tensor lp, torch::Tensor ls,
     torch::Tensor states, torch::Tensor log_alpha, long T, long use_w,
-    double H_bar) {
+    double H_bar,
+    c10::optional<torch::Tensor> dla = c10::nullopt) {
   CHECK_IN(aq1); CHECK_IN(states); CHECK_IN(log_alpha);
   const long B = aq1.size(0);
   const long A = ls.size(1);
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
   auto out = torch::empty({8}, aq1.options());
-  hipLaunchKernelGGL(k_loss_zero, dim3(1), dim3(8), 0, cur_stream(),
-                     out.data_ptr<float>(), 8);
-  const int nblk = (int)std::min<long>((B + 255) / 256, 32);
-  hipLaunchKernelGGL(k_actor_alpha_loss_fwd, dim3(nblk), dim3(256), 0,
+  float* dla_p = nullptr;
+  int dla_n = 0;
+  if (dla.has_value() && dla->numel() > 0) {
+    CHECK_IN(*dla);
+    dla_p = dla->data_ptr<float>();
+    dla_n = (int)dla->numel();
+  }
+  hipLaunchKernelGGL(k_actor_alpha_loss_fwd_1wg, dim3(1), dim3(1024), 0,
                      cur_stream(), aq1.data_ptr<float>(),
                      aq2.data_ptr<float>(), lp.data_ptr<float>(),
                      ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
-                     out.data_ptr<float>(), (int)B, (int)T, (int)A,
-                     (int)oh_stride, (int)use_w, (float)H_bar);
-  hipLaunchKernelGGL(k_actor_alpha_loss_finalize, dim3(1), dim3(1), 0,
-                     cur_stream(), out.data_ptr<float>(), (int)B,
-                     (int)use_w);
+                     out.data_ptr<float>(), dla_p, dla_n, (int)B, (int)T,
+                     (int)A, (int)oh_stride, (int)use_w, (float)H_bar);
   return out;
 }
 
@@ -1413,7 +1404,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("td_target_mt", &td_target_mt);
   mod.def("critic_loss_fwd", &critic_loss_fwd);
   mod.def("critic_loss_bwd", &critic_loss_bwd);
-  mod.def("actor_alpha_loss_fwd", &actor_alpha_loss_fwd);
+  mod.def("actor_alpha_loss_fwd", &actor_alpha_loss_fwd,
+          pybind11::arg("aq1"), pybind11::arg("aq2"),
+          pybind11::arg("lp"), pybind11::arg("ls"),
+          pybind11::arg("states"), pybind11::arg("log_alpha"),
+          pybind11::arg("T"), pybind11::arg("use_w"),
+          pybind11::arg("H_bar"),
+          pybind11::arg("dla") = pybind11::none());
   mod.def("actor_alpha_loss_bwd", &actor_alpha_loss_bwd);
   mod.def("critic_loss_bwd2", &critic_loss_bwd2);
   mod.def("actor_alpha_loss_bwd2", &actor_alpha_loss_bwd2);
