@@ -555,18 +555,25 @@ __device__ __forceinline__ void fill_task_weights(
 // in <=4 strides, LDS tree reduce, thread 0 writes the FINALIZED values —
 // one launch replaces zero + multi-block-atomic fwd + finalize (the
 // multi-block version was only ever launch-bound at these batch sizes)
-__global__ __launch_bounds__(1024) void k_critic_loss_fwd_1wg(
+// wavefront sum: 6 DPP/shuffle steps, no LDS, no barrier
+__device__ inline float wave_sum64(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+__global__ __launch_bounds__(256) void k_critic_loss_fwd_1wg(
     const float* __restrict__ q1, const float* __restrict__ q2,
     const float* __restrict__ y, const float* __restrict__ onehot,
     const float* __restrict__ log_alpha, float* __restrict__ out,
     int B, int T, int oh_stride, int use_w) {
-  __shared__ float red[3][1024];
+  __shared__ float part[3][4];
   __shared__ float smw[32];
   const int tid = threadIdx.x;
   if (use_w) fill_task_weights(smw, log_alpha, T);
-  __syncthreads();
+  if (use_w) __syncthreads();
   float s_l1 = 0.f, s_l2 = 0.f, s_w = 0.f;
-  for (int i = tid; i < B; i += 1024) {
+  for (int i = tid; i < B; i += 256) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float w_raw = use_w ? smw[t_i] : 1.f;
     s_w += w_raw;
@@ -574,20 +581,23 @@ __global__ __launch_bounds__(1024) void k_critic_loss_fwd_1wg(
     s_l1 += w_raw * d1 * d1;
     s_l2 += w_raw * d2 * d2;
   }
-  red[0][tid] = s_l1; red[1][tid] = s_l2; red[2][tid] = s_w;
-  __syncthreads();
-  for (int off = 512; off > 0; off >>= 1) {
-    if (tid < off)
-#pragma unroll
-      for (int r = 0; r < 3; ++r) red[r][tid] += red[r][tid + off];
-    __syncthreads();
+  // wave-level shuffles + ONE barrier (a 256-thread LDS reduce tree
+  // measured 16-19 us from barrier latency alone on one CU)
+  s_l1 = wave_sum64(s_l1); s_l2 = wave_sum64(s_l2); s_w = wave_sum64(s_w);
+  const int wid = tid >> 6;
+  if ((tid & 63) == 0) {
+    part[0][wid] = s_l1; part[1][wid] = s_l2; part[2][wid] = s_w;
   }
+  __syncthreads();
   if (tid == 0) {
-    const float wsum = use_w ? red[2][0] : 1.f;
+    const float r0 = part[0][0] + part[0][1] + part[0][2] + part[0][3];
+    const float r1 = part[1][0] + part[1][1] + part[1][2] + part[1][3];
+    const float r2 = part[2][0] + part[2][1] + part[2][2] + part[2][3];
+    const float wsum = use_w ? r2 : 1.f;
     const float denom = wsum * (float)B;
-    out[3] = red[0][0]; out[4] = red[1][0]; out[5] = red[2][0];
-    out[0] = red[0][0] / denom;
-    out[1] = red[1][0] / denom;
+    out[3] = r0; out[4] = r1; out[5] = r2;
+    out[0] = r0 / denom;
+    out[1] = r1 / denom;
     out[2] = wsum;
   }
 }
@@ -643,21 +653,21 @@ __global__ __launch_bounds__(256) void k_critic_loss_bwd(
 // non-null its first dla_n floats are zeroed here — the alpha gradient
 // buffer the subsequent k_actor_alpha_loss_bwd* atomics target — so the
 // engine needs no separate fill launch either.
-__global__ __launch_bounds__(1024) void k_actor_alpha_loss_fwd_1wg(
+__global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd_1wg(
     const float* __restrict__ aq1, const float* __restrict__ aq2,
     const float* __restrict__ lp, const float* __restrict__ ls,
     const float* __restrict__ onehot, const float* __restrict__ log_alpha,
     float* __restrict__ out, float* __restrict__ dla, int dla_n,
     int B, int T, int A, int oh_stride, int use_w, float H_bar) {
-  __shared__ float red[4][1024];
+  __shared__ float part[4][4];
   __shared__ float smw[32];
   const int tid = threadIdx.x;
   if (dla != nullptr && tid < dla_n) dla[tid] = 0.f;
   if (use_w) fill_task_weights(smw, log_alpha, T);
-  __syncthreads();
+  if (use_w) __syncthreads();
   constexpr float CE = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
   float s_pl = 0.f, s_w = 0.f, s_al = 0.f, s_en = 0.f;
-  for (int i = tid; i < B; i += 1024) {
+  for (int i = tid; i < B; i += 256) {
     const int t_i = task_of_row(onehot, i, oh_stride, T);
     const float la = log_alpha[t_i];
     const float alpha_i = __expf(la);
@@ -670,23 +680,25 @@ __global__ __launch_bounds__(1024) void k_actor_alpha_loss_fwd_1wg(
     for (int a = 0; a < A; ++a) ent += ls[(long)i * A + a];
     s_en += ent;
   }
-  red[0][tid] = s_pl; red[1][tid] = s_w; red[2][tid] = s_al;
-  red[3][tid] = s_en;
-  __syncthreads();
-  for (int off = 512; off > 0; off >>= 1) {
-    if (tid < off)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) red[r][tid] += red[r][tid + off];
-    __syncthreads();
+  s_pl = wave_sum64(s_pl); s_w = wave_sum64(s_w);
+  s_al = wave_sum64(s_al); s_en = wave_sum64(s_en);
+  const int wid = tid >> 6;
+  if ((tid & 63) == 0) {
+    part[0][wid] = s_pl; part[1][wid] = s_w;
+    part[2][wid] = s_al; part[3][wid] = s_en;
   }
+  __syncthreads();
   if (tid == 0) {
-    const float wsum = use_w ? red[1][0] : 1.f;
-    out[4] = red[0][0]; out[5] = red[1][0];
-    out[6] = red[2][0]; out[7] = red[3][0];
-    out[0] = red[0][0] / (wsum * (float)B);
+    const float r0 = part[0][0] + part[0][1] + part[0][2] + part[0][3];
+    const float r1 = part[1][0] + part[1][1] + part[1][2] + part[1][3];
+    const float r2 = part[2][0] + part[2][1] + part[2][2] + part[2][3];
+    const float r3 = part[3][0] + part[3][1] + part[3][2] + part[3][3];
+    const float wsum = use_w ? r1 : 1.f;
+    out[4] = r0; out[5] = r1; out[6] = r2; out[7] = r3;
+    out[0] = r0 / (wsum * (float)B);
     out[1] = wsum;
-    out[2] = -red[2][0] / (float)B;
-    out[3] = red[3][0] / (float)B;
+    out[2] = -r2 / (float)B;
+    out[3] = r3 / (float)B;
   }
 }
 
@@ -1148,7 +1160,7 @@ static std::vector<torch::Tensor> critic_loss_fwd(
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
   auto out = torch::empty({6}, q1.options());
-  hipLaunchKernelGGL(k_critic_loss_fwd_1wg, dim3(1), dim3(1024), 0,
+  hipLaunchKernelGGL(k_critic_loss_fwd_1wg, dim3(1), dim3(256), 0,
                      cur_stream(), q1.data_ptr<float>(), q2.data_ptr<float>(),
                      y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
                      out.data_ptr<float>(), (int)B, (int)T, (int)oh_stride,
@@ -1193,7 +1205,7 @@ static torch::Tensor actor_alpha_loss_fwd(
     dla_p = dla->data_ptr<float>();
     dla_n = (int)dla->numel();
   }
-  hipLaunchKernelGGL(k_actor_alpha_loss_fwd_1wg, dim3(1), dim3(1024), 0,
+  hipLaunchKernelGGL(k_actor_alpha_loss_fwd_1wg, dim3(1), dim3(256), 0,
                      cur_stream(), aq1.data_ptr<float>(),
                      aq2.data_ptr<float>(), lp.data_ptr<float>(),
                      ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
