@@ -811,7 +811,7 @@ class CAREEngine(SACEngine):
                                                    self._twin_local[1])
             head_in_dim = x.shape[1]
         closs = ext.critic_loss_fwd(q1, q2, y, states, la_det, T,
-                                    int(use_w))[0]
+                                    int(use_w), self._loss_ws)[0]
         dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
                                   int(use_w))
         wsg, bsg = self._twin_local
@@ -1003,7 +1003,8 @@ class CAREEngine(SACEngine):
         # target) — no separate fill launch
         al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls_cat[B:], states,
                                       la_det, T, int(use_w), self.H_bar_f,
-                                      self.alpha_group.flat_grad)
+                                      self.alpha_group.flat_grad,
+                                      self._aloss_ws)
         daq, dlp = ext.actor_alpha_loss_bwd2(
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)

@@ -216,6 +216,10 @@ class SACEngine:
         if not self._bf16:
             return
         dev = self.device
+        # persistent zero-initialized workspace for the one-launch loss
+        # reductions (per-block partial slots + self-resetting ticket)
+        self._loss_ws = torch.zeros(256, device=dev)
+        self._aloss_ws = torch.zeros(256, device=dev)
         self._critic_bf16 = torch.empty(self.critic_group.numel,
                                         dtype=torch.bfloat16, device=dev)
         self._target_bf16 = torch.empty(self.target_group.numel,
@@ -671,7 +675,7 @@ class SACEngine:
             q1, q2, acts_c = self._twin_fwd_manual(
                 x, self._twin_local_bf16, self._twin_local[1])
         closs = ext.critic_loss_fwd(q1, q2, y, states, la_det, T,
-                                    int(use_w))[0]
+                                    int(use_w), self._loss_ws)[0]
         dy = ext.critic_loss_bwd2(q1, q2, y, states, la_det, closs, T,
                                   int(use_w))
         wsg = [w.grad for w in self._twin_local[0]]
@@ -744,7 +748,8 @@ class SACEngine:
         # atomics target) — no separate fill launch
         al = ext.actor_alpha_loss_fwd(aq1, aq2, lp, ls, states,
                                       la_det, T, int(use_w), self.H_bar_f,
-                                      self.alpha_group.flat_grad)
+                                      self.alpha_group.flat_grad,
+                                      self._aloss_ws)
         daq, dlp = ext.actor_alpha_loss_bwd2(
             aq1, aq2, lp, states, la_det, al, self.alpha_group.flat_grad,
             T, int(use_w), self.H_bar_f)

@@ -653,6 +653,128 @@ __global__ __launch_bounds__(256) void k_critic_loss_bwd(
 // non-null its first dla_n floats are zeroed here — the alpha gradient
 // buffer the subsequent k_actor_alpha_loss_bwd* atomics target — so the
 // engine needs no separate fill launch either.
+// multi-block variant: each block reduces its stripe with wave shuffles
+// (one barrier), writes a private partial slot in ws, and the LAST block
+// (ticket in ws[0], self-resetting so graph replays stay correct) sums
+// the slots and writes the finalized outputs.  One launch, full CU
+// parallelism — the 1wg variant above is the fallback when no workspace
+// is provided (eager/legacy callers) and is latency-bound at ~4 waves.
+__global__ __launch_bounds__(256) void k_critic_loss_fwd_mb(
+    const float* __restrict__ q1, const float* __restrict__ q2,
+    const float* __restrict__ y, const float* __restrict__ onehot,
+    const float* __restrict__ log_alpha, float* __restrict__ out,
+    float* __restrict__ ws, int B, int T, int oh_stride, int use_w) {
+  __shared__ float part[3][4];
+  __shared__ float smw[32];
+  const int tid = threadIdx.x;
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  if (use_w) __syncthreads();
+  float s_l1 = 0.f, s_l2 = 0.f, s_w = 0.f;
+  for (int i = blockIdx.x * 256 + tid; i < B; i += 256 * gridDim.x) {
+    const int t_i = task_of_row(onehot, i, oh_stride, T);
+    const float w_raw = use_w ? smw[t_i] : 1.f;
+    s_w += w_raw;
+    const float d1 = y[i] - q1[i], d2 = y[i] - q2[i];
+    s_l1 += w_raw * d1 * d1;
+    s_l2 += w_raw * d2 * d2;
+  }
+  s_l1 = wave_sum64(s_l1); s_l2 = wave_sum64(s_l2); s_w = wave_sum64(s_w);
+  const int wid = tid >> 6;
+  if ((tid & 63) == 0) {
+    part[0][wid] = s_l1; part[1][wid] = s_l2; part[2][wid] = s_w;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    float* slots = ws + 1;
+    slots[blockIdx.x * 3 + 0] =
+        part[0][0] + part[0][1] + part[0][2] + part[0][3];
+    slots[blockIdx.x * 3 + 1] =
+        part[1][0] + part[1][1] + part[1][2] + part[1][3];
+    slots[blockIdx.x * 3 + 2] =
+        part[2][0] + part[2][1] + part[2][2] + part[2][3];
+    __threadfence();
+    const unsigned old = atomicAdd((unsigned*)ws, 1u);
+    if (old == (unsigned)gridDim.x - 1u) {
+      *(unsigned*)ws = 0u;  // reset the ticket for the next replay
+      __threadfence();
+      float r0 = 0.f, r1 = 0.f, r2 = 0.f;
+      for (int b = 0; b < (int)gridDim.x; ++b) {
+        r0 += slots[b * 3 + 0];
+        r1 += slots[b * 3 + 1];
+        r2 += slots[b * 3 + 2];
+      }
+      const float wsum = use_w ? r2 : 1.f;
+      const float denom = wsum * (float)B;
+      out[3] = r0; out[4] = r1; out[5] = r2;
+      out[0] = r0 / denom;
+      out[1] = r1 / denom;
+      out[2] = wsum;
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd_mb(
+    const float* __restrict__ aq1, const float* __restrict__ aq2,
+    const float* __restrict__ lp, const float* __restrict__ ls,
+    const float* __restrict__ onehot, const float* __restrict__ log_alpha,
+    float* __restrict__ out, float* __restrict__ ws,
+    float* __restrict__ dla, int dla_n,
+    int B, int T, int A, int oh_stride, int use_w, float H_bar) {
+  __shared__ float part[4][4];
+  __shared__ float smw[32];
+  const int tid = threadIdx.x;
+  if (blockIdx.x == 0 && dla != nullptr && tid < dla_n) dla[tid] = 0.f;
+  if (use_w) fill_task_weights(smw, log_alpha, T);
+  if (use_w) __syncthreads();
+  constexpr float CE = 1.4189385332046727f;  // 0.5*(1+log(2*pi))
+  float s_pl = 0.f, s_w = 0.f, s_al = 0.f, s_en = 0.f;
+  for (int i = blockIdx.x * 256 + tid; i < B; i += 256 * gridDim.x) {
+    const int t_i = task_of_row(onehot, i, oh_stride, T);
+    const float la = log_alpha[t_i];
+    const float alpha_i = __expf(la);
+    const float w_raw = use_w ? smw[t_i] : 1.f;
+    s_w += w_raw;
+    const float qmin = fminf(aq1[i], aq2[i]);
+    s_pl += w_raw * -(qmin - alpha_i * lp[i]);
+    s_al += la * (lp[i] + H_bar);
+    float ent = CE * A;
+    for (int a = 0; a < A; ++a) ent += ls[(long)i * A + a];
+    s_en += ent;
+  }
+  s_pl = wave_sum64(s_pl); s_w = wave_sum64(s_w);
+  s_al = wave_sum64(s_al); s_en = wave_sum64(s_en);
+  const int wid = tid >> 6;
+  if ((tid & 63) == 0) {
+    part[0][wid] = s_pl; part[1][wid] = s_w;
+    part[2][wid] = s_al; part[3][wid] = s_en;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    float* slots = ws + 1;
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      slots[blockIdx.x * 4 + r] =
+          part[r][0] + part[r][1] + part[r][2] + part[r][3];
+    __threadfence();
+    const unsigned old = atomicAdd((unsigned*)ws, 1u);
+    if (old == (unsigned)gridDim.x - 1u) {
+      *(unsigned*)ws = 0u;
+      __threadfence();
+      float r0 = 0.f, r1 = 0.f, r2 = 0.f, r3 = 0.f;
+      for (int b = 0; b < (int)gridDim.x; ++b) {
+        r0 += slots[b * 4 + 0]; r1 += slots[b * 4 + 1];
+        r2 += slots[b * 4 + 2]; r3 += slots[b * 4 + 3];
+      }
+      const float wsum = use_w ? r1 : 1.f;
+      out[4] = r0; out[5] = r1; out[6] = r2; out[7] = r3;
+      out[0] = r0 / (wsum * (float)B);
+      out[1] = wsum;
+      out[2] = -r2 / (float)B;
+      out[3] = r3 / (float)B;
+    }
+  }
+}
+
 __global__ __launch_bounds__(256) void k_actor_alpha_loss_fwd_1wg(
     const float* __restrict__ aq1, const float* __restrict__ aq2,
     const float* __restrict__ lp, const float* __restrict__ ls,
@@ -1154,17 +1276,30 @@ static torch::Tensor td_target_mt(torch::Tensor r, torch::Tensor d,
 
 static std::vector<torch::Tensor> critic_loss_fwd(
     torch::Tensor q1, torch::Tensor q2, torch::Tensor y,
-    torch::Tensor states, torch::Tensor log_alpha, long T, long use_w) {
+    torch::Tensor states, torch::Tensor log_alpha, long T, long use_w,
+    c10::optional<torch::Tensor> ws = c10::nullopt) {
   CHECK_IN(q1); CHECK_IN(states); CHECK_IN(log_alpha);
   const long B = q1.size(0);
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
   auto out = torch::empty({6}, q1.options());
-  hipLaunchKernelGGL(k_critic_loss_fwd_1wg, dim3(1), dim3(256), 0,
-                     cur_stream(), q1.data_ptr<float>(), q2.data_ptr<float>(),
-                     y.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
-                     out.data_ptr<float>(), (int)B, (int)T, (int)oh_stride,
-                     (int)use_w);
+  if (ws.has_value() && ws->numel() >= 1 + 32 * 3) {
+    // persistent zero-initialized workspace -> multi-block one-launch
+    // path with last-block finalize (ticket self-resets per launch)
+    const int nblk = (int)std::min<long>((B + 255) / 256, 32);
+    hipLaunchKernelGGL(k_critic_loss_fwd_mb, dim3(nblk), dim3(256), 0,
+                       cur_stream(), q1.data_ptr<float>(),
+                       q2.data_ptr<float>(), y.data_ptr<float>(), oh,
+                       log_alpha.data_ptr<float>(), out.data_ptr<float>(),
+                       ws->data_ptr<float>(), (int)B, (int)T,
+                       (int)oh_stride, (int)use_w);
+  } else {
+    hipLaunchKernelGGL(k_critic_loss_fwd_1wg, dim3(1), dim3(256), 0,
+                       cur_stream(), q1.data_ptr<float>(),
+                       q2.data_ptr<float>(), y.data_ptr<float>(), oh,
+                       log_alpha.data_ptr<float>(), out.data_ptr<float>(),
+                       (int)B, (int)T, (int)oh_stride, (int)use_w);
+  }
   return {out};
 }
 
@@ -1191,7 +1326,8 @@ static torch::Tensor actor_alpha_loss_fwd(
     torch::Tensor aq1, torch::Tensor aq2, torch::Tensor lp, torch::Tensor ls,
     torch::Tensor states, torch::Tensor log_alpha, long T, long use_w,
     double H_bar,
-    c10::optional<torch::Tensor> dla = c10::nullopt) {
+    c10::optional<torch::Tensor> dla = c10::nullopt,
+    c10::optional<torch::Tensor> ws = c10::nullopt) {
   CHECK_IN(aq1); CHECK_IN(states); CHECK_IN(log_alpha);
   const long B = aq1.size(0);
   const long A = ls.size(1);
@@ -1205,12 +1341,23 @@ static torch::Tensor actor_alpha_loss_fwd(
     dla_p = dla->data_ptr<float>();
     dla_n = (int)dla->numel();
   }
-  hipLaunchKernelGGL(k_actor_alpha_loss_fwd_1wg, dim3(1), dim3(256), 0,
-                     cur_stream(), aq1.data_ptr<float>(),
-                     aq2.data_ptr<float>(), lp.data_ptr<float>(),
-                     ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
-                     out.data_ptr<float>(), dla_p, dla_n, (int)B, (int)T,
-                     (int)A, (int)oh_stride, (int)use_w, (float)H_bar);
+  if (ws.has_value() && ws->numel() >= 1 + 32 * 4) {
+    const int nblk = (int)std::min<long>((B + 255) / 256, 32);
+    hipLaunchKernelGGL(k_actor_alpha_loss_fwd_mb, dim3(nblk), dim3(256), 0,
+                       cur_stream(), aq1.data_ptr<float>(),
+                       aq2.data_ptr<float>(), lp.data_ptr<float>(),
+                       ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
+                       out.data_ptr<float>(), ws->data_ptr<float>(), dla_p,
+                       dla_n, (int)B, (int)T, (int)A, (int)oh_stride,
+                       (int)use_w, (float)H_bar);
+  } else {
+    hipLaunchKernelGGL(k_actor_alpha_loss_fwd_1wg, dim3(1), dim3(256), 0,
+                       cur_stream(), aq1.data_ptr<float>(),
+                       aq2.data_ptr<float>(), lp.data_ptr<float>(),
+                       ls.data_ptr<float>(), oh, log_alpha.data_ptr<float>(),
+                       out.data_ptr<float>(), dla_p, dla_n, (int)B, (int)T,
+                       (int)A, (int)oh_stride, (int)use_w, (float)H_bar);
+  }
   return out;
 }
 
@@ -1414,7 +1561,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("squashed_gaussian_bwd", &squashed_gaussian_bwd);
   mod.def("td_target", &td_target);
   mod.def("td_target_mt", &td_target_mt);
-  mod.def("critic_loss_fwd", &critic_loss_fwd);
+  mod.def("critic_loss_fwd", &critic_loss_fwd,
+          pybind11::arg("q1"), pybind11::arg("q2"),
+          pybind11::arg("y"), pybind11::arg("states"),
+          pybind11::arg("log_alpha"), pybind11::arg("T"),
+          pybind11::arg("use_w"),
+          pybind11::arg("ws") = pybind11::none());
   mod.def("critic_loss_bwd", &critic_loss_bwd);
   mod.def("actor_alpha_loss_fwd", &actor_alpha_loss_fwd,
           pybind11::arg("aq1"), pybind11::arg("aq2"),
@@ -1422,7 +1574,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           pybind11::arg("states"), pybind11::arg("log_alpha"),
           pybind11::arg("T"), pybind11::arg("use_w"),
           pybind11::arg("H_bar"),
-          pybind11::arg("dla") = pybind11::none());
+          pybind11::arg("dla") = pybind11::none(),
+          pybind11::arg("ws") = pybind11::none());
   mod.def("actor_alpha_loss_bwd", &actor_alpha_loss_bwd);
   mod.def("critic_loss_bwd2", &critic_loss_bwd2);
   mod.def("actor_alpha_loss_bwd2", &actor_alpha_loss_bwd2);
