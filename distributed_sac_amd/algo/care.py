@@ -770,12 +770,18 @@ class CAREEngine(SACEngine):
             out, acts_a = self._mlp_fwd_manual(enc_cat, self._actor_ws_bf16,
                                                bs_f32)
         mu, lsr = out[:, :A], out[:, A:]
+        krng = self._use_krng and not self._eps_queue
         if self._eps_queue:
             eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
+        elif krng:
+            # eps GENERATED inside the squash kernel (counter RNG) and
+            # written here for the backward — no randn launch
+            eps = torch.empty(mu.shape, device=mu.device, dtype=mu.dtype)
         else:
             eps = torch.randn_like(mu)
         a_cat, lp_cat, tanh_u, ls_cat = ext.squashed_gaussian_fwd(
-            mu, lsr, eps, float(self.actor.k))
+            mu, lsr, eps, float(self.actor.k),
+            self._rng_ctr if krng else None)
         na, nlp = a_cat[:B], lp_cat[:B]
         sa, lp = a_cat[B:], lp_cat[B:]
 
@@ -1070,7 +1076,8 @@ class CAREEngine(SACEngine):
     def _manual_seg3(self):
         from ..ops.flat import FusedAdam as _FA
         st = self._dp_st
-        _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer])
+        _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer],
+                      rng_bump=(self._rng_ctr if self._use_krng else None))
         # self.alpha refreshed lazily outside the graph (see SACEngine)
         self._polyak_targets(mirror=self._target_bf16)
         self._refresh_mixT("target")

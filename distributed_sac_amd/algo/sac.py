@@ -220,6 +220,11 @@ class SACEngine:
         # reductions (per-block partial slots + self-resetting ticket)
         self._loss_ws = torch.zeros(256, device=dev)
         self._aloss_ws = torch.zeros(256, device=dev)
+        # persistent counter for the in-kernel RNG, seeded from the torch
+        # seed so runs stay reproducible end-to-end
+        self._rng_ctr = torch.tensor(
+            [int(torch.initial_seed()) & 0x7FFFFFFF],
+            dtype=torch.int64, device=dev)
         self._critic_bf16 = torch.empty(self.critic_group.numel,
                                         dtype=torch.bfloat16, device=dev)
         self._target_bf16 = torch.empty(self.target_group.numel,
@@ -582,6 +587,17 @@ class SACEngine:
         import os as _os
         return _os.environ.get("DSAC_CHAIN", "1") == "1"
 
+    @property
+    def _use_krng(self) -> bool:
+        """Counter-based device RNG inside the squash / replay-sample
+        kernels (round 2): replaces the torch rand+randn launches and the
+        hipGraph RNG-offset bookkeeping kernels.  DSAC_KRNG=0 restores
+        torch's philox draws."""
+        import os as _os
+        return (getattr(self, "_bf16", False)
+                and getattr(self, "_rng_ctr", None) is not None
+                and _os.environ.get("DSAC_KRNG", "1") == "1")
+
     @torch.no_grad()
     def _chain_fwd(self, x1, x2, ws_bf16, bs_f32, G, act_last=0,
                    out_f32=True, rowcat=False, save_acts=True, wps=None):
@@ -640,12 +656,18 @@ class SACEngine:
             out, acts_a = self._mlp_fwd_manual(x_cat, self._actor_ws_bf16,
                                                bs_f32)
         mu, lsr = out[:, :A], out[:, A:]
+        krng = self._use_krng and not self._eps_queue
         if self._eps_queue:
             eps = torch.cat([self._next_eps(mu[:B]), self._next_eps(mu[:B])])
+        elif krng:
+            # eps GENERATED inside the squash kernel (counter RNG) and
+            # written here for the backward — no randn launch
+            eps = torch.empty(mu.shape, device=mu.device, dtype=mu.dtype)
         else:
             eps = torch.randn_like(mu)
         a_cat, lp_cat, tanh_u, ls_cat = ext.squashed_gaussian_fwd(
-            mu, lsr, eps, float(self.actor.k))
+            mu, lsr, eps, float(self.actor.k),
+            self._rng_ctr if krng else None)
         na, nlp = a_cat[:B], lp_cat[:B]
 
         # ---- TD target ------------------------------------------------
@@ -818,7 +840,9 @@ class SACEngine:
         """Segment 3: fused actor+alpha Adam, Polyak target update."""
         st = self._dp_st
         FusedAdam.step_many([self.actor_optimizer,
-                             self.log_alpha_optimizer])
+                             self.log_alpha_optimizer],
+                            rng_bump=(self._rng_ctr if self._use_krng
+                                      else None))
         # NOTE: self.alpha is refreshed lazily (checkpoint/_per_sample_alpha
         # recompute from log_alpha) — an exp() here would replay as a
         # ~5 µs kernel every captured step just for bookkeeping
@@ -934,6 +958,8 @@ class SACEngine:
     # per-step host cost collapses to one hipGraphLaunch.
     # ------------------------------------------------------------------
     def capture(self, replay, batch_size: int, warmup_iters: int = 3):
+        if self._use_krng and hasattr(replay, "attach_rng"):
+            replay.attach_rng(self._rng_ctr)
         assert self.device.type == "cuda", "capture needs a GPU"
         torch.cuda.synchronize(self.device)
         side = torch.cuda.Stream(self.device)
@@ -968,6 +994,8 @@ class SACEngine:
     # intermediates (activations, squash state) keep stable addresses.
     # ------------------------------------------------------------------
     def capture_dp(self, replay, batch_size: int, warmup_iters: int = 3):
+        if self._use_krng and hasattr(replay, "attach_rng"):
+            replay.attach_rng(self._rng_ctr)
         assert self.device.type == "cuda", "capture needs a GPU"
         assert getattr(self, "_bf16", False), \
             "segmented DP capture runs the bf16 manual-backward path"

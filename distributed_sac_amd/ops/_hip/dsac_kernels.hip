@@ -375,11 +375,30 @@ __global__ __launch_bounds__(256) void k_reduce_partials(
 // profile).  Fields are stacked [T, cap, D]; output row i draws from shard
 // t = i / (B/T) at index floor(rand[i] * size[t]).
 // ---------------------------------------------------------------------------
+// counter-based device RNG (splitmix64 hash of (counter, index, salt)):
+// every RNG-consuming kernel derives its noise from one persistent int64
+// counter that k_adam_prolog3 bumps ONCE per update (single-block kernel,
+// so the bump is race-free by stream ordering).  Replaces the per-update
+// torch rand/randn launches AND the hipGraph RNG-offset bookkeeping
+// kernels torch inserts around them.
+__device__ __forceinline__ unsigned long long dsac_sm64(
+    unsigned long long x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+__device__ __forceinline__ float dsac_u01(unsigned long long h) {
+  // top 24 bits + 1 -> uniform in (0, 1]
+  return (float)((h >> 40) + 1ull) * 5.9604644775390625e-8f;
+}
+
 __global__ __launch_bounds__(256) void k_replay_sample(
     const float* __restrict__ states, const float* __restrict__ actions,
     const float* __restrict__ rewards, const float* __restrict__ next_states,
     const float* __restrict__ dones, const float* __restrict__ sizes,
-    const float* __restrict__ rnd, float* __restrict__ o_states,
+    const float* __restrict__ rnd, const long long* __restrict__ rng,
+    float* __restrict__ o_states,
     float* __restrict__ o_actions, float* __restrict__ o_rewards,
     float* __restrict__ o_next_states, float* __restrict__ o_dones,
     int B, int T, int per, long cap, int Ds, int Da) {
@@ -389,7 +408,16 @@ __global__ __launch_bounds__(256) void k_replay_sample(
   const int lane = threadIdx.x & 63;
   if (i >= B) return;
   const int t = min(i / per, T - 1);
-  const long idx = (long)(rnd[i] * sizes[t]);
+  float u;
+  if (rng != nullptr) {
+    const unsigned long long c = (unsigned long long)rng[0];
+    u = dsac_u01(dsac_sm64(c * 0x100000001ull
+                           + (unsigned long long)i * 2ull + 0x2ull));
+    u = u < 1.f ? u : 0.99999994f;  // keep floor(u*size) < size
+  } else {
+    u = rnd[i];
+  }
+  const long idx = (long)(u * sizes[t]);
   const long src = (long)t * cap + idx;
   for (int j = lane; j < Ds; j += 64) {
     o_states[(long)i * Ds + j] = states[src * Ds + j];
@@ -409,9 +437,10 @@ __global__ __launch_bounds__(256) void k_replay_sample(
 // ---------------------------------------------------------------------------
 __global__ __launch_bounds__(256) void k_squash_fwd(
     const float* __restrict__ mu, const float* __restrict__ lsr,
-    const float* __restrict__ eps, float* __restrict__ act,
+    float* __restrict__ eps, float* __restrict__ act,
     float* __restrict__ logp, float* __restrict__ tanh_u,
-    float* __restrict__ ls_out, int B, int A, float k) {
+    float* __restrict__ ls_out, const long long* __restrict__ rng,
+    int B, int A, float k) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= B) return;
   constexpr float C = 0.9189385332046727f;  // 0.5*log(2*pi)
@@ -420,7 +449,19 @@ __global__ __launch_bounds__(256) void k_squash_fwd(
     const long idx = (long)i * A + a;
     const float ls = fminf(fmaxf(lsr[idx], -20.f), 2.f);
     const float s = __expf(ls);
-    const float e = eps[idx];
+    float e;
+    if (rng != nullptr) {
+      // Box-Muller from two hashed uniforms; eps SAVED for the backward
+      const unsigned long long c = (unsigned long long)rng[0];
+      const unsigned long long h1 = dsac_sm64(
+          c * 0x100000001ull + (unsigned long long)idx * 2ull + 0x1ull);
+      const unsigned long long h2 = dsac_sm64(h1);
+      e = sqrtf(-2.f * __logf(dsac_u01(h1)))
+          * __cosf(6.283185307179586f * dsac_u01(h2));
+      eps[idx] = e;
+    } else {
+      e = eps[idx];
+    }
     const float u = mu[idx] + s * e;
     const float t = tanhf(u);
     act[idx] = k * t;
@@ -986,9 +1027,11 @@ struct AdamGroup {
   float* p; const float* g; float* m; float* v; const float* st; long n;
 };
 
-__global__ void k_adam_prolog3(float* s0, float* s1, float* s2,
+__global__ void k_adam_prolog3(long long* rng, float* s0, float* s1,
+                               float* s2,
                                float lr0, float lr1, float lr2,
                                float b1, float b2) {
+  if (rng != nullptr && threadIdx.x == 0) rng[0] += 1;
   const int i = threadIdx.x;
   float* s = i == 0 ? s0 : (i == 1 ? s1 : s2);
   const float lr = i == 0 ? lr0 : (i == 1 ? lr1 : lr2);
@@ -1167,8 +1210,13 @@ static std::vector<torch::Tensor> linear_bwd_dwdb(torch::Tensor dy,
 static std::vector<torch::Tensor> replay_sample(
     torch::Tensor states, torch::Tensor actions, torch::Tensor rewards,
     torch::Tensor next_states, torch::Tensor dones, torch::Tensor sizes,
-    torch::Tensor rnd, long B) {
-  CHECK_IN(states); CHECK_IN(sizes); CHECK_IN(rnd);
+    torch::Tensor rnd, long B,
+    c10::optional<torch::Tensor> rng = c10::nullopt) {
+  CHECK_IN(states); CHECK_IN(sizes);
+  const bool krng = rng.has_value() && rng->numel() > 0;
+  if (!krng) CHECK_IN(rnd);
+  TORCH_CHECK(!krng || rng->scalar_type() == torch::kInt64,
+              "rng counter must be int64");
   const long T = states.size(0), cap = states.size(1);
   const long Ds = states.size(2), Da = actions.size(2);
   const int per = (int)(B / T);
@@ -1183,7 +1231,10 @@ static std::vector<torch::Tensor> replay_sample(
                      cur_stream(), states.data_ptr<float>(),
                      actions.data_ptr<float>(), rewards.data_ptr<float>(),
                      next_states.data_ptr<float>(), dones.data_ptr<float>(),
-                     sizes.data_ptr<float>(), rnd.data_ptr<float>(),
+                     sizes.data_ptr<float>(),
+                     krng ? nullptr : rnd.data_ptr<float>(),
+                     krng ? (const long long*)rng->data_ptr<long>()
+                          : nullptr,
                      o_s.data_ptr<float>(), o_a.data_ptr<float>(),
                      o_r.data_ptr<float>(), o_ns.data_ptr<float>(),
                      o_d.data_ptr<float>(), (int)B, (int)T, per, cap,
@@ -1191,11 +1242,14 @@ static std::vector<torch::Tensor> replay_sample(
   return {o_s, o_a, o_r, o_ns, o_d};
 }
 
-static std::vector<torch::Tensor> squashed_gaussian_fwd(torch::Tensor mu,
-                                                        torch::Tensor lsr,
-                                                        torch::Tensor eps,
-                                                        double k) {
+static std::vector<torch::Tensor> squashed_gaussian_fwd(
+    torch::Tensor mu, torch::Tensor lsr, torch::Tensor eps, double k,
+    c10::optional<torch::Tensor> rng = c10::nullopt) {
   CHECK_IN(mu); CHECK_IN(lsr); CHECK_IN(eps);
+  const bool krng = rng.has_value() && rng->numel() > 0;
+  TORCH_CHECK(!krng || (rng->scalar_type() == torch::kInt64
+                        && eps.is_contiguous()),
+              "krng mode needs an int64 counter and a contiguous eps out");
   auto muc = mu.contiguous(); auto lc = lsr.contiguous();
   auto ec = eps.contiguous();
   const long B = muc.size(0), A = muc.size(1);
@@ -1209,7 +1263,10 @@ static std::vector<torch::Tensor> squashed_gaussian_fwd(torch::Tensor mu,
                      muc.data_ptr<float>(), lc.data_ptr<float>(),
                      ec.data_ptr<float>(), act.data_ptr<float>(),
                      logp.data_ptr<float>(), tanh_u.data_ptr<float>(),
-                     ls_out.data_ptr<float>(), (int)B, (int)A, (float)k);
+                     ls_out.data_ptr<float>(),
+                     krng ? (const long long*)rng->data_ptr<long>()
+                          : nullptr,
+                     (int)B, (int)A, (float)k);
   return {act, logp, tanh_u, ls_out};
 }
 
@@ -1487,7 +1544,8 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
                              std::vector<torch::Tensor> vs,
                              std::vector<torch::Tensor> states,
                              std::vector<double> lrs, double b1, double b2,
-                             double eps, std::vector<torch::Tensor> mirs) {
+                             double eps, std::vector<torch::Tensor> mirs,
+                             c10::optional<torch::Tensor> rng = c10::nullopt) {
   const size_t G = ps.size();
   TORCH_CHECK(G >= 1 && G <= 3, "1..3 groups");
   float* P[3] = {nullptr, nullptr, nullptr};
@@ -1515,8 +1573,10 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
     }
     total += N[g];
   }
+  long long* rng_p = (rng.has_value() && rng->numel() > 0)
+      ? (long long*)rng->data_ptr<long>() : nullptr;
   hipLaunchKernelGGL(k_adam_prolog3, dim3(1), dim3(3), 0, cur_stream(),
-                     St[0], St[1], St[2], (float)LR[0], (float)LR[1],
+                     rng_p, St[0], St[1], St[2], (float)LR[0], (float)LR[1],
                      (float)LR[2], (float)b1, (float)b2);
   hipLaunchKernelGGL(k_adam_multi, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(),
@@ -1556,8 +1616,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("linear_bwd_dx_g", &linear_bwd_dx_g, "grouped, sums over G");
   mod.def("linear_bwd_dwdb", &linear_bwd_dwdb, "split-K backward dW+db");
   mod.def("linear_bwd_dwdb_g", &linear_bwd_dwdb_g, "grouped split-K dW+db");
-  mod.def("replay_sample", &replay_sample, "stratified replay gather");
-  mod.def("squashed_gaussian_fwd", &squashed_gaussian_fwd);
+  mod.def("replay_sample", &replay_sample, "stratified replay gather",
+          pybind11::arg("states"), pybind11::arg("actions"),
+          pybind11::arg("rewards"), pybind11::arg("next_states"),
+          pybind11::arg("dones"), pybind11::arg("sizes"),
+          pybind11::arg("rnd"), pybind11::arg("B"),
+          pybind11::arg("rng") = pybind11::none());
+  mod.def("squashed_gaussian_fwd", &squashed_gaussian_fwd,
+          pybind11::arg("mu"), pybind11::arg("lsr"),
+          pybind11::arg("eps"), pybind11::arg("k"),
+          pybind11::arg("rng") = pybind11::none());
   mod.def("squashed_gaussian_bwd", &squashed_gaussian_bwd);
   mod.def("td_target", &td_target);
   mod.def("td_target_mt", &td_target_mt);
@@ -1590,7 +1658,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           pybind11::arg("ps"), pybind11::arg("gs"), pybind11::arg("ms"),
           pybind11::arg("vs"), pybind11::arg("states"), pybind11::arg("lrs"),
           pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
-          pybind11::arg("mirs") = std::vector<torch::Tensor>());
+          pybind11::arg("mirs") = std::vector<torch::Tensor>(),
+          pybind11::arg("rng") = pybind11::none());
   mod.def("polyak_", &polyak_,
           pybind11::arg("t"), pybind11::arg("s"), pybind11::arg("tau"),
           pybind11::arg("mir") = pybind11::none());

@@ -195,9 +195,13 @@ class FusedAdam:
 
     @staticmethod
     @torch.no_grad()
-    def step_many(opts: List["FusedAdam"]) -> None:
+    def step_many(opts: List["FusedAdam"], rng_bump=None) -> None:
         """Step up to 3 optimizers in ONE fused kernel launch (same betas/
-        eps; device-resident step state required)."""
+        eps; device-resident step state required).  rng_bump: optional
+        int64 counter the prolog kernel increments once — the per-update
+        bump for the counter-based device RNG (race-free: the prolog is a
+        single-block kernel and every RNG consumer runs in other
+        launches)."""
         live = [o for o in opts if o is not None]
         if (1 <= len(live) <= 3
                 and all(o._dev_state is not None for o in live)
@@ -212,7 +216,8 @@ class FusedAdam:
                 [o.lr for o in live],
                 live[0].betas[0], live[0].betas[1], live[0].eps,
                 [o.bf16_mirror if o.bf16_mirror is not None
-                 else torch.Tensor() for o in live])
+                 else torch.Tensor() for o in live],
+                rng_bump)
             return
         for o in live:
             o.step()

@@ -156,6 +156,14 @@ class ShardedReplay:
         if seed is not None:
             self.generator = torch.Generator(device=self.device)
             self.generator.manual_seed(seed)
+        self._rng_ctr = None     # optional int64 counter (attach_rng)
+        self._empty_rnd = None
+
+    def attach_rng(self, ctr: torch.Tensor) -> None:
+        """Use the engine's counter-based device RNG for graph-safe
+        index draws (the counter is bumped once per update by the fused
+        Adam prolog kernel)."""
+        self._rng_ctr = ctr
     def __len__(self) -> int:
         """min over shards (reference replay_buffers.__len__:102-107)."""
         return min(len(s) for s in self.shards)
@@ -207,6 +215,17 @@ class ShardedReplay:
             from ..ops import has_native, native, native_enabled
             if native_enabled() and has_native() \
                     and batch_size % self.num_tasks == 0:
+                if self._rng_ctr is not None:
+                    # counter-based in-kernel uniforms: no torch.rand
+                    # launch, no graph RNG-offset bookkeeping kernels
+                    if self._empty_rnd is None:
+                        self._empty_rnd = torch.empty(
+                            0, device=self.device)
+                    o = native().replay_sample(
+                        self.f_states, self.f_actions, self.f_rewards,
+                        self.f_next_states, self.f_dones, self.sizes_dev,
+                        self._empty_rnd, batch_size, self._rng_ctr)
+                    return dict(zip(FIELDS, o))
                 rnd = torch.rand(batch_size, device=self.device)
                 o = native().replay_sample(
                     self.f_states, self.f_actions, self.f_rewards,
