@@ -1022,7 +1022,7 @@ class CAREEngine(SACEngine):
                 acts_f[0].shape[-1], [1] * (nl_c - 1) + [0], 2, 0,
                 enc_c.shape[1], list(self._twin_local_dxp))
             dx0a = outs[-1]
-            dsa = dx0a[0] + dx0a[1]
+            dsa = dx0a              # twin heads summed inside squash bwd2
         else:
             dy = daq
             for i in range(nl_c - 1, 0, -1):
@@ -1085,7 +1085,7 @@ class CAREEngine(SACEngine):
             self.context_encoder_optimizer.step()   # refreshes ctx mirror
         self.tie_actor_state_encoder()
         closs, al = st["closs"], st["al"]
-        return {"critic_loss": closs[0] + closs[1],
+        return {"critic_loss": closs[6],  # summed in-kernel
                 "actor_loss": al[0],
                 "alpha_loss": al[2],
                 "entropy": al[3]}

@@ -783,7 +783,7 @@ class SACEngine:
                 acts_f[0].shape[-1], [1] * (nl_c - 1) + [0], 2, 0,
                 states.shape[1], list(self._twin_local_dxp))
             dx0 = outs[-1]          # [2, B, A] fp32 (action columns only)
-            dsa = dx0[0] + dx0[1]   # sum over the twin Q heads
+            dsa = dx0               # twin heads summed inside squash bwd2
         else:
             dy = daq
             for i in range(nl_c - 1, 0, -1):
@@ -850,7 +850,7 @@ class SACEngine:
                      mirror=getattr(self, "_target_bf16", None))
         closs, al = st["closs"], st["al"]
         return {
-            "critic_loss": closs[0] + closs[1],
+            "critic_loss": closs[6],  # summed in-kernel
             "actor_loss": al[0],
             "alpha_loss": al[2],
             "entropy": al[3],

@@ -440,14 +440,16 @@ __global__ __launch_bounds__(256) void k_squash_fwd(
     float* __restrict__ eps, float* __restrict__ act,
     float* __restrict__ logp, float* __restrict__ tanh_u,
     float* __restrict__ ls_out, const long long* __restrict__ rng,
-    int B, int A, float k) {
+    long mu_ld, long ls_ld, int B, int A, float k) {
+  // mu/lsr may be strided row views (the (mu|lsr) head output sliced in
+  // half) — reading through the stride kills the .contiguous() copies
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= B) return;
   constexpr float C = 0.9189385332046727f;  // 0.5*log(2*pi)
   float lp = 0.f;
   for (int a = 0; a < A; ++a) {
     const long idx = (long)i * A + a;
-    const float ls = fminf(fmaxf(lsr[idx], -20.f), 2.f);
+    const float ls = fminf(fmaxf(lsr[(long)i * ls_ld + a], -20.f), 2.f);
     const float s = __expf(ls);
     float e;
     if (rng != nullptr) {
@@ -462,7 +464,7 @@ __global__ __launch_bounds__(256) void k_squash_fwd(
     } else {
       e = eps[idx];
     }
-    const float u = mu[idx] + s * e;
+    const float u = mu[(long)i * mu_ld + a] + s * e;
     const float t = tanhf(u);
     act[idx] = k * t;
     tanh_u[idx] = t;
@@ -640,6 +642,7 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd_1wg(
     out[0] = r0 / denom;
     out[1] = r1 / denom;
     out[2] = wsum;
+    out[6] = (r0 + r1) / denom;
   }
 }
 
@@ -750,6 +753,7 @@ __global__ __launch_bounds__(256) void k_critic_loss_fwd_mb(
       out[0] = r0 / denom;
       out[1] = r1 / denom;
       out[2] = wsum;
+      out[6] = (r0 + r1) / denom;  // logged sum (no separate add launch)
     }
   }
 }
@@ -942,7 +946,8 @@ __global__ __launch_bounds__(256) void k_squash_bwd2(
     const float* __restrict__ ga, const float* __restrict__ gl,
     const float* __restrict__ lsr, const float* __restrict__ ls,
     const float* __restrict__ eps, const float* __restrict__ tanh_u,
-    unsigned short* __restrict__ dhead, int B, int A, float k) {
+    unsigned short* __restrict__ dhead, long lsr_ld, int ga_twin, int B,
+    int A, float k) {
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= B) return;
   const float g = gl[i];
@@ -959,8 +964,12 @@ __global__ __launch_bounds__(256) void k_squash_bwd2(
     const float dlp_du = 2.f * t * omt2 / (omt2 + 1e-6f);
     const float s = __expf(ls[idx]);
     const float e = eps[idx];
-    const float du = ga[idx] * k * omt2 + g * dlp_du;
-    const float raw = lsr[idx];
+    // ga_twin: ga is the [2,B,A] action-column grad straight from the
+    // twin-critic chain backward; summing the two heads here kills the
+    // dx0[0]+dx0[1] add launch
+    const float gav = ga_twin ? ga[idx] + ga[(long)B * A + idx] : ga[idx];
+    const float du = gav * k * omt2 + g * dlp_du;
+    const float raw = lsr[(long)i * lsr_ld + a];
     const float mask = (raw >= -20.f && raw <= 2.f) ? 1.f : 0.f;
     dhead[(long)i * 2 * A + a] = cvt(du);
     dhead[(long)i * 2 * A + A + a] = cvt(mask * (du * e * s - g));
@@ -1250,14 +1259,20 @@ static std::vector<torch::Tensor> squashed_gaussian_fwd(
   TORCH_CHECK(!krng || (rng->scalar_type() == torch::kInt64
                         && eps.is_contiguous()),
               "krng mode needs an int64 counter and a contiguous eps out");
-  auto muc = mu.contiguous(); auto lc = lsr.contiguous();
+  const bool mu_ok = mu.dim() == 2 && mu.stride(1) == 1;
+  const bool ls_ok = lsr.dim() == 2 && lsr.stride(1) == 1;
+  auto muc = mu_ok ? mu : mu.contiguous();
+  auto lc = ls_ok ? lsr : lsr.contiguous();
   auto ec = eps.contiguous();
   const long B = muc.size(0), A = muc.size(1);
+  const long mu_ld = mu_ok ? mu.stride(0) : A;
+  const long ls_ld = ls_ok ? lsr.stride(0) : A;
   TORCH_CHECK(A <= 32, "action_dim too large for fused kernel");
-  auto act = torch::empty_like(muc);
-  auto logp = torch::empty({B, 1}, muc.options());
-  auto tanh_u = torch::empty_like(muc);
-  auto ls_out = torch::empty_like(muc);
+  auto opt = muc.options();
+  auto act = torch::empty({B, A}, opt);
+  auto logp = torch::empty({B, 1}, opt);
+  auto tanh_u = torch::empty({B, A}, opt);
+  auto ls_out = torch::empty({B, A}, opt);
   const int grid = (B + 255) / 256;
   hipLaunchKernelGGL(k_squash_fwd, dim3(grid), dim3(256), 0, cur_stream(),
                      muc.data_ptr<float>(), lc.data_ptr<float>(),
@@ -1266,7 +1281,7 @@ static std::vector<torch::Tensor> squashed_gaussian_fwd(
                      ls_out.data_ptr<float>(),
                      krng ? (const long long*)rng->data_ptr<long>()
                           : nullptr,
-                     (int)B, (int)A, (float)k);
+                     mu_ld, ls_ld, (int)B, (int)A, (float)k);
   return {act, logp, tanh_u, ls_out};
 }
 
@@ -1339,7 +1354,7 @@ static std::vector<torch::Tensor> critic_loss_fwd(
   const long B = q1.size(0);
   const long oh_stride = states.size(1);
   const float* oh = states.data_ptr<float>() + (oh_stride - T);
-  auto out = torch::empty({6}, q1.options());
+  auto out = torch::empty({8}, q1.options());
   if (ws.has_value() && ws->numel() >= 1 + 32 * 3) {
     // persistent zero-initialized workspace -> multi-block one-launch
     // path with last-block finalize (ticket self-resets per launch)
@@ -1484,9 +1499,14 @@ static torch::Tensor squashed_gaussian_bwd2(
     torch::Tensor eps, torch::Tensor tanh_u, double k) {
   CHECK_IN(ga);
   auto gac = ga.contiguous(); auto glc = gl.contiguous();
-  auto lsrc = lsr.contiguous(); auto lsc = ls.contiguous();
+  const bool lsr_ok = lsr.dim() == 2 && lsr.stride(1) == 1;
+  auto lsrc = lsr_ok ? lsr : lsr.contiguous();
+  auto lsc = ls.contiguous();
   auto ec = eps.contiguous(); auto tc = tanh_u.contiguous();
-  const long B = gac.size(0), A = gac.size(1);
+  const int ga_twin = (gac.dim() == 3 && gac.size(0) == 2) ? 1 : 0;
+  const long B = ga_twin ? gac.size(1) : gac.size(0);
+  const long A = ga_twin ? gac.size(2) : gac.size(1);
+  const long lsr_ld = lsr_ok ? lsr.stride(0) : A;
   auto dhead = torch::empty({B, 2 * A},
                             gac.options().dtype(torch::kBFloat16));
   hipLaunchKernelGGL(k_squash_bwd2, dim3((B + 255) / 256), dim3(256), 0,
@@ -1494,8 +1514,8 @@ static torch::Tensor squashed_gaussian_bwd2(
                      glc.data_ptr<float>(), lsrc.data_ptr<float>(),
                      lsc.data_ptr<float>(), ec.data_ptr<float>(),
                      tc.data_ptr<float>(),
-                     (unsigned short*)dhead.data_ptr(), (int)B, (int)A,
-                     (float)k);
+                     (unsigned short*)dhead.data_ptr(), lsr_ld, ga_twin,
+                     (int)B, (int)A, (float)k);
   return dhead;
 }
 
