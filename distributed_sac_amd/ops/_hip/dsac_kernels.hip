@@ -993,6 +993,15 @@ __global__ __launch_bounds__(256) void k_adam(
   p[i] -= step_size * mi / (sqrtf(vi) * inv_sqrt_bc2 + eps);
 }
 
+// K9b: graph-capturable Adam — the step counter lives on-device so a
+// captured update's bias correction advances across hipGraph replays.
+__global__ void k_adam_prolog(float* __restrict__ state, float lr, float b1,
+                              float b2) {
+  const float step = state[0] + 1.f;
+  state[0] = step;
+  state[1] = lr / (1.f - __powf(b1, step));          // step_size
+  state[2] = 1.f / sqrtf(1.f - __powf(b2, step));    // inv_sqrt_bc2
+}
 
 __device__ __forceinline__ unsigned short f32_bf16_rne_d(float f) {
   union { float f; unsigned u; } v{f};
@@ -1006,32 +1015,11 @@ __device__ __forceinline__ unsigned short f32_bf16_rne_d(float f) {
 __global__ __launch_bounds__(256) void k_adam_dev(
     float* __restrict__ p, const float* __restrict__ g,
     float* __restrict__ m, float* __restrict__ v,
-    float* __restrict__ state, long n, float lr, float b1, float b2,
-    float eps, unsigned short* __restrict__ mir, float* __restrict__ ws) {
-  // fused bias correction: every block derives (step_size, inv_sqrt_bc2)
-  // from state[0]+1 itself (2 powfs per block); the LAST block (ticket
-  // in ws, self-resetting) writes the advanced state back.  All state
-  // reads precede the ticket increment, so the bump cannot race them.
-  __shared__ float sss[2];
-  if (threadIdx.x == 0) {
-    const float step = state[0] + 1.f;
-    sss[0] = lr / (1.f - __powf(b1, step));
-    sss[1] = 1.f / sqrtf(1.f - __powf(b2, step));
-  }
-  __syncthreads();
+    const float* __restrict__ state, long n, float b1, float b2, float eps,
+    unsigned short* __restrict__ mir) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (threadIdx.x == 0) {
-    const unsigned old = atomicAdd((unsigned*)ws, 1u);
-    if (old == (unsigned)gridDim.x - 1u) {
-      *(unsigned*)ws = 0u;
-      const float step = state[0] + 1.f;
-      state[0] = step;
-      state[1] = sss[0];
-      state[2] = sss[1];
-    }
-  }
   if (i >= n) return;
-  const float step_size = sss[0], inv_sqrt_bc2 = sss[1];
+  const float step_size = state[1], inv_sqrt_bc2 = state[2];
   const float gi = g[i];
   const float mi = b1 * m[i] + (1.f - b1) * gi;
   const float vi = b2 * v[i] + (1.f - b2) * gi * gi;
@@ -1048,60 +1036,40 @@ struct AdamGroup {
   float* p; const float* g; float* m; float* v; const float* st; long n;
 };
 
+__global__ void k_adam_prolog3(long long* rng, float* s0, float* s1,
+                               float* s2,
+                               float lr0, float lr1, float lr2,
+                               float b1, float b2) {
+  if (rng != nullptr && threadIdx.x == 0) rng[0] += 1;
+  const int i = threadIdx.x;
+  float* s = i == 0 ? s0 : (i == 1 ? s1 : s2);
+  const float lr = i == 0 ? lr0 : (i == 1 ? lr1 : lr2);
+  if (s == nullptr) return;
+  const float step = s[0] + 1.f;
+  s[0] = step;
+  s[1] = lr / (1.f - __powf(b1, step));
+  s[2] = 1.f / sqrtf(1.f - __powf(b2, step));
+}
 
 __global__ __launch_bounds__(256) void k_adam_multi(
-    float* p0, const float* g0, float* m0, float* v0, float* st0, long n0,
-    float* p1, const float* g1, float* m1, float* v1, float* st1, long n1,
-    float* p2, const float* g2, float* m2, float* v2, float* st2, long n2,
-    float lr0, float lr1, float lr2,
+    float* p0, const float* g0, float* m0, float* v0, const float* st0, long n0,
+    float* p1, const float* g1, float* m1, float* v1, const float* st1, long n1,
+    float* p2, const float* g2, float* m2, float* v2, const float* st2, long n2,
     float b1, float b2, float eps, unsigned short* mr0, unsigned short* mr1,
-    unsigned short* mr2, float* ws, long long* rng) {
-  // fused bias correction + last-block state writeback + per-update RNG
-  // counter bump (see k_adam_dev for the ticket-safety argument)
-  __shared__ float sss[3][2];
-  if (threadIdx.x < 3) {
-    float* st = threadIdx.x == 0 ? st0 : (threadIdx.x == 1 ? st1 : st2);
-    const float lr = threadIdx.x == 0 ? lr0
-                     : (threadIdx.x == 1 ? lr1 : lr2);
-    if (st != nullptr) {
-      const float step = st[0] + 1.f;
-      sss[threadIdx.x][0] = lr / (1.f - __powf(b1, step));
-      sss[threadIdx.x][1] = 1.f / sqrtf(1.f - __powf(b2, step));
-    }
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    const unsigned old = atomicAdd((unsigned*)ws, 1u);
-    if (old == (unsigned)gridDim.x - 1u) {
-      *(unsigned*)ws = 0u;
-      if (rng != nullptr) rng[0] += 1;
-#pragma unroll
-      for (int gidx = 0; gidx < 3; ++gidx) {
-        float* st = gidx == 0 ? st0 : (gidx == 1 ? st1 : st2);
-        if (st != nullptr) {
-          st[0] = st[0] + 1.f;
-          st[1] = sss[gidx][0];
-          st[2] = sss[gidx][1];
-        }
-      }
-    }
-  }
+    unsigned short* mr2) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  float* p; const float* g; float* m; float* v; unsigned short* mr;
-  int gsel;
-  if (i < n0) { p = p0; g = g0; m = m0; v = v0; mr = mr0; gsel = 0; }
-  else if ((i -= n0) < n1) { p = p1; g = g1; m = m1; v = v1; mr = mr1;
-                             gsel = 1; }
-  else if ((i -= n1) < n2) { p = p2; g = g2; m = m2; v = v2; mr = mr2;
-                             gsel = 2; }
+  float* p; const float* g; float* m; float* v; const float* st;
+  unsigned short* mr;
+  if (i < n0) { p = p0; g = g0; m = m0; v = v0; st = st0; mr = mr0; }
+  else if ((i -= n0) < n1) { p = p1; g = g1; m = m1; v = v1; st = st1; mr = mr1; }
+  else if ((i -= n1) < n2) { p = p2; g = g2; m = m2; v = v2; st = st2; mr = mr2; }
   else return;
   const float gi = g[i];
   const float mi = b1 * m[i] + (1.f - b1) * gi;
   const float vi = b2 * v[i] + (1.f - b2) * gi * gi;
   m[i] = mi;
   v[i] = vi;
-  const float pn = p[i]
-      - sss[gsel][0] * mi / (sqrtf(vi) * sss[gsel][1] + eps);
+  const float pn = p[i] - st[1] * mi / (sqrtf(vi) * st[2] + eps);
   p[i] = pn;
   if (mr != nullptr) mr[i] = f32_bf16_rne_d(pn);
 }
@@ -1570,26 +1538,24 @@ static void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                            torch::Tensor v, torch::Tensor state, double lr,
                            double b1, double b2, double eps,
-                           c10::optional<torch::Tensor> mir_opt,
-                           c10::optional<torch::Tensor> ws_opt) {
+                           c10::optional<torch::Tensor> mir_opt) {
   CHECK_IN(p); CHECK_IN(state);
   TORCH_CHECK(state.numel() >= 3, "state = {step, step_size, inv_sqrt_bc2}");
   const long n = p.numel();
-  TORCH_CHECK(ws_opt.has_value() && ws_opt->numel() >= 1,
-              "adam_step_dev_ needs a persistent zeroed ticket workspace");
-  auto ws = ws_opt;
   unsigned short* mp = nullptr;
   if (mir_opt.has_value() && mir_opt->defined() && mir_opt->numel() > 0) {
     TORCH_CHECK(mir_opt->numel() == n
                 && mir_opt->scalar_type() == torch::kBFloat16);
     mp = (unsigned short*)mir_opt->data_ptr();
   }
+  hipLaunchKernelGGL(k_adam_prolog, dim3(1), dim3(1), 0, cur_stream(),
+                     state.data_ptr<float>(), (float)lr, (float)b1,
+                     (float)b2);
   hipLaunchKernelGGL(k_adam_dev, dim3((n + 255) / 256), dim3(256), 0,
                      cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
-                     state.data_ptr<float>(), n, (float)lr, (float)b1,
-                     (float)b2, (float)eps, mp,
-                     ws->data_ptr<float>());
+                     state.data_ptr<float>(), n, (float)b1, (float)b2,
+                     (float)eps, mp);
 }
 
 static void adam_step_multi_(std::vector<torch::Tensor> ps,
@@ -1599,8 +1565,7 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
                              std::vector<torch::Tensor> states,
                              std::vector<double> lrs, double b1, double b2,
                              double eps, std::vector<torch::Tensor> mirs,
-                             c10::optional<torch::Tensor> rng = c10::nullopt,
-                             c10::optional<torch::Tensor> ws = c10::nullopt) {
+                             c10::optional<torch::Tensor> rng = c10::nullopt) {
   const size_t G = ps.size();
   TORCH_CHECK(G >= 1 && G <= 3, "1..3 groups");
   float* P[3] = {nullptr, nullptr, nullptr};
@@ -1630,16 +1595,15 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
   }
   long long* rng_p = (rng.has_value() && rng->numel() > 0)
       ? (long long*)rng->data_ptr<long>() : nullptr;
-  TORCH_CHECK(ws.has_value() && ws->numel() >= 1,
-              "adam_step_multi_ needs a persistent zeroed ticket workspace");
+  hipLaunchKernelGGL(k_adam_prolog3, dim3(1), dim3(3), 0, cur_stream(),
+                     rng_p, St[0], St[1], St[2], (float)LR[0], (float)LR[1],
+                     (float)LR[2], (float)b1, (float)b2);
   hipLaunchKernelGGL(k_adam_multi, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(),
                      P[0], Gr[0], M[0], V[0], St[0], N[0],
                      P[1], Gr[1], M[1], V[1], St[1], N[1],
                      P[2], Gr[2], M[2], V[2], St[2], N[2],
-                     (float)LR[0], (float)LR[1], (float)LR[2],
-                     (float)b1, (float)b2, (float)eps, MR[0], MR[1], MR[2],
-                     ws->data_ptr<float>(), rng_p);
+                     (float)b1, (float)b2, (float)eps, MR[0], MR[1], MR[2]);
 }
 
 static void polyak_(torch::Tensor t, torch::Tensor s, double tau,
@@ -1709,15 +1673,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           pybind11::arg("p"), pybind11::arg("g"), pybind11::arg("m"),
           pybind11::arg("v"), pybind11::arg("state"), pybind11::arg("lr"),
           pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
-          pybind11::arg("mir") = pybind11::none(),
-          pybind11::arg("ws") = pybind11::none());
+          pybind11::arg("mir") = pybind11::none());
   mod.def("adam_step_multi_", &adam_step_multi_,
           pybind11::arg("ps"), pybind11::arg("gs"), pybind11::arg("ms"),
           pybind11::arg("vs"), pybind11::arg("states"), pybind11::arg("lrs"),
           pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
           pybind11::arg("mirs") = std::vector<torch::Tensor>(),
-          pybind11::arg("rng") = pybind11::none(),
-          pybind11::arg("ws") = pybind11::none());
+          pybind11::arg("rng") = pybind11::none());
   mod.def("polyak_", &polyak_,
           pybind11::arg("t"), pybind11::arg("s"), pybind11::arg("tau"),
           pybind11::arg("mir") = pybind11::none());

@@ -152,11 +152,8 @@ class FusedAdam:
         # optional bf16 compute mirror refreshed in the SAME adam kernel
         # (set by the engine when mixed precision is on)
         self.bf16_mirror: Optional[torch.Tensor] = None
-        self._ticket_ws = None
         if group.flat_data.is_cuda and native_enabled() and has_native():
             self._dev_state = torch.zeros(3, device=group.flat_data.device)
-            # persistent zeroed ticket for the fused-prolog adam kernel
-            self._ticket_ws = torch.zeros(1, device=group.flat_data.device)
 
     @property
     def step_count(self) -> int:
@@ -180,9 +177,13 @@ class FusedAdam:
         m, v = self.exp_avg, self.exp_avg_sq
         b1, b2 = self.betas
         if self._dev_state is not None:
-            native().adam_step_dev_(p, g, m, v, self._dev_state,
-                                    self.lr, b1, b2, self.eps,
-                                    self.bf16_mirror, self._ticket_ws)
+            if self.bf16_mirror is not None:
+                native().adam_step_dev_(p, g, m, v, self._dev_state,
+                                        self.lr, b1, b2, self.eps,
+                                        self.bf16_mirror)
+            else:
+                native().adam_step_dev_(p, g, m, v, self._dev_state,
+                                        self.lr, b1, b2, self.eps)
             return
         self._step_count += 1
         bc1 = 1 - b1 ** self._step_count
@@ -216,7 +217,7 @@ class FusedAdam:
                 live[0].betas[0], live[0].betas[1], live[0].eps,
                 [o.bf16_mirror if o.bf16_mirror is not None
                  else torch.Tensor() for o in live],
-                rng_bump, live[0]._ticket_ws)
+                rng_bump)
             return
         for o in live:
             o.step()
