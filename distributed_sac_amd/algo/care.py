@@ -984,7 +984,9 @@ class CAREEngine(SACEngine):
         info = self._se_local
         nl_c = len(self._twin_local_bf16)
         nl_a = len(self._actor_ws_bf16)
-        self.critic_optimizer.step()   # adam kernel refreshes flat mirror
+        st["prolog"] = self._adam_prolog_all()
+        # adam kernel refreshes the flat bf16 mirror
+        self.critic_optimizer.step(pre_prologed=st["prolog"])
         self._refresh_mixT("critic")   # transposed mixture views still need it
 
         # ---- actor/alpha loss + manual backward -----------------------
@@ -1077,12 +1079,17 @@ class CAREEngine(SACEngine):
         from ..ops.flat import FusedAdam as _FA
         st = self._dp_st
         _FA.step_many([self.actor_optimizer, self.log_alpha_optimizer],
-                      rng_bump=(self._rng_ctr if self._use_krng else None))
+                      rng_bump=(None if st.get("prolog")
+                                else (self._rng_ctr if self._use_krng
+                                      else None)),
+                      pre_prologed=bool(st.get("prolog")))
         # self.alpha refreshed lazily outside the graph (see SACEngine)
         self._polyak_targets(mirror=self._target_bf16)
         self._refresh_mixT("target")
         if st["orig"]:
-            self.context_encoder_optimizer.step()   # refreshes ctx mirror
+            # refreshes ctx mirror; prolog already ran in seg2
+            self.context_encoder_optimizer.step(
+                pre_prologed=bool(st.get("prolog")))
         self.tie_actor_state_encoder()
         closs, al = st["closs"], st["al"]
         return {"critic_loss": closs[6],  # summed in-kernel

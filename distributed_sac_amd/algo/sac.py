@@ -598,6 +598,27 @@ class SACEngine:
                 and getattr(self, "_rng_ctr", None) is not None
                 and _os.environ.get("DSAC_KRNG", "1") == "1")
 
+    def _adam_prolog_all(self) -> bool:
+        """ONE launch advances every optimizer's Adam state (+ RNG bump)
+        at the top of seg2; the individual steppers then skip their own
+        prolog launches.  Returns False (fall back to per-step prologs)
+        if any optimizer lacks device state or betas differ."""
+        from ..ops import native
+        opts = [self.critic_optimizer, self.actor_optimizer,
+                self.log_alpha_optimizer]
+        ctx = getattr(self, "context_encoder_optimizer", None)
+        if ctx is not None:
+            opts.append(ctx)
+        if any(o._dev_state is None for o in opts):
+            return False
+        b1, b2 = opts[0].betas
+        if any(o.betas != (b1, b2) or o.eps != opts[0].eps for o in opts):
+            return False
+        native().adam_prolog_many(
+            [o._dev_state for o in opts], [o.lr for o in opts], b1, b2,
+            self._rng_ctr if self._use_krng else None)
+        return True
+
     @torch.no_grad()
     def _chain_fwd(self, x1, x2, ws_bf16, bs_f32, G, act_last=0,
                    out_f32=True, rowcat=False, save_acts=True, wps=None):
@@ -740,7 +761,9 @@ class SACEngine:
         from ..ops import native
         ext = native()
         st = self._dp_st
-        self.critic_optimizer.step()   # adam kernel also refreshes mirror
+        st["prolog"] = self._adam_prolog_all()
+        # adam kernel also refreshes the bf16 mirror
+        self.critic_optimizer.step(pre_prologed=st["prolog"])
 
         states, sa, lp = st["states"], st["sa"], st["lp"]
         B = st["B"]
@@ -841,8 +864,10 @@ class SACEngine:
         st = self._dp_st
         FusedAdam.step_many([self.actor_optimizer,
                              self.log_alpha_optimizer],
-                            rng_bump=(self._rng_ctr if self._use_krng
-                                      else None))
+                            rng_bump=(None if st.get("prolog")
+                                      else (self._rng_ctr if self._use_krng
+                                            else None)),
+                            pre_prologed=bool(st.get("prolog")))
         # NOTE: self.alpha is refreshed lazily (checkpoint/_per_sample_alpha
         # recompute from log_alpha) — an exp() here would replay as a
         # ~5 µs kernel every captured step just for bookkeeping

@@ -1036,6 +1036,27 @@ struct AdamGroup {
   float* p; const float* g; float* m; float* v; const float* st; long n;
 };
 
+// one launch advances EVERY optimizer's Adam state for this update
+// (critic + actor + alpha [+ context]) and bumps the device RNG counter:
+// the engine calls it once at the top of seg2, and the per-optimizer
+// steppers skip their own prologs (skip_prolog)
+__global__ void k_adam_prolog_many(long long* rng,
+                                   float* s0, float* s1, float* s2,
+                                   float* s3,
+                                   float lr0, float lr1, float lr2,
+                                   float lr3, float b1, float b2) {
+  if (rng != nullptr && threadIdx.x == 0) rng[0] += 1;
+  const int i = threadIdx.x;
+  float* sv[4] = {s0, s1, s2, s3};
+  const float lrv[4] = {lr0, lr1, lr2, lr3};
+  if (i >= 4 || sv[i] == nullptr) return;
+  float* st = sv[i];
+  const float step = st[0] + 1.f;
+  st[0] = step;
+  st[1] = lrv[i] / (1.f - __powf(b1, step));
+  st[2] = 1.f / sqrtf(1.f - __powf(b2, step));
+}
+
 __global__ void k_adam_prolog3(long long* rng, float* s0, float* s1,
                                float* s2,
                                float lr0, float lr1, float lr2,
@@ -1538,7 +1559,8 @@ static void adam_step_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                            torch::Tensor v, torch::Tensor state, double lr,
                            double b1, double b2, double eps,
-                           c10::optional<torch::Tensor> mir_opt) {
+                           c10::optional<torch::Tensor> mir_opt,
+                           long skip_prolog = 0) {
   CHECK_IN(p); CHECK_IN(state);
   TORCH_CHECK(state.numel() >= 3, "state = {step, step_size, inv_sqrt_bc2}");
   const long n = p.numel();
@@ -1548,14 +1570,35 @@ static void adam_step_dev_(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                 && mir_opt->scalar_type() == torch::kBFloat16);
     mp = (unsigned short*)mir_opt->data_ptr();
   }
-  hipLaunchKernelGGL(k_adam_prolog, dim3(1), dim3(1), 0, cur_stream(),
-                     state.data_ptr<float>(), (float)lr, (float)b1,
-                     (float)b2);
+  if (!skip_prolog)
+    hipLaunchKernelGGL(k_adam_prolog, dim3(1), dim3(1), 0, cur_stream(),
+                       state.data_ptr<float>(), (float)lr, (float)b1,
+                       (float)b2);
   hipLaunchKernelGGL(k_adam_dev, dim3((n + 255) / 256), dim3(256), 0,
                      cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      state.data_ptr<float>(), n, (float)b1, (float)b2,
                      (float)eps, mp);
+}
+
+static void adam_prolog_many(std::vector<torch::Tensor> states,
+                             std::vector<double> lrs, double b1, double b2,
+                             c10::optional<torch::Tensor> rng) {
+  const size_t G = states.size();
+  TORCH_CHECK(G >= 1 && G <= 4 && lrs.size() == G, "1..4 states");
+  float* S[4] = {nullptr, nullptr, nullptr, nullptr};
+  double L[4] = {0, 0, 0, 0};
+  for (size_t g = 0; g < G; ++g) {
+    CHECK_IN(states[g]);
+    S[g] = states[g].data_ptr<float>();
+    L[g] = lrs[g];
+  }
+  long long* rng_p = (rng.has_value() && rng->numel() > 0)
+      ? (long long*)rng->data_ptr<long>() : nullptr;
+  hipLaunchKernelGGL(k_adam_prolog_many, dim3(1), dim3(4), 0, cur_stream(),
+                     rng_p, S[0], S[1], S[2], S[3], (float)L[0],
+                     (float)L[1], (float)L[2], (float)L[3], (float)b1,
+                     (float)b2);
 }
 
 static void adam_step_multi_(std::vector<torch::Tensor> ps,
@@ -1565,7 +1608,8 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
                              std::vector<torch::Tensor> states,
                              std::vector<double> lrs, double b1, double b2,
                              double eps, std::vector<torch::Tensor> mirs,
-                             c10::optional<torch::Tensor> rng = c10::nullopt) {
+                             c10::optional<torch::Tensor> rng = c10::nullopt,
+                             long skip_prolog = 0) {
   const size_t G = ps.size();
   TORCH_CHECK(G >= 1 && G <= 3, "1..3 groups");
   float* P[3] = {nullptr, nullptr, nullptr};
@@ -1595,9 +1639,10 @@ static void adam_step_multi_(std::vector<torch::Tensor> ps,
   }
   long long* rng_p = (rng.has_value() && rng->numel() > 0)
       ? (long long*)rng->data_ptr<long>() : nullptr;
-  hipLaunchKernelGGL(k_adam_prolog3, dim3(1), dim3(3), 0, cur_stream(),
-                     rng_p, St[0], St[1], St[2], (float)LR[0], (float)LR[1],
-                     (float)LR[2], (float)b1, (float)b2);
+  if (!skip_prolog)
+    hipLaunchKernelGGL(k_adam_prolog3, dim3(1), dim3(3), 0, cur_stream(),
+                       rng_p, St[0], St[1], St[2], (float)LR[0],
+                       (float)LR[1], (float)LR[2], (float)b1, (float)b2);
   hipLaunchKernelGGL(k_adam_multi, dim3((total + 255) / 256), dim3(256), 0,
                      cur_stream(),
                      P[0], Gr[0], M[0], V[0], St[0], N[0],
@@ -1673,12 +1718,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
           pybind11::arg("p"), pybind11::arg("g"), pybind11::arg("m"),
           pybind11::arg("v"), pybind11::arg("state"), pybind11::arg("lr"),
           pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
-          pybind11::arg("mir") = pybind11::none());
+          pybind11::arg("mir") = pybind11::none(),
+          pybind11::arg("skip_prolog") = 0);
   mod.def("adam_step_multi_", &adam_step_multi_,
           pybind11::arg("ps"), pybind11::arg("gs"), pybind11::arg("ms"),
           pybind11::arg("vs"), pybind11::arg("states"), pybind11::arg("lrs"),
           pybind11::arg("b1"), pybind11::arg("b2"), pybind11::arg("eps"),
           pybind11::arg("mirs") = std::vector<torch::Tensor>(),
+          pybind11::arg("rng") = pybind11::none(),
+          pybind11::arg("skip_prolog") = 0);
+  mod.def("adam_prolog_many", &adam_prolog_many,
+          pybind11::arg("states"), pybind11::arg("lrs"),
+          pybind11::arg("b1"), pybind11::arg("b2"),
           pybind11::arg("rng") = pybind11::none());
   mod.def("polyak_", &polyak_,
           pybind11::arg("t"), pybind11::arg("s"), pybind11::arg("tau"),

@@ -172,18 +172,15 @@ class FusedAdam:
         self.group.zero_grad()
 
     @torch.no_grad()
-    def step(self) -> None:
+    def step(self, pre_prologed: bool = False) -> None:
         p, g = self.group.flat_data, self.group.flat_grad
         m, v = self.exp_avg, self.exp_avg_sq
         b1, b2 = self.betas
         if self._dev_state is not None:
-            if self.bf16_mirror is not None:
-                native().adam_step_dev_(p, g, m, v, self._dev_state,
-                                        self.lr, b1, b2, self.eps,
-                                        self.bf16_mirror)
-            else:
-                native().adam_step_dev_(p, g, m, v, self._dev_state,
-                                        self.lr, b1, b2, self.eps)
+            native().adam_step_dev_(p, g, m, v, self._dev_state,
+                                    self.lr, b1, b2, self.eps,
+                                    self.bf16_mirror,
+                                    1 if pre_prologed else 0)
             return
         self._step_count += 1
         bc1 = 1 - b1 ** self._step_count
@@ -195,7 +192,8 @@ class FusedAdam:
 
     @staticmethod
     @torch.no_grad()
-    def step_many(opts: List["FusedAdam"], rng_bump=None) -> None:
+    def step_many(opts: List["FusedAdam"], rng_bump=None,
+                  pre_prologed: bool = False) -> None:
         """Step up to 3 optimizers in ONE fused kernel launch (same betas/
         eps; device-resident step state required).  rng_bump: optional
         int64 counter the prolog kernel increments once — the per-update
@@ -217,7 +215,7 @@ class FusedAdam:
                 live[0].betas[0], live[0].betas[1], live[0].eps,
                 [o.bf16_mirror if o.bf16_mirror is not None
                  else torch.Tensor() for o in live],
-                rng_bump)
+                rng_bump, 1 if pre_prologed else 0)
             return
         for o in live:
             o.step()
