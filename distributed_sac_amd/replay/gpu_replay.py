@@ -81,16 +81,24 @@ class ReplayShard:
             rewards = rewards[-self.capacity:]
             next_states, dones = next_states[-self.capacity:], dones[-self.capacity:]
             n = self.capacity
-        idx = torch.arange(self.write_ptr, self.write_ptr + n,
-                           device=self.device) % self.capacity
-        self.states.index_copy_(0, idx, states)
-        self.actions.index_copy_(0, idx, actions)
-        self.rewards.index_copy_(0, idx, rewards.reshape(n, 1))
-        self.next_states.index_copy_(0, idx, next_states)
-        self.dones.index_copy_(0, idx, dones.reshape(n, 1))
-        self.write_ptr = (self.write_ptr + n) % self.capacity
-        self.size = min(self.size + n, self.capacity)
-        self.size_dev.fill_(float(self.size))
+        # blocks are contiguous, so ingestion is plain slice copies —
+        # straight (async DMA) H2D when the source is pinned host memory,
+        # no arange/index_copy kernels.  Sources may be host OR device.
+        pairs = ((self.states, states), (self.actions, actions),
+                 (self.rewards, rewards.reshape(n, 1)),
+                 (self.next_states, next_states),
+                 (self.dones, dones.reshape(n, 1)))
+        wp = self.write_ptr
+        head = min(n, self.capacity - wp)
+        for dst, src in pairs:
+            dst[wp:wp + head].copy_(src[:head], non_blocking=True)
+            if head < n:   # wrap: the tail goes to the front
+                dst[: n - head].copy_(src[head:], non_blocking=True)
+        self.write_ptr = (wp + n) % self.capacity
+        new_size = min(self.size + n, self.capacity)
+        if new_size != self.size:   # no kernel once the shard is full
+            self.size = new_size
+            self.size_dev.fill_(float(new_size))
 
     @torch.no_grad()
     def sample_indices(self, n: int,

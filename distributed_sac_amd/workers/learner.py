@@ -155,12 +155,10 @@ class Learner:
                 if not out:
                     break
                 task = int(out[0].item())
-                self.replay.shards[task].append(
-                    out[1].to(self.device, non_blocking=True),
-                    out[2].to(self.device, non_blocking=True),
-                    out[3].to(self.device, non_blocking=True),
-                    out[4].to(self.device, non_blocking=True),
-                    out[5].to(self.device, non_blocking=True))
+                # pinned host tensors go straight into the shard's slice
+                # copies (append does the async H2D DMA itself)
+                self.replay.shards[task].append(out[1], out[2], out[3],
+                                                out[4], out[5])
                 got = out[1].shape[0]
                 self.engine.total_step += got
                 self.ingest_count += got
