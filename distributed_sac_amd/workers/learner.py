@@ -85,8 +85,11 @@ class Learner:
         self.update_timer = StepTimer()
         self.ingest_count = 0
         # per-phase wall-time accounting (SURVEY §5.1 — absent in reference)
+        # update/publish are sub-spans of loop (timed inside train_step);
+        # other = heartbeat checks; loop - update - publish = python glue
         self.phase_seconds = {"drain": 0.0, "logs": 0.0, "update": 0.0,
-                              "sync": 0.0, "publish": 0.0, "other": 0.0}
+                              "sync": 0.0, "publish": 0.0, "other": 0.0,
+                              "loop": 0.0}
         self._pub_pinned = None
         # publish throttle: the learner updates orders of magnitude faster
         # than players poll; the reference's version-gated pull (apply only
@@ -348,7 +351,10 @@ class Learner:
             p2 = time.perf_counter()
             self.phase_seconds["logs"] += p2 - p1
             self.check_heartbeats()
+            p3 = time.perf_counter()
+            self.phase_seconds["other"] += p3 - p2
             self.train_step()
+            self.phase_seconds["loop"] += time.perf_counter() - p3
             if time.perf_counter() - last_report > 10.0:
                 last_report = time.perf_counter()
                 self.logger.print(
