@@ -72,6 +72,10 @@ def parse_args():
     p.add_argument("--async-seconds", type=float, default=30.0)
     p.add_argument("--players", type=int, default=4)
     p.add_argument("--chunk-steps", type=int, default=64)
+    p.add_argument("--graph-chunk", type=int, default=4,
+                   help="updates captured per hipGraph replay (single-GPU "
+                        "graph mode; auto-disabled unless it divides "
+                        "--steps)")
     p.add_argument("--force-ddp", action="store_true",
                    help="initialize the process group + run the DP code "
                         "path even at world_size 1 (single-GPU RCCL "
@@ -153,10 +157,14 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
     # aborts the process via the NCCL watchdog, not a catchable error) —
     # DP ranks use the SEGMENTED capture (three hipGraphs with the two
     # eager all-reduces between replays, engine.capture_dp).
+    # chunked capture (c updates per hipGraph replay) amortizes the
+    # ~0.22 ms host-side hipGraphLaunch; used only when it divides the
+    # timed step count exactly so a window is EXACTLY args.steps updates
+    chunk = args.graph_chunk if args.steps % args.graph_chunk == 0 else 1
     if device.startswith("cuda") and not args.no_graph:
         if ddp is None:
             try:
-                engine.capture(replay, cfg.batch_size)
+                engine.capture(replay, cfg.batch_size, chunk=chunk)
                 graphed = True
             except Exception as e:  # pragma: no cover
                 print(f"[bench] hipGraph capture failed, eager fallback: "
@@ -169,7 +177,7 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
                 print(f"[bench] segmented DP capture failed, eager "
                       f"fallback: {e!r}", file=sys.stderr)
 
-    def one_step():
+    def one_step():   # one GRAD UPDATE (chunked replay runs chunk of them)
         if graphed:
             engine.graphed_update()
         elif dp_graphed:
@@ -179,7 +187,8 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
                                                 graph_safe=True))
             engine.update_iteration += 1
 
-    for _ in range(args.warmup):
+    n_warm = (args.warmup + chunk - 1) // chunk if graphed else args.warmup
+    for _ in range(n_warm):
         one_step()
 
     if rank == 0 and (graphed or dp_graphed):
@@ -192,7 +201,7 @@ def measure(engine, replay, cfg, args, device, ddp, rank,
             ddp.barrier()
         sync()
         t0 = time.perf_counter()
-        for _ in range(args.steps):
+        for _ in range(args.steps // chunk if graphed else args.steps):
             one_step()
         sync()
         elapsed = time.perf_counter() - t0
@@ -367,6 +376,10 @@ def main():
                 # captured segments with eager RCCL all-reduces between
                 "hipgraph": graph_mode != "eager",
                 "launch_mode": graph_mode,
+                "graph_chunk": (args.graph_chunk
+                                if graph_mode == "graph"
+                                and args.steps % args.graph_chunk == 0
+                                else 1),
             },
         }
         print(json.dumps(result))

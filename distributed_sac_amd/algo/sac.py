@@ -982,10 +982,20 @@ class SACEngine:
     # three fused Adam steps, Polyak) replays as one captured graph —
     # per-step host cost collapses to one hipGraphLaunch.
     # ------------------------------------------------------------------
-    def capture(self, replay, batch_size: int, warmup_iters: int = 3):
+    def capture(self, replay, batch_size: int, warmup_iters: int = 3,
+                chunk: int = 1):
+        """Capture ``chunk`` FULL updates in ONE hipGraph.  hipGraphLaunch
+        costs ~9 µs/node host-side (~0.22 ms for the ~25-node update), so
+        at chunk=1 the host launch is nearly at parity with the 0.31 ms
+        GPU time; chunking amortizes it (tools/probe_graph_k.py).  Every
+        captured update is complete and distinct: the device-side RNG
+        counter and Adam states advance INSIDE the graph, so update i of
+        a chunk samples different replay indices and different eps.  The
+        metrics dict reflects the chunk's LAST update."""
         if self._use_krng and hasattr(replay, "attach_rng"):
             replay.attach_rng(self._rng_ctr)
         assert self.device.type == "cuda", "capture needs a GPU"
+        self._graph_chunk = max(1, int(chunk))
         torch.cuda.synchronize(self.device)
         side = torch.cuda.Stream(self.device)
         side.wait_stream(torch.cuda.current_stream(self.device))
@@ -996,15 +1006,17 @@ class SACEngine:
         torch.cuda.synchronize(self.device)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):
-            self._graph_metrics = self.update_tensors(
-                replay.sample(batch_size, graph_safe=True))
+            for _ in range(self._graph_chunk):
+                self._graph_metrics = self.update_tensors(
+                    replay.sample(batch_size, graph_safe=True))
         self._graph = graph
         return graph
 
     def graphed_update(self) -> Dict[str, torch.Tensor]:
-        """Replay the captured update (tensor metrics refresh in place)."""
+        """Replay the captured chunk (tensor metrics refresh in place);
+        advances ``update_iteration`` by the captured chunk size."""
         self._graph.replay()
-        self.update_iteration += 1
+        self.update_iteration += getattr(self, "_graph_chunk", 1)
         return self._graph_metrics
 
     # ------------------------------------------------------------------
