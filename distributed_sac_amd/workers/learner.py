@@ -19,6 +19,8 @@ Reference semantics kept:
 
 from __future__ import annotations
 
+import os
+
 import queue as pyqueue
 import time
 from typing import Dict, Optional
@@ -97,11 +99,13 @@ class Learner:
         # when update_iteration changed) makes intermediate publishes
         # unobservable, so we publish at most every publish_interval_s
         # (docs/DESIGN_NOTES.md) instead of syncing D2H every update.
-        self.publish_interval_s = publish_interval_s
+        self.publish_interval_s = float(
+            os.environ.get("DSAC_PUBLISH_S", publish_interval_s))
         # updates captured per hipGraph (amortizes the ~0.22 ms host
         # launch; every update in the chunk is complete and distinct)
         self.graph_chunk = max(1, int(graph_chunk))
-        self._next_metric_tick = 100
+        self._metric_every = int(os.environ.get("DSAC_METRICS_EVERY", 100))
+        self._next_metric_tick = self._metric_every
         self._last_publish = 0.0
         self._pub_stream = None
         self._pub_event = None
@@ -316,7 +320,7 @@ class Learner:
                 save_checkpoint(self.engine, self.save_dir,
                                 update_iteration=self.iteration_counter)
             if self.grad_steps >= self._next_metric_tick:
-                self._next_metric_tick = self.grad_steps + 100
+                self._next_metric_tick = self.grad_steps + self._metric_every
                 metrics = {k: float(v) for k, v in metrics_t.items()}
                 self.logger.add_scalars("learner", metrics, self.grad_steps)
                 # per-task temperature curve (reference Logger writes the
