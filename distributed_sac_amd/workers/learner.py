@@ -105,6 +105,9 @@ class Learner:
         # launch; every update in the chunk is complete and distinct)
         self.graph_chunk = max(1, int(graph_chunk))
         self._metric_every = int(os.environ.get("DSAC_METRICS_EVERY", 100))
+        self._met_pinned = None
+        self._met_event = None
+        self._met_inflight = False
         self._next_metric_tick = self._metric_every
         self._last_publish = 0.0
         self._pub_stream = None
@@ -227,22 +230,55 @@ class Learner:
             self.snapshot.publish(self._pub_pinned, self._pub_iteration)
             self._pub_iteration = 0
         # phase 2: start a fresh async copy (skip if one is in flight).
-        # Tear-safety: flat_data is snapshotted by a D2D copy into a
-        # device staging buffer first, and the MAIN stream waits on that
-        # tiny copy (µs) before it may enqueue the next Adam step that
-        # would overwrite flat_data; the slow D2H then reads only the
-        # immutable staging buffer, so the pinned snapshot can never mix
-        # parameters from two adjacent updates (ADVICE.md round-1 finding).
+        # Tear-safety: the D2D snapshot of flat_data is enqueued ON THE
+        # MAIN stream, so it is ordered before any later-enqueued Adam
+        # step that would overwrite flat_data — no main-stream wait_event
+        # needed (the earlier stage-on-side-stream + main.wait design put
+        # a full pipeline barrier on main at every publish tick, which
+        # cost ~35% of the async rate).  The slow D2H then reads only the
+        # immutable staging buffer on the side stream.
         if self._pub_iteration == 0:
             main = torch.cuda.current_stream(self.device)
-            self._pub_stream.wait_stream(main)
+            self._pub_stage.copy_(flat.reshape(-1), non_blocking=True)
+            self._pub_stage_event.record(main)
+            self._pub_stream.wait_event(self._pub_stage_event)
             with torch.cuda.stream(self._pub_stream):
-                self._pub_stage.copy_(flat.reshape(-1), non_blocking=True)
-                self._pub_stage_event.record(self._pub_stream)
                 self._pub_pinned.copy_(self._pub_stage, non_blocking=True)
             self._pub_event.record(self._pub_stream)
-            main.wait_event(self._pub_stage_event)
             self._pub_iteration = self.iteration_counter
+
+    def _metrics_tick(self, metrics_t) -> Optional[Dict[str, float]]:
+        """Pipelined metric readback: D2H into pinned staging now, log
+        the values once the copy's event has completed (next tick) —
+        metrics are one tick stale but the learner never syncs."""
+        if self._met_pinned is None:
+            self._met_pinned = {k: torch.empty_like(v.detach(),
+                                                    device="cpu",
+                                                    pin_memory=True)
+                                for k, v in metrics_t.items()}
+            la = self.engine.log_alpha.detach()
+            self._met_pinned["_log_alpha"] = torch.empty_like(
+                la, device="cpu", pin_memory=True)
+            self._met_event = torch.cuda.Event()
+            self._met_inflight = False
+        out = None
+        if self._met_inflight and self._met_event.query():
+            out = {k: float(v) for k, v in self._met_pinned.items()
+                   if not k.startswith("_")}
+            la = self._met_pinned["_log_alpha"]
+            if la.numel() > 1:
+                self.logger.add_scalars(
+                    "alpha",
+                    {f"task_{i}": float(v)
+                     for i, v in enumerate(la.exp().tolist())},
+                    self.grad_steps)
+        for k, v in metrics_t.items():
+            self._met_pinned[k].copy_(v.detach(), non_blocking=True)
+        self._met_pinned["_log_alpha"].copy_(
+            self.engine.log_alpha.detach(), non_blocking=True)
+        self._met_event.record()
+        self._met_inflight = True
+        return out
 
     def ready(self) -> bool:
         """Reference start gate: train only once the MIN per-task shard
@@ -321,20 +357,30 @@ class Learner:
                                 update_iteration=self.iteration_counter)
             if self.grad_steps >= self._next_metric_tick:
                 self._next_metric_tick = self.grad_steps + self._metric_every
-                metrics = {k: float(v) for k, v in metrics_t.items()}
-                self.logger.add_scalars("learner", metrics, self.grad_steps)
+                if self.device.type == "cuda":
+                    # non-blocking tick: enqueue a D2H of the stat buffers
+                    # into pinned staging and log the PREVIOUS tick's
+                    # completed snapshot — float() here would sync the
+                    # whole queued pipeline (~24% of the async rate)
+                    metrics = self._metrics_tick(metrics_t)
+                else:
+                    metrics = {k: float(v) for k, v in metrics_t.items()}
+                if metrics:
+                    self.logger.add_scalars("learner", metrics,
+                                            self.grad_steps)
                 # per-task temperature curve (reference Logger writes the
                 # full alpha array per update — MT10_Distributed_CARE/src/
                 # logger.py:45-132, learner.py:445-464); one D2H of a
                 # <=num_tasks vector per report tick
-                la = self.engine.log_alpha.detach()
-                if la.numel() > 1:
-                    alphas = la.exp().cpu().tolist()
-                    self.logger.add_scalars(
-                        "alpha",
-                        {f"task_{i}": float(v)
-                         for i, v in enumerate(alphas)},
-                        self.grad_steps)
+                if metrics and self.device.type != "cuda":
+                    la = self.engine.log_alpha.detach()
+                    if la.numel() > 1:
+                        alphas = la.exp().cpu().tolist()
+                        self.logger.add_scalars(
+                            "alpha",
+                            {f"task_{i}": float(v)
+                             for i, v in enumerate(alphas)},
+                            self.grad_steps)
         return metrics
 
     # -- main loop -----------------------------------------------------
