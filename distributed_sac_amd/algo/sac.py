@@ -61,7 +61,6 @@ class SACEngine:
         self._graph = None
         self._eps_queue: Optional[list] = None  # test hook: deterministic eps
         self.ddp = None  # optional DataParallelGroup (set via attach_ddp)
-        self._side_stream = None
         self._build_models()
         self._build_optimizers()
 

@@ -377,8 +377,9 @@ __global__ __launch_bounds__(256) void k_reduce_partials(
 // ---------------------------------------------------------------------------
 // counter-based device RNG (splitmix64 hash of (counter, index, salt)):
 // every RNG-consuming kernel derives its noise from one persistent int64
-// counter that k_adam_prolog3 bumps ONCE per update (single-block kernel,
-// so the bump is race-free by stream ordering).  Replaces the per-update
+// counter that k_adam_prolog_many (or prolog3) bumps ONCE per update
+// (single-block kernels, so the bump is race-free by stream ordering).
+// Replaces the per-update
 // torch rand/randn launches AND the hipGraph RNG-offset bookkeeping
 // kernels torch inserts around them.
 __device__ __forceinline__ unsigned long long dsac_sm64(
